@@ -1,0 +1,40 @@
+"""Classifiers: LogisticRegression, RandomForest, GBT, MultilayerPerceptron.
+
+The four classifier families of the reference app (BASELINE.json:5,10;
+SURVEY.md §2.2), re-implemented MI355X-first: torch-ROCm tensors + hand
+written CDNA4 HIP kernels on the hot paths, RCCL all-reduce for DP training.
+"""
+
+from flowml.classification.base import (
+    Classifier,
+    ClassificationModel,
+    ProbabilisticClassifier,
+    ProbabilisticClassificationModel,
+)
+from flowml.classification.logistic import LogisticRegression, LogisticRegressionModel
+from flowml.classification.mlp import (
+    MultilayerPerceptronClassifier,
+    MultilayerPerceptronClassificationModel,
+)
+from flowml.classification.forest import (
+    RandomForestClassifier,
+    RandomForestClassificationModel,
+    DecisionTreeClassificationModel,
+)
+from flowml.classification.gbt import GBTClassifier, GBTClassificationModel
+
+__all__ = [
+    "Classifier",
+    "ClassificationModel",
+    "ProbabilisticClassifier",
+    "ProbabilisticClassificationModel",
+    "LogisticRegression",
+    "LogisticRegressionModel",
+    "MultilayerPerceptronClassifier",
+    "MultilayerPerceptronClassificationModel",
+    "RandomForestClassifier",
+    "RandomForestClassificationModel",
+    "DecisionTreeClassificationModel",
+    "GBTClassifier",
+    "GBTClassificationModel",
+]

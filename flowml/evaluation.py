@@ -1,0 +1,162 @@
+"""MulticlassClassificationEvaluator.
+
+Behavioral spec: ``spark:.../ml/evaluation/MulticlassClassificationEvaluator.scala``
+(SURVEY.md §2.2): default ``metricName="f1"`` (the *weighted* F1), computed
+from the (prediction, label) confusion matrix. ``macroF1`` is a flowml
+extension — it is the task's headline metric (BASELINE.json:2).
+
+Confusion counts are accumulated with a single ``bincount`` on-device, so the
+evaluator runs on GPU tensors without a host round-trip.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from flowml.frame import FlowFrame
+from flowml.param import (
+    HasLabelCol,
+    HasPredictionCol,
+    HasWeightCol,
+    Param,
+    Params,
+    TypeConverters,
+)
+
+_SUPPORTED = {
+    "f1",
+    "accuracy",
+    "weightedPrecision",
+    "weightedRecall",
+    "weightedFMeasure",
+    "truePositiveRateByLabel",
+    "falsePositiveRateByLabel",
+    "precisionByLabel",
+    "recallByLabel",
+    "fMeasureByLabel",
+    "macroF1",  # flowml extension (headline metric)
+}
+
+
+class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeightCol, Params):
+    metricName: Param[str] = Param(
+        "undefined",
+        "metricName",
+        f"metric name in evaluation {sorted(_SUPPORTED)}",
+        TypeConverters.toString,
+    )
+    metricLabel: Param[float] = Param(
+        "undefined",
+        "metricLabel",
+        "The class whose metric will be computed in byLabel metrics.",
+        TypeConverters.toFloat,
+    )
+    beta: Param[float] = Param(
+        "undefined", "beta", "beta value in weightedFMeasure|fMeasureByLabel", TypeConverters.toFloat
+    )
+
+    def __init__(
+        self,
+        *,
+        predictionCol: Optional[str] = None,
+        labelCol: Optional[str] = None,
+        metricName: Optional[str] = None,
+        metricLabel: Optional[float] = None,
+        beta: Optional[float] = None,
+        weightCol: Optional[str] = None,
+    ):
+        super().__init__()
+        self._setDefault(
+            predictionCol="prediction",
+            labelCol="label",
+            metricName="f1",
+            metricLabel=0.0,
+            beta=1.0,
+        )
+        self._set(
+            predictionCol=predictionCol,
+            labelCol=labelCol,
+            metricName=metricName,
+            metricLabel=metricLabel,
+            beta=beta,
+            weightCol=weightCol,
+        )
+
+    def getMetricName(self) -> str:
+        return self.getOrDefault(self.metricName)
+
+    def setMetricName(self, value: str):
+        if value not in _SUPPORTED:
+            raise ValueError(f"Unsupported metricName {value!r}")
+        return self.set(self.metricName, value)
+
+    def isLargerBetter(self) -> bool:
+        return True
+
+    # -- core ----------------------------------------------------------------
+
+    def _confusion(self, dataset: FlowFrame) -> torch.Tensor:
+        pred = dataset.col(self.getOrDefault(self.predictionCol))
+        label = dataset.col(self.getOrDefault(self.labelCol))
+        if not isinstance(pred, torch.Tensor) or not isinstance(label, torch.Tensor):
+            raise TypeError("prediction/label columns must be numeric")
+        p = pred.reshape(-1).to(torch.int64)
+        y = label.reshape(-1).to(torch.int64)
+        k = int(torch.maximum(p.max(), y.max()).item()) + 1 if len(p) else 1
+        if self.isSet(self.weightCol):
+            w = dataset.col(self.getWeightCol()).reshape(-1).to(torch.float64)
+        else:
+            w = torch.ones_like(p, dtype=torch.float64)
+        # confusion[label, pred]
+        conf = torch.bincount(y * k + p, weights=w, minlength=k * k).reshape(k, k)
+        return conf
+
+    def evaluate(self, dataset: FlowFrame) -> float:
+        conf = self._confusion(dataset)
+        name = self.getMetricName()
+        if name not in _SUPPORTED:
+            raise ValueError(f"Unsupported metricName {name!r}")
+        total = conf.sum()
+        tp = conf.diagonal()
+        per_label_actual = conf.sum(dim=1)  # row = true label
+        per_label_pred = conf.sum(dim=0)  # col = prediction
+        eps = torch.zeros(())
+        precision = torch.where(per_label_pred > 0, tp / per_label_pred, tp.new_zeros(tp.shape))
+        recall = torch.where(per_label_actual > 0, tp / per_label_actual, tp.new_zeros(tp.shape))
+        beta = self.getOrDefault(self.beta)
+        b2 = beta * beta
+
+        def fmeasure(p, r):
+            denom = b2 * p + r
+            return torch.where(denom > 0, (1 + b2) * p * r / denom, p.new_zeros(p.shape))
+
+        if name == "accuracy":
+            return float(tp.sum() / total)
+        if name == "f1":
+            f = fmeasure(precision, recall)
+            return float((f * per_label_actual).sum() / total)
+        if name == "weightedPrecision":
+            return float((precision * per_label_actual).sum() / total)
+        if name == "weightedRecall":
+            return float((recall * per_label_actual).sum() / total)
+        if name == "weightedFMeasure":
+            f = fmeasure(precision, recall)
+            return float((f * per_label_actual).sum() / total)
+        if name == "macroF1":
+            f = fmeasure(precision, recall)
+            present = per_label_actual > 0
+            return float(f[present].mean()) if bool(present.any()) else 0.0
+        lbl = int(self.getOrDefault(self.metricLabel))
+        if name in ("truePositiveRateByLabel", "recallByLabel"):
+            return float(recall[lbl])
+        if name == "precisionByLabel":
+            return float(precision[lbl])
+        if name == "fMeasureByLabel":
+            return float(fmeasure(precision, recall)[lbl])
+        if name == "falsePositiveRateByLabel":
+            fp = per_label_pred[lbl] - tp[lbl]
+            neg = total - per_label_actual[lbl]
+            return float(fp / neg) if float(neg) > 0 else 0.0
+        raise AssertionError(name)

@@ -1,0 +1,550 @@
+"""Featurization stages: StringIndexer, OneHotEncoder, VectorAssembler, scalers.
+
+Behavioral spec: ``pyspark.ml.feature`` (upstream ``spark:.../ml/feature/``;
+SURVEY.md §2.2 rows 5-8). Semantics that matter for parity, all replicated:
+
+* StringIndexer orders labels by **descending frequency, ties broken
+  alphabetically** by default (``stringOrderType="frequencyDesc"``);
+  ``handleInvalid`` ∈ {error (default), skip, keep} with keep mapping unseen
+  labels to index ``numLabels``.
+* OneHotEncoder drops the last category by default (``dropLast=True``) so the
+  encoding has ``numCategories - 1`` slots; ``handleInvalid="keep"`` adds one
+  extra slot for invalid/unseen.
+* VectorAssembler concatenates numeric and vector columns in declared order.
+* StandardScaler defaults to ``withMean=False, withStd=True`` (!) with the
+  sample (n-1) standard deviation; zero-variance features scale to 0.
+* MinMaxScaler maps to [min,max] with constant features at the midpoint.
+
+MI355X note: the assembled feature column is a dense 2-D tensor (row-major,
+N×D) rather than per-row sparse vectors — at NSL-KDD shape (41 features →
+~120 after one-hot) dense HBM-resident columns are strictly better for the
+downstream kernels. The user-visible per-row view (``collect``) still yields
+``flowml.linalg`` vectors.
+"""
+
+from __future__ import annotations
+
+from typing import Any, Dict, List, Optional, Sequence
+
+import numpy as np
+import torch
+
+from flowml.base import Estimator, Model, Transformer
+from flowml.frame import FlowFrame
+from flowml.param import (
+    HasHandleInvalid,
+    HasInputCol,
+    HasInputCols,
+    HasOutputCol,
+    HasOutputCols,
+    Param,
+    Params,
+    TypeConverters,
+)
+
+
+# ---------------------------------------------------------------------------
+# StringIndexer
+# ---------------------------------------------------------------------------
+
+
+def _column_as_strings(df: FlowFrame, name: str) -> np.ndarray:
+    col = df.col(name)
+    if isinstance(col, np.ndarray):
+        return col.astype(str)
+    # Spark casts numeric input columns to string before indexing
+    vals = col.detach().cpu().numpy()
+    return np.asarray([str(v) for v in vals], dtype=object).astype(str)
+
+
+def _order_labels(values: np.ndarray, counts: np.ndarray, order: str) -> List[str]:
+    if order == "frequencyDesc":
+        idx = np.lexsort((values, -counts))
+    elif order == "frequencyAsc":
+        idx = np.lexsort((values, counts))
+    elif order == "alphabetAsc":
+        idx = np.argsort(values, kind="stable")
+    elif order == "alphabetDesc":
+        idx = np.argsort(values, kind="stable")[::-1]
+    else:
+        raise ValueError(f"Unsupported stringOrderType: {order}")
+    return [str(v) for v in values[idx]]
+
+
+class _StringIndexerParams(HasInputCol, HasInputCols, HasOutputCol, HasOutputCols, HasHandleInvalid):
+    stringOrderType: Param[str] = Param(
+        "undefined",
+        "stringOrderType",
+        "How to order labels of string column. The first label after ordering is assigned an index of 0. "
+        "Supported options: frequencyDesc, frequencyAsc, alphabetDesc, alphabetAsc.",
+        TypeConverters.toString,
+    )
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(handleInvalid="error", stringOrderType="frequencyDesc")
+
+    def getStringOrderType(self) -> str:
+        return self.getOrDefault(self.stringOrderType)
+
+    def _in_out_cols(self):
+        if self.isSet(self.inputCols):
+            return list(self.getInputCols()), list(self.getOrDefault(self.outputCols))
+        return [self.getInputCol()], [self.getOutputCol()]
+
+
+class StringIndexer(_StringIndexerParams, Estimator):
+    """Label → index by frequency (SURVEY.md §2.2; spark:.../feature/StringIndexer.scala)."""
+
+    def __init__(
+        self,
+        *,
+        inputCol: Optional[str] = None,
+        outputCol: Optional[str] = None,
+        inputCols: Optional[List[str]] = None,
+        outputCols: Optional[List[str]] = None,
+        handleInvalid: Optional[str] = None,
+        stringOrderType: Optional[str] = None,
+    ):
+        super().__init__()
+        self._set(
+            inputCol=inputCol,
+            outputCol=outputCol,
+            inputCols=inputCols,
+            outputCols=outputCols,
+            handleInvalid=handleInvalid,
+            stringOrderType=stringOrderType,
+        )
+
+    def setStringOrderType(self, value: str):
+        return self.set(self.stringOrderType, value)
+
+    def _fit(self, dataset: FlowFrame) -> "StringIndexerModel":
+        in_cols, _ = self._in_out_cols()
+        order = self.getStringOrderType()
+        labelsArray: List[List[str]] = []
+        for c in in_cols:
+            vals = _column_as_strings(dataset, c)
+            uniq, counts = np.unique(vals, return_counts=True)
+            labelsArray.append(_order_labels(uniq, counts, order))
+        model = StringIndexerModel(labelsArray=labelsArray)
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class StringIndexerModel(_StringIndexerParams, Model):
+    def __init__(self, labelsArray: Optional[List[List[str]]] = None):
+        super().__init__()
+        self.labelsArray: List[List[str]] = labelsArray or []
+
+    @property
+    def labels(self) -> List[str]:
+        return self.labelsArray[0] if self.labelsArray else []
+
+    @classmethod
+    def from_labels(cls, labels: List[str], inputCol: str, outputCol: str, handleInvalid="error"):
+        m = cls(labelsArray=[list(labels)])
+        m._set(inputCol=inputCol, outputCol=outputCol, handleInvalid=handleInvalid)
+        return m
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        in_cols, out_cols = self._in_out_cols()
+        handle = self.getHandleInvalid()
+        keep_mask: Optional[np.ndarray] = None
+        new_cols: Dict[str, np.ndarray] = {}
+        for c, o, labels in zip(in_cols, out_cols, self.labelsArray):
+            vals = _column_as_strings(dataset, c)
+            lookup = {lab: float(i) for i, lab in enumerate(labels)}
+            idx = np.fromiter(
+                (lookup.get(v, -1.0) for v in vals), dtype=np.float64, count=len(vals)
+            )
+            invalid = idx < 0
+            if invalid.any():
+                if handle == "error":
+                    bad = sorted(set(np.asarray(vals)[invalid].tolist()))[:5]
+                    raise ValueError(
+                        f"StringIndexer encountered unseen labels {bad} in column {c!r} "
+                        f"(handleInvalid='error'). Use 'skip' or 'keep'."
+                    )
+                elif handle == "skip":
+                    keep_mask = ~invalid if keep_mask is None else (keep_mask & ~invalid)
+                elif handle == "keep":
+                    idx = np.where(invalid, float(len(labels)), idx)
+                else:
+                    raise ValueError(f"Unsupported handleInvalid: {handle}")
+            new_cols[o] = idx
+        out = dataset
+        for o, idx in new_cols.items():
+            out = out.withColumn(o, torch.from_numpy(idx))
+        if keep_mask is not None:
+            out = out.filter(keep_mask)
+        return out
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str):
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+# ---------------------------------------------------------------------------
+# OneHotEncoder
+# ---------------------------------------------------------------------------
+
+
+class _OneHotEncoderParams(HasInputCol, HasInputCols, HasOutputCol, HasOutputCols, HasHandleInvalid):
+    dropLast: Param[bool] = Param(
+        "undefined", "dropLast", "whether to drop the last category", TypeConverters.toBoolean
+    )
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(handleInvalid="error", dropLast=True)
+
+    def getDropLast(self) -> bool:
+        return self.getOrDefault(self.dropLast)
+
+    def _in_out_cols(self):
+        if self.isSet(self.inputCols):
+            return list(self.getInputCols()), list(self.getOrDefault(self.outputCols))
+        return [self.getInputCol()], [self.getOutputCol()]
+
+
+class OneHotEncoder(_OneHotEncoderParams, Estimator):
+    """Index → one-hot vector, dropLast=True default (spark:.../feature/OneHotEncoder.scala)."""
+
+    def __init__(
+        self,
+        *,
+        inputCol: Optional[str] = None,
+        outputCol: Optional[str] = None,
+        inputCols: Optional[List[str]] = None,
+        outputCols: Optional[List[str]] = None,
+        handleInvalid: Optional[str] = None,
+        dropLast: Optional[bool] = None,
+    ):
+        super().__init__()
+        self._set(
+            inputCol=inputCol,
+            outputCol=outputCol,
+            inputCols=inputCols,
+            outputCols=outputCols,
+            handleInvalid=handleInvalid,
+            dropLast=dropLast,
+        )
+
+    def setDropLast(self, value: bool):
+        return self.set(self.dropLast, value)
+
+    def _fit(self, dataset: FlowFrame) -> "OneHotEncoderModel":
+        in_cols, _ = self._in_out_cols()
+        sizes = []
+        for c in in_cols:
+            col = dataset.col(c)
+            if isinstance(col, np.ndarray):
+                raise TypeError(f"OneHotEncoder input {c!r} must be numeric (use StringIndexer first)")
+            mx = int(col.max().item()) if len(col) else -1
+            if len(col) and float(col.min().item()) < 0:
+                raise ValueError(f"OneHotEncoder input {c!r} has negative values")
+            sizes.append(mx + 1)
+        model = OneHotEncoderModel(categorySizes=sizes)
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class OneHotEncoderModel(_OneHotEncoderParams, Model):
+    def __init__(self, categorySizes: Optional[List[int]] = None):
+        super().__init__()
+        self.categorySizes: List[int] = categorySizes or []
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        in_cols, out_cols = self._in_out_cols()
+        handle = self.getHandleInvalid()
+        drop_last = self.getDropLast()
+        out = dataset
+        keep_mask: Optional[torch.Tensor] = None
+        for c, o, n_cats in zip(in_cols, out_cols, self.categorySizes):
+            col = dataset.col(c)
+            if isinstance(col, np.ndarray):
+                raise TypeError(f"OneHotEncoder input {c!r} must be numeric")
+            idx = col.to(torch.int64)
+            # vector width: numCats (+1 if keep) (-1 if dropLast)
+            size = n_cats + (1 if handle == "keep" else 0) - (1 if drop_last else 0)
+            invalid = (idx < 0) | (idx >= n_cats)
+            if bool(invalid.any()):
+                if handle == "error":
+                    raise ValueError(
+                        f"OneHotEncoder found values out of range [0,{n_cats}) in {c!r} "
+                        f"(handleInvalid='error')."
+                    )
+                elif handle == "keep":
+                    idx = torch.where(invalid, torch.full_like(idx, n_cats), idx)
+                elif handle == "skip":
+                    m = ~invalid
+                    keep_mask = m if keep_mask is None else (keep_mask & m)
+                    idx = torch.where(invalid, torch.zeros_like(idx), idx)  # placeholder
+                else:
+                    raise ValueError(f"Unsupported handleInvalid: {handle}")
+            enc = torch.zeros((len(idx), size), dtype=torch.float32, device=idx.device)
+            in_range = idx < size  # dropLast: last category encodes as all-zeros
+            rows = torch.nonzero(in_range, as_tuple=False).flatten()
+            if rows.numel():
+                enc[rows, idx[rows]] = 1.0
+            out = out.withColumn(o, enc)
+        if keep_mask is not None:
+            out = out.filter(keep_mask)
+        return out
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str):
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+# ---------------------------------------------------------------------------
+# VectorAssembler
+# ---------------------------------------------------------------------------
+
+
+class VectorAssembler(HasInputCols, HasOutputCol, HasHandleInvalid, Transformer):
+    """Concatenate numeric + vector columns into one feature vector column.
+
+    Spec: spark:.../feature/VectorAssembler.scala (SURVEY.md §2.2).
+    Output is a dense 2-D float32 tensor column (see module docstring).
+    handleInvalid: 'error' raises on NaN, 'skip' drops rows with NaN, 'keep'
+    passes NaN through.
+    """
+
+    def __init__(
+        self,
+        *,
+        inputCols: Optional[List[str]] = None,
+        outputCol: Optional[str] = None,
+        handleInvalid: Optional[str] = None,
+    ):
+        super().__init__()
+        self._setDefault(handleInvalid="error")
+        self._set(inputCols=inputCols, outputCol=outputCol, handleInvalid=handleInvalid)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        in_cols = self.getInputCols()
+        out_col = self.getOutputCol()
+        handle = self.getHandleInvalid()
+        parts: List[torch.Tensor] = []
+        device = None
+        for c in in_cols:
+            col = dataset.col(c)
+            if isinstance(col, np.ndarray):
+                raise TypeError(
+                    f"VectorAssembler input {c!r} is a string column; index it first"
+                )
+            t = col if col.dim() == 2 else col.reshape(-1, 1)
+            parts.append(t.to(torch.float32))
+            device = t.device
+        feats = torch.cat(parts, dim=1) if len(parts) > 1 else parts[0].clone()
+        if handle != "keep":
+            bad = torch.isnan(feats).any(dim=1)
+            if bool(bad.any()):
+                if handle == "error":
+                    raise ValueError(
+                        "VectorAssembler encountered NaN/null values (handleInvalid='error')."
+                    )
+                out = dataset.withColumn(out_col, feats)
+                return out.filter(~bad)
+        return dataset.withColumn(out_col, feats)
+
+
+# ---------------------------------------------------------------------------
+# StandardScaler
+# ---------------------------------------------------------------------------
+
+
+class _StandardScalerParams(HasInputCol, HasOutputCol):
+    withMean: Param[bool] = Param(
+        "undefined", "withMean", "Center data with mean", TypeConverters.toBoolean
+    )
+    withStd: Param[bool] = Param(
+        "undefined", "withStd", "Scale to unit standard deviation", TypeConverters.toBoolean
+    )
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(withMean=False, withStd=True)
+
+    def getWithMean(self) -> bool:
+        return self.getOrDefault(self.withMean)
+
+    def getWithStd(self) -> bool:
+        return self.getOrDefault(self.withStd)
+
+
+class StandardScaler(_StandardScalerParams, Estimator):
+    """μ/σ column scaling, withMean=False default (spark:.../feature/StandardScaler.scala)."""
+
+    def __init__(
+        self,
+        *,
+        inputCol: Optional[str] = None,
+        outputCol: Optional[str] = None,
+        withMean: Optional[bool] = None,
+        withStd: Optional[bool] = None,
+    ):
+        super().__init__()
+        self._set(inputCol=inputCol, outputCol=outputCol, withMean=withMean, withStd=withStd)
+
+    def _fit(self, dataset: FlowFrame) -> "StandardScalerModel":
+        col = dataset.col(self.getInputCol())
+        if not (isinstance(col, torch.Tensor) and col.dim() == 2):
+            raise TypeError("StandardScaler input must be a vector column")
+        x = col.to(torch.float64)
+        mean = x.mean(dim=0)
+        n = x.shape[0]
+        if n > 1:
+            std = x.std(dim=0, unbiased=True)  # Spark Summarizer: sample std
+        else:
+            std = torch.zeros_like(mean)
+        model = StandardScalerModel(mean=mean.cpu(), std=std.cpu())
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class StandardScalerModel(_StandardScalerParams, Model):
+    def __init__(self, mean: Optional[torch.Tensor] = None, std: Optional[torch.Tensor] = None):
+        super().__init__()
+        self.mean = mean
+        self.std = std
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        col = dataset.col(self.getInputCol())
+        x = col.to(torch.float32)
+        mean = self.mean.to(device=x.device, dtype=x.dtype)
+        std = self.std.to(device=x.device, dtype=x.dtype)
+        if self.getWithMean():
+            x = x - mean
+        if self.getWithStd():
+            factor = torch.where(std != 0, 1.0 / std, torch.zeros_like(std))
+            x = x * factor
+        return dataset.withColumn(self.getOutputCol(), x)
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str):
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+# ---------------------------------------------------------------------------
+# MinMaxScaler
+# ---------------------------------------------------------------------------
+
+
+class _MinMaxScalerParams(HasInputCol, HasOutputCol):
+    min: Param[float] = Param(
+        "undefined", "min", "Lower bound after transformation", TypeConverters.toFloat
+    )
+    max: Param[float] = Param(
+        "undefined", "max", "Upper bound after transformation", TypeConverters.toFloat
+    )
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(min=0.0, max=1.0)
+
+    def getMin(self) -> float:
+        return self.getOrDefault(self.min)
+
+    def getMax(self) -> float:
+        return self.getOrDefault(self.max)
+
+
+class MinMaxScaler(_MinMaxScalerParams, Estimator):
+    def __init__(
+        self,
+        *,
+        inputCol: Optional[str] = None,
+        outputCol: Optional[str] = None,
+        min: Optional[float] = None,
+        max: Optional[float] = None,
+    ):
+        super().__init__()
+        self._set(inputCol=inputCol, outputCol=outputCol, min=min, max=max)
+
+    def _fit(self, dataset: FlowFrame) -> "MinMaxScalerModel":
+        col = dataset.col(self.getInputCol())
+        if not (isinstance(col, torch.Tensor) and col.dim() == 2):
+            raise TypeError("MinMaxScaler input must be a vector column")
+        x = col.to(torch.float64)
+        model = MinMaxScalerModel(
+            originalMin=x.min(dim=0).values.cpu(), originalMax=x.max(dim=0).values.cpu()
+        )
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class MinMaxScalerModel(_MinMaxScalerParams, Model):
+    def __init__(
+        self,
+        originalMin: Optional[torch.Tensor] = None,
+        originalMax: Optional[torch.Tensor] = None,
+    ):
+        super().__init__()
+        self.originalMin = originalMin
+        self.originalMax = originalMax
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        col = dataset.col(self.getInputCol())
+        x = col.to(torch.float32)
+        omin = self.originalMin.to(device=x.device, dtype=x.dtype)
+        omax = self.originalMax.to(device=x.device, dtype=x.dtype)
+        lo, hi = self.getMin(), self.getMax()
+        rng = omax - omin
+        const = rng == 0
+        scale = torch.where(const, torch.zeros_like(rng), (hi - lo) / torch.where(const, torch.ones_like(rng), rng))
+        scaled = (x - omin) * scale + lo
+        # constant feature → midpoint (Spark behavior)
+        mid = torch.full_like(scaled, 0.5 * (hi + lo))
+        scaled = torch.where(const.expand_as(scaled), mid, scaled)
+        return dataset.withColumn(self.getOutputCol(), scaled)
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str):
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)

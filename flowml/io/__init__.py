@@ -1,0 +1,543 @@
+"""Spark ML persistence: save/load in the Spark ML directory layout.
+
+Format contract (BASELINE.json:5 "model-save format"; SURVEY.md §3.5 and
+§2.2 persistence row; upstream ``spark:.../ml/util/ReadWrite.scala``):
+
+    <path>/
+      metadata/part-00000      one-line JSON {class, timestamp, sparkVersion,
+                               uid, paramMap, defaultParamMap}
+      data/part-*.parquet      learned state (model-specific schema below)
+      stages/<idx>_<uid>/      (PipelineModel / Pipeline only) nested stage dirs
+      treesMetadata/           (ensembles) per-tree {treeID, metadata, weights}
+
+Model ``data/`` schemas:
+  StringIndexerModel       {labelsArray: list<list<string>>}
+  OneHotEncoderModel       {categorySizes: list<int>}
+  StandardScalerModel      {std: VectorUDT, mean: VectorUDT}
+  MinMaxScalerModel        {originalMin: VectorUDT, originalMax: VectorUDT}
+  LogisticRegressionModel  {numClasses, numFeatures, interceptVector: VectorUDT,
+                            coefficientMatrix: MatrixUDT, isMultinomial}
+  MLPClassificationModel   {weights: VectorUDT}   (layers live in paramMap)
+  DecisionTree/RF/GBT      NodeData rows {treeID, nodeData: {id, prediction,
+                            impurity, impurityStats[], gain, leftChild,
+                            rightChild, split: {featureIndex,
+                            leftCategoriesOrThreshold[], numCategories}}}
+
+VectorUDT / MatrixUDT are encoded as the Spark struct layouts
+(type/size/indices/values and type/numRows/numCols/colPtrs/rowIndices/values/
+isTransposed). ``class`` uses the org.apache.spark.ml.* names so the layout
+matches what a Spark reader expects; ``sparkVersion`` is "3.5.0-flowml".
+Fields whose exact upstream value could not be confirmed without a live Spark
+(SURVEY.md §7 hard-part 2) are marked in comments.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import time
+from typing import Any, Dict, List, Optional
+
+import numpy as np
+import torch
+
+SPARK_VERSION = "3.5.0-flowml"
+
+
+# ---------------------------------------------------------------------------
+# class-name mapping
+# ---------------------------------------------------------------------------
+
+_CLASS_TO_SPARK: Dict[str, str] = {}
+_SPARK_TO_CLASS: Dict[str, Any] = {}
+
+
+def _register(cls_name: str, spark_name: str):
+    _CLASS_TO_SPARK[cls_name] = spark_name
+
+
+_register("Pipeline", "org.apache.spark.ml.Pipeline")
+_register("PipelineModel", "org.apache.spark.ml.PipelineModel")
+_register("StringIndexer", "org.apache.spark.ml.feature.StringIndexer")
+_register("StringIndexerModel", "org.apache.spark.ml.feature.StringIndexerModel")
+_register("OneHotEncoder", "org.apache.spark.ml.feature.OneHotEncoder")
+_register("OneHotEncoderModel", "org.apache.spark.ml.feature.OneHotEncoderModel")
+_register("VectorAssembler", "org.apache.spark.ml.feature.VectorAssembler")
+_register("StandardScaler", "org.apache.spark.ml.feature.StandardScaler")
+_register("StandardScalerModel", "org.apache.spark.ml.feature.StandardScalerModel")
+_register("MinMaxScaler", "org.apache.spark.ml.feature.MinMaxScaler")
+_register("MinMaxScalerModel", "org.apache.spark.ml.feature.MinMaxScalerModel")
+_register("LogisticRegression", "org.apache.spark.ml.classification.LogisticRegression")
+_register("LogisticRegressionModel", "org.apache.spark.ml.classification.LogisticRegressionModel")
+_register("RandomForestClassifier", "org.apache.spark.ml.classification.RandomForestClassifier")
+_register(
+    "RandomForestClassificationModel",
+    "org.apache.spark.ml.classification.RandomForestClassificationModel",
+)
+_register("GBTClassifier", "org.apache.spark.ml.classification.GBTClassifier")
+_register("GBTClassificationModel", "org.apache.spark.ml.classification.GBTClassificationModel")
+_register(
+    "DecisionTreeClassificationModel",
+    "org.apache.spark.ml.classification.DecisionTreeClassificationModel",
+)
+_register(
+    "MultilayerPerceptronClassifier",
+    "org.apache.spark.ml.classification.MultilayerPerceptronClassifier",
+)
+_register(
+    "MultilayerPerceptronClassificationModel",
+    "org.apache.spark.ml.classification.MultilayerPerceptronClassificationModel",
+)
+
+
+def _our_class(spark_name: str):
+    import flowml.feature as feature
+    import flowml.pipeline as pipeline
+    import flowml.classification as cls
+
+    simple = spark_name.rsplit(".", 1)[-1]
+    for mod in (pipeline, feature, cls):
+        if hasattr(mod, simple):
+            return getattr(mod, simple)
+    raise ValueError(f"Cannot resolve class for {spark_name!r}")
+
+
+# ---------------------------------------------------------------------------
+# metadata read/write
+# ---------------------------------------------------------------------------
+
+
+def _json_value(v: Any) -> Any:
+    if isinstance(v, (np.integer,)):
+        return int(v)
+    if isinstance(v, (np.floating,)):
+        return float(v)
+    if isinstance(v, (list, tuple)):
+        return [_json_value(x) for x in v]
+    return v
+
+
+def save_metadata(instance, path: str, extra: Optional[Dict[str, Any]] = None) -> None:
+    os.makedirs(os.path.join(path, "metadata"), exist_ok=True)
+    cls_simple = type(instance).__name__
+    spark_class = _CLASS_TO_SPARK.get(cls_simple, f"flowml.{cls_simple}")
+    paramMap = {}
+    for p, v in instance._paramMap.items():
+        if p.name == "stages":
+            continue  # Pipeline stages serialize structurally
+        paramMap[p.name] = _json_value(v)
+    defaultParamMap = {
+        p.name: _json_value(v) for p, v in instance._defaultParamMap.items()
+    }
+    meta = {
+        "class": spark_class,
+        "timestamp": int(time.time() * 1000),
+        "sparkVersion": SPARK_VERSION,
+        "uid": instance.uid,
+        "paramMap": paramMap,
+        "defaultParamMap": defaultParamMap,
+    }
+    if extra:
+        meta.update(extra)
+    with open(os.path.join(path, "metadata", "part-00000"), "w") as f:
+        f.write(json.dumps(meta) + "\n")
+    open(os.path.join(path, "metadata", "_SUCCESS"), "w").close()
+
+
+def read_metadata(path: str) -> Dict[str, Any]:
+    with open(os.path.join(path, "metadata", "part-00000")) as f:
+        return json.loads(f.readline())
+
+
+def _apply_params(instance, meta: Dict[str, Any]) -> None:
+    instance.uid = meta["uid"]
+    # re-parent instance params so set/get keep working after the uid change
+    from flowml.param import Param
+
+    for name, value in list(instance.__dict__.items()):
+        if isinstance(value, Param):
+            value.parent = instance.uid
+    # NOTE: _paramMap/_defaultParamMap keys are the (re-parented) same objects
+    for name, v in meta.get("defaultParamMap", {}).items():
+        if instance.hasParam(name):
+            p = instance.getParam(name)
+            if v is not None and p.typeConverter is not None:
+                v = p.typeConverter(v)
+            instance._defaultParamMap[p] = v
+    for name, v in meta.get("paramMap", {}).items():
+        if instance.hasParam(name):
+            p = instance.getParam(name)
+            if v is not None and p.typeConverter is not None:
+                v = p.typeConverter(v)
+            instance._paramMap[p] = v
+
+
+# ---------------------------------------------------------------------------
+# parquet helpers (VectorUDT / MatrixUDT structs)
+# ---------------------------------------------------------------------------
+
+
+def _write_parquet(path: str, table) -> None:
+    import pyarrow.parquet as pq
+
+    os.makedirs(path, exist_ok=True)
+    pq.write_table(table, os.path.join(path, "part-00000.parquet"))
+    open(os.path.join(path, "_SUCCESS"), "w").close()
+
+
+def _read_parquet(path: str):
+    import pyarrow.parquet as pq
+
+    parts = sorted(
+        f for f in os.listdir(path) if f.endswith(".parquet") or f.startswith("part-")
+    )
+    parts = [f for f in parts if not f.startswith("_")]
+    import pyarrow as pa
+
+    tables = [pq.read_table(os.path.join(path, f)) for f in parts]
+    return pa.concat_tables(tables)
+
+
+def vector_struct(values: np.ndarray) -> Dict[str, Any]:
+    """Dense VectorUDT struct row (type=1)."""
+    return {
+        "type": 1,
+        "size": None,
+        "indices": None,
+        "values": [float(v) for v in np.asarray(values, dtype=np.float64)],
+    }
+
+
+def vector_from_struct(s: Dict[str, Any]) -> np.ndarray:
+    if s["type"] == 1:
+        return np.asarray(s["values"], dtype=np.float64)
+    size = int(s["size"])
+    arr = np.zeros(size, dtype=np.float64)
+    idx = np.asarray(s["indices"], dtype=np.int64)
+    arr[idx] = np.asarray(s["values"], dtype=np.float64)
+    return arr
+
+
+def matrix_struct(m: np.ndarray) -> Dict[str, Any]:
+    """Dense MatrixUDT struct row (type=1, column-major values)."""
+    m = np.asarray(m, dtype=np.float64)
+    return {
+        "type": 1,
+        "numRows": int(m.shape[0]),
+        "numCols": int(m.shape[1]),
+        "colPtrs": None,
+        "rowIndices": None,
+        "values": [float(v) for v in m.T.ravel()],
+        "isTransposed": False,
+    }
+
+
+def matrix_from_struct(s: Dict[str, Any]) -> np.ndarray:
+    vals = np.asarray(s["values"], dtype=np.float64)
+    r, c = int(s["numRows"]), int(s["numCols"])
+    if s.get("isTransposed"):
+        return vals.reshape(r, c)
+    return vals.reshape(c, r).T
+
+
+def _table_from_rows(rows: List[Dict[str, Any]]):
+    import pyarrow as pa
+
+    return pa.Table.from_pylist(rows)
+
+
+# ---------------------------------------------------------------------------
+# per-class writers
+# ---------------------------------------------------------------------------
+
+
+def _save_instance(instance, path: str) -> None:
+    from flowml.pipeline import Pipeline, PipelineModel
+
+    cls = type(instance).__name__
+    os.makedirs(path, exist_ok=True)
+    if cls == "Pipeline":
+        _save_pipeline_stages(instance.getStages(), path)
+        save_metadata(instance, path)
+        return
+    if cls == "PipelineModel":
+        _save_pipeline_stages(instance.stages, path)
+        save_metadata(instance, path)
+        return
+    save_metadata(instance, path)
+    rows = _data_rows(instance, path)
+    if rows is not None:
+        _write_parquet(os.path.join(path, "data"), _table_from_rows(rows))
+
+
+def _save_pipeline_stages(stages, path: str) -> None:
+    stages_dir = os.path.join(path, "stages")
+    os.makedirs(stages_dir, exist_ok=True)
+    for i, stage in enumerate(stages):
+        stage_dir = os.path.join(stages_dir, f"{i}_{stage.uid}")
+        _save_instance(stage, stage_dir)
+
+
+def _tree_node_rows(tree, tree_id: int) -> List[Dict[str, Any]]:
+    """Pre-order NodeData rows (SURVEY.md §2.2 persistence schema)."""
+    rows: List[Dict[str, Any]] = []
+    order: List[int] = []
+    new_id: Dict[int, int] = {}
+
+    def visit(i: int):
+        new_id[i] = len(order)
+        order.append(i)
+        if tree.feature[i] >= 0:
+            visit(tree.left[i])
+            visit(tree.right[i])
+
+    if tree.n_nodes:
+        visit(0)
+    for i in order:
+        internal = tree.feature[i] >= 0
+        rows.append(
+            {
+                "treeID": tree_id,
+                "nodeData": {
+                    "id": new_id[i],
+                    "prediction": float(tree.prediction(i)),
+                    "impurity": float(tree.impurity[i]),
+                    "impurityStats": [float(v) for v in tree.stats[i]],
+                    "gain": float(tree.gain[i]) if internal else -1.0,
+                    "leftChild": new_id[tree.left[i]] if internal else -1,
+                    "rightChild": new_id[tree.right[i]] if internal else -1,
+                    "split": {
+                        "featureIndex": int(tree.feature[i]) if internal else -1,
+                        # continuous split: single-element threshold list
+                        # (numCategories = 0 marks continuous; categorical
+                        # split-sets are a documented divergence)
+                        "leftCategoriesOrThreshold": [float(tree.thr_raw[i])] if internal else [],
+                        "numCategories": 0 if internal else -1,
+                    },
+                },
+            }
+        )
+    return rows
+
+
+def _trees_from_node_rows(rows: List[Dict[str, Any]]):
+    """Rebuild FlatTree list from NodeData rows."""
+    from flowml.classification.tree import FlatTree
+
+    by_tree: Dict[int, List[Dict[str, Any]]] = {}
+    for r in rows:
+        by_tree.setdefault(int(r["treeID"]), []).append(r["nodeData"])
+    trees = []
+    for tid in sorted(by_tree):
+        nodes = sorted(by_tree[tid], key=lambda nd: nd["id"])
+        tr = FlatTree()
+        for nd in nodes:
+            stats = np.asarray(nd["impurityStats"], dtype=np.float64)
+            cnt = float(stats.sum() if len(stats) != 3 else stats[2])
+            tr.add_node(stats, float(nd["impurity"]), cnt)
+        for nd in nodes:
+            i = nd["id"]
+            if nd["leftChild"] >= 0:
+                tr.feature[i] = int(nd["split"]["featureIndex"])
+                tr.thr_raw[i] = float(nd["split"]["leftCategoriesOrThreshold"][0])
+                tr.left[i] = int(nd["leftChild"])
+                tr.right[i] = int(nd["rightChild"])
+                tr.gain[i] = float(nd["gain"])
+        trees.append(tr)
+    return trees
+
+
+def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
+    cls = type(instance).__name__
+    if cls == "StringIndexerModel":
+        return [{"labelsArray": [list(l) for l in instance.labelsArray]}]
+    if cls == "OneHotEncoderModel":
+        return [{"categorySizes": [int(s) for s in instance.categorySizes]}]
+    if cls == "StandardScalerModel":
+        return [
+            {
+                "std": vector_struct(instance.std.numpy()),
+                "mean": vector_struct(instance.mean.numpy()),
+            }
+        ]
+    if cls == "MinMaxScalerModel":
+        return [
+            {
+                "originalMin": vector_struct(instance.originalMin.numpy()),
+                "originalMax": vector_struct(instance.originalMax.numpy()),
+            }
+        ]
+    if cls == "LogisticRegressionModel":
+        return [
+            {
+                "numClasses": int(instance.numClasses),
+                "numFeatures": int(instance.numFeatures),
+                "interceptVector": vector_struct(instance._b.to(torch.float64).numpy()),
+                "coefficientMatrix": matrix_struct(instance._W.to(torch.float64).numpy()),
+                "isMultinomial": bool(instance.isMultinomial),
+            }
+        ]
+    if cls == "MultilayerPerceptronClassificationModel":
+        return [{"weights": vector_struct(instance._weights.to(torch.float64).numpy())}]
+    if cls in ("RandomForestClassificationModel", "GBTClassificationModel"):
+        rows: List[Dict[str, Any]] = []
+        for tid, tree in enumerate(instance._trees):
+            rows.extend(_tree_node_rows(tree, tid))
+        # treesMetadata: per-tree {treeID, metadata(json), weights}
+        tw = (
+            instance.treeWeights
+            if cls == "GBTClassificationModel"
+            else [1.0] * len(instance._trees)
+        )
+        tmeta = [
+            {"treeID": t, "metadata": json.dumps({"numFeatures": instance.numFeatures}), "weights": float(w)}
+            for t, w in enumerate(tw)
+        ]
+        _write_parquet(os.path.join(path, "treesMetadata"), _table_from_rows(tmeta))
+        return rows
+    if cls == "DecisionTreeClassificationModel":
+        return _tree_node_rows(instance._tree, 0)
+    # pure-param transformers/estimators (VectorAssembler, StringIndexer, ...)
+    return None
+
+
+# ---------------------------------------------------------------------------
+# loaders
+# ---------------------------------------------------------------------------
+
+
+def load(path: str):
+    meta = read_metadata(path)
+    spark_class = meta["class"]
+    cls = _our_class(spark_class)
+    simple = cls.__name__
+
+    if simple in ("Pipeline", "PipelineModel"):
+        stages_dir = os.path.join(path, "stages")
+        stages = []
+        if os.path.isdir(stages_dir):
+            entries = sorted(
+                os.listdir(stages_dir), key=lambda s: int(s.split("_", 1)[0])
+            )
+            for e in entries:
+                stages.append(load(os.path.join(stages_dir, e)))
+        if simple == "Pipeline":
+            inst = cls(stages=stages)
+        else:
+            inst = cls(stages)
+        _apply_params(inst, meta)
+        return inst
+
+    data_dir = os.path.join(path, "data")
+    data_rows = None
+    if os.path.isdir(data_dir):
+        data_rows = _read_parquet(data_dir).to_pylist()
+
+    inst = _construct(cls, simple, data_rows, path, meta)
+    _apply_params(inst, meta)
+    return inst
+
+
+def _construct(cls, simple: str, rows, path: str, meta):
+    if simple == "StringIndexerModel":
+        return cls(labelsArray=[list(l) for l in rows[0]["labelsArray"]])
+    if simple == "OneHotEncoderModel":
+        return cls(categorySizes=[int(v) for v in rows[0]["categorySizes"]])
+    if simple == "StandardScalerModel":
+        return cls(
+            mean=torch.from_numpy(vector_from_struct(rows[0]["mean"])),
+            std=torch.from_numpy(vector_from_struct(rows[0]["std"])),
+        )
+    if simple == "MinMaxScalerModel":
+        return cls(
+            originalMin=torch.from_numpy(vector_from_struct(rows[0]["originalMin"])),
+            originalMax=torch.from_numpy(vector_from_struct(rows[0]["originalMax"])),
+        )
+    if simple == "LogisticRegressionModel":
+        r = rows[0]
+        return cls(
+            coefficients=torch.from_numpy(matrix_from_struct(r["coefficientMatrix"])).to(torch.float32),
+            intercepts=torch.from_numpy(vector_from_struct(r["interceptVector"])).to(torch.float32),
+            numClasses=int(r["numClasses"]),
+            numFeatures=int(r["numFeatures"]),
+            isMultinomial=bool(r["isMultinomial"]),
+        )
+    if simple == "MultilayerPerceptronClassificationModel":
+        layers = meta.get("paramMap", {}).get("layers") or meta.get(
+            "defaultParamMap", {}
+        ).get("layers")
+        w = torch.from_numpy(vector_from_struct(rows[0]["weights"])).to(torch.float32)
+        activation = meta.get("paramMap", {}).get("activation", "relu")
+        return cls(layers=[int(x) for x in layers], weights=w, activation=activation)
+    if simple == "RandomForestClassificationModel":
+        trees = _trees_from_node_rows(rows)
+        n_classes = max(len(t.stats[0]) for t in trees)
+        nf = _num_features_from_trees_meta(path, trees)
+        return cls(trees=trees, numClasses=n_classes, numFeatures=nf)
+    if simple == "GBTClassificationModel":
+        trees = _trees_from_node_rows(rows)
+        tm = _read_parquet(os.path.join(path, "treesMetadata")).to_pylist()
+        weights = [r["weights"] for r in sorted(tm, key=lambda r: r["treeID"])]
+        nf = _num_features_from_trees_meta(path, trees)
+        return cls(trees=trees, treeWeights=weights, numFeatures=nf)
+    if simple == "DecisionTreeClassificationModel":
+        trees = _trees_from_node_rows(rows)
+        t = trees[0]
+        return cls(tree=t, numClasses=len(t.stats[0]), numFeatures=0)
+    # parameter-only stages
+    return cls()
+
+
+def _num_features_from_trees_meta(path: str, trees) -> int:
+    tm_path = os.path.join(path, "treesMetadata")
+    if os.path.isdir(tm_path):
+        tm = _read_parquet(tm_path).to_pylist()
+        if tm:
+            try:
+                return int(json.loads(tm[0]["metadata"]).get("numFeatures", 0))
+            except Exception:  # noqa: BLE001
+                pass
+    mx = 0
+    for t in trees:
+        for f in t.feature:
+            mx = max(mx, f + 1)
+    return mx
+
+
+# ---------------------------------------------------------------------------
+# writer facade
+# ---------------------------------------------------------------------------
+
+
+class _WriterBase:
+    def __init__(self, instance):
+        self.instance = instance
+        self._overwrite = False
+
+    def overwrite(self):
+        self._overwrite = True
+        return self
+
+    def save(self, path: str):
+        if os.path.exists(path):
+            if self._overwrite:
+                import shutil
+
+                shutil.rmtree(path)
+            else:
+                raise IOError(
+                    f"Path {path} already exists. Use write().overwrite().save(path)."
+                )
+        _save_instance(self.instance, path)
+
+
+class GenericWriter(_WriterBase):
+    pass
+
+
+class PipelineWriter(_WriterBase):
+    pass
+
+
+class PipelineModelWriter(_WriterBase):
+    pass

@@ -1,0 +1,146 @@
+"""Multi-process DP tests on the gloo backend (SURVEY.md §4
+"multi-node-without-a-cluster"): world_size=2 on CPU; a distributed fit over
+two shards must equal the single-process fit over the full data, because the
+replicated-update design all-reduces the same global statistics."""
+
+import multiprocessing as mp
+import os
+
+import numpy as np
+import pytest
+import torch
+
+
+def _make_data(n=600, f=5, k=3, seed=11):
+    r = np.random.default_rng(seed)
+    X = r.normal(size=(n, f)).astype(np.float32)
+    W = r.normal(size=(f, k))
+    y = (X @ W + 0.8 * r.normal(size=(n, k))).argmax(1).astype(np.float64)
+    return X, y
+
+
+def _worker(rank, world, port, fn_name, out_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        result = globals()[fn_name](rank, world)
+        if rank == 0:
+            out_q.put(("ok", result))
+    except Exception as e:  # noqa: BLE001
+        out_q.put(("err", f"rank{rank}: {type(e).__name__}: {e}"))
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def _run_dist(fn_name, world=2):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = int(os.environ.get("FLOWML_TEST_PORT", "29611")) + abs(hash(fn_name)) % 500
+    procs = [
+        ctx.Process(target=_worker, args=(r, world, port, fn_name, q)) for r in range(world)
+    ]
+    for p in procs:
+        p.start()
+    status, payload = q.get(timeout=180)
+    for p in procs:
+        p.join(timeout=60)
+        if p.is_alive():
+            p.terminate()
+    assert status == "ok", payload
+    return payload
+
+
+# --- worker bodies (module-level for spawn pickling) -----------------------
+
+
+def _lr_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.classification import LogisticRegression
+
+    X, y = _make_data()
+    lo, hi = dist.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi]), "label": y[lo:hi]})
+    m = LogisticRegression(maxIter=40, regParam=0.01).fit(df)
+    return m._W.numpy(), m._b.numpy()
+
+
+def _rf_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.classification import RandomForestClassifier
+
+    X, y = _make_data()
+    lo, hi = dist.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi]), "label": y[lo:hi]})
+    m = RandomForestClassifier(numTrees=4, maxDepth=3, seed=5, bootstrap=False).fit(df)
+    full = FlowFrame({"features": torch.from_numpy(X), "label": y})
+    pred = m.transform(full)["prediction"].numpy()
+    return pred
+
+
+def _mlp_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.classification import MultilayerPerceptronClassifier
+
+    X, y = _make_data()
+    lo, hi = dist.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi]), "label": y[lo:hi]})
+    m = MultilayerPerceptronClassifier(layers=[5, 8, 3], maxIter=15, seed=3).fit(df)
+    return m._weights.numpy()
+
+
+def _fused_allreduce_body(rank, world):
+    from flowml.dist import FusedAllReduce
+
+    a = torch.full((4,), float(rank + 1))
+    b = torch.full((2, 3), float(10 * (rank + 1)))
+    f = FusedAllReduce([a, b])
+    f.reduce()
+    return a.numpy(), b.numpy()
+
+
+# --- tests ------------------------------------------------------------------
+
+
+def test_lr_dist_equals_single():
+    W_dist, b_dist = _run_dist("_lr_dist_body")
+    from flowml import FlowFrame
+    from flowml.classification import LogisticRegression
+
+    X, y = _make_data()
+    m = LogisticRegression(maxIter=40, regParam=0.01).fit(
+        FlowFrame({"features": torch.from_numpy(X), "label": y})
+    )
+    assert np.allclose(W_dist, m._W.numpy(), atol=1e-2)
+    assert np.allclose(b_dist, m._b.numpy(), atol=1e-2)
+
+
+def test_rf_dist_predictions_close_to_single():
+    pred_dist = _run_dist("_rf_dist_body")
+    from flowml import FlowFrame
+    from flowml.classification import RandomForestClassifier
+
+    X, y = _make_data()
+    full = FlowFrame({"features": torch.from_numpy(X), "label": y})
+    m = RandomForestClassifier(numTrees=4, maxDepth=3, seed=5, bootstrap=False).fit(full)
+    pred_single = m.transform(full)["prediction"].numpy()
+    # identical histograms (no bootstrap) -> identical trees -> identical preds
+    assert (pred_dist == pred_single).mean() > 0.99
+
+
+def test_mlp_dist_equals_single():
+    w_dist = _run_dist("_mlp_dist_body")
+    from flowml import FlowFrame
+    from flowml.classification import MultilayerPerceptronClassifier
+
+    X, y = _make_data()
+    m = MultilayerPerceptronClassifier(layers=[5, 8, 3], maxIter=15, seed=3).fit(
+        FlowFrame({"features": torch.from_numpy(X), "label": y})
+    )
+    assert np.allclose(w_dist, m._weights.numpy(), atol=1e-3)
+
+
+def test_fused_allreduce():
+    a, b = _run_dist("_fused_allreduce_body")
+    assert np.allclose(a, 3.0)  # 1 + 2
+    assert np.allclose(b, 30.0)  # 10 + 20

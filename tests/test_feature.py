@@ -1,0 +1,158 @@
+"""Golden-row featurizer tests (SURVEY.md §4 "MLTest.testTransformer")."""
+
+import numpy as np
+import pytest
+import torch
+
+from flowml import FlowFrame
+from flowml.feature import (
+    MinMaxScaler,
+    OneHotEncoder,
+    StandardScaler,
+    StringIndexer,
+    StringIndexerModel,
+    VectorAssembler,
+)
+
+
+class TestStringIndexer:
+    def test_frequency_desc_with_alpha_ties(self):
+        # b:3, a:2, c:2 -> b=0, then tie a/c broken alphabetically: a=1, c=2
+        df = FlowFrame({"s": ["b", "b", "b", "a", "a", "c", "c"]})
+        m = StringIndexer(inputCol="s", outputCol="i").fit(df)
+        assert m.labels == ["b", "a", "c"]
+        out = m.transform(df)
+        assert out["i"].tolist() == [0.0, 0.0, 0.0, 1.0, 1.0, 2.0, 2.0]
+        assert out["i"].dtype == torch.float64
+
+    def test_order_types(self):
+        df = FlowFrame({"s": ["b", "b", "a", "c", "c", "c"]})
+        cases = {
+            "frequencyDesc": ["c", "b", "a"],
+            "frequencyAsc": ["a", "b", "c"],
+            "alphabetAsc": ["a", "b", "c"],
+            "alphabetDesc": ["c", "b", "a"],
+        }
+        for order, expect in cases.items():
+            m = StringIndexer(inputCol="s", outputCol="i", stringOrderType=order).fit(df)
+            assert m.labels == expect, order
+
+    def test_handle_invalid_error(self):
+        train = FlowFrame({"s": ["a", "b"]})
+        test = FlowFrame({"s": ["a", "zz"]})
+        m = StringIndexer(inputCol="s", outputCol="i").fit(train)
+        with pytest.raises(ValueError, match="unseen"):
+            m.transform(test)
+
+    def test_handle_invalid_skip_keep(self):
+        train = FlowFrame({"s": ["a", "a", "b"]})
+        test = FlowFrame({"s": ["a", "zz", "b"], "x": [1.0, 2.0, 3.0]})
+        m_skip = StringIndexer(inputCol="s", outputCol="i", handleInvalid="skip").fit(train)
+        out = m_skip.transform(test)
+        assert out.count() == 2
+        assert out["x"].tolist() == [1.0, 3.0]
+        m_keep = StringIndexer(inputCol="s", outputCol="i", handleInvalid="keep").fit(train)
+        out = m_keep.transform(test)
+        assert out["i"].tolist() == [0.0, 2.0, 1.0]  # unseen -> numLabels
+
+    def test_multi_col(self):
+        df = FlowFrame({"s1": ["a", "b"], "s2": ["x", "x"]})
+        m = StringIndexer(inputCols=["s1", "s2"], outputCols=["i1", "i2"]).fit(df)
+        out = m.transform(df)
+        assert "i1" in out and "i2" in out
+        assert len(m.labelsArray) == 2
+
+    def test_numeric_input_cast(self):
+        df = FlowFrame({"n": [3.0, 3.0, 1.0]})
+        m = StringIndexer(inputCol="n", outputCol="i").fit(df)
+        assert m.labels == ["3.0", "1.0"]
+
+    def test_from_labels(self):
+        m = StringIndexerModel.from_labels(["x", "y"], inputCol="s", outputCol="i")
+        out = m.transform(FlowFrame({"s": ["y", "x"]}))
+        assert out["i"].tolist() == [1.0, 0.0]
+
+
+class TestOneHotEncoder:
+    def test_drop_last_default(self):
+        df = FlowFrame({"i": [0.0, 1.0, 2.0, 1.0]})
+        m = OneHotEncoder(inputCol="i", outputCol="v").fit(df)
+        assert m.categorySizes == [3]
+        out = m.transform(df)
+        v = out["v"]
+        assert v.shape == (4, 2)  # dropLast: 3 cats -> size 2
+        assert v.tolist() == [[1, 0], [0, 1], [0, 0], [0, 1]]
+
+    def test_no_drop_last(self):
+        df = FlowFrame({"i": [0.0, 1.0, 2.0]})
+        m = OneHotEncoder(inputCol="i", outputCol="v", dropLast=False).fit(df)
+        assert m.transform(df)["v"].tolist() == [[1, 0, 0], [0, 1, 0], [0, 0, 1]]
+
+    def test_handle_invalid(self):
+        train = FlowFrame({"i": [0.0, 1.0]})
+        test = FlowFrame({"i": [0.0, 5.0]})
+        m = OneHotEncoder(inputCol="i", outputCol="v").fit(train)
+        with pytest.raises(ValueError):
+            m.transform(test)
+        m_keep = OneHotEncoder(inputCol="i", outputCol="v", handleInvalid="keep").fit(train)
+        v = m_keep.transform(test)["v"]
+        # keep: size = numCats + 1 - dropLast = 2
+        assert v.shape == (2, 2)
+        assert v.tolist() == [[1, 0], [0, 0]]  # invalid -> extra cat (dropped? no: idx=2 of size 2 -> all-zero is last cat... idx 2 == size -> zero row)
+
+
+class TestVectorAssembler:
+    def test_concat_order(self):
+        df = FlowFrame(
+            {
+                "a": [1.0, 2.0],
+                "v": torch.tensor([[10.0, 20.0], [30.0, 40.0]]),
+                "b": [5.0, 6.0],
+            }
+        )
+        out = VectorAssembler(inputCols=["a", "v", "b"], outputCol="f").transform(df)
+        assert out["f"].tolist() == [[1, 10, 20, 5], [2, 30, 40, 6]]
+        assert out["f"].dtype == torch.float32
+
+    def test_handle_invalid(self):
+        df = FlowFrame({"a": [1.0, float("nan")]})
+        va = VectorAssembler(inputCols=["a"], outputCol="f")
+        with pytest.raises(ValueError):
+            va.transform(df)
+        out = va.copy({va.handleInvalid: "skip"}).transform(df)
+        assert out.count() == 1
+        out = va.copy({va.handleInvalid: "keep"}).transform(df)
+        assert out.count() == 2
+
+    def test_string_input_rejected(self):
+        df = FlowFrame({"s": ["a", "b"]})
+        with pytest.raises(TypeError):
+            VectorAssembler(inputCols=["s"], outputCol="f").transform(df)
+
+
+class TestScalers:
+    def test_standard_scaler_defaults(self):
+        x = torch.tensor([[1.0, 10.0], [2.0, 10.0], [3.0, 10.0]])
+        df = FlowFrame({"v": x})
+        m = StandardScaler(inputCol="v", outputCol="s").fit(df)
+        out = m.transform(df)["s"]
+        # withMean=False default: divide by sample std only
+        std0 = float(x[:, 0].std(unbiased=True))
+        assert torch.allclose(out[:, 0], x[:, 0] / std0, atol=1e-6)
+        # zero-variance feature -> scaled to 0 (Spark: factor 0)
+        assert out[:, 1].abs().sum() == 0
+
+    def test_standard_scaler_with_mean(self):
+        x = torch.tensor([[1.0], [3.0]])
+        m = StandardScaler(inputCol="v", outputCol="s", withMean=True).fit(
+            FlowFrame({"v": x})
+        )
+        out = m.transform(FlowFrame({"v": x}))["s"]
+        assert abs(float(out.mean())) < 1e-6
+
+    def test_minmax_scaler(self):
+        x = torch.tensor([[0.0, 5.0], [10.0, 5.0]])
+        m = MinMaxScaler(inputCol="v", outputCol="s").fit(FlowFrame({"v": x}))
+        out = m.transform(FlowFrame({"v": x}))["s"]
+        assert out[:, 0].tolist() == [0.0, 1.0]
+        assert out[:, 1].tolist() == [0.5, 0.5]  # constant -> midpoint

@@ -1,0 +1,94 @@
+"""FlowFrame + CSV ingestion + randomSplit."""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from flowml import FlowFrame, read_csv
+
+
+def test_basic_columns():
+    df = FlowFrame({"a": [1.0, 2.0, 3.0], "s": ["x", "y", "x"]})
+    assert df.count() == 3
+    assert df.columns == ["a", "s"]
+    assert df.is_string("s") and not df.is_string("a")
+    assert isinstance(df["a"], torch.Tensor)
+
+
+def test_mismatched_lengths():
+    with pytest.raises(ValueError):
+        FlowFrame({"a": [1.0], "b": [1.0, 2.0]})
+
+
+def test_with_column_select_drop_rename():
+    df = FlowFrame({"a": [1.0, 2.0], "b": [3.0, 4.0]})
+    df2 = df.withColumn("c", torch.tensor([5.0, 6.0]))
+    assert df2.columns == ["a", "b", "c"]
+    assert df.columns == ["a", "b"]  # immutable
+    assert df2.select("c", "a").columns == ["c", "a"]
+    assert df2.drop("b").columns == ["a", "c"]
+    assert df2.withColumnRenamed("a", "z").columns == ["z", "b", "c"]
+
+
+def test_filter_and_vector_columns():
+    df = FlowFrame({"v": torch.arange(12, dtype=torch.float32).reshape(4, 3), "s": ["a", "b", "c", "d"]})
+    assert df.is_vector("v")
+    out = df.filter(np.array([True, False, True, False]))
+    assert out.count() == 2
+    assert out["s"].tolist() == ["a", "c"]
+    assert torch.equal(out["v"], torch.tensor([[0.0, 1, 2], [6, 7, 8]]))
+
+
+def test_random_split_seeded():
+    df = FlowFrame({"a": np.arange(1000, dtype=np.float64)})
+    a1, b1 = df.randomSplit([0.7, 0.3], seed=42)
+    a2, b2 = df.randomSplit([0.7, 0.3], seed=42)
+    assert a1.count() + b1.count() == 1000
+    assert torch.equal(a1["a"], a2["a"])  # deterministic
+    assert 600 < a1.count() < 800  # Binomial around 700
+    a3, _ = df.randomSplit([0.7, 0.3], seed=43)
+    assert not torch.equal(a1["a"], a3["a"])
+
+
+def test_union_and_limit():
+    df = FlowFrame({"a": [1.0, 2.0], "s": ["x", "y"]})
+    u = df.union(df)
+    assert u.count() == 4
+    assert u.limit(3).count() == 3
+
+
+def test_collect_vectors():
+    df = FlowFrame({"v": torch.tensor([[1.0, 2.0]]), "a": [3.0]})
+    row = df.collect()[0]
+    from flowml.linalg import DenseVector
+
+    assert isinstance(row["v"], DenseVector)
+    assert row["v"].toArray().tolist() == [1.0, 2.0]
+    assert row["a"] == 3.0
+
+
+def test_read_csv(tmp_path):
+    p = os.path.join(tmp_path, "flows.csv")
+    with open(p, "w") as f:
+        f.write("proto,dur,label\n")
+        f.write("tcp,1.5,0\n")
+        f.write("udp,2.0,1\n")
+        f.write("tcp,0.5,0\n")
+    df = read_csv(p)
+    assert df.count() == 3
+    assert df.is_string("proto")
+    assert df["dur"].dtype == torch.float64
+    assert df["label"].tolist() == [0, 1, 0]
+
+
+def test_read_csv_no_header_schema(tmp_path):
+    p = os.path.join(tmp_path, "nh.csv")
+    with open(p, "w") as f:
+        f.write("tcp,1.5,0\n")
+        f.write("udp,2.0,1\n")
+    df = read_csv(p, header=False)
+    assert df.columns == ["_c0", "_c1", "_c2"]
+    df2 = read_csv(p, header=False, columns=["proto", "dur", "label"])
+    assert df2.columns == ["proto", "dur", "label"]

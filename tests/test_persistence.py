@@ -1,0 +1,197 @@
+"""Spark ML save-format round trips + schema fixtures."""
+
+import json
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from flowml import FlowFrame, Pipeline
+from flowml.classification import (
+    GBTClassifier,
+    LogisticRegression,
+    MultilayerPerceptronClassifier,
+    RandomForestClassifier,
+)
+from flowml.evaluation import MulticlassClassificationEvaluator
+from flowml.feature import OneHotEncoder, StringIndexer, VectorAssembler
+from flowml.pipeline import PipelineModel
+
+
+def _fit_pipeline():
+    r = np.random.default_rng(0)
+    n = 300
+    df = FlowFrame(
+        {
+            "proto": r.choice(["tcp", "udp", "icmp"], size=n),
+            "x1": r.normal(size=n),
+            "x2": r.normal(size=n),
+            "label": r.integers(0, 3, n).astype(np.float64),
+        }
+    )
+    pipe = Pipeline(
+        stages=[
+            StringIndexer(inputCol="proto", outputCol="pIdx"),
+            OneHotEncoder(inputCol="pIdx", outputCol="pVec"),
+            VectorAssembler(inputCols=["pVec", "x1", "x2"], outputCol="features"),
+            LogisticRegression(maxIter=20),
+        ]
+    )
+    return pipe.fit(df), df
+
+
+def test_directory_layout(tmp_path):
+    model, df = _fit_pipeline()
+    p = str(tmp_path / "pm")
+    model.save(p)
+    assert os.path.isfile(os.path.join(p, "metadata", "part-00000"))
+    assert os.path.isfile(os.path.join(p, "metadata", "_SUCCESS"))
+    stages = sorted(os.listdir(os.path.join(p, "stages")))
+    assert len(stages) == 4
+    # stage dirs named <idx>_<uid>
+    for i, s in enumerate(stages):
+        assert s.startswith(f"{i}_")
+        assert os.path.isfile(os.path.join(p, "stages", s, "metadata", "part-00000"))
+    # fitted models carry data/ parquet
+    lr_dir = os.path.join(p, "stages", stages[3])
+    assert any(f.endswith(".parquet") for f in os.listdir(os.path.join(lr_dir, "data")))
+
+
+def test_metadata_json_fields(tmp_path):
+    model, _ = _fit_pipeline()
+    p = str(tmp_path / "pm")
+    model.save(p)
+    with open(os.path.join(p, "metadata", "part-00000")) as f:
+        meta = json.loads(f.readline())
+    assert meta["class"] == "org.apache.spark.ml.PipelineModel"
+    assert set(meta) >= {"class", "timestamp", "sparkVersion", "uid", "paramMap", "defaultParamMap"}
+    assert meta["uid"] == model.uid
+    # LR stage metadata: set params in paramMap, defaults in defaultParamMap
+    stages = sorted(os.listdir(os.path.join(p, "stages")))
+    with open(os.path.join(p, "stages", stages[3], "metadata", "part-00000")) as f:
+        lrmeta = json.loads(f.readline())
+    assert lrmeta["class"] == "org.apache.spark.ml.classification.LogisticRegressionModel"
+    assert lrmeta["paramMap"].get("maxIter") == 20
+    assert lrmeta["defaultParamMap"].get("regParam") == 0.0
+
+
+def test_pipeline_round_trip_predictions(tmp_path):
+    model, df = _fit_pipeline()
+    p = str(tmp_path / "pm")
+    model.save(p)
+    m2 = PipelineModel.load(p)
+    a = model.transform(df)
+    b = m2.transform(df)
+    assert torch.equal(a["prediction"], b["prediction"])
+    assert torch.allclose(a["probability"], b["probability"])
+    # params survive
+    assert m2.stages[3].getMaxIter() == 20
+    assert m2.stages[0].labels == model.stages[0].labels
+
+
+def test_overwrite_semantics(tmp_path):
+    model, _ = _fit_pipeline()
+    p = str(tmp_path / "pm")
+    model.save(p)
+    with pytest.raises(IOError):
+        model.save(p)
+    model.write().overwrite().save(p)  # works
+
+
+def _tree_df(binary=False):
+    r = np.random.default_rng(1)
+    X = r.normal(size=(400, 5)).astype(np.float32)
+    if binary:
+        y = (X[:, 0] > 0).astype(np.float64)
+    else:
+        y = r.integers(0, 3, 400).astype(np.float64)
+    return FlowFrame({"features": torch.from_numpy(X), "label": y})
+
+
+def test_rf_round_trip(tmp_path):
+    df = _tree_df()
+    m = RandomForestClassifier(numTrees=5, maxDepth=4, seed=0).fit(df)
+    p = str(tmp_path / "rf")
+    m.save(p)
+    from flowml.classification import RandomForestClassificationModel
+
+    m2 = RandomForestClassificationModel.load(p)
+    assert torch.equal(m.transform(df)["prediction"], m2.transform(df)["prediction"])
+    assert torch.allclose(
+        m.transform(df)["rawPrediction"], m2.transform(df)["rawPrediction"], atol=1e-6
+    )
+    assert m2.numClasses == 3
+    assert m2.numFeatures == 5
+    # treesMetadata exists
+    assert os.path.isdir(os.path.join(p, "treesMetadata"))
+
+
+def test_gbt_round_trip(tmp_path):
+    df = _tree_df(binary=True)
+    m = GBTClassifier(maxIter=8, maxDepth=3, seed=0).fit(df)
+    p = str(tmp_path / "gbt")
+    m.save(p)
+    m2 = GBTClassifier.load(p) if hasattr(GBTClassifier, "load") else None
+    from flowml.classification import GBTClassificationModel
+
+    m2 = GBTClassificationModel.load(p)
+    assert m2.treeWeights == m.treeWeights
+    assert torch.allclose(
+        m.transform(df)["rawPrediction"], m2.transform(df)["rawPrediction"], atol=1e-5
+    )
+
+
+def test_mlp_round_trip(tmp_path):
+    df = _tree_df()
+    m = MultilayerPerceptronClassifier(layers=[5, 8, 3], maxIter=10, seed=2).fit(df)
+    p = str(tmp_path / "mlp")
+    m.save(p)
+    from flowml.classification import MultilayerPerceptronClassificationModel
+
+    m2 = MultilayerPerceptronClassificationModel.load(p)
+    assert torch.allclose(m._weights, m2._weights, atol=1e-6)
+    assert m2.getLayers() == [5, 8, 3]
+    assert torch.equal(m.transform(df)["prediction"], m2.transform(df)["prediction"])
+
+
+def test_node_data_schema(tmp_path):
+    """NodeData parquet rows carry the Spark field set."""
+    import pyarrow.parquet as pq
+
+    df = _tree_df()
+    m = RandomForestClassifier(numTrees=2, maxDepth=3, seed=0).fit(df)
+    p = str(tmp_path / "rf")
+    m.save(p)
+    files = [f for f in os.listdir(os.path.join(p, "data")) if f.endswith(".parquet")]
+    t = pq.read_table(os.path.join(p, "data", files[0]))
+    rows = t.to_pylist()
+    nd = rows[0]["nodeData"]
+    assert set(nd) == {
+        "id",
+        "prediction",
+        "impurity",
+        "impurityStats",
+        "gain",
+        "leftChild",
+        "rightChild",
+        "split",
+    }
+    assert set(nd["split"]) == {"featureIndex", "leftCategoriesOrThreshold", "numCategories"}
+    # root node is id 0, pre-order ids contiguous per tree
+    ids = sorted(r["nodeData"]["id"] for r in rows if r["treeID"] == 0)
+    assert ids == list(range(len(ids)))
+
+
+def test_unfitted_pipeline_round_trip(tmp_path):
+    pipe = Pipeline(
+        stages=[
+            StringIndexer(inputCol="proto", outputCol="pIdx"),
+            VectorAssembler(inputCols=["pIdx", "x1"], outputCol="features"),
+        ]
+    )
+    p = str(tmp_path / "pipe")
+    pipe.save(p)
+    p2 = Pipeline.load(p)
+    assert len(p2.getStages()) == 2
+    assert p2.getStages()[0].getInputCol() == "proto"
