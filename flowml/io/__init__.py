@@ -157,7 +157,10 @@ def _apply_params(instance, meta: Dict[str, Any]) -> None:
     for name, value in list(instance.__dict__.items()):
         if isinstance(value, Param):
             value.parent = instance.uid
-    # NOTE: _paramMap/_defaultParamMap keys are the (re-parented) same objects
+    # re-parenting changed the Params' hashes — rebuild the maps so dict
+    # lookups work again
+    instance._paramMap = {p: v for p, v in instance._paramMap.items()}
+    instance._defaultParamMap = {p: v for p, v in instance._defaultParamMap.items()}
     for name, v in meta.get("defaultParamMap", {}).items():
         if instance.hasParam(name):
             p = instance.getParam(name)
