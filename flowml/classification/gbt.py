@@ -210,7 +210,7 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
             cfg.seed = (seed + 1_000_003 * rnd) & 0x7FFFFFFF
             bag = poisson_bag(N, 1, subsample, False, cfg.seed, dev) if subsample < 1.0 else None
             if train_sel is not None:
-                sel = train_sel.to(torch.float32).unsqueeze(1)
+                sel = train_sel.to(torch.float32).unsqueeze(0)  # (1, N)
                 bag = sel if bag is None else bag * sel
             tree_list = grow_trees(codes, resid, bag, thresholds, cfg)
             tree = tree_list[0]

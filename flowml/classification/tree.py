@@ -251,10 +251,44 @@ def split_scan(
 # ---------------------------------------------------------------------------
 
 
+def _hist_lds_cap(sub_F: int, B: int, S: int) -> int:
+    """Max frontier nodes of ONE tree per K2 launch (per-tree LDS budget).
+
+    The GPU kernel privatizes one tree's chunk histograms in LDS; 128 KiB of
+    the 160 KiB/CU keeps room for staging. The torch path has no such limit
+    but uses the same chunking so results are layout-identical.
+    """
+    bytes_per_node = sub_F * B * S * 4
+    return max(1, (128 * 1024) // max(bytes_per_node, 1))
+
+
+def _make_chunks(frontier, g_max: int, l_cap: int):
+    """Split the (tree-major sorted) frontier into chunks obeying both the
+    total-group budget (maxMemoryInMB) and the per-tree LDS cap."""
+    chunks = []
+    cur: List[Tuple[int, int, int]] = []
+    per_tree = 0
+    cur_tree = -1
+    for item in frontier:
+        t = item[0]
+        if t != cur_tree:
+            cur_tree = t
+            per_tree = 0
+        if cur and (len(cur) >= g_max or per_tree >= l_cap):
+            chunks.append(cur)
+            cur = []
+            per_tree = 0
+        cur.append(item)
+        per_tree += 1
+    if cur:
+        chunks.append(cur)
+    return chunks
+
+
 def grow_trees(
     codes: torch.Tensor,  # N×F uint8 (local shard)
     labels_or_targets: torch.Tensor,  # N int (classification) / N float (regression)
-    bag: Optional[torch.Tensor],  # N×T weights (Poisson) or None
+    bag: Optional[torch.Tensor],  # T×N weights (Poisson) or None
     thresholds: torch.Tensor,  # F×(B-1) raw split values
     cfg: GrowthConfig,
     row_weights: Optional[torch.Tensor] = None,  # optional extra per-row weight
@@ -271,24 +305,25 @@ def grow_trees(
     use_subset = sub_F < F
 
     if bag is not None and row_weights is not None:
-        bag = bag * row_weights.unsqueeze(1)
+        bag = bag * row_weights.unsqueeze(0)
     elif row_weights is not None:
-        bag = row_weights.unsqueeze(1).expand(N, T).contiguous()
+        bag = row_weights.unsqueeze(0).expand(T, N).contiguous()
 
-    # root stats per tree (one histogram pass with a single "feature")
+    labels_u8 = labels_or_targets.to(torch.uint8) if is_cls else torch.zeros(N, dtype=torch.uint8, device=dev)
+    targets = None if is_cls else labels_or_targets.to(torch.float32)
+
     trees = [FlatTree() for _ in range(T)]
-    nodeid = torch.zeros((N, T), dtype=torch.int32, device=dev)
+    nodeid = torch.zeros((T, N), dtype=torch.int32, device=dev)
 
-    # compute root stats via the multi-tree hist on feature 0 only
-    root_group = torch.arange(T, dtype=torch.int32, device=dev).unsqueeze(0).expand(N, T).contiguous()
+    # root stats per tree: one K2 pass over feature 0 only (lut: node0 -> t)
+    lut0 = torch.arange(T, dtype=torch.int32, device=dev).reshape(T, 1)
+    tree_lo0 = torch.arange(T, dtype=torch.int32, device=dev)
     fs0 = torch.zeros((T, 1), dtype=torch.int32, device=dev)
-    if is_cls:
-        h0 = ops.tree_hist_multi(codes, root_group, labels_or_targets, bag, T, B, S, fs0)
-    else:
-        h0 = ops.tree_hist_multi(codes, root_group, labels_or_targets, bag, T, B, 0, fs0,
-                                 targets=labels_or_targets.to(torch.float32))
-    buf = h0.reshape(-1)
-    dist.all_reduce_(buf)
+    h0 = ops.tree_hist_nodes(
+        codes, nodeid, lut0, tree_lo0, labels_u8, bag, T, 1, B, S if is_cls else 0,
+        fs0, targets=targets,
+    )
+    dist.all_reduce_(h0.reshape(-1) if h0.is_contiguous() else h0)
     root_stats = h0.sum(dim=2)[:, 0, :]  # T×S
     rs_np = root_stats.detach().cpu().numpy().astype(np.float64)
 
@@ -301,34 +336,32 @@ def grow_trees(
         if cfg.max_depth > 0 and cnt >= 2 * cfg.min_instances_per_node and imp > 0:
             frontier.append((t, nid, 0))
 
-    rng = np.random.default_rng(cfg.seed ^ 0xF0A57)
-
-    # memory budget → nodes per histogram chunk
+    # memory budget → groups per histogram chunk (maxMemoryInMB parity) and
+    # per-tree LDS cap (GPU kernel constraint)
     bytes_per_node = sub_F * B * S * 4
     g_max = max(1, int(cfg.max_memory_mb * 1024 * 1024 // max(bytes_per_node, 1)))
+    l_cap = _hist_lds_cap(sub_F, B, S)
 
+    thr_host = thresholds.cpu().numpy()
     level = 0
     while frontier and level < cfg.max_depth:
         next_frontier: List[Tuple[int, int, int]] = []
-        # per-tree lookup: local node id -> group index (-1 = inactive)
-        for chunk_start in range(0, len(frontier), g_max):
-            chunk = frontier[chunk_start : chunk_start + g_max]
+        for chunk in _make_chunks(frontier, g_max, l_cap):
             G = len(chunk)
-            # build node→group lookup per tree
-            max_nodes = max(trees[t].n_nodes for t, _, _ in chunk) + 1
+            max_nodes = max(trees[t].n_nodes for t, _, _ in chunk)
             lut = torch.full((T, max_nodes), -1, dtype=torch.int32)
+            tree_lo_np = np.zeros(T, dtype=np.int32)
+            per_tree_count = np.zeros(T, dtype=np.int64)
             for g, (t, nid, _) in enumerate(chunk):
                 lut[t, nid] = g
+                if per_tree_count[t] == 0:
+                    tree_lo_np[t] = g
+                per_tree_count[t] += 1
+            l_max = int(per_tree_count.max())
             lut_d = lut.to(dev)
-            gid = torch.where(
-                nodeid >= 0,
-                lut_d[
-                    torch.arange(T, device=dev).unsqueeze(0).expand(N, T),
-                    nodeid.to(torch.int64).clamp_min(0),
-                ],
-                torch.full_like(nodeid, -1),
-            )
-            # feature subsets per node (seeded by (seed, tree, node))
+            tree_lo = torch.from_numpy(tree_lo_np).to(dev)
+            # feature subsets per node (seeded by (seed, tree, node) —
+            # rank-invariant and reproducible)
             if use_subset:
                 fs_np = np.empty((G, sub_F), dtype=np.int32)
                 for g, (t, nid, _) in enumerate(chunk):
@@ -337,11 +370,10 @@ def grow_trees(
                 fs = torch.from_numpy(fs_np).to(dev)
             else:
                 fs = None
-            if is_cls:
-                hist = ops.tree_hist_multi(codes, gid, labels_or_targets, bag, G, B, S, fs)
-            else:
-                hist = ops.tree_hist_multi(codes, gid, labels_or_targets, bag, G, B, 0, fs,
-                                           targets=labels_or_targets.to(torch.float32))
+            hist = ops.tree_hist_nodes(
+                codes, nodeid, lut_d, tree_lo, labels_u8, bag, G, l_max, B,
+                S if is_cls else 0, fs, targets=targets,
+            )
             dist.all_reduce_(hist.reshape(-1))
             best_f, best_bin, best_gain, parent, left_stats = split_scan(
                 hist, cfg.impurity, cfg.min_instances_per_node, cfg.min_info_gain
@@ -352,7 +384,6 @@ def grow_trees(
             ls = left_stats.cpu().numpy().astype(np.float64)
             ps = parent.cpu().numpy().astype(np.float64)
             fs_host = fs.cpu().numpy() if fs is not None else None
-            thr_host = thresholds.cpu().numpy()
             for g, (t, nid, d) in enumerate(chunk):
                 if bf[g] < 0:
                     continue  # stays leaf
@@ -378,15 +409,26 @@ def grow_trees(
                         next_frontier.append((t, lid, d + 1))
                     if cnt_r >= 2 * cfg.min_instances_per_node and imp_r > 0:
                         next_frontier.append((t, rid, d + 1))
-        # advance node assignments one level (K4b) per tree
-        for t in range(T):
-            tr = trees[t]
-            M = tr.n_nodes
-            feat_of = torch.tensor(tr.feature, dtype=torch.int32, device=dev)
-            thr_of = torch.tensor(tr.thr_bin, dtype=torch.int32, device=dev)
-            left_of = torch.tensor(tr.left, dtype=torch.int32, device=dev)
-            right_of = torch.tensor(tr.right, dtype=torch.int32, device=dev)
-            nodeid[:, t] = ops.node_advance(codes, nodeid[:, t].contiguous(), feat_of, thr_of, left_of, right_of)
+        # advance node assignments one level, all trees in one launch (K4b)
+        feat_cat = torch.tensor(
+            [f for tr in trees for f in tr.feature], dtype=torch.int32, device=dev
+        )
+        thr_cat = torch.tensor(
+            [x for tr in trees for x in tr.thr_bin], dtype=torch.int32, device=dev
+        )
+        left_cat = torch.tensor(
+            [x for tr in trees for x in tr.left], dtype=torch.int32, device=dev
+        )
+        right_cat = torch.tensor(
+            [x for tr in trees for x in tr.right], dtype=torch.int32, device=dev
+        )
+        ofs = [0]
+        for tr in trees:
+            ofs.append(ofs[-1] + tr.n_nodes)
+        node_ofs = torch.tensor(ofs, dtype=torch.int32, device=dev)
+        nodeid = ops.node_advance_multi(
+            codes, nodeid, feat_cat, thr_cat, left_cat, right_cat, node_ofs
+        )
         frontier = next_frontier
         level += 1
 
@@ -477,7 +519,7 @@ class TreeEnsemble:
 def poisson_bag(
     n_rows: int, n_trees: int, rate: float, bootstrap: bool, seed: int, device
 ) -> Optional[torch.Tensor]:
-    """Per-(row,tree) Poisson bag counts (BaggedPoint).
+    """Per-(tree,row) Poisson bag counts, shape (T, N) (BaggedPoint; §3.2).
 
     Seeded per rank by (seed, rank) — deterministic for a fixed world size.
     Returns None when no subsampling is requested (weight ≡ 1).
@@ -486,7 +528,7 @@ def poisson_bag(
         return None
     g = torch.Generator(device="cpu").manual_seed((seed * 7919 + dist.get_rank()) & 0x7FFFFFFF)
     if bootstrap:
-        bag = torch.poisson(torch.full((n_rows, n_trees), rate), generator=g)
+        bag = torch.poisson(torch.full((n_trees, n_rows), rate), generator=g)
     else:
-        bag = (torch.rand((n_rows, n_trees), generator=g) < rate).to(torch.float32)
+        bag = (torch.rand((n_trees, n_rows), generator=g) < rate).to(torch.float32)
     return bag.to(torch.float32).to(device)

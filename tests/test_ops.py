@@ -56,37 +56,62 @@ class TestBinCodes:
         assert int(ops.bin_codes(X, thr)[0, 0]) == 2
 
 
-class TestTreeHistMulti:
-    def _naive(self, codes, gidx, labels, bag, G, B, C, fs):
-        N, F = codes.shape
-        T = gidx.shape[1]
-        Fp = fs.shape[1] if fs is not None else F
-        out = np.zeros((G, Fp, B, C), dtype=np.float64)
+def naive_hist(codes, nodeid, lut, labels, bag, G, B, C, fs, targets=None):
+    """Independent python-loop oracle for K2 (classification & regression)."""
+    N, F = codes.shape
+    T = nodeid.shape[0]
+    Fp = fs.shape[1] if fs is not None else F
+    S = 3 if targets is not None else C
+    out = np.zeros((G, Fp, B, S), dtype=np.float64)
+    for t in range(T):
         for i in range(N):
-            for t in range(T):
-                g = int(gidx[i, t])
-                if g < 0:
-                    continue
-                w = float(bag[i, t]) if bag is not None else 1.0
-                if w == 0:
-                    continue
-                for fp in range(Fp):
-                    f = int(fs[g, fp]) if fs is not None else fp
-                    out[g, fp, int(codes[i, f]), int(labels[i])] += w
-        return out
+            nid = int(nodeid[t, i])
+            if nid < 0:
+                continue
+            g = int(lut[t, nid])
+            if g < 0:
+                continue
+            w = float(bag[t, i]) if bag is not None else 1.0
+            if w == 0:
+                continue
+            for fp in range(Fp):
+                f = int(fs[g, fp]) if fs is not None else fp
+                b = int(codes[i, f])
+                if targets is None:
+                    out[g, fp, b, int(labels[i])] += w
+                else:
+                    tv = float(targets[i])
+                    out[g, fp, b, 0] += w * tv
+                    out[g, fp, b, 1] += w * tv * tv
+                    out[g, fp, b, 2] += w
+    return out
 
+
+class TestTreeHistNodes:
     def test_bitwise_vs_naive(self):
         r = np.random.default_rng(3)
-        N, F, T, G, B, C = 80, 5, 3, 4, 8, 3
+        N, F, T, B, C = 80, 5, 3, 8, 3
+        M = 4  # nodes per tree
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8))
-        gidx = torch.from_numpy(r.integers(-1, G, (N, T)).astype(np.int32))
+        nodeid = torch.from_numpy(r.integers(-1, M, (T, N)).astype(np.int32))
+        # groups: tree-major contiguous, 2 active nodes per tree
+        lut = torch.full((T, M), -1, dtype=torch.int32)
+        g = 0
+        tree_lo = torch.zeros(T, dtype=torch.int32)
+        for t in range(T):
+            tree_lo[t] = g
+            for nid in sorted(r.choice(M, 2, replace=False)):
+                lut[t, nid] = g
+                g += 1
+        G = g
         labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8))
-        bag = torch.from_numpy(r.integers(0, 3, (N, T)).astype(np.float32))
+        bag = torch.from_numpy(r.integers(0, 3, (T, N)).astype(np.float32))
         fs = torch.from_numpy(
             np.stack([np.sort(r.choice(F, 3, replace=False)) for _ in range(G)]).astype(np.int32)
         )
-        out = ops.tree_hist_multi(codes, gidx, labels, bag, G, B, C, fs)
-        ref = self._naive(codes.numpy(), gidx.numpy(), labels.numpy(), bag.numpy(), G, B, C, fs.numpy())
+        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, labels, bag, G, 2, B, C, fs)
+        ref = naive_hist(codes.numpy(), nodeid.numpy(), lut.numpy(), labels.numpy(),
+                         bag.numpy(), G, B, C, fs.numpy())
         # integer counts must match exactly (SURVEY.md §5.2)
         assert np.array_equal(out.numpy(), ref.astype(np.float32))
 
@@ -94,42 +119,54 @@ class TestTreeHistMulti:
         r = np.random.default_rng(4)
         N, F, B, C = 50, 4, 6, 2
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8))
-        gidx = torch.zeros((N, 1), dtype=torch.int32)
+        nodeid = torch.zeros((1, N), dtype=torch.int32)
+        lut = torch.zeros((1, 1), dtype=torch.int32)
+        tree_lo = torch.zeros(1, dtype=torch.int32)
         labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8))
-        out = ops.tree_hist_multi(codes, gidx, labels, None, 1, B, C, None)
+        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, labels, None, 1, 1, B, C, None)
         assert float(out.sum()) == N * F
-        ref = self._naive(codes.numpy(), gidx.numpy(), labels.numpy(), None, 1, B, C, None)
+        ref = naive_hist(codes.numpy(), nodeid.numpy(), lut.numpy(), labels.numpy(),
+                         None, 1, B, C, None)
         assert np.array_equal(out.numpy(), ref.astype(np.float32))
 
     def test_regression_stats(self):
         r = np.random.default_rng(5)
         N, F, B = 40, 3, 4
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8))
-        gidx = torch.zeros((N, 1), dtype=torch.int32)
+        nodeid = torch.zeros((1, N), dtype=torch.int32)
+        lut = torch.zeros((1, 1), dtype=torch.int32)
+        tree_lo = torch.zeros(1, dtype=torch.int32)
         targets = torch.from_numpy(r.normal(size=N).astype(np.float32))
-        out = ops.tree_hist_multi(codes, gidx, targets, None, 1, B, 0, None, targets=targets)
+        labels = torch.zeros(N, dtype=torch.uint8)
+        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, labels, None, 1, 1, B, 0,
+                                  None, targets=targets)
         assert out.shape == (1, F, B, 3)
-        # per feature, sums across bins must equal global sums
         for f in range(F):
             assert float(out[0, f, :, 0].sum()) == pytest.approx(float(targets.sum()), rel=1e-4)
             assert float(out[0, f, :, 2].sum()) == N
+        ref = naive_hist(codes.numpy(), nodeid.numpy(), lut.numpy(), labels.numpy(),
+                         None, 1, B, 0, None, targets=targets.numpy())
+        assert np.allclose(out.numpy(), ref, atol=1e-5)
 
 
 class TestNodeAdvance:
-    def test_advance(self):
-        # tree: node0 splits on f0 at bin<=1 -> node1(leaf), node2 splits f1<=0 -> 3,4
+    def test_advance_multi(self):
+        # tree0: node0 splits on f0 at bin<=1 -> node1(leaf), node2 splits f1<=0 -> 3,4
+        # tree1: node0 is a leaf
         codes = torch.tensor([[0, 0], [2, 0], [2, 1]], dtype=torch.uint8)
-        nodeid = torch.zeros(3, dtype=torch.int32)
-        feat = torch.tensor([0, -1, 1, -1, -1], dtype=torch.int32)
-        thr = torch.tensor([1, 0, 0, 0, 0], dtype=torch.int32)
-        left = torch.tensor([1, -1, 3, -1, -1], dtype=torch.int32)
-        right = torch.tensor([2, -1, 4, -1, -1], dtype=torch.int32)
-        n1 = ops.node_advance(codes, nodeid, feat, thr, left, right)
-        assert n1.tolist() == [1, 2, 2]
-        n2 = ops.node_advance(codes, n1, feat, thr, left, right)
-        assert n2.tolist() == [-1, 3, 4]  # leaf -> -1; internal advances
-        n3 = ops.node_advance(codes, n2, feat, thr, left, right)
-        assert n3.tolist() == [-1, -1, -1]
+        nodeid = torch.zeros((2, 3), dtype=torch.int32)
+        feat = torch.tensor([0, -1, 1, -1, -1, -1], dtype=torch.int32)
+        thr = torch.tensor([1, 0, 0, 0, 0, 0], dtype=torch.int32)
+        left = torch.tensor([1, -1, 3, -1, -1, -1], dtype=torch.int32)
+        right = torch.tensor([2, -1, 4, -1, -1, -1], dtype=torch.int32)
+        ofs = torch.tensor([0, 5, 6], dtype=torch.int32)
+        n1 = ops.node_advance_multi(codes, nodeid, feat, thr, left, right, ofs)
+        assert n1[0].tolist() == [1, 2, 2]
+        assert n1[1].tolist() == [-1, -1, -1]  # leaf root -> done
+        n2 = ops.node_advance_multi(codes, n1, feat, thr, left, right, ofs)
+        assert n2[0].tolist() == [-1, 3, 4]
+        n3 = ops.node_advance_multi(codes, n2, feat, thr, left, right, ofs)
+        assert n3[0].tolist() == [-1, -1, -1]
 
 
 class TestTreePredict:
