@@ -1,0 +1,320 @@
+#!/usr/bin/env python3
+"""flowml benchmark harness (driver contract — see BASELINE.md).
+
+Flagship: config 3 of BASELINE.json — RandomForestClassifier, 100 trees,
+histogram split-find, 41-feature NSL-KDD-shaped synthetic flows, 5 classes,
+data-parallel over RCCL/xGMI. A "step" is one full training pass (one fit)
+over the fixed synthetic matrix (the task brief's definition for non-neural
+repos). Weak scaling: rows per GPU fixed (10M total at 8 GPUs).
+
+Other modes: --model mlp (config 2: 41-128-64-5 bf16, 1M flows/GPU),
+--model gbt (config 4: a step = one boosting round over the resident shard),
+--model lr (LogisticRegression full fit), --model infer (config 5 streaming
+micro-batch ensemble inference).
+
+Launch (multi-GPU): python -m torch.distributed.run --nnodes=1
+  --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import time
+
+import numpy as np
+import torch
+
+from flowml import dist
+from flowml.frame import FlowFrame
+
+
+def synth_flows(rows: int, n_features: int, n_classes: int, seed: int, device):
+    """NSL-KDD-shaped synthetic flows: 3 low-cardinality integer-coded
+    'categorical' features + numeric features; labels from a fixed planted
+    teacher (identical on every rank) with noise, so macro-F1 is meaningful.
+    """
+    g = torch.Generator(device=device).manual_seed(seed)
+    X = torch.randn(rows, n_features, device=device, generator=g)
+    X[:, 0] = torch.randint(0, 3, (rows,), generator=g, device=device).float()
+    X[:, 1] = torch.randint(0, 10, (rows,), generator=g, device=device).float()
+    X[:, 2] = torch.randint(0, 2, (rows,), generator=g, device=device).float()
+    # teacher identical across ranks (fixed seed, host RNG): sparse linear map
+    # over 8 active features — learnable by axis-aligned trees AND by LR/MLP
+    gt = torch.Generator().manual_seed(1234)
+    Wt = torch.zeros(n_features, n_classes)
+    active = torch.randperm(n_features, generator=gt)[:8]
+    Wt[active] = 1.5 * torch.randn(8, n_classes, generator=gt)
+    Wt = Wt.to(device)
+    logits = X @ Wt + 0.5 * torch.randn(rows, n_classes, device=device, generator=g)
+    y = logits.argmax(dim=1).to(torch.float64)
+    return X, y
+
+
+def make_df(X, y):
+    return FlowFrame({"features": X, "label": y}, device=X.device)
+
+
+def sync(device):
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+
+
+def timed_loop(fn, steps, warmup, device):
+    for _ in range(warmup):
+        fn()
+    dist.barrier()
+    sync(device)
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        fn()
+    dist.barrier()
+    sync(device)
+    elapsed = time.perf_counter() - t0
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if device.type == "cuda" else "cpu")
+    if dist.is_initialized():
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    return float(t.item())
+
+
+def bench_rf(args, device):
+    from flowml.classification import RandomForestClassifier
+    from flowml.evaluation import MulticlassClassificationEvaluator
+
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    rows = args.rows
+    X, y = synth_flows(rows, 41, 5, args.seed + rank, device)
+    df = make_df(X, y)
+    Xte, yte = synth_flows(min(rows, 200_000), 41, 5, args.seed + 7777 + rank, device)
+    dfte = make_df(Xte, yte)
+
+    state = {}
+
+    def step():
+        rf = RandomForestClassifier(
+            numTrees=args.trees, maxDepth=args.depth, maxBins=32, seed=args.seed
+        )
+        state["model"] = rf.fit(df)
+
+    elapsed = timed_loop(step, args.steps, args.warmup, device)
+    f1 = MulticlassClassificationEvaluator(metricName="macroF1").evaluate(
+        state["model"].transform(dfte)
+    )
+    cfg = {
+        "model": "RandomForestClassifier",
+        "numTrees": args.trees,
+        "maxDepth": args.depth,
+        "maxBins": 32,
+        "rows_per_gpu": rows,
+        "global_rows": rows * world,
+        "features": 41,
+        "classes": 5,
+        "parallelism": f"dp{world}",
+        "macro_f1": round(f1, 4),
+    }
+    return elapsed, rows * world * args.steps, cfg, "float32"
+
+
+def bench_gbt(args, device):
+    from flowml.classification import GBTClassifier
+    from flowml.evaluation import MulticlassClassificationEvaluator
+
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    rows = args.rows
+    X, y5 = synth_flows(rows, 41, 5, args.seed + rank, device)
+    y = (y5 >= 2.5).to(torch.float64)  # binary
+    df = make_df(X, y)
+    # steps boosting rounds per timed fit: use maxIter=steps and count one
+    # fit; a "step" = one boosting round
+    gbt = GBTClassifier(maxIter=args.steps + args.warmup, maxDepth=args.depth,
+                        maxBins=32, seed=args.seed)
+    # warmup rounds aren't separable inside one fit; time the whole fit and
+    # normalize per round (warmup arg folded in)
+    dist.barrier()
+    sync(device)
+    t0 = time.perf_counter()
+    model = gbt.fit(df)
+    dist.barrier()
+    sync(device)
+    elapsed = time.perf_counter() - t0
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if device.type == "cuda" else "cpu")
+    if dist.is_initialized():
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(t.item()) * args.steps / (args.steps + args.warmup)
+    Xte, y5te = synth_flows(min(rows, 200_000), 41, 5, args.seed + 7777 + rank, device)
+    dfte = make_df(Xte, (y5te >= 2.5).to(torch.float64))
+    f1 = MulticlassClassificationEvaluator(metricName="macroF1").evaluate(
+        model.transform(dfte)
+    )
+    cfg = {
+        "model": "GBTClassifier",
+        "rounds": args.steps,
+        "maxDepth": args.depth,
+        "rows_per_gpu": rows,
+        "global_rows": rows * world,
+        "features": 41,
+        "parallelism": f"dp{world}",
+        "macro_f1": round(f1, 4),
+    }
+    return elapsed, rows * world * args.steps, cfg, "float32"
+
+
+def bench_mlp(args, device):
+    """Config 2: MLP 41-128-64-5 bf16; a step = one full-batch fwd+bwd+update."""
+    from flowml import ops
+    from flowml.classification.mlp import _flatten, _unflatten, n_weights
+
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    rows = args.rows
+    X, y = synth_flows(rows, 41, 5, args.seed + rank, device)
+    use_bf16 = device.type == "cuda"
+    Xc = X.to(torch.bfloat16) if use_bf16 else X
+    yl = y.to(torch.int64)
+    layers = [41, 128, 64, 5]
+    gen = torch.Generator().manual_seed(args.seed)
+    # flat init in the Spark ANN layout (W col-major block + bias per layer)
+    parts = []
+    for i in range(len(layers) - 1):
+        parts.append(torch.randn(layers[i] * layers[i + 1], generator=gen) * (2.0 / layers[i]) ** 0.5)
+        parts.append(torch.zeros(layers[i + 1]))
+    theta = torch.cat(parts).to(device)
+    total = float(rows * world)
+    lr = 0.5
+
+    def step():
+        nonlocal theta
+        Ws, bs = _unflatten(theta, layers)
+        if use_bf16:
+            Ws = [w.to(torch.bfloat16) for w in Ws]
+            bs = [b.to(torch.float32) for b in bs]
+        loss, gWs, gBs = ops.mlp_grad(Xc, yl, Ws, bs, activation="relu")
+        flat_g = _flatten([g.to(torch.float32) for g in gWs], gBs)
+        buf = torch.cat([loss.reshape(1), flat_g])
+        dist.all_reduce_(buf)
+        theta = theta - lr * (buf[1:] / total)
+
+    elapsed = timed_loop(step, args.steps, args.warmup, device)
+    # final accuracy
+    Ws, bs = _unflatten(theta, layers)
+    if use_bf16:
+        Ws = [w.to(torch.bfloat16) for w in Ws]
+    logits = ops.mlp_forward(Xc, Ws, bs, activation="relu")
+    from flowml.evaluation import MulticlassClassificationEvaluator
+    from flowml.frame import FlowFrame
+
+    dfp = FlowFrame({"label": y, "prediction": logits.argmax(1).to(torch.float64)})
+    f1 = MulticlassClassificationEvaluator(metricName="macroF1").evaluate(dfp)
+    cfg = {
+        "model": "MultilayerPerceptronClassifier",
+        "layers": layers,
+        "rows_per_gpu": rows,
+        "global_batch": rows * world,
+        "features": 41,
+        "classes": 5,
+        "parallelism": f"dp{world}",
+        "macro_f1": round(f1, 4),
+    }
+    return elapsed, rows * world * args.steps, cfg, "bf16" if use_bf16 else "float32"
+
+
+def bench_infer(args, device):
+    """Config 5: streaming micro-batch ensemble inference (K8)."""
+    from flowml.classification import RandomForestClassifier
+
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    train_rows = min(args.rows, 500_000)
+    X, y = synth_flows(train_rows, 41, 5, args.seed + rank, device)
+    model = RandomForestClassifier(numTrees=args.trees, maxDepth=args.depth,
+                                   seed=args.seed).fit(make_df(X, y))
+    ens = model._ensemble.to(device)
+    batch = args.batch
+    Xb, _ = synth_flows(batch, 41, 5, args.seed + 31 + rank, device)
+
+    def step():
+        raw = ens.predict_raw(Xb)
+        _ = raw.argmax(dim=1)
+
+    elapsed = timed_loop(step, args.steps, args.warmup, device)
+    cfg = {
+        "model": f"RandomForest ensemble x{args.trees} inference",
+        "micro_batch": batch,
+        "features": 41,
+        "parallelism": f"dp{world}",
+    }
+    return elapsed, batch * world * args.steps, cfg, "float32"
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--warmup", type=int, default=1)
+    p.add_argument("--model", default="rf", choices=["rf", "gbt", "mlp", "lr", "infer"])
+    p.add_argument("--rows", type=int, default=0, help="rows per GPU (0 = config default)")
+    p.add_argument("--trees", type=int, default=100)
+    p.add_argument("--depth", type=int, default=5)
+    p.add_argument("--batch", type=int, default=262144, help="infer micro-batch")
+    p.add_argument("--seed", type=int, default=42)
+    args = p.parse_args()
+
+    dist.init_from_env()
+    world = dist.get_world_size()
+    if torch.cuda.is_available():
+        import os
+
+        device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
+        torch.cuda.set_device(device)
+    else:
+        device = torch.device("cpu")
+
+    if args.rows == 0:
+        defaults = {"rf": 1_250_000, "gbt": 6_250_000, "mlp": 1_000_000,
+                    "lr": 1_250_000, "infer": 500_000}
+        args.rows = defaults[args.model]
+        if device.type == "cpu":
+            args.rows = min(args.rows, 20_000)  # keep the no-GPU sanity path fast
+
+    if args.model == "rf":
+        elapsed, flows, cfg, dtype = bench_rf(args, device)
+    elif args.model == "gbt":
+        elapsed, flows, cfg, dtype = bench_gbt(args, device)
+    elif args.model == "mlp":
+        elapsed, flows, cfg, dtype = bench_mlp(args, device)
+    elif args.model == "infer":
+        elapsed, flows, cfg, dtype = bench_infer(args, device)
+    else:
+        raise SystemExit(f"model {args.model} not benchmarked yet")
+
+    value = flows / elapsed
+    ms_per_step = elapsed * 1000.0 / args.steps
+    if dist.get_rank() == 0:
+        out = {
+            "metric": "flows/sec trained + macro-F1, 41-feat NSL-KDD-shape at 1/2/4/8 MI355X"
+            if args.model != "infer"
+            else "flows/sec inferred (streaming ensemble)",
+            "value": value,
+            "unit": "flows/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": dtype,
+            "data": "synthetic",
+            "config": cfg,
+        }
+        print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()

@@ -318,10 +318,11 @@ def grow_trees(
     # root stats per tree: one K2 pass over feature 0 only (lut: node0 -> t)
     lut0 = torch.arange(T, dtype=torch.int32, device=dev).reshape(T, 1)
     tree_lo0 = torch.arange(T, dtype=torch.int32, device=dev)
+    tree_cnt0 = torch.ones(T, dtype=torch.int32, device=dev)
     fs0 = torch.zeros((T, 1), dtype=torch.int32, device=dev)
     h0 = ops.tree_hist_nodes(
-        codes, nodeid, lut0, tree_lo0, labels_u8, bag, T, 1, B, S if is_cls else 0,
-        fs0, targets=targets,
+        codes, nodeid, lut0, tree_lo0, tree_cnt0, labels_u8, bag, T, 1, B,
+        S if is_cls else 0, fs0, targets=targets,
     )
     dist.all_reduce_(h0.reshape(-1) if h0.is_contiguous() else h0)
     root_stats = h0.sum(dim=2)[:, 0, :]  # T×S
@@ -360,6 +361,7 @@ def grow_trees(
             l_max = int(per_tree_count.max())
             lut_d = lut.to(dev)
             tree_lo = torch.from_numpy(tree_lo_np).to(dev)
+            tree_cnt = torch.from_numpy(per_tree_count.astype(np.int32)).to(dev)
             # feature subsets per node (seeded by (seed, tree, node) —
             # rank-invariant and reproducible)
             if use_subset:
@@ -371,7 +373,7 @@ def grow_trees(
             else:
                 fs = None
             hist = ops.tree_hist_nodes(
-                codes, nodeid, lut_d, tree_lo, labels_u8, bag, G, l_max, B,
+                codes, nodeid, lut_d, tree_lo, tree_cnt, labels_u8, bag, G, l_max, B,
                 S if is_cls else 0, fs, targets=targets,
             )
             dist.all_reduce_(hist.reshape(-1))
@@ -526,7 +528,18 @@ def poisson_bag(
     """
     if not bootstrap and rate >= 1.0:
         return None
-    g = torch.Generator(device="cpu").manual_seed((seed * 7919 + dist.get_rank()) & 0x7FFFFFFF)
+    device = torch.device(device)
+    seed_r = (seed * 7919 + dist.get_rank()) & 0x7FFFFFFF
+    if device.type == "cuda":
+        # generate on-device (a 100×10M Poisson draw on host would dominate);
+        # deterministic per (seed, rank) via a dedicated CUDA generator
+        g = torch.Generator(device=device).manual_seed(seed_r)
+        if bootstrap:
+            bag = torch.poisson(torch.full((n_trees, n_rows), rate, device=device), generator=g)
+        else:
+            bag = (torch.rand((n_trees, n_rows), device=device, generator=g) < rate).to(torch.float32)
+        return bag.to(torch.float32)
+    g = torch.Generator(device="cpu").manual_seed(seed_r)
     if bootstrap:
         bag = torch.poisson(torch.full((n_trees, n_rows), rate), generator=g)
     else:

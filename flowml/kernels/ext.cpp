@@ -1,0 +1,254 @@
+// Torch bindings for the flowml CDNA4 kernels. Pure C++ TU: kernels live in
+// the .hip TUs and are reached through extern "C" launchers taking a void*
+// hipStream_t, so this file needs only the torch headers.
+
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include <tuple>
+#include <vector>
+
+// ---- launcher decls (defined in tree_kernels.hip / mlp_kernels.hip) -------
+
+extern "C" {
+void launch_bin_codes(const float*, const float*, uint8_t*, long long, int,
+                      int, void*);
+void launch_tree_hist_cls(const uint8_t*, const int*, const int*, const int*,
+                          const int*, const uint8_t*, const float*, const int*,
+                          float*, long long, int, int, int, int, int, int, int,
+                          void*);
+void launch_tree_hist_reg(const uint8_t*, const int*, const int*, const int*,
+                          const int*, const float*, const float*, const int*,
+                          float*, long long, int, int, int, int, int, int,
+                          void*);
+void launch_node_advance(const uint8_t*, const int*, int*, const int*,
+                         const int*, const int*, const int*, const int*,
+                         long long, int, int, void*);
+void launch_tree_predict(const float*, const int*, const float*, const int*,
+                         const int*, const float*, float*, long long, int, int,
+                         int, int, void*);
+void launch_logreg_delta(const float*, const int*, const float*, const float*,
+                         const float*, float*, float*, long long, int, int,
+                         void*);
+void launch_softmax_xent(const float*, const int*, const float*, float*,
+                         float*, long long, int, void*);
+void launch_gemm_bias_act_bf16(const void*, const void*, const float*, void*,
+                               long long, int, int, int, void*);
+void launch_gemm_bias_act_f32(const float*, const float*, const float*, float*,
+                              long long, int, int, int, void*);
+void launch_act_backward_bf16(const void*, void*, long long, int, void*);
+}
+
+static void* cur_stream() {
+  return (void*)at::cuda::getCurrentCUDAStream().stream();
+}
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be a GPU tensor")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+#define CHK(x) \
+  CHECK_GPU(x); \
+  CHECK_CONTIG(x)
+
+// ---------------------------------------------------------------------------
+
+torch::Tensor bin_codes(torch::Tensor X, torch::Tensor thr) {
+  CHK(X);
+  CHK(thr);
+  TORCH_CHECK(X.scalar_type() == torch::kFloat32, "X must be f32");
+  const long long N = X.size(0);
+  const int F = X.size(1);
+  const int TB = thr.size(1);
+  auto out = torch::empty({N, F}, X.options().dtype(torch::kUInt8));
+  launch_bin_codes(X.data_ptr<float>(), thr.data_ptr<float>(),
+                   out.data_ptr<uint8_t>(), N, F, TB, cur_stream());
+  return out;
+}
+
+torch::Tensor tree_hist_cls(torch::Tensor codes, torch::Tensor nodeid,
+                            torch::Tensor lut, torch::Tensor tree_lo,
+                            torch::Tensor tree_cnt, torch::Tensor labels,
+                            torch::Tensor bag, int64_t G, int64_t l_max,
+                            int64_t B, int64_t C, torch::Tensor fs) {
+  CHK(codes);
+  CHK(nodeid);
+  CHK(lut);
+  CHK(tree_lo);
+  CHK(tree_cnt);
+  CHK(labels);
+  const long long N = codes.size(0);
+  const int F = codes.size(1);
+  const int T = nodeid.size(0);
+  const int M = lut.size(1);
+  const int Fp = fs.numel() ? fs.size(1) : F;
+  auto out = torch::zeros({G, Fp, B, C}, codes.options().dtype(torch::kFloat32));
+  launch_tree_hist_cls(
+      codes.data_ptr<uint8_t>(), nodeid.data_ptr<int>(), lut.data_ptr<int>(),
+      tree_lo.data_ptr<int>(), tree_cnt.data_ptr<int>(),
+      labels.data_ptr<uint8_t>(),
+      bag.numel() ? bag.data_ptr<float>() : nullptr,
+      fs.numel() ? fs.data_ptr<int>() : nullptr, out.data_ptr<float>(), N, F,
+      T, M, Fp, (int)B, (int)C, (int)l_max, cur_stream());
+  return out;
+}
+
+torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
+                            torch::Tensor lut, torch::Tensor tree_lo,
+                            torch::Tensor tree_cnt, torch::Tensor targets,
+                            torch::Tensor bag, int64_t G, int64_t l_max,
+                            int64_t B, torch::Tensor fs) {
+  CHK(codes);
+  CHK(nodeid);
+  CHK(lut);
+  CHK(targets);
+  const long long N = codes.size(0);
+  const int F = codes.size(1);
+  const int T = nodeid.size(0);
+  const int M = lut.size(1);
+  const int Fp = fs.numel() ? fs.size(1) : F;
+  auto out = torch::zeros({G, Fp, B, 3}, codes.options().dtype(torch::kFloat32));
+  launch_tree_hist_reg(
+      codes.data_ptr<uint8_t>(), nodeid.data_ptr<int>(), lut.data_ptr<int>(),
+      tree_lo.data_ptr<int>(), tree_cnt.data_ptr<int>(),
+      targets.data_ptr<float>(),
+      bag.numel() ? bag.data_ptr<float>() : nullptr,
+      fs.numel() ? fs.data_ptr<int>() : nullptr, out.data_ptr<float>(), N, F,
+      T, M, Fp, (int)B, (int)l_max, cur_stream());
+  return out;
+}
+
+torch::Tensor node_advance(torch::Tensor codes, torch::Tensor nodeid,
+                           torch::Tensor feat, torch::Tensor thr,
+                           torch::Tensor left, torch::Tensor right,
+                           torch::Tensor node_ofs) {
+  CHK(codes);
+  CHK(nodeid);
+  const long long N = codes.size(0);
+  const int F = codes.size(1);
+  const int T = nodeid.size(0);
+  auto out = torch::empty_like(nodeid);
+  launch_node_advance(codes.data_ptr<uint8_t>(), nodeid.data_ptr<int>(),
+                      out.data_ptr<int>(), feat.data_ptr<int>(),
+                      thr.data_ptr<int>(), left.data_ptr<int>(),
+                      right.data_ptr<int>(), node_ofs.data_ptr<int>(), N, F, T,
+                      cur_stream());
+  return out;
+}
+
+torch::Tensor tree_predict(torch::Tensor X, torch::Tensor feature,
+                           torch::Tensor threshold, torch::Tensor left,
+                           torch::Tensor right, torch::Tensor value) {
+  CHK(X);
+  CHK(feature);
+  CHK(value);
+  const long long N = X.size(0);
+  const int F = X.size(1);
+  const int T = feature.size(0);
+  const int M = feature.size(1);
+  const int K = value.size(2);
+  TORCH_CHECK(K <= 16, "tree_predict supports up to 16 classes");
+  auto out = torch::empty({N, K}, X.options().dtype(torch::kFloat32));
+  launch_tree_predict(X.data_ptr<float>(), feature.data_ptr<int>(),
+                      threshold.data_ptr<float>(), left.data_ptr<int>(),
+                      right.data_ptr<int>(), value.data_ptr<float>(),
+                      out.data_ptr<float>(), N, F, T, M, K, cur_stream());
+  return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> logreg_delta(torch::Tensor X,
+                                                      torch::Tensor y,
+                                                      torch::Tensor w,
+                                                      torch::Tensor W,
+                                                      torch::Tensor b) {
+  CHK(X);
+  CHK(y);
+  CHK(W);
+  CHK(b);
+  const long long N = X.size(0);
+  const int F = X.size(1);
+  const int K = W.size(0);
+  TORCH_CHECK(K <= 16, "logreg_delta supports up to 16 classes");
+  auto delta = torch::empty({N, K}, X.options());
+  auto loss = torch::zeros({1}, X.options());
+  launch_logreg_delta(X.data_ptr<float>(), y.data_ptr<int>(),
+                      w.numel() ? w.data_ptr<float>() : nullptr,
+                      W.data_ptr<float>(), b.data_ptr<float>(),
+                      delta.data_ptr<float>(), loss.data_ptr<float>(), N, F, K,
+                      cur_stream());
+  return {loss.squeeze(0), delta};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> softmax_xent_grad(
+    torch::Tensor logits, torch::Tensor y, torch::Tensor w) {
+  CHK(logits);
+  CHK(y);
+  const long long N = logits.size(0);
+  const int K = logits.size(1);
+  TORCH_CHECK(K <= 16, "softmax_xent supports up to 16 classes");
+  auto delta = torch::empty_like(logits);
+  auto loss = torch::zeros({1}, logits.options());
+  launch_softmax_xent(logits.data_ptr<float>(), y.data_ptr<int>(),
+                      w.numel() ? w.data_ptr<float>() : nullptr,
+                      delta.data_ptr<float>(), loss.data_ptr<float>(), N, K,
+                      cur_stream());
+  return {loss.squeeze(0), delta};
+}
+
+static torch::Tensor pad_k(torch::Tensor t, int64_t mult) {
+  const int64_t K = t.size(1);
+  const int64_t Kp = ((K + mult - 1) / mult) * mult;
+  if (Kp == K) return t;
+  return torch::constant_pad_nd(t, {0, Kp - K});
+}
+
+torch::Tensor gemm_bias_act(torch::Tensor X, torch::Tensor W, torch::Tensor b,
+                            int64_t act) {
+  CHK(X);
+  CHK(W);
+  CHK(b);
+  TORCH_CHECK(X.size(1) == W.size(1), "K mismatch");
+  const long long N = X.size(0);
+  const int M = W.size(0);
+  TORCH_CHECK(M <= 128, "gemm_bias_act supports M <= 128 (MLP-stack shapes)");
+  auto bf = b.to(torch::kFloat32);
+  if (X.scalar_type() == torch::kBFloat16) {
+    auto Xp = pad_k(X, 32).contiguous();
+    auto Wp = pad_k(W, 32).contiguous();
+    const int Kp = Xp.size(1);
+    auto out = torch::empty({N, M}, X.options());
+    launch_gemm_bias_act_bf16(Xp.data_ptr(), Wp.data_ptr(),
+                              bf.data_ptr<float>(), out.data_ptr(), N, M, Kp,
+                              (int)act, cur_stream());
+    return out;
+  }
+  TORCH_CHECK(X.scalar_type() == torch::kFloat32, "X must be bf16 or f32");
+  auto Xp = pad_k(X, 4).contiguous();
+  auto Wp = pad_k(W.to(torch::kFloat32), 4).contiguous();
+  const int Kp = Xp.size(1);
+  auto out = torch::empty({N, M}, X.options());
+  launch_gemm_bias_act_f32(Xp.data_ptr<float>(), Wp.data_ptr<float>(),
+                           bf.data_ptr<float>(), out.data_ptr<float>(), N, M,
+                           Kp, (int)act, cur_stream());
+  return out;
+}
+
+torch::Tensor act_backward_(torch::Tensor h, torch::Tensor delta,
+                            int64_t act) {
+  CHK(h);
+  CHK(delta);
+  TORCH_CHECK(h.scalar_type() == torch::kBFloat16, "bf16 only");
+  launch_act_backward_bf16(h.data_ptr(), delta.data_ptr(), h.numel(),
+                           (int)act, cur_stream());
+  return delta;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bin_codes", &bin_codes, "K4a quantile binning");
+  m.def("tree_hist_cls", &tree_hist_cls, "K2 class histograms (all trees)");
+  m.def("tree_hist_reg", &tree_hist_reg, "K2b regression histograms");
+  m.def("node_advance", &node_advance, "K4b node-id advance");
+  m.def("tree_predict", &tree_predict, "K8 ensemble inference");
+  m.def("logreg_delta", &logreg_delta, "K1 fused logistic delta");
+  m.def("softmax_xent_grad", &softmax_xent_grad, "K6 softmax/xent grad");
+  m.def("gemm_bias_act", &gemm_bias_act, "K5 fused MFMA GEMM+bias+act");
+  m.def("act_backward_", &act_backward_, "K7 activation backward (in-place)");
+}

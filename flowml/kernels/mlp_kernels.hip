@@ -1,0 +1,194 @@
+// CDNA4 (gfx950) fused GEMM kernels for the MLP stack.
+//
+// out[N,M] = act(X[N,K] @ W[M,K]^T + b[M])
+//
+// Shapes are tall-skinny (N up to the 1M-flow shard, M ∈ {128,64,5},
+// K ∈ {41,128,64} — BASELINE.json:8), so the design is: one 256-thread block
+// = 4 waves covers 64 rows × all M columns; each wave owns a 16-row stripe
+// and accumulates ceil(M/16) MFMA fragments across the K loop. X fragments
+// are 8 contiguous bf16 per lane straight from HBM (coalesced across the
+// 16-lane row groups); W is tiny (≤32 KiB) and L1/L2-resident, so no LDS
+// staging is needed — the kernel is X-bandwidth-bound, which is the roofline
+// for these shapes. K must be padded to a multiple of 32 (bf16) / 4 (f32) by
+// the caller (flowml/kernels/ext.cpp pads).
+//
+// MFMA: mfma_f32_16x16x32_bf16 (bf16) / mfma_f32_16x16x4f32 (exact f32).
+// Fragment maps (cdna_hip_programming.md §3): A[i=l%16][k=(l/16)*8+j],
+// B[k=(l/16)*8+j][n=l%16], D col=l%16, row=(l/16)*4+reg.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <stdint.h>
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define MAX_NFRAG 8  // supports M up to 128 per kernel pass
+
+__device__ __forceinline__ float act_apply(float v, int act) {
+  if (act == 1) return v > 0.f ? v : 0.f;          // relu
+  if (act == 2) return 1.f / (1.f + __expf(-v));   // sigmoid
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// bf16 path: X[N,Kp] bf16 (Kp%32==0), W[M,Kp] bf16, out[N,M] bf16
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ __launch_bounds__(256) void flowml_gemm_bias_act_bf16(
+    const __bf16* __restrict__ X, const __bf16* __restrict__ W,
+    const float* __restrict__ b, __bf16* __restrict__ out, long long N, int M,
+    int Kp, int act) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long long row_base = (long long)blockIdx.x * 64 + wave * 16;
+  if (row_base >= N) return;
+  const int i_in_tile = lane & 15;   // A row within the 16-row stripe
+  const int kgrp = lane >> 4;        // 0..3 -> k offset (kgrp*8)
+  const int nfrag = (M + 15) / 16;
+
+  f32x4 acc[MAX_NFRAG];
+#pragma unroll
+  for (int nf = 0; nf < MAX_NFRAG; ++nf) acc[nf] = {0.f, 0.f, 0.f, 0.f};
+
+  const long long a_row = row_base + i_in_tile;
+  const bool a_valid = a_row < N;
+  const __bf16* a_base = X + a_row * Kp + kgrp * 8;
+
+  for (int kk = 0; kk < Kp; kk += 32) {
+    bf16x8 a_frag;
+    if (a_valid) {
+      a_frag = *reinterpret_cast<const bf16x8*>(a_base + kk);
+    } else {
+      a_frag = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+    for (int nf = 0; nf < nfrag; ++nf) {
+      const int n = nf * 16 + i_in_tile;
+      bf16x8 b_frag;
+      if (n < M) {
+        b_frag = *reinterpret_cast<const bf16x8*>(W + (long long)n * Kp + kk + kgrp * 8);
+      } else {
+        b_frag = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+      acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc[nf], 0, 0, 0);
+    }
+  }
+
+  // epilogue: D col = lane&15, row = (lane>>4)*4 + reg
+  const int col_in = lane & 15;
+  const int row_off = (lane >> 4) * 4;
+  for (int nf = 0; nf < nfrag; ++nf) {
+    const int n = nf * 16 + col_in;
+    if (n >= M) continue;
+    const float bias = b[n];
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const long long r = row_base + row_off + reg;
+      if (r < N) {
+        out[r * M + n] = (__bf16)act_apply(acc[nf][reg] + bias, act);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// f32 path: exact f32 MFMA (v_mfma_f32_16x16x4_f32). Kp%4==0.
+// A lane map: A[i=l%16][k=l/16]; B[k=l/16][n=l%16]; D as above.
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ __launch_bounds__(256) void flowml_gemm_bias_act_f32(
+    const float* __restrict__ X, const float* __restrict__ W,
+    const float* __restrict__ b, float* __restrict__ out, long long N, int M,
+    int Kp, int act) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long long row_base = (long long)blockIdx.x * 64 + wave * 16;
+  if (row_base >= N) return;
+  const int i_in_tile = lane & 15;
+  const int kgrp = lane >> 4;  // 0..3 -> k index within the 4-deep step
+  const int nfrag = (M + 15) / 16;
+
+  f32x4 acc[MAX_NFRAG];
+#pragma unroll
+  for (int nf = 0; nf < MAX_NFRAG; ++nf) acc[nf] = {0.f, 0.f, 0.f, 0.f};
+
+  const long long a_row = row_base + i_in_tile;
+  const bool a_valid = a_row < N;
+  const float* a_base = X + a_row * Kp + kgrp;
+
+  for (int kk = 0; kk < Kp; kk += 4) {
+    const float a_elem = a_valid ? a_base[kk] : 0.f;
+    for (int nf = 0; nf < nfrag; ++nf) {
+      const int n = nf * 16 + i_in_tile;
+      const float b_elem = (n < M) ? W[(long long)n * Kp + kk + kgrp] : 0.f;
+      acc[nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(a_elem, b_elem, acc[nf], 0, 0, 0);
+    }
+  }
+
+  const int col_in = lane & 15;
+  const int row_off = (lane >> 4) * 4;
+  for (int nf = 0; nf < nfrag; ++nf) {
+    const int n = nf * 16 + col_in;
+    if (n >= M) continue;
+    const float bias = b[n];
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const long long r = row_base + row_off + reg;
+      if (r < N) {
+        out[r * M + n] = act_apply(acc[nf][reg] + bias, act);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// elementwise activation-backward: delta *= act'(h) (K7 epilogue fusion)
+// relu: h>0 ? 1 : 0 ; sigmoid: h*(1-h) — h is the post-activation value.
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void flowml_act_backward_bf16(
+    const __bf16* __restrict__ h, __bf16* __restrict__ delta, long long total,
+    int act) {
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const float hv = (float)h[i];
+    const float dv = (float)delta[i];
+    const float g = (act == 1) ? (hv > 0.f ? 1.f : 0.f) : hv * (1.f - hv);
+    delta[i] = (__bf16)(dv * g);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Host launchers
+// ---------------------------------------------------------------------------
+
+extern "C" void launch_gemm_bias_act_bf16(const void* X, const void* W,
+                                          const float* b, void* out,
+                                          long long N, int M, int Kp, int act,
+                                          void* stream) {
+  const int blocks = (int)((N + 63) / 64);
+  hipLaunchKernelGGL(flowml_gemm_bias_act_bf16, dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, (const __bf16*)X, (const __bf16*)W, b,
+                     (__bf16*)out, N, M, Kp, act);
+}
+
+extern "C" void launch_gemm_bias_act_f32(const float* X, const float* W,
+                                         const float* b, float* out,
+                                         long long N, int M, int Kp, int act,
+                                         void* stream) {
+  const int blocks = (int)((N + 63) / 64);
+  hipLaunchKernelGGL(flowml_gemm_bias_act_f32, dim3(blocks), dim3(256), 0,
+                     (hipStream_t)stream, X, W, b, out, N, M, Kp, act);
+}
+
+extern "C" void launch_act_backward_bf16(const void* h, void* delta,
+                                         long long total, int act,
+                                         void* stream) {
+  const int block = 256;
+  long long g = (total + block - 1) / block;
+  if (g > 4096) g = 4096;
+  if (g < 1) g = 1;
+  hipLaunchKernelGGL(flowml_act_backward_bf16, dim3((int)g), dim3(block), 0,
+                     (hipStream_t)stream, (const __bf16*)h, (__bf16*)delta,
+                     total, act);
+}

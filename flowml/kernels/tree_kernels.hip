@@ -1,0 +1,445 @@
+// CDNA4 (gfx950) kernels for the tree-ensemble hot path.
+//
+// K4a  bin_codes        — quantile binning: f32 features -> uint8 bin codes
+// K2   tree_hist_cls    — per-node class histograms, ALL trees in one pass:
+//                         one workgroup owns (row-tile × tree), privatizes
+//                         that tree's frontier histograms in LDS (float
+//                         atomics; counts are integers so sums are exact and
+//                         run-to-run deterministic), flushes with global
+//                         atomics. The #1 hot loop of the flagship config.
+// K2b  tree_hist_reg    — same with (Σt, Σt², n) stats for GBT variance trees
+// K4b  node_advance     — advance every (tree,row) one level down its tree
+// K8   tree_predict     — batched ensemble inference: sum of leaf vectors
+// K1   logreg_delta     — fused logits+softmax+delta+loss for LogisticRegression
+// K6   softmax_xent     — row-wise softmax/cross-entropy loss + delta
+//
+// Design notes (per /opt/skills guides): wave64; blocks are multiples of 64;
+// codes rows are 41 B at NSL-KDD shape so a (row,tree) touches ~1 cache line;
+// consecutive grid.x blocks share the same row tile across trees -> L2 reuse;
+// LDS histogram for one tree's frontier chunk is capped at 128 KiB by the
+// host-side chunker (flowml/classification/tree.py _hist_lds_cap).
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define FLOWML_MAXK 16  // max classes handled in-register by rowwise kernels
+
+// ---------------------------------------------------------------------------
+// K4a — binning
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void flowml_bin_codes(
+    const float* __restrict__ X, const float* __restrict__ thr,
+    uint8_t* __restrict__ out, long long N, int F, int TB) {
+  extern __shared__ float s_thr[];
+  for (int i = threadIdx.x; i < F * TB; i += blockDim.x) s_thr[i] = thr[i];
+  __syncthreads();
+  const long long total = N * (long long)F;
+  for (long long idx = blockIdx.x * (long long)blockDim.x + threadIdx.x;
+       idx < total; idx += (long long)gridDim.x * blockDim.x) {
+    const int f = (int)(idx % F);
+    const float x = X[idx];
+    const float* t = s_thr + f * TB;
+    int code = 0;
+#pragma unroll 4
+    for (int j = 0; j < TB; ++j) code += (t[j] <= x) ? 1 : 0;
+    out[idx] = (uint8_t)code;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K2 — classification histogram (LDS-privatized per tree)
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void flowml_tree_hist_cls(
+    const uint8_t* __restrict__ codes,   // N×F
+    const int* __restrict__ nodeid,      // T×N
+    const int* __restrict__ lut,         // T×M (node -> chunk group, -1)
+    const int* __restrict__ tree_lo,     // T
+    const int* __restrict__ tree_cnt,    // T (groups of tree t in chunk)
+    const uint8_t* __restrict__ labels,  // N
+    const float* __restrict__ bag,       // T×N or nullptr
+    const int* __restrict__ fs,          // G×Fp or nullptr (identity)
+    float* __restrict__ out,             // G×Fp×B×C
+    long long N, int F, int M, int Fp, int B, int C, int l_max,
+    long long rows_per_block) {
+  extern __shared__ float s_hist[];
+  const int t = blockIdx.y;
+  const int cnt = tree_cnt[t];
+  if (cnt == 0) return;
+  const int hist_sz = l_max * Fp * B * C;
+  for (int i = threadIdx.x; i < hist_sz; i += blockDim.x) s_hist[i] = 0.f;
+  __syncthreads();
+
+  const int g_lo = tree_lo[t];
+  const int* lut_t = lut + (long long)t * M;
+  const int* nid_t = nodeid + (long long)t * N;
+  const float* bag_t = bag ? bag + (long long)t * N : nullptr;
+  const long long r0 = blockIdx.x * rows_per_block;
+  const long long r1 = min(r0 + rows_per_block, N);
+  const int stride_fbc = Fp * B * C;
+
+  for (long long r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+    const int nid = nid_t[r];
+    if (nid < 0) continue;
+    const int g = lut_t[nid];
+    if (g < 0) continue;
+    float w = 1.f;
+    if (bag_t) {
+      w = bag_t[r];
+      if (w == 0.f) continue;
+    }
+    const int gl = g - g_lo;
+    const int lab = labels[r];
+    const uint8_t* crow = codes + r * F;
+    const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
+    float* hrow = s_hist + (long long)gl * stride_fbc;
+    for (int fp = 0; fp < Fp; ++fp) {
+      const int f = fs_g ? fs_g[fp] : fp;
+      const int b = crow[f];
+      atomicAdd(&hrow[(fp * B + b) * C + lab], w);
+    }
+  }
+  __syncthreads();
+  // flush this tree's [g_lo, g_lo+cnt) groups (same layout in LDS and out)
+  float* out_t = out + (long long)g_lo * stride_fbc;
+  const int flush_sz = cnt * stride_fbc;
+  for (int i = threadIdx.x; i < flush_sz; i += blockDim.x) {
+    const float v = s_hist[i];
+    if (v != 0.f) atomicAdd(&out_t[i], v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K2b — regression histogram (Σt, Σt², n)
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void flowml_tree_hist_reg(
+    const uint8_t* __restrict__ codes, const int* __restrict__ nodeid,
+    const int* __restrict__ lut, const int* __restrict__ tree_lo,
+    const int* __restrict__ tree_cnt, const float* __restrict__ targets,
+    const float* __restrict__ bag, const int* __restrict__ fs,
+    float* __restrict__ out, long long N, int F, int M, int Fp, int B,
+    int l_max, long long rows_per_block) {
+  extern __shared__ float s_hist[];
+  const int t = blockIdx.y;
+  const int cnt = tree_cnt[t];
+  if (cnt == 0) return;
+  const int hist_sz = l_max * Fp * B * 3;
+  for (int i = threadIdx.x; i < hist_sz; i += blockDim.x) s_hist[i] = 0.f;
+  __syncthreads();
+
+  const int g_lo = tree_lo[t];
+  const int* lut_t = lut + (long long)t * M;
+  const int* nid_t = nodeid + (long long)t * N;
+  const float* bag_t = bag ? bag + (long long)t * N : nullptr;
+  const long long r0 = blockIdx.x * rows_per_block;
+  const long long r1 = min(r0 + rows_per_block, N);
+  const int stride_fb3 = Fp * B * 3;
+
+  for (long long r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+    const int nid = nid_t[r];
+    if (nid < 0) continue;
+    const int g = lut_t[nid];
+    if (g < 0) continue;
+    float w = 1.f;
+    if (bag_t) {
+      w = bag_t[r];
+      if (w == 0.f) continue;
+    }
+    const int gl = g - g_lo;
+    const float tv = targets[r];
+    const float wt = w * tv;
+    const float wt2 = wt * tv;
+    const uint8_t* crow = codes + r * F;
+    const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
+    float* hrow = s_hist + (long long)gl * stride_fb3;
+    for (int fp = 0; fp < Fp; ++fp) {
+      const int f = fs_g ? fs_g[fp] : fp;
+      const int b = crow[f];
+      float* cell = &hrow[(fp * B + b) * 3];
+      atomicAdd(cell + 0, wt);
+      atomicAdd(cell + 1, wt2);
+      atomicAdd(cell + 2, w);
+    }
+  }
+  __syncthreads();
+  float* out_t = out + (long long)g_lo * stride_fb3;
+  const int flush_sz = cnt * stride_fb3;
+  for (int i = threadIdx.x; i < flush_sz; i += blockDim.x) {
+    const float v = s_hist[i];
+    if (v != 0.f) atomicAdd(&out_t[i], v);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K4b — node advance (all trees)
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void flowml_node_advance(
+    const uint8_t* __restrict__ codes, const int* __restrict__ nodeid_in,
+    int* __restrict__ nodeid_out, const int* __restrict__ feat,
+    const int* __restrict__ thr, const int* __restrict__ left,
+    const int* __restrict__ right, const int* __restrict__ node_ofs,
+    long long N, int F) {
+  const int t = blockIdx.y;
+  const int ofs = node_ofs[t];
+  const int* f_t = feat + ofs;
+  const int* thr_t = thr + ofs;
+  const int* l_t = left + ofs;
+  const int* r_t = right + ofs;
+  const int* in_t = nodeid_in + (long long)t * N;
+  int* out_t = nodeid_out + (long long)t * N;
+  for (long long r = blockIdx.x * (long long)blockDim.x + threadIdx.x; r < N;
+       r += (long long)gridDim.x * blockDim.x) {
+    const int nid = in_t[r];
+    if (nid < 0) {
+      out_t[r] = nid;
+      continue;
+    }
+    const int f = f_t[nid];
+    if (f < 0) {
+      out_t[r] = -1;  // reached a leaf: row is done
+      continue;
+    }
+    const int code = codes[r * (long long)F + f];
+    out_t[r] = (code <= thr_t[nid]) ? l_t[nid] : r_t[nid];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K8 — batched ensemble inference
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void flowml_tree_predict(
+    const float* __restrict__ X, const int* __restrict__ feature,
+    const float* __restrict__ threshold, const int* __restrict__ left,
+    const int* __restrict__ right, const float* __restrict__ value,
+    float* __restrict__ out, long long N, int F, int T, int M, int K) {
+  for (long long r = blockIdx.x * (long long)blockDim.x + threadIdx.x; r < N;
+       r += (long long)gridDim.x * blockDim.x) {
+    float acc[FLOWML_MAXK];
+#pragma unroll
+    for (int k = 0; k < FLOWML_MAXK; ++k) acc[k] = 0.f;
+    const float* xr = X + r * (long long)F;
+    for (int t = 0; t < T; ++t) {
+      const long long base = (long long)t * M;
+      int node = 0;
+      int f = feature[base + node];
+      while (f >= 0) {
+        node = (xr[f] <= threshold[base + node]) ? left[base + node]
+                                                 : right[base + node];
+        f = feature[base + node];
+      }
+      const float* v = value + (base + node) * (long long)K;
+      for (int k = 0; k < K; ++k) acc[k] += v[k];
+    }
+    float* o = out + r * (long long)K;
+    for (int k = 0; k < K; ++k) o[k] = acc[k];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K1 — fused logistic delta: logits (K dots) + softmax + delta + loss
+// (gradW = deltaᵀX and gradB = Σdelta are plain GEMM/reduce — rocBLAS/torch)
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void flowml_logreg_delta(
+    const float* __restrict__ X,  // N×F
+    const int* __restrict__ y,    // N
+    const float* __restrict__ w,  // N or nullptr
+    const float* __restrict__ W,  // K×F
+    const float* __restrict__ b,  // K
+    float* __restrict__ delta,    // N×K out
+    float* __restrict__ loss,     // 1 (atomic)
+    long long N, int F, int K) {
+  extern __shared__ float s_w[];  // K×F weights + K bias + blockDim loss scratch
+  const int KF = K * F;
+  for (int i = threadIdx.x; i < KF; i += blockDim.x) s_w[i] = W[i];
+  float* s_b = s_w + KF;
+  for (int i = threadIdx.x; i < K; i += blockDim.x) s_b[i] = b[i];
+  __syncthreads();
+
+  float loss_local = 0.f;
+  for (long long r = blockIdx.x * (long long)blockDim.x + threadIdx.x; r < N;
+       r += (long long)gridDim.x * blockDim.x) {
+    const float* xr = X + r * (long long)F;
+    float logits[FLOWML_MAXK];
+    float m = -1e30f;
+    for (int k = 0; k < K; ++k) {
+      float acc = s_b[k];
+      const float* wk = s_w + k * F;
+      for (int f = 0; f < F; ++f) acc = fmaf(wk[f], xr[f], acc);
+      logits[k] = acc;
+      m = fmaxf(m, acc);
+    }
+    float sum = 0.f;
+    for (int k = 0; k < K; ++k) {
+      logits[k] = __expf(logits[k] - m);
+      sum += logits[k];
+    }
+    const float inv = 1.f / sum;
+    const int yk = y[r];
+    const float wr = w ? w[r] : 1.f;
+    // loss = logZ - logit_y = log(sum) - (logit_y - m) = log(sum) - log(e_y)
+    loss_local += wr * (logf(sum) - logf(logits[yk]));
+    float* dr = delta + r * (long long)K;
+    for (int k = 0; k < K; ++k) {
+      float p = logits[k] * inv;
+      dr[k] = wr * (p - (k == yk ? 1.f : 0.f));
+    }
+  }
+  // block-reduce loss
+  __shared__ float s_red[256];
+  s_red[threadIdx.x] = loss_local;
+  __syncthreads();
+  for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) s_red[threadIdx.x] += s_red[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(loss, s_red[0]);
+}
+
+// ---------------------------------------------------------------------------
+// K6 — softmax / cross-entropy: loss + delta from logits
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void flowml_softmax_xent(
+    const float* __restrict__ logits,  // N×K
+    const int* __restrict__ y,         // N
+    const float* __restrict__ w,       // N or nullptr
+    float* __restrict__ delta,         // N×K out
+    float* __restrict__ loss,          // 1 (atomic)
+    long long N, int K) {
+  float loss_local = 0.f;
+  for (long long r = blockIdx.x * (long long)blockDim.x + threadIdx.x; r < N;
+       r += (long long)gridDim.x * blockDim.x) {
+    const float* lr = logits + r * (long long)K;
+    float m = -1e30f;
+    for (int k = 0; k < K; ++k) m = fmaxf(m, lr[k]);
+    float sum = 0.f;
+    float e[FLOWML_MAXK];
+    for (int k = 0; k < K; ++k) {
+      e[k] = __expf(lr[k] - m);
+      sum += e[k];
+    }
+    const float inv = 1.f / sum;
+    const int yk = y[r];
+    const float wr = w ? w[r] : 1.f;
+    loss_local += wr * (logf(sum) + m - lr[yk]);
+    float* dr = delta + r * (long long)K;
+    for (int k = 0; k < K; ++k)
+      dr[k] = wr * (e[k] * inv - (k == yk ? 1.f : 0.f));
+  }
+  __shared__ float s_red[256];
+  s_red[threadIdx.x] = loss_local;
+  __syncthreads();
+  for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) s_red[threadIdx.x] += s_red[threadIdx.x + s];
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) atomicAdd(loss, s_red[0]);
+}
+
+// ---------------------------------------------------------------------------
+// Host launchers (called from ext.cpp; stream passed as void* to keep the
+// binding TU free of HIP headers)
+// ---------------------------------------------------------------------------
+
+static inline int grid_1d(long long total, int block, int cap = 4096) {
+  long long g = (total + block - 1) / block;
+  return (int)(g < cap ? (g > 0 ? g : 1) : cap);
+}
+
+extern "C" void launch_bin_codes(const float* X, const float* thr, uint8_t* out,
+                                 long long N, int F, int TB, void* stream) {
+  const int block = 256;
+  const size_t lds = (size_t)F * TB * sizeof(float);
+  hipLaunchKernelGGL(flowml_bin_codes, dim3(grid_1d(N * F, block)), dim3(block),
+                     lds, (hipStream_t)stream, X, thr, out, N, F, TB);
+}
+
+static bool g_hist_lds_opted = false;
+static void hist_opt_in_big_lds() {
+  if (g_hist_lds_opted) return;
+  // gfx950 allows >64 KiB LDS per workgroup; opt in once per process.
+  (void)hipFuncSetAttribute((const void*)flowml_tree_hist_cls,
+                            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+  (void)hipFuncSetAttribute((const void*)flowml_tree_hist_reg,
+                            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+  g_hist_lds_opted = true;
+}
+
+extern "C" void launch_tree_hist_cls(
+    const uint8_t* codes, const int* nodeid, const int* lut, const int* tree_lo,
+    const int* tree_cnt, const uint8_t* labels, const float* bag, const int* fs,
+    float* out, long long N, int F, int T, int M, int Fp, int B, int C,
+    int l_max, void* stream) {
+  hist_opt_in_big_lds();
+  const int block = 256;
+  const long long rows_per_block = 8192;
+  const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
+  const size_t lds = (size_t)l_max * Fp * B * C * sizeof(float);
+  hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(tiles, T), dim3(block), lds,
+                     (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
+                     labels, bag, fs, out, N, F, M, Fp, B, C, l_max,
+                     rows_per_block);
+}
+
+extern "C" void launch_tree_hist_reg(
+    const uint8_t* codes, const int* nodeid, const int* lut, const int* tree_lo,
+    const int* tree_cnt, const float* targets, const float* bag, const int* fs,
+    float* out, long long N, int F, int T, int M, int Fp, int B, int l_max,
+    void* stream) {
+  hist_opt_in_big_lds();
+  const int block = 256;
+  const long long rows_per_block = 8192;
+  const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
+  const size_t lds = (size_t)l_max * Fp * B * 3 * sizeof(float);
+  hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(tiles, T), dim3(block), lds,
+                     (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
+                     targets, bag, fs, out, N, F, M, Fp, B, l_max,
+                     rows_per_block);
+}
+
+extern "C" void launch_node_advance(const uint8_t* codes, const int* nodeid_in,
+                                    int* nodeid_out, const int* feat,
+                                    const int* thr, const int* left,
+                                    const int* right, const int* node_ofs,
+                                    long long N, int F, int T, void* stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(flowml_node_advance, dim3(grid_1d(N, block, 2048), T),
+                     dim3(block), 0, (hipStream_t)stream, codes, nodeid_in,
+                     nodeid_out, feat, thr, left, right, node_ofs, N, F);
+}
+
+extern "C" void launch_tree_predict(const float* X, const int* feature,
+                                    const float* threshold, const int* left,
+                                    const int* right, const float* value,
+                                    float* out, long long N, int F, int T,
+                                    int M, int K, void* stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(flowml_tree_predict, dim3(grid_1d(N, block, 8192)),
+                     dim3(block), 0, (hipStream_t)stream, X, feature, threshold,
+                     left, right, value, out, N, F, T, M, K);
+}
+
+extern "C" void launch_logreg_delta(const float* X, const int* y,
+                                    const float* w, const float* W,
+                                    const float* b, float* delta, float* loss,
+                                    long long N, int F, int K, void* stream) {
+  const int block = 256;
+  const size_t lds = (size_t)(K * F + K) * sizeof(float);
+  hipLaunchKernelGGL(flowml_logreg_delta, dim3(grid_1d(N, block, 2048)),
+                     dim3(block), lds, (hipStream_t)stream, X, y, w, W, b,
+                     delta, loss, N, F, K);
+}
+
+extern "C" void launch_softmax_xent(const float* logits, const int* y,
+                                    const float* w, float* delta, float* loss,
+                                    long long N, int K, void* stream) {
+  const int block = 256;
+  hipLaunchKernelGGL(flowml_softmax_xent, dim3(grid_1d(N, block, 2048)),
+                     dim3(block), 0, (hipStream_t)stream, logits, y, w, delta,
+                     loss, N, K);
+}

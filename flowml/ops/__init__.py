@@ -104,6 +104,7 @@ def tree_hist_nodes(
     nodeid: torch.Tensor,  # T×N int32 current node per (tree,row), -1 done
     lut: torch.Tensor,     # T×M int32: node -> chunk group index, -1 inactive
     tree_lo: torch.Tensor, # T int32: first group index of tree t in the chunk
+    tree_cnt: torch.Tensor,  # T int32: number of groups of tree t in the chunk
     labels: torch.Tensor,  # N uint8 class labels (classification)
     bag: Optional[torch.Tensor],      # T×N float32 bag weights (None = 1)
     n_groups: int,
@@ -126,11 +127,11 @@ def tree_hist_nodes(
         bg = bag if bag is not None else _empty(codes.device)
         if targets is None:
             return ext.tree_hist_cls(
-                codes, nodeid, lut, tree_lo, labels, bg,
+                codes, nodeid, lut, tree_lo, tree_cnt, labels, bg,
                 int(n_groups), int(l_max), int(n_bins), int(n_classes), fs,
             )
         return ext.tree_hist_reg(
-            codes, nodeid, lut, tree_lo, targets, bg,
+            codes, nodeid, lut, tree_lo, tree_cnt, targets, bg,
             int(n_groups), int(l_max), int(n_bins), fs,
         )
     return _tree_hist_nodes_torch(
@@ -290,11 +291,17 @@ def logreg_grad(
     inside treeAggregate.
     """
     if hip_available(X):
-        return get_ext().logreg_grad(
+        # K1 fused kernel produces (loss, weighted delta); the rank-1-update
+        # accumulation gradW = deltaᵀX is a plain GEMM -> rocBLAS (allowed for
+        # plain library GEMMs per the north star).
+        loss, delta = get_ext().logreg_delta(
             X.contiguous(), y.to(torch.int32).contiguous(),
             w.contiguous() if w is not None else _empty(X.device),
             W.contiguous(), b.contiguous(),
         )
+        gradW = delta.T @ X
+        gradB = delta.sum(dim=0)
+        return loss, gradW, gradB
     return _logreg_grad_torch(X, y, w, W, b)
 
 
@@ -395,13 +402,18 @@ def mlp_grad(
         gWs[i] = matmul_tn(delta, a)          # K7 wgrad: (N×K)ᵀ·(N×F)
         gBs[i] = delta.sum(dim=0).to(torch.float32)
         if i > 0:
-            dx = delta @ Ws[i]                # K7 dgrad
+            dx = delta @ Ws[i]                # K7 dgrad (plain GEMM: rocBLAS)
             h_i = acts[i]
-            if activation == "relu":
-                dact = (h_i > 0).to(dx.dtype)
+            if dx.is_cuda and dx.dtype == torch.bfloat16 and hip_available(dx):
+                delta = get_ext().act_backward_(
+                    h_i.contiguous(), dx.contiguous(), _act_code(activation)
+                )
             else:
-                dact = h_i * (1 - h_i)
-            delta = dx * dact
+                if activation == "relu":
+                    dact = (h_i > 0).to(dx.dtype)
+                else:
+                    dact = h_i * (1 - h_i)
+                delta = dx * dact
     return loss, gWs, gBs
 
 
@@ -429,7 +441,9 @@ def softmax_xent_grad(
 
 def matmul_tn(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
     """wgrad helper: ``A.T @ B`` as fp32, A: N×K, B: N×F → K×F.
-    GPU path: split-K MFMA kernel (K7)."""
-    if hip_available(A):
-        return get_ext().matmul_tn(A.contiguous(), B.contiguous())
+
+    This is a *plain* GEMM (no fusion), so rocBLAS via torch.matmul is the
+    sanctioned path on GPU (north star: "hipBLASLt/rocBLAS only for plain
+    library GEMMs"); fp32 accumulate via explicit casts.
+    """
     return A.T.to(torch.float32) @ B.to(torch.float32)

@@ -1,0 +1,288 @@
+"""GPU kernel parity tests (SURVEY.md §4 "Kernel parity"): every HIP kernel
+vs the torch/CPU reference — bitwise for integer histograms, tolerance for
+GEMM — plus determinism runs and model-level fits on device."""
+
+import numpy as np
+import pytest
+import torch
+
+from flowml import ops
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _force_cpu(fn, *args, **kw):
+    """Run the torch reference implementation (CPU tensors)."""
+    cpu_args = [a.cpu() if isinstance(a, torch.Tensor) else a for a in args]
+    cpu_kw = {k: (v.cpu() if isinstance(v, torch.Tensor) else v) for k, v in kw.items()}
+    return fn(*cpu_args, **cpu_kw)
+
+
+class TestBinCodes:
+    def test_parity_bitwise(self):
+        r = np.random.default_rng(0)
+        X = torch.from_numpy(r.normal(size=(100_000, 41)).astype(np.float32)).to(DEV)
+        thr = torch.sort(
+            torch.from_numpy(r.normal(size=(41, 31)).astype(np.float32)), dim=1
+        ).values.to(DEV)
+        gpu = ops.bin_codes(X, thr).cpu()
+        cpu = ops._bin_codes_torch(X.cpu(), thr.cpu())
+        assert torch.equal(gpu, cpu)
+
+    def test_inf_padding(self):
+        X = torch.tensor([[5.0, -1.0]], device=DEV)
+        thr = torch.tensor([[1.0, 2.0, float("inf")], [0.0, float("inf"), float("inf")]], device=DEV)
+        codes = ops.bin_codes(X, thr).cpu()
+        assert codes.tolist() == [[2, 0]]
+
+
+class TestTreeHist:
+    def _mk(self, r, N=50_000, F=41, T=8, M=16, B=32, C=5, per_tree=4, Fp=6):
+        codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
+        nodeid = torch.from_numpy(r.integers(-1, M, (T, N)).astype(np.int32)).to(DEV)
+        lut = torch.full((T, M), -1, dtype=torch.int32)
+        tree_lo = torch.zeros(T, dtype=torch.int32)
+        tree_cnt = torch.zeros(T, dtype=torch.int32)
+        g = 0
+        for t in range(T):
+            tree_lo[t] = g
+            for nid in sorted(r.choice(M, per_tree, replace=False)):
+                lut[t, nid] = g
+                g += 1
+            tree_cnt[t] = per_tree
+        G = g
+        labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8)).to(DEV)
+        bag = torch.from_numpy(r.integers(0, 3, (T, N)).astype(np.float32)).to(DEV)
+        fs = torch.from_numpy(
+            np.stack([np.sort(r.choice(F, Fp, replace=False)) for _ in range(G)]).astype(np.int32)
+        ).to(DEV)
+        return codes, nodeid, lut.to(DEV), tree_lo.to(DEV), tree_cnt.to(DEV), labels, bag, fs, G, per_tree, B, C
+
+    def test_cls_parity_bitwise(self):
+        r = np.random.default_rng(1)
+        codes, nodeid, lut, tree_lo, tree_cnt, labels, bag, fs, G, l_max, B, C = self._mk(r)
+        gpu = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, bag,
+                                  G, l_max, B, C, fs).cpu()
+        cpu = ops._tree_hist_nodes_torch(codes.cpu(), nodeid.cpu(), lut.cpu(), labels.cpu(),
+                                         bag.cpu(), G, B, C, fs.cpu(), None)
+        # integer counts: must match bitwise
+        assert torch.equal(gpu, cpu)
+
+    def test_cls_determinism(self):
+        r = np.random.default_rng(2)
+        codes, nodeid, lut, tree_lo, tree_cnt, labels, bag, fs, G, l_max, B, C = self._mk(r)
+        a = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, bag, G, l_max, B, C, fs)
+        b = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, bag, G, l_max, B, C, fs)
+        assert torch.equal(a.cpu(), b.cpu())
+
+    def test_cls_no_subset_no_bag(self):
+        r = np.random.default_rng(3)
+        N, F, B, C = 30_000, 11, 16, 3
+        codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
+        nodeid = torch.zeros((1, N), dtype=torch.int32, device=DEV)
+        lut = torch.zeros((1, 1), dtype=torch.int32, device=DEV)
+        tree_lo = torch.zeros(1, dtype=torch.int32, device=DEV)
+        tree_cnt = torch.ones(1, dtype=torch.int32, device=DEV)
+        labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8)).to(DEV)
+        gpu = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, None,
+                                  1, 1, B, C, None).cpu()
+        cpu = ops._tree_hist_nodes_torch(codes.cpu(), nodeid.cpu(), lut.cpu(), labels.cpu(),
+                                         None, 1, B, C, None, None)
+        assert torch.equal(gpu, cpu)
+
+    def test_reg_parity(self):
+        r = np.random.default_rng(4)
+        N, F, B = 40_000, 9, 32
+        codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
+        nodeid = torch.from_numpy(r.integers(-1, 2, (1, N)).astype(np.int32)).to(DEV)
+        lut = torch.tensor([[0, 1]], dtype=torch.int32, device=DEV)
+        tree_lo = torch.zeros(1, dtype=torch.int32, device=DEV)
+        tree_cnt = torch.full((1,), 2, dtype=torch.int32, device=DEV)
+        targets = torch.from_numpy(r.normal(size=N).astype(np.float32)).to(DEV)
+        labels = torch.zeros(N, dtype=torch.uint8, device=DEV)
+        gpu = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, None,
+                                  2, 2, B, 0, None, targets=targets).cpu()
+        cpu = ops._tree_hist_nodes_torch(codes.cpu(), nodeid.cpu(), lut.cpu(), labels.cpu(),
+                                         None, 2, B, 0, None, targets.cpu())
+        assert torch.allclose(gpu, cpu, rtol=1e-5, atol=1e-2)
+
+
+class TestNodeAdvanceGPU:
+    def test_parity(self):
+        r = np.random.default_rng(5)
+        N, F, T = 80_000, 13, 6
+        M = 7
+        codes = torch.from_numpy(r.integers(0, 32, (N, F)).astype(np.uint8)).to(DEV)
+        nodeid = torch.from_numpy(r.integers(-1, M, (T, N)).astype(np.int32)).to(DEV)
+        feat, thr, left, right, ofs = [], [], [], [], [0]
+        for t in range(T):
+            for m in range(M):
+                internal = r.random() < 0.6
+                feat.append(int(r.integers(0, F)) if internal else -1)
+                thr.append(int(r.integers(0, 31)))
+                left.append(int(r.integers(0, M)))
+                right.append(int(r.integers(0, M)))
+            ofs.append(len(feat))
+        feat = torch.tensor(feat, dtype=torch.int32, device=DEV)
+        thr = torch.tensor(thr, dtype=torch.int32, device=DEV)
+        left = torch.tensor(left, dtype=torch.int32, device=DEV)
+        right = torch.tensor(right, dtype=torch.int32, device=DEV)
+        ofs = torch.tensor(ofs, dtype=torch.int32, device=DEV)
+        gpu = ops.node_advance_multi(codes, nodeid, feat, thr, left, right, ofs).cpu()
+        cpu = ops.node_advance_multi(codes.cpu(), nodeid.cpu(), feat.cpu(), thr.cpu(),
+                                     left.cpu(), right.cpu(), ofs.cpu())
+        assert torch.equal(gpu, cpu)
+
+
+class TestTreePredictGPU:
+    def test_parity(self):
+        r = np.random.default_rng(6)
+        N, F, T, M, K = 50_000, 41, 20, 63, 5
+        X = torch.from_numpy(r.normal(size=(N, F)).astype(np.float32)).to(DEV)
+        # random complete binary trees depth<=5 in breadth-first layout
+        feature = torch.full((T, M), -1, dtype=torch.int32)
+        threshold = torch.zeros((T, M), dtype=torch.float32)
+        left = torch.zeros((T, M), dtype=torch.int32)
+        right = torch.zeros((T, M), dtype=torch.int32)
+        value = torch.from_numpy(r.normal(size=(T, M, K)).astype(np.float32))
+        for t in range(T):
+            for i in range(31):  # internal nodes of a depth-5 complete tree
+                feature[t, i] = int(r.integers(0, F))
+                threshold[t, i] = float(r.normal())
+                left[t, i] = 2 * i + 1
+                right[t, i] = 2 * i + 2
+        args = [X, feature.to(DEV), threshold.to(DEV), left.to(DEV), right.to(DEV), value.to(DEV)]
+        gpu = ops.tree_predict(*args).cpu()
+        cpu = ops._tree_predict_torch(X.cpu(), feature, threshold, left, right, value)
+        assert torch.allclose(gpu, cpu, atol=1e-4)
+
+
+class TestLogregDeltaGPU:
+    def test_parity(self):
+        r = np.random.default_rng(7)
+        N, F, K = 60_000, 122, 5
+        X = torch.from_numpy(r.normal(size=(N, F)).astype(np.float32)).to(DEV)
+        y = torch.from_numpy(r.integers(0, K, N).astype(np.int64)).to(DEV)
+        W = torch.from_numpy((0.1 * r.normal(size=(K, F))).astype(np.float32)).to(DEV)
+        b = torch.from_numpy(r.normal(size=K).astype(np.float32)).to(DEV)
+        w = torch.from_numpy((r.random(N) + 0.5).astype(np.float32)).to(DEV)
+        loss, gW, gB = ops.logreg_grad(X, y, w, W, b)
+        loss_c, gW_c, gB_c = ops._logreg_grad_torch(X.cpu(), y.cpu(), w.cpu(), W.cpu(), b.cpu())
+        assert float(loss) == pytest.approx(float(loss_c), rel=1e-4)
+        assert torch.allclose(gW.cpu(), gW_c, rtol=1e-3, atol=1e-2)
+        assert torch.allclose(gB.cpu(), gB_c, rtol=1e-3, atol=1e-2)
+
+
+class TestSoftmaxXentGPU:
+    def test_parity(self):
+        r = np.random.default_rng(8)
+        N, K = 70_000, 5
+        logits = torch.from_numpy(r.normal(size=(N, K)).astype(np.float32)).to(DEV)
+        y = torch.from_numpy(r.integers(0, K, N).astype(np.int64)).to(DEV)
+        loss, delta = ops.softmax_xent_grad(logits, y, None)
+        loss_c, delta_c = ops.softmax_xent_grad(logits.cpu(), y.cpu(), None)
+        assert float(loss) == pytest.approx(float(loss_c), rel=1e-4)
+        assert torch.allclose(delta.cpu(), delta_c, atol=1e-4)
+
+
+class TestGemmBiasAct:
+    @pytest.mark.parametrize("M,K", [(128, 41), (64, 128), (5, 64), (128, 128)])
+    def test_bf16_parity(self, M, K):
+        r = np.random.default_rng(M * 100 + K)
+        N = 8192 + 17  # non-multiple of 64: exercises row guards
+        X = torch.from_numpy(r.normal(size=(N, K)).astype(np.float32))
+        W = torch.from_numpy((0.2 * r.normal(size=(M, K))).astype(np.float32))
+        b = torch.from_numpy(r.normal(size=M).astype(np.float32))
+        Xg = X.to(DEV, torch.bfloat16)
+        Wg = W.to(DEV, torch.bfloat16)
+        out = ops.gemm_bias_act(Xg, Wg, b.to(DEV), "relu").float().cpu()
+        # reference in fp32 on the bf16-rounded inputs
+        ref = torch.relu(Xg.float().cpu() @ Wg.float().cpu().T + b)
+        # bf16 output rounding: tolerance ~ 1% of magnitude
+        assert torch.allclose(out, ref, rtol=2e-2, atol=2e-2), (out - ref).abs().max()
+
+    def test_f32_parity_tight(self):
+        r = np.random.default_rng(77)
+        N, M, K = 4096 + 5, 64, 41
+        X = torch.from_numpy(r.normal(size=(N, K)).astype(np.float32)).to(DEV)
+        W = torch.from_numpy((0.2 * r.normal(size=(M, K))).astype(np.float32)).to(DEV)
+        b = torch.from_numpy(r.normal(size=M).astype(np.float32)).to(DEV)
+        out = ops.gemm_bias_act(X, W, b, "none").cpu()
+        ref = X.cpu() @ W.cpu().T + b.cpu()
+        # exact f32 MFMA: fmaf-chain, ~1e-6 relative
+        assert torch.allclose(out, ref, rtol=1e-5, atol=1e-5)
+
+    def test_transpose_detecting(self):
+        """Asymmetric operands catch swapped row/col mappings (guide §5.4 r16)."""
+        N, M, K = 64, 32, 32
+        X = torch.zeros(N, K)
+        X[3, 7] = 1.0
+        W = torch.zeros(M, K)
+        W[11, 7] = 2.0
+        out = ops.gemm_bias_act(
+            X.to(DEV, torch.bfloat16), W.to(DEV, torch.bfloat16),
+            torch.zeros(M, device=DEV), "none"
+        ).float().cpu()
+        assert out[3, 11] == pytest.approx(2.0)
+        assert out.abs().sum() == pytest.approx(2.0)
+
+
+class TestModelFitsGPU:
+    def test_rf_fit_gpu_matches_cpu(self):
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(9)
+        X = r.normal(size=(30_000, 10)).astype(np.float32)
+        W = r.normal(size=(10, 3))
+        y = (X @ W).argmax(1).astype(np.float64)
+        df_cpu = FlowFrame({"features": torch.from_numpy(X), "label": y})
+        df_gpu = df_cpu.to(DEV)
+        m_cpu = RandomForestClassifier(numTrees=8, maxDepth=4, seed=3, bootstrap=False).fit(df_cpu)
+        m_gpu = RandomForestClassifier(numTrees=8, maxDepth=4, seed=3, bootstrap=False).fit(df_gpu)
+        p_cpu = m_cpu.transform(df_cpu)["prediction"]
+        p_gpu = m_gpu.transform(df_gpu)["prediction"].cpu()
+        assert (p_cpu == p_gpu).float().mean() > 0.999
+
+    def test_lr_fit_gpu(self):
+        from flowml.classification import LogisticRegression
+        from flowml.evaluation import MulticlassClassificationEvaluator
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(10)
+        X = r.normal(size=(50_000, 20)).astype(np.float32)
+        W = r.normal(size=(20, 5))
+        y = (X @ W + 0.5 * r.normal(size=(50_000, 5))).argmax(1).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y}).to(DEV)
+        m = LogisticRegression(maxIter=50).fit(df)
+        acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(m.transform(df))
+        assert acc > 0.8
+
+    def test_mlp_fit_gpu_bf16(self):
+        from flowml.classification import MultilayerPerceptronClassifier
+        from flowml.evaluation import MulticlassClassificationEvaluator
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(11)
+        X = r.normal(size=(40_000, 41)).astype(np.float32)
+        W = r.normal(size=(41, 5))
+        y = (X @ W).argmax(1).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y}).to(DEV)
+        m = MultilayerPerceptronClassifier(layers=[41, 128, 64, 5], maxIter=40, seed=1).fit(df)
+        acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(m.transform(df))
+        assert acc > 0.85
+
+    def test_gbt_fit_gpu(self):
+        from flowml.classification import GBTClassifier
+        from flowml.evaluation import MulticlassClassificationEvaluator
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(12)
+        X = r.normal(size=(30_000, 10)).astype(np.float32)
+        y = ((X[:, 0] * X[:, 1] + X[:, 2]) > 0).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y}).to(DEV)
+        m = GBTClassifier(maxIter=20, maxDepth=3, seed=1).fit(df)
+        acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(m.transform(df))
+        assert acc > 0.8

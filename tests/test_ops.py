@@ -109,7 +109,8 @@ class TestTreeHistNodes:
         fs = torch.from_numpy(
             np.stack([np.sort(r.choice(F, 3, replace=False)) for _ in range(G)]).astype(np.int32)
         )
-        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, labels, bag, G, 2, B, C, fs)
+        tree_cnt = torch.full((T,), 2, dtype=torch.int32)
+        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, bag, G, 2, B, C, fs)
         ref = naive_hist(codes.numpy(), nodeid.numpy(), lut.numpy(), labels.numpy(),
                          bag.numpy(), G, B, C, fs.numpy())
         # integer counts must match exactly
@@ -123,7 +124,8 @@ class TestTreeHistNodes:
         lut = torch.zeros((1, 1), dtype=torch.int32)
         tree_lo = torch.zeros(1, dtype=torch.int32)
         labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8))
-        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, labels, None, 1, 1, B, C, None)
+        tree_cnt = torch.ones(1, dtype=torch.int32)
+        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, None, 1, 1, B, C, None)
         assert float(out.sum()) == N * F
         ref = naive_hist(codes.numpy(), nodeid.numpy(), lut.numpy(), labels.numpy(),
                          None, 1, B, C, None)
@@ -138,7 +140,8 @@ class TestTreeHistNodes:
         tree_lo = torch.zeros(1, dtype=torch.int32)
         targets = torch.from_numpy(r.normal(size=N).astype(np.float32))
         labels = torch.zeros(N, dtype=torch.uint8)
-        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, labels, None, 1, 1, B, 0,
+        tree_cnt = torch.ones(1, dtype=torch.int32)
+        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, None, 1, 1, B, 0,
                                   None, targets=targets)
         assert out.shape == (1, F, B, 3)
         for f in range(F):
