@@ -24,6 +24,8 @@ written as 0/continuous).
 from __future__ import annotations
 
 import math
+import os
+import time
 from dataclasses import dataclass, field
 from typing import List, Optional, Sequence, Tuple
 
@@ -41,11 +43,15 @@ from flowml import dist, ops
 def find_split_thresholds(
     X: torch.Tensor, max_bins: int, seed: int, sample_cap: int = 1 << 16
 ) -> torch.Tensor:
-    """Per-feature quantile thresholds, F×(max_bins-1), +inf padded.
+    """Per-feature equi-depth quantile thresholds, F×(max_bins-1), +inf padded.
 
     Samples up to ``sample_cap`` rows (seeded) per rank, gathers samples
-    across ranks (one-time, small — SURVEY.md §2.5 C6), and takes equi-depth
-    quantiles of the distinct values. All ranks produce identical thresholds.
+    across ranks (one-time, small — SURVEY.md §2.5 C6), sorts each feature
+    column and takes lower quantiles, deduplicated. Fully torch-vectorized —
+    runs on-device for the single-rank GPU path. All ranks produce identical
+    thresholds. Bin semantics are strict: code = #(thr < x), so the binned
+    split "code <= b" is exactly the raw-feature rule "x <= thr[b]" that
+    inference (K8) applies — consistent even on integer-coded features.
     """
     N, F = X.shape
     g = torch.Generator().manual_seed(seed ^ 0x5EEDB1A5)
@@ -54,33 +60,26 @@ def find_split_thresholds(
         sample = X.index_select(0, idx)
     else:
         sample = X
-    sample_np = sample.detach().cpu().numpy()
     if dist.is_initialized() and dist.get_world_size() > 1:
-        parts = dist.all_gather_object(sample_np)
-        sample_np = np.concatenate(parts, axis=0)
-        # cap the merged sample deterministically
-        if len(sample_np) > 4 * sample_cap:
+        parts = dist.all_gather_object(sample.detach().cpu().numpy())
+        merged = np.concatenate(parts, axis=0)
+        if len(merged) > 4 * sample_cap:
             rng = np.random.default_rng(seed ^ 0x5EEDB1A5)
-            keep = rng.choice(len(sample_np), size=4 * sample_cap, replace=False)
+            keep = rng.choice(len(merged), size=4 * sample_cap, replace=False)
             keep.sort()
-            sample_np = sample_np[keep]
+            merged = merged[keep]
+        sample = torch.from_numpy(np.ascontiguousarray(merged))
     B = max_bins
-    thr = np.full((F, B - 1), np.inf, dtype=np.float32)
-    qs = np.arange(1, B) / B
-    for f in range(F):
-        col = np.sort(sample_np[:, f].astype(np.float64))
-        uniq = np.unique(col)
-        if len(uniq) <= 1:
-            continue
-        if len(uniq) < B:
-            # midpoints between consecutive distinct values
-            mids = (uniq[:-1] + uniq[1:]) / 2.0
-            thr[f, : len(mids)] = mids
-        else:
-            q = np.quantile(col, qs, method="lower")
-            tq = np.unique(q)
-            thr[f, : len(tq)] = tq
-    return torch.from_numpy(thr).to(X.device)
+    S, _ = torch.sort(sample.to(torch.float32), dim=0)
+    n = S.shape[0]
+    qidx = (torch.arange(1, B, device=S.device, dtype=torch.int64) * (n - 1)) // B
+    cand = S.index_select(0, qidx)  # (B-1)×F, ascending per column
+    dup = torch.zeros_like(cand, dtype=torch.bool)
+    dup[1:] = cand[1:] == cand[:-1]
+    inf = torch.full_like(cand, float("inf"))
+    cand = torch.where(dup, inf, cand)
+    thr, _ = torch.sort(cand, dim=0)
+    return thr.T.contiguous().to(X.device)
 
 
 # ---------------------------------------------------------------------------
@@ -312,6 +311,18 @@ def grow_trees(
     labels_u8 = labels_or_targets.to(torch.uint8) if is_cls else torch.zeros(N, dtype=torch.uint8, device=dev)
     targets = None if is_cls else labels_or_targets.to(torch.float32)
 
+    timing = os.environ.get("FLOWML_TIMING", "0") == "1"
+    tmr = {}
+
+    def _tick():
+        if timing and dev.type == "cuda":
+            torch.cuda.synchronize()
+        return time.perf_counter()
+
+    def _acc(key, t0):
+        if timing:
+            tmr[key] = tmr.get(key, 0.0) + (_tick() - t0)
+
     trees = [FlatTree() for _ in range(T)]
     nodeid = torch.zeros((T, N), dtype=torch.int32, device=dev)
 
@@ -347,59 +358,93 @@ def grow_trees(
     level = 0
     while frontier and level < cfg.max_depth:
         next_frontier: List[Tuple[int, int, int]] = []
-        for chunk in _make_chunks(frontier, g_max, l_cap):
+        for ci, chunk in enumerate(_make_chunks(frontier, g_max, l_cap)):
+            t_chunk0 = _tick()
             G = len(chunk)
-            max_nodes = max(trees[t].n_nodes for t, _, _ in chunk)
-            lut = torch.full((T, max_nodes), -1, dtype=torch.int32)
+            chunk_t = np.fromiter((c[0] for c in chunk), dtype=np.int64, count=G)
+            chunk_n = np.fromiter((c[1] for c in chunk), dtype=np.int64, count=G)
+            chunk_d = np.fromiter((c[2] for c in chunk), dtype=np.int64, count=G)
+            max_nodes = int(max(trees[t].n_nodes for t, _, _ in chunk))
+            lut_np = np.full((T, max_nodes), -1, dtype=np.int32)
+            lut_np[chunk_t, chunk_n] = np.arange(G, dtype=np.int32)
+            per_tree_count = np.bincount(chunk_t, minlength=T).astype(np.int32)
+            # first group index per tree (tree-major contiguity)
             tree_lo_np = np.zeros(T, dtype=np.int32)
-            per_tree_count = np.zeros(T, dtype=np.int64)
-            for g, (t, nid, _) in enumerate(chunk):
-                lut[t, nid] = g
-                if per_tree_count[t] == 0:
-                    tree_lo_np[t] = g
-                per_tree_count[t] += 1
+            nz = per_tree_count > 0
+            starts = np.cumsum(per_tree_count) - per_tree_count
+            tree_lo_np[nz] = starts[nz]
             l_max = int(per_tree_count.max())
-            lut_d = lut.to(dev)
+            lut_d = torch.from_numpy(lut_np).to(dev)
             tree_lo = torch.from_numpy(tree_lo_np).to(dev)
-            tree_cnt = torch.from_numpy(per_tree_count.astype(np.int32)).to(dev)
-            # feature subsets per node (seeded by (seed, tree, node) —
-            # rank-invariant and reproducible)
+            tree_cnt = torch.from_numpy(per_tree_count).to(dev)
+            # per-node feature subsets: one seeded RNG per (level, chunk) —
+            # the frontier order is deterministic and identical on every rank,
+            # so this is rank-invariant and reproducible (√F draw, K2 input)
             if use_subset:
-                fs_np = np.empty((G, sub_F), dtype=np.int32)
-                for g, (t, nid, _) in enumerate(chunk):
-                    r = np.random.default_rng((cfg.seed * 1_000_003 + t * 131 + nid) & 0x7FFFFFFF)
-                    fs_np[g] = np.sort(r.choice(F, size=sub_F, replace=False)).astype(np.int32)
+                r = np.random.default_rng((cfg.seed * 1_000_003 + level * 8191 + ci) & 0x7FFFFFFF)
+                keys = r.random((G, F))
+                fs_np = np.sort(np.argpartition(keys, sub_F - 1, axis=1)[:, :sub_F], axis=1).astype(np.int32)
                 fs = torch.from_numpy(fs_np).to(dev)
             else:
+                fs_np = None
                 fs = None
+            _acc("chunk_setup", t_chunk0)
+            t0 = _tick()
             hist = ops.tree_hist_nodes(
                 codes, nodeid, lut_d, tree_lo, tree_cnt, labels_u8, bag, G, l_max, B,
                 S if is_cls else 0, fs, targets=targets,
             )
+            _acc("hist", t0)
+            t0 = _tick()
             dist.all_reduce_(hist.reshape(-1))
+            _acc("allreduce", t0)
+            t0 = _tick()
             best_f, best_bin, best_gain, parent, left_stats = split_scan(
                 hist, cfg.impurity, cfg.min_instances_per_node, cfg.min_info_gain
             )
-            bf = best_f.cpu().numpy()
-            bb = best_bin.cpu().numpy()
-            bg = best_gain.cpu().numpy()
-            ls = left_stats.cpu().numpy().astype(np.float64)
-            ps = parent.cpu().numpy().astype(np.float64)
-            fs_host = fs.cpu().numpy() if fs is not None else None
-            for g, (t, nid, d) in enumerate(chunk):
+            _acc("scan", t0)
+            t0 = _tick()
+            # vectorized bookkeeping: one host transfer per chunk
+            right_stats = parent - left_stats
+            imp_l_t = _impurity_from_stats(left_stats, cfg.impurity)
+            imp_r_t = _impurity_from_stats(right_stats, cfg.impurity)
+            cnt_l_t = _counts_of(left_stats)
+            cnt_r_t = _counts_of(right_stats)
+            packed = torch.cat(
+                [
+                    best_f.to(torch.float64).reshape(G, 1),
+                    best_bin.to(torch.float64).reshape(G, 1),
+                    best_gain.reshape(G, 1),
+                    imp_l_t.reshape(G, 1),
+                    imp_r_t.reshape(G, 1),
+                    cnt_l_t.reshape(G, 1),
+                    cnt_r_t.reshape(G, 1),
+                    left_stats,
+                    right_stats,
+                ],
+                dim=1,
+            ).cpu().numpy()
+            bf = packed[:, 0].astype(np.int64)
+            bb = packed[:, 1].astype(np.int64)
+            bg = packed[:, 2]
+            il = packed[:, 3]
+            ir = packed[:, 4]
+            cl = packed[:, 5]
+            cr = packed[:, 6]
+            Sd = left_stats.shape[1]
+            ls = packed[:, 7 : 7 + Sd]
+            rs = packed[:, 7 + Sd : 7 + 2 * Sd]
+            min_c = 2 * cfg.min_instances_per_node
+            for g in range(G):
                 if bf[g] < 0:
                     continue  # stays leaf
+                t = int(chunk_t[g])
+                nid = int(chunk_n[g])
+                d = int(chunk_d[g])
                 tr = trees[t]
-                gf = int(fs_host[g, bf[g]]) if fs_host is not None else int(bf[g])
-                stats_l = ls[g]
-                stats_p = ps[g]
-                stats_r = stats_p - stats_l
-                imp_l = float(_impurity_from_stats(torch.from_numpy(stats_l), cfg.impurity))
-                imp_r = float(_impurity_from_stats(torch.from_numpy(stats_r), cfg.impurity))
-                cnt_l = float(stats_l.sum() if is_cls else stats_l[2])
-                cnt_r = float(stats_r.sum() if is_cls else stats_r[2])
-                lid = tr.add_node(stats_l, imp_l, cnt_l)
-                rid = tr.add_node(stats_r, imp_r, cnt_r)
+                gf = int(fs_np[g, bf[g]]) if fs_np is not None else int(bf[g])
+                lid = tr.add_node(ls[g], il[g], cl[g])
+                rid = tr.add_node(rs[g], ir[g], cr[g])
                 tr.feature[nid] = gf
                 tr.thr_bin[nid] = int(bb[g])
                 tr.thr_raw[nid] = float(thr_host[gf, bb[g]])
@@ -407,11 +452,13 @@ def grow_trees(
                 tr.right[nid] = rid
                 tr.gain[nid] = float(bg[g])
                 if d + 1 < cfg.max_depth:
-                    if cnt_l >= 2 * cfg.min_instances_per_node and imp_l > 0:
+                    if cl[g] >= min_c and il[g] > 0:
                         next_frontier.append((t, lid, d + 1))
-                    if cnt_r >= 2 * cfg.min_instances_per_node and imp_r > 0:
+                    if cr[g] >= min_c and ir[g] > 0:
                         next_frontier.append((t, rid, d + 1))
+            _acc("bookkeep", t0)
         # advance node assignments one level, all trees in one launch (K4b)
+        t0 = _tick()
         feat_cat = torch.tensor(
             [f for tr in trees for f in tr.feature], dtype=torch.int32, device=dev
         )
@@ -431,9 +478,12 @@ def grow_trees(
         nodeid = ops.node_advance_multi(
             codes, nodeid, feat_cat, thr_cat, left_cat, right_cat, node_ofs
         )
+        _acc("advance", t0)
         frontier = next_frontier
         level += 1
 
+    if timing:
+        print("grow_trees timing:", {k: round(v * 1000, 1) for k, v in tmr.items()})
     return trees
 
 

@@ -42,7 +42,7 @@ extern "C" __global__ void flowml_bin_codes(
     const float* t = s_thr + f * TB;
     int code = 0;
 #pragma unroll 4
-    for (int j = 0; j < TB; ++j) code += (t[j] <= x) ? 1 : 0;
+    for (int j = 0; j < TB; ++j) code += (t[j] < x) ? 1 : 0;  // strict: see ops.bin_codes
     out[idx] = (uint8_t)code;
   }
 }

@@ -78,7 +78,9 @@ def _empty(device, dtype=torch.float32):
 
 def bin_codes(X: torch.Tensor, thresholds: torch.Tensor) -> torch.Tensor:
     """Quantize N×F float features to uint8 bin codes given per-feature
-    threshold rows (F×(B-1), +inf padded). code = #thresholds <= x.
+    threshold rows (F×(B-1), +inf padded). code = #(thresholds < x), strict —
+    so the binned split "code <= b" equals the raw rule "x <= thr[b]" used at
+    inference (K8), exactly, even for integer-coded features.
 
     Reference site: TreePoint.convertToTreeRDD.
     """
@@ -89,7 +91,7 @@ def bin_codes(X: torch.Tensor, thresholds: torch.Tensor) -> torch.Tensor:
 
 def _bin_codes_torch(X, thresholds):
     codes = torch.searchsorted(
-        thresholds.to(X.dtype), X.T.contiguous(), right=True
+        thresholds.to(X.dtype), X.T.contiguous(), right=False
     ).T
     return codes.to(torch.uint8).contiguous()
 

@@ -47,7 +47,7 @@ class TestBinCodes:
         codes = ops.bin_codes(X, thr)
         for i in range(100):
             for f in range(3):
-                expect = int((thr[f] <= X[i, f]).sum())
+                expect = int((thr[f] < X[i, f]).sum())  # strict: code<=b ⟺ x<=thr[b]
                 assert int(codes[i, f]) == expect
 
     def test_inf_padding(self):

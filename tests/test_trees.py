@@ -98,9 +98,24 @@ class TestBinning:
     def test_constant_feature(self):
         X = torch.full((10, 1), 3.14)
         thr = find_split_thresholds(X, max_bins=8, seed=0)
-        assert bool(torch.isinf(thr).all())
+        # at most one finite (useless) threshold; all rows land in bin 0
+        assert int(torch.isfinite(thr).sum()) <= 1
         codes = ops.bin_codes(X, thr)
         assert codes.max() == 0
+
+    def test_integer_feature_split_consistency(self):
+        """Binned split rule must equal the raw-feature rule at inference,
+        including x == threshold (strict-< codes)."""
+        X = torch.tensor([[0.0], [1.0], [1.0], [2.0], [2.0], [2.0]] * 10)
+        thr = find_split_thresholds(X, max_bins=4, seed=0)
+        codes = ops.bin_codes(X, thr)
+        for b in range(thr.shape[1]):
+            t = float(thr[0, b])
+            if not np.isfinite(t):
+                continue
+            left_binned = codes[:, 0] <= b
+            left_raw = X[:, 0] <= t
+            assert torch.equal(left_binned, left_raw), (b, t)
 
 
 class TestRandomForest:
