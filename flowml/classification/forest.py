@@ -144,13 +144,27 @@ class RandomForestClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
         if dist.is_initialized():
             torch.distributed.all_reduce(kmax, op=torch.distributed.ReduceOp.MAX)
         n_classes = max(int(kmax.item()) + 1, 2)
+        import os as _os
+        import time as _time
+
+        _timing = _os.environ.get("FLOWML_TIMING", "0") == "1"
+
+        def _tick():
+            if _timing and dev.type == "cuda":
+                torch.cuda.synchronize()
+            return _time.perf_counter()
+
+        t0 = _tick()
         seed = self.getOrDefault(self.seed)
         B = self.getMaxBins()
         thresholds = find_split_thresholds(X, B, seed)
+        t1 = _tick()
         codes = ops.bin_codes(X, thresholds)
+        t2 = _tick()
         T = self.getNumTrees()
         bag = poisson_bag(N, T, self.getOrDefault(self.subsamplingRate),
                           self.getOrDefault(self.bootstrap), seed, dev)
+        t3 = _tick()
         row_w = None
         if self.isSet(self.weightCol):
             row_w = dataset.col(self.getWeightCol()).reshape(-1).to(torch.float32)
@@ -170,9 +184,19 @@ class RandomForestClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
             n_classes=n_classes,
         )
         trees = grow_trees(codes, y, bag, thresholds, cfg, row_weights=row_w)
+        t4 = _tick()
         model = RandomForestClassificationModel(
             trees=trees, numClasses=n_classes, numFeatures=F
         )
+        t5 = _tick()
+        if _timing:
+            print("rf_fit timing(ms):", {
+                "thresholds": round((t1 - t0) * 1e3, 1),
+                "bin": round((t2 - t1) * 1e3, 1),
+                "bag": round((t3 - t2) * 1e3, 1),
+                "grow": round((t4 - t3) * 1e3, 1),
+                "model_build": round((t5 - t4) * 1e3, 1),
+            })
         self._copyValues(model)
         model._set_parent(self)
         return model

@@ -54,9 +54,15 @@ def find_split_thresholds(
     inference (K8) applies — consistent even on integer-coded features.
     """
     N, F = X.shape
-    g = torch.Generator().manual_seed(seed ^ 0x5EEDB1A5)
     if N > sample_cap:
-        idx = torch.randperm(N, generator=g)[:sample_cap].to(X.device)
+        if X.is_cuda:
+            # with-replacement sample on device (host randperm of the full
+            # shard costs tens of ms); deterministic per (seed, device)
+            g = torch.Generator(device=X.device).manual_seed(seed ^ 0x5EEDB1A5)
+            idx = torch.randint(0, N, (sample_cap,), device=X.device, generator=g)
+        else:
+            g = torch.Generator().manual_seed(seed ^ 0x5EEDB1A5)
+            idx = torch.randint(0, N, (sample_cap,), generator=g)
         sample = X.index_select(0, idx)
     else:
         sample = X
@@ -457,7 +463,10 @@ def grow_trees(
                     if cr[g] >= min_c and ir[g] > 0:
                         next_frontier.append((t, rid, d + 1))
             _acc("bookkeep", t0)
-        # advance node assignments one level, all trees in one launch (K4b)
+        # advance node assignments one level, all trees in one launch (K4b);
+        # skipped after the last split level (no further histogram pass reads it)
+        if not next_frontier:
+            break
         t0 = _tick()
         feat_cat = torch.tensor(
             [f for tr in trees for f in tr.feature], dtype=torch.int32, device=dev
