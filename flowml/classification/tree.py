@@ -12,8 +12,10 @@ growth is level-synchronous and *replicated* — every rank runs the identical
 driver loop; only the histograms cross GPUs (one RCCL ``all_reduce`` per
 node-group chunk, C2). One histogram pass serves ALL trees in the group
 (inner loop over trees inside the K2 kernel), the binned uint8 matrix is the
-only per-row data the hot loop reads (41 B/row), and the split scan runs
-redundantly on every rank (cheaper than a scatter).
+only per-row data the hot loop reads (41 B/row), the split scan runs
+redundantly on every rank (cheaper than a scatter), and all per-level driver
+bookkeeping is numpy-vectorized — the host never loops over individual nodes
+during growth.
 
 Divergence from Spark, documented: all features are treated as continuous
 (post-StringIndexer categorical indices are ordinal here; Spark's categorical
@@ -26,7 +28,7 @@ from __future__ import annotations
 import math
 import os
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import List, Optional, Sequence, Tuple
 
 import numpy as np
@@ -110,54 +112,149 @@ class GrowthConfig:
     n_classes: int = 2
 
 
-@dataclass
 class FlatTree:
-    """Growing/grown tree in flat arrays (node-local indices)."""
+    """Growing/grown tree in flat numpy arrays (node-local indices).
 
-    feature: List[int] = field(default_factory=list)  # -1 = leaf
-    thr_bin: List[int] = field(default_factory=list)
-    thr_raw: List[float] = field(default_factory=list)
-    left: List[int] = field(default_factory=list)
-    right: List[int] = field(default_factory=list)
-    stats: List[np.ndarray] = field(default_factory=list)  # class counts / (s,s2,n)
-    impurity: List[float] = field(default_factory=list)
-    gain: List[float] = field(default_factory=list)
-    count: List[float] = field(default_factory=list)
+    Array-backed so the grower can append whole levels at once; public
+    accessors return live slices (indexable and writable).
+    """
 
-    def add_node(self, stats: np.ndarray, impurity: float, count: float) -> int:
-        self.feature.append(-1)
-        self.thr_bin.append(0)
-        self.thr_raw.append(0.0)
-        self.left.append(-1)
-        self.right.append(-1)
-        self.stats.append(stats)
-        self.impurity.append(impurity)
-        self.gain.append(0.0)
-        self.count.append(count)
-        return len(self.feature) - 1
+    def __init__(self, stat_width: int = 0, cap: int = 8):
+        self._n = 0
+        self._sw = int(stat_width)
+        self._cap = max(int(cap), 4)
+        self._alloc(self._cap)
+
+    def _alloc(self, cap: int) -> None:
+        self._feature = np.full(cap, -1, np.int32)
+        self._thr_bin = np.zeros(cap, np.int32)
+        self._thr_raw = np.zeros(cap, np.float64)
+        self._left = np.full(cap, -1, np.int32)
+        self._right = np.full(cap, -1, np.int32)
+        self._impurity = np.zeros(cap, np.float64)
+        self._gain = np.zeros(cap, np.float64)
+        self._count = np.zeros(cap, np.float64)
+        self._stats = np.zeros((cap, self._sw), np.float64)
+
+    def _ensure(self, extra: int) -> None:
+        need = self._n + extra
+        if need <= self._cap:
+            return
+        newcap = max(need, self._cap * 2)
+        old = (self._feature, self._thr_bin, self._thr_raw, self._left,
+               self._right, self._impurity, self._gain, self._count, self._stats)
+        n = self._n
+        self._cap = newcap
+        self._alloc(newcap)
+        for dst, src in zip(
+            (self._feature, self._thr_bin, self._thr_raw, self._left,
+             self._right, self._impurity, self._gain, self._count, self._stats),
+            old,
+        ):
+            dst[:n] = src[:n]
+
+    # -- live views ---------------------------------------------------------
 
     @property
     def n_nodes(self) -> int:
-        return len(self.feature)
+        return self._n
+
+    @property
+    def feature(self) -> np.ndarray:
+        return self._feature[: self._n]
+
+    @property
+    def thr_bin(self) -> np.ndarray:
+        return self._thr_bin[: self._n]
+
+    @property
+    def thr_raw(self) -> np.ndarray:
+        return self._thr_raw[: self._n]
+
+    @property
+    def left(self) -> np.ndarray:
+        return self._left[: self._n]
+
+    @property
+    def right(self) -> np.ndarray:
+        return self._right[: self._n]
+
+    @property
+    def impurity(self) -> np.ndarray:
+        return self._impurity[: self._n]
+
+    @property
+    def gain(self) -> np.ndarray:
+        return self._gain[: self._n]
+
+    @property
+    def count(self) -> np.ndarray:
+        return self._count[: self._n]
+
+    @property
+    def stats(self) -> np.ndarray:
+        return self._stats[: self._n]
+
+    # -- mutation -------------------------------------------------------------
+
+    def add_node(self, stats, impurity: float, count: float) -> int:
+        stats = np.asarray(stats, dtype=np.float64).reshape(-1)
+        if self._sw == 0:
+            self._sw = len(stats)
+            self._stats = np.zeros((self._cap, self._sw), np.float64)
+        self._ensure(1)
+        i = self._n
+        self._stats[i] = stats
+        self._impurity[i] = impurity
+        self._count[i] = count
+        self._n += 1
+        return i
+
+    def bulk_add(self, stats2d: np.ndarray, imps: np.ndarray, cnts: np.ndarray) -> int:
+        """Append k nodes; returns the base id (new nodes are base..base+k-1)."""
+        k = len(imps)
+        if self._sw == 0 and stats2d.size:
+            self._sw = stats2d.shape[1]
+            self._stats = np.zeros((self._cap, self._sw), np.float64)
+        self._ensure(k)
+        i = self._n
+        self._stats[i : i + k] = stats2d
+        self._impurity[i : i + k] = imps
+        self._count[i : i + k] = cnts
+        self._n += k
+        return i
+
+    def set_splits(self, nids, feats, bins, raws, lefts, rights, gains) -> None:
+        self._feature[nids] = feats
+        self._thr_bin[nids] = bins
+        self._thr_raw[nids] = raws
+        self._left[nids] = lefts
+        self._right[nids] = rights
+        self._gain[nids] = gains
+
+    # -- queries --------------------------------------------------------------
 
     def prediction(self, i: int) -> float:
-        s = self.stats[i]
-        if len(s) > 1:
+        s = self._stats[i]
+        if self._sw > 1 and self._sw != 3:
             return float(np.argmax(s))
+        if self._sw == 3:
+            return float(s[0] / s[2]) if s[2] > 0 else 0.0
         return float(s[0])
 
     def depth(self) -> int:
-        # compute via BFS
-        if not self.feature:
+        if self._n == 0:
             return 0
-        depth = [0] * self.n_nodes
+        depth = np.zeros(self._n, np.int32)
         best = 0
-        for i in range(self.n_nodes):
-            if self.feature[i] >= 0:
-                for ch in (self.left[i], self.right[i]):
-                    depth[ch] = depth[i] + 1
-                    best = max(best, depth[ch])
-        return best
+        for i in range(self._n):
+            if self._feature[i] >= 0:
+                d = depth[i] + 1
+                depth[self._left[i]] = d
+                depth[self._right[i]] = d
+                if d > best:
+                    best = d
+        return int(best)
 
 
 def _feature_subset_size(strategy: str, F: int, is_classification: bool, num_trees: int) -> int:
@@ -241,11 +338,9 @@ def split_scan(
     best_gain = flat.gather(1, best.unsqueeze(1)).squeeze(1)
     best_f = (best // (B - 1)).to(torch.int64)
     best_bin = (best % (B - 1)).to(torch.int64)
-    # threshold on usable gain
     ok = (best_gain > 0) & (best_gain >= min_info_gain) & torch.isfinite(best_gain)
     best_gain = torch.where(ok, best_gain, torch.zeros_like(best_gain))
     best_f = torch.where(ok, best_f, torch.full_like(best_f, -1))
-    # left stats of the chosen split (G×S)
     gidx = torch.arange(G, device=hist.device)
     left_stats = L[gidx, best_f.clamp_min(0), best_bin.clamp_min(0), :]
     return best_f, best_bin, best_gain, parent, left_stats
@@ -267,27 +362,32 @@ def _hist_lds_cap(sub_F: int, B: int, S: int) -> int:
     return max(1, (128 * 1024) // max(bytes_per_node, 1))
 
 
-def _make_chunks(frontier, g_max: int, l_cap: int):
-    """Split the (tree-major sorted) frontier into chunks obeying both the
-    total-group budget (maxMemoryInMB) and the per-tree LDS cap."""
-    chunks = []
-    cur: List[Tuple[int, int, int]] = []
-    per_tree = 0
-    cur_tree = -1
-    for item in frontier:
-        t = item[0]
-        if t != cur_tree:
-            cur_tree = t
-            per_tree = 0
-        if cur and (len(cur) >= g_max or per_tree >= l_cap):
-            chunks.append(cur)
-            cur = []
-            per_tree = 0
-        cur.append(item)
-        per_tree += 1
-    if cur:
-        chunks.append(cur)
-    return chunks
+def _make_chunk_slices(fr_t: np.ndarray, T: int, g_max: int, l_cap: int) -> List[Tuple[int, int]]:
+    """Contiguous [lo, hi) slices of the tree-major frontier obeying both the
+    total-group budget (maxMemoryInMB) and the per-tree LDS cap. O(#trees)."""
+    counts = np.bincount(fr_t, minlength=T)
+    slices: List[Tuple[int, int]] = []
+    lo = 0
+    cur = 0
+    pos = 0
+    for t in range(T):
+        c = int(counts[t])
+        while c > 0:
+            piece = min(c, l_cap, g_max)
+            if cur and cur + piece > g_max:
+                slices.append((lo, pos))
+                lo = pos
+                cur = 0
+            cur += piece
+            pos += piece
+            c -= piece
+            if c > 0:  # same tree continues: its next piece needs a new chunk
+                slices.append((lo, pos))
+                lo = pos
+                cur = 0
+    if pos > lo:
+        slices.append((lo, pos))
+    return slices
 
 
 def grow_trees(
@@ -329,7 +429,7 @@ def grow_trees(
         if timing:
             tmr[key] = tmr.get(key, 0.0) + (_tick() - t0)
 
-    trees = [FlatTree() for _ in range(T)]
+    trees = [FlatTree(stat_width=S) for _ in range(T)]
     nodeid = torch.zeros((T, N), dtype=torch.int32, device=dev)
 
     # root stats per tree: one K2 pass over feature 0 only (lut: node0 -> t)
@@ -341,18 +441,19 @@ def grow_trees(
         codes, nodeid, lut0, tree_lo0, tree_cnt0, labels_u8, bag, T, 1, B,
         S if is_cls else 0, fs0, targets=targets,
     )
-    dist.all_reduce_(h0.reshape(-1) if h0.is_contiguous() else h0)
-    root_stats = h0.sum(dim=2)[:, 0, :]  # T×S
-    rs_np = root_stats.detach().cpu().numpy().astype(np.float64)
-
-    frontier: List[Tuple[int, int, int]] = []  # (tree, local node id, depth)
+    dist.all_reduce_(h0.reshape(-1))
+    root_stats = h0.sum(dim=2)[:, 0, :].to(torch.float64)  # T×S
+    root_imp = _impurity_from_stats(root_stats, cfg.impurity)
+    root_cnt = _counts_of(root_stats)
+    rs_np = root_stats.cpu().numpy()
+    ri_np = root_imp.cpu().numpy()
+    rc_np = root_cnt.cpu().numpy()
+    min_c = 2 * cfg.min_instances_per_node
     for t in range(T):
-        stats_t = rs_np[t]
-        imp = float(_impurity_from_stats(torch.from_numpy(stats_t), cfg.impurity))
-        cnt = float(stats_t.sum() if is_cls else stats_t[2])
-        nid = trees[t].add_node(stats_t, imp, cnt)
-        if cfg.max_depth > 0 and cnt >= 2 * cfg.min_instances_per_node and imp > 0:
-            frontier.append((t, nid, 0))
+        trees[t].add_node(rs_np[t], float(ri_np[t]), float(rc_np[t]))
+    grow_ok = (rc_np >= min_c) & (ri_np > 0) & (cfg.max_depth > 0)
+    fr_t = np.nonzero(grow_ok)[0].astype(np.int64)  # frontier tree ids
+    fr_n = np.zeros(len(fr_t), dtype=np.int64)  # frontier node ids (roots)
 
     # memory budget → groups per histogram chunk (maxMemoryInMB parity) and
     # per-tree LDS cap (GPU kernel constraint)
@@ -362,19 +463,20 @@ def grow_trees(
 
     thr_host = thresholds.cpu().numpy()
     level = 0
-    while frontier and level < cfg.max_depth:
-        next_frontier: List[Tuple[int, int, int]] = []
-        for ci, chunk in enumerate(_make_chunks(frontier, g_max, l_cap)):
+    while len(fr_t) and level < cfg.max_depth:
+        nxt_t: List[np.ndarray] = []
+        nxt_n: List[np.ndarray] = []
+        for ci, (clo, chi) in enumerate(_make_chunk_slices(fr_t, T, g_max, l_cap)):
             t_chunk0 = _tick()
-            G = len(chunk)
-            chunk_t = np.fromiter((c[0] for c in chunk), dtype=np.int64, count=G)
-            chunk_n = np.fromiter((c[1] for c in chunk), dtype=np.int64, count=G)
-            chunk_d = np.fromiter((c[2] for c in chunk), dtype=np.int64, count=G)
-            max_nodes = int(max(trees[t].n_nodes for t, _, _ in chunk))
+            chunk_t = fr_t[clo:chi]
+            chunk_n = fr_n[clo:chi]
+            G = chi - clo
+            # lut spans ALL trees' node ranges (nodeid of any tree may be
+            # looked up by the torch fallback before the -1 filter)
+            max_nodes = int(max(tr.n_nodes for tr in trees))
             lut_np = np.full((T, max_nodes), -1, dtype=np.int32)
             lut_np[chunk_t, chunk_n] = np.arange(G, dtype=np.int32)
             per_tree_count = np.bincount(chunk_t, minlength=T).astype(np.int32)
-            # first group index per tree (tree-major contiguity)
             tree_lo_np = np.zeros(T, dtype=np.int32)
             nz = per_tree_count > 0
             starts = np.cumsum(per_tree_count) - per_tree_count
@@ -408,9 +510,7 @@ def grow_trees(
             best_f, best_bin, best_gain, parent, left_stats = split_scan(
                 hist, cfg.impurity, cfg.min_instances_per_node, cfg.min_info_gain
             )
-            _acc("scan", t0)
-            t0 = _tick()
-            # vectorized bookkeeping: one host transfer per chunk
+            # vectorized bookkeeping: one packed host transfer per chunk
             right_stats = parent - left_stats
             imp_l_t = _impurity_from_stats(left_stats, cfg.impurity)
             imp_r_t = _impurity_from_stats(right_stats, cfg.impurity)
@@ -430,6 +530,8 @@ def grow_trees(
                 ],
                 dim=1,
             ).cpu().numpy()
+            _acc("scan", t0)
+            t0 = _tick()
             bf = packed[:, 0].astype(np.int64)
             bb = packed[:, 1].astype(np.int64)
             bg = packed[:, 2]
@@ -437,58 +539,79 @@ def grow_trees(
             ir = packed[:, 4]
             cl = packed[:, 5]
             cr = packed[:, 6]
-            Sd = left_stats.shape[1]
-            ls = packed[:, 7 : 7 + Sd]
-            rs = packed[:, 7 + Sd : 7 + 2 * Sd]
-            min_c = 2 * cfg.min_instances_per_node
-            for g in range(G):
-                if bf[g] < 0:
-                    continue  # stays leaf
-                t = int(chunk_t[g])
-                nid = int(chunk_n[g])
-                d = int(chunk_d[g])
-                tr = trees[t]
-                gf = int(fs_np[g, bf[g]]) if fs_np is not None else int(bf[g])
-                lid = tr.add_node(ls[g], il[g], cl[g])
-                rid = tr.add_node(rs[g], ir[g], cr[g])
-                tr.feature[nid] = gf
-                tr.thr_bin[nid] = int(bb[g])
-                tr.thr_raw[nid] = float(thr_host[gf, bb[g]])
-                tr.left[nid] = lid
-                tr.right[nid] = rid
-                tr.gain[nid] = float(bg[g])
-                if d + 1 < cfg.max_depth:
-                    if cl[g] >= min_c and il[g] > 0:
-                        next_frontier.append((t, lid, d + 1))
-                    if cr[g] >= min_c and ir[g] > 0:
-                        next_frontier.append((t, rid, d + 1))
+            ls = packed[:, 7 : 7 + S]
+            rs = packed[:, 7 + S : 7 + 2 * S]
+            has_split = bf >= 0
+            sel = np.nonzero(has_split)[0]
+            if len(sel):
+                gf = fs_np[sel, bf[sel]] if fs_np is not None else bf[sel]
+                raws = thr_host[gf, bb[sel]]
+                # per-tree bulk append (chunk is tree-major: contiguous runs)
+                st = chunk_t[sel]
+                run_starts = np.nonzero(np.diff(st, prepend=st[0] - 1))[0]
+                run_ends = np.append(run_starts[1:], len(st))
+                lid = np.empty(len(sel), np.int64)
+                rid = np.empty(len(sel), np.int64)
+                for a, b_ in zip(run_starts, run_ends):
+                    t = int(st[a])
+                    tr = trees[t]
+                    k = b_ - a
+                    stats2d = np.empty((2 * k, S))
+                    stats2d[0::2] = ls[sel[a:b_]]
+                    stats2d[1::2] = rs[sel[a:b_]]
+                    imps = np.empty(2 * k)
+                    imps[0::2] = il[sel[a:b_]]
+                    imps[1::2] = ir[sel[a:b_]]
+                    cnts = np.empty(2 * k)
+                    cnts[0::2] = cl[sel[a:b_]]
+                    cnts[1::2] = cr[sel[a:b_]]
+                    base = tr.bulk_add(stats2d, imps, cnts)
+                    lid[a:b_] = base + 2 * np.arange(k)
+                    rid[a:b_] = lid[a:b_] + 1
+                    tr.set_splits(
+                        chunk_n[sel[a:b_]], gf[a:b_], bb[sel[a:b_]], raws[a:b_],
+                        lid[a:b_], rid[a:b_], bg[sel[a:b_]],
+                    )
+                if level + 1 < cfg.max_depth:
+                    keep_l = (cl[sel] >= min_c) & (il[sel] > 0)
+                    keep_r = (cr[sel] >= min_c) & (ir[sel] > 0)
+                    nxt_t.append(np.repeat(st, 2)[np.ravel(np.column_stack([keep_l, keep_r]))])
+                    nxt_n.append(
+                        np.column_stack([lid, rid]).ravel()[
+                            np.ravel(np.column_stack([keep_l, keep_r]))
+                        ]
+                    )
             _acc("bookkeep", t0)
+        if nxt_t:
+            fr_t = np.concatenate(nxt_t)
+            fr_n = np.concatenate(nxt_n)
+        else:
+            fr_t = np.zeros(0, np.int64)
+            fr_n = np.zeros(0, np.int64)
         # advance node assignments one level, all trees in one launch (K4b);
-        # skipped after the last split level (no further histogram pass reads it)
-        if not next_frontier:
+        # skipped after the last split level (nothing reads it)
+        if not len(fr_t):
             break
         t0 = _tick()
-        feat_cat = torch.tensor(
-            [f for tr in trees for f in tr.feature], dtype=torch.int32, device=dev
-        )
-        thr_cat = torch.tensor(
-            [x for tr in trees for x in tr.thr_bin], dtype=torch.int32, device=dev
-        )
-        left_cat = torch.tensor(
-            [x for tr in trees for x in tr.left], dtype=torch.int32, device=dev
-        )
-        right_cat = torch.tensor(
-            [x for tr in trees for x in tr.right], dtype=torch.int32, device=dev
-        )
-        ofs = [0]
-        for tr in trees:
-            ofs.append(ofs[-1] + tr.n_nodes)
-        node_ofs = torch.tensor(ofs, dtype=torch.int32, device=dev)
+        feat_cat = torch.from_numpy(
+            np.concatenate([tr.feature for tr in trees])
+        ).to(dev)
+        thr_cat = torch.from_numpy(
+            np.concatenate([tr.thr_bin for tr in trees])
+        ).to(dev)
+        left_cat = torch.from_numpy(
+            np.concatenate([tr.left for tr in trees])
+        ).to(dev)
+        right_cat = torch.from_numpy(
+            np.concatenate([tr.right for tr in trees])
+        ).to(dev)
+        ofs = np.zeros(T + 1, np.int32)
+        ofs[1:] = np.cumsum([tr.n_nodes for tr in trees])
+        node_ofs = torch.from_numpy(ofs).to(dev)
         nodeid = ops.node_advance_multi(
             codes, nodeid, feat_cat, thr_cat, left_cat, right_cat, node_ofs
         )
         _acc("advance", t0)
-        frontier = next_frontier
         level += 1
 
     if timing:
@@ -518,18 +641,23 @@ class TreeEnsemble:
         value = np.zeros((T, M, K), dtype=np.float32)
         for ti, tr in enumerate(trees):
             n = tr.n_nodes
+            if n == 0:
+                continue
             feature[ti, :n] = tr.feature
             threshold[ti, :n] = tr.thr_raw
             left[ti, :n] = tr.left
             right[ti, :n] = tr.right
-            for i in range(n):
-                s = tr.stats[i]
-                if mode == "prob_sum":
-                    tot = s.sum()
-                    value[ti, i] = (s / tot) if tot > 0 else 0.0
-                else:
-                    # regression leaf: mean target
-                    value[ti, i, 0] = s[0] / s[2] if s[2] > 0 else 0.0
+            stats = tr.stats
+            if mode == "prob_sum":
+                tot = stats.sum(axis=1, keepdims=True)
+                value[ti, :n] = np.divide(
+                    stats, tot, out=np.zeros_like(stats), where=tot > 0
+                )
+            else:
+                cnt = stats[:, 2]
+                value[ti, :n, 0] = np.divide(
+                    stats[:, 0], cnt, out=np.zeros(n), where=cnt > 0
+                )
         self._feature = torch.from_numpy(feature)
         self._threshold = torch.from_numpy(threshold)
         self._left = torch.from_numpy(left)
@@ -555,7 +683,7 @@ class TreeEnsemble:
         self.to(X.device)
         value = self._value
         if self._tree_weights is not None:
-            value = value * self._tree_weights.view(-1, 1, 1)
+            value = (value * self._tree_weights.view(-1, 1, 1)).contiguous()
         return ops.tree_predict(
             X.to(torch.float32), self._feature, self._threshold, self._left, self._right, value
         )
@@ -565,14 +693,12 @@ class TreeEnsemble:
         tree, summed, normalized again (spark: TreeEnsembleModel.featureImportances)."""
         total = np.zeros(num_features, dtype=np.float64)
         for tr in self.trees:
-            imp = np.zeros(num_features, dtype=np.float64)
-            for i in range(tr.n_nodes):
-                f = tr.feature[i]
-                if f >= 0:
-                    imp[f] += tr.gain[i] * tr.count[i]
-            s = imp.sum()
+            internal = tr.feature >= 0
+            contrib = np.zeros(num_features, dtype=np.float64)
+            np.add.at(contrib, tr.feature[internal], tr.gain[internal] * tr.count[internal])
+            s = contrib.sum()
             if s > 0:
-                total += imp / s
+                total += contrib / s
         s = total.sum()
         return total / s if s > 0 else total
 
