@@ -92,13 +92,20 @@ def bench_rf(args, device):
     Xte, yte = synth_flows(min(rows, 200_000), 41, 5, args.seed + 7777 + rank, device)
     dfte = make_df(Xte, yte)
 
+    import os as _os
+
+    _timing = _os.environ.get("FLOWML_TIMING", "0") == "1"
     state = {}
 
     def step():
+        t0 = time.perf_counter()
         rf = RandomForestClassifier(
             numTrees=args.trees, maxDepth=args.depth, maxBins=32, seed=args.seed
         )
         state["model"] = rf.fit(df)
+        if _timing:
+            sync(device)
+            print(f"step total: {(time.perf_counter() - t0) * 1e3:.1f} ms")
 
     elapsed = timed_loop(step, args.steps, args.warmup, device)
     f1 = MulticlassClassificationEvaluator(metricName="macroF1").evaluate(

@@ -376,8 +376,12 @@ extern "C" void launch_tree_hist_cls(
     float* out, long long N, int F, int T, int M, int Fp, int B, int C,
     int l_max, void* stream) {
   hist_opt_in_big_lds();
-  const int block = 256;
-  const long long rows_per_block = 8192;
+  // 512 threads/block: 8 waves share one LDS histogram copy -> 16 waves/CU
+  // at the 60 KB deep-level histograms (2x the 256-thread occupancy); the
+  // kernel is gather-latency-bound, so resident waves are the lever (PMC:
+  // 49 G wave-cycles at 8 waves/CU — profiles/pmc_summary notes).
+  const int block = 512;
+  const long long rows_per_block = 16384;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t lds = (size_t)l_max * Fp * B * C * sizeof(float);
   hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(tiles, T), dim3(block), lds,
@@ -392,8 +396,8 @@ extern "C" void launch_tree_hist_reg(
     float* out, long long N, int F, int T, int M, int Fp, int B, int l_max,
     void* stream) {
   hist_opt_in_big_lds();
-  const int block = 256;
-  const long long rows_per_block = 8192;
+  const int block = 512;
+  const long long rows_per_block = 16384;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t lds = (size_t)l_max * Fp * B * 3 * sizeof(float);
   hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(tiles, T), dim3(block), lds,
