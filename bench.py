@@ -231,6 +231,48 @@ def bench_mlp(args, device):
     return elapsed, rows * world * args.steps, cfg, "bf16" if use_bf16 else "float32"
 
 
+def bench_lr(args, device):
+    """LogisticRegression: a step = one full-shard fused gradient (K1) +
+    all-reduce + replicated update (one LBFGS-style objective evaluation)."""
+    from flowml import ops
+
+    rank = dist.get_rank()
+    world = dist.get_world_size()
+    rows = args.rows
+    X, y = synth_flows(rows, 41, 5, args.seed + rank, device)
+    yl = y.to(torch.int64)
+    K, F = 5, 41
+    W = torch.zeros(K, F, device=device)
+    b = torch.zeros(K, device=device)
+    total = float(rows * world)
+    lr = 0.5
+
+    def step():
+        nonlocal W, b
+        loss, gW, gB = ops.logreg_grad(X, yl, None, W, b)
+        buf = torch.cat([loss.reshape(1), gW.reshape(-1), gB])
+        dist.all_reduce_(buf)
+        W = W - lr * buf[1 : 1 + K * F].view(K, F) / total
+        b = b - lr * buf[1 + K * F :] / total
+
+    elapsed = timed_loop(step, args.steps, args.warmup, device)
+    pred = (X @ W.T + b).argmax(dim=1).to(torch.float64)
+    from flowml.evaluation import MulticlassClassificationEvaluator
+
+    dfp = FlowFrame({"label": y, "prediction": pred})
+    f1 = MulticlassClassificationEvaluator(metricName="macroF1").evaluate(dfp)
+    cfg = {
+        "model": "LogisticRegression",
+        "rows_per_gpu": rows,
+        "global_rows": rows * world,
+        "features": 41,
+        "classes": 5,
+        "parallelism": f"dp{world}",
+        "macro_f1": round(f1, 4),
+    }
+    return elapsed, rows * world * args.steps, cfg, "float32"
+
+
 def bench_infer(args, device):
     """Config 5: streaming micro-batch ensemble inference (K8)."""
     from flowml.classification import RandomForestClassifier
@@ -295,6 +337,8 @@ def main():
         elapsed, flows, cfg, dtype = bench_gbt(args, device)
     elif args.model == "mlp":
         elapsed, flows, cfg, dtype = bench_mlp(args, device)
+    elif args.model == "lr":
+        elapsed, flows, cfg, dtype = bench_lr(args, device)
     elif args.model == "infer":
         elapsed, flows, cfg, dtype = bench_infer(args, device)
     else:
