@@ -283,19 +283,30 @@ def bench_infer(args, device):
     X, y = synth_flows(train_rows, 41, 5, args.seed + rank, device)
     model = RandomForestClassifier(numTrees=args.trees, maxDepth=args.depth,
                                    seed=args.seed).fit(make_df(X, y))
-    ens = model._ensemble.to(device)
     batch = args.batch
     Xb, _ = synth_flows(batch, 41, 5, args.seed + 31 + rank, device)
+    use_graph = device.type == "cuda"
+    if use_graph:
+        from flowml.streaming import StreamingClassifier
 
-    def step():
-        raw = ens.predict_raw(Xb)
-        _ = raw.argmax(dim=1)
+        sc = StreamingClassifier(model, batch_size=batch, num_features=41,
+                                 device=device, use_graph=True)
+
+        def step():
+            sc.infer(Xb)
+    else:
+        ens = model._ensemble.to(device)
+
+        def step():
+            raw = ens.predict_raw(Xb)
+            _ = raw.argmax(dim=1)
 
     elapsed = timed_loop(step, args.steps, args.warmup, device)
     cfg = {
         "model": f"RandomForest ensemble x{args.trees} inference",
         "micro_batch": batch,
         "features": 41,
+        "hipgraph": use_graph,
         "parallelism": f"dp{world}",
     }
     return elapsed, batch * world * args.steps, cfg, "float32"

@@ -286,3 +286,27 @@ class TestModelFitsGPU:
         m = GBTClassifier(maxIter=20, maxDepth=3, seed=1).fit(df)
         acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(m.transform(df))
         assert acc > 0.8
+
+
+class TestStreamingGraph:
+    def test_hipgraph_capture_matches_eager(self):
+        """Config 5: hipGraph-captured micro-batch path == eager path."""
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+        from flowml.streaming import StreamingClassifier
+
+        r = np.random.default_rng(21)
+        X = torch.from_numpy(r.normal(size=(20_000, 12)).astype(np.float32)).to(DEV)
+        y = ((X[:, 0] + X[:, 3]) > 0).to(torch.float64) + (X[:, 1] > 1).to(torch.float64)
+        model = RandomForestClassifier(numTrees=16, maxDepth=4, seed=0).fit(
+            FlowFrame({"features": X, "label": y}, device=DEV)
+        )
+        B = 4096
+        sc = StreamingClassifier(model, batch_size=B, num_features=12, device=DEV)
+        assert sc._graph is not None
+        for i in range(3):
+            xb = X[i * B : (i + 1) * B]
+            pred, raw = sc.infer(xb)
+            eager = model.predictRaw(xb)
+            assert torch.allclose(raw, eager, atol=1e-5)
+            assert torch.equal(pred.cpu(), eager.argmax(dim=1).cpu())
