@@ -33,7 +33,6 @@ from flowml.classification.tree import (
     TreeEnsemble,
     find_split_thresholds,
     grow_trees,
-    poisson_bag,
 )
 from flowml.frame import FlowFrame
 from flowml.linalg import DenseVector
@@ -162,8 +161,20 @@ class RandomForestClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
         codes = ops.bin_codes(X, thresholds)
         t2 = _tick()
         T = self.getNumTrees()
-        bag = poisson_bag(N, T, self.getOrDefault(self.subsamplingRate),
-                          self.getOrDefault(self.bootstrap), seed, dev)
+        # BaggedPoint weights are computed IN the K2 kernel from a splitmix64
+        # hash of (seed, global row id, tree) — no T×N bag tensor, and the
+        # draw is rank-count-invariant (SURVEY.md §7 hard-part 5)
+        rate = self.getOrDefault(self.subsamplingRate)
+        bootstrap = self.getOrDefault(self.bootstrap)
+        hash_bag = None
+        if bootstrap or rate < 1.0:
+            mode = 2 if bootstrap else 3
+            cdf = ops.poisson_cdf_table(rate, "poisson" if bootstrap else "bernoulli")
+            row_offset = 0
+            if dist.is_initialized() and dist.get_world_size() > 1:
+                counts = dist.all_gather_object(int(N))
+                row_offset = int(sum(counts[: dist.get_rank()]))
+            hash_bag = (mode, cdf.to(dev), (seed * 0x9E3779B9) & 0x7FFFFFFFFFFF, row_offset)
         t3 = _tick()
         row_w = None
         if self.isSet(self.weightCol):
@@ -183,7 +194,8 @@ class RandomForestClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
             is_classification=True,
             n_classes=n_classes,
         )
-        trees = grow_trees(codes, y, bag, thresholds, cfg, row_weights=row_w)
+        trees = grow_trees(codes, y, None, thresholds, cfg, row_weights=row_w,
+                           hash_bag=hash_bag)
         t4 = _tick()
         model = RandomForestClassificationModel(
             trees=trees, numClasses=n_classes, numFeatures=F

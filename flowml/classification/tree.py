@@ -393,10 +393,11 @@ def _make_chunk_slices(fr_t: np.ndarray, T: int, g_max: int, l_cap: int) -> List
 def grow_trees(
     codes: torch.Tensor,  # N×F uint8 (local shard)
     labels_or_targets: torch.Tensor,  # N int (classification) / N float (regression)
-    bag: Optional[torch.Tensor],  # T×N weights (Poisson) or None
+    bag: Optional[torch.Tensor],  # T×N weights or None (explicit tensor path)
     thresholds: torch.Tensor,  # F×(B-1) raw split values
     cfg: GrowthConfig,
     row_weights: Optional[torch.Tensor] = None,  # optional extra per-row weight
+    hash_bag=None,  # (mode, cdf 8×f32, seed, global row offset): in-kernel bagging
 ) -> List[FlatTree]:
     """Grow ``cfg.num_trees`` trees level-synchronously. Replicated across ranks:
     identical return on every rank (histograms are all-reduced)."""
@@ -463,6 +464,7 @@ def grow_trees(
 
     thr_host = thresholds.cpu().numpy()
     level = 0
+    pending_adv = None  # fused K4b: applied inside the first K2 chunk of a level
     while len(fr_t) and level < cfg.max_depth:
         nxt_t: List[np.ndarray] = []
         nxt_n: List[np.ndarray] = []
@@ -501,7 +503,10 @@ def grow_trees(
             hist = ops.tree_hist_nodes(
                 codes, nodeid, lut_d, tree_lo, tree_cnt, labels_u8, bag, G, l_max, B,
                 S if is_cls else 0, fs, targets=targets,
+                advance=pending_adv if ci == 0 else None, hash_bag=hash_bag,
             )
+            if ci == 0:
+                pending_adv = None
             _acc("hist", t0)
             t0 = _tick()
             dist.all_reduce_(hist.reshape(-1))
@@ -588,29 +593,19 @@ def grow_trees(
         else:
             fr_t = np.zeros(0, np.int64)
             fr_n = np.zeros(0, np.int64)
-        # advance node assignments one level, all trees in one launch (K4b);
-        # skipped after the last split level (nothing reads it)
         if not len(fr_t):
             break
+        # stage the K4b advance arrays for the next level's first K2 chunk
+        # (fused into the histogram kernel: one pass over the shard, not two)
         t0 = _tick()
-        feat_cat = torch.from_numpy(
-            np.concatenate([tr.feature for tr in trees])
-        ).to(dev)
-        thr_cat = torch.from_numpy(
-            np.concatenate([tr.thr_bin for tr in trees])
-        ).to(dev)
-        left_cat = torch.from_numpy(
-            np.concatenate([tr.left for tr in trees])
-        ).to(dev)
-        right_cat = torch.from_numpy(
-            np.concatenate([tr.right for tr in trees])
-        ).to(dev)
+        feat_cat = torch.from_numpy(np.concatenate([tr.feature for tr in trees])).to(dev)
+        thr_cat = torch.from_numpy(np.concatenate([tr.thr_bin for tr in trees])).to(dev)
+        left_cat = torch.from_numpy(np.concatenate([tr.left for tr in trees])).to(dev)
+        right_cat = torch.from_numpy(np.concatenate([tr.right for tr in trees])).to(dev)
         ofs = np.zeros(T + 1, np.int32)
         ofs[1:] = np.cumsum([tr.n_nodes for tr in trees])
         node_ofs = torch.from_numpy(ofs).to(dev)
-        nodeid = ops.node_advance_multi(
-            codes, nodeid, feat_cat, thr_cat, left_cat, right_cat, node_ofs
-        )
+        pending_adv = (feat_cat, thr_cat, left_cat, right_cat, node_ofs)
         _acc("advance", t0)
         level += 1
 

@@ -13,14 +13,18 @@
 extern "C" {
 void launch_bin_codes(const float*, const float*, uint8_t*, long long, int,
                       int, void*);
-void launch_tree_hist_cls(const uint8_t*, const int*, const int*, const int*,
+void launch_tree_hist_cls(const uint8_t*, int*, const int*, const int*,
                           const int*, const uint8_t*, const float*, const int*,
-                          float*, long long, int, int, int, int, int, int, int,
-                          void*);
-void launch_tree_hist_reg(const uint8_t*, const int*, const int*, const int*,
+                          float*, const int*, const int*, const int*,
+                          const int*, const int*, int, const float*,
+                          unsigned long long, long long, long long, int, int,
+                          int, int, int, int, int, void*);
+void launch_tree_hist_reg(const uint8_t*, int*, const int*, const int*,
                           const int*, const float*, const float*, const int*,
-                          float*, long long, int, int, int, int, int, int,
-                          void*);
+                          float*, const int*, const int*, const int*,
+                          const int*, const int*, int, const float*,
+                          unsigned long long, long long, long long, int, int,
+                          int, int, int, int, void*);
 void launch_node_advance(const uint8_t*, const int*, int*, const int*,
                          const int*, const int*, const int*, const int*,
                          long long, int, int, void*);
@@ -68,7 +72,12 @@ torch::Tensor tree_hist_cls(torch::Tensor codes, torch::Tensor nodeid,
                             torch::Tensor lut, torch::Tensor tree_lo,
                             torch::Tensor tree_cnt, torch::Tensor labels,
                             torch::Tensor bag, int64_t G, int64_t l_max,
-                            int64_t B, int64_t C, torch::Tensor fs) {
+                            int64_t B, int64_t C, torch::Tensor fs,
+                            torch::Tensor adv_feat, torch::Tensor adv_thr,
+                            torch::Tensor adv_left, torch::Tensor adv_right,
+                            torch::Tensor adv_ofs, int64_t bag_mode,
+                            torch::Tensor bag_cdf, int64_t bag_seed,
+                            int64_t row_offset) {
   CHK(codes);
   CHK(nodeid);
   CHK(lut);
@@ -80,13 +89,21 @@ torch::Tensor tree_hist_cls(torch::Tensor codes, torch::Tensor nodeid,
   const int T = nodeid.size(0);
   const int M = lut.size(1);
   const int Fp = fs.numel() ? fs.size(1) : F;
+  const bool adv = adv_feat.numel() > 0;
   auto out = torch::zeros({G, Fp, B, C}, codes.options().dtype(torch::kFloat32));
   launch_tree_hist_cls(
       codes.data_ptr<uint8_t>(), nodeid.data_ptr<int>(), lut.data_ptr<int>(),
       tree_lo.data_ptr<int>(), tree_cnt.data_ptr<int>(),
       labels.data_ptr<uint8_t>(),
       bag.numel() ? bag.data_ptr<float>() : nullptr,
-      fs.numel() ? fs.data_ptr<int>() : nullptr, out.data_ptr<float>(), N, F,
+      fs.numel() ? fs.data_ptr<int>() : nullptr, out.data_ptr<float>(),
+      adv ? adv_feat.data_ptr<int>() : nullptr,
+      adv ? adv_thr.data_ptr<int>() : nullptr,
+      adv ? adv_left.data_ptr<int>() : nullptr,
+      adv ? adv_right.data_ptr<int>() : nullptr,
+      adv ? adv_ofs.data_ptr<int>() : nullptr, (int)bag_mode,
+      bag_cdf.numel() ? bag_cdf.data_ptr<float>() : nullptr,
+      (unsigned long long)bag_seed, (long long)row_offset, N, F,
       T, M, Fp, (int)B, (int)C, (int)l_max, cur_stream());
   return out;
 }
@@ -95,7 +112,12 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
                             torch::Tensor lut, torch::Tensor tree_lo,
                             torch::Tensor tree_cnt, torch::Tensor targets,
                             torch::Tensor bag, int64_t G, int64_t l_max,
-                            int64_t B, torch::Tensor fs) {
+                            int64_t B, torch::Tensor fs,
+                            torch::Tensor adv_feat, torch::Tensor adv_thr,
+                            torch::Tensor adv_left, torch::Tensor adv_right,
+                            torch::Tensor adv_ofs, int64_t bag_mode,
+                            torch::Tensor bag_cdf, int64_t bag_seed,
+                            int64_t row_offset) {
   CHK(codes);
   CHK(nodeid);
   CHK(lut);
@@ -105,13 +127,21 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
   const int T = nodeid.size(0);
   const int M = lut.size(1);
   const int Fp = fs.numel() ? fs.size(1) : F;
+  const bool adv = adv_feat.numel() > 0;
   auto out = torch::zeros({G, Fp, B, 3}, codes.options().dtype(torch::kFloat32));
   launch_tree_hist_reg(
       codes.data_ptr<uint8_t>(), nodeid.data_ptr<int>(), lut.data_ptr<int>(),
       tree_lo.data_ptr<int>(), tree_cnt.data_ptr<int>(),
       targets.data_ptr<float>(),
       bag.numel() ? bag.data_ptr<float>() : nullptr,
-      fs.numel() ? fs.data_ptr<int>() : nullptr, out.data_ptr<float>(), N, F,
+      fs.numel() ? fs.data_ptr<int>() : nullptr, out.data_ptr<float>(),
+      adv ? adv_feat.data_ptr<int>() : nullptr,
+      adv ? adv_thr.data_ptr<int>() : nullptr,
+      adv ? adv_left.data_ptr<int>() : nullptr,
+      adv ? adv_right.data_ptr<int>() : nullptr,
+      adv ? adv_ofs.data_ptr<int>() : nullptr, (int)bag_mode,
+      bag_cdf.numel() ? bag_cdf.data_ptr<float>() : nullptr,
+      (unsigned long long)bag_seed, (long long)row_offset, N, F,
       T, M, Fp, (int)B, (int)l_max, cur_stream());
   return out;
 }

@@ -51,9 +51,29 @@ extern "C" __global__ void flowml_bin_codes(
 // K2 — classification histogram (LDS-privatized per tree)
 // ---------------------------------------------------------------------------
 
+// splitmix64-based per-(row,tree) bag weight: Poisson via inverse-CDF table
+// (cdf[k] = P(X<=k)) or Bernoulli(cdf[0]=rate). Keyed by the GLOBAL row id,
+// so bagging is rank-count-invariant (SURVEY.md §7 hard-part 5). The torch
+// reference (ops._hash_bag_torch) computes the identical function bitwise.
+__device__ __forceinline__ float flowml_hash_bag(
+    int mode, const float* cdf, unsigned long long seed, long long grow, int t) {
+  unsigned long long x = seed ^ (0x9E3779B97F4A7C15ULL * (unsigned long long)(grow + 1))
+                              ^ (0xBF58476D1CE4E5B9ULL * (unsigned long long)(t + 1));
+  x += 0x9E3779B97F4A7C15ULL;
+  x = (x ^ (x >> 30)) * 0xBF58476D1CE4E5B9ULL;
+  x = (x ^ (x >> 27)) * 0x94D049BB133111EBULL;
+  x = x ^ (x >> 31);
+  const float u = (float)(x >> 40) * (1.0f / 16777216.0f);  // 24-bit uniform
+  if (mode == 3) return u < cdf[0] ? 1.f : 0.f;
+  int k = 0;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) k += (u >= cdf[i]) ? 1 : 0;
+  return (float)k;
+}
+
 extern "C" __global__ void flowml_tree_hist_cls(
     const uint8_t* __restrict__ codes,   // N×F
-    const int* __restrict__ nodeid,      // T×N
+    int* __restrict__ nodeid,            // T×N (updated IN PLACE when advancing)
     const int* __restrict__ lut,         // T×M (node -> chunk group, -1)
     const int* __restrict__ tree_lo,     // T
     const int* __restrict__ tree_cnt,    // T (groups of tree t in chunk)
@@ -61,33 +81,69 @@ extern "C" __global__ void flowml_tree_hist_cls(
     const float* __restrict__ bag,       // T×N or nullptr
     const int* __restrict__ fs,          // G×Fp or nullptr (identity)
     float* __restrict__ out,             // G×Fp×B×C
+    // fused K4b level advance (first chunk of a level): node arrays or null
+    const int* __restrict__ adv_feat, const int* __restrict__ adv_thr,
+    const int* __restrict__ adv_left, const int* __restrict__ adv_right,
+    const int* __restrict__ adv_ofs,
+    // in-kernel hash bagging (BaggedPoint without the T×N tensor):
+    // bag_mode 0 = none/tensor, 2 = Poisson(cdf), 3 = Bernoulli(cdf[0])
+    int bag_mode, const float* __restrict__ bag_cdf,
+    unsigned long long bag_seed, long long row_offset,
     long long N, int F, int M, int Fp, int B, int C, int l_max,
     long long rows_per_block) {
   extern __shared__ float s_hist[];
   const int t = blockIdx.y;
   const int cnt = tree_cnt[t];
-  if (cnt == 0) return;
-  const int hist_sz = l_max * Fp * B * C;
+  const bool advancing = adv_feat != nullptr;
+  if (cnt == 0 && !advancing) return;
+  const int hist_sz = cnt ? l_max * Fp * B * C : 0;
   for (int i = threadIdx.x; i < hist_sz; i += blockDim.x) s_hist[i] = 0.f;
   __syncthreads();
 
   const int g_lo = tree_lo[t];
   const int* lut_t = lut + (long long)t * M;
-  const int* nid_t = nodeid + (long long)t * N;
+  int* nid_t = nodeid + (long long)t * N;
   const float* bag_t = bag ? bag + (long long)t * N : nullptr;
+  const int aofs = advancing ? adv_ofs[t] : 0;
+  const int* af = advancing ? adv_feat + aofs : nullptr;
+  const int* at = advancing ? adv_thr + aofs : nullptr;
+  const int* al = advancing ? adv_left + aofs : nullptr;
+  const int* ar = advancing ? adv_right + aofs : nullptr;
   const long long r0 = blockIdx.x * rows_per_block;
   const long long r1 = min(r0 + rows_per_block, N);
   const int stride_fbc = Fp * B * C;
+  float cdf[8];
+  if (bag_mode >= 2) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) cdf[i] = bag_cdf[i];
+  }
 
   for (long long r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
-    const int nid = nid_t[r];
+    int nid = nid_t[r];
+    if (advancing) {
+      if (nid >= 0) {
+        const int f = af[nid];
+        if (f < 0) {
+          nid = -1;  // reached a leaf: row is done
+        } else {
+          nid = (codes[r * F + f] <= at[nid]) ? al[nid] : ar[nid];
+        }
+        nid_t[r] = nid;
+      }
+      if (cnt == 0) continue;
+    }
     if (nid < 0) continue;
     const int g = lut_t[nid];
     if (g < 0) continue;
     float w = 1.f;
-    if (bag_t) {
-      w = bag_t[r];
+    if (bag_mode >= 2) {
+      w = flowml_hash_bag(bag_mode, cdf, bag_seed, row_offset + r, t);
       if (w == 0.f) continue;
+    }
+    if (bag_t) {
+      const float bw = bag_t[r];
+      if (bw == 0.f) continue;
+      w *= bw;
     }
     const int gl = g - g_lo;
     const int lab = labels[r];
@@ -100,6 +156,7 @@ extern "C" __global__ void flowml_tree_hist_cls(
       atomicAdd(&hrow[(fp * B + b) * C + lab], w);
     }
   }
+  if (cnt == 0) return;
   __syncthreads();
   // flush this tree's [g_lo, g_lo+cnt) groups (same layout in LDS and out)
   float* out_t = out + (long long)g_lo * stride_fbc;
@@ -115,37 +172,71 @@ extern "C" __global__ void flowml_tree_hist_cls(
 // ---------------------------------------------------------------------------
 
 extern "C" __global__ void flowml_tree_hist_reg(
-    const uint8_t* __restrict__ codes, const int* __restrict__ nodeid,
+    const uint8_t* __restrict__ codes, int* __restrict__ nodeid,
     const int* __restrict__ lut, const int* __restrict__ tree_lo,
     const int* __restrict__ tree_cnt, const float* __restrict__ targets,
     const float* __restrict__ bag, const int* __restrict__ fs,
-    float* __restrict__ out, long long N, int F, int M, int Fp, int B,
+    float* __restrict__ out,
+    const int* __restrict__ adv_feat, const int* __restrict__ adv_thr,
+    const int* __restrict__ adv_left, const int* __restrict__ adv_right,
+    const int* __restrict__ adv_ofs,
+    int bag_mode, const float* __restrict__ bag_cdf,
+    unsigned long long bag_seed, long long row_offset,
+    long long N, int F, int M, int Fp, int B,
     int l_max, long long rows_per_block) {
   extern __shared__ float s_hist[];
   const int t = blockIdx.y;
   const int cnt = tree_cnt[t];
-  if (cnt == 0) return;
-  const int hist_sz = l_max * Fp * B * 3;
+  const bool advancing = adv_feat != nullptr;
+  if (cnt == 0 && !advancing) return;
+  const int hist_sz = cnt ? l_max * Fp * B * 3 : 0;
   for (int i = threadIdx.x; i < hist_sz; i += blockDim.x) s_hist[i] = 0.f;
   __syncthreads();
 
   const int g_lo = tree_lo[t];
   const int* lut_t = lut + (long long)t * M;
-  const int* nid_t = nodeid + (long long)t * N;
+  int* nid_t = nodeid + (long long)t * N;
   const float* bag_t = bag ? bag + (long long)t * N : nullptr;
+  const int aofs = advancing ? adv_ofs[t] : 0;
+  const int* af = advancing ? adv_feat + aofs : nullptr;
+  const int* at = advancing ? adv_thr + aofs : nullptr;
+  const int* al = advancing ? adv_left + aofs : nullptr;
+  const int* ar = advancing ? adv_right + aofs : nullptr;
   const long long r0 = blockIdx.x * rows_per_block;
   const long long r1 = min(r0 + rows_per_block, N);
   const int stride_fb3 = Fp * B * 3;
+  float cdf[8];
+  if (bag_mode >= 2) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) cdf[i] = bag_cdf[i];
+  }
 
   for (long long r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
-    const int nid = nid_t[r];
+    int nid = nid_t[r];
+    if (advancing) {
+      if (nid >= 0) {
+        const int f = af[nid];
+        if (f < 0) {
+          nid = -1;
+        } else {
+          nid = (codes[r * F + f] <= at[nid]) ? al[nid] : ar[nid];
+        }
+        nid_t[r] = nid;
+      }
+      if (cnt == 0) continue;
+    }
     if (nid < 0) continue;
     const int g = lut_t[nid];
     if (g < 0) continue;
     float w = 1.f;
-    if (bag_t) {
-      w = bag_t[r];
+    if (bag_mode >= 2) {
+      w = flowml_hash_bag(bag_mode, cdf, bag_seed, row_offset + r, t);
       if (w == 0.f) continue;
+    }
+    if (bag_t) {
+      const float bw = bag_t[r];
+      if (bw == 0.f) continue;
+      w *= bw;
     }
     const int gl = g - g_lo;
     const float tv = targets[r];
@@ -163,6 +254,7 @@ extern "C" __global__ void flowml_tree_hist_reg(
       atomicAdd(cell + 2, w);
     }
   }
+  if (cnt == 0) return;
   __syncthreads();
   float* out_t = out + (long long)g_lo * stride_fb3;
   const int flush_sz = cnt * stride_fb3;
@@ -371,29 +463,35 @@ static void hist_opt_in_big_lds() {
 }
 
 extern "C" void launch_tree_hist_cls(
-    const uint8_t* codes, const int* nodeid, const int* lut, const int* tree_lo,
+    const uint8_t* codes, int* nodeid, const int* lut, const int* tree_lo,
     const int* tree_cnt, const uint8_t* labels, const float* bag, const int* fs,
-    float* out, long long N, int F, int T, int M, int Fp, int B, int C,
+    float* out, const int* adv_feat, const int* adv_thr, const int* adv_left,
+    const int* adv_right, const int* adv_ofs, int bag_mode,
+    const float* bag_cdf, unsigned long long bag_seed, long long row_offset,
+    long long N, int F, int T, int M, int Fp, int B, int C,
     int l_max, void* stream) {
   hist_opt_in_big_lds();
   // 512 threads/block: 8 waves share one LDS histogram copy -> 16 waves/CU
-  // at the 60 KB deep-level histograms (2x the 256-thread occupancy); the
-  // kernel is gather-latency-bound, so resident waves are the lever (PMC:
-  // 49 G wave-cycles at 8 waves/CU — profiles/pmc_summary notes).
+  // at the 60 KB deep-level histograms; the kernel is gather-latency-bound,
+  // so resident waves are the lever (profiles/NOTES.md).
   const int block = 512;
   const long long rows_per_block = 16384;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t lds = (size_t)l_max * Fp * B * C * sizeof(float);
   hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(tiles, T), dim3(block), lds,
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
-                     labels, bag, fs, out, N, F, M, Fp, B, C, l_max,
-                     rows_per_block);
+                     labels, bag, fs, out, adv_feat, adv_thr, adv_left,
+                     adv_right, adv_ofs, bag_mode, bag_cdf, bag_seed,
+                     row_offset, N, F, M, Fp, B, C, l_max, rows_per_block);
 }
 
 extern "C" void launch_tree_hist_reg(
-    const uint8_t* codes, const int* nodeid, const int* lut, const int* tree_lo,
+    const uint8_t* codes, int* nodeid, const int* lut, const int* tree_lo,
     const int* tree_cnt, const float* targets, const float* bag, const int* fs,
-    float* out, long long N, int F, int T, int M, int Fp, int B, int l_max,
+    float* out, const int* adv_feat, const int* adv_thr, const int* adv_left,
+    const int* adv_right, const int* adv_ofs, int bag_mode,
+    const float* bag_cdf, unsigned long long bag_seed, long long row_offset,
+    long long N, int F, int T, int M, int Fp, int B, int l_max,
     void* stream) {
   hist_opt_in_big_lds();
   const int block = 512;
@@ -402,8 +500,9 @@ extern "C" void launch_tree_hist_reg(
   const size_t lds = (size_t)l_max * Fp * B * 3 * sizeof(float);
   hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(tiles, T), dim3(block), lds,
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
-                     targets, bag, fs, out, N, F, M, Fp, B, l_max,
-                     rows_per_block);
+                     targets, bag, fs, out, adv_feat, adv_thr, adv_left,
+                     adv_right, adv_ofs, bag_mode, bag_cdf, bag_seed,
+                     row_offset, N, F, M, Fp, B, l_max, rows_per_block);
 }
 
 extern "C" void launch_node_advance(const uint8_t* codes, const int* nodeid_in,

@@ -101,6 +101,47 @@ def _bin_codes_torch(X, thresholds):
 # ---------------------------------------------------------------------------
 
 
+def poisson_cdf_table(rate: float, mode: str) -> torch.Tensor:
+    """8-entry float32 inverse-CDF table for the in-kernel hash bagging.
+    Poisson: cdf[k] = P(X<=k); Bernoulli: cdf[0] = rate, rest 2.0 (never hit)."""
+    import math
+
+    if mode == "bernoulli":
+        t = [float(rate)] + [2.0] * 7
+    else:
+        acc = 0.0
+        t = []
+        term = math.exp(-rate)
+        for k in range(8):
+            acc += term
+            t.append(min(acc, 1.0))
+            term *= rate / (k + 1)
+    return torch.tensor(t, dtype=torch.float32)
+
+
+def _hash_bag_torch(mode: int, cdf: torch.Tensor, seed: int, grow_ids, t: int):
+    """Bitwise-identical CPU reference of the kernel's splitmix64 bag weight."""
+    import numpy as np
+
+    g = (np.asarray(grow_ids, dtype=np.uint64) + np.uint64(1))
+    with np.errstate(over="ignore"):
+        x = (np.uint64(seed)
+             ^ (np.uint64(0x9E3779B97F4A7C15) * g)
+             ^ (np.uint64(0xBF58476D1CE4E5B9) * np.uint64(t + 1)))
+        x = x + np.uint64(0x9E3779B97F4A7C15)
+        x = (x ^ (x >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+        x = (x ^ (x >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+        x = x ^ (x >> np.uint64(31))
+    u = (x >> np.uint64(40)).astype(np.float32) * np.float32(1.0 / 16777216.0)
+    c = cdf.cpu().numpy()
+    if mode == 3:
+        return (u < c[0]).astype(np.float32)
+    k = np.zeros(u.shape, dtype=np.int32)
+    for i in range(8):
+        k += (u >= c[i]).astype(np.int32)
+    return k.astype(np.float32)
+
+
 def tree_hist_nodes(
     codes: torch.Tensor,   # N×F uint8
     nodeid: torch.Tensor,  # T×N int32 current node per (tree,row), -1 done
@@ -115,34 +156,58 @@ def tree_hist_nodes(
     n_classes: int,
     feat_subset: Optional[torch.Tensor] = None,  # G×F' int32
     targets: Optional[torch.Tensor] = None,      # N float32 (regression)
+    advance=None,          # (feat_cat, thr_cat, left_cat, right_cat, node_ofs)
+    hash_bag=None,         # (mode: 2=poisson|3=bernoulli, cdf 8×f32, seed, row_offset)
 ) -> torch.Tensor:
     """Classification: out[g,f',b,c] += w  (G×F'×B×C float32).
     Regression (targets given): out[g,f',b,(s,s²,n)] += w·(t,t²,1) (G×F'×B×3).
+
+    ``advance``: fused K4b — each (tree,row) is advanced one level down its
+    tree (``nodeid`` is updated IN PLACE) before accumulation; pass it on the
+    FIRST chunk of each level. ``hash_bag``: in-kernel Poisson/Bernoulli
+    BaggedPoint weights keyed by (seed, global row, tree) — no T×N bag tensor.
 
     Reference site: DTStatsAggregator bin update inside findBestSplits
     (SURVEY.md §2.4 K2). The GPU kernel privatizes one tree's frontier
     histograms in LDS per workgroup and flushes with global atomics.
     """
+    dev = codes.device
     if hip_available(codes):
         ext = get_ext()
-        fs = feat_subset if feat_subset is not None else _empty(codes.device, torch.int32)
-        bg = bag if bag is not None else _empty(codes.device)
+        fs = feat_subset if feat_subset is not None else _empty(dev, torch.int32)
+        bg = bag if bag is not None else _empty(dev)
+        if advance is not None:
+            af, at, al, ar, ao = advance
+        else:
+            e = _empty(dev, torch.int32)
+            af = at = al = ar = ao = e
+        if hash_bag is not None:
+            bm, cdf, bseed, rofs = hash_bag
+            cdf = cdf.to(dev)
+        else:
+            bm, cdf, bseed, rofs = 0, _empty(dev), 0, 0
         if targets is None:
             return ext.tree_hist_cls(
                 codes, nodeid, lut, tree_lo, tree_cnt, labels, bg,
                 int(n_groups), int(l_max), int(n_bins), int(n_classes), fs,
+                af, at, al, ar, ao, int(bm), cdf, int(bseed), int(rofs),
             )
         return ext.tree_hist_reg(
             codes, nodeid, lut, tree_lo, tree_cnt, targets, bg,
             int(n_groups), int(l_max), int(n_bins), fs,
+            af, at, al, ar, ao, int(bm), cdf, int(bseed), int(rofs),
         )
+    if advance is not None:
+        af, at, al, ar, ao = advance
+        nodeid.copy_(node_advance_multi(codes, nodeid, af, at, al, ar, ao))
     return _tree_hist_nodes_torch(
-        codes, nodeid, lut, labels, bag, n_groups, n_bins, n_classes, feat_subset, targets
+        codes, nodeid, lut, labels, bag, n_groups, n_bins, n_classes,
+        feat_subset, targets, hash_bag,
     )
 
 
 def _tree_hist_nodes_torch(codes, nodeid, lut, labels, bag, n_groups, n_bins,
-                           n_classes, feat_subset, targets):
+                           n_classes, feat_subset, targets, hash_bag=None):
     N, F = codes.shape
     T = nodeid.shape[0]
     dev = codes.device
@@ -154,6 +219,15 @@ def _tree_hist_nodes_torch(codes, nodeid, lut, labels, bag, n_groups, n_bins,
         live = nid >= 0
         g_t = torch.where(live, lut[t].to(torch.int64)[nid.clamp_min(0)], nid.new_full((N,), -1))
         active = g_t >= 0
+        hw_t = None
+        if hash_bag is not None:
+            bm, cdf, bseed, rofs = hash_bag
+            import numpy as _np
+
+            hw_t = torch.from_numpy(
+                _hash_bag_torch(bm, cdf, bseed, _np.arange(rofs, rofs + N), t)
+            ).to(dev)
+            active = active & (hw_t > 0)
         if bag is not None:
             w_t = bag[t].to(torch.float64)
             active = active & (w_t > 0)
@@ -164,6 +238,8 @@ def _tree_hist_nodes_torch(codes, nodeid, lut, labels, bag, n_groups, n_bins,
         c_a = codes[rows].to(torch.int64)
         w_a = (bag[t][rows].to(torch.float64) if bag is not None
                else torch.ones(len(rows), dtype=torch.float64, device=dev))
+        if hw_t is not None:
+            w_a = w_a * hw_t[rows].to(torch.float64)
         if feat_subset is not None and feat_subset.numel():
             feats = feat_subset.to(torch.int64)[g_a]
             c_a = c_a.gather(1, feats)

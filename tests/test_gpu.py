@@ -310,3 +310,60 @@ class TestStreamingGraph:
             eager = model.predictRaw(xb)
             assert torch.allclose(raw, eager, atol=1e-5)
             assert torch.equal(pred.cpu(), eager.argmax(dim=1).cpu())
+
+
+class TestFusedHistFeatures:
+    def test_hash_bag_parity_bitwise(self):
+        """In-kernel splitmix64 bagging == CPU reference, bitwise."""
+        r = np.random.default_rng(31)
+        N, F, B, C, T = 40_000, 9, 16, 3, 4
+        codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
+        nodeid = torch.zeros((T, N), dtype=torch.int32, device=DEV)
+        lut = torch.arange(T, dtype=torch.int32).reshape(T, 1).to(DEV)
+        tree_lo = torch.arange(T, dtype=torch.int32).to(DEV)
+        tree_cnt = torch.ones(T, dtype=torch.int32).to(DEV)
+        labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8)).to(DEV)
+        cdf = ops.poisson_cdf_table(1.0, "poisson")
+        hb = (2, cdf, 123456789, 1000)
+        gpu = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, None,
+                                  T, 1, B, C, None, hash_bag=hb).cpu()
+        cpu = ops._tree_hist_nodes_torch(codes.cpu(), nodeid.cpu(), lut.cpu(), labels.cpu(),
+                                         None, T, B, C, None, None, hash_bag=hb)
+        assert torch.equal(gpu, cpu)
+        assert float(gpu.sum()) > 0
+
+    def test_fused_advance_parity(self):
+        """advance-inside-K2 == separate node_advance + hist, and nodeid is
+        updated in place identically."""
+        r = np.random.default_rng(32)
+        N, F, B, C, T, M = 30_000, 7, 8, 2, 3, 5
+        codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
+        nodeid0 = torch.zeros((T, N), dtype=torch.int32, device=DEV)
+        # simple trees: root splits f0 at bin 3 -> children 1,2 (leaves at id1, id2)
+        feat = torch.tensor([0, -1, -1] * T, dtype=torch.int32, device=DEV)
+        thr = torch.tensor([3, 0, 0] * T, dtype=torch.int32, device=DEV)
+        left = torch.tensor([1, -1, -1] * T, dtype=torch.int32, device=DEV)
+        right = torch.tensor([2, -1, -1] * T, dtype=torch.int32, device=DEV)
+        ofs = torch.tensor([0, 3, 6, 9], dtype=torch.int32, device=DEV)
+        adv = (feat, thr, left, right, ofs)
+        labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8)).to(DEV)
+        # group = children: lut maps node 1,2 of each tree
+        lut = torch.full((T, 3), -1, dtype=torch.int32)
+        g = 0
+        tree_lo = torch.zeros(T, dtype=torch.int32)
+        tree_cnt = torch.full((T,), 2, dtype=torch.int32)
+        for t in range(T):
+            tree_lo[t] = g
+            lut[t, 1] = g
+            lut[t, 2] = g + 1
+            g += 2
+        lut, tree_lo, tree_cnt = lut.to(DEV), tree_lo.to(DEV), tree_cnt.to(DEV)
+
+        nid_fused = nodeid0.clone()
+        h_fused = ops.tree_hist_nodes(codes, nid_fused, lut, tree_lo, tree_cnt,
+                                      labels, None, g, 2, B, C, None, advance=adv)
+        nid_sep = ops.node_advance_multi(codes, nodeid0.clone(), feat, thr, left, right, ofs)
+        h_sep = ops.tree_hist_nodes(codes, nid_sep, lut, tree_lo, tree_cnt,
+                                    labels, None, g, 2, B, C, None)
+        assert torch.equal(nid_fused.cpu(), nid_sep.cpu())
+        assert torch.equal(h_fused.cpu(), h_sep.cpu())
