@@ -400,7 +400,13 @@ def grow_trees(
     hash_bag=None,  # (mode, cdf 8×f32, seed, global row offset): in-kernel bagging
 ) -> List[FlatTree]:
     """Grow ``cfg.num_trees`` trees level-synchronously. Replicated across ranks:
-    identical return on every rank (histograms are all-reduced)."""
+    identical return on every rank (histograms are all-reduced).
+
+    All nodes of all trees live in ONE struct-of-arrays arena (a FlatTree with
+    global ids + a tree_of map): per-level bookkeeping is a handful of bulk
+    numpy ops, the kernel's node->group lut is 1-D, and the K4b advance arrays
+    are just the arena arrays — no per-tree host loops during growth.
+    """
     N, F = codes.shape
     T = cfg.num_trees
     dev = codes.device
@@ -430,17 +436,21 @@ def grow_trees(
         if timing:
             tmr[key] = tmr.get(key, 0.0) + (_tick() - t0)
 
-    trees = [FlatTree(stat_width=S) for _ in range(T)]
-    nodeid = torch.zeros((T, N), dtype=torch.int32, device=dev)
+    arena = FlatTree(stat_width=S, cap=max(4 * T, 64))
+    tree_of_parts: List[np.ndarray] = [np.arange(T, dtype=np.int64)]
+    # roots are arena nodes 0..T-1; every (tree,row) starts at its root
+    nodeid = (
+        torch.arange(T, dtype=torch.int32, device=dev).unsqueeze(1).expand(T, N).contiguous()
+    )
 
-    # root stats per tree: one K2 pass over feature 0 only (lut: node0 -> t)
-    lut0 = torch.arange(T, dtype=torch.int32, device=dev).reshape(T, 1)
+    # root stats per tree: one K2 pass over feature 0 only (lut: root t -> t)
+    lut0 = torch.arange(T, dtype=torch.int32, device=dev)
     tree_lo0 = torch.arange(T, dtype=torch.int32, device=dev)
     tree_cnt0 = torch.ones(T, dtype=torch.int32, device=dev)
     fs0 = torch.zeros((T, 1), dtype=torch.int32, device=dev)
     h0 = ops.tree_hist_nodes(
         codes, nodeid, lut0, tree_lo0, tree_cnt0, labels_u8, bag, T, 1, B,
-        S if is_cls else 0, fs0, targets=targets,
+        S if is_cls else 0, fs0, targets=targets, hash_bag=hash_bag,
     )
     dist.all_reduce_(h0.reshape(-1))
     root_stats = h0.sum(dim=2)[:, 0, :].to(torch.float64)  # T×S
@@ -450,11 +460,10 @@ def grow_trees(
     ri_np = root_imp.cpu().numpy()
     rc_np = root_cnt.cpu().numpy()
     min_c = 2 * cfg.min_instances_per_node
-    for t in range(T):
-        trees[t].add_node(rs_np[t], float(ri_np[t]), float(rc_np[t]))
+    arena.bulk_add(rs_np, ri_np, rc_np)
     grow_ok = (rc_np >= min_c) & (ri_np > 0) & (cfg.max_depth > 0)
-    fr_t = np.nonzero(grow_ok)[0].astype(np.int64)  # frontier tree ids
-    fr_n = np.zeros(len(fr_t), dtype=np.int64)  # frontier node ids (roots)
+    fr_t = np.nonzero(grow_ok)[0].astype(np.int64)  # frontier tree ids (sorted)
+    fr_n = fr_t.copy()  # global arena ids of the roots
 
     # memory budget → groups per histogram chunk (maxMemoryInMB parity) and
     # per-tree LDS cap (GPU kernel constraint)
@@ -473,11 +482,8 @@ def grow_trees(
             chunk_t = fr_t[clo:chi]
             chunk_n = fr_n[clo:chi]
             G = chi - clo
-            # lut spans ALL trees' node ranges (nodeid of any tree may be
-            # looked up by the torch fallback before the -1 filter)
-            max_nodes = int(max(tr.n_nodes for tr in trees))
-            lut_np = np.full((T, max_nodes), -1, dtype=np.int32)
-            lut_np[chunk_t, chunk_n] = np.arange(G, dtype=np.int32)
+            lut_np = np.full(arena.n_nodes, -1, dtype=np.int32)
+            lut_np[chunk_n] = np.arange(G, dtype=np.int32)
             per_tree_count = np.bincount(chunk_t, minlength=T).astype(np.int32)
             tree_lo_np = np.zeros(T, dtype=np.int32)
             nz = per_tree_count > 0
@@ -546,46 +552,32 @@ def grow_trees(
             cr = packed[:, 6]
             ls = packed[:, 7 : 7 + S]
             rs = packed[:, 7 + S : 7 + 2 * S]
-            has_split = bf >= 0
-            sel = np.nonzero(has_split)[0]
+            sel = np.nonzero(bf >= 0)[0]
             if len(sel):
+                k = len(sel)
                 gf = fs_np[sel, bf[sel]] if fs_np is not None else bf[sel]
                 raws = thr_host[gf, bb[sel]]
-                # per-tree bulk append (chunk is tree-major: contiguous runs)
-                st = chunk_t[sel]
-                run_starts = np.nonzero(np.diff(st, prepend=st[0] - 1))[0]
-                run_ends = np.append(run_starts[1:], len(st))
-                lid = np.empty(len(sel), np.int64)
-                rid = np.empty(len(sel), np.int64)
-                for a, b_ in zip(run_starts, run_ends):
-                    t = int(st[a])
-                    tr = trees[t]
-                    k = b_ - a
-                    stats2d = np.empty((2 * k, S))
-                    stats2d[0::2] = ls[sel[a:b_]]
-                    stats2d[1::2] = rs[sel[a:b_]]
-                    imps = np.empty(2 * k)
-                    imps[0::2] = il[sel[a:b_]]
-                    imps[1::2] = ir[sel[a:b_]]
-                    cnts = np.empty(2 * k)
-                    cnts[0::2] = cl[sel[a:b_]]
-                    cnts[1::2] = cr[sel[a:b_]]
-                    base = tr.bulk_add(stats2d, imps, cnts)
-                    lid[a:b_] = base + 2 * np.arange(k)
-                    rid[a:b_] = lid[a:b_] + 1
-                    tr.set_splits(
-                        chunk_n[sel[a:b_]], gf[a:b_], bb[sel[a:b_]], raws[a:b_],
-                        lid[a:b_], rid[a:b_], bg[sel[a:b_]],
-                    )
+                # bulk-append ALL children of the chunk to the arena at once
+                stats2d = np.empty((2 * k, S))
+                stats2d[0::2] = ls[sel]
+                stats2d[1::2] = rs[sel]
+                imps = np.empty(2 * k)
+                imps[0::2] = il[sel]
+                imps[1::2] = ir[sel]
+                cnts = np.empty(2 * k)
+                cnts[0::2] = cl[sel]
+                cnts[1::2] = cr[sel]
+                base = arena.bulk_add(stats2d, imps, cnts)
+                lid = base + 2 * np.arange(k, dtype=np.int64)
+                rid = lid + 1
+                arena.set_splits(chunk_n[sel], gf, bb[sel], raws, lid, rid, bg[sel])
+                tree_of_parts.append(np.repeat(chunk_t[sel], 2))
                 if level + 1 < cfg.max_depth:
                     keep_l = (cl[sel] >= min_c) & (il[sel] > 0)
                     keep_r = (cr[sel] >= min_c) & (ir[sel] > 0)
-                    nxt_t.append(np.repeat(st, 2)[np.ravel(np.column_stack([keep_l, keep_r]))])
-                    nxt_n.append(
-                        np.column_stack([lid, rid]).ravel()[
-                            np.ravel(np.column_stack([keep_l, keep_r]))
-                        ]
-                    )
+                    mask = np.ravel(np.column_stack([keep_l, keep_r]))
+                    nxt_t.append(np.repeat(chunk_t[sel], 2)[mask])
+                    nxt_n.append(np.column_stack([lid, rid]).ravel()[mask])
             _acc("bookkeep", t0)
         if nxt_t:
             fr_t = np.concatenate(nxt_t)
@@ -595,19 +587,44 @@ def grow_trees(
             fr_n = np.zeros(0, np.int64)
         if not len(fr_t):
             break
-        # stage the K4b advance arrays for the next level's first K2 chunk
-        # (fused into the histogram kernel: one pass over the shard, not two)
+        # stage the K4b advance arrays (= the arena arrays) for the next
+        # level's first K2 chunk — fused: one pass over the shard, not two
         t0 = _tick()
-        feat_cat = torch.from_numpy(np.concatenate([tr.feature for tr in trees])).to(dev)
-        thr_cat = torch.from_numpy(np.concatenate([tr.thr_bin for tr in trees])).to(dev)
-        left_cat = torch.from_numpy(np.concatenate([tr.left for tr in trees])).to(dev)
-        right_cat = torch.from_numpy(np.concatenate([tr.right for tr in trees])).to(dev)
-        ofs = np.zeros(T + 1, np.int32)
-        ofs[1:] = np.cumsum([tr.n_nodes for tr in trees])
-        node_ofs = torch.from_numpy(ofs).to(dev)
-        pending_adv = (feat_cat, thr_cat, left_cat, right_cat, node_ofs)
+        pending_adv = (
+            torch.from_numpy(arena.feature.copy()).to(dev),
+            torch.from_numpy(arena.thr_bin.copy()).to(dev),
+            torch.from_numpy(arena.left.copy()).to(dev),
+            torch.from_numpy(arena.right.copy()).to(dev),
+        )
         _acc("advance", t0)
         level += 1
+
+    # extract per-tree FlatTrees from the arena (once per fit, vectorized)
+    t0 = _tick()
+    tree_of = np.concatenate(tree_of_parts)
+    order = np.argsort(tree_of, kind="stable")  # per-tree contiguous, BFS order
+    bounds = np.searchsorted(tree_of[order], np.arange(T + 1))
+    g2l = np.empty(arena.n_nodes, dtype=np.int64)
+    g2l[order] = np.concatenate(
+        [np.arange(bounds[t + 1] - bounds[t]) for t in range(T)]
+    ) if T else np.zeros(0, np.int64)
+    trees: List[FlatTree] = []
+    for t in range(T):
+        ids = order[bounds[t] : bounds[t + 1]]
+        n = len(ids)
+        tr = FlatTree(stat_width=S, cap=max(n, 4))
+        tr.bulk_add(arena.stats[ids], arena.impurity[ids], arena.count[ids])
+        internal = arena.feature[ids] >= 0
+        li = arena.left[ids]
+        ri = arena.right[ids]
+        tr.feature[:] = arena.feature[ids]
+        tr.thr_bin[:] = arena.thr_bin[ids]
+        tr.thr_raw[:] = arena.thr_raw[ids]
+        tr.gain[:] = arena.gain[ids]
+        tr.left[:] = np.where(internal, g2l[np.where(internal, li, 0)], -1)
+        tr.right[:] = np.where(internal, g2l[np.where(internal, ri, 0)], -1)
+        trees.append(tr)
+    _acc("extract", t0)
 
     if timing:
         print("grow_trees timing:", {k: round(v * 1000, 1) for k, v in tmr.items()})

@@ -16,17 +16,17 @@ void launch_bin_codes(const float*, const float*, uint8_t*, long long, int,
 void launch_tree_hist_cls(const uint8_t*, int*, const int*, const int*,
                           const int*, const uint8_t*, const float*, const int*,
                           float*, const int*, const int*, const int*,
-                          const int*, const int*, int, const float*,
+                          const int*, int, const float*,
                           unsigned long long, long long, long long, int, int,
-                          int, int, int, int, int, void*);
+                          int, int, int, int, void*);
 void launch_tree_hist_reg(const uint8_t*, int*, const int*, const int*,
                           const int*, const float*, const float*, const int*,
                           float*, const int*, const int*, const int*,
-                          const int*, const int*, int, const float*,
+                          const int*, int, const float*,
                           unsigned long long, long long, long long, int, int,
-                          int, int, int, int, void*);
+                          int, int, int, void*);
 void launch_node_advance(const uint8_t*, const int*, int*, const int*,
-                         const int*, const int*, const int*, const int*,
+                         const int*, const int*, const int*,
                          long long, int, int, void*);
 void launch_tree_predict(const float*, const int*, const float*, const int*,
                          const int*, const float*, float*, long long, int, int,
@@ -75,7 +75,7 @@ torch::Tensor tree_hist_cls(torch::Tensor codes, torch::Tensor nodeid,
                             int64_t B, int64_t C, torch::Tensor fs,
                             torch::Tensor adv_feat, torch::Tensor adv_thr,
                             torch::Tensor adv_left, torch::Tensor adv_right,
-                            torch::Tensor adv_ofs, int64_t bag_mode,
+                            int64_t bag_mode,
                             torch::Tensor bag_cdf, int64_t bag_seed,
                             int64_t row_offset) {
   CHK(codes);
@@ -87,7 +87,6 @@ torch::Tensor tree_hist_cls(torch::Tensor codes, torch::Tensor nodeid,
   const long long N = codes.size(0);
   const int F = codes.size(1);
   const int T = nodeid.size(0);
-  const int M = lut.size(1);
   const int Fp = fs.numel() ? fs.size(1) : F;
   const bool adv = adv_feat.numel() > 0;
   auto out = torch::zeros({G, Fp, B, C}, codes.options().dtype(torch::kFloat32));
@@ -100,11 +99,10 @@ torch::Tensor tree_hist_cls(torch::Tensor codes, torch::Tensor nodeid,
       adv ? adv_feat.data_ptr<int>() : nullptr,
       adv ? adv_thr.data_ptr<int>() : nullptr,
       adv ? adv_left.data_ptr<int>() : nullptr,
-      adv ? adv_right.data_ptr<int>() : nullptr,
-      adv ? adv_ofs.data_ptr<int>() : nullptr, (int)bag_mode,
+      adv ? adv_right.data_ptr<int>() : nullptr, (int)bag_mode,
       bag_cdf.numel() ? bag_cdf.data_ptr<float>() : nullptr,
       (unsigned long long)bag_seed, (long long)row_offset, N, F,
-      T, M, Fp, (int)B, (int)C, (int)l_max, cur_stream());
+      T, Fp, (int)B, (int)C, (int)l_max, cur_stream());
   return out;
 }
 
@@ -115,7 +113,7 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
                             int64_t B, torch::Tensor fs,
                             torch::Tensor adv_feat, torch::Tensor adv_thr,
                             torch::Tensor adv_left, torch::Tensor adv_right,
-                            torch::Tensor adv_ofs, int64_t bag_mode,
+                            int64_t bag_mode,
                             torch::Tensor bag_cdf, int64_t bag_seed,
                             int64_t row_offset) {
   CHK(codes);
@@ -125,7 +123,6 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
   const long long N = codes.size(0);
   const int F = codes.size(1);
   const int T = nodeid.size(0);
-  const int M = lut.size(1);
   const int Fp = fs.numel() ? fs.size(1) : F;
   const bool adv = adv_feat.numel() > 0;
   auto out = torch::zeros({G, Fp, B, 3}, codes.options().dtype(torch::kFloat32));
@@ -138,18 +135,16 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
       adv ? adv_feat.data_ptr<int>() : nullptr,
       adv ? adv_thr.data_ptr<int>() : nullptr,
       adv ? adv_left.data_ptr<int>() : nullptr,
-      adv ? adv_right.data_ptr<int>() : nullptr,
-      adv ? adv_ofs.data_ptr<int>() : nullptr, (int)bag_mode,
+      adv ? adv_right.data_ptr<int>() : nullptr, (int)bag_mode,
       bag_cdf.numel() ? bag_cdf.data_ptr<float>() : nullptr,
       (unsigned long long)bag_seed, (long long)row_offset, N, F,
-      T, M, Fp, (int)B, (int)l_max, cur_stream());
+      T, Fp, (int)B, (int)l_max, cur_stream());
   return out;
 }
 
 torch::Tensor node_advance(torch::Tensor codes, torch::Tensor nodeid,
                            torch::Tensor feat, torch::Tensor thr,
-                           torch::Tensor left, torch::Tensor right,
-                           torch::Tensor node_ofs) {
+                           torch::Tensor left, torch::Tensor right) {
   CHK(codes);
   CHK(nodeid);
   const long long N = codes.size(0);
@@ -159,8 +154,7 @@ torch::Tensor node_advance(torch::Tensor codes, torch::Tensor nodeid,
   launch_node_advance(codes.data_ptr<uint8_t>(), nodeid.data_ptr<int>(),
                       out.data_ptr<int>(), feat.data_ptr<int>(),
                       thr.data_ptr<int>(), left.data_ptr<int>(),
-                      right.data_ptr<int>(), node_ofs.data_ptr<int>(), N, F, T,
-                      cur_stream());
+                      right.data_ptr<int>(), N, F, T, cur_stream());
   return out;
 }
 

@@ -2,22 +2,33 @@
 //
 // K4a  bin_codes        — quantile binning: f32 features -> uint8 bin codes
 // K2   tree_hist_cls    — per-node class histograms, ALL trees in one pass:
-//                         one workgroup owns (row-tile × tree), privatizes
+//                         one workgroup owns (tree × row-tile), privatizes
 //                         that tree's frontier histograms in LDS (float
 //                         atomics; counts are integers so sums are exact and
 //                         run-to-run deterministic), flushes with global
-//                         atomics. The #1 hot loop of the flagship config.
+//                         atomics. Optionally fuses the K4b level advance
+//                         (nodeid updated in place) and computes BaggedPoint
+//                         weights from a splitmix64 hash (no bag tensor).
+//                         The #1 hot loop of the flagship config.
 // K2b  tree_hist_reg    — same with (Σt, Σt², n) stats for GBT variance trees
-// K4b  node_advance     — advance every (tree,row) one level down its tree
+// K4b  node_advance     — standalone advance (CPU-parity and GBT paths)
 // K8   tree_predict     — batched ensemble inference: sum of leaf vectors
 // K1   logreg_delta     — fused logits+softmax+delta+loss for LogisticRegression
 // K6   softmax_xent     — row-wise softmax/cross-entropy loss + delta
 //
+// Node identifiers are GLOBAL arena indices (one struct-of-arrays node store
+// for the whole forest — flowml/classification/tree.py), so the node->group
+// lut is 1-D and the advance arrays need no per-tree offsets.
+//
+// Grid layout: blockIdx.x = tree, blockIdx.y = row tile — consecutive blocks
+// process the SAME codes tile for different trees, so the 672 KB tile stays
+// in the XCD's 4 MB L2 across all 100 trees instead of streaming the whole
+// codes matrix from HBM once per tree.
+//
 // Design notes (per /opt/skills guides): wave64; blocks are multiples of 64;
-// codes rows are 41 B at NSL-KDD shape so a (row,tree) touches ~1 cache line;
-// consecutive grid.x blocks share the same row tile across trees -> L2 reuse;
-// LDS histogram for one tree's frontier chunk is capped at 128 KiB by the
-// host-side chunker (flowml/classification/tree.py _hist_lds_cap).
+// 512-thread blocks so 8 waves share one LDS histogram copy (16 waves/CU at
+// the 60 KB deep-level histograms); LDS histogram capped at 128 KiB by the
+// host-side chunker (_hist_lds_cap).
 
 #include <hip/hip_runtime.h>
 #include <stdint.h>
@@ -47,10 +58,6 @@ extern "C" __global__ void flowml_bin_codes(
   }
 }
 
-// ---------------------------------------------------------------------------
-// K2 — classification histogram (LDS-privatized per tree)
-// ---------------------------------------------------------------------------
-
 // splitmix64-based per-(row,tree) bag weight: Poisson via inverse-CDF table
 // (cdf[k] = P(X<=k)) or Bernoulli(cdf[0]=rate). Keyed by the GLOBAL row id,
 // so bagging is rank-count-invariant (SURVEY.md §7 hard-part 5). The torch
@@ -71,28 +78,29 @@ __device__ __forceinline__ float flowml_hash_bag(
   return (float)k;
 }
 
+// ---------------------------------------------------------------------------
+// K2 — classification histogram (LDS-privatized per tree, fused advance/bag)
+// ---------------------------------------------------------------------------
+
 extern "C" __global__ void flowml_tree_hist_cls(
     const uint8_t* __restrict__ codes,   // N×F
-    int* __restrict__ nodeid,            // T×N (updated IN PLACE when advancing)
-    const int* __restrict__ lut,         // T×M (node -> chunk group, -1)
+    int* __restrict__ nodeid,            // T×N global arena ids (advance: in place)
+    const int* __restrict__ lut,         // n_nodes: arena id -> chunk group, -1
     const int* __restrict__ tree_lo,     // T
     const int* __restrict__ tree_cnt,    // T (groups of tree t in chunk)
     const uint8_t* __restrict__ labels,  // N
     const float* __restrict__ bag,       // T×N or nullptr
     const int* __restrict__ fs,          // G×Fp or nullptr (identity)
     float* __restrict__ out,             // G×Fp×B×C
-    // fused K4b level advance (first chunk of a level): node arrays or null
-    const int* __restrict__ adv_feat, const int* __restrict__ adv_thr,
-    const int* __restrict__ adv_left, const int* __restrict__ adv_right,
-    const int* __restrict__ adv_ofs,
-    // in-kernel hash bagging (BaggedPoint without the T×N tensor):
-    // bag_mode 0 = none/tensor, 2 = Poisson(cdf), 3 = Bernoulli(cdf[0])
+    const int* __restrict__ adv_feat,    // arena split arrays or null
+    const int* __restrict__ adv_thr, const int* __restrict__ adv_left,
+    const int* __restrict__ adv_right,
     int bag_mode, const float* __restrict__ bag_cdf,
     unsigned long long bag_seed, long long row_offset,
-    long long N, int F, int M, int Fp, int B, int C, int l_max,
+    long long N, int F, int Fp, int B, int C, int l_max,
     long long rows_per_block) {
   extern __shared__ float s_hist[];
-  const int t = blockIdx.y;
+  const int t = blockIdx.x;
   const int cnt = tree_cnt[t];
   const bool advancing = adv_feat != nullptr;
   if (cnt == 0 && !advancing) return;
@@ -101,15 +109,9 @@ extern "C" __global__ void flowml_tree_hist_cls(
   __syncthreads();
 
   const int g_lo = tree_lo[t];
-  const int* lut_t = lut + (long long)t * M;
   int* nid_t = nodeid + (long long)t * N;
   const float* bag_t = bag ? bag + (long long)t * N : nullptr;
-  const int aofs = advancing ? adv_ofs[t] : 0;
-  const int* af = advancing ? adv_feat + aofs : nullptr;
-  const int* at = advancing ? adv_thr + aofs : nullptr;
-  const int* al = advancing ? adv_left + aofs : nullptr;
-  const int* ar = advancing ? adv_right + aofs : nullptr;
-  const long long r0 = blockIdx.x * rows_per_block;
+  const long long r0 = blockIdx.y * rows_per_block;
   const long long r1 = min(r0 + rows_per_block, N);
   const int stride_fbc = Fp * B * C;
   float cdf[8];
@@ -122,18 +124,18 @@ extern "C" __global__ void flowml_tree_hist_cls(
     int nid = nid_t[r];
     if (advancing) {
       if (nid >= 0) {
-        const int f = af[nid];
+        const int f = adv_feat[nid];
         if (f < 0) {
           nid = -1;  // reached a leaf: row is done
         } else {
-          nid = (codes[r * F + f] <= at[nid]) ? al[nid] : ar[nid];
+          nid = (codes[r * F + f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
         }
         nid_t[r] = nid;
       }
       if (cnt == 0) continue;
     }
     if (nid < 0) continue;
-    const int g = lut_t[nid];
+    const int g = lut[nid];
     if (g < 0) continue;
     float w = 1.f;
     if (bag_mode >= 2) {
@@ -179,13 +181,12 @@ extern "C" __global__ void flowml_tree_hist_reg(
     float* __restrict__ out,
     const int* __restrict__ adv_feat, const int* __restrict__ adv_thr,
     const int* __restrict__ adv_left, const int* __restrict__ adv_right,
-    const int* __restrict__ adv_ofs,
     int bag_mode, const float* __restrict__ bag_cdf,
     unsigned long long bag_seed, long long row_offset,
-    long long N, int F, int M, int Fp, int B,
+    long long N, int F, int Fp, int B,
     int l_max, long long rows_per_block) {
   extern __shared__ float s_hist[];
-  const int t = blockIdx.y;
+  const int t = blockIdx.x;
   const int cnt = tree_cnt[t];
   const bool advancing = adv_feat != nullptr;
   if (cnt == 0 && !advancing) return;
@@ -194,15 +195,9 @@ extern "C" __global__ void flowml_tree_hist_reg(
   __syncthreads();
 
   const int g_lo = tree_lo[t];
-  const int* lut_t = lut + (long long)t * M;
   int* nid_t = nodeid + (long long)t * N;
   const float* bag_t = bag ? bag + (long long)t * N : nullptr;
-  const int aofs = advancing ? adv_ofs[t] : 0;
-  const int* af = advancing ? adv_feat + aofs : nullptr;
-  const int* at = advancing ? adv_thr + aofs : nullptr;
-  const int* al = advancing ? adv_left + aofs : nullptr;
-  const int* ar = advancing ? adv_right + aofs : nullptr;
-  const long long r0 = blockIdx.x * rows_per_block;
+  const long long r0 = blockIdx.y * rows_per_block;
   const long long r1 = min(r0 + rows_per_block, N);
   const int stride_fb3 = Fp * B * 3;
   float cdf[8];
@@ -215,18 +210,18 @@ extern "C" __global__ void flowml_tree_hist_reg(
     int nid = nid_t[r];
     if (advancing) {
       if (nid >= 0) {
-        const int f = af[nid];
+        const int f = adv_feat[nid];
         if (f < 0) {
           nid = -1;
         } else {
-          nid = (codes[r * F + f] <= at[nid]) ? al[nid] : ar[nid];
+          nid = (codes[r * F + f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
         }
         nid_t[r] = nid;
       }
       if (cnt == 0) continue;
     }
     if (nid < 0) continue;
-    const int g = lut_t[nid];
+    const int g = lut[nid];
     if (g < 0) continue;
     float w = 1.f;
     if (bag_mode >= 2) {
@@ -265,21 +260,15 @@ extern "C" __global__ void flowml_tree_hist_reg(
 }
 
 // ---------------------------------------------------------------------------
-// K4b — node advance (all trees)
+// K4b — standalone node advance (global arena ids)
 // ---------------------------------------------------------------------------
 
 extern "C" __global__ void flowml_node_advance(
     const uint8_t* __restrict__ codes, const int* __restrict__ nodeid_in,
     int* __restrict__ nodeid_out, const int* __restrict__ feat,
     const int* __restrict__ thr, const int* __restrict__ left,
-    const int* __restrict__ right, const int* __restrict__ node_ofs,
-    long long N, int F) {
+    const int* __restrict__ right, long long N, int F) {
   const int t = blockIdx.y;
-  const int ofs = node_ofs[t];
-  const int* f_t = feat + ofs;
-  const int* thr_t = thr + ofs;
-  const int* l_t = left + ofs;
-  const int* r_t = right + ofs;
   const int* in_t = nodeid_in + (long long)t * N;
   int* out_t = nodeid_out + (long long)t * N;
   for (long long r = blockIdx.x * (long long)blockDim.x + threadIdx.x; r < N;
@@ -289,13 +278,13 @@ extern "C" __global__ void flowml_node_advance(
       out_t[r] = nid;
       continue;
     }
-    const int f = f_t[nid];
+    const int f = feat[nid];
     if (f < 0) {
       out_t[r] = -1;  // reached a leaf: row is done
       continue;
     }
     const int code = codes[r * (long long)F + f];
-    out_t[r] = (code <= thr_t[nid]) ? l_t[nid] : r_t[nid];
+    out_t[r] = (code <= thr[nid]) ? left[nid] : right[nid];
   }
 }
 
@@ -345,7 +334,7 @@ extern "C" __global__ void flowml_logreg_delta(
     float* __restrict__ delta,    // N×K out
     float* __restrict__ loss,     // 1 (atomic)
     long long N, int F, int K) {
-  extern __shared__ float s_w[];  // K×F weights + K bias + blockDim loss scratch
+  extern __shared__ float s_w[];  // K×F weights + K bias
   const int KF = K * F;
   for (int i = threadIdx.x; i < KF; i += blockDim.x) s_w[i] = W[i];
   float* s_b = s_w + KF;
@@ -466,54 +455,51 @@ extern "C" void launch_tree_hist_cls(
     const uint8_t* codes, int* nodeid, const int* lut, const int* tree_lo,
     const int* tree_cnt, const uint8_t* labels, const float* bag, const int* fs,
     float* out, const int* adv_feat, const int* adv_thr, const int* adv_left,
-    const int* adv_right, const int* adv_ofs, int bag_mode,
-    const float* bag_cdf, unsigned long long bag_seed, long long row_offset,
-    long long N, int F, int T, int M, int Fp, int B, int C,
+    const int* adv_right, int bag_mode, const float* bag_cdf,
+    unsigned long long bag_seed, long long row_offset,
+    long long N, int F, int T, int Fp, int B, int C,
     int l_max, void* stream) {
   hist_opt_in_big_lds();
-  // 512 threads/block: 8 waves share one LDS histogram copy -> 16 waves/CU
-  // at the 60 KB deep-level histograms; the kernel is gather-latency-bound,
-  // so resident waves are the lever (profiles/NOTES.md).
   const int block = 512;
   const long long rows_per_block = 16384;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t lds = (size_t)l_max * Fp * B * C * sizeof(float);
-  hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(tiles, T), dim3(block), lds,
+  hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(T, tiles), dim3(block), lds,
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
                      labels, bag, fs, out, adv_feat, adv_thr, adv_left,
-                     adv_right, adv_ofs, bag_mode, bag_cdf, bag_seed,
-                     row_offset, N, F, M, Fp, B, C, l_max, rows_per_block);
+                     adv_right, bag_mode, bag_cdf, bag_seed, row_offset, N, F,
+                     Fp, B, C, l_max, rows_per_block);
 }
 
 extern "C" void launch_tree_hist_reg(
     const uint8_t* codes, int* nodeid, const int* lut, const int* tree_lo,
     const int* tree_cnt, const float* targets, const float* bag, const int* fs,
     float* out, const int* adv_feat, const int* adv_thr, const int* adv_left,
-    const int* adv_right, const int* adv_ofs, int bag_mode,
-    const float* bag_cdf, unsigned long long bag_seed, long long row_offset,
-    long long N, int F, int T, int M, int Fp, int B, int l_max,
+    const int* adv_right, int bag_mode, const float* bag_cdf,
+    unsigned long long bag_seed, long long row_offset,
+    long long N, int F, int T, int Fp, int B, int l_max,
     void* stream) {
   hist_opt_in_big_lds();
   const int block = 512;
   const long long rows_per_block = 16384;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t lds = (size_t)l_max * Fp * B * 3 * sizeof(float);
-  hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(tiles, T), dim3(block), lds,
+  hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(T, tiles), dim3(block), lds,
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
                      targets, bag, fs, out, adv_feat, adv_thr, adv_left,
-                     adv_right, adv_ofs, bag_mode, bag_cdf, bag_seed,
-                     row_offset, N, F, M, Fp, B, l_max, rows_per_block);
+                     adv_right, bag_mode, bag_cdf, bag_seed, row_offset, N, F,
+                     Fp, B, l_max, rows_per_block);
 }
 
 extern "C" void launch_node_advance(const uint8_t* codes, const int* nodeid_in,
                                     int* nodeid_out, const int* feat,
                                     const int* thr, const int* left,
-                                    const int* right, const int* node_ofs,
-                                    long long N, int F, int T, void* stream) {
+                                    const int* right, long long N, int F,
+                                    int T, void* stream) {
   const int block = 256;
   hipLaunchKernelGGL(flowml_node_advance, dim3(grid_1d(N, block, 2048), T),
                      dim3(block), 0, (hipStream_t)stream, codes, nodeid_in,
-                     nodeid_out, feat, thr, left, right, node_ofs, N, F);
+                     nodeid_out, feat, thr, left, right, N, F);
 }
 
 extern "C" void launch_tree_predict(const float* X, const int* feature,

@@ -144,8 +144,8 @@ def _hash_bag_torch(mode: int, cdf: torch.Tensor, seed: int, grow_ids, t: int):
 
 def tree_hist_nodes(
     codes: torch.Tensor,   # N×F uint8
-    nodeid: torch.Tensor,  # T×N int32 current node per (tree,row), -1 done
-    lut: torch.Tensor,     # T×M int32: node -> chunk group index, -1 inactive
+    nodeid: torch.Tensor,  # T×N int32 GLOBAL arena node id per (tree,row), -1 done
+    lut: torch.Tensor,     # n_nodes int32: arena id -> chunk group index, -1
     tree_lo: torch.Tensor, # T int32: first group index of tree t in the chunk
     tree_cnt: torch.Tensor,  # T int32: number of groups of tree t in the chunk
     labels: torch.Tensor,  # N uint8 class labels (classification)
@@ -156,7 +156,7 @@ def tree_hist_nodes(
     n_classes: int,
     feat_subset: Optional[torch.Tensor] = None,  # G×F' int32
     targets: Optional[torch.Tensor] = None,      # N float32 (regression)
-    advance=None,          # (feat_cat, thr_cat, left_cat, right_cat, node_ofs)
+    advance=None,          # (feat, thr_bin, left, right) — global arena arrays
     hash_bag=None,         # (mode: 2=poisson|3=bernoulli, cdf 8×f32, seed, row_offset)
 ) -> torch.Tensor:
     """Classification: out[g,f',b,c] += w  (G×F'×B×C float32).
@@ -177,10 +177,10 @@ def tree_hist_nodes(
         fs = feat_subset if feat_subset is not None else _empty(dev, torch.int32)
         bg = bag if bag is not None else _empty(dev)
         if advance is not None:
-            af, at, al, ar, ao = advance
+            af, at, al, ar = advance
         else:
             e = _empty(dev, torch.int32)
-            af = at = al = ar = ao = e
+            af = at = al = ar = e
         if hash_bag is not None:
             bm, cdf, bseed, rofs = hash_bag
             cdf = cdf.to(dev)
@@ -190,16 +190,16 @@ def tree_hist_nodes(
             return ext.tree_hist_cls(
                 codes, nodeid, lut, tree_lo, tree_cnt, labels, bg,
                 int(n_groups), int(l_max), int(n_bins), int(n_classes), fs,
-                af, at, al, ar, ao, int(bm), cdf, int(bseed), int(rofs),
+                af, at, al, ar, int(bm), cdf, int(bseed), int(rofs),
             )
         return ext.tree_hist_reg(
             codes, nodeid, lut, tree_lo, tree_cnt, targets, bg,
             int(n_groups), int(l_max), int(n_bins), fs,
-            af, at, al, ar, ao, int(bm), cdf, int(bseed), int(rofs),
+            af, at, al, ar, int(bm), cdf, int(bseed), int(rofs),
         )
     if advance is not None:
-        af, at, al, ar, ao = advance
-        nodeid.copy_(node_advance_multi(codes, nodeid, af, at, al, ar, ao))
+        af, at, al, ar = advance
+        nodeid.copy_(node_advance(codes, nodeid, af, at, al, ar))
     return _tree_hist_nodes_torch(
         codes, nodeid, lut, labels, bag, n_groups, n_bins, n_classes,
         feat_subset, targets, hash_bag,
@@ -214,10 +214,11 @@ def _tree_hist_nodes_torch(codes, nodeid, lut, labels, bag, n_groups, n_bins,
     stat_w = 3 if targets is not None else n_classes
     Fp = feat_subset.shape[1] if (feat_subset is not None and feat_subset.numel()) else F
     out = torch.zeros(n_groups * Fp * n_bins * stat_w, dtype=torch.float64, device=dev)
+    lut64 = lut.to(torch.int64)
     for t in range(T):
         nid = nodeid[t].to(torch.int64)
         live = nid >= 0
-        g_t = torch.where(live, lut[t].to(torch.int64)[nid.clamp_min(0)], nid.new_full((N,), -1))
+        g_t = torch.where(live, lut64[nid.clamp_min(0)], nid.new_full((N,), -1))
         active = g_t >= 0
         hw_t = None
         if hash_bag is not None:
@@ -265,28 +266,21 @@ def _tree_hist_nodes_torch(codes, nodeid, lut, labels, bag, n_groups, n_bins,
 # ---------------------------------------------------------------------------
 
 
-def node_advance_multi(
-    codes: torch.Tensor,     # N×F uint8
-    nodeid: torch.Tensor,    # T×N int32 (updated out-of-place)
-    feat_cat: torch.Tensor,  # ΣM int32: per-node split feature (-1 leaf), trees concatenated
-    thr_cat: torch.Tensor,   # ΣM int32: split bin (go left iff code <= thr)
-    left_cat: torch.Tensor,  # ΣM int32 (node-local child ids)
-    right_cat: torch.Tensor, # ΣM int32
-    node_ofs: torch.Tensor,  # T+1 int32: tree t's nodes at [ofs[t], ofs[t+1])
+def node_advance(
+    codes: torch.Tensor,   # N×F uint8
+    nodeid: torch.Tensor,  # T×N int32 GLOBAL arena ids (out-of-place result)
+    feat: torch.Tensor,    # n_nodes int32: split feature per arena node (-1 leaf)
+    thr: torch.Tensor,     # n_nodes int32: split bin (go left iff code <= thr)
+    left: torch.Tensor,    # n_nodes int32 (global child ids)
+    right: torch.Tensor,   # n_nodes int32
 ) -> torch.Tensor:
     """Advance every (tree,row) one level; leaves become -1 (K4b)."""
     if hip_available(codes):
-        return get_ext().node_advance(
-            codes, nodeid, feat_cat, thr_cat, left_cat, right_cat, node_ofs
-        )
+        return get_ext().node_advance(codes, nodeid, feat, thr, left, right)
     T, N = nodeid.shape
     out = torch.empty_like(nodeid)
-    ofs = node_ofs.tolist()
     for t in range(T):
-        lo, hi = ofs[t], ofs[t + 1]
-        out[t] = _node_advance_torch(
-            codes, nodeid[t], feat_cat[lo:hi], thr_cat[lo:hi], left_cat[lo:hi], right_cat[lo:hi]
-        )
+        out[t] = _node_advance_torch(codes, nodeid[t], feat, thr, left, right)
     return out
 
 

@@ -41,15 +41,17 @@ class TestBinCodes:
 class TestTreeHist:
     def _mk(self, r, N=50_000, F=41, T=8, M=16, B=32, C=5, per_tree=4, Fp=6):
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
-        nodeid = torch.from_numpy(r.integers(-1, M, (T, N)).astype(np.int32)).to(DEV)
-        lut = torch.full((T, M), -1, dtype=torch.int32)
+        local = r.integers(-1, M, (T, N)).astype(np.int32)
+        glob = np.where(local >= 0, local + (np.arange(T)[:, None] * M), -1)
+        nodeid = torch.from_numpy(glob.astype(np.int32)).to(DEV)
+        lut = torch.full((T * M,), -1, dtype=torch.int32)
         tree_lo = torch.zeros(T, dtype=torch.int32)
         tree_cnt = torch.zeros(T, dtype=torch.int32)
         g = 0
         for t in range(T):
             tree_lo[t] = g
             for nid in sorted(r.choice(M, per_tree, replace=False)):
-                lut[t, nid] = g
+                lut[t * M + nid] = g
                 g += 1
             tree_cnt[t] = per_tree
         G = g
@@ -82,7 +84,7 @@ class TestTreeHist:
         N, F, B, C = 30_000, 11, 16, 3
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
         nodeid = torch.zeros((1, N), dtype=torch.int32, device=DEV)
-        lut = torch.zeros((1, 1), dtype=torch.int32, device=DEV)
+        lut = torch.zeros(1, dtype=torch.int32, device=DEV)
         tree_lo = torch.zeros(1, dtype=torch.int32, device=DEV)
         tree_cnt = torch.ones(1, dtype=torch.int32, device=DEV)
         labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8)).to(DEV)
@@ -97,7 +99,7 @@ class TestTreeHist:
         N, F, B = 40_000, 9, 32
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
         nodeid = torch.from_numpy(r.integers(-1, 2, (1, N)).astype(np.int32)).to(DEV)
-        lut = torch.tensor([[0, 1]], dtype=torch.int32, device=DEV)
+        lut = torch.tensor([0, 1], dtype=torch.int32, device=DEV)
         tree_lo = torch.zeros(1, dtype=torch.int32, device=DEV)
         tree_cnt = torch.full((1,), 2, dtype=torch.int32, device=DEV)
         targets = torch.from_numpy(r.normal(size=N).astype(np.float32)).to(DEV)
@@ -113,26 +115,26 @@ class TestNodeAdvanceGPU:
     def test_parity(self):
         r = np.random.default_rng(5)
         N, F, T = 80_000, 13, 6
-        M = 7
+        M = 7  # arena nodes per tree: global id = t*M + m
         codes = torch.from_numpy(r.integers(0, 32, (N, F)).astype(np.uint8)).to(DEV)
-        nodeid = torch.from_numpy(r.integers(-1, M, (T, N)).astype(np.int32)).to(DEV)
-        feat, thr, left, right, ofs = [], [], [], [], [0]
+        local = r.integers(-1, M, (T, N)).astype(np.int32)
+        glob = np.where(local >= 0, local + (np.arange(T)[:, None] * M), -1)
+        nodeid = torch.from_numpy(glob.astype(np.int32)).to(DEV)
+        feat, thr, left, right = [], [], [], []
         for t in range(T):
             for m in range(M):
                 internal = r.random() < 0.6
                 feat.append(int(r.integers(0, F)) if internal else -1)
                 thr.append(int(r.integers(0, 31)))
-                left.append(int(r.integers(0, M)))
-                right.append(int(r.integers(0, M)))
-            ofs.append(len(feat))
+                left.append(t * M + int(r.integers(0, M)))
+                right.append(t * M + int(r.integers(0, M)))
         feat = torch.tensor(feat, dtype=torch.int32, device=DEV)
         thr = torch.tensor(thr, dtype=torch.int32, device=DEV)
         left = torch.tensor(left, dtype=torch.int32, device=DEV)
         right = torch.tensor(right, dtype=torch.int32, device=DEV)
-        ofs = torch.tensor(ofs, dtype=torch.int32, device=DEV)
-        gpu = ops.node_advance_multi(codes, nodeid, feat, thr, left, right, ofs).cpu()
-        cpu = ops.node_advance_multi(codes.cpu(), nodeid.cpu(), feat.cpu(), thr.cpu(),
-                                     left.cpu(), right.cpu(), ofs.cpu())
+        gpu = ops.node_advance(codes, nodeid, feat, thr, left, right).cpu()
+        cpu = ops.node_advance(codes.cpu(), nodeid.cpu(), feat.cpu(), thr.cpu(),
+                               left.cpu(), right.cpu())
         assert torch.equal(gpu, cpu)
 
 
@@ -318,8 +320,9 @@ class TestFusedHistFeatures:
         r = np.random.default_rng(31)
         N, F, B, C, T = 40_000, 9, 16, 3, 4
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
-        nodeid = torch.zeros((T, N), dtype=torch.int32, device=DEV)
-        lut = torch.arange(T, dtype=torch.int32).reshape(T, 1).to(DEV)
+        nodeid = (torch.arange(T, dtype=torch.int32).unsqueeze(1)
+                  .expand(T, N).contiguous().to(DEV))
+        lut = torch.arange(T, dtype=torch.int32).to(DEV)
         tree_lo = torch.arange(T, dtype=torch.int32).to(DEV)
         tree_cnt = torch.ones(T, dtype=torch.int32).to(DEV)
         labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8)).to(DEV)
@@ -336,33 +339,32 @@ class TestFusedHistFeatures:
         """advance-inside-K2 == separate node_advance + hist, and nodeid is
         updated in place identically."""
         r = np.random.default_rng(32)
-        N, F, B, C, T, M = 30_000, 7, 8, 2, 3, 5
+        N, F, B, C, T, M = 30_000, 7, 8, 2, 3, 3
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
-        nodeid0 = torch.zeros((T, N), dtype=torch.int32, device=DEV)
-        # simple trees: root splits f0 at bin 3 -> children 1,2 (leaves at id1, id2)
+        # arena: per tree t, root = 3t splits f0 at bin 3 -> leaves 3t+1, 3t+2
+        nodeid0 = (torch.arange(T, dtype=torch.int32) * M).unsqueeze(1) \
+            .expand(T, N).contiguous().to(DEV)
         feat = torch.tensor([0, -1, -1] * T, dtype=torch.int32, device=DEV)
         thr = torch.tensor([3, 0, 0] * T, dtype=torch.int32, device=DEV)
-        left = torch.tensor([1, -1, -1] * T, dtype=torch.int32, device=DEV)
-        right = torch.tensor([2, -1, -1] * T, dtype=torch.int32, device=DEV)
-        ofs = torch.tensor([0, 3, 6, 9], dtype=torch.int32, device=DEV)
-        adv = (feat, thr, left, right, ofs)
+        left = torch.cat([torch.tensor([t * M + 1, -1, -1], dtype=torch.int32) for t in range(T)]).to(DEV)
+        right = torch.cat([torch.tensor([t * M + 2, -1, -1], dtype=torch.int32) for t in range(T)]).to(DEV)
+        adv = (feat, thr, left, right)
         labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8)).to(DEV)
-        # group = children: lut maps node 1,2 of each tree
-        lut = torch.full((T, 3), -1, dtype=torch.int32)
+        lut = torch.full((T * M,), -1, dtype=torch.int32)
         g = 0
         tree_lo = torch.zeros(T, dtype=torch.int32)
         tree_cnt = torch.full((T,), 2, dtype=torch.int32)
         for t in range(T):
             tree_lo[t] = g
-            lut[t, 1] = g
-            lut[t, 2] = g + 1
+            lut[t * M + 1] = g
+            lut[t * M + 2] = g + 1
             g += 2
         lut, tree_lo, tree_cnt = lut.to(DEV), tree_lo.to(DEV), tree_cnt.to(DEV)
 
         nid_fused = nodeid0.clone()
         h_fused = ops.tree_hist_nodes(codes, nid_fused, lut, tree_lo, tree_cnt,
                                       labels, None, g, 2, B, C, None, advance=adv)
-        nid_sep = ops.node_advance_multi(codes, nodeid0.clone(), feat, thr, left, right, ofs)
+        nid_sep = ops.node_advance(codes, nodeid0.clone(), feat, thr, left, right)
         h_sep = ops.tree_hist_nodes(codes, nid_sep, lut, tree_lo, tree_cnt,
                                     labels, None, g, 2, B, C, None)
         assert torch.equal(nid_fused.cpu(), nid_sep.cpu())

@@ -57,7 +57,8 @@ class TestBinCodes:
 
 
 def naive_hist(codes, nodeid, lut, labels, bag, G, B, C, fs, targets=None):
-    """Independent python-loop oracle for K2 (classification & regression)."""
+    """Independent python-loop oracle for K2 (classification & regression).
+    nodeid holds GLOBAL arena ids; lut is 1-D arena id -> group."""
     N, F = codes.shape
     T = nodeid.shape[0]
     Fp = fs.shape[1] if fs is not None else F
@@ -68,7 +69,7 @@ def naive_hist(codes, nodeid, lut, labels, bag, G, B, C, fs, targets=None):
             nid = int(nodeid[t, i])
             if nid < 0:
                 continue
-            g = int(lut[t, nid])
+            g = int(lut[nid])
             if g < 0:
                 continue
             w = float(bag[t, i]) if bag is not None else 1.0
@@ -91,17 +92,19 @@ class TestTreeHistNodes:
     def test_bitwise_vs_naive(self):
         r = np.random.default_rng(3)
         N, F, T, B, C = 80, 5, 3, 8, 3
-        M = 4  # nodes per tree
+        M = 4  # nodes per tree (global arena id = t*M + local)
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8))
-        nodeid = torch.from_numpy(r.integers(-1, M, (T, N)).astype(np.int32))
-        # groups: tree-major contiguous, 2 active nodes per tree
-        lut = torch.full((T, M), -1, dtype=torch.int32)
+        local = r.integers(-1, M, (T, N)).astype(np.int32)
+        glob = np.where(local >= 0, local + (np.arange(T)[:, None] * M), -1)
+        nodeid = torch.from_numpy(glob.astype(np.int32))
+        # groups: tree-major contiguous, 2 active nodes per tree; 1-D lut
+        lut = torch.full((T * M,), -1, dtype=torch.int32)
         g = 0
         tree_lo = torch.zeros(T, dtype=torch.int32)
         for t in range(T):
             tree_lo[t] = g
             for nid in sorted(r.choice(M, 2, replace=False)):
-                lut[t, nid] = g
+                lut[t * M + nid] = g
                 g += 1
         G = g
         labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8))
@@ -121,7 +124,7 @@ class TestTreeHistNodes:
         N, F, B, C = 50, 4, 6, 2
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8))
         nodeid = torch.zeros((1, N), dtype=torch.int32)
-        lut = torch.zeros((1, 1), dtype=torch.int32)
+        lut = torch.zeros(1, dtype=torch.int32)
         tree_lo = torch.zeros(1, dtype=torch.int32)
         labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8))
         tree_cnt = torch.ones(1, dtype=torch.int32)
@@ -136,7 +139,7 @@ class TestTreeHistNodes:
         N, F, B = 40, 3, 4
         codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8))
         nodeid = torch.zeros((1, N), dtype=torch.int32)
-        lut = torch.zeros((1, 1), dtype=torch.int32)
+        lut = torch.zeros(1, dtype=torch.int32)
         tree_lo = torch.zeros(1, dtype=torch.int32)
         targets = torch.from_numpy(r.normal(size=N).astype(np.float32))
         labels = torch.zeros(N, dtype=torch.uint8)
@@ -153,22 +156,21 @@ class TestTreeHistNodes:
 
 
 class TestNodeAdvance:
-    def test_advance_multi(self):
-        # tree0: node0 splits on f0 at bin<=1 -> node1(leaf), node2 splits f1<=0 -> 3,4
-        # tree1: node0 is a leaf
+    def test_advance_global(self):
+        # arena: tree0 root=node0 splits f0 at bin<=1 -> 1(leaf), 2 splits f1<=0 -> 3,4
+        #        tree1 root=node5 is a leaf
         codes = torch.tensor([[0, 0], [2, 0], [2, 1]], dtype=torch.uint8)
-        nodeid = torch.zeros((2, 3), dtype=torch.int32)
+        nodeid = torch.tensor([[0, 0, 0], [5, 5, 5]], dtype=torch.int32)
         feat = torch.tensor([0, -1, 1, -1, -1, -1], dtype=torch.int32)
         thr = torch.tensor([1, 0, 0, 0, 0, 0], dtype=torch.int32)
         left = torch.tensor([1, -1, 3, -1, -1, -1], dtype=torch.int32)
         right = torch.tensor([2, -1, 4, -1, -1, -1], dtype=torch.int32)
-        ofs = torch.tensor([0, 5, 6], dtype=torch.int32)
-        n1 = ops.node_advance_multi(codes, nodeid, feat, thr, left, right, ofs)
+        n1 = ops.node_advance(codes, nodeid, feat, thr, left, right)
         assert n1[0].tolist() == [1, 2, 2]
         assert n1[1].tolist() == [-1, -1, -1]  # leaf root -> done
-        n2 = ops.node_advance_multi(codes, n1, feat, thr, left, right, ofs)
+        n2 = ops.node_advance(codes, n1, feat, thr, left, right)
         assert n2[0].tolist() == [-1, 3, 4]
-        n3 = ops.node_advance_multi(codes, n2, feat, thr, left, right, ofs)
+        n3 = ops.node_advance(codes, n2, feat, thr, left, right)
         assert n3[0].tolist() == [-1, -1, -1]
 
 
