@@ -170,6 +170,16 @@ class MultilayerPerceptronClassifier(_MLPParams, ProbabilisticClassifier):
                 f"features dim {X.shape[1]} != layers[0] {layers[0]}"
             )
         activation = self.getActivation()
+        # north-star compute dtype on GPU is bf16 (BASELINE.json:8): cast and
+        # pre-pad the shard once; master weights/optimizer state stay fp32
+        use_bf16 = X.is_cuda
+        if use_bf16:
+            K0 = layers[0]
+            Kp = ((K0 + 31) // 32) * 32
+            Xc = torch.zeros(X.shape[0], Kp, dtype=torch.bfloat16, device=dev)
+            Xc[:, :K0] = X.to(torch.bfloat16)
+        else:
+            Xc = X
 
         # global row count (loss normalizer)
         cnt = torch.tensor([float(len(X))], dtype=torch.float64, device=dev)
@@ -193,8 +203,12 @@ class MultilayerPerceptronClassifier(_MLPParams, ProbabilisticClassifier):
 
         def f_grad(theta: torch.Tensor):
             Ws, bs = _unflatten(theta, layers)
-            loss_s, gWs, gBs = ops.mlp_grad(X, y, Ws, bs, activation=activation)
-            flat_g = _flatten(gWs, gBs)
+            if use_bf16:
+                Ws = [w.to(torch.bfloat16) for w in Ws]
+            loss_s, gWs, gBs = ops.mlp_grad(
+                Xc, y, Ws, bs, activation=activation, in_features=layers[0]
+            )
+            flat_g = _flatten([g.to(torch.float32) for g in gWs], gBs)
             buf = torch.cat([loss_s.reshape(1), flat_g])
             dist.all_reduce_(buf)
             return float(buf[0]) / total, buf[1:] / total

@@ -229,7 +229,13 @@ torch::Tensor gemm_bias_act(torch::Tensor X, torch::Tensor W, torch::Tensor b,
   CHK(X);
   CHK(W);
   CHK(b);
-  TORCH_CHECK(X.size(1) == W.size(1), "K mismatch");
+  // X may come PRE-padded to the MFMA K multiple (32 bf16 / 4 f32) so the
+  // big operand isn't re-padded on every call; W (small) is padded here.
+  const int64_t kmult = X.scalar_type() == torch::kBFloat16 ? 32 : 4;
+  const int64_t Kw = W.size(1);
+  const int64_t Kw_p = ((Kw + kmult - 1) / kmult) * kmult;
+  TORCH_CHECK(X.size(1) == Kw || X.size(1) == Kw_p,
+              "K mismatch (X may be pre-padded to the next multiple of ", kmult, ")");
   const long long N = X.size(0);
   const int M = W.size(0);
   TORCH_CHECK(M <= 128, "gemm_bias_act supports M <= 128 (MLP-stack shapes)");

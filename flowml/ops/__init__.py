@@ -451,6 +451,7 @@ def mlp_grad(
     bs: List[torch.Tensor],
     activation: str = "relu",
     w: Optional[torch.Tensor] = None,
+    in_features: Optional[int] = None,
 ):
     """Fused forward+backward for the softmax/cross-entropy MLP.
 
@@ -472,6 +473,8 @@ def mlp_grad(
     for i in range(L - 1, -1, -1):
         a = acts[i]
         gWs[i] = matmul_tn(delta, a)          # K7 wgrad: (N×K)ᵀ·(N×F)
+        if i == 0 and in_features is not None and gWs[0].shape[1] != in_features:
+            gWs[0] = gWs[0][:, :in_features].contiguous()  # drop X pad columns
         gBs[i] = delta.sum(dim=0).to(torch.float32)
         if i > 0:
             dx = delta @ Ws[i]                # K7 dgrad (plain GEMM: rocBLAS)
@@ -516,6 +519,10 @@ def matmul_tn(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
 
     This is a *plain* GEMM (no fusion), so rocBLAS via torch.matmul is the
     sanctioned path on GPU (north star: "hipBLASLt/rocBLAS only for plain
-    library GEMMs"); fp32 accumulate via explicit casts.
+    library GEMMs"). bf16 operands stay bf16 through rocBLAS (fp32 internal
+    accumulation) — the fp32 up-cast of a 1M-row activation would cost more
+    HBM than the GEMM itself; only the small K×F result is up-cast.
     """
+    if A.is_cuda and A.dtype == torch.bfloat16 and B.dtype == torch.bfloat16:
+        return (A.T @ B).to(torch.float32)
     return A.T.to(torch.float32) @ B.to(torch.float32)
