@@ -518,29 +518,10 @@ def grow_trees(
             dist.all_reduce_(hist.reshape(-1))
             _acc("allreduce", t0)
             t0 = _tick()
-            best_f, best_bin, best_gain, parent, left_stats = split_scan(
+            # K3 split-gain scan: packed per-node results, one host transfer
+            packed = ops.split_scan_packed(
                 hist, cfg.impurity, cfg.min_instances_per_node, cfg.min_info_gain
-            )
-            # vectorized bookkeeping: one packed host transfer per chunk
-            right_stats = parent - left_stats
-            imp_l_t = _impurity_from_stats(left_stats, cfg.impurity)
-            imp_r_t = _impurity_from_stats(right_stats, cfg.impurity)
-            cnt_l_t = _counts_of(left_stats)
-            cnt_r_t = _counts_of(right_stats)
-            packed = torch.cat(
-                [
-                    best_f.to(torch.float64).reshape(G, 1),
-                    best_bin.to(torch.float64).reshape(G, 1),
-                    best_gain.reshape(G, 1),
-                    imp_l_t.reshape(G, 1),
-                    imp_r_t.reshape(G, 1),
-                    cnt_l_t.reshape(G, 1),
-                    cnt_r_t.reshape(G, 1),
-                    left_stats,
-                    right_stats,
-                ],
-                dim=1,
-            ).cpu().numpy()
+            ).cpu().numpy().astype(np.float64)
             _acc("scan", t0)
             t0 = _tick()
             bf = packed[:, 0].astype(np.int64)

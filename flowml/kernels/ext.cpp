@@ -36,6 +36,8 @@ void launch_logreg_delta(const float*, const int*, const float*, const float*,
                          void*);
 void launch_softmax_xent(const float*, const int*, const float*, float*,
                          float*, long long, int, void*);
+void launch_split_scan(const float*, float*, int, int, int, int, int, int,
+                       double, void*);
 void launch_gemm_bias_act_bf16(const void*, const void*, const float*, void*,
                                long long, int, int, int, void*);
 void launch_gemm_bias_act_f32(const float*, const float*, const float*, float*,
@@ -140,6 +142,21 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
       (unsigned long long)bag_seed, (long long)row_offset, N, F,
       T, Fp, (int)B, (int)l_max, cur_stream());
   return out;
+}
+
+torch::Tensor split_scan(torch::Tensor hist, int64_t imp_code,
+                         int64_t min_instances, double min_info_gain) {
+  CHK(hist);
+  const int G = hist.size(0);
+  const int Fp = hist.size(1);
+  const int B = hist.size(2);
+  const int S = hist.size(3);
+  TORCH_CHECK(S <= 8, "split_scan supports up to 8 stats");
+  auto packed = torch::empty({G, 7 + 2 * S}, hist.options());
+  launch_split_scan(hist.data_ptr<float>(), packed.data_ptr<float>(), G, Fp, B,
+                    S, (int)imp_code, (int)min_instances, min_info_gain,
+                    cur_stream());
+  return packed;
 }
 
 torch::Tensor node_advance(torch::Tensor codes, torch::Tensor nodeid,
@@ -276,6 +293,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tree_hist_cls", &tree_hist_cls, "K2 class histograms (all trees)");
   m.def("tree_hist_reg", &tree_hist_reg, "K2b regression histograms");
   m.def("node_advance", &node_advance, "K4b node-id advance");
+  m.def("split_scan", &split_scan, "K3 split-gain scan (packed)");
   m.def("tree_predict", &tree_predict, "K8 ensemble inference");
   m.def("logreg_delta", &logreg_delta, "K1 fused logistic delta");
   m.def("softmax_xent_grad", &softmax_xent_grad, "K6 softmax/xent grad");

@@ -532,3 +532,161 @@ extern "C" void launch_softmax_xent(const float* logits, const int* y,
                      dim3(block), 0, (hipStream_t)stream, logits, y, w, delta,
                      loss, N, K);
 }
+
+// ---------------------------------------------------------------------------
+// K3 — split-gain scan: per (node, feature) prefix sums over bins, impurity
+// gain, argmax over (feature, bin). One workgroup per node; the node's
+// histogram slab is staged in LDS; gains are computed in DOUBLE with exactly
+// the torch reference's formula order (flowml.classification.tree.split_scan)
+// so the chosen splits match the CPU path bitwise (counts are integer-valued
+// f32 -> exact in double). Output per node: packed row
+// [best_f, best_bin, gain, imp_l, imp_r, cnt_l, cnt_r, left_stats[S], right_stats[S]].
+// ---------------------------------------------------------------------------
+
+__device__ __forceinline__ double flowml_impurity(const double* s, int S, int code) {
+  if (code <= 1) {  // 0 = gini, 1 = entropy (class counts)
+    double n = 0;
+    for (int i = 0; i < S; ++i) n += s[i];
+    if (n < 1e-12) n = 1e-12;
+    if (code == 0) {
+      double acc = 1.0;
+      for (int i = 0; i < S; ++i) {
+        const double p = s[i] / n;
+        acc -= p * p;
+      }
+      return acc;
+    }
+    double acc = 0.0;
+    for (int i = 0; i < S; ++i) {
+      const double p = s[i] / n;
+      if (p > 0) {
+        const double pc = p < 1e-30 ? 1e-30 : p;
+        acc -= p * log2(pc);
+      }
+    }
+    return acc;
+  }
+  // 2 = variance on (sum, sum2, n)
+  double nn = s[2] < 1e-12 ? 1e-12 : s[2];
+  const double m = s[0] / nn;
+  double v = s[1] / nn - m * m;
+  return v < 0 ? 0 : v;
+}
+
+__device__ __forceinline__ double flowml_count_of(const double* s, int S, int code) {
+  if (code == 2) return s[2];
+  double n = 0;
+  for (int i = 0; i < S; ++i) n += s[i];
+  return n;
+}
+
+#define K3_MAXS 8
+
+extern "C" __global__ void flowml_split_scan(
+    const float* __restrict__ hist,  // G×Fp×B×S
+    float* __restrict__ packed,      // G×(7+2S)
+    int G, int Fp, int B, int S, int imp_code, int min_instances,
+    double min_info_gain) {
+  extern __shared__ float s_h[];  // Fp×B×S staged slab
+  const int g = blockIdx.x;
+  const int slab = Fp * B * S;
+  const float* hg = hist + (long long)g * slab;
+  for (int i = threadIdx.x; i < slab; i += blockDim.x) s_h[i] = hg[i];
+  __syncthreads();
+
+  // parent stats from feature 0 (identical across features)
+  double parent[K3_MAXS];
+  for (int i = 0; i < S; ++i) {
+    double acc = 0;
+    for (int b = 0; b < B; ++b) acc += (double)s_h[(0 * B + b) * S + i];
+    parent[i] = acc;
+  }
+  const double n_parent_raw = flowml_count_of(parent, S, imp_code);
+  const double n_parent = n_parent_raw < 1e-12 ? 1e-12 : n_parent_raw;
+  const double imp_parent = flowml_impurity(parent, S, imp_code);
+
+  // each thread scans a subset of features; prefix over bins is sequential
+  // per feature (B=32: trivially cheap from LDS)
+  double best_gain = -1.0 / 0.0;
+  int best_idx = 0x7FFFFFFF;  // (f*(B-1)+b); smaller wins ties (torch argmax)
+  for (int f = threadIdx.x; f < Fp; f += blockDim.x) {
+    double L[K3_MAXS];
+    for (int i = 0; i < S; ++i) L[i] = 0;
+    for (int b = 0; b < B - 1; ++b) {
+      for (int i = 0; i < S; ++i) L[i] += (double)s_h[(f * B + b) * S + i];
+      double R[K3_MAXS];
+      for (int i = 0; i < S; ++i) R[i] = parent[i] - L[i];
+      const double nL = flowml_count_of(L, S, imp_code);
+      const double nR = flowml_count_of(R, S, imp_code);
+      if (nL < min_instances || nR < min_instances) continue;
+      const double child =
+          (nL * flowml_impurity(L, S, imp_code) + nR * flowml_impurity(R, S, imp_code)) /
+          n_parent;
+      const double gain = imp_parent - child;
+      const int idx = f * (B - 1) + b;
+      if (gain > best_gain || (gain == best_gain && idx < best_idx)) {
+        best_gain = gain;
+        best_idx = idx;
+      }
+    }
+  }
+
+  // block argmax (gain desc, idx asc)
+  __shared__ double r_gain[256];
+  __shared__ int r_idx[256];
+  r_gain[threadIdx.x] = best_gain;
+  r_idx[threadIdx.x] = best_idx;
+  __syncthreads();
+  for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) {
+      const double og = r_gain[threadIdx.x + s];
+      const int oi = r_idx[threadIdx.x + s];
+      if (og > r_gain[threadIdx.x] ||
+          (og == r_gain[threadIdx.x] && oi < r_idx[threadIdx.x])) {
+        r_gain[threadIdx.x] = og;
+        r_idx[threadIdx.x] = oi;
+      }
+    }
+    __syncthreads();
+  }
+
+  if (threadIdx.x == 0) {
+    double gain = r_gain[0];
+    int idx = r_idx[0];
+    float* out = packed + (long long)g * (7 + 2 * S);
+    const bool ok = (idx != 0x7FFFFFFF) && (gain > 0) && (gain >= min_info_gain) &&
+                    !isinf(gain) && !isnan(gain);
+    if (!ok) {
+      out[0] = -1.f;
+      for (int i = 1; i < 7 + 2 * S; ++i) out[i] = 0.f;
+      return;
+    }
+    const int f = idx / (B - 1);
+    const int b = idx % (B - 1);
+    double L[K3_MAXS], R[K3_MAXS];
+    for (int i = 0; i < S; ++i) L[i] = 0;
+    for (int bb = 0; bb <= b; ++bb)
+      for (int i = 0; i < S; ++i) L[i] += (double)s_h[(f * B + bb) * S + i];
+    for (int i = 0; i < S; ++i) R[i] = parent[i] - L[i];
+    out[0] = (float)f;
+    out[1] = (float)b;
+    out[2] = (float)gain;
+    out[3] = (float)flowml_impurity(L, S, imp_code);
+    out[4] = (float)flowml_impurity(R, S, imp_code);
+    out[5] = (float)flowml_count_of(L, S, imp_code);
+    out[6] = (float)flowml_count_of(R, S, imp_code);
+    for (int i = 0; i < S; ++i) out[7 + i] = (float)L[i];
+    for (int i = 0; i < S; ++i) out[7 + S + i] = (float)R[i];
+  }
+}
+
+extern "C" void launch_split_scan(const float* hist, float* packed, int G,
+                                  int Fp, int B, int S, int imp_code,
+                                  int min_instances, double min_info_gain,
+                                  void* stream) {
+  const int block = 256;
+  const size_t lds = (size_t)Fp * B * S * sizeof(float);
+  hipLaunchKernelGGL(flowml_split_scan, dim3(G), dim3(block), lds,
+                     (hipStream_t)stream, hist, packed, G, Fp, B, S, imp_code,
+                     min_instances, min_info_gain);
+}

@@ -369,3 +369,27 @@ class TestFusedHistFeatures:
                                     labels, None, g, 2, B, C, None)
         assert torch.equal(nid_fused.cpu(), nid_sep.cpu())
         assert torch.equal(h_fused.cpu(), h_sep.cpu())
+
+
+class TestSplitScanGPU:
+    @pytest.mark.parametrize("imp,S", [("gini", 5), ("entropy", 3), ("variance", 3)])
+    def test_packed_parity_bitwise(self, imp, S):
+        """K3 kernel (double gain math) == torch fp64 reference, bitwise."""
+        r = np.random.default_rng(41)
+        G, Fp, B = 64, 6, 32
+        # integer-valued counts like real histograms
+        hist = torch.from_numpy(
+            r.integers(0, 50, (G, Fp, B, S)).astype(np.float32)
+        )
+        if imp == "variance":
+            # (s, s2, n) stats: make s2 >= s^2/n plausible
+            hist[..., 1] = hist[..., 1] + hist[..., 0] ** 2
+        gpu = ops.split_scan_packed(hist.to(DEV), imp, 2, 0.0).cpu()
+        cpu = ops.split_scan_packed(hist, imp, 2, 0.0)
+        assert torch.equal(gpu[:, :2], cpu[:, :2]), "split choice differs"
+        assert torch.allclose(gpu, cpu, rtol=1e-6, atol=1e-6)
+
+    def test_no_valid_split(self):
+        hist = torch.zeros(4, 2, 8, 3, device=DEV)
+        out = ops.split_scan_packed(hist, "gini", 1, 0.0).cpu()
+        assert (out[:, 0] == -1).all()
