@@ -544,27 +544,30 @@ extern "C" void launch_softmax_xent(const float* logits, const int* y,
 // ---------------------------------------------------------------------------
 
 __device__ __forceinline__ double flowml_impurity(const double* s, int S, int code) {
+  // NOTE: operation ORDER mirrors the torch reference exactly (sum the
+  // series first, then the final subtract/negate) so gains match bitwise
+  // and argmax tie-breaks agree with the CPU path.
   if (code <= 1) {  // 0 = gini, 1 = entropy (class counts)
     double n = 0;
     for (int i = 0; i < S; ++i) n += s[i];
     if (n < 1e-12) n = 1e-12;
     if (code == 0) {
-      double acc = 1.0;
+      double acc = 0.0;
       for (int i = 0; i < S; ++i) {
         const double p = s[i] / n;
-        acc -= p * p;
+        acc += p * p;
       }
-      return acc;
+      return 1.0 - acc;
     }
     double acc = 0.0;
     for (int i = 0; i < S; ++i) {
       const double p = s[i] / n;
       if (p > 0) {
         const double pc = p < 1e-30 ? 1e-30 : p;
-        acc -= p * log2(pc);
+        acc += p * log2(pc);
       }
     }
-    return acc;
+    return -acc;
   }
   // 2 = variance on (sum, sum2, n)
   double nn = s[2] < 1e-12 ? 1e-12 : s[2];
@@ -610,12 +613,19 @@ extern "C" __global__ void flowml_split_scan(
   double best_gain = -1.0 / 0.0;
   int best_idx = 0x7FFFFFFF;  // (f*(B-1)+b); smaller wins ties (torch argmax)
   for (int f = threadIdx.x; f < Fp; f += blockDim.x) {
+    // per-feature totals (torch subtracts the FEATURE's own cumsum total)
+    double total_f[K3_MAXS];
+    for (int i = 0; i < S; ++i) {
+      double acc = 0;
+      for (int b = 0; b < B; ++b) acc += (double)s_h[(f * B + b) * S + i];
+      total_f[i] = acc;
+    }
     double L[K3_MAXS];
     for (int i = 0; i < S; ++i) L[i] = 0;
     for (int b = 0; b < B - 1; ++b) {
       for (int i = 0; i < S; ++i) L[i] += (double)s_h[(f * B + b) * S + i];
       double R[K3_MAXS];
-      for (int i = 0; i < S; ++i) R[i] = parent[i] - L[i];
+      for (int i = 0; i < S; ++i) R[i] = total_f[i] - L[i];
       const double nL = flowml_count_of(L, S, imp_code);
       const double nR = flowml_count_of(R, S, imp_code);
       if (nL < min_instances || nR < min_instances) continue;
@@ -663,11 +673,16 @@ extern "C" __global__ void flowml_split_scan(
     }
     const int f = idx / (B - 1);
     const int b = idx % (B - 1);
-    double L[K3_MAXS], R[K3_MAXS];
+    double L[K3_MAXS], R[K3_MAXS], total_f[K3_MAXS];
+    for (int i = 0; i < S; ++i) {
+      double acc = 0;
+      for (int bb = 0; bb < B; ++bb) acc += (double)s_h[(f * B + bb) * S + i];
+      total_f[i] = acc;
+    }
     for (int i = 0; i < S; ++i) L[i] = 0;
     for (int bb = 0; bb <= b; ++bb)
       for (int i = 0; i < S; ++i) L[i] += (double)s_h[(f * B + bb) * S + i];
-    for (int i = 0; i < S; ++i) R[i] = parent[i] - L[i];
+    for (int i = 0; i < S; ++i) R[i] = total_f[i] - L[i];
     out[0] = (float)f;
     out[1] = (float)b;
     out[2] = (float)gain;
