@@ -544,6 +544,7 @@ extern "C" void launch_softmax_xent(const float* logits, const int* y,
 // ---------------------------------------------------------------------------
 
 __device__ __forceinline__ double flowml_impurity(const double* s, int S, int code) {
+#pragma clang fp contract(off)
   // NOTE: operation ORDER mirrors the torch reference exactly (sum the
   // series first, then the final subtract/negate) so gains match bitwise
   // and argmax tie-breaks agree with the CPU path.
@@ -577,6 +578,7 @@ __device__ __forceinline__ double flowml_impurity(const double* s, int S, int co
 }
 
 __device__ __forceinline__ double flowml_count_of(const double* s, int S, int code) {
+#pragma clang fp contract(off)
   if (code == 2) return s[2];
   double n = 0;
   for (int i = 0; i < S; ++i) n += s[i];
@@ -590,6 +592,7 @@ extern "C" __global__ void flowml_split_scan(
     float* __restrict__ packed,      // G×(7+2S)
     int G, int Fp, int B, int S, int imp_code, int min_instances,
     double min_info_gain) {
+#pragma clang fp contract(off)
   extern __shared__ float s_h[];  // Fp×B×S staged slab
   const int g = blockIdx.x;
   const int slab = Fp * B * S;

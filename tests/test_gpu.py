@@ -377,10 +377,15 @@ class TestSplitScanGPU:
         """K3 kernel (double gain math) == torch fp64 reference, bitwise."""
         r = np.random.default_rng(41)
         G, Fp, B = 64, 6, 32
-        # integer-valued counts like real histograms
-        hist = torch.from_numpy(
-            r.integers(0, 50, (G, Fp, B, S)).astype(np.float32)
-        )
+        # integer-valued counts with CONSISTENT per-feature totals (each row
+        # contributes once to every feature, like a real histogram): one base
+        # bin×stat table per node, bins permuted per feature
+        base = r.integers(0, 50, (G, 1, B, S))
+        hist = np.empty((G, Fp, B, S), dtype=np.float32)
+        for g in range(G):
+            for f in range(Fp):
+                hist[g, f] = base[g, 0, r.permutation(B)]
+        hist = torch.from_numpy(hist)
         if imp == "variance":
             # (s, s2, n) stats: make s2 >= s^2/n plausible
             hist[..., 1] = hist[..., 1] + hist[..., 0] ** 2
