@@ -301,8 +301,12 @@ def _impurity_from_stats(stats: torch.Tensor, impurity: str) -> torch.Tensor:
     return (s2 / nn - (s / nn) ** 2).clamp_min(0)
 
 
-def _counts_of(stats: torch.Tensor) -> torch.Tensor:
-    if stats.shape[-1] == 3:
+def _counts_of(stats: torch.Tensor, impurity: str) -> torch.Tensor:
+    """Row count from a stats vector. MUST be keyed by the impurity kind:
+    3-class classification stats also have width 3 — shape sniffing here
+    silently read the class-2 count as n (bug found by the K3 kernel's
+    explicit imp_code parity)."""
+    if impurity == "variance":
         return stats[..., 2]
     return stats.sum(dim=-1)
 
@@ -323,9 +327,9 @@ def split_scan(
     parent = total[:, 0, :]  # G×S  (same for all features)
     L = leftcum[:, :, : B - 1, :]  # candidate splits at bins 0..B-2
     R = total.unsqueeze(2) - L
-    nL = _counts_of(L)
-    nR = _counts_of(R)
-    n = _counts_of(parent).clamp_min(1e-12)  # G
+    nL = _counts_of(L, impurity)
+    nR = _counts_of(R, impurity)
+    n = _counts_of(parent, impurity).clamp_min(1e-12)  # G
     impL = _impurity_from_stats(L, impurity)
     impR = _impurity_from_stats(R, impurity)
     imp_parent = _impurity_from_stats(parent, impurity)  # G
@@ -455,7 +459,7 @@ def grow_trees(
     dist.all_reduce_(h0.reshape(-1))
     root_stats = h0.sum(dim=2)[:, 0, :].to(torch.float64)  # T×S
     root_imp = _impurity_from_stats(root_stats, cfg.impurity)
-    root_cnt = _counts_of(root_stats)
+    root_cnt = _counts_of(root_stats, cfg.impurity)
     rs_np = root_stats.cpu().numpy()
     ri_np = root_imp.cpu().numpy()
     rc_np = root_cnt.cpu().numpy()

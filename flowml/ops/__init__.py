@@ -561,8 +561,8 @@ def split_scan_packed(
     right_stats = torch.where(ok.unsqueeze(1), right_stats, z(right_stats))
     imp_l = torch.where(ok, _impurity_from_stats(left_stats, impurity), z(best_gain))
     imp_r = torch.where(ok, _impurity_from_stats(right_stats, impurity), z(best_gain))
-    cnt_l = torch.where(ok, _counts_of(left_stats), z(best_gain))
-    cnt_r = torch.where(ok, _counts_of(right_stats), z(best_gain))
+    cnt_l = torch.where(ok, _counts_of(left_stats, impurity), z(best_gain))
+    cnt_r = torch.where(ok, _counts_of(right_stats, impurity), z(best_gain))
     return torch.cat(
         [
             best_f.to(torch.float64).reshape(G, 1),
