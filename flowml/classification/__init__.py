@@ -19,6 +19,7 @@ from flowml.classification.mlp import (
 from flowml.classification.forest import (
     RandomForestClassifier,
     RandomForestClassificationModel,
+    DecisionTreeClassifier,
     DecisionTreeClassificationModel,
 )
 from flowml.classification.gbt import GBTClassifier, GBTClassificationModel
@@ -34,6 +35,7 @@ __all__ = [
     "MultilayerPerceptronClassificationModel",
     "RandomForestClassifier",
     "RandomForestClassificationModel",
+    "DecisionTreeClassifier",
     "DecisionTreeClassificationModel",
     "GBTClassifier",
     "GBTClassificationModel",

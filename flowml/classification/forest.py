@@ -201,6 +201,13 @@ class RandomForestClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
             trees=trees, numClasses=n_classes, numFeatures=F
         )
         t5 = _tick()
+        from flowml.utils.logging import get_logger
+
+        get_logger().log(
+            "rf_fit", trees=T, total_nodes=int(sum(tr.n_nodes for tr in trees)),
+            rows=int(N), features=int(F), classes=int(n_classes),
+            fit_ms=round((t5 - t0) * 1e3, 2),
+        )
         if _timing:
             print("rf_fit timing(ms):", {
                 "thresholds": round((t1 - t0) * 1e3, 1),
@@ -298,3 +305,56 @@ class DecisionTreeClassificationModel(_TreeEnsembleParams, ProbabilisticClassifi
     def raw2probability(self, raw: torch.Tensor) -> torch.Tensor:
         s = raw.sum(dim=1, keepdim=True).clamp_min(1e-30)
         return raw / s
+
+
+class DecisionTreeClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
+    """Single decision tree (spark:.../classification/DecisionTreeClassifier.scala):
+    a 1-tree forest with all features considered at every node and no
+    bootstrap — defaults maxDepth=5, maxBins=32, impurity=gini."""
+
+    def __init__(
+        self,
+        *,
+        featuresCol: Optional[str] = None,
+        labelCol: Optional[str] = None,
+        predictionCol: Optional[str] = None,
+        rawPredictionCol: Optional[str] = None,
+        probabilityCol: Optional[str] = None,
+        maxDepth: Optional[int] = None,
+        maxBins: Optional[int] = None,
+        minInstancesPerNode: Optional[int] = None,
+        minInfoGain: Optional[float] = None,
+        impurity: Optional[str] = None,
+        seed: Optional[int] = None,
+        weightCol: Optional[str] = None,
+    ):
+        super().__init__()
+        self._set(
+            featuresCol=featuresCol,
+            labelCol=labelCol,
+            predictionCol=predictionCol,
+            rawPredictionCol=rawPredictionCol,
+            probabilityCol=probabilityCol,
+            maxDepth=maxDepth,
+            maxBins=maxBins,
+            minInstancesPerNode=minInstancesPerNode,
+            minInfoGain=minInfoGain,
+            impurity=impurity,
+            seed=seed,
+            weightCol=weightCol,
+        )
+        self._set(numTrees=1, bootstrap=False, subsamplingRate=1.0,
+                  featureSubsetStrategy="all")
+
+    def _fit(self, dataset: FlowFrame) -> "DecisionTreeClassificationModel":
+        rf = RandomForestClassifier()
+        self._copyValues(rf)
+        forest = rf._fit(dataset)
+        model = DecisionTreeClassificationModel(
+            tree=forest._trees[0],
+            numClasses=forest.numClasses,
+            numFeatures=forest.numFeatures,
+        )
+        self._copyValues(model)
+        model._set_parent(self)
+        return model

@@ -215,6 +215,9 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
             tree_list = grow_trees(codes, resid, bag, thresholds, cfg)
             tree = tree_list[0]
             trees.append(tree)
+            from flowml.utils.logging import get_logger
+
+            get_logger().log("gbt_round", round=rnd, nodes=tree.n_nodes)
             weights.append(1.0 if rnd == 0 else step)  # Spark: first tree unshrunk
             te = TreeEnsemble([tree], 2, mode="scalar")
             pred = te.predict_raw(X)[:, 0]

@@ -306,6 +306,13 @@ class LogisticRegression(_LogisticRegressionParams, ProbabilisticClassifier):
             objectiveHistory=[float(h) for h in history],
             totalIterations=max(len(history) - 1, 0),
         )
+        from flowml.utils.logging import get_logger
+
+        get_logger().log(
+            "lr_fit", iterations=model._summary.totalIterations,
+            final_objective=float(history[-1]) if history else None,
+            num_classes=num_classes, num_features=F,
+        )
         return model
 
 

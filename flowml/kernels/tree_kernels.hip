@@ -32,6 +32,13 @@
 
 #include <hip/hip_runtime.h>
 #include <stdint.h>
+#include <stdlib.h>
+
+// launch-shape knobs (read once; for on-box tuning sweeps without rebuilds)
+static int flowml_env_int(const char* name, int dflt) {
+  const char* v = getenv(name);
+  return v ? atoi(v) : dflt;
+}
 
 #define FLOWML_MAXK 16  // max classes handled in-register by rowwise kernels
 
@@ -460,8 +467,10 @@ extern "C" void launch_tree_hist_cls(
     long long N, int F, int T, int Fp, int B, int C,
     int l_max, void* stream) {
   hist_opt_in_big_lds();
-  const int block = 512;
-  const long long rows_per_block = 16384;
+  static const int cfg_block = flowml_env_int("FLOWML_HIST_BLOCK", 512);
+  static const int cfg_rpb = flowml_env_int("FLOWML_HIST_RPB", 16384);
+  const int block = cfg_block;
+  const long long rows_per_block = cfg_rpb;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t lds = (size_t)l_max * Fp * B * C * sizeof(float);
   hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(T, tiles), dim3(block), lds,
@@ -480,8 +489,10 @@ extern "C" void launch_tree_hist_reg(
     long long N, int F, int T, int Fp, int B, int l_max,
     void* stream) {
   hist_opt_in_big_lds();
-  const int block = 512;
-  const long long rows_per_block = 16384;
+  static const int cfg_block = flowml_env_int("FLOWML_HIST_BLOCK", 512);
+  static const int cfg_rpb = flowml_env_int("FLOWML_HIST_RPB", 16384);
+  const int block = cfg_block;
+  const long long rows_per_block = cfg_rpb;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t lds = (size_t)l_max * Fp * B * 3 * sizeof(float);
   hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(T, tiles), dim3(block), lds,

@@ -259,3 +259,25 @@ class TestCheckpointResume:
         p_full = m_full.transform(df)["prediction"]
         p_res = m_res.transform(df)["prediction"]
         assert torch.equal(p_full, p_res)
+
+
+class TestDecisionTree:
+    def test_single_tree_fit(self):
+        from flowml.classification import DecisionTreeClassifier
+
+        df, X, y = _multi_df(n=1500)
+        m = DecisionTreeClassifier(maxDepth=5, seed=0).fit(df)
+        acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(m.transform(df))
+        assert acc > 0.5
+        assert m.depth <= 5
+        assert m.numNodes >= 3
+
+    def test_matches_one_tree_forest(self):
+        from flowml.classification import DecisionTreeClassifier, RandomForestClassifier
+
+        df, _, _ = _multi_df(n=800)
+        dt = DecisionTreeClassifier(maxDepth=4, seed=5).fit(df)
+        rf = RandomForestClassifier(
+            numTrees=1, maxDepth=4, seed=5, bootstrap=False, featureSubsetStrategy="all"
+        ).fit(df)
+        assert torch.equal(dt.transform(df)["prediction"], rf.transform(df)["prediction"])
