@@ -204,7 +204,23 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
 
         train_sel = (~val_mask) if val_mask is not None else None
 
+        import os as _os
+        import time as _time
+
+        _timing = _os.environ.get("FLOWML_TIMING", "0") == "1"
+        _tmr = {}
+
+        def _tick():
+            if _timing and dev.type == "cuda":
+                torch.cuda.synchronize()
+            return _time.perf_counter()
+
+        def _acc(k, t0):
+            if _timing:
+                _tmr[k] = _tmr.get(k, 0.0) + (_tick() - t0)
+
         for rnd in range(start_round, max_iter):
+            _t0 = _tick()
             # pseudo-residual: -dL/dF for L = log(1+exp(-2 y F)) → 2y/(1+exp(2yF))
             resid = 2.0 * ysign / (1.0 + torch.exp(2.0 * ysign * margin))
             cfg.seed = (seed + 1_000_003 * rnd) & 0x7FFFFFFF
@@ -212,16 +228,23 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
             if train_sel is not None:
                 sel = train_sel.to(torch.float32).unsqueeze(0)  # (1, N)
                 bag = sel if bag is None else bag * sel
+            _acc("resid", _t0)
+            _t1 = _tick()
             tree_list = grow_trees(codes, resid, bag, thresholds, cfg)
             tree = tree_list[0]
             trees.append(tree)
+            _acc("grow", _t1)
             from flowml.utils.logging import get_logger
 
             get_logger().log("gbt_round", round=rnd, nodes=tree.n_nodes)
             weights.append(1.0 if rnd == 0 else step)  # Spark: first tree unshrunk
+            _t2 = _tick()
             te = TreeEnsemble([tree], 2, mode="scalar")
+            _acc("ensemble_build", _t2)
+            _t3 = _tick()
             pred = te.predict_raw(X)[:, 0]
             margin = margin + weights[-1] * pred
+            _acc("margin", _t3)
             if val_mask is not None:
                 vl = torch.log1p(torch.exp(-2.0 * ysign[val_mask] * margin[val_mask])).mean()
                 buf = torch.tensor([float(vl), 1.0], device=dev)
@@ -239,6 +262,8 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
             ):
                 self._write_checkpoint(trees, weights, rnd + 1, F)
 
+        if _timing and _tmr:
+            print("gbt rounds timing:", {k: round(v * 1000, 1) for k, v in _tmr.items()})
         model = GBTClassificationModel(
             trees=trees, treeWeights=weights, numFeatures=F
         )
