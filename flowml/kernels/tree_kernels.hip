@@ -470,7 +470,12 @@ extern "C" void launch_tree_hist_cls(
   static const int cfg_block = flowml_env_int("FLOWML_HIST_BLOCK", 512);
   static const int cfg_rpb = flowml_env_int("FLOWML_HIST_RPB", 16384);
   const int block = cfg_block;
-  const long long rows_per_block = cfg_rpb;
+  // target >= 1024 blocks so a single-tree launch (GBT) still fills the
+  // 256-CU chip; large T (RF) keeps the big tile for flush amortization
+  long long rows_per_block = cfg_rpb;
+  long long ideal = (N * (long long)T) / 1024;  // tiles*T ~ 1024 blocks
+  if (ideal < 2048) ideal = 2048;
+  if (ideal < rows_per_block) rows_per_block = ideal;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t lds = (size_t)l_max * Fp * B * C * sizeof(float);
   hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(T, tiles), dim3(block), lds,
@@ -492,7 +497,12 @@ extern "C" void launch_tree_hist_reg(
   static const int cfg_block = flowml_env_int("FLOWML_HIST_BLOCK", 512);
   static const int cfg_rpb = flowml_env_int("FLOWML_HIST_RPB", 16384);
   const int block = cfg_block;
-  const long long rows_per_block = cfg_rpb;
+  // target >= 1024 blocks so a single-tree launch (GBT) still fills the
+  // 256-CU chip; large T (RF) keeps the big tile for flush amortization
+  long long rows_per_block = cfg_rpb;
+  long long ideal = (N * (long long)T) / 1024;  // tiles*T ~ 1024 blocks
+  if (ideal < 2048) ideal = 2048;
+  if (ideal < rows_per_block) rows_per_block = ideal;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t lds = (size_t)l_max * Fp * B * 3 * sizeof(float);
   hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(T, tiles), dim3(block), lds,
