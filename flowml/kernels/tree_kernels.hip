@@ -127,6 +127,10 @@ extern "C" __global__ void flowml_tree_hist_cls(
     for (int i = 0; i < 8; ++i) cdf[i] = bag_cdf[i];
   }
 
+#ifndef FLOWML_HIST_UNROLL
+#define FLOWML_HIST_UNROLL 1
+#endif
+#pragma unroll FLOWML_HIST_UNROLL
   for (long long r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
     int nid = nid_t[r];
     if (advancing) {

@@ -144,3 +144,31 @@ def test_fused_allreduce():
     a, b = _run_dist("_fused_allreduce_body")
     assert np.allclose(a, 3.0)  # 1 + 2
     assert np.allclose(b, 30.0)  # 10 + 20
+
+
+def _gbt_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.classification import GBTClassifier
+
+    X, y = _make_data()
+    yb = (y >= 1.5).astype(np.float64)
+    lo, hi = dist.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi]), "label": yb[lo:hi]})
+    m = GBTClassifier(maxIter=8, maxDepth=3, seed=11).fit(df)
+    full = FlowFrame({"features": torch.from_numpy(X), "label": yb})
+    return m.transform(full)["rawPrediction"][:, 1].numpy()
+
+
+def test_gbt_dist_close_to_single():
+    raw_dist = _run_dist("_gbt_dist_body")
+    from flowml import FlowFrame
+    from flowml.classification import GBTClassifier
+
+    X, y = _make_data()
+    yb = (y >= 1.5).astype(np.float64)
+    full = FlowFrame({"features": torch.from_numpy(X), "label": yb})
+    m = GBTClassifier(maxIter=8, maxDepth=3, seed=11).fit(full)
+    raw_single = m.transform(full)["rawPrediction"][:, 1].numpy()
+    # identical global histograms (hash bagging off for GBT; thresholds from
+    # the gathered sample) -> identical trees up to fp reduction order
+    assert np.corrcoef(raw_dist, raw_single)[0, 1] > 0.99

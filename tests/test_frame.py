@@ -92,3 +92,21 @@ def test_read_csv_no_header_schema(tmp_path):
     assert df.columns == ["_c0", "_c1", "_c2"]
     df2 = read_csv(p, header=False, columns=["proto", "dur", "label"])
     assert df2.columns == ["proto", "dur", "label"]
+
+
+def test_streaming_classifier_cpu_fallback():
+    """StreamingClassifier without a GPU: eager path, fixed-shape batches."""
+    from flowml.classification import RandomForestClassifier
+    from flowml.streaming import StreamingClassifier
+
+    r = np.random.default_rng(3)
+    X = torch.from_numpy(r.normal(size=(2000, 6)).astype(np.float32))
+    y = (X[:, 0] > 0).to(torch.float64)
+    m = RandomForestClassifier(numTrees=4, maxDepth=3, seed=0).fit(
+        FlowFrame({"features": X, "label": y})
+    )
+    sc = StreamingClassifier(m, batch_size=500, num_features=6, device="cpu", use_graph=False)
+    pred, raw = sc.infer(X[:500])
+    eager = m.predictRaw(X[:500])
+    assert torch.allclose(raw, eager)
+    assert torch.equal(pred, eager.argmax(dim=1))
