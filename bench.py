@@ -383,7 +383,9 @@ def main():
             "data": "synthetic",
             "config": cfg,
         }
-        print(json.dumps(out))
+        print(json.dumps(out), flush=True)
+    if dist.is_initialized():
+        torch.distributed.destroy_process_group()
 
 
 if __name__ == "__main__":

@@ -176,7 +176,7 @@ class StringIndexerModel(_StringIndexerParams, Model):
             new_cols[o] = idx
         out = dataset
         for o, idx in new_cols.items():
-            out = out.withColumn(o, torch.from_numpy(idx))
+            out = out.withColumn(o, torch.from_numpy(idx).to(dataset.device))
         if keep_mask is not None:
             out = out.filter(keep_mask)
         return out
