@@ -202,8 +202,9 @@ def bench_mlp(args, device):
     total = float(rows * world)
     lr = 0.5
 
-    def step():
-        nonlocal theta
+    def _train_ops():
+        # every op here is capture-safe (no host syncs): the whole step is
+        # hipGraph-captured below and replayed per iteration
         Ws, bs = _unflatten(theta, layers)
         if use_bf16:
             Ws = [w.to(torch.bfloat16) for w in Ws]
@@ -213,7 +214,30 @@ def bench_mlp(args, device):
         flat_g = _flatten([g.to(torch.float32) for g in gWs], gBs)
         buf = torch.cat([loss.reshape(1), flat_g])
         dist.all_reduce_(buf)
-        theta = theta - lr * (buf[1:] / total)
+        theta.sub_(lr * (buf[1:] / total))
+
+    graph = None
+    if device.type == "cuda":
+        try:
+            s = torch.cuda.Stream(device=device)
+            s.wait_stream(torch.cuda.current_stream(device))
+            with torch.cuda.stream(s):
+                for _ in range(3):
+                    _train_ops()
+            torch.cuda.current_stream(device).wait_stream(s)
+            torch.cuda.synchronize(device)
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                _train_ops()
+        except Exception as e:  # noqa: BLE001
+            print(f"# hipGraph capture unavailable ({e}); eager steps", flush=True)
+            graph = None
+
+    def step():
+        if graph is not None:
+            graph.replay()
+        else:
+            _train_ops()
 
     elapsed = timed_loop(step, args.steps, args.warmup, device)
     # final accuracy

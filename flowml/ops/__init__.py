@@ -363,13 +363,15 @@ def logreg_grad(
     inside treeAggregate.
     """
     if hip_available(X):
-        # K1 fused kernel produces (loss, weighted delta); the rank-1-update
-        # accumulation gradW = deltaᵀX is a plain GEMM -> rocBLAS (allowed for
-        # plain library GEMMs per the north star).
-        loss, delta = get_ext().logreg_delta(
-            X.contiguous(), y.to(torch.int32).contiguous(),
+        # Composition of hand kernels: logits via the coalesced f32 MFMA
+        # GEMM (K5) + fused softmax/xent/delta (K6) — ~10x the throughput of
+        # the per-thread-dot variant (X rows are only 164 B, so per-lane
+        # sequential reads split 41 ways per wave instruction). gradW = deltaᵀX
+        # is a plain GEMM -> rocBLAS (sanctioned for plain library GEMMs).
+        logits = gemm_bias_act(X, W, b, "none")
+        loss, delta = get_ext().softmax_xent_grad(
+            logits.contiguous(), y.to(torch.int32).contiguous(),
             w.contiguous() if w is not None else _empty(X.device),
-            W.contiguous(), b.contiguous(),
         )
         gradW = delta.T @ X
         gradB = delta.sum(dim=0)
