@@ -43,6 +43,8 @@ void launch_gemm_bias_act_bf16(const void*, const void*, const float*, void*,
 void launch_gemm_bias_act_f32(const float*, const float*, const float*, float*,
                               long long, int, int, int, void*);
 void launch_act_backward_bf16(const void*, void*, long long, int, void*);
+void launch_wgrad_bias_bf16(const void*, const void*, float*, long long, int,
+                            int, void*);
 }
 
 static void* cur_stream() {
@@ -278,6 +280,24 @@ torch::Tensor gemm_bias_act(torch::Tensor X, torch::Tensor W, torch::Tensor b,
   return out;
 }
 
+std::tuple<torch::Tensor, torch::Tensor> wgrad_bias(torch::Tensor delta,
+                                                    torch::Tensor act) {
+  CHK(delta);
+  CHK(act);
+  TORCH_CHECK(delta.scalar_type() == torch::kBFloat16 &&
+                  act.scalar_type() == torch::kBFloat16,
+              "bf16 only");
+  const long long N = delta.size(0);
+  const int K = delta.size(1);
+  const int F = act.size(1);
+  auto out = torch::zeros({K, F + 1}, delta.options().dtype(torch::kFloat32));
+  launch_wgrad_bias_bf16(delta.data_ptr(), act.data_ptr(),
+                         out.data_ptr<float>(), N, K, F, cur_stream());
+  auto gW = out.narrow(1, 0, F).contiguous();
+  auto gB = out.select(1, F).contiguous();
+  return {gW, gB};
+}
+
 torch::Tensor act_backward_(torch::Tensor h, torch::Tensor delta,
                             int64_t act) {
   CHK(h);
@@ -299,4 +319,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_grad", &softmax_xent_grad, "K6 softmax/xent grad");
   m.def("gemm_bias_act", &gemm_bias_act, "K5 fused MFMA GEMM+bias+act");
   m.def("act_backward_", &act_backward_, "K7 activation backward (in-place)");
+  m.def("wgrad_bias", &wgrad_bias, "K7 split-N wgrad + fused bias grad");
 }

@@ -192,3 +192,142 @@ extern "C" void launch_act_backward_bf16(const void* h, void* delta,
                      (hipStream_t)stream, (const __bf16*)h, (__bf16*)delta,
                      total, act);
 }
+
+// ---------------------------------------------------------------------------
+// K7 — wgrad with fused bias-grad: gW[K,F] = deltaᵀ[K,N] @ act[N,F],
+// gB[K] = Σ_n delta[n,:].  N is huge (the 1M-flow shard) and K,F are tiny
+// (≤128), so the rocBLAS heuristic (no split-K → 2 workgroups!) collapses;
+// this kernel splits N across ~768 blocks, stages TRANSPOSED 32-row tiles of
+// both operands in LDS (bank-padded rows), runs bf16 MFMA over the n-depth,
+// and atomically accumulates fp32 partials. The bias gradient rides along as
+// an extra all-ones act column (out column F), so no separate reduce pass.
+// ---------------------------------------------------------------------------
+
+#define WG_NPAD 40  // LDS row stride in bf16 elems: 80 B, 16B-aligned, bank-clean
+
+extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
+    const __bf16* __restrict__ delta,  // N×K
+    const __bf16* __restrict__ act,    // N×F
+    float* __restrict__ out,           // K×(F+1) fp32 (zeroed; col F = gB)
+    long long N, int K, int F, int Kp, int Fp,  // 16-padded; Fp covers F+1
+    int waves_k, int waves_f,          // wave grid (waves_k*waves_f = 4)
+    long long chunk) {                 // rows per block
+  extern __shared__ __bf16 s_t[];  // dt[Kp][WG_NPAD] then at[Fp][WG_NPAD]
+  __bf16* dt = s_t;
+  __bf16* at = s_t + (long long)Kp * WG_NPAD;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  // zero-fill LDS once: rows >= K (or beyond F) stay zero for the whole loop
+  for (int i = tid; i < (Kp + Fp) * WG_NPAD; i += blockDim.x) s_t[i] = (__bf16)0.f;
+  __syncthreads();
+
+  const int nk16 = Kp / 16;
+  const int nf16 = Fp / 16;
+  const int kfrag = (nk16 + waves_k - 1) / waves_k;  // <= 4 (host guarantees)
+  const int ffrag = (nf16 + waves_f - 1) / waves_f;  // <= 8
+  const int wk = wave / waves_f;
+  const int wf = wave % waves_f;
+  const int kf0 = wk * kfrag;  // first k-fragment of this wave
+  const int ff0 = wf * ffrag;
+
+  f32x4 acc[4][8];
+#pragma unroll
+  for (int a = 0; a < 4; ++a)
+#pragma unroll
+    for (int b = 0; b < 8; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+
+  const long long n_lo = blockIdx.x * chunk;
+  const long long n_hi = min(n_lo + chunk, N);
+  const int lrow = lane & 15;        // fragment row within 16
+  const int lgrp = (lane >> 4) * 8;  // n offset of this lane's 8 elems
+
+  for (long long n0 = n_lo; n0 < n_hi; n0 += 32) {
+    const int nrows = (int)min((long long)32, n_hi - n0);
+    __syncthreads();  // previous fragments consumed
+    // stage transposed tiles (lanes contiguous in k/f -> coalesced reads)
+    for (int i = tid; i < 32 * K; i += blockDim.x) {
+      const int n = i / K;
+      const int k = i % K;
+      dt[k * WG_NPAD + n] =
+          (n < nrows) ? delta[(n0 + n) * (long long)K + k] : (__bf16)0.f;
+    }
+    for (int i = tid; i < 32 * F; i += blockDim.x) {
+      const int n = i / F;
+      const int f = i % F;
+      at[f * WG_NPAD + n] =
+          (n < nrows) ? act[(n0 + n) * (long long)F + f] : (__bf16)0.f;
+    }
+    // bias column: act_t row F = 1 for valid rows (fused gB)
+    if (tid < 32) at[F * WG_NPAD + tid] = (tid < nrows) ? (__bf16)1.f : (__bf16)0.f;
+    __syncthreads();
+    // one MFMA (n-depth 32) per fragment pair; STATIC acc indexing with
+    // runtime guards (runtime-indexed register arrays spill to scratch)
+#pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      if (a >= kfrag || kf0 + a >= nk16) continue;
+      const __bf16* arow = dt + ((kf0 + a) * 16 + lrow) * WG_NPAD + lgrp;
+      bf16x8 afrag = *reinterpret_cast<const bf16x8*>(arow);
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        if (b >= ffrag || ff0 + b >= nf16) continue;
+        const __bf16* brow = at + ((ff0 + b) * 16 + lrow) * WG_NPAD + lgrp;
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(brow);
+        acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[a][b], 0, 0, 0);
+      }
+    }
+  }
+
+  // epilogue: D col = lane&15 (f), row = (lane>>4)*4 + reg (k)
+  const int col = lane & 15;
+  const int row_off = (lane >> 4) * 4;
+#pragma unroll
+  for (int a = 0; a < 4; ++a) {
+    if (a >= kfrag || kf0 + a >= nk16) continue;
+#pragma unroll
+    for (int b = 0; b < 8; ++b) {
+      if (b >= ffrag || ff0 + b >= nf16) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int k = (kf0 + a) * 16 + row_off + reg;
+        const int f = (ff0 + b) * 16 + col;
+        if (k < K && f <= F) {
+          atomicAdd(&out[(long long)k * (F + 1) + f], acc[a][b][reg]);
+        }
+      }
+    }
+  }
+}
+
+extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
+                                       float* out, long long N, int K, int F,
+                                       void* stream) {
+  auto pad16 = [](int v) { return (v + 15) / 16 * 16; };
+  const int Kp = pad16(K);
+  const int Fp = pad16(F + 1);
+  const int nk16 = Kp / 16;
+  const int nf16 = Fp / 16;
+  // wave grid: first feasible of (2,2), (1,4), (4,1) with kfrag<=4, ffrag<=8
+  int waves_k = 2, waves_f = 2;
+  auto ok = [&](int wk2, int wf2) {
+    return (nk16 + wk2 - 1) / wk2 <= 4 && (nf16 + wf2 - 1) / wf2 <= 8;
+  };
+  if (!ok(2, 2)) {
+    if (ok(1, 4)) {
+      waves_k = 1;
+      waves_f = 4;
+    } else {
+      waves_k = 4;
+      waves_f = 1;
+    }
+  }
+  const int blocks = 768;
+  long long chunk = (N + blocks - 1) / blocks;
+  chunk = (chunk + 31) / 32 * 32;
+  const int grid = (int)((N + chunk - 1) / chunk);
+  const size_t lds = (size_t)(Kp + Fp) * WG_NPAD * sizeof(__bf16);
+  hipLaunchKernelGGL(flowml_wgrad_bias_bf16, dim3(grid), dim3(256), lds,
+                     (hipStream_t)stream, (const __bf16*)delta,
+                     (const __bf16*)act, out, N, K, F, Kp, Fp, waves_k,
+                     waves_f, chunk);
+}

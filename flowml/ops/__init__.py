@@ -472,12 +472,20 @@ def mlp_grad(
     delta = delta.to(X.dtype)
     gWs: List[Optional[torch.Tensor]] = [None] * L
     gBs: List[Optional[torch.Tensor]] = [None] * L
+    use_wgrad_kernel = (
+        X.is_cuda and X.dtype == torch.bfloat16 and hip_available(X)
+    )
     for i in range(L - 1, -1, -1):
         a = acts[i]
-        gWs[i] = matmul_tn(delta, a)          # K7 wgrad: (N×K)ᵀ·(N×F)
+        if use_wgrad_kernel:
+            # K7 split-N MFMA wgrad with fused bias grad (rocBLAS picks a
+            # 2-workgroup no-split-K tile for these K=1M outer products)
+            gWs[i], gBs[i] = get_ext().wgrad_bias(delta.contiguous(), a.contiguous())
+        else:
+            gWs[i] = matmul_tn(delta, a)      # K7 wgrad: (N×K)ᵀ·(N×F)
+            gBs[i] = delta.sum(dim=0).to(torch.float32)
         if i == 0 and in_features is not None and gWs[0].shape[1] != in_features:
             gWs[0] = gWs[0][:, :in_features].contiguous()  # drop X pad columns
-        gBs[i] = delta.sum(dim=0).to(torch.float32)
         if i > 0:
             dx = delta @ Ws[i]                # K7 dgrad (plain GEMM: rocBLAS)
             h_i = acts[i]

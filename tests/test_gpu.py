@@ -398,3 +398,18 @@ class TestSplitScanGPU:
         hist = torch.zeros(4, 2, 8, 3, device=DEV)
         out = ops.split_scan_packed(hist, "gini", 1, 0.0).cpu()
         assert (out[:, 0] == -1).all()
+
+
+class TestWgradBias:
+    @pytest.mark.parametrize("K,F", [(128, 64), (64, 128), (5, 64), (128, 128)])
+    def test_parity(self, K, F):
+        r = np.random.default_rng(K * 1000 + F)
+        N = 100_000 + 37
+        delta = torch.from_numpy((0.1 * r.normal(size=(N, K))).astype(np.float32)).to(DEV, torch.bfloat16)
+        act = torch.from_numpy(r.normal(size=(N, F)).astype(np.float32)).to(DEV, torch.bfloat16)
+        gW, gB = ops.get_ext().wgrad_bias(delta.contiguous(), act.contiguous())
+        refW = delta.float().T @ act.float()
+        refB = delta.float().sum(dim=0)
+        assert torch.allclose(gW, refW, rtol=1e-2, atol=refW.abs().max() * 1e-3), \
+            (gW - refW).abs().max()
+        assert torch.allclose(gB, refB, rtol=1e-2, atol=refB.abs().max() * 1e-3)
