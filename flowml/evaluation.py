@@ -19,6 +19,7 @@ from flowml.frame import FlowFrame
 from flowml.param import (
     HasLabelCol,
     HasPredictionCol,
+    HasRawPredictionCol,
     HasWeightCol,
     Param,
     Params,
@@ -160,3 +161,78 @@ class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeight
             neg = total - per_label_actual[lbl]
             return float(fp / neg) if float(neg) > 0 else 0.0
         raise AssertionError(name)
+
+
+class BinaryClassificationEvaluator(HasLabelCol, HasRawPredictionCol, HasWeightCol, Params):
+    """Binary evaluator: areaUnderROC (default) / areaUnderPR.
+
+    Spec: spark:.../ml/evaluation/BinaryClassificationEvaluator.scala — scores
+    come from the rawPrediction column (N×2 margins or a 1-D score).
+    """
+
+    metricName: Param[str] = Param(
+        "undefined",
+        "metricName",
+        "metric name in evaluation (areaUnderROC|areaUnderPR)",
+        TypeConverters.toString,
+    )
+
+    def __init__(
+        self,
+        *,
+        rawPredictionCol: Optional[str] = None,
+        labelCol: Optional[str] = None,
+        metricName: Optional[str] = None,
+        weightCol: Optional[str] = None,
+    ):
+        super().__init__()
+        self._setDefault(
+            rawPredictionCol="rawPrediction", labelCol="label", metricName="areaUnderROC"
+        )
+        self._set(
+            rawPredictionCol=rawPredictionCol,
+            labelCol=labelCol,
+            metricName=metricName,
+            weightCol=weightCol,
+        )
+
+    def getMetricName(self) -> str:
+        return self.getOrDefault(self.metricName)
+
+    def isLargerBetter(self) -> bool:
+        return True
+
+    def evaluate(self, dataset: FlowFrame) -> float:
+        raw = dataset.col(self.getOrDefault(self.rawPredictionCol))
+        y = dataset.col(self.getOrDefault(self.labelCol)).reshape(-1).to(torch.float64)
+        score = raw[:, 1] if raw.dim() == 2 else raw.reshape(-1)
+        score = score.to(torch.float64)
+        if self.isSet(self.weightCol):
+            w = dataset.col(self.getWeightCol()).reshape(-1).to(torch.float64)
+        else:
+            w = torch.ones_like(y)
+        order = torch.argsort(score, descending=True)
+        y = y[order]
+        w = w[order]
+        tp = torch.cumsum(w * y, dim=0)
+        fp = torch.cumsum(w * (1 - y), dim=0)
+        # collapse ties: keep the last point of each distinct score
+        s_sorted = score[order]
+        last = torch.ones_like(y, dtype=torch.bool)
+        last[:-1] = s_sorted[:-1] != s_sorted[1:]
+        tp = tp[last]
+        fp = fp[last]
+        P = float((w * y).sum())
+        Nn = float((w * (1 - y)).sum())
+        if P == 0 or Nn == 0:
+            return 0.0
+        tpr = torch.cat([torch.zeros(1, dtype=torch.float64), tp / P])
+        fpr = torch.cat([torch.zeros(1, dtype=torch.float64), fp / Nn])
+        if self.getMetricName() == "areaUnderROC":
+            return float(torch.trapz(tpr, fpr))
+        # areaUnderPR: precision-recall curve, first point (r0, p=1)
+        prec = tp / (tp + fp)
+        rec = tp / P
+        prec = torch.cat([torch.ones(1, dtype=torch.float64), prec])
+        rec = torch.cat([torch.zeros(1, dtype=torch.float64), rec])
+        return float(torch.trapz(prec, rec))
