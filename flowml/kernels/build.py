@@ -29,12 +29,19 @@ def build(verbose: bool = True) -> str:
         os.path.join(_HERE, "tree_kernels.hip"),
         os.path.join(_HERE, "mlp_kernels.hip"),
     ]
+    cflags = ["-O3"]
+    hipflags = ["-O3", "--offload-arch=gfx950"]
+    if os.environ.get("FLOWML_ASAN", "0") == "1":
+        # device+host address sanitizer build: run the GPU
+        # test suite against this to catch OOB in the atomic kernels
+        cflags += ["-fsanitize=address", "-shared-libasan"]
+        hipflags += ["-fsanitize=address", "-shared-libasan"]
     mod_path = cpp_extension.load(
         name="_flowml_hip",
         sources=sources,
         build_directory=build_dir,
-        extra_cflags=["-O3"],
-        extra_cuda_cflags=["-O3", "--offload-arch=gfx950"],
+        extra_cflags=cflags,
+        extra_cuda_cflags=hipflags,
         verbose=verbose,
         is_python_module=False,
         is_standalone=False,
