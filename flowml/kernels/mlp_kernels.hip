@@ -203,7 +203,8 @@ extern "C" void launch_act_backward_bf16(const void* h, void* delta,
 // an extra all-ones act column (out column F), so no separate reduce pass.
 // ---------------------------------------------------------------------------
 
-#define WG_NPAD 40  // LDS row stride in bf16 elems: 80 B, 16B-aligned, bank-clean
+// LDS row strides: +8 bf16 keeps rows 16B-aligned and de-patterns banks
+#define WG_PAD 8
 
 extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
     const __bf16* __restrict__ delta,  // N×K
@@ -212,14 +213,18 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
     long long N, int K, int F, int Kp, int Fp,  // 16-padded; Fp covers F+1
     int waves_k, int waves_f,          // wave grid (waves_k*waves_f = 4)
     long long chunk) {                 // rows per block
-  extern __shared__ __bf16 s_t[];  // dt[Kp][WG_NPAD] then at[Fp][WG_NPAD]
+  // row-major 32-row tiles (vectorized coalesced staging); fragments are
+  // gathered TRANSPOSED by scalar LDS reads (no scatter on the write side)
+  extern __shared__ __bf16 s_t[];  // dt[32][Kp+PAD] then at[32][Fp+PAD]
+  const int dstride = Kp + WG_PAD;
+  const int astride = Fp + WG_PAD;
   __bf16* dt = s_t;
-  __bf16* at = s_t + (long long)Kp * WG_NPAD;
+  __bf16* at = s_t + 32 * dstride;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  // zero-fill LDS once: rows >= K (or beyond F) stay zero for the whole loop
-  for (int i = tid; i < (Kp + Fp) * WG_NPAD; i += blockDim.x) s_t[i] = (__bf16)0.f;
+  for (int i = tid; i < 32 * (dstride + astride); i += blockDim.x)
+    s_t[i] = (__bf16)0.f;
   __syncthreads();
 
   const int nk16 = Kp / 16;
@@ -228,7 +233,7 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
   const int ffrag = (nf16 + waves_f - 1) / waves_f;  // <= 8
   const int wk = wave / waves_f;
   const int wf = wave % waves_f;
-  const int kf0 = wk * kfrag;  // first k-fragment of this wave
+  const int kf0 = wk * kfrag;
   const int ff0 = wf * ffrag;
 
   f32x4 acc[4][8];
@@ -239,40 +244,67 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
 
   const long long n_lo = blockIdx.x * chunk;
   const long long n_hi = min(n_lo + chunk, N);
-  const int lrow = lane & 15;        // fragment row within 16
-  const int lgrp = (lane >> 4) * 8;  // n offset of this lane's 8 elems
+  const int lrow = lane & 15;        // fragment k/f index within 16
+  const int lgrp = (lane >> 4) * 8;  // fragment n offset of this lane
+
+  const int kvec = K >> 3;           // bf16x8 chunks per delta row (K%8 tail below)
+  const int ktail = K & 7;
+  const int fvec = F >> 3;
+  const int ftail = F & 7;
 
   for (long long n0 = n_lo; n0 < n_hi; n0 += 32) {
     const int nrows = (int)min((long long)32, n_hi - n0);
     __syncthreads();  // previous fragments consumed
-    // stage transposed tiles (lanes contiguous in k/f -> coalesced reads)
-    for (int i = tid; i < 32 * K; i += blockDim.x) {
-      const int n = i / K;
-      const int k = i % K;
-      dt[k * WG_NPAD + n] =
-          (n < nrows) ? delta[(n0 + n) * (long long)K + k] : (__bf16)0.f;
+    // coalesced row-major staging, 16 B per thread-op
+    for (int i = tid; i < 32 * kvec; i += blockDim.x) {
+      const int n = i / kvec;
+      const int c = i % kvec;
+      bf16x8 v = (n < nrows)
+          ? *reinterpret_cast<const bf16x8*>(delta + (n0 + n) * (long long)K + c * 8)
+          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      *reinterpret_cast<bf16x8*>(dt + n * dstride + c * 8) = v;
     }
-    for (int i = tid; i < 32 * F; i += blockDim.x) {
-      const int n = i / F;
-      const int f = i % F;
-      at[f * WG_NPAD + n] =
-          (n < nrows) ? act[(n0 + n) * (long long)F + f] : (__bf16)0.f;
+    if (ktail) {
+      for (int i = tid; i < 32 * ktail; i += blockDim.x) {
+        const int n = i / ktail;
+        const int k = kvec * 8 + i % ktail;
+        dt[n * dstride + k] =
+            (n < nrows) ? delta[(n0 + n) * (long long)K + k] : (__bf16)0.f;
+      }
     }
-    // bias column: act_t row F = 1 for valid rows (fused gB)
-    if (tid < 32) at[F * WG_NPAD + tid] = (tid < nrows) ? (__bf16)1.f : (__bf16)0.f;
+    for (int i = tid; i < 32 * fvec; i += blockDim.x) {
+      const int n = i / fvec;
+      const int c = i % fvec;
+      bf16x8 v = (n < nrows)
+          ? *reinterpret_cast<const bf16x8*>(act + (n0 + n) * (long long)F + c * 8)
+          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      *reinterpret_cast<bf16x8*>(at + n * astride + c * 8) = v;
+    }
+    if (ftail) {
+      for (int i = tid; i < 32 * ftail; i += blockDim.x) {
+        const int n = i / ftail;
+        const int f = fvec * 8 + i % ftail;
+        at[n * astride + f] =
+            (n < nrows) ? act[(n0 + n) * (long long)F + f] : (__bf16)0.f;
+      }
+    }
+    if (tid < 32) at[tid * astride + F] = (tid < nrows) ? (__bf16)1.f : (__bf16)0.f;
     __syncthreads();
-    // one MFMA (n-depth 32) per fragment pair; STATIC acc indexing with
-    // runtime guards (runtime-indexed register arrays spill to scratch)
+    // transposed fragment gathers: lane reads its 8 n-rows at a fixed k/f col
 #pragma unroll
     for (int a = 0; a < 4; ++a) {
       if (a >= kfrag || kf0 + a >= nk16) continue;
-      const __bf16* arow = dt + ((kf0 + a) * 16 + lrow) * WG_NPAD + lgrp;
-      bf16x8 afrag = *reinterpret_cast<const bf16x8*>(arow);
+      const int kcol = (kf0 + a) * 16 + lrow;
+      bf16x8 afrag;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) afrag[j] = dt[(lgrp + j) * dstride + kcol];
 #pragma unroll
       for (int b = 0; b < 8; ++b) {
         if (b >= ffrag || ff0 + b >= nf16) continue;
-        const __bf16* brow = at + ((ff0 + b) * 16 + lrow) * WG_NPAD + lgrp;
-        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(brow);
+        const int fcol = (ff0 + b) * 16 + lrow;
+        bf16x8 bfrag;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) bfrag[j] = at[(lgrp + j) * astride + fcol];
         acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[a][b], 0, 0, 0);
       }
     }
@@ -307,7 +339,6 @@ extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
   const int Fp = pad16(F + 1);
   const int nk16 = Kp / 16;
   const int nf16 = Fp / 16;
-  // wave grid: first feasible of (2,2), (1,4), (4,1) with kfrag<=4, ffrag<=8
   int waves_k = 2, waves_f = 2;
   auto ok = [&](int wk2, int wf2) {
     return (nk16 + wk2 - 1) / wk2 <= 4 && (nf16 + wf2 - 1) / wf2 <= 8;
@@ -325,7 +356,8 @@ extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
   long long chunk = (N + blocks - 1) / blocks;
   chunk = (chunk + 31) / 32 * 32;
   const int grid = (int)((N + chunk - 1) / chunk);
-  const size_t lds = (size_t)(Kp + Fp) * WG_NPAD * sizeof(__bf16);
+  const size_t lds =
+      (size_t)32 * (Kp + Fp + 2 * WG_PAD) * sizeof(__bf16);
   hipLaunchKernelGGL(flowml_wgrad_bias_bf16, dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (const __bf16*)delta,
                      (const __bf16*)act, out, N, K, F, Kp, Fp, waves_k,
