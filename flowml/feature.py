@@ -548,3 +548,96 @@ class MinMaxScalerModel(_MinMaxScalerParams, Model):
         from flowml.io import load as _load
 
         return _load(path)
+
+
+# ---------------------------------------------------------------------------
+# IndexToString (maps prediction indices back to label strings)
+# ---------------------------------------------------------------------------
+
+
+class IndexToString(HasInputCol, HasOutputCol, Transformer):
+    """Inverse of StringIndexer (spark:.../feature/StringIndexerModel — the
+    companion IndexToString transformer): maps a double index column back to
+    the original label strings via ``labels``."""
+
+    labels: Param[List[str]] = Param(
+        "undefined", "labels", "Optional array of labels specifying index-string mapping.",
+        TypeConverters.toListString,
+    )
+
+    def __init__(self, *, inputCol: Optional[str] = None, outputCol: Optional[str] = None,
+                 labels: Optional[List[str]] = None):
+        super().__init__()
+        self._set(inputCol=inputCol, outputCol=outputCol, labels=labels)
+
+    def getLabels(self) -> List[str]:
+        return self.getOrDefault(self.labels)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        col = dataset.col(self.getInputCol())
+        if isinstance(col, np.ndarray):
+            raise TypeError("IndexToString input must be a numeric index column")
+        labels = self.getLabels()
+        idx = col.detach().cpu().numpy().astype(np.int64)
+        if idx.size and (idx.min() < 0 or idx.max() >= len(labels)):
+            raise ValueError(
+                f"index out of range [0,{len(labels)}) in {self.getInputCol()!r}"
+            )
+        out = np.asarray(labels, dtype=object)[idx]
+        return dataset.withColumn(self.getOutputCol(), out)
+
+
+# ---------------------------------------------------------------------------
+# Normalizer (per-row p-norm scaling)
+# ---------------------------------------------------------------------------
+
+
+class Normalizer(HasInputCol, HasOutputCol, Transformer):
+    """Normalize each feature row to unit p-norm (spark:.../feature/Normalizer.scala,
+    default p=2)."""
+
+    p: Param[float] = Param("undefined", "p", "the p norm value.", TypeConverters.toFloat)
+
+    def __init__(self, *, inputCol: Optional[str] = None, outputCol: Optional[str] = None,
+                 p: Optional[float] = None):
+        super().__init__()
+        self._setDefault(p=2.0)
+        self._set(inputCol=inputCol, outputCol=outputCol, p=p)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getInputCol())
+        if not (isinstance(x, torch.Tensor) and x.dim() == 2):
+            raise TypeError("Normalizer input must be a vector column")
+        pv = self.getOrDefault(self.p)
+        norm = torch.norm(x.to(torch.float32), p=pv, dim=1, keepdim=True)
+        out = torch.where(norm > 0, x / norm, x)
+        return dataset.withColumn(self.getOutputCol(), out)
+
+
+# ---------------------------------------------------------------------------
+# Binarizer (threshold a numeric/vector column to 0/1)
+# ---------------------------------------------------------------------------
+
+
+class Binarizer(HasInputCol, HasOutputCol, Transformer):
+    """Binarize continuous features by threshold (spark:.../feature/Binarizer.scala:
+    strictly greater than threshold -> 1.0)."""
+
+    threshold: Param[float] = Param(
+        "undefined", "threshold", "threshold used to binarize continuous features",
+        TypeConverters.toFloat,
+    )
+
+    def __init__(self, *, inputCol: Optional[str] = None, outputCol: Optional[str] = None,
+                 threshold: Optional[float] = None):
+        super().__init__()
+        self._setDefault(threshold=0.0)
+        self._set(inputCol=inputCol, outputCol=outputCol, threshold=threshold)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getInputCol())
+        if isinstance(x, np.ndarray):
+            raise TypeError("Binarizer input must be numeric")
+        thr = self.getOrDefault(self.threshold)
+        out = (x > thr).to(torch.float64 if x.dim() == 1 else torch.float32)
+        return dataset.withColumn(self.getOutputCol(), out)

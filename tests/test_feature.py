@@ -156,3 +156,50 @@ class TestScalers:
         out = m.transform(FlowFrame({"v": x}))["s"]
         assert out[:, 0].tolist() == [0.0, 1.0]
         assert out[:, 1].tolist() == [0.5, 0.5]  # constant -> midpoint
+
+
+class TestIndexToString:
+    def test_round_trip_with_indexer(self):
+        from flowml.feature import IndexToString
+
+        df = FlowFrame({"s": ["b", "a", "b", "c"]})
+        m = StringIndexer(inputCol="s", outputCol="i").fit(df)
+        out = m.transform(df)
+        back = IndexToString(inputCol="i", outputCol="s2", labels=m.labels).transform(out)
+        assert back["s2"].tolist() == ["b", "a", "b", "c"]
+
+    def test_out_of_range(self):
+        from flowml.feature import IndexToString
+
+        df = FlowFrame({"i": [0.0, 5.0]})
+        with pytest.raises(ValueError):
+            IndexToString(inputCol="i", outputCol="s", labels=["x"]).transform(df)
+
+
+class TestNormalizer:
+    def test_l2_default(self):
+        from flowml.feature import Normalizer
+
+        x = torch.tensor([[3.0, 4.0], [0.0, 0.0]])
+        out = Normalizer(inputCol="v", outputCol="n").transform(FlowFrame({"v": x}))["n"]
+        assert torch.allclose(out[0], torch.tensor([0.6, 0.8]))
+        assert torch.equal(out[1], torch.zeros(2))  # zero row passes through
+
+    def test_l1(self):
+        from flowml.feature import Normalizer
+
+        x = torch.tensor([[1.0, 3.0]])
+        out = Normalizer(inputCol="v", outputCol="n", p=1.0).transform(FlowFrame({"v": x}))["n"]
+        assert torch.allclose(out[0], torch.tensor([0.25, 0.75]))
+
+
+class TestBinarizer:
+    def test_scalar_and_vector(self):
+        from flowml.feature import Binarizer
+
+        df = FlowFrame({"a": [0.0, 0.5, 1.0], "v": torch.tensor([[-1.0, 2.0]] * 3)})
+        b = Binarizer(inputCol="a", outputCol="ab", threshold=0.5)
+        out = b.transform(df)
+        assert out["ab"].tolist() == [0.0, 0.0, 1.0]  # strictly greater
+        b2 = Binarizer(inputCol="v", outputCol="vb")
+        assert b2.transform(df)["vb"].tolist() == [[0.0, 1.0]] * 3
