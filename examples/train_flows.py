@@ -144,6 +144,13 @@ def main():
 
     model = pipe.fit(train)
     pred = model.transform(test)
+    # map predicted indices back to label strings (IndexToString)
+    from flowml.feature import IndexToString
+
+    label_stage = model.stages[3]  # the label StringIndexerModel
+    pred = IndexToString(
+        inputCol="prediction", outputCol="predictedLabel", labels=label_stage.labels
+    ).transform(pred)
     acc = MulticlassClassificationEvaluator(
         labelCol="labelIdx", metricName="accuracy"
     ).evaluate(pred)
@@ -152,7 +159,9 @@ def main():
         labelCol="labelIdx", metricName="macroF1"
     ).evaluate(pred)
     if rank == 0:
+        head = pred.select("predictedLabel").head(3)
         print(f"model={args.model} test accuracy={acc:.4f} weighted-F1={f1:.4f} macro-F1={mf1:.4f}")
+        print("sample predictions:", [r["predictedLabel"] for r in head])
         if args.save:
             model.write().overwrite().save(args.save)
             print(f"saved PipelineModel to {args.save}")

@@ -63,6 +63,9 @@ _register("StringIndexerModel", "org.apache.spark.ml.feature.StringIndexerModel"
 _register("OneHotEncoder", "org.apache.spark.ml.feature.OneHotEncoder")
 _register("OneHotEncoderModel", "org.apache.spark.ml.feature.OneHotEncoderModel")
 _register("VectorAssembler", "org.apache.spark.ml.feature.VectorAssembler")
+_register("IndexToString", "org.apache.spark.ml.feature.IndexToString")
+_register("Normalizer", "org.apache.spark.ml.feature.Normalizer")
+_register("Binarizer", "org.apache.spark.ml.feature.Binarizer")
 _register("StandardScaler", "org.apache.spark.ml.feature.StandardScaler")
 _register("StandardScalerModel", "org.apache.spark.ml.feature.StandardScalerModel")
 _register("MinMaxScaler", "org.apache.spark.ml.feature.MinMaxScaler")
