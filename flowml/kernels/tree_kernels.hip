@@ -105,7 +105,7 @@ extern "C" __global__ void flowml_tree_hist_cls(
     int bag_mode, const float* __restrict__ bag_cdf,
     unsigned long long bag_seed, long long row_offset,
     long long N, int F, int Fp, int B, int C, int l_max,
-    long long rows_per_block) {
+    long long rows_per_block, int stage_bytes) {
   extern __shared__ float s_hist[];
   const int t = blockIdx.x;
   const int cnt = tree_cnt[t];
@@ -121,17 +121,40 @@ extern "C" __global__ void flowml_tree_hist_cls(
   const long long r0 = blockIdx.y * rows_per_block;
   const long long r1 = min(r0 + rows_per_block, N);
   const int stride_fbc = Fp * B * C;
+  // staged codes tile right after the histograms (16-B aligned carve):
+  // per-lane byte gathers from GLOBAL split into ~F cache lines per wave
+  // instruction (the TA-throughput floor); a coalesced 16-B block copy into
+  // LDS followed by LDS byte reads removes that (stage_bytes==0 disables it
+  // when F is too wide to fit beside the histograms).
+  uint8_t* s_codes =
+      (uint8_t*)(s_hist + ((l_max * (long long)stride_fbc + 3) & ~3LL));
+  const bool staged = stage_bytes > 0;
   float cdf[8];
   if (bag_mode >= 2) {
 #pragma unroll
     for (int i = 0; i < 8; ++i) cdf[i] = bag_cdf[i];
   }
 
-#ifndef FLOWML_HIST_UNROLL
-#define FLOWML_HIST_UNROLL 1
-#endif
-#pragma unroll FLOWML_HIST_UNROLL
-  for (long long r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+  for (long long sub = r0; sub < r1; sub += blockDim.x) {
+    const int rows_in = (int)min((long long)blockDim.x, r1 - sub);
+    const long long r = sub + threadIdx.x;
+    const uint8_t* crow;
+    if (staged) {
+      __syncthreads();  // previous sub-tile fully consumed
+      const uint8_t* src = codes + sub * F;
+      const int total = rows_in * F;
+      const int total16 = total & ~15;
+      for (int i = threadIdx.x * 16; i < total16; i += blockDim.x * 16)
+        *reinterpret_cast<uint4*>(s_codes + i) =
+            *reinterpret_cast<const uint4*>(src + i);
+      for (int i = total16 + threadIdx.x; i < total; i += blockDim.x)
+        s_codes[i] = src[i];
+      __syncthreads();
+      crow = s_codes + threadIdx.x * F;
+    } else {
+      crow = codes + r * F;
+    }
+    if (threadIdx.x >= rows_in) continue;
     int nid = nid_t[r];
     if (advancing) {
       if (nid >= 0) {
@@ -139,7 +162,7 @@ extern "C" __global__ void flowml_tree_hist_cls(
         if (f < 0) {
           nid = -1;  // reached a leaf: row is done
         } else {
-          nid = (codes[r * F + f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
+          nid = (crow[f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
         }
         nid_t[r] = nid;
       }
@@ -160,7 +183,6 @@ extern "C" __global__ void flowml_tree_hist_cls(
     }
     const int gl = g - g_lo;
     const int lab = labels[r];
-    const uint8_t* crow = codes + r * F;
     const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
     float* hrow = s_hist + (long long)gl * stride_fbc;
     for (int fp = 0; fp < Fp; ++fp) {
@@ -195,7 +217,7 @@ extern "C" __global__ void flowml_tree_hist_reg(
     int bag_mode, const float* __restrict__ bag_cdf,
     unsigned long long bag_seed, long long row_offset,
     long long N, int F, int Fp, int B,
-    int l_max, long long rows_per_block) {
+    int l_max, long long rows_per_block, int stage_bytes) {
   extern __shared__ float s_hist[];
   const int t = blockIdx.x;
   const int cnt = tree_cnt[t];
@@ -211,13 +233,35 @@ extern "C" __global__ void flowml_tree_hist_reg(
   const long long r0 = blockIdx.y * rows_per_block;
   const long long r1 = min(r0 + rows_per_block, N);
   const int stride_fb3 = Fp * B * 3;
+  uint8_t* s_codes =
+      (uint8_t*)(s_hist + ((l_max * (long long)stride_fb3 + 3) & ~3LL));
+  const bool staged = stage_bytes > 0;
   float cdf[8];
   if (bag_mode >= 2) {
 #pragma unroll
     for (int i = 0; i < 8; ++i) cdf[i] = bag_cdf[i];
   }
 
-  for (long long r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+  for (long long sub = r0; sub < r1; sub += blockDim.x) {
+    const int rows_in = (int)min((long long)blockDim.x, r1 - sub);
+    const long long r = sub + threadIdx.x;
+    const uint8_t* crow;
+    if (staged) {
+      __syncthreads();
+      const uint8_t* src = codes + sub * F;
+      const int total = rows_in * F;
+      const int total16 = total & ~15;
+      for (int i = threadIdx.x * 16; i < total16; i += blockDim.x * 16)
+        *reinterpret_cast<uint4*>(s_codes + i) =
+            *reinterpret_cast<const uint4*>(src + i);
+      for (int i = total16 + threadIdx.x; i < total; i += blockDim.x)
+        s_codes[i] = src[i];
+      __syncthreads();
+      crow = s_codes + threadIdx.x * F;
+    } else {
+      crow = codes + r * F;
+    }
+    if (threadIdx.x >= rows_in) continue;
     int nid = nid_t[r];
     if (advancing) {
       if (nid >= 0) {
@@ -225,7 +269,7 @@ extern "C" __global__ void flowml_tree_hist_reg(
         if (f < 0) {
           nid = -1;
         } else {
-          nid = (codes[r * F + f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
+          nid = (crow[f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
         }
         nid_t[r] = nid;
       }
@@ -248,7 +292,6 @@ extern "C" __global__ void flowml_tree_hist_reg(
     const float tv = targets[r];
     const float wt = w * tv;
     const float wt2 = wt * tv;
-    const uint8_t* crow = codes + r * F;
     const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
     float* hrow = s_hist + (long long)gl * stride_fb3;
     for (int fp = 0; fp < Fp; ++fp) {
@@ -481,12 +524,15 @@ extern "C" void launch_tree_hist_cls(
   if (ideal < 2048) ideal = 2048;
   if (ideal < rows_per_block) rows_per_block = ideal;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
-  const size_t lds = (size_t)l_max * Fp * B * C * sizeof(float);
+  const size_t hist_b = ((size_t)l_max * Fp * B * C * sizeof(float) + 15) & ~15ULL;
+  int stage_bytes = block * F;
+  if (hist_b + (size_t)stage_bytes > 150 * 1024) stage_bytes = 0;  // fall back
+  const size_t lds = hist_b + stage_bytes;
   hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(T, tiles), dim3(block), lds,
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
                      labels, bag, fs, out, adv_feat, adv_thr, adv_left,
                      adv_right, bag_mode, bag_cdf, bag_seed, row_offset, N, F,
-                     Fp, B, C, l_max, rows_per_block);
+                     Fp, B, C, l_max, rows_per_block, stage_bytes);
 }
 
 extern "C" void launch_tree_hist_reg(
@@ -508,12 +554,15 @@ extern "C" void launch_tree_hist_reg(
   if (ideal < 2048) ideal = 2048;
   if (ideal < rows_per_block) rows_per_block = ideal;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
-  const size_t lds = (size_t)l_max * Fp * B * 3 * sizeof(float);
+  const size_t hist_b = ((size_t)l_max * Fp * B * 3 * sizeof(float) + 15) & ~15ULL;
+  int stage_bytes = block * F;
+  if (hist_b + (size_t)stage_bytes > 150 * 1024) stage_bytes = 0;  // fall back
+  const size_t lds = hist_b + stage_bytes;
   hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(T, tiles), dim3(block), lds,
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
                      targets, bag, fs, out, adv_feat, adv_thr, adv_left,
                      adv_right, bag_mode, bag_cdf, bag_seed, row_offset, N, F,
-                     Fp, B, l_max, rows_per_block);
+                     Fp, B, l_max, rows_per_block, stage_bytes);
 }
 
 extern "C" void launch_node_advance(const uint8_t* codes, const int* nodeid_in,
