@@ -525,7 +525,12 @@ extern "C" void launch_tree_hist_cls(
   if (ideal < rows_per_block) rows_per_block = ideal;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t hist_b = ((size_t)l_max * Fp * B * C * sizeof(float) + 15) & ~15ULL;
-  int stage_bytes = block * F;
+  // measured: LDS-staged code tiles LOSE ~25% here — adjacent lanes' 41-B
+  // rows share 64-B cache lines, so L1 already coalesces the byte gathers
+  // and the staging barriers/occupancy cost more than they save. Off by
+  // default; keep for wider-F workloads via FLOWML_HIST_STAGE=1.
+  static const int cfg_stage = flowml_env_int("FLOWML_HIST_STAGE", 0);
+  int stage_bytes = cfg_stage ? block * F : 0;
   if (hist_b + (size_t)stage_bytes > 150 * 1024) stage_bytes = 0;  // fall back
   const size_t lds = hist_b + stage_bytes;
   hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(T, tiles), dim3(block), lds,
@@ -555,7 +560,8 @@ extern "C" void launch_tree_hist_reg(
   if (ideal < rows_per_block) rows_per_block = ideal;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
   const size_t hist_b = ((size_t)l_max * Fp * B * 3 * sizeof(float) + 15) & ~15ULL;
-  int stage_bytes = block * F;
+  static const int cfg_stage = flowml_env_int("FLOWML_HIST_STAGE", 0);
+  int stage_bytes = cfg_stage ? block * F : 0;
   if (hist_b + (size_t)stage_bytes > 150 * 1024) stage_bytes = 0;  // fall back
   const size_t lds = hist_b + stage_bytes;
   hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(T, tiles), dim3(block), lds,
