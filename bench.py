@@ -40,14 +40,22 @@ def synth_flows(rows: int, n_features: int, n_classes: int, seed: int, device):
     X[:, 0] = torch.randint(0, 3, (rows,), generator=g, device=device).float()
     X[:, 1] = torch.randint(0, 10, (rows,), generator=g, device=device).float()
     X[:, 2] = torch.randint(0, 2, (rows,), generator=g, device=device).float()
-    # teacher identical across ranks (fixed seed, host RNG): sparse linear map
-    # over 8 active features — learnable by axis-aligned trees AND by LR/MLP
+    # teacher identical across ranks (fixed seed, host RNG): a mix of a
+    # sparse linear map and axis-aligned threshold rules, so linear models
+    # (LR/MLP) AND trees both have learnable structure
     gt = torch.Generator().manual_seed(1234)
     Wt = torch.zeros(n_features, n_classes)
     active = torch.randperm(n_features, generator=gt)[:8]
-    Wt[active] = 1.5 * torch.randn(8, n_classes, generator=gt)
+    Wt[active] = 1.2 * torch.randn(8, n_classes, generator=gt)
     Wt = Wt.to(device)
-    logits = X @ Wt + 0.5 * torch.randn(rows, n_classes, device=device, generator=g)
+    logits = X @ Wt
+    # axis-aligned rules on 6 features: per class, indicator(x_f > thr)·w
+    rule_f = torch.randperm(n_features, generator=gt)[:6]
+    rule_t = 0.5 * torch.randn(6, generator=gt)
+    rule_w = (2.0 * torch.randn(6, n_classes, generator=gt)).to(device)
+    ind = (X[:, rule_f.to(device)] > rule_t.to(device)).to(torch.float32)
+    logits = logits + ind @ rule_w
+    logits = logits + 0.5 * torch.randn(rows, n_classes, device=device, generator=g)
     y = logits.argmax(dim=1).to(torch.float64)
     return X, y
 

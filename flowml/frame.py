@@ -313,6 +313,36 @@ def read_csv(
     return from_arrow(table, device=device)
 
 
+def read_parquet(path: str, columns: Optional[List[str]] = None, device=None) -> FlowFrame:
+    """Columnar Parquet load (pyarrow C++ reader; SURVEY.md L3 'CSV/Parquet I/O')."""
+    import pyarrow.parquet as pq
+
+    table = pq.read_table(path, columns=columns)
+    return from_arrow(table, device=device)
+
+
+def write_parquet(df: FlowFrame, path: str, compression: str = "zstd") -> None:
+    """Write a FlowFrame to Parquet (vector columns as FixedSizeList)."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    arrays = []
+    names = []
+    for n in df.columns:
+        c = df[n]
+        if isinstance(c, np.ndarray):
+            arrays.append(pa.array(c.tolist(), type=pa.string()))
+        else:
+            t = c.detach().cpu()
+            if t.dim() == 2:
+                flat = pa.array(t.reshape(-1).numpy())
+                arrays.append(pa.FixedSizeListArray.from_arrays(flat, t.shape[1]))
+            else:
+                arrays.append(pa.array(t.numpy()))
+        names.append(n)
+    pq.write_table(pa.table(arrays, names=names), path, compression=compression)
+
+
 def from_arrow(table, device=None) -> FlowFrame:
     """Build a FlowFrame from a pyarrow Table (zero-copy where possible)."""
     import pyarrow as pa
@@ -322,6 +352,9 @@ def from_arrow(table, device=None) -> FlowFrame:
         combined = col.combine_chunks()
         if pa.types.is_string(combined.type) or pa.types.is_large_string(combined.type):
             cols[name] = np.asarray(combined.to_pylist(), dtype=object)
+        elif pa.types.is_fixed_size_list(combined.type) or pa.types.is_list(combined.type):
+            np_arr = np.stack(combined.to_numpy(zero_copy_only=False))
+            cols[name] = torch.from_numpy(np.ascontiguousarray(np_arr))
         else:
             np_arr = combined.to_numpy(zero_copy_only=False)
             cols[name] = torch.from_numpy(np.ascontiguousarray(np_arr))

@@ -110,3 +110,23 @@ def test_streaming_classifier_cpu_fallback():
     eager = m.predictRaw(X[:500])
     assert torch.allclose(raw, eager)
     assert torch.equal(pred, eager.argmax(dim=1))
+
+
+def test_parquet_round_trip(tmp_path):
+    from flowml.frame import read_parquet, write_parquet
+
+    df = FlowFrame(
+        {
+            "s": ["a", "b", "c"],
+            "x": [1.5, 2.5, 3.5],
+            "v": torch.tensor([[1.0, 2.0], [3.0, 4.0], [5.0, 6.0]]),
+        }
+    )
+    p = str(tmp_path / "t.parquet")
+    write_parquet(df, p)
+    back = read_parquet(p)
+    assert back["s"].tolist() == ["a", "b", "c"]
+    assert torch.allclose(back["x"], df["x"])
+    assert torch.allclose(back["v"], df["v"])
+    sel = read_parquet(p, columns=["x"])
+    assert sel.columns == ["x"]
