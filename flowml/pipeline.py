@@ -17,6 +17,24 @@ from flowml.param import Param, Params
 
 
 class Pipeline(Estimator):
+    """A pipeline of featurization + estimator stages.
+
+    >>> import torch
+    >>> from flowml import FlowFrame, Pipeline
+    >>> from flowml.feature import StringIndexer, VectorAssembler
+    >>> from flowml.classification import LogisticRegression
+    >>> df = FlowFrame({"proto": ["tcp", "udp", "tcp", "udp"],
+    ...                 "x": [1.0, -1.0, 2.0, -2.0],
+    ...                 "label": [1.0, 0.0, 1.0, 0.0]})
+    >>> pipe = Pipeline(stages=[
+    ...     StringIndexer(inputCol="proto", outputCol="p"),
+    ...     VectorAssembler(inputCols=["p", "x"], outputCol="features"),
+    ...     LogisticRegression(maxIter=10)])
+    >>> model = pipe.fit(df)
+    >>> model.transform(df)["prediction"].tolist()
+    [1.0, 0.0, 1.0, 0.0]
+    """
+
     stages: Param[List[Any]] = Param("undefined", "stages", "a list of pipeline stages")
 
     def __init__(self, stages: Optional[List[Any]] = None):

@@ -21,7 +21,14 @@ from flowml.param import Param, Params, TypeConverters
 
 
 class ParamGridBuilder:
-    """Builder for a param grid used in grid search."""
+    """Builder for a param grid used in grid search.
+
+    >>> from flowml.classification import LogisticRegression
+    >>> lr = LogisticRegression()
+    >>> grid = ParamGridBuilder().addGrid(lr.regParam, [0.0, 0.1]).build()
+    >>> len(grid)
+    2
+    """
 
     def __init__(self):
         self._grid: Dict[Param, List[Any]] = {}
