@@ -421,3 +421,22 @@ class TestWgradBias:
         assert torch.allclose(gW, refW, rtol=1e-2, atol=refW.abs().max() * 1e-3), \
             (gW - refW).abs().max()
         assert torch.allclose(gB, refB, rtol=1e-2, atol=refB.abs().max() * 1e-3)
+
+
+class TestDeepTreesGPU:
+    def test_depth8_chunked_equals_default_budget(self):
+        """Deep frontier forces multi-chunk K2 launches (per-tree LDS cap +
+        maxMemoryInMB): results must not depend on chunking."""
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(77)
+        X = torch.from_numpy(r.normal(size=(60_000, 12)).astype(np.float32)).to(DEV)
+        y = ((X[:, 0] * X[:, 1] + X[:, 2] - X[:, 3] ** 2) > 0).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y}, device=DEV)
+        m1 = RandomForestClassifier(numTrees=6, maxDepth=8, seed=4, maxMemoryInMB=256).fit(df)
+        m2 = RandomForestClassifier(numTrees=6, maxDepth=8, seed=4, maxMemoryInMB=1).fit(df)
+        p1 = m1.transform(df)["prediction"]
+        p2 = m2.transform(df)["prediction"]
+        assert torch.equal(p1.cpu(), p2.cpu())
+        assert max(t.depth() for t in m1._trees) > 5  # actually went deep
