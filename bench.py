@@ -20,10 +20,8 @@ from __future__ import annotations
 
 import argparse
 import json
-import sys
 import time
 
-import numpy as np
 import torch
 
 from flowml import dist
@@ -183,7 +181,7 @@ def bench_gbt(args, device):
 def bench_mlp(args, device):
     """Config 2: MLP 41-128-64-5 bf16; a step = one full-batch fwd+bwd+update."""
     from flowml import ops
-    from flowml.classification.mlp import _flatten, _unflatten, n_weights
+    from flowml.classification.mlp import _flatten, _unflatten
 
     rank = dist.get_rank()
     world = dist.get_world_size()

@@ -26,7 +26,6 @@ from flowml.param import (
     HasPredictionCol,
     HasProbabilityCol,
     HasRawPredictionCol,
-    HasWeightCol,
 )
 
 

@@ -17,7 +17,6 @@ from __future__ import annotations
 
 from typing import List, Optional
 
-import numpy as np
 import torch
 
 from flowml import ops

@@ -17,11 +17,8 @@ probability = sigmoid(2 F(x)), prediction by sign. predictRaw = [-2F, 2F]
 
 from __future__ import annotations
 
-import json
-import os
 from typing import List, Optional
 
-import numpy as np
 import torch
 
 from flowml import dist, ops

@@ -30,7 +30,7 @@ from flowml.classification.base import (
     _label_tensor,
 )
 from flowml.frame import FlowFrame
-from flowml.linalg import DenseMatrix, DenseVector, Vectors
+from flowml.linalg import DenseMatrix, DenseVector
 from flowml.optim import lbfgs, owlqn
 from flowml.param import (
     HasMaxIter,

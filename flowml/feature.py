@@ -24,7 +24,7 @@ downstream kernels. The user-visible per-row view (``collect``) still yields
 
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional, Sequence
+from typing import Dict, List, Optional
 
 import numpy as np
 import torch
@@ -38,7 +38,6 @@ from flowml.param import (
     HasOutputCol,
     HasOutputCols,
     Param,
-    Params,
     TypeConverters,
 )
 

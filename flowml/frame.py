@@ -15,8 +15,7 @@ replacing ``spark.read.csv`` (SURVEY.md §2.2 "CSV ingestion").
 
 from __future__ import annotations
 
-import math
-from typing import Any, Dict, Iterable, List, Optional, Sequence, Tuple, Union
+from typing import Any, Dict, List, Optional, Sequence, Tuple, Union
 
 import numpy as np
 import torch

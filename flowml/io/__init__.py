@@ -258,8 +258,6 @@ def _table_from_rows(rows: List[Dict[str, Any]]):
 
 
 def _save_instance(instance, path: str) -> None:
-    from flowml.pipeline import Pipeline, PipelineModel
-
     cls = type(instance).__name__
     os.makedirs(path, exist_ok=True)
     if cls == "Pipeline":

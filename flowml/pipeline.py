@@ -13,7 +13,7 @@ from __future__ import annotations
 from typing import Any, List, Optional
 
 from flowml.base import Estimator, Model, Transformer
-from flowml.param import Param, Params
+from flowml.param import Param
 
 
 class Pipeline(Estimator):

@@ -13,7 +13,7 @@ from __future__ import annotations
 import json
 import os
 import shutil
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 
 def save_gbt_checkpoint(ckpt_dir: str, trees, weights: List[float], rounds: int, num_features: int) -> str:
