@@ -14,8 +14,6 @@ argmax), exactly once per transform.
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 
 from flowml.base import Estimator, Model
