@@ -478,10 +478,26 @@ def grow_trees(
     thr_host = thresholds.cpu().numpy()
     level = 0
     pending_adv = None  # fused K4b: applied inside the first K2 chunk of a level
+    hist_global_env = int(os.environ.get("FLOWML_HIST_GLOBAL", "-1"))
     while len(fr_t) and level < cfg.max_depth:
         nxt_t: List[np.ndarray] = []
         nxt_n: List[np.ndarray] = []
-        for ci, (clo, chi) in enumerate(_make_chunk_slices(fr_t, T, g_max, l_cap)):
+        # Deep levels overflow the per-tree LDS budget and would force one
+        # full pass over the N×F codes PER LDS-sized chunk. For
+        # classification the histogram entries are integer counts, so the
+        # kernel can instead accumulate with exact fp32 HBM atomics in ONE
+        # pass (l_max=0 → GLOBAL mode); only the maxMemoryInMB budget still
+        # chunks. Decision is device-independent (counts only) so the torch
+        # oracle sees identical chunking → bitwise parity. Regression (GBT)
+        # keeps LDS: float target sums are order-sensitive under atomics.
+        slices = _make_chunk_slices(fr_t, T, g_max, l_cap)
+        use_global = False
+        if is_cls and hist_global_env != 0 and len(slices) > 2:
+            budget_slices = _make_chunk_slices(fr_t, T, g_max, g_max)
+            if hist_global_env == 1 or len(slices) >= 2 * len(budget_slices):
+                slices = budget_slices
+                use_global = True
+        for ci, (clo, chi) in enumerate(slices):
             t_chunk0 = _tick()
             chunk_t = fr_t[clo:chi]
             chunk_n = fr_n[clo:chi]
@@ -511,7 +527,8 @@ def grow_trees(
             _acc("chunk_setup", t_chunk0)
             t0 = _tick()
             hist = ops.tree_hist_nodes(
-                codes, nodeid, lut_d, tree_lo, tree_cnt, labels_u8, bag, G, l_max, B,
+                codes, nodeid, lut_d, tree_lo, tree_cnt, labels_u8, bag, G,
+                0 if use_global else l_max, B,
                 S if is_cls else 0, fs, targets=targets,
                 advance=pending_adv if ci == 0 else None, hash_bag=hash_bag,
             )

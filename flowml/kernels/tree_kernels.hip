@@ -111,7 +111,13 @@ extern "C" __global__ void flowml_tree_hist_cls(
   const int cnt = tree_cnt[t];
   const bool advancing = adv_feat != nullptr;
   if (cnt == 0 && !advancing) return;
-  const int hist_sz = cnt ? l_max * Fp * B * C : 0;
+  // l_max == 0 ⇒ GLOBAL mode: accumulate straight into `out` with HBM
+  // atomics instead of LDS privatization. Used for deep levels where the
+  // frontier no longer fits LDS: one pass over the codes instead of one per
+  // LDS-sized chunk. Counts are integers accumulated in fp32 (≤2^24), so
+  // float atomics stay exact ⇒ bitwise deterministic in any order.
+  const bool use_lds = l_max > 0;
+  const int hist_sz = (cnt && use_lds) ? l_max * Fp * B * C : 0;
   for (int i = threadIdx.x; i < hist_sz; i += blockDim.x) s_hist[i] = 0.f;
   __syncthreads();
 
@@ -181,17 +187,17 @@ extern "C" __global__ void flowml_tree_hist_cls(
       if (bw == 0.f) continue;
       w *= bw;
     }
-    const int gl = g - g_lo;
     const int lab = labels[r];
     const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
-    float* hrow = s_hist + (long long)gl * stride_fbc;
+    float* hrow = use_lds ? s_hist + (long long)(g - g_lo) * stride_fbc
+                          : out + (long long)g * stride_fbc;
     for (int fp = 0; fp < Fp; ++fp) {
       const int f = fs_g ? fs_g[fp] : fp;
       const int b = crow[f];
       atomicAdd(&hrow[(fp * B + b) * C + lab], w);
     }
   }
-  if (cnt == 0) return;
+  if (cnt == 0 || !use_lds) return;
   __syncthreads();
   // flush this tree's [g_lo, g_lo+cnt) groups (same layout in LDS and out)
   float* out_t = out + (long long)g_lo * stride_fbc;

@@ -440,3 +440,31 @@ class TestDeepTreesGPU:
         p2 = m2.transform(df)["prediction"]
         assert torch.equal(p1.cpu(), p2.cpu())
         assert max(t.depth() for t in m1._trees) > 5  # actually went deep
+
+
+@pytest.mark.gpu
+class TestGlobalHistGPU:
+    def test_global_hist_mode_bitwise(self, monkeypatch):
+        """GLOBAL (HBM-atomic, one-pass) deep-level histograms vs the
+        LDS-chunked path vs the torch oracle — all bitwise identical
+        (counts are integers, fp32 atomics exact below 2^24)."""
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(123)
+        X = torch.from_numpy(r.normal(size=(80_000, 12)).astype(np.float32)).to(DEV)
+        y = ((X[:, 0] * X[:, 1] - X[:, 4] + X[:, 7] ** 2) > 0.5).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y}, device=DEV)
+
+        def fit(mode, force_torch="0"):
+            monkeypatch.setenv("FLOWML_HIST_GLOBAL", mode)
+            monkeypatch.setenv("FLOWML_FORCE_TORCH_OPS", force_torch)
+            m = RandomForestClassifier(numTrees=8, maxDepth=8, seed=9,
+                                       featureSubsetStrategy="all").fit(df)
+            return m.transform(df)["prediction"].cpu()
+
+        p_lds = fit("0")
+        p_glob = fit("1")
+        p_torch = fit("1", force_torch="1")
+        assert torch.equal(p_lds, p_glob)
+        assert torch.equal(p_glob, p_torch)

@@ -281,3 +281,33 @@ class TestDecisionTree:
             numTrees=1, maxDepth=4, seed=5, bootstrap=False, featureSubsetStrategy="all"
         ).fit(df)
         assert torch.equal(dt.transform(df)["prediction"], rf.transform(df)["prediction"])
+
+
+class TestGlobalHistMode:
+    """Deep-level GLOBAL histogram mode (l_max=0 → HBM atomics, one pass)
+    must be bitwise-equivalent to the LDS-chunked path. With
+    featureSubsetStrategy='all' the chunk layout cannot influence the model,
+    so forcing either mode must give identical predictions."""
+
+    def _fit(self, monkeypatch, mode: str):
+        import numpy as np
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+
+        monkeypatch.setenv("FLOWML_HIST_GLOBAL", mode)
+        r = np.random.default_rng(3)
+        X = torch.from_numpy(r.normal(size=(4000, 10)).astype(np.float32))
+        y = ((X[:, 0] + X[:, 1] * X[:, 2]) > 0).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y})
+        m = RandomForestClassifier(
+            numTrees=8, maxDepth=7, seed=11, featureSubsetStrategy="all",
+            maxMemoryInMB=1,
+        ).fit(df)
+        return m.transform(df)["prediction"]
+
+    def test_global_equals_lds_chunked(self, monkeypatch):
+        p_lds = self._fit(monkeypatch, "0")
+        p_glob = self._fit(monkeypatch, "1")
+        p_auto = self._fit(monkeypatch, "-1")
+        assert torch.equal(p_lds, p_glob)
+        assert torch.equal(p_lds, p_auto)
