@@ -468,3 +468,41 @@ class TestGlobalHistGPU:
         p_torch = fit("1", force_torch="1")
         assert torch.equal(p_lds, p_glob)
         assert torch.equal(p_glob, p_torch)
+
+
+@pytest.mark.gpu
+class TestOddSizesGPU:
+    """Oversubscription / boundary coverage: grids are
+    sized in tiles, so N far from multiples of block·rows_per_block must not
+    read or write out of bounds — checked by bitwise parity vs the oracle."""
+
+    @pytest.mark.parametrize("N", [1, 7, 511, 513, 16385])
+    def test_hist_tiny_and_odd_N(self, N):
+        r = np.random.default_rng(N)
+        F, B, C, T = 5, 8, 3, 2
+        codes = torch.from_numpy(r.integers(0, B, (N, F)).astype(np.uint8)).to(DEV)
+        nodeid = torch.from_numpy(r.integers(0, 2, (T, N)).astype(np.int32)).to(DEV)
+        lut = torch.arange(2, dtype=torch.int32, device=DEV)
+        tree_lo = torch.zeros(T, dtype=torch.int32, device=DEV)
+        tree_cnt = torch.full((T,), 2, dtype=torch.int32, device=DEV)
+        labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8)).to(DEV)
+        gpu = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels,
+                                  None, 2, 2, B, C, None).cpu()
+        cpu = ops._tree_hist_nodes_torch(codes.cpu(), nodeid.cpu(), lut.cpu(),
+                                         labels.cpu(), None, 2, B, C, None, None)
+        assert torch.equal(gpu, cpu)
+
+    @pytest.mark.parametrize("N", [1, 513])
+    def test_predict_odd_N(self, N):
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(42)
+        Xf = torch.from_numpy(r.normal(size=(2000, 6)).astype(np.float32)).to(DEV)
+        yf = (Xf[:, 0] > 0).to(torch.float64)
+        m = RandomForestClassifier(numTrees=4, maxDepth=4, seed=1).fit(
+            FlowFrame({"features": Xf, "label": yf}, device=DEV))
+        Xq = torch.from_numpy(r.normal(size=(N, 6)).astype(np.float32)).to(DEV)
+        p_gpu = m.predictRaw(Xq).cpu()
+        p_cpu = m.predictRaw(Xq.cpu())
+        assert torch.allclose(p_gpu, p_cpu)
