@@ -483,16 +483,17 @@ def grow_trees(
         nxt_t: List[np.ndarray] = []
         nxt_n: List[np.ndarray] = []
         # Deep levels overflow the per-tree LDS budget and would force one
-        # full pass over the N×F codes PER LDS-sized chunk. For
-        # classification the histogram entries are integer counts, so the
-        # kernel can instead accumulate with exact fp32 HBM atomics in ONE
-        # pass (l_max=0 → GLOBAL mode); only the maxMemoryInMB budget still
-        # chunks. Decision is device-independent (counts only) so the torch
-        # oracle sees identical chunking → bitwise parity. Regression (GBT)
-        # keeps LDS: float target sums are order-sensitive under atomics.
+        # full pass over the N×F codes PER LDS-sized chunk. The kernel can
+        # instead accumulate with fp32 HBM atomics in ONE pass (l_max=0 →
+        # GLOBAL mode); only the maxMemoryInMB budget still chunks.
+        # Classification counts are integers ⇒ exact/bitwise in any order;
+        # regression sums were already order-dependent at the LDS flush, so
+        # GLOBAL mode is the same accuracy class. Decision is
+        # device-independent (counts only) so the torch oracle sees
+        # identical chunking → parity holds.
         slices = _make_chunk_slices(fr_t, T, g_max, l_cap)
         use_global = False
-        if is_cls and hist_global_env != 0 and len(slices) > 2:
+        if hist_global_env != 0 and len(slices) > 2:
             budget_slices = _make_chunk_slices(fr_t, T, g_max, g_max)
             if hist_global_env == 1 or len(slices) >= 2 * len(budget_slices):
                 slices = budget_slices

@@ -229,7 +229,12 @@ extern "C" __global__ void flowml_tree_hist_reg(
   const int cnt = tree_cnt[t];
   const bool advancing = adv_feat != nullptr;
   if (cnt == 0 && !advancing) return;
-  const int hist_sz = cnt ? l_max * Fp * B * 3 : 0;
+  // l_max == 0 ⇒ GLOBAL mode (see flowml_tree_hist_cls). Regression sums
+  // are floats, so atomic order shifts ulps — but the LDS path's flush
+  // atomics already make it order-dependent across blocks, so GLOBAL mode
+  // is the same accuracy class (parity tests are tolerance-based).
+  const bool use_lds = l_max > 0;
+  const int hist_sz = (cnt && use_lds) ? l_max * Fp * B * 3 : 0;
   for (int i = threadIdx.x; i < hist_sz; i += blockDim.x) s_hist[i] = 0.f;
   __syncthreads();
 
@@ -294,12 +299,12 @@ extern "C" __global__ void flowml_tree_hist_reg(
       if (bw == 0.f) continue;
       w *= bw;
     }
-    const int gl = g - g_lo;
     const float tv = targets[r];
     const float wt = w * tv;
     const float wt2 = wt * tv;
     const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
-    float* hrow = s_hist + (long long)gl * stride_fb3;
+    float* hrow = use_lds ? s_hist + (long long)(g - g_lo) * stride_fb3
+                          : out + (long long)g * stride_fb3;
     for (int fp = 0; fp < Fp; ++fp) {
       const int f = fs_g ? fs_g[fp] : fp;
       const int b = crow[f];
@@ -309,7 +314,7 @@ extern "C" __global__ void flowml_tree_hist_reg(
       atomicAdd(cell + 2, w);
     }
   }
-  if (cnt == 0) return;
+  if (cnt == 0 || !use_lds) return;
   __syncthreads();
   float* out_t = out + (long long)g_lo * stride_fb3;
   const int flush_sz = cnt * stride_fb3;

@@ -506,3 +506,26 @@ class TestOddSizesGPU:
         p_gpu = m.predictRaw(Xq).cpu()
         p_cpu = m.predictRaw(Xq.cpu())
         assert torch.allclose(p_gpu, p_cpu)
+
+    def test_reg_global_mode_parity(self, monkeypatch):
+        """GLOBAL mode on the regression histogram (GBT deep levels): same
+        accuracy class as LDS (tolerance parity vs fp64 torch oracle)."""
+        from flowml.classification import GBTClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(5)
+        X = torch.from_numpy(r.normal(size=(60_000, 10)).astype(np.float32)).to(DEV)
+        y = ((X[:, 0] + X[:, 1] * X[:, 2] - X[:, 5]) > 0).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y}, device=DEV)
+
+        def fit(mode):
+            monkeypatch.setenv("FLOWML_HIST_GLOBAL", mode)
+            m = GBTClassifier(maxIter=5, maxDepth=8, seed=2).fit(df)
+            return m.transform(df)["prediction"].cpu()
+
+        p_lds = fit("0")
+        p_glob = fit("1")
+        # same splits except ulp-level ties: predictions should agree almost
+        # everywhere (both paths are the same accuracy class)
+        agree = (p_lds == p_glob).double().mean().item()
+        assert agree > 0.999
