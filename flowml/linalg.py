@@ -11,7 +11,7 @@ view only.
 
 from __future__ import annotations
 
-from typing import Any, Iterable, List, Sequence, Union
+from typing import Sequence, Union
 
 import numpy as np
 
