@@ -37,6 +37,10 @@ _SUPPORTED = {
     "precisionByLabel",
     "recallByLabel",
     "fMeasureByLabel",
+    "weightedTruePositiveRate",
+    "weightedFalsePositiveRate",
+    "hammingLoss",
+    "logLoss",
     "macroF1",  # flowml extension (headline metric)
 }
 
@@ -57,6 +61,12 @@ class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeight
     beta: Param[float] = Param(
         "undefined", "beta", "beta value in weightedFMeasure|fMeasureByLabel", TypeConverters.toFloat
     )
+    probabilityCol: Param[str] = Param(
+        "undefined", "probabilityCol", "probability column (for logLoss)", TypeConverters.toString
+    )
+    eps: Param[float] = Param(
+        "undefined", "eps", "probability clamp for logLoss", TypeConverters.toFloat
+    )
 
     def __init__(
         self,
@@ -67,6 +77,8 @@ class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeight
         metricLabel: Optional[float] = None,
         beta: Optional[float] = None,
         weightCol: Optional[str] = None,
+        probabilityCol: Optional[str] = None,
+        eps: Optional[float] = None,
     ):
         super().__init__()
         self._setDefault(
@@ -75,6 +87,8 @@ class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeight
             metricName="f1",
             metricLabel=0.0,
             beta=1.0,
+            probabilityCol="probability",
+            eps=1e-15,
         )
         self._set(
             predictionCol=predictionCol,
@@ -83,6 +97,8 @@ class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeight
             metricLabel=metricLabel,
             beta=beta,
             weightCol=weightCol,
+            probabilityCol=probabilityCol,
+            eps=eps,
         )
 
     def getMetricName(self) -> str:
@@ -94,9 +110,23 @@ class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeight
         return self.set(self.metricName, value)
 
     def isLargerBetter(self) -> bool:
-        return True
+        return self.getMetricName() not in ("hammingLoss", "logLoss")
 
     # -- core ----------------------------------------------------------------
+
+    def _log_loss(self, dataset: FlowFrame) -> float:
+        """Weighted multiclass log loss from probabilityCol (Spark ≥3.1)."""
+        prob = dataset.col(self.getOrDefault(self.probabilityCol))
+        label = dataset.col(self.getOrDefault(self.labelCol))
+        if not isinstance(prob, torch.Tensor) or prob.dim() != 2:
+            raise TypeError("probability column must be a vector column for logLoss")
+        y = label.reshape(-1).to(torch.int64)
+        e = self.getOrDefault(self.eps)
+        p = prob.to(torch.float64).gather(1, y.unsqueeze(1)).squeeze(1).clamp_min(e)
+        if self.isSet(self.weightCol):
+            w = dataset.col(self.getWeightCol()).reshape(-1).to(torch.float64)
+            return float(-(w * p.log()).sum() / w.sum())
+        return float(-p.log().mean())
 
     def _confusion(self, dataset: FlowFrame) -> torch.Tensor:
         pred = dataset.col(self.getOrDefault(self.predictionCol))
@@ -115,10 +145,12 @@ class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeight
         return conf
 
     def evaluate(self, dataset: FlowFrame) -> float:
-        conf = self._confusion(dataset)
         name = self.getMetricName()
         if name not in _SUPPORTED:
             raise ValueError(f"Unsupported metricName {name!r}")
+        if name == "logLoss":
+            return self._log_loss(dataset)
+        conf = self._confusion(dataset)
         total = conf.sum()
         tp = conf.diagonal()
         per_label_actual = conf.sum(dim=1)  # row = true label
@@ -140,8 +172,15 @@ class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeight
             return float((f * per_label_actual).sum() / total)
         if name == "weightedPrecision":
             return float((precision * per_label_actual).sum() / total)
-        if name == "weightedRecall":
+        if name in ("weightedRecall", "weightedTruePositiveRate"):
             return float((recall * per_label_actual).sum() / total)
+        if name == "hammingLoss":
+            return float(1.0 - tp.sum() / total)
+        if name == "weightedFalsePositiveRate":
+            fp = per_label_pred - tp
+            neg = total - per_label_actual
+            fpr = torch.where(neg > 0, fp / neg, fp.new_zeros(fp.shape))
+            return float((fpr * per_label_actual).sum() / total)
         if name == "weightedFMeasure":
             f = fmeasure(precision, recall)
             return float((f * per_label_actual).sum() / total)

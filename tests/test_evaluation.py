@@ -68,3 +68,52 @@ def test_weighted_rows():
 
 def test_is_larger_better():
     assert MulticlassClassificationEvaluator().isLargerBetter()
+
+
+class TestExtendedMetrics:
+    def _df(self):
+        import torch
+        from flowml.frame import FlowFrame
+
+        pred = torch.tensor([0.0, 1.0, 2.0, 0.0, 1.0, 2.0, 0.0, 1.0])
+        label = torch.tensor([0.0, 1.0, 2.0, 1.0, 0.0, 2.0, 0.0, 2.0])
+        prob = torch.tensor([
+            [0.7, 0.2, 0.1], [0.1, 0.8, 0.1], [0.2, 0.2, 0.6], [0.5, 0.3, 0.2],
+            [0.3, 0.6, 0.1], [0.1, 0.1, 0.8], [0.9, 0.05, 0.05], [0.2, 0.5, 0.3],
+        ])
+        return FlowFrame({"prediction": pred, "label": label, "probability": prob})
+
+    def test_hamming_loss(self):
+        from flowml.evaluation import MulticlassClassificationEvaluator
+
+        ev = MulticlassClassificationEvaluator(metricName="hammingLoss")
+        df = self._df()
+        acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(df)
+        assert abs(ev.evaluate(df) - (1 - acc)) < 1e-12
+        assert not ev.isLargerBetter()
+
+    def test_log_loss(self):
+        import numpy as np
+        from flowml.evaluation import MulticlassClassificationEvaluator
+
+        df = self._df()
+        ev = MulticlassClassificationEvaluator(metricName="logLoss")
+        p = df.col("probability").numpy()
+        y = df.col("label").numpy().astype(int)
+        expect = -np.log(np.clip(p[np.arange(len(y)), y], 1e-15, None)).mean()
+        assert abs(ev.evaluate(df) - expect) < 1e-12
+        assert not ev.isLargerBetter()
+
+    def test_weighted_tpr_equals_weighted_recall(self):
+        from flowml.evaluation import MulticlassClassificationEvaluator
+
+        df = self._df()
+        a = MulticlassClassificationEvaluator(metricName="weightedTruePositiveRate").evaluate(df)
+        b = MulticlassClassificationEvaluator(metricName="weightedRecall").evaluate(df)
+        assert a == b
+
+    def test_weighted_fpr_range(self):
+        from flowml.evaluation import MulticlassClassificationEvaluator
+
+        v = MulticlassClassificationEvaluator(metricName="weightedFalsePositiveRate").evaluate(self._df())
+        assert 0.0 <= v <= 1.0
