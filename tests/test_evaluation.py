@@ -98,7 +98,7 @@ class TestExtendedMetrics:
 
         df = self._df()
         ev = MulticlassClassificationEvaluator(metricName="logLoss")
-        p = df.col("probability").numpy()
+        p = df.col("probability").numpy().astype(np.float64)
         y = df.col("label").numpy().astype(int)
         expect = -np.log(np.clip(p[np.arange(len(y)), y], 1e-15, None)).mean()
         assert abs(ev.evaluate(df) - expect) < 1e-12
