@@ -189,12 +189,20 @@ extern "C" __global__ void flowml_tree_hist_cls(
     }
     const int lab = labels[r];
     const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
-    float* hrow = use_lds ? s_hist + (long long)(g - g_lo) * stride_fbc
-                          : out + (long long)g * stride_fbc;
-    for (int fp = 0; fp < Fp; ++fp) {
-      const int f = fs_g ? fs_g[fp] : fp;
-      const int b = crow[f];
-      atomicAdd(&hrow[(fp * B + b) * C + lab], w);
+    // branch on the address space (not a ternary pointer): keeps the LDS
+    // path's atomic a ds_add_f32 instead of a generic flat_atomic_add_f32
+    if (use_lds) {
+      float* hrow = s_hist + (long long)(g - g_lo) * stride_fbc;
+      for (int fp = 0; fp < Fp; ++fp) {
+        const int f = fs_g ? fs_g[fp] : fp;
+        atomicAdd(&hrow[(fp * B + crow[f]) * C + lab], w);
+      }
+    } else {
+      float* hrow = out + (long long)g * stride_fbc;
+      for (int fp = 0; fp < Fp; ++fp) {
+        const int f = fs_g ? fs_g[fp] : fp;
+        atomicAdd(&hrow[(fp * B + crow[f]) * C + lab], w);
+      }
     }
   }
   if (cnt == 0 || !use_lds) return;
@@ -303,15 +311,25 @@ extern "C" __global__ void flowml_tree_hist_reg(
     const float wt = w * tv;
     const float wt2 = wt * tv;
     const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
-    float* hrow = use_lds ? s_hist + (long long)(g - g_lo) * stride_fb3
-                          : out + (long long)g * stride_fb3;
-    for (int fp = 0; fp < Fp; ++fp) {
-      const int f = fs_g ? fs_g[fp] : fp;
-      const int b = crow[f];
-      float* cell = &hrow[(fp * B + b) * 3];
-      atomicAdd(cell + 0, wt);
-      atomicAdd(cell + 1, wt2);
-      atomicAdd(cell + 2, w);
+    // branch on address space: ds_add_f32 on the LDS path (see cls kernel)
+    if (use_lds) {
+      float* hrow = s_hist + (long long)(g - g_lo) * stride_fb3;
+      for (int fp = 0; fp < Fp; ++fp) {
+        const int f = fs_g ? fs_g[fp] : fp;
+        float* cell = &hrow[(fp * B + crow[f]) * 3];
+        atomicAdd(cell + 0, wt);
+        atomicAdd(cell + 1, wt2);
+        atomicAdd(cell + 2, w);
+      }
+    } else {
+      float* hrow = out + (long long)g * stride_fb3;
+      for (int fp = 0; fp < Fp; ++fp) {
+        const int f = fs_g ? fs_g[fp] : fp;
+        float* cell = &hrow[(fp * B + crow[f]) * 3];
+        atomicAdd(cell + 0, wt);
+        atomicAdd(cell + 1, wt2);
+        atomicAdd(cell + 2, w);
+      }
     }
   }
   if (cnt == 0 || !use_lds) return;
