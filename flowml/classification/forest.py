@@ -20,6 +20,7 @@ from typing import List, Optional
 import torch
 
 from flowml import ops
+from flowml.classification.summary import _HasTrainingSummary
 from flowml.classification.base import (
     ProbabilisticClassificationModel,
     ProbabilisticClassifier,
@@ -217,10 +218,15 @@ class RandomForestClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
             })
         self._copyValues(model)
         model._set_parent(self)
+        from flowml.classification.summary import ClassificationTrainingSummary
+
+        model._summary = ClassificationTrainingSummary(
+            model, dataset, self.getLabelCol(), self.getPredictionCol()
+        )
         return model
 
 
-class RandomForestClassificationModel(_TreeEnsembleParams, ProbabilisticClassificationModel):
+class RandomForestClassificationModel(_TreeEnsembleParams, _HasTrainingSummary, ProbabilisticClassificationModel):
     def __init__(
         self,
         trees: Optional[List[FlatTree]] = None,
@@ -278,7 +284,7 @@ class RandomForestClassificationModel(_TreeEnsembleParams, ProbabilisticClassifi
         return _load(path)
 
 
-class DecisionTreeClassificationModel(_TreeEnsembleParams, ProbabilisticClassificationModel):
+class DecisionTreeClassificationModel(_TreeEnsembleParams, _HasTrainingSummary, ProbabilisticClassificationModel):
     """Single decision tree (also the per-tree element of a forest)."""
 
     def __init__(self, tree: Optional[FlatTree] = None, numClasses: int = 2, numFeatures: int = 0):
@@ -356,4 +362,9 @@ class DecisionTreeClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
         )
         self._copyValues(model)
         model._set_parent(self)
+        from flowml.classification.summary import ClassificationTrainingSummary
+
+        model._summary = ClassificationTrainingSummary(
+            model, dataset, self.getLabelCol(), self.getPredictionCol()
+        )
         return model

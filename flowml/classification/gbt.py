@@ -22,6 +22,7 @@ from typing import List, Optional
 import torch
 
 from flowml import dist, ops
+from flowml.classification.summary import _HasTrainingSummary
 from flowml.classification.base import (
     ProbabilisticClassificationModel,
     ProbabilisticClassifier,
@@ -266,6 +267,11 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
         )
         self._copyValues(model)
         model._set_parent(self)
+        from flowml.classification.summary import ClassificationTrainingSummary
+
+        model._summary = ClassificationTrainingSummary(
+            model, dataset, self.getLabelCol(), self.getPredictionCol()
+        )
         return model
 
     def _write_checkpoint(self, trees, weights, rounds, num_features):
@@ -274,7 +280,7 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
         save_gbt_checkpoint(self._checkpoint_dir, trees, weights, rounds, num_features)
 
 
-class GBTClassificationModel(_GBTParams, ProbabilisticClassificationModel):
+class GBTClassificationModel(_GBTParams, _HasTrainingSummary, ProbabilisticClassificationModel):
     def __init__(
         self,
         trees: Optional[List[FlatTree]] = None,

@@ -30,6 +30,7 @@ from flowml.classification.base import (
     _label_tensor,
 )
 from flowml.frame import FlowFrame
+from flowml.classification.summary import ClassificationTrainingSummary
 from flowml.linalg import DenseMatrix, DenseVector
 from flowml.optim import lbfgs, owlqn
 from flowml.param import (
@@ -305,6 +306,10 @@ class LogisticRegression(_LogisticRegressionParams, ProbabilisticClassifier):
         model._summary = LogisticRegressionTrainingSummary(
             objectiveHistory=[float(h) for h in history],
             totalIterations=max(len(history) - 1, 0),
+            model=model,
+            dataset=dataset,
+            labelCol=self.getLabelCol(),
+            predictionCol=self.getPredictionCol(),
         )
         from flowml.utils.logging import get_logger
 
@@ -407,9 +412,14 @@ class LogisticRegressionModel(ProbabilisticClassificationModel, _LogisticRegress
         return _load(path)
 
 
-class LogisticRegressionTrainingSummary:
-    """Mirror of Spark's training summary (objectiveHistory etc.)."""
+class LogisticRegressionTrainingSummary(ClassificationTrainingSummary):
+    """Mirror of Spark's training summary: ``objectiveHistory`` /
+    ``totalIterations`` plus the shared training-set metrics
+    (accuracy, weighted*/byLabel, predictions)."""
 
-    def __init__(self, objectiveHistory: List[float], totalIterations: int):
+    def __init__(self, objectiveHistory: List[float], totalIterations: int,
+                 model=None, dataset=None, labelCol: str = "label",
+                 predictionCol: str = "prediction"):
+        super().__init__(model, dataset, labelCol, predictionCol)
         self.objectiveHistory = objectiveHistory
         self.totalIterations = totalIterations

@@ -26,6 +26,7 @@ from typing import List, Optional
 import torch
 
 from flowml import dist, ops
+from flowml.classification.summary import _HasTrainingSummary
 from flowml.classification.base import (
     ProbabilisticClassificationModel,
     ProbabilisticClassifier,
@@ -233,11 +234,16 @@ class MultilayerPerceptronClassifier(_MLPParams, ProbabilisticClassifier):
         )
         self._copyValues(model)
         model._set_parent(self)
+        from flowml.classification.summary import ClassificationTrainingSummary
+
+        model._summary = ClassificationTrainingSummary(
+            model, dataset, self.getLabelCol(), self.getPredictionCol()
+        )
         model._objective_history = [float(h) for h in history]
         return model
 
 
-class MultilayerPerceptronClassificationModel(_MLPParams, ProbabilisticClassificationModel):
+class MultilayerPerceptronClassificationModel(_MLPParams, _HasTrainingSummary, ProbabilisticClassificationModel):
     def __init__(
         self,
         layers: Optional[List[int]] = None,

@@ -117,3 +117,59 @@ class TestExtendedMetrics:
 
         v = MulticlassClassificationEvaluator(metricName="weightedFalsePositiveRate").evaluate(self._df())
         assert 0.0 <= v <= 1.0
+
+
+class TestTrainingSummaries:
+    def _df(self):
+        import numpy as np
+        import torch
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(0)
+        X = torch.from_numpy(r.normal(size=(400, 5)).astype(np.float32))
+        y = (X[:, 0] + 0.5 * X[:, 1] > 0).to(torch.float64)
+        return FlowFrame({"features": X, "label": y})
+
+    def test_rf_summary(self):
+        from flowml.classification import RandomForestClassifier
+
+        df = self._df()
+        m = RandomForestClassifier(numTrees=5, maxDepth=3, seed=1).fit(df)
+        assert m.hasSummary()
+        s = m.summary
+        assert 0.5 < s.accuracy <= 1.0
+        assert s.labels == [0.0, 1.0]
+        assert len(s.precisionByLabel) == 2
+        assert 0.0 <= s.weightedFalsePositiveRate <= 1.0
+        assert abs(s.weightedTruePositiveRate - s.weightedRecall) < 1e-12
+        assert s.predictions.count() == df.count()
+
+    def test_lr_summary_metrics(self):
+        from flowml.classification import LogisticRegression
+
+        m = LogisticRegression(maxIter=20).fit(self._df())
+        s = m.summary
+        assert s.totalIterations >= 1
+        assert s.objectiveHistory[0] >= s.objectiveHistory[-1]
+        assert 0.5 < s.accuracy <= 1.0
+        assert len(s.fMeasureByLabel()) == 2
+
+    def test_mlp_gbt_summary(self):
+        from flowml.classification import GBTClassifier, MultilayerPerceptronClassifier
+
+        df = self._df()
+        g = GBTClassifier(maxIter=3, maxDepth=2, seed=0).fit(df)
+        assert g.hasSummary() and 0.5 < g.summary.accuracy <= 1.0
+        p = MultilayerPerceptronClassifier(layers=[5, 8, 2], maxIter=10, seed=0).fit(df)
+        assert p.hasSummary() and 0.0 < p.summary.accuracy <= 1.0
+
+    def test_loaded_model_has_no_summary(self, tmp_path):
+        import pytest
+        from flowml.classification import RandomForestClassifier, RandomForestClassificationModel
+
+        m = RandomForestClassifier(numTrees=3, maxDepth=2, seed=1).fit(self._df())
+        m.save(str(tmp_path / "m"))
+        loaded = RandomForestClassificationModel.load(str(tmp_path / "m"))
+        assert not loaded.hasSummary()
+        with pytest.raises(RuntimeError):
+            _ = loaded.summary
