@@ -440,6 +440,8 @@ def grow_trees(
         if timing:
             tmr[key] = tmr.get(key, 0.0) + (_tick() - t0)
 
+    hist_by_level = []  # per-level hist ms (timing only)
+
     arena = FlatTree(stat_width=S, cap=max(4 * T, 64))
     tree_of_parts: List[np.ndarray] = [np.arange(T, dtype=np.int64)]
     # roots are arena nodes 0..T-1; every (tree,row) starts at its root
@@ -542,6 +544,10 @@ def grow_trees(
             )
             if ci == 0:
                 pending_adv = None
+            if timing:
+                while len(hist_by_level) <= level:
+                    hist_by_level.append(0.0)
+                hist_by_level[level] += (_tick() - t0) * 1000
             _acc("hist", t0)
             t0 = _tick()
             dist.all_reduce_(hist.reshape(-1))
@@ -648,6 +654,7 @@ def grow_trees(
 
     if timing:
         print("grow_trees timing:", {k: round(v * 1000, 1) for k, v in tmr.items()})
+        print("hist ms by level:", [round(v, 2) for v in hist_by_level])
     return trees
 
 
