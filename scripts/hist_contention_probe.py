@@ -4,11 +4,14 @@ Same data, same launch shape; only the nodeid spread differs. If 1-node
 (level-0-like) launches are much slower than 16-node launches, early levels
 are contention-bound and wave-private LDS replication would pay.
 """
+import os
+import sys
 import time
 
 import numpy as np
 import torch
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from flowml import ops
 
 dev = "cuda"
