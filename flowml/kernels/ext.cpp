@@ -45,6 +45,8 @@ void launch_gemm_bias_act_f32(const float*, const float*, const float*, float*,
 void launch_act_backward_bf16(const void*, void*, long long, int, void*);
 void launch_wgrad_bias_bf16(const void*, const void*, float*, long long, int,
                             int, void*);
+void launch_wgrad_f32_splitk(const float*, const float*, float*, long long,
+                             int, int, int, long long, void*);
 }
 
 static void* cur_stream() {
@@ -308,6 +310,34 @@ torch::Tensor act_backward_(torch::Tensor h, torch::Tensor delta,
   return delta;
 }
 
+
+// deterministic split-K f32 wgrad: gradW = Dᵀ·X (+bias grad), for the LR
+// outer product where rocBLAS picks a no-split-K tile (2 workgroups).
+std::tuple<torch::Tensor, torch::Tensor> wgrad_f32(torch::Tensor X,
+                                                   torch::Tensor D) {
+  CHK(X);
+  CHK(D);
+  TORCH_CHECK(X.scalar_type() == torch::kFloat32 &&
+                  D.scalar_type() == torch::kFloat32,
+              "f32 only");
+  const long long N = X.size(0);
+  const int F = X.size(1);
+  const int C = D.size(1);
+  TORCH_CHECK(D.size(0) == N, "row mismatch");
+  const int FC = C * (F + 1);
+  TORCH_CHECK(FC <= 1024, "C*(F+1) must be <= 1024");
+  long long chunk = (N + 767) / 768;
+  if (chunk < 128) chunk = 128;
+  const int grid = (int)((N + chunk - 1) / chunk);
+  auto part = torch::empty({grid, FC}, X.options());
+  launch_wgrad_f32_splitk(X.data_ptr<float>(), D.data_ptr<float>(),
+                          part.data_ptr<float>(), N, F, C, grid, chunk,
+                          cur_stream());
+  auto full = part.sum(0);  // fixed-order reduction -> deterministic
+  auto m = full.reshape({C, F + 1});
+  return {m.slice(1, 0, F).contiguous(), m.select(1, F).contiguous()};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bin_codes", &bin_codes, "K4a quantile binning");
   m.def("tree_hist_cls", &tree_hist_cls, "K2 class histograms (all trees)");
@@ -316,6 +346,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("split_scan", &split_scan, "K3 split-gain scan (packed)");
   m.def("tree_predict", &tree_predict, "K8 ensemble inference");
   m.def("logreg_delta", &logreg_delta, "K1 fused logistic delta");
+  m.def("wgrad_f32", &wgrad_f32, "split-K f32 wgrad + bias grad (LR)");
   m.def("softmax_xent_grad", &softmax_xent_grad, "K6 softmax/xent grad");
   m.def("gemm_bias_act", &gemm_bias_act, "K5 fused MFMA GEMM+bias+act");
   m.def("act_backward_", &act_backward_, "K7 activation backward (in-place)");

@@ -363,3 +363,69 @@ extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
                      (const __bf16*)act, out, N, K, F, Kp, Fp, waves_k,
                      waves_f, chunk);
 }
+
+// ---------------------------------------------------------------------------
+// K7f32 — deterministic split-K f32 wgrad for tall-skinny outer products
+// (LR gradient: gradW[c,f] = Σ_r D[r,c]·X[r,f], gradB[c] = Σ_r D[r,c]).
+// rocBLAS picks a no-split-K tile for C×F×N = 5×42×1.25M (2 workgroups on a
+// 256-CU chip); here ~768 blocks each reduce a row-chunk through an LDS tile
+// and write an EXACT partial (fixed sequential order, no atomics) to
+// part[s, c*(F+1)+f]; the host sums over s in fixed order → deterministic.
+// Bias grad fused as implicit all-ones column f == F.
+// ---------------------------------------------------------------------------
+
+#define WGF32_TILE 128
+
+extern "C" __global__ void flowml_wgrad_f32_splitk(
+    const float* __restrict__ X,  // N×F
+    const float* __restrict__ D,  // N×C
+    float* __restrict__ part,     // S×(C·(F+1))
+    long long N, int F, int C, long long rows_per_block) {
+  extern __shared__ float lds[];
+  const int Fp = F + 1;
+  const int FC = C * Fp;
+  const int ncell = (FC + (int)blockDim.x - 1) / (int)blockDim.x;  // ≤4 enforced by launcher
+  float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  const long long r0 = (long long)blockIdx.x * rows_per_block;
+  const long long r1 = min(r0 + rows_per_block, N);
+  float* ldsX = lds;                   // TILE×F
+  float* ldsD = lds + WGF32_TILE * F;  // TILE×C
+  for (long long base = r0; base < r1; base += WGF32_TILE) {
+    const int rows = (int)min((long long)WGF32_TILE, r1 - base);
+    __syncthreads();
+    for (int i = threadIdx.x; i < rows * F; i += blockDim.x)
+      ldsX[i] = X[base * F + i];
+    for (int i = threadIdx.x; i < rows * C; i += blockDim.x)
+      ldsD[i] = D[base * C + i];
+    __syncthreads();
+    for (int k = 0; k < 4; ++k) {
+      if (k >= ncell) break;
+      const int cell = (int)threadIdx.x + k * (int)blockDim.x;
+      if (cell >= FC) break;
+      const int c = cell / Fp, f = cell % Fp;
+      float s = acc[k];
+      if (f == F) {
+        for (int r = 0; r < rows; ++r) s += ldsD[r * C + c];
+      } else {
+        for (int r = 0; r < rows; ++r) s += ldsX[r * F + f] * ldsD[r * C + c];
+      }
+      acc[k] = s;
+    }
+  }
+  float* dst = part + (long long)blockIdx.x * FC;
+  for (int k = 0; k < 4; ++k) {
+    if (k >= ncell) break;
+    const int cell = (int)threadIdx.x + k * (int)blockDim.x;
+    if (cell < FC) dst[cell] = acc[k];
+  }
+}
+
+// caller (ext.cpp) picks grid/chunk so it can size `part` first:
+//   chunk = max(ceil(N/768), 128); grid = ceil(N/chunk); FC = C*(F+1) ≤ 1024
+extern "C" void launch_wgrad_f32_splitk(const float* X, const float* D,
+                                        float* part, long long N, int F, int C,
+                                        int grid, long long chunk, void* stream) {
+  const size_t lds_b = (size_t)WGF32_TILE * (F + C) * sizeof(float);
+  hipLaunchKernelGGL(flowml_wgrad_f32_splitk, dim3(grid), dim3(256), lds_b,
+                     (hipStream_t)stream, X, D, part, N, F, C, chunk);
+}

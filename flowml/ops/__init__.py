@@ -373,8 +373,15 @@ def logreg_grad(
             logits.contiguous(), y.to(torch.int32).contiguous(),
             w.contiguous() if w is not None else _empty(X.device),
         )
-        gradW = delta.T @ X
-        gradB = delta.sum(dim=0)
+        # gradW = deltaᵀX is a C×F×N outer product; rocBLAS picks a
+        # no-split-K tile for it (2 workgroups on 256 CUs, ~2 ms) — the
+        # split-K kernel writes exact per-block partials and sums them in
+        # fixed order (deterministic), with the bias grad fused.
+        if delta.size(1) * (X.size(1) + 1) <= 1024:
+            gradW, gradB = get_ext().wgrad_f32(X, delta.contiguous())
+        else:
+            gradW = delta.T @ X
+            gradB = delta.sum(dim=0)
         return loss, gradW, gradB
     return _logreg_grad_torch(X, y, w, W, b)
 

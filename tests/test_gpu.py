@@ -529,3 +529,25 @@ class TestOddSizesGPU:
         # everywhere (both paths are the same accuracy class)
         agree = (p_lds == p_glob).double().mean().item()
         assert agree > 0.999
+
+
+@pytest.mark.gpu
+class TestWgradF32GPU:
+    @pytest.mark.parametrize("N,F,C", [(1_250_000, 41, 5), (100_001, 13, 7), (257, 41, 5)])
+    def test_parity_vs_rocblas(self, N, F, C):
+        r = np.random.default_rng(N % 1000)
+        X = torch.from_numpy(r.normal(size=(N, F)).astype(np.float32)).to(DEV)
+        D = torch.from_numpy(r.normal(size=(N, C)).astype(np.float32)).to(DEV)
+        gW, gB = ops.get_ext().wgrad_f32(X, D)
+        ref_W = (D.double().T @ X.double()).float()
+        ref_B = D.double().sum(0).float()
+        assert torch.allclose(gW, ref_W, rtol=2e-5, atol=2e-3), (gW - ref_W).abs().max()
+        assert torch.allclose(gB, ref_B, rtol=2e-5, atol=2e-3)
+
+    def test_deterministic(self):
+        r = np.random.default_rng(9)
+        X = torch.from_numpy(r.normal(size=(500_000, 41)).astype(np.float32)).to(DEV)
+        D = torch.from_numpy(r.normal(size=(500_000, 5)).astype(np.float32)).to(DEV)
+        a = ops.get_ext().wgrad_f32(X, D)
+        b = ops.get_ext().wgrad_f32(X, D)
+        assert torch.equal(a[0], b[0]) and torch.equal(a[1], b[1])
