@@ -201,6 +201,12 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
         val_tol = self.getOrDefault(self.validationTol)
 
         train_sel = (~val_mask) if val_mask is not None else None
+        inst_w = None
+        if self.isSet(self.weightCol):
+            # instance weights enter exactly like bag weights: the histogram
+            # stats become (Σ w·r, Σ w·r², Σ w) so leaf values are weighted
+            # mean residuals (Spark 3.0 GBT weightCol semantics)
+            inst_w = dataset.col(self.getWeightCol()).reshape(-1).to(torch.float32)
 
         import os as _os
         import time as _time
@@ -226,6 +232,9 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
             if train_sel is not None:
                 sel = train_sel.to(torch.float32).unsqueeze(0)  # (1, N)
                 bag = sel if bag is None else bag * sel
+            if inst_w is not None:
+                wrow = inst_w.unsqueeze(0)
+                bag = wrow.clone() if bag is None else bag * wrow
             _acc("resid", _t0)
             _t1 = _tick()
             tree_list = grow_trees(codes, resid, bag, thresholds, cfg)
@@ -244,8 +253,12 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
             margin = margin + weights[-1] * pred
             _acc("margin", _t3)
             if val_mask is not None:
-                vl = torch.log1p(torch.exp(-2.0 * ysign[val_mask] * margin[val_mask])).mean()
-                buf = torch.tensor([float(vl), 1.0], device=dev)
+                vls = torch.log1p(torch.exp(-2.0 * ysign[val_mask] * margin[val_mask]))
+                if inst_w is not None:
+                    wv = inst_w[val_mask]
+                    buf = torch.tensor([float((vls * wv).sum()), float(wv.sum())], device=dev)
+                else:
+                    buf = torch.tensor([float(vls.sum()), float(len(vls))], device=dev)
                 dist.all_reduce_(buf)
                 vloss = float(buf[0] / buf[1])
                 if vloss < best_val - val_tol * max(best_val, 1e-12):

@@ -311,3 +311,64 @@ class TestGlobalHistMode:
         p_auto = self._fit(monkeypatch, "-1")
         assert torch.equal(p_lds, p_glob)
         assert torch.equal(p_lds, p_auto)
+
+
+class TestWeightCol:
+    def test_gbt_uniform_weights_equal_duplication(self):
+        """Uniform weight w=2 must train the identical model as doubling
+        every row: the quantile thresholds of a uniformly-replicated multiset
+        are unchanged and 2·t is exact in fp32, so all histogram stats match
+        bitwise and every split and leaf value is identical. (w=3 would NOT
+        be bitwise — 3·t rounds in fp32; non-uniform weights additionally
+        change the threshold sketch.)"""
+        import numpy as np
+        from flowml.classification import GBTClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(4)
+        # integer-grid features: few distinct values per column, so the
+        # equi-depth threshold sketch is duplication-invariant (continuous
+        # data would shift quantile positions under replication)
+        X = torch.from_numpy(r.integers(0, 10, (600, 6)).astype(np.float32))
+        y = ((X[:, 0] + X[:, 1]) > 9).to(torch.float64)
+        w = torch.full((600,), 2.0)
+        df_w = FlowFrame({"features": X, "label": y, "w": w})
+        idx = torch.repeat_interleave(torch.arange(600), 2)
+        df_dup = FlowFrame({"features": X[idx], "label": y[idx]})
+        m_w = GBTClassifier(maxIter=4, maxDepth=3, seed=0, weightCol="w").fit(df_w)
+        m_d = GBTClassifier(maxIter=4, maxDepth=3, seed=0).fit(df_dup)
+        p_w = m_w.transform(df_w)["prediction"]
+        p_d = m_d.transform(df_w)["prediction"]
+        assert torch.equal(p_w, p_d)
+
+    def test_gbt_nonuniform_weights_change_fit(self):
+        import numpy as np
+        from flowml.classification import GBTClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(6)
+        X = torch.from_numpy(r.normal(size=(600, 6)).astype(np.float32))
+        y = ((X[:, 0] + X[:, 1]) > 0).to(torch.float64)
+        w = torch.where(y > 0, torch.full((600,), 8.0), torch.ones(600))
+        df = FlowFrame({"features": X, "label": y, "w": w})
+        m_w = GBTClassifier(maxIter=4, maxDepth=3, seed=0, weightCol="w").fit(df)
+        m_u = GBTClassifier(maxIter=4, maxDepth=3, seed=0).fit(df)
+        assert not torch.equal(m_w.transform(df)["prediction"],
+                               m_u.transform(df)["prediction"])
+
+    def test_rf_weightcol_changes_fit(self):
+        import numpy as np
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(5)
+        X = torch.from_numpy(r.normal(size=(500, 5)).astype(np.float32))
+        y = (X[:, 0] > 0).to(torch.float64)
+        w = torch.where(y > 0, torch.full((500,), 10.0), torch.ones(500))
+        df = FlowFrame({"features": X, "label": y, "w": w})
+        m = RandomForestClassifier(numTrees=5, maxDepth=3, seed=1, weightCol="w",
+                                   bootstrap=False).fit(df)
+        # upweighting class 1 must push recall for class 1 up
+        pred = m.transform(df)["prediction"]
+        rec1 = float(((pred == 1) & (y == 1)).sum() / (y == 1).sum())
+        assert rec1 > 0.9
