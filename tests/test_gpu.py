@@ -551,3 +551,35 @@ class TestWgradF32GPU:
         a = ops.get_ext().wgrad_f32(X, D)
         b = ops.get_ext().wgrad_f32(X, D)
         assert torch.equal(a[0], b[0]) and torch.equal(a[1], b[1])
+
+
+@pytest.mark.gpu
+class TestPipelineGPU:
+    def test_full_pipeline_on_device(self):
+        """End-to-end featurization + fit + transform with every tensor on
+        the GPU: StringIndexer → OneHotEncoder → VectorAssembler → RF."""
+        from flowml import FlowFrame, Pipeline
+        from flowml.classification import RandomForestClassifier
+        from flowml.evaluation import MulticlassClassificationEvaluator
+        from flowml.feature import OneHotEncoder, StringIndexer, VectorAssembler
+
+        r = np.random.default_rng(2)
+        n = 20_000
+        proto = np.array(["tcp", "udp", "icmp"])[r.integers(0, 3, n)]
+        x1 = r.normal(size=n).astype(np.float32)
+        x2 = r.normal(size=n).astype(np.float32)
+        label = ((x1 > 0) ^ (proto == "udp")).astype(np.float64)
+        df = FlowFrame({"proto": proto, "x1": torch.from_numpy(x1),
+                        "x2": torch.from_numpy(x2),
+                        "label": torch.from_numpy(label)}, device=DEV)
+        pipe = Pipeline(stages=[
+            StringIndexer(inputCol="proto", outputCol="protoIdx"),
+            OneHotEncoder(inputCols=["protoIdx"], outputCols=["protoVec"]),
+            VectorAssembler(inputCols=["protoVec", "x1", "x2"], outputCol="features"),
+            RandomForestClassifier(numTrees=10, maxDepth=5, seed=3),
+        ])
+        model = pipe.fit(df)
+        out = model.transform(df)
+        assert out.col("prediction").is_cuda
+        acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(out)
+        assert acc > 0.9
