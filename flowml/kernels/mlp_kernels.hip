@@ -19,6 +19,7 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <stdint.h>
+#include <stdlib.h>
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
@@ -158,6 +159,12 @@ extern "C" __global__ void flowml_act_backward_bf16(
   }
 }
 
+
+static int flowml_env_int_mlp(const char* name, int dflt) {
+  const char* v = getenv(name);
+  return v && *v ? atoi(v) : dflt;
+}
+
 // ---------------------------------------------------------------------------
 // Host launchers
 // ---------------------------------------------------------------------------
@@ -206,7 +213,7 @@ extern "C" void launch_act_backward_bf16(const void* h, void* delta,
 // LDS row strides: +8 bf16 keeps rows 16B-aligned and de-patterns banks
 #define WG_PAD 8
 
-extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
+extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16_v2(
     const __bf16* __restrict__ delta,  // N×K
     const __bf16* __restrict__ act,    // N×F
     float* __restrict__ out,           // K×(F+1) fp32 (zeroed; col F = gB)
@@ -331,6 +338,10 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
   }
 }
 
+extern "C" __global__ void flowml_wgrad_bias_bf16(
+    const __bf16*, const __bf16*, float*, long long, int, int, int, int, int,
+    int, long long);
+
 extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
                                        float* out, long long N, int K, int F,
                                        void* stream) {
@@ -352,12 +363,21 @@ extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
       waves_f = 1;
     }
   }
-  const int blocks = 768;
+  static const int blocks = flowml_env_int_mlp("FLOWML_WGRAD_BLOCKS", 768);
+  static const int version = flowml_env_int_mlp("FLOWML_WGRAD_V", 3);
   long long chunk = (N + blocks - 1) / blocks;
   chunk = (chunk + 31) / 32 * 32;
   const int grid = (int)((N + chunk - 1) / chunk);
-  const size_t lds =
-      (size_t)32 * (Kp + Fp + 2 * WG_PAD) * sizeof(__bf16);
+  if (version == 2) {
+    const size_t lds =
+        (size_t)32 * (Kp + Fp + 2 * WG_PAD) * sizeof(__bf16);
+    hipLaunchKernelGGL(flowml_wgrad_bias_bf16_v2, dim3(grid), dim3(256), lds,
+                       (hipStream_t)stream, (const __bf16*)delta,
+                       (const __bf16*)act, out, N, K, F, Kp, Fp, waves_k,
+                       waves_f, chunk);
+    return;
+  }
+  const size_t lds = (size_t)32 * (Kp + Fp) * sizeof(__bf16);
   hipLaunchKernelGGL(flowml_wgrad_bias_bf16, dim3(grid), dim3(256), lds,
                      (hipStream_t)stream, (const __bf16*)delta,
                      (const __bf16*)act, out, N, K, F, Kp, Fp, waves_k,
@@ -428,4 +448,157 @@ extern "C" void launch_wgrad_f32_splitk(const float* X, const float* D,
   const size_t lds_b = (size_t)WGF32_TILE * (F + C) * sizeof(float);
   hipLaunchKernelGGL(flowml_wgrad_f32_splitk, dim3(grid), dim3(256), lds_b,
                      (hipStream_t)stream, X, D, part, N, F, C, chunk);
+}
+
+// ---------------------------------------------------------------------------
+// K7 v3 — same split-N wgrad, fragments loaded with the gfx950 hardware
+// transpose read. v2's transposed SCALAR gathers are the measured limiter
+// (96 ds_read_u16/tile/wave with 4-way group conflicts; all three layer
+// shapes flatline at ~327 µs). v3 stages each operand into per-16-col tiles
+// whose 32 rows are PERMUTED so that `ds_read_b64_tr_b16` (per 16-lane
+// group: a contiguous 128 B block read as a 4×16 bf16 matrix, lane i gets
+// column i — mapping verified empirically, scripts/tr_probe.hip) delivers
+// an MFMA fragment in 2 instructions instead of 8 scalar reads per bf16x8.
+// Tile layout: elem(n, c) = tile*512 + ((n>>2)&1)*256 + (n>>3)*64 + (n&3)*16 + c
+// so window w reads rows n = 8g + 4w + j for lane group g, elem j.
+// ---------------------------------------------------------------------------
+
+typedef __bf16 bf16x4_t __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ int wg3_elem(int n, int c) {
+  return ((n >> 2) & 1) * 256 + (n >> 3) * 64 + (n & 3) * 16 + c;
+}
+
+extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
+    const __bf16* __restrict__ delta,  // N×K
+    const __bf16* __restrict__ act,    // N×F
+    float* __restrict__ out,           // K×(F+1) fp32 (zeroed; col F = gB)
+    long long N, int K, int F, int Kp, int Fp,
+    int waves_k, int waves_f, long long chunk) {
+  extern __shared__ __bf16 s_t[];  // dt tiles [nk16][512] then at [nf16][512]
+  const int nk16 = Kp / 16;
+  const int nf16 = Fp / 16;
+  __bf16* dt = s_t;
+  __bf16* at = s_t + nk16 * 512;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  for (int i = tid; i < (nk16 + nf16) * 512; i += blockDim.x)
+    s_t[i] = (__bf16)0.f;
+  __syncthreads();
+
+  const int kfrag = (nk16 + waves_k - 1) / waves_k;  // <= 4
+  const int ffrag = (nf16 + waves_f - 1) / waves_f;  // <= 8
+  const int wk = wave / waves_f;
+  const int wf = wave % waves_f;
+  const int kf0 = wk * kfrag;
+  const int ff0 = wf * ffrag;
+
+  f32x4 acc[4][8];
+#pragma unroll
+  for (int a = 0; a < 4; ++a)
+#pragma unroll
+    for (int b = 0; b < 8; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+
+  const long long n_lo = blockIdx.x * chunk;
+  const long long n_hi = min(n_lo + chunk, N);
+  const int kvec = K >> 3;
+  const int ktail = K & 7;
+  const int fvec = F >> 3;
+  const int ftail = F & 7;
+
+  auto dt3 = (__attribute__((address_space(3))) bf16x4_t*)dt;
+  auto at3 = (__attribute__((address_space(3))) bf16x4_t*)at;
+
+  for (long long n0 = n_lo; n0 < n_hi; n0 += 32) {
+    const int nrows = (int)min((long long)32, n_hi - n0);
+    __syncthreads();
+    // staging: 8-elem (16 B) pieces land inside one tile row (c0 % 8 == 0)
+    for (int i = tid; i < 32 * kvec; i += blockDim.x) {
+      const int n = i / kvec;
+      const int c = i % kvec;
+      bf16x8 v = (n < nrows)
+          ? *reinterpret_cast<const bf16x8*>(delta + (n0 + n) * (long long)K + c * 8)
+          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      const int c0 = c * 8;
+      *reinterpret_cast<bf16x8*>(dt + (c0 >> 4) * 512 + wg3_elem(n, c0 & 15)) = v;
+    }
+    if (ktail) {
+      for (int i = tid; i < 32 * ktail; i += blockDim.x) {
+        const int n = i / ktail;
+        const int k = kvec * 8 + i % ktail;
+        dt[(k >> 4) * 512 + wg3_elem(n, k & 15)] =
+            (n < nrows) ? delta[(n0 + n) * (long long)K + k] : (__bf16)0.f;
+      }
+    }
+    for (int i = tid; i < 32 * fvec; i += blockDim.x) {
+      const int n = i / fvec;
+      const int c = i % fvec;
+      bf16x8 v = (n < nrows)
+          ? *reinterpret_cast<const bf16x8*>(act + (n0 + n) * (long long)F + c * 8)
+          : bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      const int c0 = c * 8;
+      *reinterpret_cast<bf16x8*>(at + (c0 >> 4) * 512 + wg3_elem(n, c0 & 15)) = v;
+    }
+    if (ftail) {
+      for (int i = tid; i < 32 * ftail; i += blockDim.x) {
+        const int n = i / ftail;
+        const int f = fvec * 8 + i % ftail;
+        at[(f >> 4) * 512 + wg3_elem(n, f & 15)] =
+            (n < nrows) ? act[(n0 + n) * (long long)F + f] : (__bf16)0.f;
+      }
+    }
+    if (tid < 32)
+      at[(F >> 4) * 512 + wg3_elem(tid, F & 15)] =
+          (tid < nrows) ? (__bf16)1.f : (__bf16)0.f;
+    __syncthreads();
+
+    // hoisted B fragments: 2 tr reads each, reused across the a-loop
+    bf16x8 bfr[8];
+#pragma unroll
+    for (int b = 0; b < 8; ++b) {
+      if (b >= ffrag || ff0 + b >= nf16) continue;
+      const int base = (ff0 + b) * 128;  // tile offset in bf16x4 units
+      bf16x4_t u0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(at3 + base + lane);
+      bf16x4_t u1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(at3 + base + 64 + lane);
+      bf16x8 f8;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) { f8[j] = u0[j]; f8[4 + j] = u1[j]; }
+      bfr[b] = f8;
+    }
+#pragma unroll
+    for (int a = 0; a < 4; ++a) {
+      if (a >= kfrag || kf0 + a >= nk16) continue;
+      const int base = (kf0 + a) * 128;
+      bf16x4_t v0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(dt3 + base + lane);
+      bf16x4_t v1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(dt3 + base + 64 + lane);
+      bf16x8 af;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) { af[j] = v0[j]; af[4 + j] = v1[j]; }
+#pragma unroll
+      for (int b = 0; b < 8; ++b) {
+        if (b >= ffrag || ff0 + b >= nf16) continue;
+        acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr[b], acc[a][b], 0, 0, 0);
+      }
+    }
+  }
+
+  const int col = lane & 15;
+  const int row_off = (lane >> 4) * 4;
+#pragma unroll
+  for (int a = 0; a < 4; ++a) {
+    if (a >= kfrag || kf0 + a >= nk16) continue;
+#pragma unroll
+    for (int b = 0; b < 8; ++b) {
+      if (b >= ffrag || ff0 + b >= nf16) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int k = (kf0 + a) * 16 + row_off + reg;
+        const int f = (ff0 + b) * 16 + col;
+        if (k < K && f <= F) {
+          atomicAdd(&out[(long long)k * (F + 1) + f], acc[a][b][reg]);
+        }
+      }
+    }
+  }
 }
