@@ -24,6 +24,7 @@ from flowml.param import (
     HasPredictionCol,
     HasProbabilityCol,
     HasRawPredictionCol,
+    HasThresholds,
 )
 
 
@@ -39,7 +40,7 @@ class _ClassifierParams(_PredictorParams, HasRawPredictionCol):
         self._setDefault(rawPredictionCol="rawPrediction")
 
 
-class _ProbabilisticClassifierParams(_ClassifierParams, HasProbabilityCol):
+class _ProbabilisticClassifierParams(_ClassifierParams, HasProbabilityCol, HasThresholds):
     def __init__(self):
         super().__init__()
         self._setDefault(probabilityCol="probability")
@@ -117,8 +118,23 @@ class ProbabilisticClassificationModel(_ProbabilisticClassifierParams, Classific
         if prob_col:
             out = out.withColumn(prob_col, prob)
         if pred_col:
-            # Spark predicts from probability for probabilistic models
-            out = out.withColumn(pred_col, prob.argmax(dim=1).to(torch.float64))
+            # Spark predicts from probability for probabilistic models;
+            # with thresholds set, predict argmax(p/t) (Spark semantics)
+            scored = prob
+            if self.isSet(self.thresholds):
+                t = torch.as_tensor(self.getThresholds(), dtype=prob.dtype,
+                                    device=prob.device)
+                if t.numel() != prob.shape[1]:
+                    raise ValueError(
+                        f"thresholds length {t.numel()} != numClasses {prob.shape[1]}")
+                zero = t == 0
+                if bool(zero.any()):
+                    # Spark: a zero threshold wins outright (at most one)
+                    pred = torch.full((prob.shape[0],), float(int(zero.nonzero()[0])),
+                                      dtype=torch.float64, device=prob.device)
+                    return out.withColumn(pred_col, pred)
+                scored = prob / t
+            out = out.withColumn(pred_col, scored.argmax(dim=1).to(torch.float64))
         return out
 
     def predictProbability(self, features) -> torch.Tensor:

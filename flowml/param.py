@@ -375,6 +375,26 @@ class HasRawPredictionCol(Params):
         return self.set(self.rawPredictionCol, value)
 
 
+class HasThresholds(Params):
+    thresholds: Param[list] = Param(
+        "undefined",
+        "thresholds",
+        "Thresholds in multi-class classification to adjust the probability "
+        "of predicting each class. Array must have length equal to the "
+        "number of classes, with values > 0 excepting that at most one "
+        "value may be 0. The class with largest value p/t is predicted, "
+        "where p is the original probability of that class and t is the "
+        "class's threshold.",
+        TypeConverters.toListFloat,
+    )
+
+    def getThresholds(self) -> list:
+        return self.getOrDefault(self.thresholds)
+
+    def setThresholds(self, value):
+        return self.set(self.thresholds, value)
+
+
 class HasProbabilityCol(Params):
     probabilityCol: Param[str] = Param(
         "undefined",

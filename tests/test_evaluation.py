@@ -173,3 +173,32 @@ class TestTrainingSummaries:
         assert not loaded.hasSummary()
         with pytest.raises(RuntimeError):
             _ = loaded.summary
+
+
+class TestThresholds:
+    def test_multiclass_thresholds(self):
+        import numpy as np
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+        import torch
+
+        r = np.random.default_rng(1)
+        X = torch.from_numpy(r.normal(size=(300, 4)).astype(np.float32))
+        y = (X[:, 0] > 0).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y})
+        m = RandomForestClassifier(numTrees=5, maxDepth=3, seed=0).fit(df)
+        base = m.transform(df)["prediction"]
+        # huge threshold on class 1 -> class 1 nearly never predicted
+        m.setThresholds([1e-6, 1.0])
+        t1 = m.transform(df)["prediction"]
+        assert float((t1 == 1).sum()) <= float((base == 1).sum())
+        assert float((t1 == 0).sum()) > float((base == 0).sum()) * 0.9
+        # zero threshold wins outright
+        m.setThresholds([0.0, 1.0])
+        t0 = m.transform(df)["prediction"]
+        assert torch.all(t0 == 0)
+        # wrong length raises
+        m.setThresholds([0.5, 0.5, 0.5])
+        import pytest
+        with pytest.raises(ValueError):
+            m.transform(df)
