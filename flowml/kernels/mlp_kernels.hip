@@ -363,7 +363,7 @@ extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
       waves_f = 1;
     }
   }
-  static const int blocks = flowml_env_int_mlp("FLOWML_WGRAD_BLOCKS", 768);
+  static const int blocks = flowml_env_int_mlp("FLOWML_WGRAD_BLOCKS", 1536);
   static const int version = flowml_env_int_mlp("FLOWML_WGRAD_V", 3);
   long long chunk = (N + blocks - 1) / blocks;
   chunk = (chunk + 31) / 32 * 32;
