@@ -30,8 +30,15 @@ class FlowFrame:
         self._cols: Dict[str, Column] = {}
         self._nrows: Optional[int] = None
         self.device = torch.device(device) if device is not None else torch.device("cpu")
+        # device is honored, not just recorded: tensor columns are moved so
+        # FlowFrame({...}, device="cuda") behaves like .to("cuda") (string
+        # columns stay host-side numpy by design)
+        move = device is not None
         for name, col in columns.items():
-            self._cols[name] = self._coerce(col)
+            c = self._coerce(col)
+            if move and isinstance(c, torch.Tensor) and c.device != self.device:
+                c = c.to(self.device, non_blocking=True)
+            self._cols[name] = c
         for name, col in self._cols.items():
             n = len(col)
             if self._nrows is None:
