@@ -459,9 +459,20 @@ def grow_trees(
         S if is_cls else 0, fs0, targets=targets, hash_bag=hash_bag,
     )
     dist.all_reduce_(h0.reshape(-1))
-    root_stats = h0.sum(dim=2)[:, 0, :].to(torch.float64)  # T×S
-    root_imp = _impurity_from_stats(root_stats, cfg.impurity)
-    root_cnt = _counts_of(root_stats, cfg.impurity)
+    if is_cls:
+        root_stats = h0.sum(dim=2)[:, 0, :].to(torch.float64)  # T×S
+        root_imp = _impurity_from_stats(root_stats, cfg.impurity)
+        root_cnt = _counts_of(root_stats, cfg.impurity)
+    else:
+        cells0, ns20 = ops.reg_hist_views(h0, T, 1, B)
+        s0 = cells0[:, 0, :, 0].to(torch.float64).sum(-1)
+        n0 = cells0[:, 0, :, 1].to(torch.float64).sum(-1)
+        s20 = ns20.to(torch.float64)
+        n0c = n0.clamp_min(1e-12)
+        m0 = s0 / n0c
+        root_imp = (s20 / n0c - m0 * m0).clamp_min(0.0)
+        root_cnt = n0
+        root_stats = torch.stack([s0, s20, n0], dim=1)  # T×3 (s, s², n)
     rs_np = root_stats.cpu().numpy()
     ri_np = root_imp.cpu().numpy()
     rc_np = root_cnt.cpu().numpy()
@@ -473,9 +484,10 @@ def grow_trees(
 
     # memory budget → groups per histogram chunk (maxMemoryInMB parity) and
     # per-tree LDS cap (GPU kernel constraint)
-    bytes_per_node = sub_F * B * S * 4
+    S_hist = S if is_cls else 2  # reg cells are (Σwt, Σw); Σt² is per-node
+    bytes_per_node = sub_F * B * S_hist * 4
     g_max = max(1, int(cfg.max_memory_mb * 1024 * 1024 // max(bytes_per_node, 1)))
-    l_cap = _hist_lds_cap(sub_F, B, S)
+    l_cap = _hist_lds_cap(sub_F, B, S_hist)
 
     thr_host = thresholds.cpu().numpy()
     level = 0
@@ -554,20 +566,44 @@ def grow_trees(
             _acc("allreduce", t0)
             t0 = _tick()
             # K3 split-gain scan: packed per-node results, one host transfer
-            packed = ops.split_scan_packed(
-                hist, cfg.impurity, cfg.min_instances_per_node, cfg.min_info_gain
-            ).cpu().numpy().astype(np.float64)
+            if is_cls:
+                packed = ops.split_scan_packed(
+                    hist, cfg.impurity, cfg.min_instances_per_node, cfg.min_info_gain
+                ).cpu().numpy().astype(np.float64)
+            else:
+                packed = ops.split_scan_packed_reg(
+                    hist, G, sub_F, B, cfg.min_instances_per_node, cfg.min_info_gain
+                ).cpu().numpy().astype(np.float64)
             _acc("scan", t0)
             t0 = _tick()
             bf = packed[:, 0].astype(np.int64)
             bb = packed[:, 1].astype(np.int64)
             bg = packed[:, 2]
-            il = packed[:, 3]
-            ir = packed[:, 4]
-            cl = packed[:, 5]
-            cr = packed[:, 6]
-            ls = packed[:, 7 : 7 + S]
-            rs = packed[:, 7 + S : 7 + 2 * S]
+            if is_cls:
+                il = packed[:, 3]
+                ir = packed[:, 4]
+                cl = packed[:, 5]
+                cr = packed[:, 6]
+                ls = packed[:, 7 : 7 + S]
+                rs = packed[:, 7 + S : 7 + 2 * S]
+            else:
+                # [f, b, gain, sL, nL, sR, nR, imp_self, s_self, n_self]:
+                # child impurities are unknown here (backfilled after growth)
+                cl = packed[:, 4]
+                cr = packed[:, 6]
+                il = np.full(len(packed), np.nan)
+                ir = np.full(len(packed), np.nan)
+                ls = np.zeros((len(packed), S))
+                ls[:, 0] = packed[:, 3]
+                ls[:, 2] = packed[:, 4]
+                rs = np.zeros((len(packed), S))
+                rs[:, 0] = packed[:, 5]
+                rs[:, 2] = packed[:, 6]
+                # this node's own impurity/Σt² are now known exactly: record
+                arena.impurity[chunk_n] = packed[:, 7]
+                arena.stats[chunk_n, 1] = np.asarray(
+                    ops.reg_hist_views(hist, G, sub_F, B)[1].cpu().numpy(),
+                    dtype=np.float64)
             sel = np.nonzero(bf >= 0)[0]
             if len(sel):
                 k = len(sel)
@@ -589,8 +625,14 @@ def grow_trees(
                 arena.set_splits(chunk_n[sel], gf, bb[sel], raws, lid, rid, bg[sel])
                 tree_of_parts.append(np.repeat(chunk_t[sel], 2))
                 if level + 1 < cfg.max_depth:
-                    keep_l = (cl[sel] >= min_c) & (il[sel] > 0)
-                    keep_r = (cr[sel] >= min_c) & (ir[sel] > 0)
+                    if is_cls:
+                        keep_l = (cl[sel] >= min_c) & (il[sel] > 0)
+                        keep_r = (cr[sel] >= min_c) & (ir[sel] > 0)
+                    else:
+                        # child impurity is unknown until its own scan; a
+                        # pure child resolves next level via gain == 0
+                        keep_l = cl[sel] >= min_c
+                        keep_r = cr[sel] >= min_c
                     mask = np.ravel(np.column_stack([keep_l, keep_r]))
                     nxt_t.append(np.repeat(chunk_t[sel], 2)[mask])
                     nxt_n.append(np.column_stack([lid, rid]).ravel()[mask])
@@ -614,6 +656,39 @@ def grow_trees(
         )
         _acc("advance", t0)
         level += 1
+
+    if not is_cls:
+        # backfill leaf Σt²/impurity: rows sit at their leaves (keep-at-leaf
+        # advance), so one weighted index_add per tree gives every
+        # never-scanned node its Σwt². Values are informational (NodeData
+        # impurity / impurityStats) — split decisions never read them.
+        t0 = _tick()
+        need = np.isnan(arena.impurity)
+        if need.any():
+            s2g = torch.zeros(arena.n_nodes, dtype=torch.float64, device=dev)
+            tf = targets.to(torch.float64)
+            t2 = tf * tf
+            for t in range(T):
+                wv = t2 if bag is None else t2 * bag[t].to(torch.float64)
+                if hash_bag is not None:
+                    bm, cdf, bseed, rofs = hash_bag
+                    import numpy as _np
+
+                    hw = torch.from_numpy(
+                        ops._hash_bag_torch(bm, cdf, bseed,
+                                            _np.arange(rofs, rofs + N), t)
+                    ).to(dev)
+                    wv = wv * hw.to(torch.float64)
+                s2g.index_add_(0, nodeid[t].to(torch.int64), wv)
+            dist.all_reduce_(s2g)
+            s2_np = s2g.cpu().numpy()
+            stats = arena.stats
+            stats[need, 1] = s2_np[need]
+            n_np = np.maximum(stats[:, 2], 1e-12)
+            m_np = stats[:, 0] / n_np
+            imp_np = np.maximum(stats[:, 1] / n_np - m_np * m_np, 0.0)
+            arena.impurity[need] = imp_np[need]
+        _acc("leaf_backfill", t0)
 
     # extract per-tree FlatTrees from the arena (once per fit, vectorized)
     t0 = _tick()

@@ -24,7 +24,7 @@ void launch_tree_hist_reg(const uint8_t*, int*, const int*, const int*,
                           float*, const int*, const int*, const int*,
                           const int*, int, const float*,
                           unsigned long long, long long, long long, int, int,
-                          int, int, int, void*);
+                          int, int, int, int, void*);
 void launch_node_advance(const uint8_t*, const int*, int*, const int*,
                          const int*, const int*, const int*,
                          long long, int, int, void*);
@@ -38,6 +38,8 @@ void launch_softmax_xent(const float*, const int*, const float*, float*,
                          float*, long long, int, void*);
 void launch_split_scan(const float*, float*, int, int, int, int, int, int,
                        double, void*);
+void launch_split_scan_reg(const float*, float*, int, int, int, int, double,
+                           void*);
 void launch_gemm_bias_act_bf16(const void*, const void*, const float*, void*,
                                long long, int, int, int, void*);
 void launch_gemm_bias_act_f32(const float*, const float*, const float*, float*,
@@ -131,7 +133,9 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
   const int T = nodeid.size(0);
   const int Fp = fs.numel() ? fs.size(1) : F;
   const bool adv = adv_feat.numel() > 0;
-  auto out = torch::zeros({G, Fp, B, 3}, codes.options().dtype(torch::kFloat32));
+  // flat: G*Fp*B*2 cells then G node-s2 slots (ops splits the views)
+  auto out = torch::zeros({G * Fp * B * 2 + G},
+                          codes.options().dtype(torch::kFloat32));
   launch_tree_hist_reg(
       codes.data_ptr<uint8_t>(), nodeid.data_ptr<int>(), lut.data_ptr<int>(),
       tree_lo.data_ptr<int>(), tree_cnt.data_ptr<int>(),
@@ -144,8 +148,20 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
       adv ? adv_right.data_ptr<int>() : nullptr, (int)bag_mode,
       bag_cdf.numel() ? bag_cdf.data_ptr<float>() : nullptr,
       (unsigned long long)bag_seed, (long long)row_offset, N, F,
-      T, Fp, (int)B, (int)l_max, cur_stream());
+      T, Fp, (int)B, (int)G, (int)l_max, cur_stream());
   return out;
+}
+
+torch::Tensor split_scan_reg(torch::Tensor hist_flat, int64_t G, int64_t Fp,
+                             int64_t B, int64_t min_instances,
+                             double min_info_gain) {
+  CHK(hist_flat);
+  TORCH_CHECK(hist_flat.numel() == G * Fp * B * 2 + G, "reg hist layout");
+  auto packed = torch::empty({G, 10}, hist_flat.options());
+  launch_split_scan_reg(hist_flat.data_ptr<float>(), packed.data_ptr<float>(),
+                        (int)G, (int)Fp, (int)B, (int)min_instances,
+                        min_info_gain, cur_stream());
+  return packed;
 }
 
 torch::Tensor split_scan(torch::Tensor hist, int64_t imp_code,
@@ -344,6 +360,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("tree_hist_reg", &tree_hist_reg, "K2b regression histograms");
   m.def("node_advance", &node_advance, "K4b node-id advance");
   m.def("split_scan", &split_scan, "K3 split-gain scan (packed)");
+  m.def("split_scan_reg", &split_scan_reg, "K3b variance scan (2-wide + node s2)");
   m.def("tree_predict", &tree_predict, "K8 ensemble inference");
   m.def("logreg_delta", &logreg_delta, "K1 fused logistic delta");
   m.def("wgrad_f32", &wgrad_f32, "split-K f32 wgrad + bias grad (LR)");

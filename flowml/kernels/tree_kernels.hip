@@ -165,12 +165,13 @@ extern "C" __global__ void flowml_tree_hist_cls(
     if (advancing) {
       if (nid >= 0) {
         const int f = adv_feat[nid];
-        if (f < 0) {
-          nid = -1;  // reached a leaf: row is done
-        } else {
+        // rows KEEP their node id at leaves (lut[leaf] = -1 already skips
+        // them in the histograms) so the final per-leaf stats pass can
+        // still address every row
+        if (f >= 0) {
           nid = (crow[f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
+          nid_t[r] = nid;
         }
-        nid_t[r] = nid;
       }
       if (cnt == 0) continue;
     }
@@ -217,7 +218,13 @@ extern "C" __global__ void flowml_tree_hist_cls(
 }
 
 // ---------------------------------------------------------------------------
-// K2b — regression histogram (Σt, Σt², n)
+// K2b — regression histogram. Per-bin cells are (Σwt, Σw) ONLY — the
+// variance split gain is algebraically free of per-bin Σt²
+// (gain = (sL²/nL + sR²/nR − sP²/nP)/nP), so the third atomic per
+// (row,feature) of the naive (Σt,Σt²,n) layout is replaced by ONE per-row
+// atomic into a per-NODE Σwt² slot (needed only for node impurity values):
+// 2·Fp+1 atomics per row instead of 3·Fp. Output layout:
+//   out[0 .. G·Fp·B·2)  = cells, out[G·Fp·B·2 + g] = node Σwt².
 // ---------------------------------------------------------------------------
 
 extern "C" __global__ void flowml_tree_hist_reg(
@@ -230,7 +237,7 @@ extern "C" __global__ void flowml_tree_hist_reg(
     const int* __restrict__ adv_left, const int* __restrict__ adv_right,
     int bag_mode, const float* __restrict__ bag_cdf,
     unsigned long long bag_seed, long long row_offset,
-    long long N, int F, int Fp, int B,
+    long long N, int F, int Fp, int B, int Gtot,
     int l_max, long long rows_per_block, int stage_bytes) {
   extern __shared__ float s_hist[];
   const int t = blockIdx.x;
@@ -242,7 +249,8 @@ extern "C" __global__ void flowml_tree_hist_reg(
   // atomics already make it order-dependent across blocks, so GLOBAL mode
   // is the same accuracy class (parity tests are tolerance-based).
   const bool use_lds = l_max > 0;
-  const int hist_sz = (cnt && use_lds) ? l_max * Fp * B * 3 : 0;
+  const int stride_fb2 = Fp * B * 2;
+  const int hist_sz = (cnt && use_lds) ? l_max * (stride_fb2 + 1) : 0;
   for (int i = threadIdx.x; i < hist_sz; i += blockDim.x) s_hist[i] = 0.f;
   __syncthreads();
 
@@ -251,9 +259,8 @@ extern "C" __global__ void flowml_tree_hist_reg(
   const float* bag_t = bag ? bag + (long long)t * N : nullptr;
   const long long r0 = blockIdx.y * rows_per_block;
   const long long r1 = min(r0 + rows_per_block, N);
-  const int stride_fb3 = Fp * B * 3;
   uint8_t* s_codes =
-      (uint8_t*)(s_hist + ((l_max * (long long)stride_fb3 + 3) & ~3LL));
+      (uint8_t*)(s_hist + ((l_max * (long long)(stride_fb2 + 1) + 3) & ~3LL));
   const bool staged = stage_bytes > 0;
   float cdf[8];
   if (bag_mode >= 2) {
@@ -285,12 +292,10 @@ extern "C" __global__ void flowml_tree_hist_reg(
     if (advancing) {
       if (nid >= 0) {
         const int f = adv_feat[nid];
-        if (f < 0) {
-          nid = -1;
-        } else {
+        if (f >= 0) {  // keep-at-leaf (see cls kernel)
           nid = (crow[f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
+          nid_t[r] = nid;
         }
-        nid_t[r] = nid;
       }
       if (cnt == 0) continue;
     }
@@ -313,32 +318,38 @@ extern "C" __global__ void flowml_tree_hist_reg(
     const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
     // branch on address space: ds_add_f32 on the LDS path (see cls kernel)
     if (use_lds) {
-      float* hrow = s_hist + (long long)(g - g_lo) * stride_fb3;
+      const int gl = g - g_lo;
+      float* hrow = s_hist + (long long)gl * stride_fb2;
       for (int fp = 0; fp < Fp; ++fp) {
         const int f = fs_g ? fs_g[fp] : fp;
-        float* cell = &hrow[(fp * B + crow[f]) * 3];
+        float* cell = &hrow[(fp * B + crow[f]) * 2];
         atomicAdd(cell + 0, wt);
-        atomicAdd(cell + 1, wt2);
-        atomicAdd(cell + 2, w);
+        atomicAdd(cell + 1, w);
       }
+      atomicAdd(&s_hist[l_max * stride_fb2 + gl], wt2);
     } else {
-      float* hrow = out + (long long)g * stride_fb3;
+      float* hrow = out + (long long)g * stride_fb2;
       for (int fp = 0; fp < Fp; ++fp) {
         const int f = fs_g ? fs_g[fp] : fp;
-        float* cell = &hrow[(fp * B + crow[f]) * 3];
+        float* cell = &hrow[(fp * B + crow[f]) * 2];
         atomicAdd(cell + 0, wt);
-        atomicAdd(cell + 1, wt2);
-        atomicAdd(cell + 2, w);
+        atomicAdd(cell + 1, w);
       }
+      atomicAdd(&out[(long long)Gtot * stride_fb2 + g], wt2);
     }
   }
   if (cnt == 0 || !use_lds) return;
   __syncthreads();
-  float* out_t = out + (long long)g_lo * stride_fb3;
-  const int flush_sz = cnt * stride_fb3;
+  float* out_t = out + (long long)g_lo * stride_fb2;
+  const int flush_sz = cnt * stride_fb2;
   for (int i = threadIdx.x; i < flush_sz; i += blockDim.x) {
     const float v = s_hist[i];
     if (v != 0.f) atomicAdd(&out_t[i], v);
+  }
+  float* s2_out = out + (long long)Gtot * stride_fb2 + g_lo;
+  for (int i = threadIdx.x; i < cnt; i += blockDim.x) {
+    const float v = s_hist[l_max * stride_fb2 + i];
+    if (v != 0.f) atomicAdd(&s2_out[i], v);
   }
 }
 
@@ -363,7 +374,7 @@ extern "C" __global__ void flowml_node_advance(
     }
     const int f = feat[nid];
     if (f < 0) {
-      out_t[r] = -1;  // reached a leaf: row is done
+      out_t[r] = nid;  // keep-at-leaf (rows stay addressable for stats passes)
       continue;
     }
     const int code = codes[r * (long long)F + f];
@@ -575,7 +586,7 @@ extern "C" void launch_tree_hist_reg(
     float* out, const int* adv_feat, const int* adv_thr, const int* adv_left,
     const int* adv_right, int bag_mode, const float* bag_cdf,
     unsigned long long bag_seed, long long row_offset,
-    long long N, int F, int T, int Fp, int B, int l_max,
+    long long N, int F, int T, int Fp, int B, int Gtot, int l_max,
     void* stream) {
   hist_opt_in_big_lds();
   static const int cfg_block = flowml_env_int("FLOWML_HIST_BLOCK", 512);
@@ -588,7 +599,9 @@ extern "C" void launch_tree_hist_reg(
   if (ideal < 2048) ideal = 2048;
   if (ideal < rows_per_block) rows_per_block = ideal;
   const int tiles = (int)((N + rows_per_block - 1) / rows_per_block);
-  const size_t hist_b = ((size_t)l_max * Fp * B * 3 * sizeof(float) + 15) & ~15ULL;
+  // cells (Fp*B*2 per node) + one node-s2 slot per node
+  const size_t hist_b =
+      ((size_t)l_max * (Fp * B * 2 + 1) * sizeof(float) + 15) & ~15ULL;
   static const int cfg_stage = flowml_env_int("FLOWML_HIST_STAGE", 0);
   int stage_bytes = cfg_stage ? block * F : 0;
   if (hist_b + (size_t)stage_bytes > 150 * 1024) stage_bytes = 0;  // fall back
@@ -597,7 +610,7 @@ extern "C" void launch_tree_hist_reg(
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
                      targets, bag, fs, out, adv_feat, adv_thr, adv_left,
                      adv_right, bag_mode, bag_cdf, bag_seed, row_offset, N, F,
-                     Fp, B, l_max, rows_per_block, stage_bytes);
+                     Fp, B, Gtot, l_max, rows_per_block, stage_bytes);
 }
 
 extern "C" void launch_node_advance(const uint8_t* codes, const int* nodeid_in,
@@ -815,5 +828,129 @@ extern "C" void launch_split_scan(const float* hist, float* packed, int G,
   const size_t lds = (size_t)Fp * B * S * sizeof(float);
   hipLaunchKernelGGL(flowml_split_scan, dim3(G), dim3(block), lds,
                      (hipStream_t)stream, hist, packed, G, Fp, B, S, imp_code,
+                     min_instances, min_info_gain);
+}
+
+// ---------------------------------------------------------------------------
+// K3b — variance split scan on the 2-wide regression layout
+// (cells (Σwt,Σw) + per-node Σwt²; see K2b). Canonical gain — identical op
+// order to ops.split_scan_packed_reg so CPU/GPU splits match bitwise:
+//   gain = (sL·sL/nL + sR·sR/nR − sP·sP/nP) / nP
+// Packed per-node row (10 f32): [f, b, gain, sL, nL, sR, nR, imp_self,
+// s_self, n_self]; no-split rows set f = −1 but STILL carry imp_self/s/n.
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void flowml_split_scan_reg(
+    const float* __restrict__ hist,  // G·Fp·B·2 cells then G node-s2
+    float* __restrict__ packed,      // G×10
+    int G, int Fp, int B, int min_instances, double min_info_gain) {
+#pragma clang fp contract(off)
+  extern __shared__ float s_h[];  // Fp×B×2 staged slab
+  const int g = blockIdx.x;
+  const int slab = Fp * B * 2;
+  const float* hg = hist + (long long)g * slab;
+  for (int i = threadIdx.x; i < slab; i += blockDim.x) s_h[i] = hg[i];
+  __syncthreads();
+
+  // parent totals from feature 0 (ascending-b order = torch cumsum order)
+  double sP = 0, nP_raw = 0;
+  for (int b = 0; b < B; ++b) {
+    sP += (double)s_h[(0 * B + b) * 2 + 0];
+    nP_raw += (double)s_h[(0 * B + b) * 2 + 1];
+  }
+  const double nP = nP_raw < 1e-12 ? 1e-12 : nP_raw;
+  const double s2P = (double)hist[(long long)G * slab + g];
+  const double m = sP / nP;
+  double imp_self = s2P / nP - m * m;
+  if (imp_self < 0) imp_self = 0;
+  const double gP = sP * sP / nP;
+
+  double best_gain = -1.0 / 0.0;
+  int best_idx = 0x7FFFFFFF;
+  for (int f = threadIdx.x; f < Fp; f += blockDim.x) {
+    double totS = 0, totN = 0;
+    for (int b = 0; b < B; ++b) {
+      totS += (double)s_h[(f * B + b) * 2 + 0];
+      totN += (double)s_h[(f * B + b) * 2 + 1];
+    }
+    double sL = 0, nL = 0;
+    for (int b = 0; b < B - 1; ++b) {
+      sL += (double)s_h[(f * B + b) * 2 + 0];
+      nL += (double)s_h[(f * B + b) * 2 + 1];
+      const double sR = totS - sL;
+      const double nR = totN - nL;
+      if (nL < min_instances || nR < min_instances) continue;
+      const double gL = sL * sL / nL;
+      const double gR = sR * sR / nR;
+      const double gain = (gL + gR - gP) / nP;
+      const int idx = f * (B - 1) + b;
+      if (gain > best_gain || (gain == best_gain && idx < best_idx)) {
+        best_gain = gain;
+        best_idx = idx;
+      }
+    }
+  }
+
+  __shared__ double r_gain[256];
+  __shared__ int r_idx[256];
+  r_gain[threadIdx.x] = best_gain;
+  r_idx[threadIdx.x] = best_idx;
+  __syncthreads();
+  for (int s = blockDim.x / 2; s > 0; s >>= 1) {
+    if (threadIdx.x < s) {
+      const double og = r_gain[threadIdx.x + s];
+      const int oi = r_idx[threadIdx.x + s];
+      if (og > r_gain[threadIdx.x] ||
+          (og == r_gain[threadIdx.x] && oi < r_idx[threadIdx.x])) {
+        r_gain[threadIdx.x] = og;
+        r_idx[threadIdx.x] = oi;
+      }
+    }
+    __syncthreads();
+  }
+
+  if (threadIdx.x == 0) {
+    const double gain = r_gain[0];
+    const int idx = r_idx[0];
+    float* out = packed + (long long)g * 10;
+    const bool ok = (idx != 0x7FFFFFFF) && (gain > 0) &&
+                    (gain >= min_info_gain) && !isinf(gain) && !isnan(gain);
+    out[7] = (float)imp_self;
+    out[8] = (float)sP;
+    out[9] = (float)nP_raw;
+    if (!ok) {
+      out[0] = -1.f;
+      for (int i = 1; i < 7; ++i) out[i] = 0.f;
+      return;
+    }
+    const int f = idx / (B - 1);
+    const int b = idx % (B - 1);
+    double totS = 0, totN = 0;
+    for (int bb = 0; bb < B; ++bb) {
+      totS += (double)s_h[(f * B + bb) * 2 + 0];
+      totN += (double)s_h[(f * B + bb) * 2 + 1];
+    }
+    double sL = 0, nL = 0;
+    for (int bb = 0; bb <= b; ++bb) {
+      sL += (double)s_h[(f * B + bb) * 2 + 0];
+      nL += (double)s_h[(f * B + bb) * 2 + 1];
+    }
+    out[0] = (float)f;
+    out[1] = (float)b;
+    out[2] = (float)gain;
+    out[3] = (float)sL;
+    out[4] = (float)nL;
+    out[5] = (float)(totS - sL);
+    out[6] = (float)(totN - nL);
+  }
+}
+
+extern "C" void launch_split_scan_reg(const float* hist, float* packed, int G,
+                                      int Fp, int B, int min_instances,
+                                      double min_info_gain, void* stream) {
+  const int block = 256;
+  const size_t lds = (size_t)Fp * B * 2 * sizeof(float);
+  hipLaunchKernelGGL(flowml_split_scan_reg, dim3(G), dim3(block), lds,
+                     (hipStream_t)stream, hist, packed, G, Fp, B,
                      min_instances, min_info_gain);
 }

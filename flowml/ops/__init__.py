@@ -160,7 +160,10 @@ def tree_hist_nodes(
     hash_bag=None,         # (mode: 2=poisson|3=bernoulli, cdf 8×f32, seed, row_offset)
 ) -> torch.Tensor:
     """Classification: out[g,f',b,c] += w  (G×F'×B×C float32).
-    Regression (targets given): out[g,f',b,(s,s²,n)] += w·(t,t²,1) (G×F'×B×3).
+    Regression (targets given): FLAT float32 of length G·F'·B·2 + G —
+    cells out[g,f',b,(Σwt, Σw)] followed by per-node Σwt² slots (variance
+    split gain needs no per-bin Σt²; see tree_kernels.hip K2b).
+    Use ``reg_hist_views`` to split the flat tensor.
 
     ``advance``: fused K4b — each (tree,row) is advanced one level down its
     tree (``nodeid`` is updated IN PLACE) before accumulation; pass it on the
@@ -211,9 +214,10 @@ def _tree_hist_nodes_torch(codes, nodeid, lut, labels, bag, n_groups, n_bins,
     N, F = codes.shape
     T = nodeid.shape[0]
     dev = codes.device
-    stat_w = 3 if targets is not None else n_classes
+    stat_w = 2 if targets is not None else n_classes
     Fp = feat_subset.shape[1] if (feat_subset is not None and feat_subset.numel()) else F
     out = torch.zeros(n_groups * Fp * n_bins * stat_w, dtype=torch.float64, device=dev)
+    ns2 = torch.zeros(n_groups, dtype=torch.float64, device=dev)
     lut64 = lut.to(torch.int64)
     for t in range(T):
         nid = nodeid[t].to(torch.int64)
@@ -255,10 +259,83 @@ def _tree_hist_nodes_torch(codes, nodeid, lut, labels, bag, n_groups, n_bins,
             out.scatter_add_(0, base.reshape(-1),
                              (w_a * tv).unsqueeze(1).expand(-1, Fp).reshape(-1))
             out.scatter_add_(0, (base + 1).reshape(-1),
-                             (w_a * tv * tv).unsqueeze(1).expand(-1, Fp).reshape(-1))
-            out.scatter_add_(0, (base + 2).reshape(-1),
                              w_a.unsqueeze(1).expand(-1, Fp).reshape(-1))
+            ns2.scatter_add_(0, g_a, w_a * tv * tv)
+    if targets is not None:
+        # flat [cells | node_s2] — the layout the HIP kernel writes
+        return torch.cat([out, ns2]).to(torch.float32)
     return out.reshape(n_groups, Fp, n_bins, stat_w).to(torch.float32)
+
+
+def split_scan_packed_reg(
+    flat_hist: torch.Tensor,  # G·Fp·B·2 + G flat (already all-reduced)
+    G: int, Fp: int, B: int,
+    min_instances: int,
+    min_info_gain: float,
+) -> torch.Tensor:
+    """K3 variance scan on the 2-wide regression layout. Per-node packed row
+    (G×10 float32): [best_f, best_bin, gain, sL, nL, sR, nR, imp_self,
+    s_self, n_self]. Children impurities are NOT computable here (no per-bin
+    Σt²) — the grower backfills leaf impurities from the per-leaf stats pass.
+    Canonical gain (identical op order on both paths, all f64):
+        gain = (sL·sL/nL + sR·sR/nR − sP·sP/nP) / nP
+    which equals Spark's imp_p − wL·imp_L − wR·imp_R algebraically.
+    """
+    if hip_available(flat_hist):
+        return get_ext().split_scan_reg(
+            flat_hist.contiguous(), int(G), int(Fp), int(B),
+            int(min_instances), float(min_info_gain)
+        )
+    cells, ns2 = reg_hist_views(flat_hist, G, Fp, B)
+    sb = cells[..., 0].to(torch.float64)
+    nb = cells[..., 1].to(torch.float64)
+    sP = sb[:, 0, :].sum(-1)                       # G
+    nP_raw = nb[:, 0, :].sum(-1)
+    nP = nP_raw.clamp_min(1e-12)
+    s2P = ns2.to(torch.float64)
+    m = sP / nP
+    imp_self = (s2P / nP - m * m).clamp_min(0.0)
+    sl = sb.cumsum(-1)[..., : B - 1]               # G,Fp,B-1
+    cl = nb.cumsum(-1)[..., : B - 1]
+    sr = sb.sum(-1, keepdim=True) - sl
+    cr = nb.sum(-1, keepdim=True) - cl
+    valid = (cl >= min_instances) & (cr >= min_instances)
+    clc = cl.clamp_min(1e-300)
+    crc = cr.clamp_min(1e-300)
+    gP = sP * sP / nP
+    gain = (sl * sl / clc + sr * sr / crc - gP.reshape(G, 1, 1)) / nP.reshape(G, 1, 1)
+    gain = torch.where(valid, gain, torch.full_like(gain, -float("inf")))
+    flat = gain.reshape(G, -1)
+    idx = flat.argmax(dim=1)                       # first max wins ties
+    bg = flat.gather(1, idx.unsqueeze(1)).squeeze(1)
+    bf = idx // (B - 1)
+    bb = idx % (B - 1)
+    ok = torch.isfinite(bg) & (bg > 0) & (bg >= min_info_gain)
+    ar = torch.arange(G, device=flat_hist.device)
+    slb = sl[ar, bf, bb]
+    clb = cl[ar, bf, bb]
+    srb = sr[ar, bf, bb]
+    crb = cr[ar, bf, bb]
+    z = torch.zeros_like(bg)
+    out = torch.stack([
+        torch.where(ok, bf.to(torch.float64), torch.full_like(bg, -1.0)),
+        torch.where(ok, bb.to(torch.float64), z),
+        torch.where(ok, bg, z),
+        torch.where(ok, slb, z),
+        torch.where(ok, clb, z),
+        torch.where(ok, srb, z),
+        torch.where(ok, crb, z),
+        imp_self,
+        sP,
+        nP_raw,
+    ], dim=1)
+    return out.to(torch.float32)
+
+
+def reg_hist_views(flat: torch.Tensor, G: int, Fp: int, B: int):
+    """Split the flat regression hist into (cells (G,Fp,B,2), node_s2 (G,))."""
+    cells = flat[: G * Fp * B * 2].view(G, Fp, B, 2)
+    return cells, flat[G * Fp * B * 2 :]
 
 
 # ---------------------------------------------------------------------------
@@ -274,7 +351,7 @@ def node_advance(
     left: torch.Tensor,    # n_nodes int32 (global child ids)
     right: torch.Tensor,   # n_nodes int32
 ) -> torch.Tensor:
-    """Advance every (tree,row) one level; leaves become -1 (K4b)."""
+    """Advance every (tree,row) one level; rows at leaves keep their id (K4b)."""
     if hip_available(codes):
         return get_ext().node_advance(codes, nodeid, feat, thr, left, right)
     T, N = nodeid.shape
@@ -293,8 +370,9 @@ def _node_advance_torch(codes, nodeid, feat_of, thrbin_of, left_of, right_of):
     fx = codes.gather(1, f.clamp_min(0).unsqueeze(1)).squeeze(1).to(torch.int32)
     go_left = fx <= thrbin_of[nid_c]
     nxt = torch.where(go_left, left_of.to(torch.int64)[nid_c], right_of.to(torch.int64)[nid_c])
-    out = torch.where(internal, nxt, torch.full_like(nid, -1))
-    out = torch.where(live, out, nid)
+    # keep-at-leaf: rows at leaves retain their node id (histograms skip
+    # them via lut = -1; the final per-leaf stats pass needs them live)
+    out = torch.where(internal, nxt, nid)
     return out.to(torch.int32)
 
 

@@ -372,7 +372,7 @@ class TestFusedHistFeatures:
 
 
 class TestSplitScanGPU:
-    @pytest.mark.parametrize("imp,S", [("gini", 5), ("entropy", 3), ("variance", 3)])
+    @pytest.mark.parametrize("imp,S", [("gini", 5), ("entropy", 3)])
     def test_packed_parity_bitwise(self, imp, S):
         """K3 kernel (double gain math) == torch fp64 reference, bitwise."""
         r = np.random.default_rng(41)
@@ -583,3 +583,31 @@ class TestPipelineGPU:
         assert out.col("prediction").is_cuda
         acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(out)
         assert acc > 0.9
+
+
+@pytest.mark.gpu
+class TestSplitScanRegGPU:
+    def test_packed_parity_bitwise(self):
+        """K3b (variance, 2-wide cells + node Σt²): GPU kernel vs the torch
+        fp64 reference — identical op order ⇒ bitwise equal packed rows."""
+        r = np.random.default_rng(8)
+        G, Fp, B = 37, 9, 32
+        cells = r.integers(0, 50, (G, Fp, B, 2)).astype(np.float32)
+        ns2 = r.integers(0, 5000, G).astype(np.float32)
+        flat = np.concatenate([cells.reshape(-1), ns2])
+        gpu = ops.split_scan_packed_reg(torch.from_numpy(flat).to(DEV), G, Fp, B, 1, 0.0).cpu()
+        cpu = ops.split_scan_packed_reg(torch.from_numpy(flat), G, Fp, B, 1, 0.0)
+        assert torch.equal(gpu, cpu)
+
+    def test_min_instances_and_no_split(self):
+        # single bin occupied -> no valid split; imp_self still emitted
+        G, Fp, B = 3, 2, 8
+        cells = np.zeros((G, Fp, B, 2), dtype=np.float32)
+        cells[:, :, 0, 0] = 10.0  # all mass in bin 0
+        cells[:, :, 0, 1] = 5.0
+        ns2 = np.full(G, 30.0, dtype=np.float32)
+        flat = np.concatenate([cells.reshape(-1), ns2])
+        out = ops.split_scan_packed_reg(torch.from_numpy(flat).to(DEV), G, Fp, B, 1, 0.0).cpu()
+        assert torch.all(out[:, 0] == -1)
+        expect_imp = 30.0 / 5.0 - 2.0 * 2.0  # s2/n - (s/n)^2
+        assert torch.allclose(out[:, 7], torch.full((G,), expect_imp))

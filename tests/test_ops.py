@@ -62,8 +62,9 @@ def naive_hist(codes, nodeid, lut, labels, bag, G, B, C, fs, targets=None):
     N, F = codes.shape
     T = nodeid.shape[0]
     Fp = fs.shape[1] if fs is not None else F
-    S = 3 if targets is not None else C
+    S = 2 if targets is not None else C
     out = np.zeros((G, Fp, B, S), dtype=np.float64)
+    ns2 = np.zeros(G, dtype=np.float64)
     for t in range(T):
         for i in range(N):
             nid = int(nodeid[t, i])
@@ -75,6 +76,8 @@ def naive_hist(codes, nodeid, lut, labels, bag, G, B, C, fs, targets=None):
             w = float(bag[t, i]) if bag is not None else 1.0
             if w == 0:
                 continue
+            if targets is not None:
+                ns2[g] += w * float(targets[i]) * float(targets[i])
             for fp in range(Fp):
                 f = int(fs[g, fp]) if fs is not None else fp
                 b = int(codes[i, f])
@@ -83,8 +86,9 @@ def naive_hist(codes, nodeid, lut, labels, bag, G, B, C, fs, targets=None):
                 else:
                     tv = float(targets[i])
                     out[g, fp, b, 0] += w * tv
-                    out[g, fp, b, 1] += w * tv * tv
-                    out[g, fp, b, 2] += w
+                    out[g, fp, b, 1] += w
+    if targets is not None:
+        return out, ns2
     return out
 
 
@@ -144,15 +148,18 @@ class TestTreeHistNodes:
         targets = torch.from_numpy(r.normal(size=N).astype(np.float32))
         labels = torch.zeros(N, dtype=torch.uint8)
         tree_cnt = torch.ones(1, dtype=torch.int32)
-        out = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, None, 1, 1, B, 0,
-                                  None, targets=targets)
-        assert out.shape == (1, F, B, 3)
+        flat = ops.tree_hist_nodes(codes, nodeid, lut, tree_lo, tree_cnt, labels, None, 1, 1, B, 0,
+                                   None, targets=targets)
+        out, ns2 = ops.reg_hist_views(flat, 1, F, B)
+        assert out.shape == (1, F, B, 2)
         for f in range(F):
             assert float(out[0, f, :, 0].sum()) == pytest.approx(float(targets.sum()), rel=1e-4)
-            assert float(out[0, f, :, 2].sum()) == N
-        ref = naive_hist(codes.numpy(), nodeid.numpy(), lut.numpy(), labels.numpy(),
-                         None, 1, B, 0, None, targets=targets.numpy())
+            assert float(out[0, f, :, 1].sum()) == N
+        assert float(ns2[0]) == pytest.approx(float((targets.double() ** 2).sum()), rel=1e-5)
+        ref, ref_ns2 = naive_hist(codes.numpy(), nodeid.numpy(), lut.numpy(), labels.numpy(),
+                                  None, 1, B, 0, None, targets=targets.numpy())
         assert np.allclose(out.numpy(), ref, atol=1e-5)
+        assert np.allclose(ns2.numpy(), ref_ns2, atol=1e-4)
 
 
 class TestNodeAdvance:
@@ -167,11 +174,11 @@ class TestNodeAdvance:
         right = torch.tensor([2, -1, 4, -1, -1, -1], dtype=torch.int32)
         n1 = ops.node_advance(codes, nodeid, feat, thr, left, right)
         assert n1[0].tolist() == [1, 2, 2]
-        assert n1[1].tolist() == [-1, -1, -1]  # leaf root -> done
+        assert n1[1].tolist() == [5, 5, 5]  # keep-at-leaf: rows stay addressable
         n2 = ops.node_advance(codes, n1, feat, thr, left, right)
-        assert n2[0].tolist() == [-1, 3, 4]
+        assert n2[0].tolist() == [1, 3, 4]
         n3 = ops.node_advance(codes, n2, feat, thr, left, right)
-        assert n3[0].tolist() == [-1, -1, -1]
+        assert n3[0].tolist() == [1, 3, 4]  # stable at leaves
 
 
 class TestTreePredict:
