@@ -665,11 +665,15 @@ def grow_trees(
         t0 = _tick()
         need = np.isnan(arena.impurity)
         if need.any():
-            s2g = torch.zeros(arena.n_nodes, dtype=torch.float64, device=dev)
-            tf = targets.to(torch.float64)
+            # fp32 accumulation on GPU (native atomicAdd; fp64 index_add on
+            # ROCm is a CAS loop that serializes under leaf contention —
+            # measured 1.5 s vs ~ms); fp64 on CPU (deterministic reference)
+            acc_dt = torch.float32 if dev.type == "cuda" else torch.float64
+            s2g = torch.zeros(arena.n_nodes, dtype=acc_dt, device=dev)
+            tf = targets.to(acc_dt)
             t2 = tf * tf
             for t in range(T):
-                wv = t2 if bag is None else t2 * bag[t].to(torch.float64)
+                wv = t2 if bag is None else t2 * bag[t].to(acc_dt)
                 if hash_bag is not None:
                     bm, cdf, bseed, rofs = hash_bag
                     import numpy as _np
@@ -678,8 +682,9 @@ def grow_trees(
                         ops._hash_bag_torch(bm, cdf, bseed,
                                             _np.arange(rofs, rofs + N), t)
                     ).to(dev)
-                    wv = wv * hw.to(torch.float64)
+                    wv = wv * hw.to(acc_dt)
                 s2g.index_add_(0, nodeid[t].to(torch.int64), wv)
+            s2g = s2g.to(torch.float64)
             dist.all_reduce_(s2g)
             s2_np = s2g.cpu().numpy()
             stats = arena.stats
