@@ -449,38 +449,47 @@ def grow_trees(
         torch.arange(T, dtype=torch.int32, device=dev).unsqueeze(1).expand(T, N).contiguous()
     )
 
-    # root stats per tree: one K2 pass over feature 0 only (lut: root t -> t)
-    lut0 = torch.arange(T, dtype=torch.int32, device=dev)
-    tree_lo0 = torch.arange(T, dtype=torch.int32, device=dev)
-    tree_cnt0 = torch.ones(T, dtype=torch.int32, device=dev)
-    fs0 = torch.zeros((T, 1), dtype=torch.int32, device=dev)
-    h0 = ops.tree_hist_nodes(
-        codes, nodeid, lut0, tree_lo0, tree_cnt0, labels_u8, bag, T, 1, B,
-        S if is_cls else 0, fs0, targets=targets, hash_bag=hash_bag,
-    )
-    dist.all_reduce_(h0.reshape(-1))
-    if is_cls:
-        root_stats = h0.sum(dim=2)[:, 0, :].to(torch.float64)  # T×S
-        root_imp = _impurity_from_stats(root_stats, cfg.impurity)
-        root_cnt = _counts_of(root_stats, cfg.impurity)
-    else:
-        cells0, ns20 = ops.reg_hist_views(h0, T, 1, B)
-        s0 = cells0[:, 0, :, 0].to(torch.float64).sum(-1)
-        n0 = cells0[:, 0, :, 1].to(torch.float64).sum(-1)
-        s20 = ns20.to(torch.float64)
-        n0c = n0.clamp_min(1e-12)
-        m0 = s0 / n0c
-        root_imp = (s20 / n0c - m0 * m0).clamp_min(0.0)
-        root_cnt = n0
-        root_stats = torch.stack([s0, s20, n0], dim=1)  # T×3 (s, s², n)
-    rs_np = root_stats.cpu().numpy()
-    ri_np = root_imp.cpu().numpy()
-    rc_np = root_cnt.cpu().numpy()
     min_c = 2 * cfg.min_instances_per_node
-    arena.bulk_add(rs_np, ri_np, rc_np)
-    grow_ok = (rc_np >= min_c) & (ri_np > 0) & (cfg.max_depth > 0)
-    fr_t = np.nonzero(grow_ok)[0].astype(np.int64)  # frontier tree ids (sorted)
-    fr_n = fr_t.copy()  # global arena ids of the roots
+    if is_cls or cfg.max_depth == 0:
+        # root stats per tree: one K2 pass over feature 0 (lut: root t -> t)
+        lut0 = torch.arange(T, dtype=torch.int32, device=dev)
+        tree_lo0 = torch.arange(T, dtype=torch.int32, device=dev)
+        tree_cnt0 = torch.ones(T, dtype=torch.int32, device=dev)
+        fs0 = torch.zeros((T, 1), dtype=torch.int32, device=dev)
+        h0 = ops.tree_hist_nodes(
+            codes, nodeid, lut0, tree_lo0, tree_cnt0, labels_u8, bag, T, 1, B,
+            S if is_cls else 0, fs0, targets=targets, hash_bag=hash_bag,
+        )
+        dist.all_reduce_(h0.reshape(-1))
+        if is_cls:
+            root_stats = h0.sum(dim=2)[:, 0, :].to(torch.float64)  # T×S
+            root_imp = _impurity_from_stats(root_stats, cfg.impurity)
+            root_cnt = _counts_of(root_stats, cfg.impurity)
+        else:
+            cells0, ns20 = ops.reg_hist_views(h0, T, 1, B)
+            s0 = cells0[:, 0, :, 0].to(torch.float64).sum(-1)
+            n0 = cells0[:, 0, :, 1].to(torch.float64).sum(-1)
+            s20 = ns20.to(torch.float64)
+            n0c = n0.clamp_min(1e-12)
+            m0 = s0 / n0c
+            root_imp = (s20 / n0c - m0 * m0).clamp_min(0.0)
+            root_cnt = n0
+            root_stats = torch.stack([s0, s20, n0], dim=1)  # T×3 (s, s², n)
+        rs_np = root_stats.cpu().numpy()
+        ri_np = root_imp.cpu().numpy()
+        rc_np = root_cnt.cpu().numpy()
+        arena.bulk_add(rs_np, ri_np, rc_np)
+        grow_ok = (rc_np >= min_c) & (ri_np > 0) & (cfg.max_depth > 0)
+        fr_t = np.nonzero(grow_ok)[0].astype(np.int64)  # frontier tree ids
+        fr_n = fr_t.copy()  # global arena ids of the roots
+    else:
+        # regression: the dedicated root pass is redundant — the level-0
+        # scan computes each root's own (s, n, imp) from its histogram, and
+        # the scan's min_instances check turns too-small roots into leaves.
+        # Saves one full K2 pass per GBT round.
+        arena.bulk_add(np.zeros((T, S)), np.full(T, np.nan), np.zeros(T))
+        fr_t = np.arange(T, dtype=np.int64)
+        fr_n = fr_t.copy()
 
     # memory budget → groups per histogram chunk (maxMemoryInMB parity) and
     # per-tree LDS cap (GPU kernel constraint)
@@ -599,8 +608,12 @@ def grow_trees(
                 rs = np.zeros((len(packed), S))
                 rs[:, 0] = packed[:, 5]
                 rs[:, 2] = packed[:, 6]
-                # this node's own impurity/Σt² are now known exactly: record
+                # this node's own impurity/stats are now known exactly:
+                # record (also covers roots — no dedicated root pass for reg)
                 arena.impurity[chunk_n] = packed[:, 7]
+                arena.stats[chunk_n, 0] = packed[:, 8]
+                arena.stats[chunk_n, 2] = packed[:, 9]
+                arena.count[chunk_n] = packed[:, 9]
                 arena.stats[chunk_n, 1] = np.asarray(
                     ops.reg_hist_views(hist, G, sub_F, B)[1].cpu().numpy(),
                     dtype=np.float64)
