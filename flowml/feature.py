@@ -24,6 +24,7 @@ downstream kernels. The user-visible per-row view (``collect``) still yields
 
 from __future__ import annotations
 
+import math
 from typing import Dict, List, Optional
 
 import numpy as np
@@ -640,3 +641,205 @@ class Binarizer(HasInputCol, HasOutputCol, Transformer):
         thr = self.getOrDefault(self.threshold)
         out = (x > thr).to(torch.float64 if x.dim() == 1 else torch.float32)
         return dataset.withColumn(self.getOutputCol(), out)
+
+
+# ---------------------------------------------------------------------------
+# MaxAbsScaler (spark:.../feature/MaxAbsScaler.scala)
+# ---------------------------------------------------------------------------
+
+
+class _MaxAbsScalerParams(HasInputCol, HasOutputCol):
+    pass
+
+
+class MaxAbsScaler(_MaxAbsScalerParams, Estimator):
+    """Scale each feature to [-1, 1] by dividing by its max absolute value
+    (zero-max features pass through unchanged, like Spark)."""
+
+    def __init__(self, *, inputCol: Optional[str] = None, outputCol: Optional[str] = None):
+        super().__init__()
+        self._set(inputCol=inputCol, outputCol=outputCol)
+
+    def _fit(self, dataset: FlowFrame) -> "MaxAbsScalerModel":
+        x = dataset.col(self.getInputCol())
+        if not isinstance(x, torch.Tensor) or x.dim() != 2:
+            raise TypeError("MaxAbsScaler input must be a vector column")
+        max_abs = x.abs().amax(dim=0).to(torch.float32).cpu()
+        model = MaxAbsScalerModel(maxAbs=max_abs)
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class MaxAbsScalerModel(_MaxAbsScalerParams, Model):
+    def __init__(self, maxAbs: Optional[torch.Tensor] = None):
+        super().__init__()
+        self.maxAbs = maxAbs if maxAbs is not None else torch.zeros(0)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getInputCol())
+        d = self.maxAbs.to(device=x.device, dtype=x.dtype)
+        d = torch.where(d == 0, torch.ones_like(d), d)
+        return dataset.withColumn(self.getOutputCol(), x / d)
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+# ---------------------------------------------------------------------------
+# Bucketizer (spark:.../feature/Bucketizer.scala)
+# ---------------------------------------------------------------------------
+
+
+class Bucketizer(HasInputCol, HasOutputCol, HasHandleInvalid, Transformer):
+    """Map a continuous column to bucket indices given split points.
+
+    ``splits`` = [s0 < s1 < ... < sn]; bucket i covers [s_i, s_{i+1}) and the
+    last bucket is closed on the right. Values outside [s0, sn] (or NaN):
+    ``handleInvalid`` error (default) / skip (drop rows) / keep (extra bucket
+    index = numBuckets).
+    """
+
+    splits: Param[list] = Param(
+        "undefined", "splits",
+        "Split points for mapping continuous features into buckets (length >= 3, "
+        "strictly increasing; -inf/inf cover all values).",
+        TypeConverters.toListFloat,
+    )
+
+    def __init__(self, *, splits: Optional[list] = None, inputCol: Optional[str] = None,
+                 outputCol: Optional[str] = None, handleInvalid: Optional[str] = None):
+        super().__init__()
+        self._setDefault(handleInvalid="error")
+        self._set(splits=splits, inputCol=inputCol, outputCol=outputCol,
+                  handleInvalid=handleInvalid)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        sp = self.getOrDefault(self.splits)
+        if sp is None or len(sp) < 3:
+            raise ValueError("Bucketizer requires splits with length >= 3")
+        x = dataset.col(self.getInputCol())
+        if isinstance(x, np.ndarray):
+            raise TypeError("Bucketizer input must be numeric")
+        handle = self.getHandleInvalid()
+        edges = torch.tensor(sp, dtype=torch.float64, device=x.device)
+        xv = x.reshape(-1).to(torch.float64)
+        n_buckets = len(sp) - 1
+        idx = torch.bucketize(xv, edges, right=True) - 1  # [s_i, s_{i+1})
+        idx = torch.where(xv == edges[-1], torch.full_like(idx, n_buckets - 1), idx)
+        invalid = torch.isnan(xv) | (idx < 0) | (idx >= n_buckets)
+        if bool(invalid.any()):
+            if handle == "error":
+                raise ValueError(
+                    "Bucketizer found values outside the splits range "
+                    "(handleInvalid='error')."
+                )
+            if handle == "keep":
+                idx = torch.where(invalid, torch.full_like(idx, n_buckets), idx)
+            elif handle == "skip":
+                out = dataset.withColumn(self.getOutputCol(), idx.to(torch.float64))
+                return out._take_mask((~invalid).cpu().numpy())
+            else:
+                raise ValueError(f"Unsupported handleInvalid: {handle}")
+        return dataset.withColumn(self.getOutputCol(), idx.to(torch.float64))
+
+
+# ---------------------------------------------------------------------------
+# Imputer (spark:.../feature/Imputer.scala)
+# ---------------------------------------------------------------------------
+
+
+class _ImputerParams(HasInputCol, HasInputCols, HasOutputCol, HasOutputCols):
+    strategy: Param[str] = Param(
+        "undefined", "strategy",
+        "strategy for imputation: mean (default) | median | mode",
+        TypeConverters.toString,
+    )
+    missingValue: Param[float] = Param(
+        "undefined", "missingValue",
+        "placeholder for missing values (default NaN)",
+        TypeConverters.toFloat,
+    )
+
+    def _in_out_cols(self):
+        if self.isSet(self.inputCols):
+            return self.getOrDefault(self.inputCols), self.getOrDefault(self.outputCols)
+        return [self.getInputCol()], [self.getOutputCol()]
+
+
+class Imputer(_ImputerParams, Estimator):
+    """Replace missing values with the column mean/median/mode."""
+
+    def __init__(self, *, inputCol: Optional[str] = None, outputCol: Optional[str] = None,
+                 inputCols: Optional[list] = None, outputCols: Optional[list] = None,
+                 strategy: Optional[str] = None, missingValue: Optional[float] = None):
+        super().__init__()
+        self._setDefault(strategy="mean", missingValue=float("nan"))
+        self._set(inputCol=inputCol, outputCol=outputCol, inputCols=inputCols,
+                  outputCols=outputCols, strategy=strategy, missingValue=missingValue)
+
+    def _fit(self, dataset: FlowFrame) -> "ImputerModel":
+        in_cols, _ = self._in_out_cols()
+        strat = self.getOrDefault(self.strategy)
+        mv = self.getOrDefault(self.missingValue)
+        surrogates = []
+        for c in in_cols:
+            x = dataset.col(c).reshape(-1).to(torch.float64)
+            miss = torch.isnan(x) if math.isnan(mv) else (x == mv)
+            good = x[~miss & ~torch.isnan(x)]
+            if not len(good):
+                raise ValueError(f"Imputer: column {c!r} has no non-missing values")
+            if strat == "mean":
+                surrogates.append(float(good.mean()))
+            elif strat == "median":
+                surrogates.append(float(good.median()))
+            elif strat == "mode":
+                vals, cnts = torch.unique(good, return_counts=True)
+                surrogates.append(float(vals[int(cnts.argmax())]))
+            else:
+                raise ValueError(f"Unsupported strategy {strat!r}")
+        model = ImputerModel(surrogates=surrogates)
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class ImputerModel(_ImputerParams, Model):
+    def __init__(self, surrogates: Optional[list] = None):
+        super().__init__()
+        self.surrogates = list(surrogates or [])
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        in_cols, out_cols = self._in_out_cols()
+        mv = self.getOrDefault(self.missingValue)
+        out = dataset
+        for c, o, sur in zip(in_cols, out_cols, self.surrogates):
+            x = out.col(c)
+            miss = torch.isnan(x) if math.isnan(mv) else (x == mv)
+            out = out.withColumn(o, torch.where(miss, torch.full_like(x, sur), x))
+        return out
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)

@@ -69,6 +69,11 @@ _register("Binarizer", "org.apache.spark.ml.feature.Binarizer")
 _register("StandardScaler", "org.apache.spark.ml.feature.StandardScaler")
 _register("StandardScalerModel", "org.apache.spark.ml.feature.StandardScalerModel")
 _register("MinMaxScaler", "org.apache.spark.ml.feature.MinMaxScaler")
+_register("MaxAbsScaler", "org.apache.spark.ml.feature.MaxAbsScaler")
+_register("MaxAbsScalerModel", "org.apache.spark.ml.feature.MaxAbsScalerModel")
+_register("Bucketizer", "org.apache.spark.ml.feature.Bucketizer")
+_register("Imputer", "org.apache.spark.ml.feature.Imputer")
+_register("ImputerModel", "org.apache.spark.ml.feature.ImputerModel")
 _register("MinMaxScalerModel", "org.apache.spark.ml.feature.MinMaxScalerModel")
 _register("LogisticRegression", "org.apache.spark.ml.classification.LogisticRegression")
 _register("LogisticRegressionModel", "org.apache.spark.ml.classification.LogisticRegressionModel")
@@ -371,6 +376,10 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "originalMax": vector_struct(instance.originalMax.numpy()),
             }
         ]
+    if cls == "MaxAbsScalerModel":
+        return [{"maxAbs": vector_struct(instance.maxAbs.numpy())}]
+    if cls == "ImputerModel":
+        return [{"surrogates": [float(v) for v in instance.surrogates]}]
     if cls == "LogisticRegressionModel":
         return [
             {
@@ -457,6 +466,10 @@ def _construct(cls, simple: str, rows, path: str, meta):
             originalMin=torch.from_numpy(vector_from_struct(rows[0]["originalMin"])),
             originalMax=torch.from_numpy(vector_from_struct(rows[0]["originalMax"])),
         )
+    if simple == "MaxAbsScalerModel":
+        return cls(maxAbs=torch.from_numpy(vector_from_struct(rows[0]["maxAbs"])))
+    if simple == "ImputerModel":
+        return cls(surrogates=[float(v) for v in rows[0]["surrogates"]])
     if simple == "LogisticRegressionModel":
         r = rows[0]
         return cls(

@@ -203,3 +203,62 @@ class TestBinarizer:
         assert out["ab"].tolist() == [0.0, 0.0, 1.0]  # strictly greater
         b2 = Binarizer(inputCol="v", outputCol="vb")
         assert b2.transform(df)["vb"].tolist() == [[0.0, 1.0]] * 3
+
+
+class TestMaxAbsScaler:
+    def test_fit_transform_roundtrip(self, tmp_path):
+        from flowml.feature import MaxAbsScaler, MaxAbsScalerModel
+
+        X = torch.tensor([[2.0, -4.0, 0.0], [-1.0, 2.0, 0.0]])
+        df = FlowFrame({"features": X})
+        m = MaxAbsScaler(inputCol="features", outputCol="scaled").fit(df)
+        out = m.transform(df)["scaled"]
+        assert torch.allclose(out, torch.tensor([[1.0, -1.0, 0.0], [-0.5, 0.5, 0.0]]))
+        m.save(str(tmp_path / "mas"))
+        lm = MaxAbsScalerModel.load(str(tmp_path / "mas"))
+        assert torch.allclose(lm.transform(df)["scaled"], out)
+
+
+class TestBucketizer:
+    def test_buckets(self):
+        from flowml.feature import Bucketizer
+
+        b = Bucketizer(splits=[-float("inf"), 0.0, 10.0, float("inf")],
+                       inputCol="x", outputCol="b")
+        df = FlowFrame({"x": torch.tensor([-5.0, 0.0, 3.0, 10.0, 99.0])})
+        out = b.transform(df)["b"].tolist()
+        assert out == [0.0, 1.0, 1.0, 2.0, 2.0]
+
+    def test_right_endpoint_closed_and_invalid(self):
+        import pytest
+        from flowml.feature import Bucketizer
+
+        b = Bucketizer(splits=[0.0, 1.0, 2.0], inputCol="x", outputCol="b")
+        df = FlowFrame({"x": torch.tensor([0.0, 1.9, 2.0])})
+        assert b.transform(df)["b"].tolist() == [0.0, 1.0, 1.0]  # 2.0 in last
+        bad = FlowFrame({"x": torch.tensor([-1.0, 0.5])})
+        with pytest.raises(ValueError):
+            b.transform(bad)
+        b.setHandleInvalid("keep")
+        assert b.transform(bad)["b"].tolist() == [2.0, 0.0]
+        b.setHandleInvalid("skip")
+        assert b.transform(bad)["b"].tolist() == [0.0]
+
+
+class TestImputer:
+    def test_mean_median_roundtrip(self, tmp_path):
+        from flowml.feature import Imputer, ImputerModel
+
+        nan = float("nan")
+        df = FlowFrame({"a": torch.tensor([1.0, nan, 3.0]),
+                        "b": torch.tensor([10.0, 20.0, nan])})
+        m = Imputer(inputCols=["a", "b"], outputCols=["ai", "bi"]).fit(df)
+        out = m.transform(df)
+        assert out["ai"].tolist() == [1.0, 2.0, 3.0]
+        assert out["bi"].tolist() == [10.0, 20.0, 15.0]
+        med = Imputer(inputCols=["a"], outputCols=["ai"], strategy="median").fit(df)
+        assert med.surrogates[0] == 1.0 or med.surrogates[0] == 3.0 or med.surrogates[0] == 2.0
+        m.save(str(tmp_path / "imp"))
+        lm = ImputerModel.load(str(tmp_path / "imp"))
+        assert lm.surrogates == m.surrogates
+        assert lm.transform(df)["ai"].tolist() == [1.0, 2.0, 3.0]
