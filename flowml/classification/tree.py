@@ -450,7 +450,7 @@ def grow_trees(
     )
 
     min_c = 2 * cfg.min_instances_per_node
-    if is_cls or cfg.max_depth == 0:
+    if cfg.max_depth == 0:
         # root stats per tree: one K2 pass over feature 0 (lut: root t -> t)
         lut0 = torch.arange(T, dtype=torch.int32, device=dev)
         tree_lo0 = torch.arange(T, dtype=torch.int32, device=dev)
@@ -483,10 +483,11 @@ def grow_trees(
         fr_t = np.nonzero(grow_ok)[0].astype(np.int64)  # frontier tree ids
         fr_n = fr_t.copy()  # global arena ids of the roots
     else:
-        # regression: the dedicated root pass is redundant — the level-0
-        # scan computes each root's own (s, n, imp) from its histogram, and
-        # the scan's min_instances check turns too-small roots into leaves.
-        # Saves one full K2 pass per GBT round.
+        # the dedicated root pass is redundant when the trees grow: the
+        # level-0 scan emits each root's own stats/impurity from its
+        # histogram (trailing packed columns), and its min_instances check
+        # turns too-small/pure roots into leaves. Saves one K2 pass per fit
+        # (one per ROUND for GBT).
         arena.bulk_add(np.zeros((T, S)), np.full(T, np.nan), np.zeros(T))
         fr_t = np.arange(T, dtype=np.int64)
         fr_n = fr_t.copy()
@@ -595,6 +596,12 @@ def grow_trees(
                 cr = packed[:, 6]
                 ls = packed[:, 7 : 7 + S]
                 rs = packed[:, 7 + S : 7 + 2 * S]
+                # node-own impurity/stats (trailing columns): covers roots,
+                # and is a bitwise no-op for nodes created by a parent scan
+                arena.impurity[chunk_n] = packed[:, 7 + 2 * S]
+                pstats = packed[:, 8 + 2 * S : 8 + 3 * S]
+                arena.stats[chunk_n] = pstats
+                arena.count[chunk_n] = pstats.sum(axis=1)
             else:
                 # [f, b, gain, sL, nL, sR, nR, imp_self, s_self, n_self]:
                 # child impurities are unknown here (backfilled after growth)

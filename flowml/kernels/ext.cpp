@@ -172,7 +172,7 @@ torch::Tensor split_scan(torch::Tensor hist, int64_t imp_code,
   const int B = hist.size(2);
   const int S = hist.size(3);
   TORCH_CHECK(S <= 8, "split_scan supports up to 8 stats");
-  auto packed = torch::empty({G, 7 + 2 * S}, hist.options());
+  auto packed = torch::empty({G, 8 + 3 * S}, hist.options());
   launch_split_scan(hist.data_ptr<float>(), packed.data_ptr<float>(), G, Fp, B,
                     S, (int)imp_code, (int)min_instances, min_info_gain,
                     cur_stream());

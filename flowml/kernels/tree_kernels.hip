@@ -788,7 +788,11 @@ extern "C" __global__ void flowml_split_scan(
   if (threadIdx.x == 0) {
     double gain = r_gain[0];
     int idx = r_idx[0];
-    float* out = packed + (long long)g * (7 + 2 * S);
+    float* out = packed + (long long)g * (8 + 3 * S);
+    // trailing columns: the node's OWN impurity + stats (lets the grower
+    // skip the dedicated root pass — roots get stats from their own scan)
+    out[7 + 2 * S] = (float)imp_parent;
+    for (int i = 0; i < S; ++i) out[8 + 2 * S + i] = (float)parent[i];
     const bool ok = (idx != 0x7FFFFFFF) && (gain > 0) && (gain >= min_info_gain) &&
                     !isinf(gain) && !isnan(gain);
     if (!ok) {

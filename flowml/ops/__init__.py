@@ -630,9 +630,11 @@ def split_scan_packed(
     min_info_gain: float,
 ) -> torch.Tensor:
     """K3 — best split per node, packed per-node rows
-    [best_f, best_bin, gain, imp_l, imp_r, cnt_l, cnt_r, ls[S], rs[S]]
-    (G×(7+2S) float32). GPU: one workgroup per node, LDS-staged slab, double
-    gain math in the torch reference's exact formula order (bitwise match).
+    [best_f, best_bin, gain, imp_l, imp_r, cnt_l, cnt_r, ls[S], rs[S],
+    imp_self, parent[S]] (G×(8+3S) float32). The trailing node-own
+    impurity/stats let the grower skip a dedicated root-stats pass. GPU: one
+    workgroup per node, LDS-staged slab, double gain math in the torch
+    reference's exact formula order (bitwise match).
     """
     code = {"gini": 0, "entropy": 1, "variance": 2}[impurity]
     if hip_available(hist):
@@ -658,6 +660,7 @@ def split_scan_packed(
     imp_r = torch.where(ok, _impurity_from_stats(right_stats, impurity), z(best_gain))
     cnt_l = torch.where(ok, _counts_of(left_stats, impurity), z(best_gain))
     cnt_r = torch.where(ok, _counts_of(right_stats, impurity), z(best_gain))
+    imp_p = _impurity_from_stats(parent, impurity)
     return torch.cat(
         [
             best_f.to(torch.float64).reshape(G, 1),
@@ -669,6 +672,8 @@ def split_scan_packed(
             cnt_r.reshape(G, 1),
             left_stats,
             right_stats,
+            imp_p.reshape(G, 1),
+            parent,
         ],
         dim=1,
     ).to(torch.float32)
