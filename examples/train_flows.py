@@ -111,10 +111,14 @@ def main():
     p.add_argument("--test-fraction", type=float, default=0.2)
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--rows", type=int, default=20_000, help="synthetic rows if no --data")
+    p.add_argument("--device", default=None, help="cpu | cuda:N (default: auto)")
     args = p.parse_args()
 
     dist.init_from_env()
-    device = RuntimeConfig().resolve_device()
+    cfg = RuntimeConfig()
+    if args.device:
+        cfg.device = args.device
+    device = cfg.resolve_device()
     rank = dist.get_rank()
 
     data = args.data

@@ -677,38 +677,42 @@ def grow_trees(
         _acc("advance", t0)
         level += 1
 
+    tree_of = np.concatenate(tree_of_parts)
+
     if not is_cls:
         # backfill leaf Σt²/impurity: rows sit at their leaves (keep-at-leaf
-        # advance), so one weighted index_add per tree gives every
-        # never-scanned node its Σwt². Values are informational (NodeData
-        # impurity / impurityStats) — split decisions never read them.
+        # advance), so ONE single-feature K2b pass over the never-scanned
+        # nodes yields their per-node Σwt² slots. (A torch index_add was
+        # 1.27 ms/round here — 6M rows funneling into ~60 leaf slots
+        # serializes atomics; the hist kernel privatizes per-block in LDS.)
+        # Values are informational (NodeData impurity / impurityStats) —
+        # split decisions never read them.
         t0 = _tick()
         need = np.isnan(arena.impurity)
         if need.any():
-            # fp32 accumulation on GPU (native atomicAdd; fp64 index_add on
-            # ROCm is a CAS loop that serializes under leaf contention —
-            # measured 1.5 s vs ~ms); fp64 on CPU (deterministic reference)
-            acc_dt = torch.float32 if dev.type == "cuda" else torch.float64
-            s2g = torch.zeros(arena.n_nodes, dtype=acc_dt, device=dev)
-            tf = targets.to(acc_dt)
-            t2 = tf * tf
-            for t in range(T):
-                wv = t2 if bag is None else t2 * bag[t].to(acc_dt)
-                if hash_bag is not None:
-                    bm, cdf, bseed, rofs = hash_bag
-                    import numpy as _np
-
-                    hw = torch.from_numpy(
-                        ops._hash_bag_torch(bm, cdf, bseed,
-                                            _np.arange(rofs, rofs + N), t)
-                    ).to(dev)
-                    wv = wv * hw.to(acc_dt)
-                s2g.index_add_(0, nodeid[t].to(torch.int64), wv)
-            s2g = s2g.to(torch.float64)
-            dist.all_reduce_(s2g)
-            s2_np = s2g.cpu().numpy()
+            need_idx = np.nonzero(need)[0]
+            grp = need_idx[np.argsort(tree_of[need_idx], kind="stable")]
+            Gn = len(grp)
+            lut_np2 = np.full(arena.n_nodes, -1, dtype=np.int32)
+            lut_np2[grp] = np.arange(Gn, dtype=np.int32)
+            cnt_np2 = np.bincount(tree_of[grp], minlength=T).astype(np.int32)
+            lo_np2 = (np.cumsum(cnt_np2) - cnt_np2).astype(np.int32)
+            l_max_n = int(cnt_np2.max())
+            if l_max_n * (B * 2 + 1) * 4 > 128 * 1024:
+                l_max_n = 0  # deep forest: GLOBAL mode
+            hb = ops.tree_hist_nodes(
+                codes, nodeid,
+                torch.from_numpy(lut_np2).to(dev),
+                torch.from_numpy(lo_np2).to(dev),
+                torch.from_numpy(cnt_np2).to(dev),
+                labels_u8, bag, Gn, l_max_n, B, 0,
+                torch.zeros((Gn, 1), dtype=torch.int32, device=dev),
+                targets=targets, hash_bag=hash_bag,
+            )
+            dist.all_reduce_(hb.reshape(-1))
+            ns2b = ops.reg_hist_views(hb, Gn, 1, B)[1].cpu().numpy().astype(np.float64)
             stats = arena.stats
-            stats[need, 1] = s2_np[need]
+            stats[grp, 1] = ns2b
             n_np = np.maximum(stats[:, 2], 1e-12)
             m_np = stats[:, 0] / n_np
             imp_np = np.maximum(stats[:, 1] / n_np - m_np * m_np, 0.0)
@@ -717,7 +721,6 @@ def grow_trees(
 
     # extract per-tree FlatTrees from the arena (once per fit, vectorized)
     t0 = _tick()
-    tree_of = np.concatenate(tree_of_parts)
     order = np.argsort(tree_of, kind="stable")  # per-tree contiguous, BFS order
     bounds = np.searchsorted(tree_of[order], np.arange(T + 1))
     g2l = np.empty(arena.n_nodes, dtype=np.int64)
