@@ -550,8 +550,19 @@ def grow_trees(
             # so this is rank-invariant and reproducible (√F draw, K2 input)
             if use_subset:
                 r = np.random.default_rng((cfg.seed * 1_000_003 + level * 8191 + ci) & 0x7FFFFFFF)
-                keys = r.random((G, F))
-                fs_np = np.sort(np.argpartition(keys, sub_F - 1, axis=1)[:, :sub_F], axis=1).astype(np.int32)
+                # vectorized partial Fisher-Yates: sub_F int draws per node
+                # instead of F random keys + argpartition (the fs draw was
+                # the largest host-side cost of the flagship fit)
+                pool = np.broadcast_to(
+                    np.arange(F, dtype=np.int32), (G, F)
+                ).copy()
+                rows_ar = np.arange(G)
+                for j in range(sub_F):
+                    k = r.integers(j, F, G)
+                    tmp = pool[rows_ar, j].copy()
+                    pool[rows_ar, j] = pool[rows_ar, k]
+                    pool[rows_ar, k] = tmp
+                fs_np = np.sort(pool[:, :sub_F], axis=1)
                 fs = torch.from_numpy(fs_np).to(dev)
             else:
                 fs_np = None
