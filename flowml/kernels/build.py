@@ -18,7 +18,10 @@ _HERE = os.path.dirname(os.path.abspath(__file__))
 
 
 def build(verbose: bool = True) -> str:
-    os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+    asan = os.environ.get("FLOWML_ASAN", "0") == "1"
+    # force (not setdefault): an inherited PYTORCH_ROCM_ARCH=gfx950 would
+    # conflict with the xnack+ objects the device sanitizer needs
+    os.environ["PYTORCH_ROCM_ARCH"] = "gfx950:xnack+" if asan else "gfx950"
     os.environ.setdefault("MAX_JOBS", "8")
     from torch.utils import cpp_extension
 
@@ -31,12 +34,38 @@ def build(verbose: bool = True) -> str:
     ]
     cflags = ["-O3"]
     hipflags = ["-O3", "--offload-arch=gfx950"]
-    if os.environ.get("FLOWML_ASAN", "0") == "1":
-        # device+host address sanitizer build: run the GPU
-        # test suite against this to catch OOB in the atomic kernels
-        cflags += ["-fsanitize=address", "-shared-libasan"]
+    if asan:
+        # DEVICE address-sanitizer build: catches OOB in the
+        # atomic kernels. Device ASAN needs xnack+ code objects
+        # (PYTORCH_ROCM_ARCH=gfx950:xnack+ above); the host TU is left
+        # un-instrumented (g++ lacks clang's -shared-libasan, and the
+        # kernels are the risk surface).
+        hipflags = [f.replace("--offload-arch=gfx950",
+                              "--offload-arch=gfx950:xnack+") for f in hipflags]
         hipflags += ["-fsanitize=address", "-shared-libasan"]
-    mod_path = cpp_extension.load(
+    try:
+        mod_path = _load(cpp_extension, sources, build_dir, cflags, hipflags,
+                         verbose)
+    except OSError as e:
+        if asan:
+            raise RuntimeError(
+                "ASAN build compiled but needs the sanitizer runtime "
+                "preloaded to import. Run under:\n  LD_PRELOAD=$("
+                "/opt/rocm/lib/llvm/bin/clang -print-file-name="
+                "libclang_rt.asan-x86_64.so) ASAN_OPTIONS=detect_leaks=0 "
+                "HSA_XNACK=1 python ..."
+            ) from e
+        raise
+    so_path = os.path.join(build_dir, "_flowml_hip.so")
+    dst = os.path.join(_HERE, "_flowml_hip.so")
+    shutil.copy2(so_path, dst)
+    if verbose:
+        print(f"built {dst}")
+    return mod_path
+
+
+def _load(cpp_extension, sources, build_dir, cflags, hipflags, verbose):
+    return cpp_extension.load(
         name="_flowml_hip",
         sources=sources,
         build_directory=build_dir,
@@ -46,15 +75,8 @@ def build(verbose: bool = True) -> str:
         is_python_module=False,
         is_standalone=False,
     )
-    so_src = os.path.join(build_dir, "_flowml_hip.so")
-    if not os.path.exists(so_src):
-        raise RuntimeError(f"build produced no {so_src}")
-    so_dst = os.path.join(_HERE, "_flowml_hip.so")
-    shutil.copy2(so_src, so_dst)
-    return so_dst
 
 
 if __name__ == "__main__":
-    path = build()
-    print(f"built {path}")
+    build()
     sys.exit(0)
