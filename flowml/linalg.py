@@ -116,8 +116,8 @@ class SparseVector(Vector):
             if self.indices[-1] >= self.size:
                 raise ValueError("Index out of bounds")
 
-    @property
     def numNonzeros(self) -> int:
+        # a METHOD, matching pyspark.ml.linalg.SparseVector.numNonzeros()
         return int(np.count_nonzero(self.values))
 
     def toArray(self) -> np.ndarray:
