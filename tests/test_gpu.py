@@ -611,3 +611,33 @@ class TestSplitScanRegGPU:
         assert torch.all(out[:, 0] == -1)
         expect_imp = 30.0 / 5.0 - 2.0 * 2.0  # s2/n - (s/n)^2
         assert torch.allclose(out[:, 7], torch.full((G,), expect_imp))
+
+
+@pytest.mark.gpu
+class TestFullFitDeterminismGPU:
+    def test_rf_fit_twice_bitwise(self):
+        """Two identical RF fits on the GPU must produce bitwise-identical
+        predictions (integer-count histograms + fixed-order reductions)."""
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(21)
+        X = torch.from_numpy(r.normal(size=(200_000, 20)).astype(np.float32)).to(DEV)
+        y = ((X[:, 0] + X[:, 3] * X[:, 7]) > 0).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y}, device=DEV)
+        est = RandomForestClassifier(numTrees=16, maxDepth=6, seed=9)
+        p1 = est.fit(df).transform(df)["prediction"]
+        p2 = est.fit(df).transform(df)["prediction"]
+        assert torch.equal(p1.cpu(), p2.cpu())
+
+    def test_lr_fit_twice_bitwise(self):
+        from flowml.classification import LogisticRegression
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(22)
+        X = torch.from_numpy(r.normal(size=(300_000, 41)).astype(np.float32)).to(DEV)
+        y = (X[:, 0] > 0.3).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y}, device=DEV)
+        est = LogisticRegression(maxIter=15)
+        m1, m2 = est.fit(df), est.fit(df)
+        assert torch.equal(m1._W, m2._W) and torch.equal(m1._b, m2._b)
