@@ -338,9 +338,10 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16_v2(
   }
 }
 
-extern "C" __global__ void flowml_wgrad_bias_bf16(
-    const __bf16*, const __bf16*, float*, long long, int, int, int, int, int,
-    int, long long);
+template <int KF, int FF>
+__global__ void flowml_wgrad_v3(const __bf16*, const __bf16*, float*,
+                                long long, int, int, int, int, int, int,
+                                long long);
 
 extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
                                        float* out, long long N, int K, int F,
@@ -378,10 +379,18 @@ extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
     return;
   }
   const size_t lds = (size_t)32 * (Kp + Fp) * sizeof(__bf16);
-  hipLaunchKernelGGL(flowml_wgrad_bias_bf16, dim3(grid), dim3(256), lds,
-                     (hipStream_t)stream, (const __bf16*)delta,
-                     (const __bf16*)act, out, N, K, F, Kp, Fp, waves_k,
-                     waves_f, chunk);
+  const int kfrag = (nk16 + waves_k - 1) / waves_k;
+  const int ffrag = (nf16 + waves_f - 1) / waves_f;
+#define WG3_LAUNCH(KF, FF)                                                  \
+  hipLaunchKernelGGL((flowml_wgrad_v3<KF, FF>), dim3(grid), dim3(256), lds, \
+                     (hipStream_t)stream, (const __bf16*)delta,             \
+                     (const __bf16*)act, out, N, K, F, Kp, Fp, waves_k,     \
+                     waves_f, chunk)
+  if (kfrag <= 1 && ffrag <= 8) WG3_LAUNCH(1, 8);
+  else if (kfrag <= 2 && ffrag <= 6) WG3_LAUNCH(2, 6);
+  else if (kfrag <= 4 && ffrag <= 4) WG3_LAUNCH(4, 4);
+  else WG3_LAUNCH(4, 8);
+#undef WG3_LAUNCH
 }
 
 // ---------------------------------------------------------------------------
@@ -469,7 +478,12 @@ __device__ __forceinline__ int wg3_elem(int n, int c) {
   return ((n >> 2) & 1) * 256 + (n >> 3) * 64 + (n & 3) * 16 + c;
 }
 
-extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
+// Templated on the per-wave fragment grid: a static acc[4][8] allocated
+// 252 VGPR + 128 AGPR = 1 wave/SIMD (no latency hiding). Instantiating the
+// exact (KF, FF) needed by each layer shape keeps the accumulators small
+// enough for 2-3 waves/SIMD.
+template <int KF, int FF>
+__global__ __launch_bounds__(256) void flowml_wgrad_v3(
     const __bf16* __restrict__ delta,  // N×K
     const __bf16* __restrict__ act,    // N×F
     float* __restrict__ out,           // K×(F+1) fp32 (zeroed; col F = gB)
@@ -494,11 +508,11 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
   const int kf0 = wk * kfrag;
   const int ff0 = wf * ffrag;
 
-  f32x4 acc[4][8];
+  f32x4 acc[KF][FF];
 #pragma unroll
-  for (int a = 0; a < 4; ++a)
+  for (int a = 0; a < KF; ++a)
 #pragma unroll
-    for (int b = 0; b < 8; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+    for (int b = 0; b < FF; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
 
   const long long n_lo = blockIdx.x * chunk;
   const long long n_hi = min(n_lo + chunk, N);
@@ -554,9 +568,9 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
     __syncthreads();
 
     // hoisted B fragments: 2 tr reads each, reused across the a-loop
-    bf16x8 bfr[8];
+    bf16x8 bfr[FF];
 #pragma unroll
-    for (int b = 0; b < 8; ++b) {
+    for (int b = 0; b < FF; ++b) {
       if (b >= ffrag || ff0 + b >= nf16) continue;
       const int base = (ff0 + b) * 128;  // tile offset in bf16x4 units
       bf16x4_t u0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(at3 + base + lane);
@@ -567,7 +581,7 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
       bfr[b] = f8;
     }
 #pragma unroll
-    for (int a = 0; a < 4; ++a) {
+    for (int a = 0; a < KF; ++a) {
       if (a >= kfrag || kf0 + a >= nk16) continue;
       const int base = (kf0 + a) * 128;
       bf16x4_t v0 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(dt3 + base + lane);
@@ -576,7 +590,7 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
 #pragma unroll
       for (int j = 0; j < 4; ++j) { af[j] = v0[j]; af[4 + j] = v1[j]; }
 #pragma unroll
-      for (int b = 0; b < 8; ++b) {
+      for (int b = 0; b < FF; ++b) {
         if (b >= ffrag || ff0 + b >= nf16) continue;
         acc[a][b] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bfr[b], acc[a][b], 0, 0, 0);
       }
@@ -586,10 +600,10 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16(
   const int col = lane & 15;
   const int row_off = (lane >> 4) * 4;
 #pragma unroll
-  for (int a = 0; a < 4; ++a) {
+  for (int a = 0; a < KF; ++a) {
     if (a >= kfrag || kf0 + a >= nk16) continue;
 #pragma unroll
-    for (int b = 0; b < 8; ++b) {
+    for (int b = 0; b < FF; ++b) {
       if (b >= ffrag || ff0 + b >= nf16) continue;
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
