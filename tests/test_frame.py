@@ -130,3 +130,24 @@ def test_parquet_round_trip(tmp_path):
     assert torch.allclose(back["v"], df["v"])
     sel = read_parquet(p, columns=["x"])
     assert sel.columns == ["x"]
+
+
+class TestCsvMissingValues:
+    def test_na_to_nan_and_imputer_pipeline(self, tmp_path):
+        """Empty CSV fields load as NaN (numeric) / '' (string) and flow
+        through Imputer + StringIndexer(handleInvalid) like in Spark."""
+        import math
+
+        import flowml
+        from flowml.feature import Imputer, StringIndexer
+
+        p = tmp_path / "f.csv"
+        p.write_text("a,b,proto\n1.0,,tcp\n,2.5,udp\n3.0,4.5,tcp\n")
+        df = flowml.read_csv(str(p))
+        assert math.isnan(df.col("a")[1].item())
+        assert math.isnan(df.col("b")[0].item())
+        out = Imputer(inputCols=["a", "b"], outputCols=["a2", "b2"]).fit(df).transform(df)
+        assert out.col("a2").tolist() == [1.0, 2.0, 3.0]
+        assert out.col("b2").tolist() == [3.5, 2.5, 4.5]
+        m = StringIndexer(inputCol="proto", outputCol="p").fit(out)
+        assert m.transform(out).col("p").tolist() == [0.0, 1.0, 0.0]
