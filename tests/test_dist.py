@@ -36,7 +36,12 @@ def _worker(rank, world, port, fn_name, out_q):
 def _run_dist(fn_name, world=2):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = int(os.environ.get("FLOWML_TEST_PORT", "29611")) + abs(hash(fn_name)) % 500
+    # stable per-test spread + per-process offset so parallel pytest-xdist
+    # workers cannot collide on a rendezvous port
+    import hashlib
+
+    h = int(hashlib.md5(fn_name.encode()).hexdigest()[:4], 16) % 500
+    port = int(os.environ.get("FLOWML_TEST_PORT", "29611")) + h + (os.getpid() % 97) * 5
     procs = [
         ctx.Process(target=_worker, args=(r, world, port, fn_name, q)) for r in range(world)
     ]
