@@ -46,7 +46,8 @@ void launch_gemm_bias_act_f32(const float*, const float*, const float*, float*,
                               long long, int, int, int, void*);
 void launch_act_backward_bf16(const void*, void*, long long, int, void*);
 void launch_wgrad_bias_bf16(const void*, const void*, float*, long long, int,
-                            int, void*);
+                            int, int, long long, int, void*);
+void wgrad_bias_plan(long long, int*, long long*, int*);
 void launch_wgrad_f32_splitk(const float*, const float*, float*, long long,
                              int, int, int, long long, void*);
 }
@@ -308,9 +309,24 @@ std::tuple<torch::Tensor, torch::Tensor> wgrad_bias(torch::Tensor delta,
   const long long N = delta.size(0);
   const int K = delta.size(1);
   const int F = act.size(1);
-  auto out = torch::zeros({K, F + 1}, delta.options().dtype(torch::kFloat32));
-  launch_wgrad_bias_bf16(delta.data_ptr(), act.data_ptr(),
-                         out.data_ptr<float>(), N, K, F, cur_stream());
+  int grid = 0, version = 4;
+  long long chunk = 0;
+  wgrad_bias_plan(N, &grid, &chunk, &version);
+  torch::Tensor out;
+  if (version >= 4) {
+    // exact per-block partial slabs, summed in fixed order -> deterministic
+    auto part = torch::zeros({grid, (int64_t)K * (F + 1)},
+                             delta.options().dtype(torch::kFloat32));
+    launch_wgrad_bias_bf16(delta.data_ptr(), act.data_ptr(),
+                           part.data_ptr<float>(), N, K, F, grid, chunk,
+                           version, cur_stream());
+    out = part.sum(0).reshape({K, F + 1});
+  } else {
+    out = torch::zeros({K, F + 1}, delta.options().dtype(torch::kFloat32));
+    launch_wgrad_bias_bf16(delta.data_ptr(), act.data_ptr(),
+                           out.data_ptr<float>(), N, K, F, grid, chunk,
+                           version, cur_stream());
+  }
   auto gW = out.narrow(1, 0, F).contiguous();
   auto gB = out.select(1, F).contiguous();
   return {gW, gB};

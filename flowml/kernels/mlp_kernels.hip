@@ -338,13 +338,26 @@ extern "C" __global__ __launch_bounds__(256) void flowml_wgrad_bias_bf16_v2(
   }
 }
 
-template <int KF, int FF>
+template <int KF, int FF, bool PARTIAL>
 __global__ void flowml_wgrad_v3(const __bf16*, const __bf16*, float*,
                                 long long, int, int, int, int, int, int,
                                 long long);
 
+// exported so the binding can size the partials buffer before launching
+extern "C" void wgrad_bias_plan(long long N, int* grid_out,
+                                long long* chunk_out, int* version_out) {
+  static const int blocks = flowml_env_int_mlp("FLOWML_WGRAD_BLOCKS", 1536);
+  static const int version = flowml_env_int_mlp("FLOWML_WGRAD_V", 4);
+  long long chunk = (N + blocks - 1) / blocks;
+  chunk = (chunk + 31) / 32 * 32;
+  *chunk_out = chunk;
+  *grid_out = (int)((N + chunk - 1) / chunk);
+  *version_out = version;
+}
+
 extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
                                        float* out, long long N, int K, int F,
+                                       int grid, long long chunk, int version,
                                        void* stream) {
   auto pad16 = [](int v) { return (v + 15) / 16 * 16; };
   const int Kp = pad16(K);
@@ -364,11 +377,6 @@ extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
       waves_f = 1;
     }
   }
-  static const int blocks = flowml_env_int_mlp("FLOWML_WGRAD_BLOCKS", 1536);
-  static const int version = flowml_env_int_mlp("FLOWML_WGRAD_V", 3);
-  long long chunk = (N + blocks - 1) / blocks;
-  chunk = (chunk + 31) / 32 * 32;
-  const int grid = (int)((N + chunk - 1) / chunk);
   if (version == 2) {
     const size_t lds =
         (size_t)32 * (Kp + Fp + 2 * WG_PAD) * sizeof(__bf16);
@@ -381,15 +389,18 @@ extern "C" void launch_wgrad_bias_bf16(const void* delta, const void* act,
   const size_t lds = (size_t)32 * (Kp + Fp) * sizeof(__bf16);
   const int kfrag = (nk16 + waves_k - 1) / waves_k;
   const int ffrag = (nf16 + waves_f - 1) / waves_f;
-#define WG3_LAUNCH(KF, FF)                                                  \
-  hipLaunchKernelGGL((flowml_wgrad_v3<KF, FF>), dim3(grid), dim3(256), lds, \
-                     (hipStream_t)stream, (const __bf16*)delta,             \
-                     (const __bf16*)act, out, N, K, F, Kp, Fp, waves_k,     \
+#define WG3_LAUNCH(KF, FF, P)                                                  \
+  hipLaunchKernelGGL((flowml_wgrad_v3<KF, FF, P>), dim3(grid), dim3(256), lds, \
+                     (hipStream_t)stream, (const __bf16*)delta,                \
+                     (const __bf16*)act, out, N, K, F, Kp, Fp, waves_k,        \
                      waves_f, chunk)
-  if (kfrag <= 1 && ffrag <= 8) WG3_LAUNCH(1, 8);
-  else if (kfrag <= 2 && ffrag <= 6) WG3_LAUNCH(2, 6);
-  else if (kfrag <= 4 && ffrag <= 4) WG3_LAUNCH(4, 4);
-  else WG3_LAUNCH(4, 8);
+#define WG3_PICK(P)                                        \
+  if (kfrag <= 1 && ffrag <= 8) WG3_LAUNCH(1, 8, P);       \
+  else if (kfrag <= 2 && ffrag <= 6) WG3_LAUNCH(2, 6, P);  \
+  else if (kfrag <= 4 && ffrag <= 4) WG3_LAUNCH(4, 4, P);  \
+  else WG3_LAUNCH(4, 8, P)
+  if (version >= 4) { WG3_PICK(true); } else { WG3_PICK(false); }
+#undef WG3_PICK
 #undef WG3_LAUNCH
 }
 
@@ -482,11 +493,13 @@ __device__ __forceinline__ int wg3_elem(int n, int c) {
 // 252 VGPR + 128 AGPR = 1 wave/SIMD (no latency hiding). Instantiating the
 // exact (KF, FF) needed by each layer shape keeps the accumulators small
 // enough for 2-3 waves/SIMD.
-template <int KF, int FF>
+template <int KF, int FF, bool PARTIAL>
 __global__ __launch_bounds__(256) void flowml_wgrad_v3(
     const __bf16* __restrict__ delta,  // N×K
     const __bf16* __restrict__ act,    // N×F
-    float* __restrict__ out,           // K×(F+1) fp32 (zeroed; col F = gB)
+    float* __restrict__ out,  // PARTIAL: grid×K×(F+1) exact per-block slabs
+                              // (zeroed; summed host-side in fixed order ⇒
+                              // DETERMINISTIC); else K×(F+1) atomic target
     long long N, int K, int F, int Kp, int Fp,
     int waves_k, int waves_f, long long chunk) {
   extern __shared__ __bf16 s_t[];  // dt tiles [nk16][512] then at [nf16][512]
@@ -599,6 +612,7 @@ __global__ __launch_bounds__(256) void flowml_wgrad_v3(
 
   const int col = lane & 15;
   const int row_off = (lane >> 4) * 4;
+  float* dst = PARTIAL ? out + (long long)blockIdx.x * K * (F + 1) : out;
 #pragma unroll
   for (int a = 0; a < KF; ++a) {
     if (a >= kfrag || kf0 + a >= nk16) continue;
@@ -610,7 +624,11 @@ __global__ __launch_bounds__(256) void flowml_wgrad_v3(
         const int k = (kf0 + a) * 16 + row_off + reg;
         const int f = (ff0 + b) * 16 + col;
         if (k < K && f <= F) {
-          atomicAdd(&out[(long long)k * (F + 1) + f], acc[a][b][reg]);
+          if (PARTIAL) {
+            dst[(long long)k * (F + 1) + f] = acc[a][b][reg];
+          } else {
+            atomicAdd(&dst[(long long)k * (F + 1) + f], acc[a][b][reg]);
+          }
         }
       }
     }

@@ -641,3 +641,16 @@ class TestFullFitDeterminismGPU:
         est = LogisticRegression(maxIter=15)
         m1, m2 = est.fit(df), est.fit(df)
         assert torch.equal(m1._W, m2._W) and torch.equal(m1._b, m2._b)
+
+    def test_mlp_grad_deterministic(self):
+        """wgrad v4 (exact per-block partials, fixed-order sum): two
+        identical bf16 backward passes must match bitwise."""
+        from flowml import ops as _ops
+
+        r = np.random.default_rng(13)
+        N, K, F = 300_000, 128, 41
+        delta = torch.from_numpy(r.normal(size=(N, K)).astype(np.float32)).bfloat16().to(DEV)
+        act = torch.from_numpy(r.normal(size=(N, F)).astype(np.float32)).bfloat16().to(DEV)
+        a = _ops.get_ext().wgrad_bias(delta, act)
+        b = _ops.get_ext().wgrad_bias(delta, act)
+        assert torch.equal(a[0], b[0]) and torch.equal(a[1], b[1])
