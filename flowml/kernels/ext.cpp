@@ -36,6 +36,7 @@ void launch_logreg_delta(const float*, const int*, const float*, const float*,
                          void*);
 void launch_softmax_xent(const float*, const int*, const float*, float*,
                          float*, long long, int, void*);
+int softmax_xent_grid(long long);
 void launch_split_scan(const float*, float*, int, int, int, int, int, int,
                        double, void*);
 void launch_split_scan_reg(const float*, float*, int, int, int, int, double,
@@ -247,12 +248,13 @@ std::tuple<torch::Tensor, torch::Tensor> softmax_xent_grad(
   const int K = logits.size(1);
   TORCH_CHECK(K <= 16, "softmax_xent supports up to 16 classes");
   auto delta = torch::empty_like(logits);
-  auto loss = torch::zeros({1}, logits.options());
+  // per-block loss partials summed in fixed order -> deterministic
+  auto loss = torch::zeros({softmax_xent_grid(N)}, logits.options());
   launch_softmax_xent(logits.data_ptr<float>(), y.data_ptr<int>(),
                       w.numel() ? w.data_ptr<float>() : nullptr,
                       delta.data_ptr<float>(), loss.data_ptr<float>(), N, K,
                       cur_stream());
-  return {loss.squeeze(0), delta};
+  return {loss.sum(), delta};
 }
 
 static torch::Tensor pad_k(torch::Tensor t, int64_t mult) {

@@ -472,7 +472,8 @@ extern "C" __global__ void flowml_logreg_delta(
     if (threadIdx.x < s) s_red[threadIdx.x] += s_red[threadIdx.x + s];
     __syncthreads();
   }
-  if (threadIdx.x == 0) atomicAdd(loss, s_red[0]);
+  // per-block partial (fixed-order host sum ⇒ deterministic loss)
+  if (threadIdx.x == 0) loss[blockIdx.x] = s_red[0];
 }
 
 // ---------------------------------------------------------------------------
@@ -513,7 +514,8 @@ extern "C" __global__ void flowml_softmax_xent(
     if (threadIdx.x < s) s_red[threadIdx.x] += s_red[threadIdx.x + s];
     __syncthreads();
   }
-  if (threadIdx.x == 0) atomicAdd(loss, s_red[0]);
+  // per-block partial (fixed-order host sum ⇒ deterministic loss)
+  if (threadIdx.x == 0) loss[blockIdx.x] = s_red[0];
 }
 
 // ---------------------------------------------------------------------------
@@ -646,11 +648,15 @@ extern "C" void launch_logreg_delta(const float* X, const int* y,
                      delta, loss, N, F, K);
 }
 
+extern "C" int softmax_xent_grid(long long N) {
+  return grid_1d(N, 256, 2048);
+}
+
 extern "C" void launch_softmax_xent(const float* logits, const int* y,
                                     const float* w, float* delta, float* loss,
                                     long long N, int K, void* stream) {
   const int block = 256;
-  hipLaunchKernelGGL(flowml_softmax_xent, dim3(grid_1d(N, block, 2048)),
+  hipLaunchKernelGGL(flowml_softmax_xent, dim3(softmax_xent_grid(N)),
                      dim3(block), 0, (hipStream_t)stream, logits, y, w, delta,
                      loss, N, K);
 }
