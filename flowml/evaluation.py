@@ -275,3 +275,52 @@ class BinaryClassificationEvaluator(HasLabelCol, HasRawPredictionCol, HasWeightC
         prec = torch.cat([torch.ones(1, dtype=torch.float64), prec])
         rec = torch.cat([torch.zeros(1, dtype=torch.float64), rec])
         return float(torch.trapz(prec, rec))
+
+
+class RegressionEvaluator(HasLabelCol, HasPredictionCol, HasWeightCol, Params):
+    """Spec: spark:.../ml/evaluation/RegressionEvaluator.scala —
+    metricName ∈ rmse (default) | mse | mae | r2 | var."""
+
+    metricName: Param[str] = Param(
+        "undefined", "metricName", "metric name in evaluation (rmse|mse|r2|mae|var)",
+        TypeConverters.toString,
+    )
+
+    def __init__(self, *, predictionCol: Optional[str] = None,
+                 labelCol: Optional[str] = None, metricName: Optional[str] = None,
+                 weightCol: Optional[str] = None):
+        super().__init__()
+        self._setDefault(predictionCol="prediction", labelCol="label",
+                         metricName="rmse")
+        self._set(predictionCol=predictionCol, labelCol=labelCol,
+                  metricName=metricName, weightCol=weightCol)
+
+    def isLargerBetter(self) -> bool:
+        return self.getOrDefault(self.metricName) in ("r2", "var")
+
+    def evaluate(self, dataset: FlowFrame) -> float:
+        name = self.getOrDefault(self.metricName)
+        if name not in ("rmse", "mse", "mae", "r2", "var"):
+            raise ValueError(f"Unsupported metricName {name!r}")
+        pred = dataset.col(self.getOrDefault(self.predictionCol)).reshape(-1).to(torch.float64)
+        lab = dataset.col(self.getOrDefault(self.labelCol)).reshape(-1).to(torch.float64)
+        if self.isSet(self.weightCol):
+            w = dataset.col(self.getWeightCol()).reshape(-1).to(torch.float64)
+        else:
+            w = torch.ones_like(lab)
+        wsum = w.sum()
+        err = lab - pred
+        mse = float((w * err * err).sum() / wsum)
+        if name == "mse":
+            return mse
+        if name == "rmse":
+            return float(mse ** 0.5)
+        if name == "mae":
+            return float((w * err.abs()).sum() / wsum)
+        mean_l = (w * lab).sum() / wsum
+        if name == "var":  # explained variance
+            mean_p = (w * pred).sum() / wsum
+            return float((w * (pred - mean_p) ** 2).sum() / wsum)
+        ss_tot = float((w * (lab - mean_l) ** 2).sum())
+        ss_res = float((w * err * err).sum())
+        return 1.0 - ss_res / ss_tot if ss_tot > 0 else 0.0

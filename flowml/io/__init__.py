@@ -73,6 +73,12 @@ _register("MaxAbsScaler", "org.apache.spark.ml.feature.MaxAbsScaler")
 _register("MaxAbsScalerModel", "org.apache.spark.ml.feature.MaxAbsScalerModel")
 _register("Bucketizer", "org.apache.spark.ml.feature.Bucketizer")
 _register("Imputer", "org.apache.spark.ml.feature.Imputer")
+_register("RandomForestRegressor", "org.apache.spark.ml.regression.RandomForestRegressor")
+_register("RandomForestRegressionModel", "org.apache.spark.ml.regression.RandomForestRegressionModel")
+_register("DecisionTreeRegressor", "org.apache.spark.ml.regression.DecisionTreeRegressor")
+_register("DecisionTreeRegressionModel", "org.apache.spark.ml.regression.DecisionTreeRegressionModel")
+_register("GBTRegressor", "org.apache.spark.ml.regression.GBTRegressor")
+_register("GBTRegressionModel", "org.apache.spark.ml.regression.GBTRegressionModel")
 _register("ImputerModel", "org.apache.spark.ml.feature.ImputerModel")
 _register("MinMaxScalerModel", "org.apache.spark.ml.feature.MinMaxScalerModel")
 _register("LogisticRegression", "org.apache.spark.ml.classification.LogisticRegression")
@@ -102,9 +108,10 @@ def _our_class(spark_name: str):
     import flowml.feature as feature
     import flowml.pipeline as pipeline
     import flowml.classification as cls
+    import flowml.regression as reg
 
     simple = spark_name.rsplit(".", 1)[-1]
-    for mod in (pipeline, feature, cls):
+    for mod in (pipeline, feature, cls, reg):
         if hasattr(mod, simple):
             return getattr(mod, simple)
     raise ValueError(f"Cannot resolve class for {spark_name!r}")
@@ -392,14 +399,16 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
         ]
     if cls == "MultilayerPerceptronClassificationModel":
         return [{"weights": vector_struct(instance._weights.to(torch.float64).numpy())}]
-    if cls in ("RandomForestClassificationModel", "GBTClassificationModel"):
+    if cls in ("RandomForestClassificationModel", "GBTClassificationModel",
+               "RandomForestRegressionModel", "GBTRegressionModel",
+               "DecisionTreeRegressionModel"):
         rows: List[Dict[str, Any]] = []
         for tid, tree in enumerate(instance._trees):
             rows.extend(_tree_node_rows(tree, tid))
         # treesMetadata: per-tree {treeID, metadata(json), weights}
         tw = (
             instance.treeWeights
-            if cls == "GBTClassificationModel"
+            if cls in ("GBTClassificationModel", "GBTRegressionModel")
             else [1.0] * len(instance._trees)
         )
         tmeta = [
@@ -501,6 +510,19 @@ def _construct(cls, simple: str, rows, path: str, meta):
         trees = _trees_from_node_rows(rows)
         t = trees[0]
         return cls(tree=t, numClasses=len(t.stats[0]), numFeatures=0)
+    if simple == "RandomForestRegressionModel":
+        trees = _trees_from_node_rows(rows)
+        return cls(trees=trees, numFeatures=_num_features_from_trees_meta(path, trees))
+    if simple == "GBTRegressionModel":
+        trees = _trees_from_node_rows(rows)
+        tm = _read_parquet(os.path.join(path, "treesMetadata")).to_pylist()
+        weights = [r["weights"] for r in sorted(tm, key=lambda r: r["treeID"])]
+        return cls(trees=trees, treeWeights=weights,
+                   numFeatures=_num_features_from_trees_meta(path, trees))
+    if simple == "DecisionTreeRegressionModel":
+        trees = _trees_from_node_rows(rows)
+        return cls(tree=trees[0],
+                   numFeatures=_num_features_from_trees_meta(path, trees))
     # parameter-only stages
     return cls()
 

@@ -1,0 +1,104 @@
+"""Regression tree models + RegressionEvaluator."""
+
+import numpy as np
+import pytest
+import torch
+
+from flowml.evaluation import RegressionEvaluator
+from flowml.frame import FlowFrame
+from flowml.regression import (
+    DecisionTreeRegressor,
+    DecisionTreeRegressionModel,
+    GBTRegressor,
+    GBTRegressionModel,
+    RandomForestRegressor,
+    RandomForestRegressionModel,
+)
+
+
+def _reg_df(n=2000, seed=0):
+    r = np.random.default_rng(seed)
+    X = torch.from_numpy(r.normal(size=(n, 6)).astype(np.float32))
+    y = (2.0 * X[:, 0] - X[:, 1] * X[:, 2] + 0.1 * torch.from_numpy(
+        r.normal(size=n).astype(np.float32))).to(torch.float64)
+    return FlowFrame({"features": X, "label": y}), X, y
+
+
+class TestRegressionEvaluator:
+    def test_metrics_vs_numpy(self):
+        r = np.random.default_rng(3)
+        lab = r.normal(size=200)
+        pred = lab + r.normal(size=200) * 0.3
+        df = FlowFrame({"label": torch.from_numpy(lab), "prediction": torch.from_numpy(pred)})
+        err = lab - pred
+        assert RegressionEvaluator(metricName="mse").evaluate(df) == pytest.approx((err ** 2).mean())
+        assert RegressionEvaluator(metricName="rmse").evaluate(df) == pytest.approx(np.sqrt((err ** 2).mean()))
+        assert RegressionEvaluator(metricName="mae").evaluate(df) == pytest.approx(np.abs(err).mean())
+        r2 = 1 - (err ** 2).sum() / ((lab - lab.mean()) ** 2).sum()
+        assert RegressionEvaluator(metricName="r2").evaluate(df) == pytest.approx(r2)
+        assert RegressionEvaluator(metricName="rmse").isLargerBetter() is False
+        assert RegressionEvaluator(metricName="r2").isLargerBetter() is True
+
+
+class TestDecisionTreeRegressor:
+    def test_fit_quality_vs_sklearn(self):
+        from sklearn.tree import DecisionTreeRegressor as SkDT
+
+        df, X, y = _reg_df()
+        m = DecisionTreeRegressor(maxDepth=5, seed=0).fit(df)
+        r2 = RegressionEvaluator(metricName="r2").evaluate(m.transform(df))
+        sk = SkDT(max_depth=5).fit(X.numpy(), y.numpy())
+        sk_r2 = sk.score(X.numpy(), y.numpy())
+        assert r2 > sk_r2 - 0.1  # binned splits vs exact: close
+        assert m.numNodes >= 3 and m.depth <= 5
+
+    def test_leaf_value_is_mean(self):
+        # single split on one feature: leaves predict the group means
+        X = torch.tensor([[0.0], [0.0], [10.0], [10.0]])
+        y = torch.tensor([1.0, 3.0, 10.0, 14.0], dtype=torch.float64)
+        df = FlowFrame({"features": X, "label": y})
+        m = DecisionTreeRegressor(maxDepth=1, maxBins=4).fit(df)
+        p = m.transform(df)["prediction"].tolist()
+        assert p[0] == p[1] == pytest.approx(2.0)
+        assert p[2] == p[3] == pytest.approx(12.0)
+
+
+class TestRandomForestRegressor:
+    def test_fit_and_importances(self):
+        df, X, y = _reg_df()
+        m = RandomForestRegressor(numTrees=20, maxDepth=5, seed=1).fit(df)
+        r2 = RegressionEvaluator(metricName="r2").evaluate(m.transform(df))
+        assert r2 > 0.5
+        imp = m.featureImportances.toArray()
+        assert imp.shape == (6,) and imp[0] == imp.max()  # x0 dominates
+
+    def test_persistence_roundtrip(self, tmp_path):
+        df, _, _ = _reg_df(500)
+        m = RandomForestRegressor(numTrees=4, maxDepth=3, seed=2).fit(df)
+        m.save(str(tmp_path / "rfr"))
+        lm = RandomForestRegressionModel.load(str(tmp_path / "rfr"))
+        assert torch.allclose(lm.transform(df)["prediction"], m.transform(df)["prediction"])
+
+
+class TestGBTRegressor:
+    def test_monotone_training_rmse(self):
+        df, _, _ = _reg_df()
+        e = RegressionEvaluator(metricName="rmse")
+        r5 = e.evaluate(GBTRegressor(maxIter=5, maxDepth=3, seed=0).fit(df).transform(df))
+        r30 = e.evaluate(GBTRegressor(maxIter=30, maxDepth=3, seed=0).fit(df).transform(df))
+        assert r30 < r5
+
+    def test_beats_single_tree(self):
+        df, _, _ = _reg_df()
+        e = RegressionEvaluator(metricName="r2")
+        gbt = e.evaluate(GBTRegressor(maxIter=30, maxDepth=3, seed=0).fit(df).transform(df))
+        dt = e.evaluate(DecisionTreeRegressor(maxDepth=3, seed=0).fit(df).transform(df))
+        assert gbt > dt
+
+    def test_persistence_roundtrip(self, tmp_path):
+        df, _, _ = _reg_df(500)
+        m = GBTRegressor(maxIter=4, maxDepth=2, seed=1).fit(df)
+        m.save(str(tmp_path / "gbtr"))
+        lm = GBTRegressionModel.load(str(tmp_path / "gbtr"))
+        assert lm.treeWeights == pytest.approx(m.treeWeights)
+        assert torch.allclose(lm.transform(df)["prediction"], m.transform(df)["prediction"])
