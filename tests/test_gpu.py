@@ -654,3 +654,22 @@ class TestFullFitDeterminismGPU:
         a = _ops.get_ext().wgrad_bias(delta, act)
         b = _ops.get_ext().wgrad_bias(delta, act)
         assert torch.equal(a[0], b[0]) and torch.equal(a[1], b[1])
+
+
+@pytest.mark.gpu
+class TestRegressionGPU:
+    def test_rf_regressor_gpu_matches_cpu(self):
+        from flowml.evaluation import RegressionEvaluator
+        from flowml.regression import GBTRegressor, RandomForestRegressor
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(6)
+        X = torch.from_numpy(r.normal(size=(100_000, 8)).astype(np.float32))
+        y = (2.0 * X[:, 0] - X[:, 1] * X[:, 2]).to(torch.float64)
+        df_c = FlowFrame({"features": X, "label": y})
+        df_g = df_c.to(DEV)
+        e = RegressionEvaluator(metricName="r2")
+        m_g = RandomForestRegressor(numTrees=10, maxDepth=5, seed=3).fit(df_g)
+        assert e.evaluate(m_g.transform(df_g)) > 0.5
+        g = GBTRegressor(maxIter=10, maxDepth=4, seed=3).fit(df_g)
+        assert e.evaluate(g.transform(df_g)) > 0.7
