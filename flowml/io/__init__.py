@@ -79,6 +79,8 @@ _register("DecisionTreeRegressor", "org.apache.spark.ml.regression.DecisionTreeR
 _register("DecisionTreeRegressionModel", "org.apache.spark.ml.regression.DecisionTreeRegressionModel")
 _register("GBTRegressor", "org.apache.spark.ml.regression.GBTRegressor")
 _register("GBTRegressionModel", "org.apache.spark.ml.regression.GBTRegressionModel")
+_register("LinearRegression", "org.apache.spark.ml.regression.LinearRegression")
+_register("LinearRegressionModel", "org.apache.spark.ml.regression.LinearRegressionModel")
 _register("ImputerModel", "org.apache.spark.ml.feature.ImputerModel")
 _register("MinMaxScalerModel", "org.apache.spark.ml.feature.MinMaxScalerModel")
 _register("LogisticRegression", "org.apache.spark.ml.classification.LogisticRegression")
@@ -397,6 +399,14 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "isMultinomial": bool(instance.isMultinomial),
             }
         ]
+    if cls == "LinearRegressionModel":
+        return [
+            {
+                "intercept": float(instance.intercept),
+                "coefficients": vector_struct(instance._coef.to(torch.float64).numpy()),
+                "scale": float(instance.scale),
+            }
+        ]
     if cls == "MultilayerPerceptronClassificationModel":
         return [{"weights": vector_struct(instance._weights.to(torch.float64).numpy())}]
     if cls in ("RandomForestClassificationModel", "GBTClassificationModel",
@@ -488,6 +498,11 @@ def _construct(cls, simple: str, rows, path: str, meta):
             numFeatures=int(r["numFeatures"]),
             isMultinomial=bool(r["isMultinomial"]),
         )
+    if simple == "LinearRegressionModel":
+        r0 = rows[0]
+        coef = torch.from_numpy(vector_from_struct(r0["coefficients"])).to(torch.float32)
+        return cls(coefficients=coef, intercept=float(r0["intercept"]),
+                   numFeatures=int(coef.numel()))
     if simple == "MultilayerPerceptronClassificationModel":
         layers = meta.get("paramMap", {}).get("layers") or meta.get(
             "defaultParamMap", {}

@@ -34,6 +34,12 @@ from flowml.frame import FlowFrame
 from flowml.param import HasMaxIter, Param, TypeConverters
 
 
+def np_asarray_f32(v):
+    import numpy as _np
+
+    return _np.asarray(v, dtype=_np.float32)
+
+
 class _RegressorIOMixin:
     def write(self):
         from flowml.io import GenericWriter
@@ -276,3 +282,223 @@ class GBTRegressionModel(_TreeEnsembleParams, RegressionModel, _RegressorIOMixin
 
     def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
         return self._ensemble.predict_raw(features)[:, 0]
+
+
+# ---------------------------------------------------------------------------
+# LinearRegression (spark:.../ml/regression/LinearRegression.scala)
+# ---------------------------------------------------------------------------
+
+
+from flowml.param import HasTol, HasWeightCol  # noqa: E402
+
+
+class _LinearRegressionParams(_PredictorParams, HasMaxIter, HasTol, HasWeightCol):
+    regParam: Param[float] = Param(
+        "undefined", "regParam", "regularization parameter (>= 0).",
+        TypeConverters.toFloat,
+    )
+    elasticNetParam: Param[float] = Param(
+        "undefined", "elasticNetParam",
+        "the ElasticNet mixing parameter, in range [0, 1] (0 = L2, 1 = L1).",
+        TypeConverters.toFloat,
+    )
+    fitIntercept: Param[bool] = Param(
+        "undefined", "fitIntercept", "whether to fit an intercept term.",
+        TypeConverters.toBoolean,
+    )
+    standardization: Param[bool] = Param(
+        "undefined", "standardization",
+        "whether to standardize the training features before fitting.",
+        TypeConverters.toBoolean,
+    )
+
+
+class LinearRegression(_LinearRegressionParams, Estimator, _RegressorIOMixin):
+    """Least-squares linear regression with elastic-net regularization,
+    solved by LBFGS / OWL-QN in standardized-feature space (the same
+    optimizer stack as LogisticRegression; gradients all-reduce across
+    ranks per iteration — SURVEY.md §2.5 C1 pattern).
+
+    Documented divergence: Spark additionally scales the objective by the
+    label std (so regParam is relative to a unit-variance label); here the
+    objective is ½·Σw(ŷ−y)²/ΣW — identical solutions at regParam=0, and
+    regParam differs by a yStd factor otherwise.
+    """
+
+    def __init__(self, **kwargs):
+        super().__init__()
+        self._setDefault(maxIter=100, regParam=0.0, elasticNetParam=0.0,
+                         tol=1e-6, fitIntercept=True, standardization=True)
+        self._set(**kwargs)
+
+    def _fit(self, dataset: FlowFrame) -> "LinearRegressionModel":
+        from flowml.optim import lbfgs, owlqn
+
+        X = _features_tensor(dataset, self.getFeaturesCol()).to(torch.float32)
+        y = _label_tensor(dataset, self.getLabelCol()).to(torch.float32)
+        dev = X.device
+        N, F = X.shape
+        w = None
+        if self.isSet(self.weightCol):
+            w = dataset.col(self.getWeightCol()).reshape(-1).to(torch.float32)
+        wcol = w if w is not None else torch.ones(N, device=dev)
+
+        stats = torch.zeros(2 + 3 * F, dtype=torch.float64, device=dev)
+        w64 = wcol.to(torch.float64)
+        stats[0] = w64.sum()
+        stats[1] = (w64 * y.to(torch.float64)).sum()
+        X64 = X.to(torch.float64)
+        stats[2 : 2 + F] = (X64 * w64.unsqueeze(1)).sum(dim=0)
+        stats[2 + F : 2 + 2 * F] = (X64**2 * w64.unsqueeze(1)).sum(dim=0)
+        stats[2 + 2 * F] = N
+        dist.all_reduce_(stats)
+        sumW = float(stats[0])
+        y_mean = float(stats[1]) / max(sumW, 1e-300)
+        mean = (stats[2 : 2 + F] / max(sumW, 1e-300)).to(torch.float32)
+        var = (stats[2 + F : 2 + 2 * F] / max(sumW, 1e-300)).to(torch.float32) - mean**2
+        totalN = float(stats[2 + 2 * F])
+        if totalN > 1:
+            var = var * (totalN / (totalN - 1.0))
+        std = var.clamp_min(0).sqrt()
+        inv_std = torch.where(std > 0, 1.0 / std, torch.zeros_like(std))
+        Xs = X * inv_std
+
+        fit_intercept = self.getOrDefault(self.fitIntercept)
+        reg = self.getOrDefault(self.regParam)
+        alpha = self.getOrDefault(self.elasticNetParam)
+        standardization = self.getOrDefault(self.standardization)
+        l2_w = torch.ones(F, device=dev) if standardization else inv_std**2
+        l1_w = torch.ones(F, device=dev) if standardization else inv_std
+        l2_coef = reg * (1.0 - alpha)
+        l1_coef = reg * alpha
+        n_params = F + (1 if fit_intercept else 0)
+        theta0 = torch.zeros(n_params, dtype=torch.float32, device=dev)
+        if fit_intercept:
+            theta0[F] = y_mean
+
+        use_hip = X.is_cuda and ops.hip_available(X)
+
+        def f_grad(theta: torch.Tensor):
+            beta = theta[:F]
+            b = theta[F] if fit_intercept else torch.zeros((), device=dev)
+            r = Xs @ beta + b - y
+            wr = wcol * r if w is not None else r
+            loss_s = 0.5 * (wr * r).sum() if w is not None else 0.5 * (r * r).sum()
+            if use_hip:
+                gW, gB = ops.get_ext().wgrad_f32(Xs, wr.unsqueeze(1).contiguous())
+                gW = gW[0]
+                gB = gB[0:1]
+            else:
+                gW = wr @ Xs
+                gB = wr.sum().reshape(1)
+            buf = torch.cat([loss_s.reshape(1), gW.reshape(-1), gB])
+            dist.all_reduce_(buf)
+            loss = float(buf[0]) / sumW
+            gW = buf[1 : 1 + F] / sumW
+            gB = buf[1 + F :] / sumW
+            if l2_coef > 0:
+                loss += 0.5 * l2_coef * float((l2_w * beta**2).sum())
+                gW = gW + l2_coef * l2_w * beta
+            grad = torch.cat([gW, gB]) if fit_intercept else gW
+            return loss, grad
+
+        max_iter = self.getMaxIter()
+        tol = self.getTol()
+        if l1_coef > 0:
+            l1_vec = torch.zeros(n_params, device=dev)
+            l1_vec[:F] = l1_coef * l1_w
+            theta, history = owlqn(f_grad, theta0, l1_vec, max_iter=max_iter, tol=tol)
+        else:
+            theta, history = lbfgs(f_grad, theta0, max_iter=max_iter, tol=tol)
+
+        coef = (theta[:F] * inv_std).detach().cpu()
+        intercept = float(theta[F]) if fit_intercept else 0.0
+        model = LinearRegressionModel(coefficients=coef, intercept=intercept,
+                                      numFeatures=F)
+        self._copyValues(model)
+        model._set_parent(self)
+        model._summary = LinearRegressionTrainingSummary(
+            objectiveHistory=[float(h) for h in history],
+            totalIterations=max(len(history) - 1, 0),
+            model=model, dataset=dataset,
+            labelCol=self.getLabelCol(),
+            predictionCol=self.getOrDefault(self.predictionCol),
+        )
+        return model
+
+
+class LinearRegressionModel(_LinearRegressionParams, RegressionModel, _RegressorIOMixin):
+    def __init__(self, coefficients=None, intercept: float = 0.0, numFeatures: int = 0):
+        super().__init__()
+        if coefficients is None:
+            self._coef = torch.zeros(0)
+        elif isinstance(coefficients, torch.Tensor):
+            self._coef = coefficients.to(torch.float32)
+        else:
+            self._coef = torch.as_tensor(np_asarray_f32(coefficients))
+        self.intercept = float(intercept)
+        self.numFeatures = numFeatures
+        self.scale = 1.0  # Spark field (non-huber -> 1.0)
+        self._summary = None
+
+    @property
+    def coefficients(self):
+        from flowml.linalg import DenseVector
+
+        return DenseVector(self._coef.to(torch.float64).numpy())
+
+    def hasSummary(self) -> bool:
+        return self._summary is not None
+
+    @property
+    def summary(self) -> "LinearRegressionTrainingSummary":
+        if self._summary is None:
+            raise RuntimeError("No training summary available for this model")
+        return self._summary
+
+    def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
+        c = self._coef.to(device=features.device, dtype=torch.float32)
+        return features.to(torch.float32) @ c + self.intercept
+
+
+class LinearRegressionTrainingSummary:
+    """objectiveHistory + training-set regression metrics (Spark parity:
+    rootMeanSquaredError, meanSquaredError, meanAbsoluteError, r2)."""
+
+    def __init__(self, objectiveHistory, totalIterations, model=None,
+                 dataset=None, labelCol="label", predictionCol="prediction"):
+        self.objectiveHistory = objectiveHistory
+        self.totalIterations = totalIterations
+        self._model = model
+        self._dataset = dataset
+        self.labelCol = labelCol
+        self.predictionCol = predictionCol
+        self._pred = None
+
+    @property
+    def predictions(self):
+        if self._pred is None:
+            self._pred = self._model.transform(self._dataset)
+        return self._pred
+
+    def _eval(self, name):
+        from flowml.evaluation import RegressionEvaluator
+
+        return RegressionEvaluator(metricName=name, labelCol=self.labelCol,
+                                   predictionCol=self.predictionCol).evaluate(self.predictions)
+
+    @property
+    def rootMeanSquaredError(self):
+        return self._eval("rmse")
+
+    @property
+    def meanSquaredError(self):
+        return self._eval("mse")
+
+    @property
+    def meanAbsoluteError(self):
+        return self._eval("mae")
+
+    @property
+    def r2(self):
+        return self._eval("r2")

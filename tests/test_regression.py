@@ -102,3 +102,57 @@ class TestGBTRegressor:
         lm = GBTRegressionModel.load(str(tmp_path / "gbtr"))
         assert lm.treeWeights == pytest.approx(m.treeWeights)
         assert torch.allclose(lm.transform(df)["prediction"], m.transform(df)["prediction"])
+
+
+class TestLinearRegression:
+    def test_recovers_coefficients(self):
+        from flowml.regression import LinearRegression
+
+        r = np.random.default_rng(0)
+        X = torch.from_numpy(r.normal(size=(2000, 5)).astype(np.float32))
+        w = torch.tensor([2.0, -1.0, 0.5, 0.0, 3.0])
+        y = (X @ w + 1.5).to(torch.float64)
+        m = LinearRegression(maxIter=100).fit(FlowFrame({"features": X, "label": y}))
+        assert np.allclose(m.coefficients.toArray(), w.numpy(), atol=1e-3)
+        assert m.intercept == pytest.approx(1.5, abs=1e-3)
+        assert m.summary.r2 > 0.9999
+        assert m.summary.totalIterations >= 1
+
+    def test_l1_sparsifies(self):
+        from flowml.regression import LinearRegression
+
+        r = np.random.default_rng(1)
+        X = torch.from_numpy(r.normal(size=(3000, 8)).astype(np.float32))
+        y = (3.0 * X[:, 0] + 0.05 * torch.from_numpy(r.normal(size=3000).astype(np.float32))).to(torch.float64)
+        m = LinearRegression(maxIter=150, regParam=0.3, elasticNetParam=1.0).fit(
+            FlowFrame({"features": X, "label": y}))
+        c = m.coefficients.toArray()
+        assert abs(c[0]) > 1.0
+        assert np.abs(c[1:]).max() < 1e-3  # L1 zeroed the noise features
+
+    def test_weightcol_duplication(self):
+        from flowml.regression import LinearRegression
+
+        r = np.random.default_rng(2)
+        X = torch.from_numpy(r.normal(size=(400, 3)).astype(np.float32))
+        y = (X[:, 0] - X[:, 2]).to(torch.float64)
+        w = torch.from_numpy(r.integers(1, 4, 400).astype(np.float32))
+        df_w = FlowFrame({"features": X, "label": y, "w": w})
+        idx = torch.repeat_interleave(torch.arange(400), w.to(torch.int64))
+        df_d = FlowFrame({"features": X[idx], "label": y[idx]})
+        mw = LinearRegression(maxIter=60, weightCol="w").fit(df_w)
+        md = LinearRegression(maxIter=60).fit(df_d)
+        assert np.allclose(mw.coefficients.toArray(), md.coefficients.toArray(), atol=1e-4)
+
+    def test_persistence_roundtrip(self, tmp_path):
+        from flowml.regression import LinearRegression, LinearRegressionModel
+
+        r = np.random.default_rng(3)
+        X = torch.from_numpy(r.normal(size=(300, 4)).astype(np.float32))
+        y = (X[:, 1] * 2).to(torch.float64)
+        m = LinearRegression(maxIter=40).fit(FlowFrame({"features": X, "label": y}))
+        m.save(str(tmp_path / "linr"))
+        lm = LinearRegressionModel.load(str(tmp_path / "linr"))
+        assert np.allclose(lm.coefficients.toArray(), m.coefficients.toArray())
+        assert lm.intercept == pytest.approx(m.intercept)
+        assert not lm.hasSummary()
