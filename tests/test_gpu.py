@@ -673,3 +673,15 @@ class TestRegressionGPU:
         assert e.evaluate(m_g.transform(df_g)) > 0.5
         g = GBTRegressor(maxIter=10, maxDepth=4, seed=3).fit(df_g)
         assert e.evaluate(g.transform(df_g)) > 0.7
+
+    def test_linear_regression_gpu(self):
+        from flowml.regression import LinearRegression
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(9)
+        X = torch.from_numpy(r.normal(size=(500_000, 41)).astype(np.float32)).to(DEV)
+        w = torch.from_numpy(r.normal(size=41).astype(np.float32)).to(DEV)
+        y = (X @ w + 0.7).to(torch.float64)
+        m = LinearRegression(maxIter=60).fit(FlowFrame({"features": X, "label": y}, device=DEV))
+        assert np.allclose(m.coefficients.toArray(), w.cpu().numpy(), atol=2e-3)
+        assert m.intercept == pytest.approx(0.7, abs=2e-3)
