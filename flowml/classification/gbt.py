@@ -237,7 +237,8 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
                 bag = wrow.clone() if bag is None else bag * wrow
             _acc("resid", _t0)
             _t1 = _tick()
-            tree_list = grow_trees(codes, resid, bag, thresholds, cfg)
+            tree_list, row_vals = grow_trees(codes, resid, bag, thresholds, cfg,
+                                             return_row_values=True)
             tree = tree_list[0]
             trees.append(tree)
             _acc("grow", _t1)
@@ -245,12 +246,11 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
 
             get_logger().log("gbt_round", round=rnd, nodes=tree.n_nodes)
             weights.append(1.0 if rnd == 0 else step)  # Spark: first tree unshrunk
-            _t2 = _tick()
-            te = TreeEnsemble([tree], 2, mode="scalar")
-            _acc("ensemble_build", _t2)
             _t3 = _tick()
-            pred = te.predict_raw(X)[:, 0]
-            margin = margin + weights[-1] * pred
+            # margin update by leaf-value gather over the settled node ids —
+            # bit-identical to re-predicting the new tree (strict-< binning)
+            # without the K8 pass over raw X
+            margin = margin + weights[-1] * row_vals[0]
             _acc("margin", _t3)
             if val_mask is not None:
                 vls = torch.log1p(torch.exp(-2.0 * ysign[val_mask] * margin[val_mask]))

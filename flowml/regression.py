@@ -249,11 +249,13 @@ class GBTRegressor(_TreeEnsembleParams, _PredictorParams, HasMaxIter, Estimator,
                 n_classes=2,
             )
             bag = poisson_bag(N, 1, subsample, False, cfg.seed, dev) if subsample < 1.0 else None
-            tree = grow_trees(codes, resid, bag, thresholds, cfg)[0]
+            tree_list, row_vals = grow_trees(codes, resid, bag, thresholds, cfg,
+                                             return_row_values=True)
+            tree = tree_list[0]
             trees.append(tree)
             weights.append(1.0 if rnd == 0 else step)
-            te = TreeEnsemble([tree], 1, mode="scalar")
-            margin = margin + weights[-1] * te.predict_raw(X)[:, 0]
+            # leaf-value gather (bit-identical to re-predicting the tree)
+            margin = margin + weights[-1] * row_vals[0]
         model = GBTRegressionModel(trees=trees, treeWeights=weights, numFeatures=F)
         self._copyValues(model)
         model._set_parent(self)
