@@ -23,6 +23,7 @@ from flowml.classification.forest import (
     DecisionTreeClassificationModel,
 )
 from flowml.classification.gbt import GBTClassifier, GBTClassificationModel
+from flowml.classification.naive_bayes import NaiveBayes, NaiveBayesModel
 
 __all__ = [
     "Classifier",
@@ -39,4 +40,6 @@ __all__ = [
     "DecisionTreeClassificationModel",
     "GBTClassifier",
     "GBTClassificationModel",
+    "NaiveBayes",
+    "NaiveBayesModel",
 ]

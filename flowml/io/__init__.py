@@ -81,6 +81,8 @@ _register("GBTRegressor", "org.apache.spark.ml.regression.GBTRegressor")
 _register("GBTRegressionModel", "org.apache.spark.ml.regression.GBTRegressionModel")
 _register("LinearRegression", "org.apache.spark.ml.regression.LinearRegression")
 _register("LinearRegressionModel", "org.apache.spark.ml.regression.LinearRegressionModel")
+_register("NaiveBayes", "org.apache.spark.ml.classification.NaiveBayes")
+_register("NaiveBayesModel", "org.apache.spark.ml.classification.NaiveBayesModel")
 _register("ImputerModel", "org.apache.spark.ml.feature.ImputerModel")
 _register("MinMaxScalerModel", "org.apache.spark.ml.feature.MinMaxScalerModel")
 _register("LogisticRegression", "org.apache.spark.ml.classification.LogisticRegression")
@@ -399,6 +401,17 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "isMultinomial": bool(instance.isMultinomial),
             }
         ]
+    if cls == "NaiveBayesModel":
+        return [
+            {
+                "pi": vector_struct(instance._pi.to(torch.float64).numpy()),
+                "theta": matrix_struct(instance._theta.to(torch.float64).numpy()),
+                "sigma": matrix_struct(
+                    instance._sigma.to(torch.float64).numpy().reshape(
+                        instance._theta.shape if instance._sigma.numel() else (0, 0))),
+                "modelType": instance._mtype,
+            }
+        ]
     if cls == "LinearRegressionModel":
         return [
             {
@@ -498,6 +511,14 @@ def _construct(cls, simple: str, rows, path: str, meta):
             numFeatures=int(r["numFeatures"]),
             isMultinomial=bool(r["isMultinomial"]),
         )
+    if simple == "NaiveBayesModel":
+        r0 = rows[0]
+        th = torch.from_numpy(matrix_from_struct(r0["theta"]))
+        sg = torch.from_numpy(matrix_from_struct(r0["sigma"]))
+        return cls(pi=torch.from_numpy(vector_from_struct(r0["pi"])), theta=th,
+                   sigma=sg if sg.numel() else torch.zeros(0),
+                   numClasses=th.shape[0], numFeatures=th.shape[1],
+                   modelType=r0["modelType"])
     if simple == "LinearRegressionModel":
         r0 = rows[0]
         coef = torch.from_numpy(vector_from_struct(r0["coefficients"])).to(torch.float32)

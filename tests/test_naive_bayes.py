@@ -1,0 +1,95 @@
+"""NaiveBayes vs sklearn oracles + persistence."""
+
+import numpy as np
+import pytest
+import torch
+
+from flowml.classification import NaiveBayes, NaiveBayesModel
+from flowml.frame import FlowFrame
+
+
+class TestMultinomial:
+    def test_matches_sklearn(self):
+        from sklearn.naive_bayes import MultinomialNB
+
+        r = np.random.default_rng(0)
+        X = r.poisson(3.0, size=(500, 8)).astype(np.float32)
+        y = r.integers(0, 3, 500).astype(np.float64)
+        # make classes separable-ish
+        X[y == 1, :3] += 4
+        X[y == 2, 5:] += 4
+        df = FlowFrame({"features": torch.from_numpy(X), "label": torch.from_numpy(y)})
+        m = NaiveBayes(smoothing=1.0).fit(df)
+        sk = MultinomialNB(alpha=1.0).fit(X, y)
+        ours = m.transform(df)["prediction"].numpy()
+        theirs = sk.predict(X)
+        assert (ours == theirs).mean() > 0.99
+        # log class priors match
+        assert np.allclose(m.pi.toArray(), sk.class_log_prior_, atol=1e-9)
+        assert np.allclose(m.theta.toArray(), sk.feature_log_prob_, atol=1e-9)
+
+    def test_probability_normalized(self):
+        r = np.random.default_rng(1)
+        X = r.poisson(2.0, size=(100, 5)).astype(np.float32)
+        y = r.integers(0, 2, 100).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": torch.from_numpy(y)})
+        out = NaiveBayes().fit(df).transform(df)
+        p = out["probability"]
+        assert torch.allclose(p.sum(dim=1), torch.ones(100, dtype=p.dtype))
+
+
+class TestBernoulli:
+    def test_matches_sklearn(self):
+        from sklearn.naive_bayes import BernoulliNB
+
+        r = np.random.default_rng(2)
+        X = (r.random((400, 10)) < 0.3).astype(np.float32)
+        y = r.integers(0, 2, 400).astype(np.float64)
+        X[y == 1, :4] = (r.random((int((y == 1).sum()), 4)) < 0.8).astype(np.float32)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": torch.from_numpy(y)})
+        m = NaiveBayes(modelType="bernoulli", smoothing=1.0).fit(df)
+        sk = BernoulliNB(alpha=1.0).fit(X, y)
+        assert (m.transform(df)["prediction"].numpy() == sk.predict(X)).mean() > 0.99
+
+    def test_rejects_non_binary(self):
+        X = torch.tensor([[0.0, 2.0]])
+        df = FlowFrame({"features": X, "label": torch.tensor([0.0])})
+        with pytest.raises(ValueError):
+            NaiveBayes(modelType="bernoulli").fit(df)
+
+
+class TestGaussian:
+    def test_matches_sklearn(self):
+        from sklearn.naive_bayes import GaussianNB
+
+        r = np.random.default_rng(3)
+        X = r.normal(size=(600, 6)).astype(np.float32)
+        y = r.integers(0, 3, 600).astype(np.float64)
+        X[y == 1] += 2.0
+        X[y == 2, :3] -= 2.0
+        df = FlowFrame({"features": torch.from_numpy(X), "label": torch.from_numpy(y)})
+        m = NaiveBayes(modelType="gaussian").fit(df)
+        sk = GaussianNB().fit(X, y)
+        assert (m.transform(df)["prediction"].numpy() == sk.predict(X)).mean() > 0.99
+
+
+class TestPersistence:
+    def test_roundtrip(self, tmp_path):
+        r = np.random.default_rng(4)
+        X = r.poisson(2.0, size=(200, 5)).astype(np.float32)
+        y = r.integers(0, 2, 200).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": torch.from_numpy(y)})
+        m = NaiveBayes().fit(df)
+        m.save(str(tmp_path / "nb"))
+        lm = NaiveBayesModel.load(str(tmp_path / "nb"))
+        assert torch.equal(lm.transform(df)["prediction"], m.transform(df)["prediction"])
+
+    def test_gaussian_roundtrip(self, tmp_path):
+        r = np.random.default_rng(5)
+        X = r.normal(size=(200, 4)).astype(np.float32)
+        y = (X[:, 0] > 0).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": torch.from_numpy(y)})
+        m = NaiveBayes(modelType="gaussian").fit(df)
+        m.save(str(tmp_path / "gnb"))
+        lm = NaiveBayesModel.load(str(tmp_path / "gnb"))
+        assert torch.allclose(lm.transform(df)["probability"], m.transform(df)["probability"])
