@@ -24,6 +24,7 @@ from flowml.classification.forest import (
 )
 from flowml.classification.gbt import GBTClassifier, GBTClassificationModel
 from flowml.classification.naive_bayes import NaiveBayes, NaiveBayesModel
+from flowml.classification.ovr import OneVsRest, OneVsRestModel
 
 __all__ = [
     "Classifier",
@@ -42,4 +43,6 @@ __all__ = [
     "GBTClassificationModel",
     "NaiveBayes",
     "NaiveBayesModel",
+    "OneVsRest",
+    "OneVsRestModel",
 ]

@@ -93,3 +93,30 @@ class TestPersistence:
         m.save(str(tmp_path / "gnb"))
         lm = NaiveBayesModel.load(str(tmp_path / "gnb"))
         assert torch.allclose(lm.transform(df)["probability"], m.transform(df)["probability"])
+
+
+class TestOneVsRest:
+    def test_matches_direct_multinomial_quality(self):
+        from flowml.classification import LogisticRegression, OneVsRest
+        from flowml.evaluation import MulticlassClassificationEvaluator
+
+        r = np.random.default_rng(7)
+        X = torch.from_numpy(r.normal(size=(600, 5)).astype(np.float32))
+        y = torch.from_numpy(r.integers(0, 3, 600).astype(np.float64))
+        X[y == 1, 0] += 2.5
+        X[y == 2, 1] -= 2.5
+        df = FlowFrame({"features": X, "label": y})
+        ovr = OneVsRest(classifier=LogisticRegression(maxIter=30)).fit(df)
+        acc_ovr = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(ovr.transform(df))
+        direct = LogisticRegression(maxIter=30).fit(df)
+        acc_dir = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(direct.transform(df))
+        assert acc_ovr > 0.85
+        assert abs(acc_ovr - acc_dir) < 0.1
+        assert len(ovr.models) == 3
+
+    def test_requires_classifier(self):
+        from flowml.classification import OneVsRest
+
+        df = FlowFrame({"features": torch.zeros(4, 2), "label": torch.zeros(4, dtype=torch.float64)})
+        with pytest.raises(ValueError):
+            OneVsRest().fit(df)
