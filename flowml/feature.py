@@ -843,3 +843,44 @@ class ImputerModel(_ImputerParams, Model):
         from flowml.io import load as _load
 
         return _load(path)
+
+
+# ---------------------------------------------------------------------------
+# QuantileDiscretizer (spark:.../feature/QuantileDiscretizer.scala)
+# ---------------------------------------------------------------------------
+
+
+class QuantileDiscretizer(HasInputCol, HasOutputCol, HasHandleInvalid, Estimator):
+    """Fit a Bucketizer whose splits are approximate quantiles of the input
+    column (numBuckets equi-depth buckets, -inf/+inf end splits, duplicate
+    quantiles collapsed — so the fitted model can have fewer buckets)."""
+
+    numBuckets: Param[int] = Param(
+        "undefined", "numBuckets",
+        "Number of buckets (quantiles) into which data points are grouped (>= 2).",
+        TypeConverters.toInt,
+    )
+
+    def __init__(self, *, inputCol: Optional[str] = None, outputCol: Optional[str] = None,
+                 numBuckets: Optional[int] = None, handleInvalid: Optional[str] = None):
+        super().__init__()
+        self._setDefault(numBuckets=2, handleInvalid="error")
+        self._set(inputCol=inputCol, outputCol=outputCol, numBuckets=numBuckets,
+                  handleInvalid=handleInvalid)
+
+    def _fit(self, dataset: FlowFrame) -> "Bucketizer":
+        x = dataset.col(self.getInputCol())
+        if isinstance(x, np.ndarray):
+            raise TypeError("QuantileDiscretizer input must be numeric")
+        nb = self.getOrDefault(self.numBuckets)
+        if nb < 2:
+            raise ValueError("numBuckets must be >= 2")
+        xv = x.reshape(-1).to(torch.float64)
+        xv = xv[~torch.isnan(xv)]
+        q = torch.arange(1, nb, dtype=torch.float64, device=xv.device) / nb
+        inner = torch.quantile(xv, q) if len(xv) else torch.zeros(0, dtype=torch.float64)
+        vals = sorted(set(float(v) for v in inner))
+        splits = [-float("inf")] + vals + [float("inf")]
+        return Bucketizer(splits=splits, inputCol=self.getInputCol(),
+                          outputCol=self.getOutputCol(),
+                          handleInvalid=self.getHandleInvalid())

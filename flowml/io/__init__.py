@@ -72,6 +72,7 @@ _register("MinMaxScaler", "org.apache.spark.ml.feature.MinMaxScaler")
 _register("MaxAbsScaler", "org.apache.spark.ml.feature.MaxAbsScaler")
 _register("MaxAbsScalerModel", "org.apache.spark.ml.feature.MaxAbsScalerModel")
 _register("Bucketizer", "org.apache.spark.ml.feature.Bucketizer")
+_register("QuantileDiscretizer", "org.apache.spark.ml.feature.QuantileDiscretizer")
 _register("Imputer", "org.apache.spark.ml.feature.Imputer")
 _register("RandomForestRegressor", "org.apache.spark.ml.regression.RandomForestRegressor")
 _register("RandomForestRegressionModel", "org.apache.spark.ml.regression.RandomForestRegressionModel")

@@ -262,3 +262,27 @@ class TestImputer:
         lm = ImputerModel.load(str(tmp_path / "imp"))
         assert lm.surrogates == m.surrogates
         assert lm.transform(df)["ai"].tolist() == [1.0, 2.0, 3.0]
+
+
+class TestQuantileDiscretizer:
+    def test_equi_depth_buckets(self):
+        from flowml.feature import QuantileDiscretizer
+
+        x = torch.arange(100, dtype=torch.float32)
+        df = FlowFrame({"x": x})
+        model = QuantileDiscretizer(inputCol="x", outputCol="b", numBuckets=4).fit(df)
+        out = model.transform(df)["b"]
+        counts = torch.bincount(out.to(torch.int64))
+        assert len(counts) == 4
+        assert int(counts.min()) >= 20  # roughly equi-depth
+        # fitted model is a plain Bucketizer (Spark returns Bucketizer too)
+        from flowml.feature import Bucketizer
+        assert isinstance(model, Bucketizer)
+
+    def test_duplicate_quantiles_collapse(self):
+        from flowml.feature import QuantileDiscretizer
+
+        x = torch.tensor([1.0] * 50 + [2.0] * 50)
+        df = FlowFrame({"x": x})
+        m = QuantileDiscretizer(inputCol="x", outputCol="b", numBuckets=10).fit(df)
+        assert len(m.getOrDefault(m.splits)) <= 5
