@@ -685,3 +685,20 @@ class TestRegressionGPU:
         m = LinearRegression(maxIter=60).fit(FlowFrame({"features": X, "label": y}, device=DEV))
         assert np.allclose(m.coefficients.toArray(), w.cpu().numpy(), atol=2e-3)
         assert m.intercept == pytest.approx(0.7, abs=2e-3)
+
+    def test_naive_bayes_and_ovr_gpu(self):
+        from flowml.classification import LogisticRegression, NaiveBayes, OneVsRest
+        from flowml.evaluation import MulticlassClassificationEvaluator
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(30)
+        X = torch.from_numpy(np.abs(r.normal(size=(200_000, 12))).astype(np.float32)).to(DEV)
+        y = torch.from_numpy(r.integers(0, 3, 200_000).astype(np.float64)).to(DEV)
+        X[y == 1, 0] += 3
+        X[y == 2, 5] += 3
+        df = FlowFrame({"features": X, "label": y}, device=DEV)
+        ev = MulticlassClassificationEvaluator(metricName="accuracy")
+        nb = NaiveBayes(modelType="gaussian").fit(df)
+        assert ev.evaluate(nb.transform(df)) > 0.8
+        ovr = OneVsRest(classifier=LogisticRegression(maxIter=20)).fit(df)
+        assert ev.evaluate(ovr.transform(df)) > 0.8
