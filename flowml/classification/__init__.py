@@ -25,6 +25,7 @@ from flowml.classification.forest import (
 from flowml.classification.gbt import GBTClassifier, GBTClassificationModel
 from flowml.classification.naive_bayes import NaiveBayes, NaiveBayesModel
 from flowml.classification.ovr import OneVsRest, OneVsRestModel
+from flowml.classification.svc import LinearSVC, LinearSVCModel
 
 __all__ = [
     "Classifier",
@@ -45,4 +46,6 @@ __all__ = [
     "NaiveBayesModel",
     "OneVsRest",
     "OneVsRestModel",
+    "LinearSVC",
+    "LinearSVCModel",
 ]

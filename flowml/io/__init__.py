@@ -84,6 +84,8 @@ _register("LinearRegression", "org.apache.spark.ml.regression.LinearRegression")
 _register("LinearRegressionModel", "org.apache.spark.ml.regression.LinearRegressionModel")
 _register("NaiveBayes", "org.apache.spark.ml.classification.NaiveBayes")
 _register("NaiveBayesModel", "org.apache.spark.ml.classification.NaiveBayesModel")
+_register("LinearSVC", "org.apache.spark.ml.classification.LinearSVC")
+_register("LinearSVCModel", "org.apache.spark.ml.classification.LinearSVCModel")
 _register("ImputerModel", "org.apache.spark.ml.feature.ImputerModel")
 _register("MinMaxScalerModel", "org.apache.spark.ml.feature.MinMaxScalerModel")
 _register("LogisticRegression", "org.apache.spark.ml.classification.LogisticRegression")
@@ -413,6 +415,13 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "modelType": instance._mtype,
             }
         ]
+    if cls == "LinearSVCModel":
+        return [
+            {
+                "intercept": float(instance.intercept),
+                "coefficients": vector_struct(instance._coef.to(torch.float64).numpy()),
+            }
+        ]
     if cls == "LinearRegressionModel":
         return [
             {
@@ -520,6 +529,11 @@ def _construct(cls, simple: str, rows, path: str, meta):
                    sigma=sg if sg.numel() else torch.zeros(0),
                    numClasses=th.shape[0], numFeatures=th.shape[1],
                    modelType=r0["modelType"])
+    if simple == "LinearSVCModel":
+        r0 = rows[0]
+        coef = torch.from_numpy(vector_from_struct(r0["coefficients"])).to(torch.float32)
+        return cls(coefficients=coef, intercept=float(r0["intercept"]),
+                   numFeatures=int(coef.numel()))
     if simple == "LinearRegressionModel":
         r0 = rows[0]
         coef = torch.from_numpy(vector_from_struct(r0["coefficients"])).to(torch.float32)
