@@ -46,6 +46,16 @@ _SUPPORTED = {
 
 
 class MulticlassClassificationEvaluator(HasLabelCol, HasPredictionCol, HasWeightCol, Params):
+    """Confusion-matrix multiclass metrics (weighted f1 default).
+
+    >>> import torch
+    >>> from flowml.frame import FlowFrame
+    >>> df = FlowFrame({"prediction": torch.tensor([0.0, 1.0, 1.0, 0.0]),
+    ...                 "label":      torch.tensor([0.0, 1.0, 0.0, 0.0])})
+    >>> MulticlassClassificationEvaluator(metricName="accuracy").evaluate(df)
+    0.75
+    """
+
     metricName: Param[str] = Param(
         "undefined",
         "metricName",

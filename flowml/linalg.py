@@ -85,7 +85,8 @@ class DenseVector(Vector):
         return int(result)
 
     def __repr__(self) -> str:
-        vals = ",".join(_format_float(v) for v in self.values)
+        # pyspark formats with ", " separators
+        vals = ", ".join(_format_float(v) for v in self.values)
         return f"DenseVector([{vals}])"
 
     def __str__(self) -> str:
@@ -237,6 +238,14 @@ class SparseMatrix(Matrix):
 
 
 class Vectors:
+    """Factory (pyspark.ml.linalg.Vectors).
+
+    >>> Vectors.dense([1.0, 0.0, 3.0])
+    DenseVector([1.0, 0.0, 3.0])
+    >>> Vectors.sparse(3, [0, 2], [1.0, 3.0]).toArray().tolist()
+    [1.0, 0.0, 3.0]
+    """
+
     @staticmethod
     def dense(*elements) -> DenseVector:
         if len(elements) == 1 and not np.isscalar(elements[0]):

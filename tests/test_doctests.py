@@ -2,11 +2,13 @@
 
 import doctest
 
+import flowml.evaluation
+import flowml.linalg
 import flowml.pipeline
 import flowml.tuning
 
 
 def test_module_doctests():
-    for mod in (flowml.pipeline, flowml.tuning):
+    for mod in (flowml.pipeline, flowml.tuning, flowml.evaluation, flowml.linalg):
         results = doctest.testmod(mod, verbose=False)
         assert results.failed == 0, f"{mod.__name__}: {results.failed} doctest failures"
