@@ -884,3 +884,97 @@ class QuantileDiscretizer(HasInputCol, HasOutputCol, HasHandleInvalid, Estimator
         return Bucketizer(splits=splits, inputCol=self.getInputCol(),
                           outputCol=self.getOutputCol(),
                           handleInvalid=self.getHandleInvalid())
+
+
+# ---------------------------------------------------------------------------
+# PCA (spark:.../feature/PCA.scala)
+# ---------------------------------------------------------------------------
+
+
+class _PCAParams(HasInputCol, HasOutputCol):
+    k: Param[int] = Param(
+        "undefined", "k", "the number of principal components", TypeConverters.toInt
+    )
+
+
+class PCA(_PCAParams, Estimator):
+    """Project vectors onto the top-k principal components. The covariance
+    moments are per-rank reductions all-reduced across GPUs (SURVEY.md §2.5
+    C6); the K×K eigendecomposition runs replicated on every rank."""
+
+    def __init__(self, *, k: Optional[int] = None, inputCol: Optional[str] = None,
+                 outputCol: Optional[str] = None):
+        super().__init__()
+        self._set(k=k, inputCol=inputCol, outputCol=outputCol)
+
+    def _fit(self, dataset: FlowFrame) -> "PCAModel":
+        from flowml import dist
+
+        x = dataset.col(self.getInputCol())
+        if not isinstance(x, torch.Tensor) or x.dim() != 2:
+            raise TypeError("PCA input must be a vector column")
+        k = self.getOrDefault(self.k)
+        N, F = x.shape
+        if k is None or not (1 <= k <= F):
+            raise ValueError(f"k must be in [1, {F}]")
+        X = x.to(torch.float64)
+        buf = torch.zeros(1 + F + F * F, dtype=torch.float64, device=X.device)
+        buf[0] = N
+        buf[1 : 1 + F] = X.sum(dim=0)
+        buf[1 + F :] = (X.T @ X).reshape(-1)
+        dist.all_reduce_(buf)
+        n = float(buf[0])
+        mean = buf[1 : 1 + F] / n
+        cov = (buf[1 + F :].view(F, F) - n * torch.outer(mean, mean)) / max(n - 1, 1.0)
+        evals, evecs = torch.linalg.eigh(cov.cpu())  # ascending
+        order = torch.argsort(evals, descending=True)[:k]
+        comps = evecs[:, order]                       # F×k
+        ev = evals[order].clamp_min(0)
+        total = float(evals.clamp_min(0).sum())
+        model = PCAModel(
+            pc=comps.to(torch.float64),
+            explainedVariance=(ev / total if total > 0 else ev).to(torch.float64),
+        )
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class PCAModel(_PCAParams, Model):
+    def __init__(self, pc: Optional[torch.Tensor] = None,
+                 explainedVariance: Optional[torch.Tensor] = None):
+        super().__init__()
+        self._pc = pc if pc is not None else torch.zeros(0, 0, dtype=torch.float64)
+        self._ev = explainedVariance if explainedVariance is not None else torch.zeros(0)
+
+    @property
+    def pc(self):
+        from flowml.linalg import Matrices
+
+        return Matrices.dense(self._pc.shape[0], self._pc.shape[1],
+                              self._pc.numpy().T.flatten())
+
+    @property
+    def explainedVariance(self):
+        from flowml.linalg import DenseVector
+
+        return DenseVector(self._ev.numpy())
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getInputCol())
+        pc = self._pc.to(device=x.device, dtype=x.dtype)
+        return dataset.withColumn(self.getOutputCol(), x @ pc)
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)

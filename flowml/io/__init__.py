@@ -73,6 +73,8 @@ _register("MaxAbsScaler", "org.apache.spark.ml.feature.MaxAbsScaler")
 _register("MaxAbsScalerModel", "org.apache.spark.ml.feature.MaxAbsScalerModel")
 _register("Bucketizer", "org.apache.spark.ml.feature.Bucketizer")
 _register("QuantileDiscretizer", "org.apache.spark.ml.feature.QuantileDiscretizer")
+_register("PCA", "org.apache.spark.ml.feature.PCA")
+_register("PCAModel", "org.apache.spark.ml.feature.PCAModel")
 _register("Imputer", "org.apache.spark.ml.feature.Imputer")
 _register("RandomForestRegressor", "org.apache.spark.ml.regression.RandomForestRegressor")
 _register("RandomForestRegressionModel", "org.apache.spark.ml.regression.RandomForestRegressionModel")
@@ -415,6 +417,13 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "modelType": instance._mtype,
             }
         ]
+    if cls == "PCAModel":
+        return [
+            {
+                "pc": matrix_struct(instance._pc.numpy()),
+                "explainedVariance": vector_struct(instance._ev.numpy()),
+            }
+        ]
     if cls == "LinearSVCModel":
         return [
             {
@@ -529,6 +538,10 @@ def _construct(cls, simple: str, rows, path: str, meta):
                    sigma=sg if sg.numel() else torch.zeros(0),
                    numClasses=th.shape[0], numFeatures=th.shape[1],
                    modelType=r0["modelType"])
+    if simple == "PCAModel":
+        r0 = rows[0]
+        return cls(pc=torch.from_numpy(matrix_from_struct(r0["pc"])),
+                   explainedVariance=torch.from_numpy(vector_from_struct(r0["explainedVariance"])))
     if simple == "LinearSVCModel":
         r0 = rows[0]
         coef = torch.from_numpy(vector_from_struct(r0["coefficients"])).to(torch.float32)

@@ -286,3 +286,28 @@ class TestQuantileDiscretizer:
         df = FlowFrame({"x": x})
         m = QuantileDiscretizer(inputCol="x", outputCol="b", numBuckets=10).fit(df)
         assert len(m.getOrDefault(m.splits)) <= 5
+
+
+class TestPCA:
+    def test_matches_sklearn(self, tmp_path):
+        from sklearn.decomposition import PCA as SkPCA
+
+        from flowml.feature import PCA, PCAModel
+
+        r = np.random.default_rng(5)
+        base = r.normal(size=(500, 2))
+        X = np.hstack([base, base @ r.normal(size=(2, 3))]).astype(np.float32)
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = PCA(k=2, inputCol="features", outputCol="p").fit(df)
+        ours = m.transform(df)["p"].numpy()
+        sk = SkPCA(n_components=2).fit(X)
+        theirs = sk.transform(X)
+        # components are sign/center ambiguous: compare |corr| per component
+        for j in range(2):
+            c = np.corrcoef(ours[:, j], theirs[:, j])[0, 1]
+            assert abs(c) > 0.9999
+        assert np.allclose(m.explainedVariance.toArray(),
+                           sk.explained_variance_ratio_, atol=1e-4)
+        m.save(str(tmp_path / "pca"))
+        lm = PCAModel.load(str(tmp_path / "pca"))
+        assert np.allclose(lm.transform(df)["p"].numpy(), ours)
