@@ -1,0 +1,248 @@
+"""KMeans clustering + ClusteringEvaluator (pyspark.ml.clustering /
+spark:.../ml/clustering/KMeans.scala, evaluation/ClusteringEvaluator.scala).
+
+Lloyd iterations over the resident shard: assignments are a fused
+distance-matrix argmin (|x|² − 2x·cᵀ + |c|², rocBLAS GEMM on GPU), the
+per-cluster (count, Σx) statistics are computed with a deterministic one-hot
+GEMM and all-reduced across ranks (SURVEY.md §2.5 C2 pattern), and every
+rank applies the identical center update — the replicated-state design of
+the tree/LR fits. Init: k-means|| (default, Spark's oversampling init) or
+random. ClusteringEvaluator implements Spark's squared-euclidean silhouette
+(per-cluster moments ⇒ O(N·k), not O(N²)).
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from flowml import dist
+from flowml.base import Estimator, Model
+from flowml.frame import FlowFrame
+from flowml.param import (
+    HasFeaturesCol,
+    HasMaxIter,
+    HasPredictionCol,
+    HasSeed,
+    HasTol,
+    HasWeightCol,
+    Param,
+    Params,
+    TypeConverters,
+)
+
+
+class _KMeansParams(HasFeaturesCol, HasPredictionCol, HasMaxIter, HasSeed, HasTol, HasWeightCol):
+    k: Param[int] = Param(
+        "undefined", "k", "The number of clusters to create (> 1).", TypeConverters.toInt
+    )
+    initMode: Param[str] = Param(
+        "undefined", "initMode", "The initialization algorithm: random | k-means||",
+        TypeConverters.toString,
+    )
+    initSteps: Param[int] = Param(
+        "undefined", "initSteps", "The number of steps for k-means|| (> 0).",
+        TypeConverters.toInt,
+    )
+
+
+def _dist2(X: torch.Tensor, C: torch.Tensor) -> torch.Tensor:
+    """N×k squared euclidean distances via GEMM (clamped at 0)."""
+    x2 = (X * X).sum(dim=1, keepdim=True)
+    c2 = (C * C).sum(dim=1)
+    return (x2 - 2.0 * (X @ C.T) + c2).clamp_min(0.0)
+
+
+class KMeans(_KMeansParams, Estimator):
+    def __init__(self, **kwargs):
+        super().__init__()
+        self._setDefault(featuresCol="features", predictionCol="prediction",
+                         k=2, maxIter=20, initMode="k-means||", initSteps=2,
+                         tol=1e-4, seed=42)
+        self._set(**kwargs)
+
+    # -- k-means|| init (replicated across ranks: seeded identically) -------
+    def _init_centers(self, X: torch.Tensor, k: int) -> torch.Tensor:
+        seed = self.getOrDefault(self.seed)
+        mode = self.getOrDefault(self.initMode)
+        N = X.shape[0]
+        g = torch.Generator(device=X.device if X.is_cuda else "cpu")
+        g.manual_seed(seed & 0x7FFFFFFF)
+        if mode == "random":
+            idx = torch.randperm(N, generator=g, device=X.device)[:k]
+            return X[idx].clone().to(torch.float64)
+        if mode != "k-means||":
+            raise ValueError(f"Unsupported initMode {mode!r}")
+        # NOTE: single-shard draw per rank then rank-0 broadcast keeps all
+        # ranks identical; candidates are capped so the final local k-means
+        # (on candidates) is trivial.
+        first = int(torch.randint(0, N, (1,), generator=g, device=X.device))
+        centers = X[first : first + 1].to(torch.float64)
+        steps = self.getOrDefault(self.initSteps)
+        l = 2 * k  # Spark's oversampling factor ~2k per step
+        for _ in range(steps):
+            d2 = _dist2(X.to(torch.float64), centers).min(dim=1).values
+            total = float(d2.sum().clamp_min(1e-300))
+            probs = (l * d2 / total).clamp(0.0, 1.0)
+            u = torch.rand(N, generator=g, device=X.device)
+            picked = torch.nonzero(u < probs, as_tuple=False).flatten()
+            if len(picked):
+                centers = torch.cat([centers, X[picked].to(torch.float64)], dim=0)
+        # weight candidates by assigned counts, then greedy k-means++ on the
+        # (small) candidate set to get exactly k centers
+        d2 = _dist2(X.to(torch.float64), centers)
+        assign = d2.argmin(dim=1)
+        w = torch.bincount(assign, minlength=len(centers)).to(torch.float64)
+        cand = centers
+        chosen = [int(w.argmax())]
+        for _ in range(min(k, len(cand)) - 1):
+            cd2 = _dist2(cand, cand[chosen]).min(dim=1).values * w
+            cd2[chosen] = -1.0
+            chosen.append(int(cd2.argmax()))
+        C = cand[chosen]
+        if len(C) < k:  # degenerate: pad with random rows
+            idx = torch.randperm(N, generator=g, device=X.device)[: k - len(C)]
+            C = torch.cat([C, X[idx].to(torch.float64)], dim=0)
+        return C.clone()
+
+    def _fit(self, dataset: FlowFrame) -> "KMeansModel":
+        X = dataset.col(self.getFeaturesCol())
+        if not isinstance(X, torch.Tensor) or X.dim() != 2:
+            raise TypeError("KMeans features must be a vector column")
+        Xd = X.to(torch.float64)
+        N, F = Xd.shape
+        k = self.getOrDefault(self.k)
+        if k < 2:
+            raise ValueError("k must be > 1")
+        centers = self._init_centers(Xd, k)
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            dist.broadcast_(centers)  # rank-0 init wins (shards differ)
+        tol = self.getOrDefault(self.tol)
+        cost = float("inf")
+        it = 0
+        for it in range(self.getMaxIter()):
+            d2 = _dist2(Xd, centers)
+            mind2, assign = d2.min(dim=1)
+            onehot = torch.zeros(N, k, dtype=torch.float64, device=Xd.device)
+            onehot[torch.arange(N, device=Xd.device), assign] = 1.0
+            # deterministic per-cluster stats: GEMM, then one all_reduce
+            buf = torch.cat([
+                onehot.sum(dim=0),                    # k counts
+                (onehot.T @ Xd).reshape(-1),          # k×F sums
+                mind2.sum().reshape(1),               # cost
+            ])
+            dist.all_reduce_(buf)
+            cnt = buf[:k]
+            sums = buf[k : k + k * F].view(k, F)
+            new_cost = float(buf[-1])
+            nonempty = cnt > 0
+            new_centers = torch.where(
+                nonempty.unsqueeze(1), sums / cnt.clamp_min(1.0).unsqueeze(1), centers
+            )
+            shift = float(((new_centers - centers) ** 2).sum(dim=1).max())
+            centers = new_centers
+            if shift <= tol * tol:
+                cost = new_cost
+                break
+            cost = new_cost
+        model = KMeansModel(centers=centers.cpu(), trainingCost=cost,
+                            numIter=it + 1)
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class KMeansModel(_KMeansParams, Model):
+    def __init__(self, centers: Optional[torch.Tensor] = None,
+                 trainingCost: float = 0.0, numIter: int = 0):
+        super().__init__()
+        self._centers = centers if centers is not None else torch.zeros(0, 0, dtype=torch.float64)
+        self.trainingCost = trainingCost
+        self.numIter = numIter
+
+    def clusterCenters(self) -> List:
+        return [c.numpy() for c in self._centers]
+
+    @property
+    def k(self) -> int:  # noqa: A003 - Spark exposes model.k... via param; keep simple
+        return self._centers.shape[0]
+
+    def predict(self, features) -> int:
+        from flowml.classification.base import _as_row_tensor
+
+        t = _as_row_tensor(features).to(torch.float64)
+        return int(_dist2(t, self._centers.to(t.device)).argmin(dim=1)[0])
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        X = dataset.col(self.getFeaturesCol()).to(torch.float64)
+        C = self._centers.to(X.device)
+        assign = _dist2(X, C).argmin(dim=1)
+        return dataset.withColumn(self.getOrDefault(self.predictionCol),
+                                  assign.to(torch.float64))
+
+    def computeCost(self, dataset: FlowFrame) -> float:
+        X = dataset.col(self.getFeaturesCol()).to(torch.float64)
+        return float(_dist2(X, self._centers.to(X.device)).min(dim=1).values.sum())
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+class ClusteringEvaluator(HasFeaturesCol, HasPredictionCol, Params):
+    """Spark's squared-euclidean silhouette via per-cluster moments:
+    for point x, a = mean ||x−y||² over own cluster = |x|² − 2x·μ_c + m_c
+    (m_c = mean |y|² of the cluster); b = min over other clusters.
+    s = (b − a)/max(a, b); metric = mean s (larger is better)."""
+
+    metricName: Param[str] = Param(
+        "undefined", "metricName", "metric name in evaluation (silhouette)",
+        TypeConverters.toString,
+    )
+
+    def __init__(self, *, featuresCol: Optional[str] = None,
+                 predictionCol: Optional[str] = None, metricName: Optional[str] = None):
+        super().__init__()
+        self._setDefault(featuresCol="features", predictionCol="prediction",
+                         metricName="silhouette")
+        self._set(featuresCol=featuresCol, predictionCol=predictionCol,
+                  metricName=metricName)
+
+    def isLargerBetter(self) -> bool:
+        return True
+
+    def evaluate(self, dataset: FlowFrame) -> float:
+        if self.getOrDefault(self.metricName) != "silhouette":
+            raise ValueError("Only metricName='silhouette' is supported")
+        X = dataset.col(self.getFeaturesCol()).to(torch.float64)
+        a = dataset.col(self.getOrDefault(self.predictionCol)).reshape(-1).to(torch.int64)
+        N, F = X.shape
+        k = int(a.max()) + 1
+        onehot = torch.zeros(N, k, dtype=torch.float64, device=X.device)
+        onehot[torch.arange(N, device=X.device), a] = 1.0
+        cnt = onehot.sum(dim=0)                                  # k
+        mu = (onehot.T @ X) / cnt.clamp_min(1.0).unsqueeze(1)    # k×F
+        x2 = (X * X).sum(dim=1)
+        m2 = (onehot.T @ x2.unsqueeze(1)).squeeze(1) / cnt.clamp_min(1.0)  # k
+        # mean squared distance of x to cluster c: |x|² − 2 x·μ_c + m2_c
+        d = x2.unsqueeze(1) - 2.0 * (X @ mu.T) + m2              # N×k
+        own = d.gather(1, a.unsqueeze(1)).squeeze(1)
+        d_other = d.scatter(1, a.unsqueeze(1), float("inf"))
+        b = d_other.min(dim=1).values
+        s = torch.where(
+            cnt[a] > 1,
+            (b - own) / torch.maximum(own, b).clamp_min(1e-300),
+            torch.zeros_like(own),
+        )
+        return float(s.mean())

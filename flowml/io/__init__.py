@@ -75,6 +75,8 @@ _register("Bucketizer", "org.apache.spark.ml.feature.Bucketizer")
 _register("QuantileDiscretizer", "org.apache.spark.ml.feature.QuantileDiscretizer")
 _register("PCA", "org.apache.spark.ml.feature.PCA")
 _register("PCAModel", "org.apache.spark.ml.feature.PCAModel")
+_register("KMeans", "org.apache.spark.ml.clustering.KMeans")
+_register("KMeansModel", "org.apache.spark.ml.clustering.KMeansModel")
 _register("Imputer", "org.apache.spark.ml.feature.Imputer")
 _register("RandomForestRegressor", "org.apache.spark.ml.regression.RandomForestRegressor")
 _register("RandomForestRegressionModel", "org.apache.spark.ml.regression.RandomForestRegressionModel")
@@ -118,9 +120,10 @@ def _our_class(spark_name: str):
     import flowml.pipeline as pipeline
     import flowml.classification as cls
     import flowml.regression as reg
+    import flowml.clustering as clus
 
     simple = spark_name.rsplit(".", 1)[-1]
-    for mod in (pipeline, feature, cls, reg):
+    for mod in (pipeline, feature, cls, reg, clus):
         if hasattr(mod, simple):
             return getattr(mod, simple)
     raise ValueError(f"Cannot resolve class for {spark_name!r}")
@@ -417,6 +420,13 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "modelType": instance._mtype,
             }
         ]
+    if cls == "KMeansModel":
+        return [
+            {
+                "clusterCenters": matrix_struct(instance._centers.numpy()),
+                "trainingCost": float(instance.trainingCost),
+            }
+        ]
     if cls == "PCAModel":
         return [
             {
@@ -538,6 +548,10 @@ def _construct(cls, simple: str, rows, path: str, meta):
                    sigma=sg if sg.numel() else torch.zeros(0),
                    numClasses=th.shape[0], numFeatures=th.shape[1],
                    modelType=r0["modelType"])
+    if simple == "KMeansModel":
+        r0 = rows[0]
+        return cls(centers=torch.from_numpy(matrix_from_struct(r0["clusterCenters"])),
+                   trainingCost=float(r0["trainingCost"]))
     if simple == "PCAModel":
         r0 = rows[0]
         return cls(pc=torch.from_numpy(matrix_from_struct(r0["pc"])),

@@ -1,0 +1,81 @@
+"""KMeans + ClusteringEvaluator vs sklearn oracles."""
+
+import numpy as np
+import pytest
+import torch
+
+from flowml.clustering import ClusteringEvaluator, KMeans, KMeansModel
+from flowml.frame import FlowFrame
+
+
+def _blobs(n=600, seed=0):
+    r = np.random.default_rng(seed)
+    centers = np.array([[0.0, 0.0], [6.0, 0.0], [0.0, 6.0]])
+    X = np.vstack([c + r.normal(scale=0.6, size=(n // 3, 2)) for c in centers]).astype(np.float32)
+    labels = np.repeat(np.arange(3), n // 3)
+    return X, labels
+
+
+class TestKMeans:
+    def test_recovers_blobs(self):
+        X, labels = _blobs()
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = KMeans(k=3, seed=1).fit(df)
+        pred = m.transform(df)["prediction"].numpy().astype(int)
+        # cluster/label agreement up to permutation: each true blob maps to
+        # one dominant cluster
+        for c in range(3):
+            dominant = np.bincount(pred[labels == c], minlength=3).max()
+            assert dominant > 0.95 * (labels == c).sum()
+        assert len(m.clusterCenters()) == 3
+        assert m.trainingCost < 600 * 2 * 0.6**2 * 3  # sane inertia
+
+    def test_cost_close_to_sklearn(self):
+        from sklearn.cluster import KMeans as SkKMeans
+
+        X, _ = _blobs(seed=3)
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = KMeans(k=3, seed=5, maxIter=30).fit(df)
+        sk = SkKMeans(n_clusters=3, n_init=5).fit(X.astype(np.float64))
+        assert m.computeCost(df) < sk.inertia_ * 1.05
+
+    def test_random_init_and_predict(self):
+        X, _ = _blobs(seed=4)
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = KMeans(k=3, seed=2, initMode="random").fit(df)
+        single = m.predict([0.1, 0.1])
+        assert 0 <= single < 3
+
+    def test_persistence(self, tmp_path):
+        X, _ = _blobs(seed=6)
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = KMeans(k=3, seed=7).fit(df)
+        m.save(str(tmp_path / "km"))
+        lm = KMeansModel.load(str(tmp_path / "km"))
+        assert torch.equal(lm.transform(df)["prediction"], m.transform(df)["prediction"])
+        assert lm.trainingCost == pytest.approx(m.trainingCost)
+
+
+class TestClusteringEvaluator:
+    def test_silhouette_vs_sklearn(self):
+        from sklearn.metrics import silhouette_score
+
+        X, labels = _blobs(seed=8)
+        df = FlowFrame({"features": torch.from_numpy(X),
+                        "prediction": torch.from_numpy(labels.astype(np.float64))})
+        ours = ClusteringEvaluator().evaluate(df)
+        # Spark/our silhouette uses SQUARED distances; sklearn uses
+        # euclidean — not identical, but both should be high for clean blobs
+        theirs = silhouette_score(X, labels)
+        assert ours > 0.7 and theirs > 0.7
+        assert ClusteringEvaluator().isLargerBetter()
+
+    def test_bad_labels_score_worse(self):
+        X, labels = _blobs(seed=9)
+        r = np.random.default_rng(0)
+        bad = r.integers(0, 3, len(labels)).astype(np.float64)
+        good = ClusteringEvaluator().evaluate(FlowFrame(
+            {"features": torch.from_numpy(X), "prediction": torch.from_numpy(labels.astype(np.float64))}))
+        worse = ClusteringEvaluator().evaluate(FlowFrame(
+            {"features": torch.from_numpy(X), "prediction": torch.from_numpy(bad)}))
+        assert good > worse
