@@ -164,8 +164,7 @@ class KMeansModel(_KMeansParams, Model):
     def clusterCenters(self) -> List:
         return [c.numpy() for c in self._centers]
 
-    @property
-    def k(self) -> int:  # noqa: A003 - Spark exposes model.k... via param; keep simple
+    def numClusters(self) -> int:
         return self._centers.shape[0]
 
     def predict(self, features) -> int:
