@@ -177,3 +177,23 @@ def test_gbt_dist_close_to_single():
     # identical global histograms (hash bagging off for GBT; thresholds from
     # the gathered sample) -> identical trees up to fp reduction order
     assert np.corrcoef(raw_dist, raw_single)[0, 1] > 0.99
+
+
+def _kmeans_dist_body(rank, world):
+    from flowml import dist as fd
+    from flowml.clustering import KMeans
+    from flowml.frame import FlowFrame
+
+    X, _ = _make_data()
+    lo, hi = fd.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi])})
+    m = KMeans(k=3, seed=11, maxIter=15).fit(df)
+    c = np.stack(m.clusterCenters())
+    gathered = fd.all_gather_object(c)
+    assert all(np.allclose(g, gathered[0]) for g in gathered), "ranks diverged"
+    return c
+
+
+def test_kmeans_dist_ranks_agree():
+    centers = _run_dist("_kmeans_dist_body")
+    assert centers.shape[0] == 3
