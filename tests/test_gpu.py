@@ -702,3 +702,24 @@ class TestRegressionGPU:
         assert ev.evaluate(nb.transform(df)) > 0.8
         ovr = OneVsRest(classifier=LogisticRegression(maxIter=20)).fit(df)
         assert ev.evaluate(ovr.transform(df)) > 0.8
+
+    def test_kmeans_pca_svc_gpu(self):
+        from flowml.classification import LinearSVC
+        from flowml.clustering import ClusteringEvaluator, KMeans
+        from flowml.feature import PCA
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(40)
+        centers = np.array([[0, 0, 0], [8, 0, 0], [0, 8, 0]], dtype=np.float64)
+        X = np.vstack([c + r.normal(scale=0.5, size=(50_000, 3)) for c in centers]).astype(np.float32)
+        df = FlowFrame({"features": torch.from_numpy(X)}, device=DEV)
+        km = KMeans(k=3, seed=1).fit(df)
+        sil = ClusteringEvaluator().evaluate(km.transform(df))
+        assert sil > 0.8
+        p = PCA(k=2, inputCol="features", outputCol="pc").fit(df)
+        assert p.transform(df)["pc"].shape == (150_000, 2)
+        y = torch.from_numpy((X[:, 0] > 4).astype(np.float64)).to(DEV)
+        df2 = FlowFrame({"features": torch.from_numpy(X), "label": y}, device=DEV)
+        svc = LinearSVC(maxIter=30).fit(df2)
+        acc = (svc.transform(df2)["prediction"] == y).double().mean()
+        assert float(acc) > 0.99
