@@ -10,7 +10,6 @@ scores (Spark's ProbabilisticClassifier contract).
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 
