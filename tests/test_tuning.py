@@ -95,3 +95,29 @@ class TestBinaryEvaluator:
         m = LogisticRegression(maxIter=30).fit(df)
         auc = BinaryClassificationEvaluator().evaluate(m.transform(df))
         assert auc > 0.9
+
+
+class TestSmallerIsBetterMetric:
+    def test_cv_with_rmse_picks_lower(self):
+        """CrossValidator must respect isLargerBetter()=False (rmse)."""
+        import numpy as np
+
+        from flowml.evaluation import RegressionEvaluator
+        from flowml.regression import LinearRegression
+        from flowml.tuning import CrossValidator, ParamGridBuilder
+
+        r = np.random.default_rng(0)
+        X = torch.from_numpy(r.normal(size=(400, 4)).astype(np.float32))
+        y = (X @ torch.tensor([1.0, -2.0, 0.5, 0.0]) + 0.05 *
+             torch.from_numpy(r.normal(size=400).astype(np.float32))).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y})
+        lr = LinearRegression(maxIter=50)
+        grid = (ParamGridBuilder()
+                .addGrid(lr.regParam, [0.0, 50.0])  # huge reg must lose on rmse
+                .build())
+        cv = CrossValidator(estimator=lr, estimatorParamMaps=grid,
+                            evaluator=RegressionEvaluator(metricName="rmse"),
+                            numFolds=3, seed=1)
+        m = cv.fit(df)
+        assert m.avgMetrics[0] < m.avgMetrics[1]
+        assert m.bestModel.getOrDefault(m.bestModel.regParam) == 0.0
