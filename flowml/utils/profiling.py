@@ -3,8 +3,8 @@
 Replaces the Spark UI/event-log with hipEvent-pair phase timing: every
 ``with timer.phase("hist"):`` records a GPU-accurate duration (torch.cuda
 events on ROCm; wall time on CPU) and ``dump_chrome_trace`` writes a
-chrome://tracing-loadable JSON per rank. rocprofv3 counter runs are driven by
-``bench.py --profile``.
+chrome://tracing-loadable JSON per rank. rocprofv3 counter runs wrap ``bench.py``
+directly (see profiles/NOTES.md for the recipes).
 """
 
 from __future__ import annotations
