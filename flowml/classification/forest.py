@@ -302,6 +302,10 @@ class DecisionTreeClassificationModel(_TreeEnsembleParams, _HasTrainingSummary, 
     def numNodes(self) -> int:
         return self._tree.n_nodes
 
+    @property
+    def featureImportances(self) -> DenseVector:
+        return DenseVector(self._ensemble.feature_importances(self.numFeatures))
+
     def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
         # single tree raw = leaf class counts (Spark) — we keep normalized
         # probabilities × count to match impurityStats-based raw closely

@@ -192,6 +192,12 @@ class DecisionTreeRegressionModel(_TreeEnsembleParams, RegressionModel, _Regress
     def numNodes(self) -> int:
         return self._trees[0].n_nodes if self._trees else 0
 
+    @property
+    def featureImportances(self):
+        from flowml.linalg import DenseVector
+
+        return DenseVector(self._ensemble.feature_importances(self.numFeatures))
+
     def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
         return self._ensemble.predict_raw(features)[:, 0]
 
@@ -281,6 +287,12 @@ class GBTRegressionModel(_TreeEnsembleParams, RegressionModel, _RegressorIOMixin
 
     def getNumTrees(self) -> int:
         return len(self._trees)
+
+    @property
+    def featureImportances(self):
+        from flowml.linalg import DenseVector
+
+        return DenseVector(self._ensemble.feature_importances(self.numFeatures))
 
     def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
         return self._ensemble.predict_raw(features)[:, 0]

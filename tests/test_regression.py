@@ -156,3 +156,22 @@ class TestLinearRegression:
         assert np.allclose(lm.coefficients.toArray(), m.coefficients.toArray())
         assert lm.intercept == pytest.approx(m.intercept)
         assert not lm.hasSummary()
+
+
+class TestFeatureImportancesEverywhere:
+    def test_gbt_and_dt_importances(self):
+        from flowml.classification import DecisionTreeClassifier, GBTClassifier
+        from flowml.regression import DecisionTreeRegressor, GBTRegressor
+
+        df, X, y = _reg_df(800)
+        ycls = (y > y.median()).to(torch.float64)
+        dfc = FlowFrame({"features": X, "label": ycls})
+        for m in (
+            GBTRegressor(maxIter=3, maxDepth=3, seed=0).fit(df),
+            DecisionTreeRegressor(maxDepth=3, seed=0).fit(df),
+            GBTClassifier(maxIter=3, maxDepth=3, seed=0).fit(dfc),
+            DecisionTreeClassifier(maxDepth=3, seed=0).fit(dfc),
+        ):
+            imp = m.featureImportances.toArray()
+            assert imp.shape == (6,)
+            assert abs(imp.sum() - 1.0) < 1e-9 or imp.sum() == 0.0
