@@ -437,8 +437,12 @@ class StandardScalerModel(_StandardScalerParams, Model):
     def _transform(self, dataset: FlowFrame) -> FlowFrame:
         col = dataset.col(self.getInputCol())
         x = col.to(torch.float32)
-        mean = self.mean.to(device=x.device, dtype=x.dtype)
-        std = self.std.to(device=x.device, dtype=x.dtype)
+        # persist the device migration: repeat transforms (and hipGraph
+        # capture after one warm-up) see resident tensors, no H2D per call
+        self.mean = self.mean.to(device=x.device, dtype=x.dtype)
+        self.std = self.std.to(device=x.device, dtype=x.dtype)
+        mean = self.mean
+        std = self.std
         if self.getWithMean():
             x = x - mean
         if self.getWithStd():
@@ -523,8 +527,10 @@ class MinMaxScalerModel(_MinMaxScalerParams, Model):
     def _transform(self, dataset: FlowFrame) -> FlowFrame:
         col = dataset.col(self.getInputCol())
         x = col.to(torch.float32)
-        omin = self.originalMin.to(device=x.device, dtype=x.dtype)
-        omax = self.originalMax.to(device=x.device, dtype=x.dtype)
+        self.originalMin = self.originalMin.to(device=x.device, dtype=x.dtype)
+        self.originalMax = self.originalMax.to(device=x.device, dtype=x.dtype)
+        omin = self.originalMin
+        omax = self.originalMax
         lo, hi = self.getMin(), self.getMax()
         rng = omax - omin
         const = rng == 0
@@ -678,8 +684,8 @@ class MaxAbsScalerModel(_MaxAbsScalerParams, Model):
 
     def _transform(self, dataset: FlowFrame) -> FlowFrame:
         x = dataset.col(self.getInputCol())
-        d = self.maxAbs.to(device=x.device, dtype=x.dtype)
-        d = torch.where(d == 0, torch.ones_like(d), d)
+        self.maxAbs = self.maxAbs.to(device=x.device, dtype=x.dtype)
+        d = torch.where(self.maxAbs == 0, torch.ones_like(self.maxAbs), self.maxAbs)
         return dataset.withColumn(self.getOutputCol(), x / d)
 
     def write(self):

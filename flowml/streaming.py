@@ -33,7 +33,20 @@ class StreamingClassifier:
         num_features: Optional[int] = None,
         device=None,
         use_graph: bool = True,
+        featurizer=None,
+        input_col: str = "features",
+        feature_col: str = "features",
     ):
+        """``featurizer``: optional fitted transformer/PipelineModel of
+        NUMERIC, sync-free stages (scalers, Normalizer, Binarizer,
+        VectorAssembler over tensor cols...) applied to the raw micro-batch
+        INSIDE the captured graph — the SURVEY.md §2.4 K8
+        "featurize + ensemble-forward" capture. ``num_features`` then sizes
+        the RAW input (``input_col``); ``feature_col`` names the featurized
+        output the model consumes. Stages that touch the host (StringIndexer
+        on string columns, handleInvalid='error' scans) cannot be captured —
+        capture failures fall back to eager replay with a warning.
+        """
         self.model = model
         self.batch_size = int(batch_size)
         self.device = torch.device(device) if device is not None else torch.device("cuda:0")
@@ -42,6 +55,9 @@ class StreamingClassifier:
             raise ValueError("num_features required (model has no numFeatures)")
         self.num_features = int(nf)
         self.use_graph = bool(use_graph) and self.device.type == "cuda"
+        self._featurizer = featurizer
+        self._input_col = input_col
+        self._feature_col = feature_col
 
         self._static_in = torch.zeros(
             self.batch_size, self.num_features, dtype=torch.float32, device=self.device
@@ -50,10 +66,25 @@ class StreamingClassifier:
         self._static_raw = None
         self._static_pred = None
         if self.use_graph:
-            self._capture()
+            try:
+                self._capture()
+            except Exception as e:  # noqa: BLE001 - sync-in-capture etc.
+                import warnings
+
+                warnings.warn(
+                    f"hipGraph capture failed ({type(e).__name__}: {e}); "
+                    "falling back to eager micro-batches"
+                )
+                self._graph = None
 
     def _forward(self):
-        raw = self.model.predictRaw(self._static_in)
+        feats = self._static_in
+        if self._featurizer is not None:
+            from flowml.frame import FlowFrame
+
+            df = FlowFrame({self._input_col: self._static_in})
+            feats = self._featurizer.transform(df).col(self._feature_col)
+        raw = self.model.predictRaw(feats)
         pred = raw.argmax(dim=1)
         return raw, pred
 

@@ -723,3 +723,39 @@ class TestRegressionGPU:
         svc = LinearSVC(maxIter=30).fit(df2)
         acc = (svc.transform(df2)["prediction"] == y).double().mean()
         assert float(acc) > 0.99
+
+
+@pytest.mark.gpu
+class TestStreamingFeaturizedGraph:
+    def test_pipeline_capture_matches_eager(self):
+        """Featurization (scaler) captured INSIDE the hipGraph must match the
+        uncaptured eager path bitwise."""
+        from flowml.classification import RandomForestClassifier
+        from flowml.feature import StandardScaler
+        from flowml.frame import FlowFrame
+        from flowml.pipeline import Pipeline
+        from flowml.streaming import StreamingClassifier
+
+        r = np.random.default_rng(50)
+        X = torch.from_numpy(r.normal(size=(40_000, 8)).astype(np.float32) * 5).to(DEV)
+        y = (X[:, 0] > 0).to(torch.float64)
+        df = FlowFrame({"raw": X, "label": y}, device=DEV)
+        pipe = Pipeline(stages=[
+            StandardScaler(inputCol="raw", outputCol="features"),
+            RandomForestClassifier(numTrees=8, maxDepth=4, seed=1),
+        ]).fit(df)
+        scaler, rf = pipe.stages
+        B = 4096
+        sc_graph = StreamingClassifier(rf, B, num_features=8, device=DEV,
+                                       featurizer=scaler, input_col="raw",
+                                       feature_col="features", use_graph=True)
+        sc_eager = StreamingClassifier(rf, B, num_features=8, device=DEV,
+                                       featurizer=scaler, input_col="raw",
+                                       feature_col="features", use_graph=False)
+        assert sc_graph._graph is not None, "capture fell back to eager"
+        for i in range(3):
+            batch = X[i * B : (i + 1) * B]
+            pg, rg = sc_graph.infer(batch)
+            pe, re_ = sc_eager.infer(batch)
+            assert torch.equal(pg.cpu(), pe.cpu())
+            assert torch.equal(rg.cpu(), re_.cpu())
