@@ -82,7 +82,8 @@ class StreamingClassifier:
         if self._featurizer is not None:
             from flowml.frame import FlowFrame
 
-            df = FlowFrame({self._input_col: self._static_in})
+            df = FlowFrame({self._input_col: self._static_in},
+                           device=self.device)
             feats = self._featurizer.transform(df).col(self._feature_col)
         raw = self.model.predictRaw(feats)
         pred = raw.argmax(dim=1)
