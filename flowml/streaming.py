@@ -113,8 +113,8 @@ class StreamingClassifier:
         if self._graph is not None:
             self._graph.replay()
             return self._static_pred, self._static_raw
-        raw = self.model.predictRaw(self._static_in)
-        return raw.argmax(dim=1), raw
+        raw, pred = self._forward()
+        return pred, raw
 
     def infer_stream(self, batches):
         """Generator over a stream of host/device batches (pipelined H2D)."""
