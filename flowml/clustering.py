@@ -228,6 +228,8 @@ class ClusteringEvaluator(HasFeaturesCol, HasPredictionCol, Params):
         a = dataset.col(self.getOrDefault(self.predictionCol)).reshape(-1).to(torch.int64)
         N, F = X.shape
         k = int(a.max()) + 1
+        if k < 2:
+            raise ValueError("silhouette requires at least 2 clusters")
         onehot = torch.zeros(N, k, dtype=torch.float64, device=X.device)
         onehot[torch.arange(N, device=X.device), a] = 1.0
         cnt = onehot.sum(dim=0)                                  # k
