@@ -236,6 +236,9 @@ class GBTRegressor(_TreeEnsembleParams, _PredictorParams, HasMaxIter, Estimator,
         step = self.getOrDefault(self.stepSize)
         max_iter = self.getOrDefault(self.maxIter)
         subsample = self.getOrDefault(self.subsamplingRate)
+        inst_w = None
+        if self.isSet(self.weightCol):
+            inst_w = dataset.col(self.getWeightCol()).reshape(-1).to(torch.float32)
         trees: List[FlatTree] = []
         weights: List[float] = []
         margin = torch.zeros(N, dtype=torch.float32, device=dev)
@@ -255,6 +258,9 @@ class GBTRegressor(_TreeEnsembleParams, _PredictorParams, HasMaxIter, Estimator,
                 n_classes=2,
             )
             bag = poisson_bag(N, 1, subsample, False, cfg.seed, dev) if subsample < 1.0 else None
+            if inst_w is not None:
+                wrow = inst_w.unsqueeze(0)
+                bag = wrow.clone() if bag is None else bag * wrow
             tree_list, row_vals = grow_trees(codes, resid, bag, thresholds, cfg,
                                              return_row_values=True)
             tree = tree_list[0]

@@ -175,3 +175,18 @@ class TestFeatureImportancesEverywhere:
             imp = m.featureImportances.toArray()
             assert imp.shape == (6,)
             assert abs(imp.sum() - 1.0) < 1e-9 or imp.sum() == 0.0
+
+
+class TestGBTRegressorWeightCol:
+    def test_uniform_weight_equals_duplication(self):
+        r = np.random.default_rng(10)
+        X = torch.from_numpy(r.integers(0, 10, (400, 4)).astype(np.float32))
+        y = (X[:, 0] * 2 - X[:, 1]).to(torch.float64)
+        w = torch.full((400,), 2.0)
+        df_w = FlowFrame({"features": X, "label": y, "w": w})
+        idx = torch.repeat_interleave(torch.arange(400), 2)
+        df_d = FlowFrame({"features": X[idx], "label": y[idx]})
+        mw = GBTRegressor(maxIter=4, maxDepth=3, seed=0, weightCol="w").fit(df_w)
+        md = GBTRegressor(maxIter=4, maxDepth=3, seed=0).fit(df_d)
+        assert torch.equal(mw.transform(df_w)["prediction"],
+                           md.transform(df_w)["prediction"])
