@@ -52,10 +52,18 @@ class ParamGridBuilder:
 
 class _ValidatorParams(Params):
     seed: Param[int] = Param("undefined", "seed", "random seed.", TypeConverters.toInt)
+    parallelism: Param[int] = Param(
+        "undefined", "parallelism",
+        "the number of threads to use when running parallel algorithms (>= 1). "
+        "Accepted for Spark API parity; sub-fits run sequentially here — each "
+        "fit already saturates the GPU(s), so thread-level fit parallelism "
+        "would only interleave kernels.",
+        TypeConverters.toInt,
+    )
 
     def __init__(self):
         super().__init__()
-        self._setDefault(seed=42)
+        self._setDefault(seed=42, parallelism=1)
         self.estimator: Optional[Estimator] = None
         self.estimatorParamMaps: List[Dict[Param, Any]] = []
         self.evaluator = None
