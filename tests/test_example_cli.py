@@ -1,0 +1,25 @@
+"""The app-level driver (examples/train_flows.py) — the reference app's
+equivalent — must run end-to-end on CPU for every model family."""
+
+import os
+import subprocess
+import sys
+
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+CLI = os.path.join(REPO, "examples", "train_flows.py")
+
+
+@pytest.mark.parametrize("model", ["lr", "rf", "gbt", "mlp"])
+def test_cli_end_to_end(model, tmp_path):
+    save = str(tmp_path / f"m_{model}")
+    r = subprocess.run(
+        [sys.executable, CLI, "--rows", "3000", "--model", model,
+         "--save", save, "--device", "cpu"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert r.returncode == 0, r.stdout[-800:] + r.stderr[-800:]
+    assert "test accuracy=" in r.stdout
+    assert os.path.isdir(os.path.join(save, "metadata"))
+    assert os.path.isdir(os.path.join(save, "stages"))
