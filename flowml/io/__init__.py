@@ -77,6 +77,8 @@ _register("PCA", "org.apache.spark.ml.feature.PCA")
 _register("PCAModel", "org.apache.spark.ml.feature.PCAModel")
 _register("KMeans", "org.apache.spark.ml.clustering.KMeans")
 _register("KMeansModel", "org.apache.spark.ml.clustering.KMeansModel")
+_register("CrossValidatorModel", "org.apache.spark.ml.tuning.CrossValidatorModel")
+_register("TrainValidationSplitModel", "org.apache.spark.ml.tuning.TrainValidationSplitModel")
 _register("Imputer", "org.apache.spark.ml.feature.Imputer")
 _register("RandomForestRegressor", "org.apache.spark.ml.regression.RandomForestRegressor")
 _register("RandomForestRegressionModel", "org.apache.spark.ml.regression.RandomForestRegressionModel")
@@ -121,9 +123,10 @@ def _our_class(spark_name: str):
     import flowml.classification as cls
     import flowml.regression as reg
     import flowml.clustering as clus
+    import flowml.tuning as tuning
 
     simple = spark_name.rsplit(".", 1)[-1]
-    for mod in (pipeline, feature, cls, reg, clus):
+    for mod in (pipeline, feature, cls, reg, clus, tuning):
         if hasattr(mod, simple):
             return getattr(mod, simple)
     raise ValueError(f"Cannot resolve class for {spark_name!r}")
@@ -291,6 +294,17 @@ def _save_instance(instance, path: str) -> None:
     if cls == "PipelineModel":
         _save_pipeline_stages(instance.stages, path)
         save_metadata(instance, path)
+        return
+    if cls in ("CrossValidatorModel", "TrainValidationSplitModel"):
+        # Spark layout: metadata/ + bestModel/ subdir (we persist the
+        # metrics in data/; the estimator/evaluator/param-maps needed only
+        # for refitting are not persisted — documented divergence)
+        save_metadata(instance, path)
+        _save_instance(instance.bestModel, os.path.join(path, "bestModel"))
+        metrics = (instance.avgMetrics if cls == "CrossValidatorModel"
+                   else instance.validationMetrics)
+        _write_parquet(os.path.join(path, "data"),
+                       _table_from_rows([{"metrics": [float(m) for m in metrics]}]))
         return
     save_metadata(instance, path)
     rows = _data_rows(instance, path)
@@ -485,6 +499,17 @@ def load(path: str):
     spark_class = meta["class"]
     cls = _our_class(spark_class)
     simple = cls.__name__
+
+    if simple in ("CrossValidatorModel", "TrainValidationSplitModel"):
+        best = load(os.path.join(path, "bestModel"))
+        rows = _read_parquet(os.path.join(path, "data")).to_pylist()
+        metrics = [float(m) for m in rows[0]["metrics"]]
+        if simple == "CrossValidatorModel":
+            inst = cls(bestModel=best, avgMetrics=metrics)
+        else:
+            inst = cls(bestModel=best, validationMetrics=metrics)
+        _apply_params(inst, meta)
+        return inst
 
     if simple in ("Pipeline", "PipelineModel"):
         stages_dir = os.path.join(path, "stages")

@@ -138,6 +138,20 @@ class CrossValidatorModel(Model):
     def _transform(self, dataset):
         return self.bestModel.transform(dataset)
 
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str) -> "CrossValidatorModel":
+        from flowml.io import load as _load
+
+        return _load(path)
+
 
 class TrainValidationSplit(_ValidatorParams, Estimator):
     trainRatio: Param[float] = Param(
@@ -182,3 +196,17 @@ class TrainValidationSplitModel(Model):
 
     def _transform(self, dataset):
         return self.bestModel.transform(dataset)
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str) -> "TrainValidationSplitModel":
+        from flowml.io import load as _load
+
+        return _load(path)

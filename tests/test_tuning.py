@@ -121,3 +121,27 @@ class TestSmallerIsBetterMetric:
         m = cv.fit(df)
         assert m.avgMetrics[0] < m.avgMetrics[1]
         assert m.bestModel.getOrDefault(m.bestModel.regParam) == 0.0
+
+
+class TestValidatorModelPersistence:
+    def test_cv_model_roundtrip(self, tmp_path):
+        import numpy as np
+
+        from flowml.classification import LogisticRegression
+        from flowml.evaluation import MulticlassClassificationEvaluator
+        from flowml.tuning import CrossValidator, CrossValidatorModel, ParamGridBuilder
+
+        r = np.random.default_rng(2)
+        X = torch.from_numpy(r.normal(size=(300, 4)).astype(np.float32))
+        y = (X[:, 0] > 0).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y})
+        lr = LogisticRegression(maxIter=15)
+        cv = CrossValidator(estimator=lr,
+                            estimatorParamMaps=ParamGridBuilder().addGrid(lr.regParam, [0.0, 0.1]).build(),
+                            evaluator=MulticlassClassificationEvaluator(metricName="accuracy"),
+                            numFolds=2, seed=3)
+        m = cv.fit(df)
+        m.save(str(tmp_path / "cvm"))
+        lm = CrossValidatorModel.load(str(tmp_path / "cvm"))
+        assert lm.avgMetrics == pytest.approx(m.avgMetrics)
+        assert torch.equal(lm.transform(df)["prediction"], m.transform(df)["prediction"])
