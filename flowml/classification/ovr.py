@@ -78,3 +78,17 @@ class OneVsRestModel(_ClassifierParams, Model):
         if pred_col:
             out = out.withColumn(pred_col, raw.argmax(dim=1).to(torch.float64))
         return out
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str) -> "OneVsRestModel":
+        from flowml.io import load as _load
+
+        return _load(path)

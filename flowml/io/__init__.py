@@ -79,6 +79,7 @@ _register("KMeans", "org.apache.spark.ml.clustering.KMeans")
 _register("KMeansModel", "org.apache.spark.ml.clustering.KMeansModel")
 _register("CrossValidatorModel", "org.apache.spark.ml.tuning.CrossValidatorModel")
 _register("TrainValidationSplitModel", "org.apache.spark.ml.tuning.TrainValidationSplitModel")
+_register("OneVsRestModel", "org.apache.spark.ml.classification.OneVsRestModel")
 _register("Imputer", "org.apache.spark.ml.feature.Imputer")
 _register("RandomForestRegressor", "org.apache.spark.ml.regression.RandomForestRegressor")
 _register("RandomForestRegressionModel", "org.apache.spark.ml.regression.RandomForestRegressionModel")
@@ -295,6 +296,14 @@ def _save_instance(instance, path: str) -> None:
         _save_pipeline_stages(instance.stages, path)
         save_metadata(instance, path)
         return
+    if cls == "OneVsRestModel":
+        # Spark layout: metadata/ + models/<i>_<uid>/ per binary sub-model
+        save_metadata(instance, path)
+        models_dir = os.path.join(path, "models")
+        os.makedirs(models_dir, exist_ok=True)
+        for i, m in enumerate(instance.models):
+            _save_instance(m, os.path.join(models_dir, f"{i}_{m.uid}"))
+        return
     if cls in ("CrossValidatorModel", "TrainValidationSplitModel"):
         # Spark layout: metadata/ + bestModel/ subdir (we persist the
         # metrics in data/; the estimator/evaluator/param-maps needed only
@@ -310,6 +319,16 @@ def _save_instance(instance, path: str) -> None:
     rows = _data_rows(instance, path)
     if rows is not None:
         _write_parquet(os.path.join(path, "data"), _table_from_rows(rows))
+    else:
+        from flowml.base import Model
+
+        if isinstance(instance, Model):
+            # every fitted Model must have an explicit data writer —
+            # silently persisting only the params would lose learned state
+            raise NotImplementedError(
+                f"No persistence writer for fitted model {cls}; "
+                "refusing to save params-only"
+            )
 
 
 def _save_pipeline_stages(stages, path: str) -> None:
@@ -499,6 +518,14 @@ def load(path: str):
     spark_class = meta["class"]
     cls = _our_class(spark_class)
     simple = cls.__name__
+
+    if simple == "OneVsRestModel":
+        models_dir = os.path.join(path, "models")
+        entries = sorted(os.listdir(models_dir), key=lambda s: int(s.split("_", 1)[0]))
+        models = [load(os.path.join(models_dir, e)) for e in entries]
+        inst = cls(models=models, numClasses=len(models))
+        _apply_params(inst, meta)
+        return inst
 
     if simple in ("CrossValidatorModel", "TrainValidationSplitModel"):
         best = load(os.path.join(path, "bestModel"))

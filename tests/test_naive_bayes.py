@@ -163,3 +163,18 @@ class TestLinearSVC:
         lm = LinearSVCModel.load(str(tmp_path / "svc"))
         assert np.allclose(lm.coefficients.toArray(), m.coefficients.toArray())
         assert lm.intercept == pytest.approx(m.intercept)
+
+    def test_ovr_model_persistence(self, tmp_path):
+        from flowml.classification import LinearSVC, OneVsRest, OneVsRestModel
+
+        r = np.random.default_rng(20)
+        X = torch.from_numpy(r.normal(size=(300, 4)).astype(np.float32))
+        y = torch.from_numpy(r.integers(0, 3, 300).astype(np.float64))
+        X[y == 1, 0] += 3
+        X[y == 2, 1] -= 3
+        df = FlowFrame({"features": X, "label": y})
+        m = OneVsRest(classifier=LinearSVC(maxIter=20)).fit(df)
+        m.save(str(tmp_path / "ovr"))
+        lm = OneVsRestModel.load(str(tmp_path / "ovr"))
+        assert len(lm.models) == 3
+        assert torch.equal(lm.transform(df)["prediction"], m.transform(df)["prediction"])
