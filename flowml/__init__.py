@@ -27,6 +27,7 @@ from flowml.base import Estimator, Transformer, Model
 from flowml.pipeline import Pipeline, PipelineModel
 from flowml.frame import FlowFrame, read_csv
 from flowml import linalg
+from flowml import classification, clustering, evaluation, feature, regression, tuning  # noqa: F401 (namespace re-exports)
 
 __all__ = [
     "Param",
