@@ -10,10 +10,14 @@
 //                         (nodeid updated in place) and computes BaggedPoint
 //                         weights from a splitmix64 hash (no bag tensor).
 //                         The #1 hot loop of the flagship config.
-// K2b  tree_hist_reg    — same with (Σt, Σt², n) stats for GBT variance trees
+// K2b  tree_hist_reg    — regression variant: 2-wide (Σwt, Σw) cells + one
+//                         per-node Σwt² slot (variance gain needs no per-bin
+//                         Σt²) for GBT trees
 // K4b  node_advance     — standalone advance (CPU-parity and GBT paths)
 // K8   tree_predict     — batched ensemble inference: sum of leaf vectors
 // K1   logreg_delta     — fused logits+softmax+delta+loss for LogisticRegression
+// K3   split_scan        — classification split-gain scan (packed rows)
+// K3b  split_scan_reg    — variance scan on the 2-wide regression layout
 // K6   softmax_xent     — row-wise softmax/cross-entropy loss + delta
 //
 // Node identifiers are GLOBAL arena indices (one struct-of-arrays node store
