@@ -363,7 +363,12 @@ def from_arrow(table, device=None) -> FlowFrame:
             cols[name] = torch.from_numpy(np.ascontiguousarray(np_arr))
         else:
             np_arr = combined.to_numpy(zero_copy_only=False)
-            cols[name] = torch.from_numpy(np.ascontiguousarray(np_arr))
+            # pyarrow hands back a read-only view; copy so torch owns
+            # writable memory (silences the non-writable-tensor warning)
+            cols[name] = torch.from_numpy(
+                np.ascontiguousarray(np_arr).copy()
+                if not np_arr.flags.writeable else np.ascontiguousarray(np_arr)
+            )
     ff = FlowFrame(cols)
     if device is not None:
         ff = ff.to(device)
