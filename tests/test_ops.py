@@ -241,5 +241,5 @@ class TestMlpGrad:
         h = torch.sigmoid(X @ W1.T + b1)
         ref = torch.nn.functional.cross_entropy(h @ W2.T + b2, y, reduction="sum")
         ref.backward()
-        assert float(loss) == pytest.approx(float(ref), rel=1e-4)
+        assert float(loss) == pytest.approx(float(ref.detach()), rel=1e-4)
         assert torch.allclose(gWs[0], W1.grad, atol=1e-4)
