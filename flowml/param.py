@@ -151,6 +151,10 @@ class Params:
             key=lambda p: p.name,
         )
 
+    def __repr__(self) -> str:
+        # pyspark shows the uid (e.g. "LogisticRegression_4db9da9f2932")
+        return self.uid
+
     def hasParam(self, paramName: str) -> bool:
         p = getattr(self, paramName, None)
         return isinstance(p, Param)
