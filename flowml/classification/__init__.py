@@ -26,6 +26,7 @@ from flowml.classification.gbt import GBTClassifier, GBTClassificationModel
 from flowml.classification.naive_bayes import NaiveBayes, NaiveBayesModel
 from flowml.classification.ovr import OneVsRest, OneVsRestModel
 from flowml.classification.svc import LinearSVC, LinearSVCModel
+from flowml.fm import FMClassifier, FMClassificationModel
 
 __all__ = [
     "Classifier",
@@ -48,4 +49,6 @@ __all__ = [
     "OneVsRestModel",
     "LinearSVC",
     "LinearSVCModel",
+    "FMClassifier",
+    "FMClassificationModel",
 ]

@@ -80,6 +80,10 @@ _register("KMeansModel", "org.apache.spark.ml.clustering.KMeansModel")
 _register("CrossValidatorModel", "org.apache.spark.ml.tuning.CrossValidatorModel")
 _register("TrainValidationSplitModel", "org.apache.spark.ml.tuning.TrainValidationSplitModel")
 _register("OneVsRestModel", "org.apache.spark.ml.classification.OneVsRestModel")
+_register("FMClassifier", "org.apache.spark.ml.classification.FMClassifier")
+_register("FMClassificationModel", "org.apache.spark.ml.classification.FMClassificationModel")
+_register("FMRegressor", "org.apache.spark.ml.regression.FMRegressor")
+_register("FMRegressionModel", "org.apache.spark.ml.regression.FMRegressionModel")
 _register("Imputer", "org.apache.spark.ml.feature.Imputer")
 _register("RandomForestRegressor", "org.apache.spark.ml.regression.RandomForestRegressor")
 _register("RandomForestRegressionModel", "org.apache.spark.ml.regression.RandomForestRegressionModel")
@@ -125,9 +129,10 @@ def _our_class(spark_name: str):
     import flowml.regression as reg
     import flowml.clustering as clus
     import flowml.tuning as tuning
+    import flowml.fm as fm
 
     simple = spark_name.rsplit(".", 1)[-1]
-    for mod in (pipeline, feature, cls, reg, clus, tuning):
+    for mod in (pipeline, feature, cls, reg, clus, tuning, fm):
         if hasattr(mod, simple):
             return getattr(mod, simple)
     raise ValueError(f"Cannot resolve class for {spark_name!r}")
@@ -467,6 +472,14 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "explainedVariance": vector_struct(instance._ev.numpy()),
             }
         ]
+    if cls in ("FMClassificationModel", "FMRegressionModel"):
+        return [
+            {
+                "intercept": float(instance.intercept),
+                "linear": vector_struct(instance._linear.to(torch.float64).numpy()),
+                "factors": matrix_struct(instance._factors.to(torch.float64).numpy()),
+            }
+        ]
     if cls == "LinearSVCModel":
         return [
             {
@@ -608,6 +621,12 @@ def _construct(cls, simple: str, rows, path: str, meta):
         r0 = rows[0]
         return cls(pc=torch.from_numpy(matrix_from_struct(r0["pc"])),
                    explainedVariance=torch.from_numpy(vector_from_struct(r0["explainedVariance"])))
+    if simple in ("FMClassificationModel", "FMRegressionModel"):
+        r0 = rows[0]
+        lin = torch.from_numpy(vector_from_struct(r0["linear"])).to(torch.float32)
+        fac = torch.from_numpy(matrix_from_struct(r0["factors"])).to(torch.float32)
+        return cls(intercept=float(r0["intercept"]), linear=lin, factors=fac,
+                   numFeatures=int(lin.numel()))
     if simple == "LinearSVCModel":
         r0 = rows[0]
         coef = torch.from_numpy(vector_from_struct(r0["coefficients"])).to(torch.float32)
