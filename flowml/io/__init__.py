@@ -84,6 +84,8 @@ _register("FMClassifier", "org.apache.spark.ml.classification.FMClassifier")
 _register("FMClassificationModel", "org.apache.spark.ml.classification.FMClassificationModel")
 _register("FMRegressor", "org.apache.spark.ml.regression.FMRegressor")
 _register("FMRegressionModel", "org.apache.spark.ml.regression.FMRegressionModel")
+_register("IsotonicRegression", "org.apache.spark.ml.regression.IsotonicRegression")
+_register("IsotonicRegressionModel", "org.apache.spark.ml.regression.IsotonicRegressionModel")
 _register("Imputer", "org.apache.spark.ml.feature.Imputer")
 _register("RandomForestRegressor", "org.apache.spark.ml.regression.RandomForestRegressor")
 _register("RandomForestRegressionModel", "org.apache.spark.ml.regression.RandomForestRegressionModel")
@@ -472,6 +474,13 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "explainedVariance": vector_struct(instance._ev.numpy()),
             }
         ]
+    if cls == "IsotonicRegressionModel":
+        return [
+            {
+                "boundaries": vector_struct(instance._bounds.numpy()),
+                "predictions": vector_struct(instance._preds.numpy()),
+            }
+        ]
     if cls in ("FMClassificationModel", "FMRegressionModel"):
         return [
             {
@@ -621,6 +630,10 @@ def _construct(cls, simple: str, rows, path: str, meta):
         r0 = rows[0]
         return cls(pc=torch.from_numpy(matrix_from_struct(r0["pc"])),
                    explainedVariance=torch.from_numpy(vector_from_struct(r0["explainedVariance"])))
+    if simple == "IsotonicRegressionModel":
+        r0 = rows[0]
+        return cls(boundaries=torch.from_numpy(vector_from_struct(r0["boundaries"])),
+                   predictions=torch.from_numpy(vector_from_struct(r0["predictions"])))
     if simple in ("FMClassificationModel", "FMRegressionModel"):
         r0 = rows[0]
         lin = torch.from_numpy(vector_from_struct(r0["linear"])).to(torch.float32)

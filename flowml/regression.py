@@ -522,3 +522,133 @@ class LinearRegressionTrainingSummary:
     @property
     def r2(self):
         return self._eval("r2")
+
+
+# ---------------------------------------------------------------------------
+# IsotonicRegression (spark:.../ml/regression/IsotonicRegression.scala)
+# ---------------------------------------------------------------------------
+
+
+class IsotonicRegression(_PredictorParams, HasWeightCol, Estimator, _RegressorIOMixin):
+    """Weighted isotonic (or antitonic) regression by pool-adjacent-violators;
+    prediction linearly interpolates between fitted boundaries (Spark
+    semantics, incl. the featureIndex selector for vector features)."""
+
+    isotonic: Param[bool] = Param(
+        "undefined", "isotonic",
+        "whether the output sequence should be isotonic/increasing (true) "
+        "or antitonic/decreasing (false)", TypeConverters.toBoolean,
+    )
+    featureIndex: Param[int] = Param(
+        "undefined", "featureIndex",
+        "the index of the feature if featuresCol is a vector column",
+        TypeConverters.toInt,
+    )
+
+    def __init__(self, **kwargs):
+        super().__init__()
+        self._setDefault(isotonic=True, featureIndex=0)
+        self._set(**kwargs)
+
+    def _x_col(self, dataset: FlowFrame) -> torch.Tensor:
+        x = dataset.col(self.getFeaturesCol())
+        if isinstance(x, torch.Tensor) and x.dim() == 2:
+            x = x[:, self.getOrDefault(self.featureIndex)]
+        return x.reshape(-1).to(torch.float64)
+
+    def _fit(self, dataset: FlowFrame) -> "IsotonicRegressionModel":
+        import numpy as np
+
+        x = self._x_col(dataset).cpu().numpy()
+        y = _label_tensor(dataset, self.getLabelCol()).to(torch.float64).cpu().numpy()
+        if self.isSet(self.weightCol):
+            w = dataset.col(self.getWeightCol()).reshape(-1).to(torch.float64).cpu().numpy()
+        else:
+            w = np.ones_like(y)
+        inc = self.getOrDefault(self.isotonic)
+        order = np.argsort(x, kind="stable")
+        xs, ys, ws = x[order], y[order], w[order]
+        if not inc:
+            ys = -ys
+        # PAVA with a block stack: (sum_wy, sum_w, x_lo, x_hi)
+        blocks: list = []
+        for xi, yi, wi in zip(xs, ys, ws):
+            blocks.append([yi * wi, wi, xi, xi])
+            while len(blocks) > 1 and (
+                blocks[-2][0] / blocks[-2][1] >= blocks[-1][0] / blocks[-1][1]
+            ):
+                b = blocks.pop()
+                blocks[-1][0] += b[0]
+                blocks[-1][1] += b[1]
+                blocks[-1][3] = b[3]
+            # merge equal-x runs so boundaries stay strictly increasing
+            while len(blocks) > 1 and blocks[-2][3] >= blocks[-1][2]:
+                b = blocks.pop()
+                blocks[-1][0] += b[0]
+                blocks[-1][1] += b[1]
+                blocks[-1][3] = b[3]
+        bounds, preds = [], []
+        for swy, sw, lo, hi in blocks:
+            v = swy / sw
+            bounds.append(lo)
+            preds.append(v)
+            if hi != lo:
+                bounds.append(hi)
+                preds.append(v)
+        if not inc:
+            preds = [-v for v in preds]
+        model = IsotonicRegressionModel(
+            boundaries=torch.tensor(bounds, dtype=torch.float64),
+            predictions=torch.tensor(preds, dtype=torch.float64),
+        )
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class IsotonicRegressionModel(HasWeightCol, RegressionModel, _RegressorIOMixin):
+    isotonic = IsotonicRegression.isotonic
+    featureIndex = IsotonicRegression.featureIndex
+
+    def __init__(self, boundaries: Optional[torch.Tensor] = None,
+                 predictions: Optional[torch.Tensor] = None):
+        super().__init__()
+        self._bounds = boundaries if boundaries is not None else torch.zeros(0, dtype=torch.float64)
+        self._preds = predictions if predictions is not None else torch.zeros(0, dtype=torch.float64)
+
+    @property
+    def boundaries(self):
+        from flowml.linalg import DenseVector
+
+        return DenseVector(self._bounds.numpy())
+
+    @property
+    def predictions(self):
+        from flowml.linalg import DenseVector
+
+        return DenseVector(self._preds.numpy())
+
+    def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
+        x = features
+        if x.dim() == 2:
+            x = x[:, self.getOrDefault(self.featureIndex)]
+        x = x.reshape(-1).to(torch.float64)
+        b = self._bounds.to(x.device)
+        p = self._preds.to(x.device)
+        if len(b) == 0:
+            return torch.zeros_like(x)
+        if len(b) == 1:
+            return p[0].expand_as(x).clone()
+        idx = torch.searchsorted(b, x).clamp(1, len(b) - 1)
+        lo, hi = b[idx - 1], b[idx]
+        plo, phi = p[idx - 1], p[idx]
+        t = torch.where(hi > lo, (x - lo) / (hi - lo), torch.zeros_like(x))
+        out = plo + t.clamp(0, 1) * (phi - plo)
+        out = torch.where(x <= b[0], p[0], out)
+        out = torch.where(x >= b[-1], p[-1], out)
+        return out
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getFeaturesCol())
+        pred = self.predictRaw(x if isinstance(x, torch.Tensor) else torch.as_tensor(x))
+        return dataset.withColumn(self.getOrDefault(self.predictionCol), pred)

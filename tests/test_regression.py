@@ -190,3 +190,41 @@ class TestGBTRegressorWeightCol:
         md = GBTRegressor(maxIter=4, maxDepth=3, seed=0).fit(df_d)
         assert torch.equal(mw.transform(df_w)["prediction"],
                            md.transform(df_w)["prediction"])
+
+
+class TestIsotonicRegression:
+    def test_matches_sklearn(self, tmp_path):
+        from sklearn.isotonic import IsotonicRegression as SkIso
+
+        from flowml.regression import IsotonicRegression, IsotonicRegressionModel
+
+        r = np.random.default_rng(0)
+        x = np.sort(r.uniform(0, 10, 300)).astype(np.float32)
+        y = (np.log1p(x) + r.normal(scale=0.2, size=300)).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(x),
+                        "label": torch.from_numpy(y)})
+        m = IsotonicRegression().fit(df)
+        ours = m.transform(df)["prediction"].numpy()
+        sk = SkIso(out_of_bounds="clip").fit(x, y)
+        theirs = sk.predict(x)
+        assert np.allclose(ours, theirs, atol=1e-6)
+        # monotone
+        assert (np.diff(ours) >= -1e-12).all()
+        m.save(str(tmp_path / "iso"))
+        lm = IsotonicRegressionModel.load(str(tmp_path / "iso"))
+        assert np.allclose(lm.transform(df)["prediction"].numpy(), ours)
+
+    def test_antitonic_and_vector_feature(self):
+        from flowml.regression import IsotonicRegression
+
+        r = np.random.default_rng(1)
+        x = r.uniform(0, 5, 200).astype(np.float32)
+        X = np.stack([r.normal(size=200).astype(np.float32), x], axis=1)
+        y = (-2.0 * x + r.normal(scale=0.1, size=200)).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": torch.from_numpy(y)})
+        m = IsotonicRegression(isotonic=False, featureIndex=1).fit(df)
+        pred = m.transform(df)["prediction"].numpy()
+        order = np.argsort(x)
+        assert (np.diff(pred[order]) <= 1e-9).all()  # decreasing in x
+        corr = np.corrcoef(pred, y)[0, 1]
+        assert corr > 0.95
