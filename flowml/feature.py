@@ -984,3 +984,60 @@ class PCAModel(_PCAParams, Model):
         from flowml.io import load as _load
 
         return _load(path)
+
+
+# ---------------------------------------------------------------------------
+# VectorSlicer / Interaction (spark:.../feature/VectorSlicer.scala, Interaction.scala)
+# ---------------------------------------------------------------------------
+
+
+class VectorSlicer(HasInputCol, HasOutputCol, Transformer):
+    """Select a sub-vector by indices (Spark's name-based selection is not
+    supported — our vector columns carry no attribute metadata)."""
+
+    indices: Param[list] = Param(
+        "undefined", "indices",
+        "An array of indices to select features from a vector column.",
+        TypeConverters.toListInt,
+    )
+
+    def __init__(self, *, inputCol: Optional[str] = None, outputCol: Optional[str] = None,
+                 indices: Optional[list] = None):
+        super().__init__()
+        self._set(inputCol=inputCol, outputCol=outputCol, indices=indices)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getInputCol())
+        if not isinstance(x, torch.Tensor) or x.dim() != 2:
+            raise TypeError("VectorSlicer input must be a vector column")
+        idx = self.getOrDefault(self.indices)
+        if not idx:
+            raise ValueError("VectorSlicer requires non-empty indices")
+        sel = torch.tensor(idx, dtype=torch.int64, device=x.device)
+        return dataset.withColumn(self.getOutputCol(), x.index_select(1, sel))
+
+
+class Interaction(HasInputCols, HasOutputCol, Transformer):
+    """Elementwise product-expansion of the input columns: the output vector
+    holds every cross-product combination of one element from each input
+    (scalars are width-1 vectors) — Spark's Interaction transformer."""
+
+    def __init__(self, *, inputCols: Optional[list] = None, outputCol: Optional[str] = None):
+        super().__init__()
+        self._set(inputCols=inputCols, outputCol=outputCol)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        cols = self.getInputCols()
+        if not cols or len(cols) < 2:
+            raise ValueError("Interaction requires at least two inputCols")
+        mats = []
+        for c in cols:
+            t = dataset.col(c)
+            if isinstance(t, np.ndarray):
+                raise TypeError(f"Interaction input {c!r} must be numeric")
+            mats.append((t if t.dim() == 2 else t.reshape(-1, 1)).to(torch.float32))
+        out = mats[0]
+        for t in mats[1:]:
+            # (N, a) x (N, b) -> (N, a*b)
+            out = (out.unsqueeze(2) * t.unsqueeze(1)).reshape(out.shape[0], -1)
+        return dataset.withColumn(self.getOutputCol(), out)

@@ -75,6 +75,8 @@ _register("Bucketizer", "org.apache.spark.ml.feature.Bucketizer")
 _register("QuantileDiscretizer", "org.apache.spark.ml.feature.QuantileDiscretizer")
 _register("PCA", "org.apache.spark.ml.feature.PCA")
 _register("PCAModel", "org.apache.spark.ml.feature.PCAModel")
+_register("VectorSlicer", "org.apache.spark.ml.feature.VectorSlicer")
+_register("Interaction", "org.apache.spark.ml.feature.Interaction")
 _register("KMeans", "org.apache.spark.ml.clustering.KMeans")
 _register("KMeansModel", "org.apache.spark.ml.clustering.KMeansModel")
 _register("CrossValidatorModel", "org.apache.spark.ml.tuning.CrossValidatorModel")

@@ -311,3 +311,28 @@ class TestPCA:
         m.save(str(tmp_path / "pca"))
         lm = PCAModel.load(str(tmp_path / "pca"))
         assert np.allclose(lm.transform(df)["p"].numpy(), ours)
+
+
+class TestVectorSlicerAndInteraction:
+    def test_slicer(self):
+        from flowml.feature import VectorSlicer
+
+        X = torch.arange(12, dtype=torch.float32).reshape(3, 4)
+        df = FlowFrame({"v": X})
+        out = VectorSlicer(inputCol="v", outputCol="s", indices=[3, 1]).transform(df)
+        assert torch.equal(out["s"], X[:, [3, 1]])
+
+    def test_interaction(self):
+        from flowml.feature import Interaction
+
+        a = torch.tensor([[1.0, 2.0], [3.0, 4.0]])
+        b = torch.tensor([10.0, 100.0])
+        df = FlowFrame({"a": a, "b": b})
+        out = Interaction(inputCols=["a", "b"], outputCol="i").transform(df)
+        assert torch.equal(out["i"], torch.tensor([[10.0, 20.0], [300.0, 400.0]]))
+        # three-way: widths multiply
+        c = torch.tensor([[1.0, -1.0], [2.0, 0.5]])
+        df2 = FlowFrame({"a": a, "b": b, "c": c})
+        out2 = Interaction(inputCols=["a", "b", "c"], outputCol="i").transform(df2)
+        assert out2["i"].shape == (2, 4)
+        assert torch.allclose(out2["i"][0], torch.tensor([10.0, -10.0, 20.0, -20.0]))
