@@ -336,3 +336,20 @@ class TestVectorSlicerAndInteraction:
         out2 = Interaction(inputCols=["a", "b", "c"], outputCol="i").transform(df2)
         assert out2["i"].shape == (2, 4)
         assert torch.allclose(out2["i"][0], torch.tensor([10.0, -10.0, 20.0, -20.0]))
+
+
+class TestStringOrderTypes:
+    def test_all_four_orders(self):
+        from flowml.feature import StringIndexer
+
+        vals = ["b", "b", "b", "a", "c", "c"]  # freq: b=3, c=2, a=1
+        df = FlowFrame({"s": np.array(vals, dtype=object)})
+
+        def labels(order):
+            return StringIndexer(inputCol="s", outputCol="i",
+                                 stringOrderType=order).fit(df).labels
+
+        assert labels("frequencyDesc") == ["b", "c", "a"]
+        assert labels("frequencyAsc") == ["a", "c", "b"]
+        assert labels("alphabetAsc") == ["a", "b", "c"]
+        assert labels("alphabetDesc") == ["c", "b", "a"]
