@@ -202,3 +202,38 @@ class TestThresholds:
         import pytest
         with pytest.raises(ValueError):
             m.transform(df)
+
+
+class TestWeightedEvaluators:
+    def test_weighted_accuracy_equals_duplication(self):
+        import torch
+
+        from flowml.evaluation import MulticlassClassificationEvaluator
+
+        pred = torch.tensor([0.0, 1.0, 1.0, 0.0])
+        lab = torch.tensor([0.0, 1.0, 0.0, 0.0])
+        w = torch.tensor([3.0, 1.0, 2.0, 1.0])
+        df_w = FlowFrame({"prediction": pred, "label": lab, "w": w})
+        idx = torch.repeat_interleave(torch.arange(4), w.to(torch.int64))
+        df_d = FlowFrame({"prediction": pred[idx], "label": lab[idx]})
+        ev_w = MulticlassClassificationEvaluator(metricName="accuracy", weightCol="w")
+        ev = MulticlassClassificationEvaluator(metricName="accuracy")
+        assert ev_w.evaluate(df_w) == pytest.approx(ev.evaluate(df_d))
+        f1_w = MulticlassClassificationEvaluator(metricName="f1", weightCol="w").evaluate(df_w)
+        f1_d = MulticlassClassificationEvaluator(metricName="f1").evaluate(df_d)
+        assert f1_w == pytest.approx(f1_d)
+
+    def test_weighted_rmse_equals_duplication(self):
+        import torch
+
+        from flowml.evaluation import RegressionEvaluator
+
+        pred = torch.tensor([1.0, 2.0, 4.0])
+        lab = torch.tensor([1.5, 2.0, 3.0])
+        w = torch.tensor([2.0, 1.0, 3.0])
+        df_w = FlowFrame({"prediction": pred, "label": lab, "w": w})
+        idx = torch.repeat_interleave(torch.arange(3), w.to(torch.int64))
+        df_d = FlowFrame({"prediction": pred[idx], "label": lab[idx]})
+        a = RegressionEvaluator(metricName="rmse", weightCol="w").evaluate(df_w)
+        b = RegressionEvaluator(metricName="rmse").evaluate(df_d)
+        assert a == pytest.approx(b)
