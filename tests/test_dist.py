@@ -197,3 +197,16 @@ def _kmeans_dist_body(rank, world):
 def test_kmeans_dist_ranks_agree():
     centers = _run_dist("_kmeans_dist_body")
     assert centers.shape[0] == 3
+
+
+def test_shard_range_covers_and_partitions():
+    from flowml.dist import shard_range
+
+    for n in (0, 1, 7, 100, 101, 1000003):
+        for world in (1, 2, 3, 4, 7, 8):
+            spans = [shard_range(n, r, world) for r in range(world)]
+            assert spans[0][0] == 0 and spans[-1][1] == n
+            for (a, b), (c, d) in zip(spans, spans[1:]):
+                assert b == c  # contiguous, disjoint
+            sizes = [b - a for a, b in spans]
+            assert max(sizes) - min(sizes) <= 1  # balanced
