@@ -210,3 +210,30 @@ def test_shard_range_covers_and_partitions():
                 assert b == c  # contiguous, disjoint
             sizes = [b - a for a, b in spans]
             assert max(sizes) - min(sizes) <= 1  # balanced
+
+
+def _linreg_dist_body(rank, world):
+    from flowml import dist as fd
+    from flowml.frame import FlowFrame
+    from flowml.regression import LinearRegression
+
+    X, y = _make_data()
+    yr = (X[:, 0] * 2 - X[:, 2]).astype(np.float64)
+    lo, hi = fd.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi]),
+                    "label": torch.from_numpy(yr[lo:hi])})
+    m = LinearRegression(maxIter=60).fit(df)
+    return np.concatenate([m.coefficients.toArray(), [m.intercept]])
+
+
+def test_linear_regression_dist_equals_single():
+    out = _run_dist("_linreg_dist_body")
+    from flowml.frame import FlowFrame
+    from flowml.regression import LinearRegression
+
+    X, y = _make_data()
+    yr = (X[:, 0] * 2 - X[:, 2]).astype(np.float64)
+    m = LinearRegression(maxIter=60).fit(
+        FlowFrame({"features": torch.from_numpy(X), "label": torch.from_numpy(yr)}))
+    expect = np.concatenate([m.coefficients.toArray(), [m.intercept]])
+    assert np.allclose(out, expect, atol=1e-3)
