@@ -237,3 +237,30 @@ def test_linear_regression_dist_equals_single():
         FlowFrame({"features": torch.from_numpy(X), "label": torch.from_numpy(yr)}))
     expect = np.concatenate([m.coefficients.toArray(), [m.intercept]])
     assert np.allclose(out, expect, atol=1e-3)
+
+
+def _nb_dist_body(rank, world):
+    from flowml import dist as fd
+    from flowml.classification import NaiveBayes
+    from flowml.frame import FlowFrame
+
+    X, y = _make_data()
+    Xp = np.abs(X)
+    lo, hi = fd.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(Xp[lo:hi]),
+                    "label": torch.from_numpy(y[lo:hi].astype(np.float64))})
+    m = NaiveBayes().fit(df)
+    return np.concatenate([m.pi.toArray(), m.theta.toArray().ravel()])
+
+
+def test_naive_bayes_dist_equals_single():
+    out = _run_dist("_nb_dist_body")
+    from flowml.classification import NaiveBayes
+    from flowml.frame import FlowFrame
+
+    X, y = _make_data()
+    Xp = np.abs(X)
+    m = NaiveBayes().fit(FlowFrame({"features": torch.from_numpy(Xp),
+                                    "label": torch.from_numpy(y.astype(np.float64))}))
+    expect = np.concatenate([m.pi.toArray(), m.theta.toArray().ravel()])
+    assert np.allclose(out, expect, atol=1e-10)
