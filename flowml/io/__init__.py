@@ -446,18 +446,18 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
             {
                 "numClasses": int(instance.numClasses),
                 "numFeatures": int(instance.numFeatures),
-                "interceptVector": vector_struct(instance._b.to(torch.float64).numpy()),
-                "coefficientMatrix": matrix_struct(instance._W.to(torch.float64).numpy()),
+                "interceptVector": vector_struct(instance._b.to(torch.float64).cpu().numpy()),
+                "coefficientMatrix": matrix_struct(instance._W.to(torch.float64).cpu().numpy()),
                 "isMultinomial": bool(instance.isMultinomial),
             }
         ]
     if cls == "NaiveBayesModel":
         return [
             {
-                "pi": vector_struct(instance._pi.to(torch.float64).numpy()),
-                "theta": matrix_struct(instance._theta.to(torch.float64).numpy()),
+                "pi": vector_struct(instance._pi.to(torch.float64).cpu().numpy()),
+                "theta": matrix_struct(instance._theta.to(torch.float64).cpu().numpy()),
                 "sigma": matrix_struct(
-                    instance._sigma.to(torch.float64).numpy().reshape(
+                    instance._sigma.to(torch.float64).cpu().numpy().reshape(
                         instance._theta.shape if instance._sigma.numel() else (0, 0))),
                 "modelType": instance._mtype,
             }
@@ -487,27 +487,27 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
         return [
             {
                 "intercept": float(instance.intercept),
-                "linear": vector_struct(instance._linear.to(torch.float64).numpy()),
-                "factors": matrix_struct(instance._factors.to(torch.float64).numpy()),
+                "linear": vector_struct(instance._linear.to(torch.float64).cpu().numpy()),
+                "factors": matrix_struct(instance._factors.to(torch.float64).cpu().numpy()),
             }
         ]
     if cls == "LinearSVCModel":
         return [
             {
                 "intercept": float(instance.intercept),
-                "coefficients": vector_struct(instance._coef.to(torch.float64).numpy()),
+                "coefficients": vector_struct(instance._coef.to(torch.float64).cpu().numpy()),
             }
         ]
     if cls == "LinearRegressionModel":
         return [
             {
                 "intercept": float(instance.intercept),
-                "coefficients": vector_struct(instance._coef.to(torch.float64).numpy()),
+                "coefficients": vector_struct(instance._coef.to(torch.float64).cpu().numpy()),
                 "scale": float(instance.scale),
             }
         ]
     if cls == "MultilayerPerceptronClassificationModel":
-        return [{"weights": vector_struct(instance._weights.to(torch.float64).numpy())}]
+        return [{"weights": vector_struct(instance._weights.to(torch.float64).cpu().numpy())}]
     if cls in ("RandomForestClassificationModel", "GBTClassificationModel",
                "RandomForestRegressionModel", "GBTRegressionModel",
                "DecisionTreeRegressionModel"):
