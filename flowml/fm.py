@@ -3,7 +3,8 @@ regression/FMRegressor.scala).
 
 Model: ŷ(x) = w₀ + Σᵢ wᵢxᵢ + ½ Σ_f [(Σᵢ v_{if} xᵢ)² − Σᵢ v_{if}² xᵢ²]
 (the O(N·F·k) pairwise-interaction identity). Spark trains with mini-batch
-gradient descent / adamW on the shard; here the same adamW runs FULL-batch —
+gradient descent / adamW on the shard; here Adam (with coupled L2 — the
+regParam enters through the gradient) runs FULL-batch —
 the gradient is a handful of GEMMs over the resident shard (rocBLAS on GPU),
 all-reduced across ranks like every other fit (SURVEY.md §2.5 C1 pattern).
 Binary classification uses the logistic loss on ±1 labels, regression the
@@ -46,7 +47,7 @@ class _FMParams(HasMaxIter, HasSeed, HasTol):
         TypeConverters.toFloat,
     )
     stepSize: Param[float] = Param(
-        "undefined", "stepSize", "initial step size for adamW (> 0)",
+        "undefined", "stepSize", "initial step size for the Adam solver (> 0)",
         TypeConverters.toFloat,
     )
     initStd: Param[float] = Param(
@@ -67,7 +68,7 @@ def _fm_raw(X: torch.Tensor, w0: torch.Tensor, w: torch.Tensor,
 
 
 def _fit_fm(est, X: torch.Tensor, target: torch.Tensor, loss_kind: str):
-    """Shared adamW loop. target: ±1 (logistic) or y (squared)."""
+    """Shared full-batch Adam loop. target: ±1 (logistic) or y (squared)."""
     dev = X.device
     N, F = X.shape
     k = est.getOrDefault(est.factorSize)
@@ -132,7 +133,7 @@ def _fit_fm(est, X: torch.Tensor, target: torch.Tensor, loss_kind: str):
 
 
 class FMClassifier(_FMParams, ProbabilisticClassifier):
-    """Binary FM classifier (logistic loss on ±1 labels, adamW)."""
+    """Binary FM classifier (logistic loss on ±1 labels, Adam)."""
 
     def __init__(self, **kwargs):
         super().__init__()
@@ -210,7 +211,7 @@ from flowml.regression import RegressionModel, _RegressorIOMixin  # noqa: E402
 
 
 class FMRegressor(_FMParams, _PredictorParams, Estimator, _RegressorIOMixin):
-    """FM regressor (squared loss, adamW)."""
+    """FM regressor (squared loss, Adam)."""
 
     def __init__(self, **kwargs):
         super().__init__()
