@@ -85,7 +85,7 @@ def _fit_fm(est, X: torch.Tensor, target: torch.Tensor, loss_kind: str):
     n_total = torch.tensor([float(N)], dtype=torch.float64, device=dev)
     dist.all_reduce_(n_total)
     n_all = float(n_total)
-    # adamW state
+    # Adam state
     params = [w0, w, V]
     m = [torch.zeros_like(p) for p in params]
     v = [torch.zeros_like(p) for p in params]
