@@ -94,6 +94,18 @@ class _StringIndexerParams(HasInputCol, HasInputCols, HasOutputCol, HasOutputCol
 
 
 class StringIndexer(_StringIndexerParams, Estimator):
+    """Map string labels to double indices by descending frequency.
+
+    >>> import numpy as np
+    >>> from flowml.frame import FlowFrame
+    >>> df = FlowFrame({"s": np.array(["b", "a", "b", "c", "b", "a"], dtype=object)})
+    >>> m = StringIndexer(inputCol="s", outputCol="i").fit(df)
+    >>> m.labels
+    ['b', 'a', 'c']
+    >>> m.transform(df)["i"].tolist()
+    [0.0, 1.0, 0.0, 2.0, 0.0, 1.0]
+    """
+
     """Label → index by frequency (SURVEY.md §2.2; spark:.../feature/StringIndexer.scala)."""
 
     def __init__(

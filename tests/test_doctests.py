@@ -3,12 +3,13 @@
 import doctest
 
 import flowml.evaluation
+import flowml.feature
 import flowml.linalg
 import flowml.pipeline
 import flowml.tuning
 
 
 def test_module_doctests():
-    for mod in (flowml.pipeline, flowml.tuning, flowml.evaluation, flowml.linalg):
+    for mod in (flowml.pipeline, flowml.tuning, flowml.evaluation, flowml.linalg, flowml.feature):
         results = doctest.testmod(mod, verbose=False)
         assert results.failed == 0, f"{mod.__name__}: {results.failed} doctest failures"
