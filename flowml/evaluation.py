@@ -328,9 +328,10 @@ class RegressionEvaluator(HasLabelCol, HasPredictionCol, HasWeightCol, Params):
         if name == "mae":
             return float((w * err.abs()).sum() / wsum)
         mean_l = (w * lab).sum() / wsum
-        if name == "var":  # explained variance
-            mean_p = (w * pred).sum() / wsum
-            return float((w * (pred - mean_p) ** 2).sum() / wsum)
+        if name == "var":
+            # Spark's explainedVariance: SS_reg/n with the LABEL mean
+            # (RegressionMetrics.explainedVariance), not the prediction mean
+            return float((w * (pred - mean_l) ** 2).sum() / wsum)
         ss_tot = float((w * (lab - mean_l) ** 2).sum())
         ss_res = float((w * err * err).sum())
         return 1.0 - ss_res / ss_tot if ss_tot > 0 else 0.0

@@ -228,3 +228,16 @@ class TestIsotonicRegression:
         assert (np.diff(pred[order]) <= 1e-9).all()  # decreasing in x
         corr = np.corrcoef(pred, y)[0, 1]
         assert corr > 0.95
+
+
+class TestExplainedVariance:
+    def test_var_uses_label_mean(self):
+        from flowml.evaluation import RegressionEvaluator
+
+        lab = torch.tensor([0.0, 2.0, 4.0])
+        pred = torch.tensor([1.0, 2.0, 3.0])
+        df = FlowFrame({"label": lab, "prediction": pred})
+        v = RegressionEvaluator(metricName="var").evaluate(df)
+        ybar = float(lab.mean())
+        expect = float(((pred - ybar) ** 2).mean())
+        assert v == pytest.approx(expect)
