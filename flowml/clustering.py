@@ -247,3 +247,6 @@ class ClusteringEvaluator(HasFeaturesCol, HasPredictionCol, Params):
             torch.zeros_like(own),
         )
         return float(s.mean())
+
+
+__all__ = ["KMeans", "KMeansModel", "ClusteringEvaluator"]

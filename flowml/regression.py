@@ -652,3 +652,22 @@ class IsotonicRegressionModel(HasWeightCol, RegressionModel, _RegressorIOMixin):
         x = dataset.col(self.getFeaturesCol())
         pred = self.predictRaw(x if isinstance(x, torch.Tensor) else torch.as_tensor(x))
         return dataset.withColumn(self.getOrDefault(self.predictionCol), pred)
+
+
+__all__ = [
+    "RegressionModel",
+    "LinearRegression", "LinearRegressionModel", "LinearRegressionTrainingSummary",
+    "DecisionTreeRegressor", "DecisionTreeRegressionModel",
+    "RandomForestRegressor", "RandomForestRegressionModel",
+    "GBTRegressor", "GBTRegressionModel",
+    "IsotonicRegression", "IsotonicRegressionModel",
+]
+__all__ += ["FMRegressor", "FMRegressionModel"]
+
+
+def __getattr__(name):  # PEP 562: lazy FM re-export (flowml.fm imports us)
+    if name in ("FMRegressor", "FMRegressionModel"):
+        import flowml.fm as _fm
+
+        return getattr(_fm, name)
+    raise AttributeError(f"module {__name__!r} has no attribute {name!r}")
