@@ -119,10 +119,14 @@ class FlatTree:
     accessors return live slices (indexable and writable).
     """
 
-    def __init__(self, stat_width: int = 0, cap: int = 8):
+    def __init__(self, stat_width: int = 0, cap: int = 8, regression: bool = False):
         self._n = 0
         self._sw = int(stat_width)
         self._cap = max(int(cap), 4)
+        # Explicit impurity kind: a width-3 stats row is ambiguous between
+        # 3-class classification counts and regression (s, s2, n), so the
+        # kind must be threaded in, never shape-sniffed (ADVICE r1, high).
+        self.regression = bool(regression)
         self._alloc(self._cap)
 
     def _alloc(self, cap: int) -> None:
@@ -236,10 +240,11 @@ class FlatTree:
 
     def prediction(self, i: int) -> float:
         s = self._stats[i]
-        if self._sw > 1 and self._sw != 3:
+        if self.regression:
+            # stats = (sum, sum_sq, n)
+            return float(s[0] / s[2]) if self._sw == 3 and s[2] > 0 else float(s[0])
+        if self._sw > 1:
             return float(np.argmax(s))
-        if self._sw == 3:
-            return float(s[0] / s[2]) if s[2] > 0 else 0.0
         return float(s[0])
 
     def depth(self) -> int:
@@ -443,7 +448,7 @@ def grow_trees(
 
     hist_by_level = []  # per-level hist ms (timing only)
 
-    arena = FlatTree(stat_width=S, cap=max(4 * T, 64))
+    arena = FlatTree(stat_width=S, cap=max(4 * T, 64), regression=not is_cls)
     tree_of_parts: List[np.ndarray] = [np.arange(T, dtype=np.int64)]
     # roots are arena nodes 0..T-1; every (tree,row) starts at its root
     nodeid = (
@@ -782,7 +787,7 @@ def grow_trees(
     for t in range(T):
         sl = slice(bounds[t], bounds[t + 1])
         n = bounds[t + 1] - bounds[t]
-        tr = FlatTree(stat_width=S, cap=max(int(n), 4))
+        tr = FlatTree(stat_width=S, cap=max(int(n), 4), regression=not is_cls)
         tr.bulk_add(st_o[sl], imp_o[sl], cnt_o[sl])
         tr.feature[:] = feat_o[sl]
         tr.thr_bin[:] = thb_o[sl]

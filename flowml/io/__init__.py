@@ -390,8 +390,14 @@ def _tree_node_rows(tree, tree_id: int) -> List[Dict[str, Any]]:
     return rows
 
 
-def _trees_from_node_rows(rows: List[Dict[str, Any]]):
-    """Rebuild FlatTree list from NodeData rows."""
+def _trees_from_node_rows(rows: List[Dict[str, Any]], regression: bool = False):
+    """Rebuild FlatTree list from NodeData rows.
+
+    ``regression`` is threaded from the model class being loaded (regression
+    trees — incl. GBT classification's internal trees — store (s, s2, n)
+    stats; classification stores per-class counts). Width-3 stats are
+    ambiguous between the two, so this must never be shape-sniffed.
+    """
     from flowml.classification.tree import FlatTree
 
     by_tree: Dict[int, List[Dict[str, Any]]] = {}
@@ -400,10 +406,10 @@ def _trees_from_node_rows(rows: List[Dict[str, Any]]):
     trees = []
     for tid in sorted(by_tree):
         nodes = sorted(by_tree[tid], key=lambda nd: nd["id"])
-        tr = FlatTree()
+        tr = FlatTree(regression=regression)
         for nd in nodes:
             stats = np.asarray(nd["impurityStats"], dtype=np.float64)
-            cnt = float(stats.sum() if len(stats) != 3 else stats[2])
+            cnt = float(stats[2] if regression else stats.sum())
             tr.add_node(stats, float(nd["impurity"]), cnt)
         for nd in nodes:
             i = nd["id"]
@@ -660,31 +666,33 @@ def _construct(cls, simple: str, rows, path: str, meta):
         activation = meta.get("paramMap", {}).get("activation", "relu")
         return cls(layers=[int(x) for x in layers], weights=w, activation=activation)
     if simple == "RandomForestClassificationModel":
-        trees = _trees_from_node_rows(rows)
+        trees = _trees_from_node_rows(rows, regression=False)
         n_classes = max(len(t.stats[0]) for t in trees)
         nf = _num_features_from_trees_meta(path, trees)
         return cls(trees=trees, numClasses=n_classes, numFeatures=nf)
     if simple == "GBTClassificationModel":
-        trees = _trees_from_node_rows(rows)
+        # GBT classification trees are internal regression trees on
+        # pseudo-residuals (variance impurity, (s, s2, n) stats).
+        trees = _trees_from_node_rows(rows, regression=True)
         tm = _read_parquet(os.path.join(path, "treesMetadata")).to_pylist()
         weights = [r["weights"] for r in sorted(tm, key=lambda r: r["treeID"])]
         nf = _num_features_from_trees_meta(path, trees)
         return cls(trees=trees, treeWeights=weights, numFeatures=nf)
     if simple == "DecisionTreeClassificationModel":
-        trees = _trees_from_node_rows(rows)
+        trees = _trees_from_node_rows(rows, regression=False)
         t = trees[0]
         return cls(tree=t, numClasses=len(t.stats[0]), numFeatures=0)
     if simple == "RandomForestRegressionModel":
-        trees = _trees_from_node_rows(rows)
+        trees = _trees_from_node_rows(rows, regression=True)
         return cls(trees=trees, numFeatures=_num_features_from_trees_meta(path, trees))
     if simple == "GBTRegressionModel":
-        trees = _trees_from_node_rows(rows)
+        trees = _trees_from_node_rows(rows, regression=True)
         tm = _read_parquet(os.path.join(path, "treesMetadata")).to_pylist()
         weights = [r["weights"] for r in sorted(tm, key=lambda r: r["treeID"])]
         return cls(trees=trees, treeWeights=weights,
                    numFeatures=_num_features_from_trees_meta(path, trees))
     if simple == "DecisionTreeRegressionModel":
-        trees = _trees_from_node_rows(rows)
+        trees = _trees_from_node_rows(rows, regression=True)
         return cls(tree=trees[0],
                    numFeatures=_num_features_from_trees_meta(path, trees))
     # parameter-only stages

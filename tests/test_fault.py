@@ -96,7 +96,7 @@ def test_checkpoint_atomicity(tmp_path):
     from flowml.utils.checkpoint import save_gbt_checkpoint, load_gbt_checkpoint
     from flowml.classification.tree import FlatTree
 
-    tr = FlatTree(stat_width=3)
+    tr = FlatTree(stat_width=3, regression=True)
     tr.add_node(np.array([1.0, 2.0, 4.0]), 0.5, 4.0)
     path = save_gbt_checkpoint(str(tmp_path / "ck"), [tr], [1.0], 4, 6)
     assert os.path.basename(path) == "round_000004"

@@ -127,6 +127,36 @@ def test_rf_round_trip(tmp_path):
     assert os.path.isdir(os.path.join(p, "treesMetadata"))
 
 
+def test_rf_3class_round_trip_importances_and_prediction(tmp_path):
+    """Width-3 impurityStats must not be mistaken for regression (s, s2, n).
+
+    A 3-class classification tree has 3-wide stats rows; shape-sniffing used
+    to (a) write leaf predictions with the regression formula s[0]/s[2] and
+    (b) reload node counts as stats[2], silently changing featureImportances
+    after a save/load round trip.
+    """
+    import pyarrow.parquet as pq
+
+    df = _tree_df()  # 3 classes
+    m = RandomForestClassifier(numTrees=5, maxDepth=4, seed=0).fit(df)
+    p = str(tmp_path / "rf3")
+    m.save(p)
+    from flowml.classification import RandomForestClassificationModel
+
+    m2 = RandomForestClassificationModel.load(p)
+    fi, fi2 = m.featureImportances.toArray(), m2.featureImportances.toArray()
+    assert np.allclose(fi, fi2), (fi, fi2)
+    # written NodeData prediction is the argmax of the class counts
+    tbl = pq.read_table(os.path.join(p, "data")).to_pylist()
+    for row in tbl:
+        nd = row["nodeData"]
+        stats = nd["impurityStats"]
+        assert len(stats) == 3
+        assert nd["prediction"] == float(np.argmax(stats))
+    # reloaded trees know they are classification trees
+    assert all(not t.regression for t in m2._trees)
+
+
 def test_gbt_round_trip(tmp_path):
     df = _tree_df(binary=True)
     m = GBTClassifier(maxIter=8, maxDepth=3, seed=0).fit(df)
