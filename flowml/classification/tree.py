@@ -263,19 +263,25 @@ class FlatTree:
 
 
 def _feature_subset_size(strategy: str, F: int, is_classification: bool, num_trees: int) -> int:
+    # Spark RandomForest.numFeaturesPerNode uses ceil for sqrt/log2/onethird
+    # (e.g. F=41 → sqrt gives 7, not 6) — ADVICE r1 parity fix
     s = str(strategy)
     if s == "auto":
         if num_trees == 1:
             return F
-        return max(1, int(math.sqrt(F))) if is_classification else max(1, F // 3)
+        return (
+            max(1, math.ceil(math.sqrt(F)))
+            if is_classification
+            else max(1, math.ceil(F / 3.0))
+        )
     if s == "all":
         return F
     if s == "sqrt":
-        return max(1, int(math.sqrt(F)))
+        return max(1, math.ceil(math.sqrt(F)))
     if s == "onethird":
-        return max(1, F // 3)
+        return max(1, math.ceil(F / 3.0))
     if s == "log2":
-        return max(1, int(math.log2(F)))
+        return max(1, math.ceil(math.log2(F)))
     try:
         iv = int(s)
         if iv > 0:

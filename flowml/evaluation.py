@@ -275,15 +275,15 @@ class BinaryClassificationEvaluator(HasLabelCol, HasRawPredictionCol, HasWeightC
         Nn = float((w * (1 - y)).sum())
         if P == 0 or Nn == 0:
             return 0.0
-        tpr = torch.cat([torch.zeros(1, dtype=torch.float64), tp / P])
-        fpr = torch.cat([torch.zeros(1, dtype=torch.float64), fp / Nn])
+        tpr = torch.cat([tp.new_zeros(1), tp / P])
+        fpr = torch.cat([fp.new_zeros(1), fp / Nn])
         if self.getMetricName() == "areaUnderROC":
             return float(torch.trapz(tpr, fpr))
         # areaUnderPR: precision-recall curve, first point (r0, p=1)
         prec = tp / (tp + fp)
         rec = tp / P
-        prec = torch.cat([torch.ones(1, dtype=torch.float64), prec])
-        rec = torch.cat([torch.zeros(1, dtype=torch.float64), rec])
+        prec = torch.cat([prec.new_ones(1), prec])
+        rec = torch.cat([rec.new_zeros(1), rec])
         return float(torch.trapz(prec, rec))
 
 

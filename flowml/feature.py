@@ -132,12 +132,25 @@ class StringIndexer(_StringIndexerParams, Estimator):
         return self.set(self.stringOrderType, value)
 
     def _fit(self, dataset: FlowFrame) -> "StringIndexerModel":
+        from flowml import dist
+
         in_cols, _ = self._in_out_cols()
         order = self.getStringOrderType()
         labelsArray: List[List[str]] = []
         for c in in_cols:
             vals = _column_as_strings(dataset, c)
             uniq, counts = np.unique(vals, return_counts=True)
+            if dist.get_world_size() > 1:
+                # merge per-rank frequency dicts so every
+                # rank derives the identical label order on sharded data
+                merged: Dict[str, int] = {}
+                for part in dist.all_gather_object(
+                    dict(zip(uniq.tolist(), (int(v) for v in counts)))
+                ):
+                    for k, v in part.items():
+                        merged[k] = merged.get(k, 0) + v
+                uniq = np.array(sorted(merged.keys()), dtype=object)
+                counts = np.array([merged[k] for k in uniq], dtype=np.int64)
             labelsArray.append(_order_labels(uniq, counts, order))
         model = StringIndexerModel(labelsArray=labelsArray)
         self._copyValues(model)
@@ -268,6 +281,10 @@ class OneHotEncoder(_OneHotEncoderParams, Estimator):
             if len(col) and float(col.min().item()) < 0:
                 raise ValueError(f"OneHotEncoder input {c!r} has negative values")
             sizes.append(mx + 1)
+        from flowml import dist
+
+        if dist.get_world_size() > 1:
+            sizes = [max(s) for s in zip(*dist.all_gather_object(sizes))]
         model = OneHotEncoderModel(categorySizes=sizes)
         self._copyValues(model)
         model._set_parent(self)
@@ -427,13 +444,31 @@ class StandardScaler(_StandardScalerParams, Estimator):
         col = dataset.col(self.getInputCol())
         if not (isinstance(col, torch.Tensor) and col.dim() == 2):
             raise TypeError("StandardScaler input must be a vector column")
+        from flowml import dist
+
         x = col.to(torch.float64)
-        mean = x.mean(dim=0)
-        n = x.shape[0]
-        if n > 1:
-            std = x.std(dim=0, unbiased=True)  # Spark Summarizer: sample std
+        if dist.get_world_size() > 1:
+            # all-reduce (n, Σx, Σx²) moments so sharded fits agree (C6)
+            F = x.shape[1]
+            buf = torch.zeros(1 + 2 * F, dtype=torch.float64, device=x.device)
+            buf[0] = x.shape[0]
+            buf[1 : 1 + F] = x.sum(dim=0)
+            buf[1 + F :] = (x * x).sum(dim=0)
+            dist.all_reduce_(buf)
+            n = float(buf[0])
+            mean = buf[1 : 1 + F] / max(n, 1.0)
+            if n > 1:
+                var = (buf[1 + F :] - n * mean * mean) / (n - 1)
+                std = var.clamp_min(0).sqrt()
+            else:
+                std = torch.zeros_like(mean)
         else:
-            std = torch.zeros_like(mean)
+            mean = x.mean(dim=0)
+            n = x.shape[0]
+            if n > 1:
+                std = x.std(dim=0, unbiased=True)  # Spark Summarizer: sample std
+            else:
+                std = torch.zeros_like(mean)
         model = StandardScalerModel(mean=mean.cpu(), std=std.cpu())
         self._copyValues(model)
         model._set_parent(self)
@@ -517,10 +552,12 @@ class MinMaxScaler(_MinMaxScalerParams, Estimator):
         col = dataset.col(self.getInputCol())
         if not (isinstance(col, torch.Tensor) and col.dim() == 2):
             raise TypeError("MinMaxScaler input must be a vector column")
+        from flowml import dist
+
         x = col.to(torch.float64)
-        model = MinMaxScalerModel(
-            originalMin=x.min(dim=0).values.cpu(), originalMax=x.max(dim=0).values.cpu()
-        )
+        mn = dist.all_reduce_(x.min(dim=0).values.contiguous(), op="min")
+        mx = dist.all_reduce_(x.max(dim=0).values.contiguous(), op="max")
+        model = MinMaxScalerModel(originalMin=mn.cpu(), originalMax=mx.cpu())
         self._copyValues(model)
         model._set_parent(self)
         return model
@@ -682,7 +719,11 @@ class MaxAbsScaler(_MaxAbsScalerParams, Estimator):
         x = dataset.col(self.getInputCol())
         if not isinstance(x, torch.Tensor) or x.dim() != 2:
             raise TypeError("MaxAbsScaler input must be a vector column")
-        max_abs = x.abs().amax(dim=0).to(torch.float32).cpu()
+        from flowml import dist
+
+        max_abs = dist.all_reduce_(
+            x.abs().amax(dim=0).to(torch.float32).contiguous(), op="max"
+        ).cpu()
         model = MaxAbsScalerModel(maxAbs=max_abs)
         self._copyValues(model)
         model._set_parent(self)
@@ -808,23 +849,48 @@ class Imputer(_ImputerParams, Estimator):
                   outputCols=outputCols, strategy=strategy, missingValue=missingValue)
 
     def _fit(self, dataset: FlowFrame) -> "ImputerModel":
+        from flowml import dist
+
         in_cols, _ = self._in_out_cols()
         strat = self.getOrDefault(self.strategy)
         mv = self.getOrDefault(self.missingValue)
+        world = dist.get_world_size()
         surrogates = []
         for c in in_cols:
             x = dataset.col(c).reshape(-1).to(torch.float64)
             miss = torch.isnan(x) if math.isnan(mv) else (x == mv)
             good = x[~miss & ~torch.isnan(x)]
-            if not len(good):
+            n_good = torch.tensor([float(len(good))], dtype=torch.float64, device=x.device)
+            if world > 1:
+                dist.all_reduce_(n_good)
+            if float(n_good) == 0:
                 raise ValueError(f"Imputer: column {c!r} has no non-missing values")
             if strat == "mean":
-                surrogates.append(float(good.mean()))
+                s = torch.zeros(1, dtype=torch.float64, device=x.device)
+                if len(good):
+                    s[0] = good.sum()
+                if world > 1:
+                    dist.all_reduce_(s)
+                surrogates.append(float(s / n_good))
             elif strat == "median":
-                surrogates.append(float(good.median()))
+                if world > 1:
+                    surrogates.append(float(dist.dist_quantiles(good, [0.5])[0]))
+                else:
+                    surrogates.append(float(good.median()))
             elif strat == "mode":
                 vals, cnts = torch.unique(good, return_counts=True)
-                surrogates.append(float(vals[int(cnts.argmax())]))
+                if world > 1:
+                    merged: Dict[float, int] = {}
+                    for part in dist.all_gather_object(
+                        dict(zip((float(v) for v in vals), (int(v) for v in cnts)))
+                    ):
+                        for k, v in part.items():
+                            merged[k] = merged.get(k, 0) + v
+                    # ties: smallest value wins (deterministic across ranks)
+                    best = max(sorted(merged), key=lambda k: merged[k])
+                    surrogates.append(float(best))
+                else:
+                    surrogates.append(float(vals[int(cnts.argmax())]))
             else:
                 raise ValueError(f"Unsupported strategy {strat!r}")
         model = ImputerModel(surrogates=surrogates)
@@ -893,10 +959,18 @@ class QuantileDiscretizer(HasInputCol, HasOutputCol, HasHandleInvalid, Estimator
         nb = self.getOrDefault(self.numBuckets)
         if nb < 2:
             raise ValueError("numBuckets must be >= 2")
+        from flowml import dist
+
         xv = x.reshape(-1).to(torch.float64)
         xv = xv[~torch.isnan(xv)]
         q = torch.arange(1, nb, dtype=torch.float64, device=xv.device) / nb
-        inner = torch.quantile(xv, q) if len(xv) else torch.zeros(0, dtype=torch.float64)
+        if dist.get_world_size() > 1:
+            # sharded data: global type-1 quantiles (Spark itself uses
+            # approxQuantile here, so the interpolation difference vs the
+            # local torch.quantile path is within the contract)
+            inner = dist.dist_quantiles(xv, [float(p) for p in q])
+        else:
+            inner = torch.quantile(xv, q) if len(xv) else torch.zeros(0, dtype=torch.float64)
         vals = sorted(set(float(v) for v in inner))
         splits = [-float("inf")] + vals + [float("inf")]
         return Bucketizer(splits=splits, inputCol=self.getInputCol(),

@@ -289,6 +289,22 @@ class TestModelFitsGPU:
         acc = MulticlassClassificationEvaluator(metricName="accuracy").evaluate(m.transform(df))
         assert acc > 0.8
 
+    def test_binary_evaluator_gpu_frame(self):
+        """AUC on a GPU-resident frame equals the CPU result (ADVICE r1:
+        curve endpoints used to be CPU tensors → device-mismatch on GPU)."""
+        from flowml.evaluation import BinaryClassificationEvaluator
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(13)
+        score = r.normal(size=2000).astype(np.float32)
+        y = (score + 0.8 * r.normal(size=2000) > 0).astype(np.float64)
+        cols = {"rawPrediction": torch.from_numpy(score), "label": y}
+        df_cpu = FlowFrame(dict(cols))
+        df_gpu = FlowFrame(dict(cols)).to(DEV)
+        for metric in ("areaUnderROC", "areaUnderPR"):
+            e = BinaryClassificationEvaluator(metricName=metric)
+            assert abs(e.evaluate(df_gpu) - e.evaluate(df_cpu)) < 1e-9
+
 
 class TestStreamingGraph:
     def test_hipgraph_capture_matches_eager(self):
