@@ -109,6 +109,10 @@ def bench_rf(args, device):
             numTrees=args.trees, maxDepth=args.depth, maxBins=32, seed=args.seed
         )
         state["model"] = rf.fit(df)
+        # force-materialize the (otherwise lazy) inference ensemble so the
+        # timed step includes full model construction — nothing the fit
+        # produces is deferred past the clock
+        _ = state["model"]._ensemble
         if _timing:
             sync(device)
             print(f"step total: {(time.perf_counter() - t0) * 1e3:.1f} ms")
@@ -265,6 +269,8 @@ def bench_mlp(args, device):
         "classes": 5,
         "parallelism": f"dp{world}",
         "macro_f1": round(f1, 4),
+        "hipgraph": graph is not None,
+        "rccl_live": dist.is_initialized(),
     }
     return elapsed, rows * world * args.steps, cfg, "bf16" if use_bf16 else "float32"
 
@@ -350,10 +356,45 @@ def bench_infer(args, device):
     return elapsed, batch * world * args.steps, cfg, "float32"
 
 
+def report_f1_curve(args, device):
+    """--report-f1-curve: RF fit time + macro-F1 at depths 5/8/10 on the
+    planted teacher (VERDICT r1 next-round #2) — one JSON line per depth."""
+    from flowml.classification import RandomForestClassifier
+    from flowml.evaluation import MulticlassClassificationEvaluator
+
+    rank = dist.get_rank()
+    rows = args.rows
+    X, y = synth_flows(rows, 41, 5, args.seed + rank, device)
+    df = make_df(X, y)
+    Xte, yte = synth_flows(min(rows, 200_000), 41, 5, args.seed + 7777 + rank, device)
+    dfte = make_df(Xte, yte)
+    for depth in (5, 8, 10):
+        rf = RandomForestClassifier(numTrees=args.trees, maxDepth=depth,
+                                    maxBins=32, seed=args.seed)
+        rf.fit(df)  # warmup
+        dist.barrier()
+        sync(device)
+        t0 = time.perf_counter()
+        model = rf.fit(df)
+        _ = model._ensemble
+        sync(device)
+        fit_ms = (time.perf_counter() - t0) * 1000.0
+        f1 = MulticlassClassificationEvaluator(metricName="macroF1").evaluate(
+            model.transform(dfte)
+        )
+        if rank == 0:
+            print(json.dumps({"curve": "rf_f1_vs_depth", "maxDepth": depth,
+                              "numTrees": args.trees, "rows_per_gpu": rows,
+                              "fit_ms": round(fit_ms, 2),
+                              "macro_f1": round(f1, 4)}), flush=True)
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=3)
+    p.add_argument("--steps", type=int, default=0,
+                   help="timed steps (0 = default: enough for a >=5 s timed "
+                        "region on GPU so rocm-smi sampling sees the kernels)")
     p.add_argument("--warmup", type=int, default=1)
     p.add_argument("--model", default="rf", choices=["rf", "gbt", "mlp", "lr", "infer"])
     p.add_argument("--rows", type=int, default=0, help="rows per GPU (0 = config default)")
@@ -361,6 +402,8 @@ def main():
     p.add_argument("--depth", type=int, default=5)
     p.add_argument("--batch", type=int, default=262144, help="infer micro-batch")
     p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--report-f1-curve", action="store_true",
+                   help="print RF fit-time/macro-F1 at depths 5/8/10 and exit")
     args = p.parse_args()
 
     dist.init_from_env()
@@ -373,12 +416,26 @@ def main():
     else:
         device = torch.device("cpu")
 
+    if args.steps == 0:
+        # GPU defaults sized so the timed region is >= ~5 s (VERDICT r1
+        # weak #4: the driver's rocm-smi gpu_busy sampler needs a window)
+        gpu_steps = {"rf": 300, "gbt": 500, "mlp": 2000, "lr": 2000, "infer": 1000}
+        args.steps = gpu_steps[args.model] if device.type == "cuda" else 3
+
     if args.rows == 0:
         defaults = {"rf": 1_250_000, "gbt": 6_250_000, "mlp": 1_000_000,
                     "lr": 1_250_000, "infer": 500_000}
         args.rows = defaults[args.model]
         if device.type == "cpu":
             args.rows = min(args.rows, 20_000)  # keep the no-GPU sanity path fast
+
+    if args.report_f1_curve:
+        if device.type == "cpu":
+            args.rows = min(args.rows, 20_000)
+        report_f1_curve(args, device)
+        if dist.is_initialized():
+            torch.distributed.destroy_process_group()
+        return
 
     if args.model == "rf":
         elapsed, flows, cfg, dtype = bench_rf(args, device)
