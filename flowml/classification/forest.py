@@ -31,6 +31,8 @@ from flowml.classification.tree import (
     FlatTree,
     GrowthConfig,
     TreeEnsemble,
+    apply_categorical_codes,
+    cat_arity_from_meta,
     find_split_thresholds,
     grow_trees,
 )
@@ -159,6 +161,9 @@ class RandomForestClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
         thresholds = find_split_thresholds(X, B, seed)
         t1 = _tick()
         codes = ops.bin_codes(X, thresholds)
+        cat_arity = cat_arity_from_meta(dataset, self.getFeaturesCol(), F)
+        if cat_arity is not None:
+            apply_categorical_codes(codes, X, cat_arity)
         t2 = _tick()
         T = self.getNumTrees()
         # BaggedPoint weights are computed IN the K2 kernel from a splitmix64
@@ -193,6 +198,7 @@ class RandomForestClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
             seed=seed,
             is_classification=True,
             n_classes=n_classes,
+            cat_arity=cat_arity,
         )
         trees = grow_trees(codes, y, None, thresholds, cfg, row_weights=row_w,
                            hash_bag=hash_bag)

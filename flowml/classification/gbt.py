@@ -34,6 +34,8 @@ from flowml.classification.tree import (
     FlatTree,
     GrowthConfig,
     TreeEnsemble,
+    apply_categorical_codes,
+    cat_arity_from_meta,
     find_split_thresholds,
     grow_trees,
     poisson_bag,
@@ -167,6 +169,9 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
         B = self.getMaxBins()
         thresholds = find_split_thresholds(X, B, seed)
         codes = ops.bin_codes(X, thresholds)
+        cat_arity = cat_arity_from_meta(dataset, self.getFeaturesCol(), F)
+        if cat_arity is not None:
+            apply_categorical_codes(codes, X, cat_arity)
 
         step = self.getStepSize()
         max_iter = self.getOrDefault(self.maxIter)
@@ -194,6 +199,7 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
             seed=seed,
             is_classification=False,
             n_classes=2,
+            cat_arity=cat_arity,
         )
         subsample = self.getOrDefault(self.subsamplingRate)
 

@@ -17,10 +17,12 @@ redundantly on every rank (cheaper than a scatter), and all per-level driver
 bookkeeping is numpy-vectorized — the host never loops over individual nodes
 during growth.
 
-Divergence from Spark, documented: all features are treated as continuous
-(post-StringIndexer categorical indices are ordinal here; Spark's categorical
-split-sets are not implemented — the saved-model ``split.numCategories`` is
-written as 0/continuous).
+Categorical features (declared via the features column's metadata — set by
+VectorIndexerModel, or by VectorAssembler from nominal input columns) use
+Spark's categorical split-sets: identity binning (bin = category), ordered
+centroid-sort scan or unordered subset enumeration in the split search
+(``binsToBestSplit`` categorical branches), and set-membership routing in
+advance/inference. Arity is capped at min(maxBins, 63) (int64 masks).
 """
 
 from __future__ import annotations
@@ -110,6 +112,10 @@ class GrowthConfig:
     seed: int = 0
     is_classification: bool = True
     n_classes: int = 2
+    # per-feature categorical arity (len-F int array; 0 = continuous).
+    # Categorical features use identity binning (bin = category) and Spark's
+    # categorical split-set semantics in the scan.
+    cat_arity: Optional[np.ndarray] = None
 
 
 class FlatTree:
@@ -139,6 +145,11 @@ class FlatTree:
         self._gain = np.zeros(cap, np.float64)
         self._count = np.zeros(cap, np.float64)
         self._stats = np.zeros((cap, self._sw), np.float64)
+        # categorical split-set per node: bit c of cat_mask set = category c
+        # goes LEFT; 0 = continuous split (or leaf). num_cats = feature arity
+        # at a categorical split (0 otherwise) — persisted as numCategories.
+        self._cat_mask = np.zeros(cap, np.int64)
+        self._num_cats = np.zeros(cap, np.int32)
 
     def _ensure(self, extra: int) -> None:
         need = self._n + extra
@@ -146,13 +157,15 @@ class FlatTree:
             return
         newcap = max(need, self._cap * 2)
         old = (self._feature, self._thr_bin, self._thr_raw, self._left,
-               self._right, self._impurity, self._gain, self._count, self._stats)
+               self._right, self._impurity, self._gain, self._count, self._stats,
+               self._cat_mask, self._num_cats)
         n = self._n
         self._cap = newcap
         self._alloc(newcap)
         for dst, src in zip(
             (self._feature, self._thr_bin, self._thr_raw, self._left,
-             self._right, self._impurity, self._gain, self._count, self._stats),
+             self._right, self._impurity, self._gain, self._count, self._stats,
+             self._cat_mask, self._num_cats),
             old,
         ):
             dst[:n] = src[:n]
@@ -199,6 +212,14 @@ class FlatTree:
     def stats(self) -> np.ndarray:
         return self._stats[: self._n]
 
+    @property
+    def cat_mask(self) -> np.ndarray:
+        return self._cat_mask[: self._n]
+
+    @property
+    def num_cats(self) -> np.ndarray:
+        return self._num_cats[: self._n]
+
     # -- mutation -------------------------------------------------------------
 
     def add_node(self, stats, impurity: float, count: float) -> int:
@@ -228,13 +249,17 @@ class FlatTree:
         self._n += k
         return i
 
-    def set_splits(self, nids, feats, bins, raws, lefts, rights, gains) -> None:
+    def set_splits(self, nids, feats, bins, raws, lefts, rights, gains,
+                   cat_masks=None, num_cats=None) -> None:
         self._feature[nids] = feats
         self._thr_bin[nids] = bins
         self._thr_raw[nids] = raws
         self._left[nids] = lefts
         self._right[nids] = rights
         self._gain[nids] = gains
+        if cat_masks is not None:
+            self._cat_mask[nids] = cat_masks
+            self._num_cats[nids] = num_cats
 
     # -- queries --------------------------------------------------------------
 
@@ -362,6 +387,221 @@ def split_scan(
 
 
 # ---------------------------------------------------------------------------
+# Categorical split-sets (spark:.../tree/impl/RandomForest.binsToBestSplit,
+# categorical branches; SURVEY.md §2.2 RF row / VERDICT r1 next-round #4)
+# ---------------------------------------------------------------------------
+
+
+def cat_arity_from_meta(dataset, features_col: str, F: int) -> Optional[np.ndarray]:
+    """Per-feature arity array from the features column's metadata
+    ({"categorical": {featIdx: arity}}, set by VectorIndexerModel or by
+    VectorAssembler from nominal input columns — the stand-in for Spark's
+    ML attribute groups). None when no feature is categorical."""
+    m = dataset.meta(features_col) if hasattr(dataset, "meta") else {}
+    cats = m.get("categorical")
+    if not cats:
+        return None
+    arr = np.zeros(F, np.int64)
+    for k, v in cats.items():
+        k, v = int(k), int(v)
+        if not (0 <= k < F):
+            raise ValueError(f"categorical feature index {k} out of range [0, {F})")
+        if v < 2:
+            raise ValueError(f"categorical feature {k} has arity {v} (< 2)")
+        arr[k] = v
+    return arr
+
+
+def max_unordered_arity(max_bins: int) -> int:
+    """Spark DecisionTreeMetadata.maxCategoriesForUnorderedFeature:
+    arity k is 'unordered' (full subset enumeration) iff
+    2·(2^(k−1) − 1) ≤ maxBins, i.e. k ≤ floor(log2(maxBins/2 + 1)) + 1."""
+    return int(math.floor(math.log2(max_bins / 2 + 1))) + 1 if max_bins >= 2 else 1
+
+
+def apply_categorical_codes(
+    codes: torch.Tensor, X: torch.Tensor, cat_arity: np.ndarray
+) -> torch.Tensor:
+    """Overwrite categorical feature columns of the binned code matrix with
+    the raw category index (identity binning — TreePoint semantics for
+    categorical features: bin = category). Values are validated to be
+    integral in [0, arity)."""
+    idx = np.nonzero(cat_arity > 0)[0]
+    if not len(idx):
+        return codes
+    dev = X.device
+    cols = torch.from_numpy(idx).to(dev)
+    vals = X.index_select(1, cols)
+    arity_t = torch.from_numpy(cat_arity[idx].astype(np.float32)).to(dev)
+    bad = (vals < 0) | (vals >= arity_t) | (vals != vals.floor())
+    if bool(bad.any()):
+        j = int(torch.nonzero(bad)[0, 1])
+        raise ValueError(
+            f"categorical feature {int(idx[j])} has values outside "
+            f"[0, {int(cat_arity[idx[j]])}) or non-integral values"
+        )
+    codes[:, cols] = vals.to(codes.dtype)
+    return codes
+
+
+def _categorical_best_splits(
+    hist: torch.Tensor,      # G×Fp×B×S float64-able (cls) or G×Fp×B×2 (reg cells)
+    slot_feat: np.ndarray,   # G×Fp global feature id per slot
+    cat_arity: np.ndarray,   # F arity per feature (0 = continuous)
+    cfg: "GrowthConfig",
+):
+    """Best categorical split per node over its categorical feature slots.
+
+    Spark semantics: ordered-categorical features (binary classification,
+    regression, or high-arity multiclass) sort categories by centroid —
+    P(class 1), target mean, or label-distribution impurity respectively,
+    empty categories last — and scan the arity−1 prefix splits (Breiman's
+    trick: exact for binary/regression, heuristic for multiclass);
+    low-arity multiclass features are 'unordered': all 2^(k−1)−1 subset
+    splits are enumerated. Returns per-node numpy arrays
+    (slot, gain, mask, arity, lstats, rstats, impL, impR, cntL, cntR);
+    slot = −1 / gain = −inf where no categorical candidate exists.
+    """
+    dev = hist.device
+    G, Fp = hist.shape[0], hist.shape[1]
+    Sw = hist.shape[3]
+    is_cls = cfg.is_classification
+    imp_kind = cfg.impurity
+    min_inst = cfg.min_instances_per_node
+    arity_slot = cat_arity[slot_feat]  # G×Fp
+    # per-position candidate buffers (first-max-wins merge per node at the end)
+    gain_pos = np.full((G, Fp), -np.inf)
+    mask_pos = np.zeros((G, Fp), np.int64)
+    ls_pos = np.zeros((G, Fp, Sw))
+    rs_pos = np.zeros((G, Fp, Sw))
+    il_pos = np.zeros((G, Fp))
+    ir_pos = np.zeros((G, Fp))
+    cl_pos = np.zeros((G, Fp))
+    cr_pos = np.zeros((G, Fp))
+    unord_max = max_unordered_arity(cfg.max_bins)
+
+    def _gain_cls(Lst, parent):
+        # Lst: M×C×S prefix/subset left stats; parent: M×S
+        Rst = parent.unsqueeze(1) - Lst
+        nL = Lst.sum(-1)
+        nR = Rst.sum(-1)
+        n = parent.sum(-1).clamp_min(1e-12)
+        iL = _impurity_from_stats(Lst, imp_kind)
+        iR = _impurity_from_stats(Rst, imp_kind)
+        iP = _impurity_from_stats(parent, imp_kind)
+        g = iP.unsqueeze(1) - (nL * iL + nR * iR) / n.unsqueeze(1)
+        valid = (nL >= min_inst) & (nR >= min_inst)
+        return torch.where(valid, g, torch.full_like(g, -float("inf"))), Rst, iL, iR, nL, nR
+
+    def _gain_reg(Lst, parent):
+        # stats are (Σwt, Σw); canonical variance-gain formula — identical
+        # op order to split_scan_packed_reg so merged gains are comparable
+        Rst = parent.unsqueeze(1) - Lst
+        sL, nL = Lst[..., 0], Lst[..., 1]
+        sR, nR = Rst[..., 0], Rst[..., 1]
+        sP, nP = parent[..., 0], parent[..., 1].clamp_min(1e-12)
+        g = (sL * sL / nL.clamp_min(1e-300) + sR * sR / nR.clamp_min(1e-300)
+             - (sP * sP / nP).unsqueeze(1)) / nP.unsqueeze(1)
+        valid = (nL >= min_inst) & (nR >= min_inst)
+        return torch.where(valid, g, torch.full_like(g, -float("inf"))), Rst, None, None, nL, nR
+
+    H = hist.to(torch.float64)
+    for k in np.unique(arity_slot[arity_slot > 0]):
+        k = int(k)
+        gi_np, si_np = np.nonzero(arity_slot == k)
+        if not len(gi_np):
+            continue
+        gi = torch.from_numpy(gi_np).to(dev)
+        si = torch.from_numpy(si_np).to(dev)
+        Hs = H[gi, si][:, :k, :]  # M×k×S
+        parent = Hs.sum(1)        # M×S
+        M = Hs.shape[0]
+        unordered = is_cls and cfg.n_classes > 2 and k <= unord_max
+        if unordered and k >= 2:
+            n_masks = (1 << (k - 1)) - 1
+            m_t = torch.arange(1, n_masks + 1, dtype=torch.int64, device=dev)
+            bits = ((m_t.unsqueeze(1) >> torch.arange(k, device=dev)) & 1).to(
+                torch.float64
+            )  # n_masks×k (bit k−1 always 0: complement symmetry)
+            Lst = torch.einsum("mk,nks->nms", bits, Hs)  # M×n_masks×S
+            g, Rst, iL, iR, nL, nR = _gain_cls(Lst, parent)
+            bi = g.argmax(dim=1)  # first max wins
+            ar = torch.arange(M, device=dev)
+            best_g = g[ar, bi]
+            best_mask = m_t[bi]
+        else:
+            cnt = Hs.sum(-1) if is_cls else Hs[..., 1]
+            if not is_cls:
+                centroid = Hs[..., 0] / cnt.clamp_min(1e-300)
+            elif cfg.n_classes == 2:
+                centroid = Hs[..., 1] / cnt.clamp_min(1e-300)
+            else:
+                centroid = _impurity_from_stats(Hs, imp_kind)
+            centroid = torch.where(
+                cnt > 0, centroid, torch.full_like(centroid, float("inf"))
+            )
+            order = torch.argsort(centroid, dim=1, stable=True)  # M×k
+            Hsort = Hs.gather(1, order.unsqueeze(-1).expand(-1, -1, Sw))
+            pref = Hsort.cumsum(dim=1)[:, : k - 1, :]  # M×(k−1)×S prefixes
+            gain_fn = _gain_cls if is_cls else _gain_reg
+            g, Rst, iL, iR, nL, nR = gain_fn(pref, parent)
+            bi = g.argmax(dim=1)
+            ar = torch.arange(M, device=dev)
+            best_g = g[ar, bi]
+            # mask of the winning prefix: OR of bits of order[:, :bi+1]
+            bitvals = (torch.ones(1, dtype=torch.int64, device=dev) << order)
+            maskpref = bitvals.cumsum(dim=1)  # exact: bits are disjoint
+            best_mask = maskpref[ar, bi]
+            Lst = pref
+        best_ls = Lst[ar, bi]
+        best_rs = Rst[ar, bi]
+        gain_pos[gi_np, si_np] = best_g.cpu().numpy()
+        mask_pos[gi_np, si_np] = best_mask.cpu().numpy()
+        ls_pos[gi_np, si_np] = best_ls.cpu().numpy()
+        rs_pos[gi_np, si_np] = best_rs.cpu().numpy()
+        if is_cls:
+            il_pos[gi_np, si_np] = iL[ar, bi].cpu().numpy()
+            ir_pos[gi_np, si_np] = iR[ar, bi].cpu().numpy()
+        cl_pos[gi_np, si_np] = nL[ar, bi].cpu().numpy()
+        cr_pos[gi_np, si_np] = nR[ar, bi].cpu().numpy()
+
+    best_slot = gain_pos.argmax(axis=1)  # first max wins (numpy argmax)
+    ar_np = np.arange(G)
+    best_gain = gain_pos[ar_np, best_slot]
+    none = ~np.isfinite(best_gain)
+    best_slot = np.where(none, -1, best_slot)
+    return {
+        "slot": best_slot.astype(np.int64),
+        "gain": best_gain,
+        "mask": mask_pos[ar_np, np.maximum(best_slot, 0)],
+        "arity": arity_slot[ar_np, np.maximum(best_slot, 0)].astype(np.int32),
+        "ls": ls_pos[ar_np, np.maximum(best_slot, 0)],
+        "rs": rs_pos[ar_np, np.maximum(best_slot, 0)],
+        "il": il_pos[ar_np, np.maximum(best_slot, 0)],
+        "ir": ir_pos[ar_np, np.maximum(best_slot, 0)],
+        "cl": cl_pos[ar_np, np.maximum(best_slot, 0)],
+        "cr": cr_pos[ar_np, np.maximum(best_slot, 0)],
+    }
+
+
+def _neutralize_cat_slots(hist: torch.Tensor, slot_feat: np.ndarray,
+                          cat_arity: np.ndarray) -> None:
+    """In place: collapse each categorical slot's histogram to bin 0 so the
+    continuous scan (HIP K3 or torch) sees correct parent/self stats but
+    never emits a split there (every prefix has an empty right side)."""
+    arity_slot = cat_arity[slot_feat]
+    gi_np, si_np = np.nonzero(arity_slot > 0)
+    if not len(gi_np):
+        return
+    dev = hist.device
+    gi = torch.from_numpy(gi_np).to(dev)
+    si = torch.from_numpy(si_np).to(dev)
+    tot = hist[gi, si].sum(dim=1)  # M×S
+    hist[gi, si] = 0.0
+    hist[gi, si, 0] = tot
+
+
+# ---------------------------------------------------------------------------
 # The level-synchronous grower
 # ---------------------------------------------------------------------------
 
@@ -431,6 +671,18 @@ def grow_trees(
     S = cfg.n_classes if is_cls else 3
     sub_F = _feature_subset_size(cfg.feature_subset, F, is_cls, T)
     use_subset = sub_F < F
+    cat_arity_np = None
+    if cfg.cat_arity is not None and (np.asarray(cfg.cat_arity) > 0).any():
+        cat_arity_np = np.asarray(cfg.cat_arity, dtype=np.int64)
+        amax = int(cat_arity_np.max())
+        if amax > min(cfg.max_bins, 63):
+            raise ValueError(
+                f"categorical feature arity {amax} exceeds "
+                f"min(maxBins={cfg.max_bins}, 63)"
+            )
+    all_feats_np = np.broadcast_to(
+        np.arange(F, dtype=np.int64), (1, F)
+    )  # slot map when no feature subsetting
 
     if bag is not None and row_weights is not None:
         bag = bag * row_weights.unsqueeze(0)
@@ -598,6 +850,21 @@ def grow_trees(
             dist.all_reduce_(hist.reshape(-1))
             _acc("allreduce", t0)
             t0 = _tick()
+            cat_best = None
+            if cat_arity_np is not None:
+                sf_np = fs_np if fs_np is not None else np.broadcast_to(
+                    all_feats_np, (G, F))
+                if is_cls:
+                    cat_best = _categorical_best_splits(
+                        hist, sf_np, cat_arity_np, cfg)
+                    _neutralize_cat_slots(hist, sf_np, cat_arity_np)
+                else:
+                    cells_v = ops.reg_hist_views(hist, G, sub_F, B)[0]
+                    cat_best = _categorical_best_splits(
+                        cells_v, sf_np, cat_arity_np, cfg)
+                    _neutralize_cat_slots(cells_v, sf_np, cat_arity_np)
+            _acc("cat_scan", t0)
+            t0 = _tick()
             # K3 split-gain scan: packed per-node results, one host transfer
             if is_cls:
                 packed = ops.split_scan_packed(
@@ -647,11 +914,55 @@ def grow_trees(
                 arena.stats[chunk_n, 1] = np.asarray(
                     ops.reg_hist_views(hist, G, sub_F, B)[1].cpu().numpy(),
                     dtype=np.float64)
+            cat_mask_arr = np.zeros(G, np.int64)
+            cat_ncat_arr = np.zeros(G, np.int32)
+            if cat_best is not None:
+                # merge: first-max-wins across ALL slots — categorical wins on
+                # strictly greater gain, or equal gain at a smaller slot index
+                cg = cat_best["gain"]
+                c_ok = np.isfinite(cg) & (cg > 0) & (cg >= cfg.min_info_gain)
+                cont_gain = np.where(bf >= 0, bg, -np.inf)
+                take = c_ok & (
+                    (cg > cont_gain)
+                    | ((cg == cont_gain) & (cat_best["slot"] < bf))
+                )
+                if take.any():
+                    tk = np.nonzero(take)[0]
+                    bf[tk] = cat_best["slot"][tk]
+                    bb[tk] = 0
+                    bg[tk] = cg[tk]
+                    cat_mask_arr[tk] = cat_best["mask"][tk]
+                    cat_ncat_arr[tk] = cat_best["arity"][tk]
+                    cl[tk] = cat_best["cl"][tk]
+                    cr[tk] = cat_best["cr"][tk]
+                    if is_cls:
+                        il[tk] = cat_best["il"][tk]
+                        ir[tk] = cat_best["ir"][tk]
+                        ls[tk] = cat_best["ls"][tk]
+                        rs[tk] = cat_best["rs"][tk]
+                    else:
+                        # cat reg stats are (Σwt, Σw) → map into (s, s², n)
+                        ls[tk] = 0.0
+                        rs[tk] = 0.0
+                        ls[tk, 0] = cat_best["ls"][tk, 0]
+                        ls[tk, 2] = cat_best["ls"][tk, 1]
+                        rs[tk, 0] = cat_best["rs"][tk, 0]
+                        rs[tk, 2] = cat_best["rs"][tk, 1]
             sel = np.nonzero(bf >= 0)[0]
             if len(sel):
                 k = len(sel)
                 gf = fs_np[sel, bf[sel]] if fs_np is not None else bf[sel]
-                raws = thr_host[gf, bb[sel]]
+                raws = thr_host[gf, np.maximum(bb[sel], 0)]
+                sel_mask = cat_mask_arr[sel]
+                sel_ncat = cat_ncat_arr[sel]
+                is_cat_sel = sel_mask != 0
+                if is_cat_sel.any():
+                    raws = raws.copy()
+                    raws[is_cat_sel] = np.nan  # categorical: mask routes, not thr
+                    bb_sel = bb[sel].copy()
+                    bb_sel[is_cat_sel] = -1
+                else:
+                    bb_sel = bb[sel]
                 # bulk-append ALL children of the chunk to the arena at once
                 stats2d = np.empty((2 * k, S))
                 stats2d[0::2] = ls[sel]
@@ -665,7 +976,8 @@ def grow_trees(
                 base = arena.bulk_add(stats2d, imps, cnts)
                 lid = base + 2 * np.arange(k, dtype=np.int64)
                 rid = lid + 1
-                arena.set_splits(chunk_n[sel], gf, bb[sel], raws, lid, rid, bg[sel])
+                arena.set_splits(chunk_n[sel], gf, bb_sel, raws, lid, rid, bg[sel],
+                                 cat_masks=sel_mask, num_cats=sel_ncat)
                 tree_of_parts.append(np.repeat(chunk_t[sel], 2))
                 if level + 1 < cfg.max_depth:
                     if is_cls:
@@ -696,6 +1008,7 @@ def grow_trees(
             torch.from_numpy(arena.thr_bin.copy()).to(dev),
             torch.from_numpy(arena.left.copy()).to(dev),
             torch.from_numpy(arena.right.copy()).to(dev),
+            torch.from_numpy(arena.cat_mask.copy()).to(dev),
         )
         _acc("advance", t0)
         level += 1
@@ -714,6 +1027,7 @@ def grow_trees(
             torch.from_numpy(arena.thr_bin.copy()).to(dev),
             torch.from_numpy(arena.left.copy()).to(dev),
             torch.from_numpy(arena.right.copy()).to(dev),
+            torch.from_numpy(arena.cat_mask.copy()).to(dev),
         ))
         _acc("final_advance", t0)
 
@@ -786,6 +1100,8 @@ def grow_trees(
     thb_o = arena.thr_bin[order]
     thr_o = arena.thr_raw[order]
     gain_o = arena.gain[order]
+    catm_o = arena.cat_mask[order]
+    ncat_o = arena.num_cats[order]
     internal_o = feat_o >= 0
     left_o = np.where(internal_o, g2l[np.where(internal_o, arena.left[order], 0)], -1)
     right_o = np.where(internal_o, g2l[np.where(internal_o, arena.right[order], 0)], -1)
@@ -799,6 +1115,8 @@ def grow_trees(
         tr.thr_bin[:] = thb_o[sl]
         tr.thr_raw[:] = thr_o[sl]
         tr.gain[:] = gain_o[sl]
+        tr.cat_mask[:] = catm_o[sl]
+        tr.num_cats[:] = ncat_o[sl]
         tr.left[:] = left_o[sl]
         tr.right[:] = right_o[sl]
         trees.append(tr)
@@ -831,6 +1149,7 @@ class TreeEnsemble:
         threshold = np.zeros((T, M), dtype=np.float32)
         left = np.zeros((T, M), dtype=np.int32)
         right = np.zeros((T, M), dtype=np.int32)
+        catmask = np.zeros((T, M), dtype=np.int64)
         value = np.zeros((T, M, K), dtype=np.float32)
         for ti, tr in enumerate(trees):
             n = tr.n_nodes
@@ -840,6 +1159,7 @@ class TreeEnsemble:
             threshold[ti, :n] = tr.thr_raw
             left[ti, :n] = tr.left
             right[ti, :n] = tr.right
+            catmask[ti, :n] = tr.cat_mask
             stats = tr.stats
             if mode == "prob_sum":
                 tot = stats.sum(axis=1, keepdims=True)
@@ -855,6 +1175,7 @@ class TreeEnsemble:
         self._threshold = torch.from_numpy(threshold)
         self._left = torch.from_numpy(left)
         self._right = torch.from_numpy(right)
+        self._catmask = torch.from_numpy(catmask)
         self._value = torch.from_numpy(value)
         self._tree_weights: Optional[torch.Tensor] = None  # GBT stepSize weights
 
@@ -866,6 +1187,7 @@ class TreeEnsemble:
         self._threshold = self._threshold.to(device)
         self._left = self._left.to(device)
         self._right = self._right.to(device)
+        self._catmask = self._catmask.to(device)
         self._value = self._value.to(device)
         if self._tree_weights is not None:
             self._tree_weights = self._tree_weights.to(device)
@@ -878,7 +1200,8 @@ class TreeEnsemble:
         if self._tree_weights is not None:
             value = (value * self._tree_weights.view(-1, 1, 1)).contiguous()
         return ops.tree_predict(
-            X.to(torch.float32), self._feature, self._threshold, self._left, self._right, value
+            X.to(torch.float32), self._feature, self._threshold, self._left,
+            self._right, value, self._catmask
         )
 
     def feature_importances(self, num_features: int) -> np.ndarray:

@@ -200,8 +200,17 @@ class StringIndexerModel(_StringIndexerParams, Model):
                     raise ValueError(f"Unsupported handleInvalid: {handle}")
             new_cols[o] = idx
         out = dataset
-        for o, idx in new_cols.items():
-            out = out.withColumn(o, torch.from_numpy(idx).to(dataset.device))
+        for o, idx, labels in zip(
+            new_cols.keys(), new_cols.values(), self.labelsArray
+        ):
+            # nominal attribute (Spark NominalAttribute.withValues): lets
+            # VectorAssembler declare this slot categorical for the trees;
+            # handleInvalid='keep' adds one extra index = numLabels
+            n_vals = len(labels) + (1 if handle == "keep" else 0)
+            out = out.withColumn(
+                o, torch.from_numpy(idx).to(dataset.device),
+                meta={"nominal": n_vals},
+            )
         if keep_mask is not None:
             out = out.filter(keep_mask)
         return out
@@ -380,6 +389,8 @@ class VectorAssembler(HasInputCols, HasOutputCol, HasHandleInvalid, Transformer)
         handle = self.getHandleInvalid()
         parts: List[torch.Tensor] = []
         device = None
+        categorical: Dict[int, int] = {}
+        offset = 0
         for c in in_cols:
             col = dataset.col(c)
             if isinstance(col, np.ndarray):
@@ -387,9 +398,19 @@ class VectorAssembler(HasInputCols, HasOutputCol, HasHandleInvalid, Transformer)
                     f"VectorAssembler input {c!r} is a string column; index it first"
                 )
             t = col if col.dim() == 2 else col.reshape(-1, 1)
+            # merge ML attributes (Spark AttributeGroup): a nominal scalar
+            # input (StringIndexer output) or a categorical vector input
+            # (VectorIndexer output) marks its slots categorical
+            cmeta = dataset.meta(c)
+            if t.shape[1] == 1 and cmeta.get("nominal"):
+                categorical[offset] = int(cmeta["nominal"])
+            for j, a in (cmeta.get("categorical") or {}).items():
+                categorical[offset + int(j)] = int(a)
+            offset += t.shape[1]
             parts.append(t.to(torch.float32))
             device = t.device
         feats = torch.cat(parts, dim=1) if len(parts) > 1 else parts[0].clone()
+        out_meta = {"categorical": categorical} if categorical else None
         if handle != "keep":
             bad = torch.isnan(feats).any(dim=1)
             if bool(bad.any()):
@@ -397,9 +418,172 @@ class VectorAssembler(HasInputCols, HasOutputCol, HasHandleInvalid, Transformer)
                     raise ValueError(
                         "VectorAssembler encountered NaN/null values (handleInvalid='error')."
                     )
-                out = dataset.withColumn(out_col, feats)
+                out = dataset.withColumn(out_col, feats, meta=out_meta)
                 return out.filter(~bad)
-        return dataset.withColumn(out_col, feats)
+        return dataset.withColumn(out_col, feats, meta=out_meta)
+
+
+# ---------------------------------------------------------------------------
+# VectorIndexer (spark:.../feature/VectorIndexer.scala)
+# ---------------------------------------------------------------------------
+
+
+class _VectorIndexerParams(HasInputCol, HasOutputCol, HasHandleInvalid):
+    maxCategories: Param[int] = Param(
+        "undefined", "maxCategories",
+        "Threshold for the number of values a categorical feature can take "
+        "(>= 2). If a feature is found to have > maxCategories values, then "
+        "it is declared continuous.",
+        TypeConverters.toInt,
+    )
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(maxCategories=20, handleInvalid="error")
+
+    def getMaxCategories(self) -> int:
+        return self.getOrDefault(self.maxCategories)
+
+
+class VectorIndexer(_VectorIndexerParams, Estimator):
+    """Automatically decide which features are categorical and index them.
+
+    Spark semantics (spark:.../feature/VectorIndexer.scala): a feature with
+    at most ``maxCategories`` distinct values is declared categorical; its
+    values are re-indexed with **0.0 kept at index 0** (to preserve
+    sparsity) and the remaining values sorted ascending. The output column
+    carries categorical metadata ({"categorical": {featIdx: arity}}) that
+    the tree estimators read — the stand-in for Spark's ML attributes.
+    Distributed: per-rank distinct sets are merged across ranks so every
+    rank fits the identical model.
+    """
+
+    def __init__(self, *, inputCol: Optional[str] = None,
+                 outputCol: Optional[str] = None,
+                 maxCategories: Optional[int] = None,
+                 handleInvalid: Optional[str] = None):
+        super().__init__()
+        self._set(inputCol=inputCol, outputCol=outputCol,
+                  maxCategories=maxCategories, handleInvalid=handleInvalid)
+
+    def setMaxCategories(self, v: int):
+        return self.set(self.maxCategories, v)
+
+    def _fit(self, dataset: FlowFrame) -> "VectorIndexerModel":
+        from flowml import dist
+
+        x = dataset.col(self.getInputCol())
+        if not isinstance(x, torch.Tensor) or x.dim() != 2:
+            raise TypeError("VectorIndexer input must be a vector column")
+        maxc = self.getMaxCategories()
+        if maxc < 2:
+            raise ValueError("maxCategories must be >= 2")
+        F = x.shape[1]
+        xd = x.to(torch.float64)
+        distinct: List[Optional[np.ndarray]] = []
+        for j in range(F):
+            u = torch.unique(xd[:, j])
+            # cutoff: once a feature exceeds maxCategories it is continuous;
+            # keep maxc+1 values so the merged set can still exceed the cap
+            distinct.append(u.cpu().numpy()[: maxc + 1] if len(u) <= maxc else None)
+        if dist.get_world_size() > 1:
+            parts = dist.all_gather_object(
+                [d.tolist() if d is not None else None for d in distinct]
+            )
+            merged: List[Optional[np.ndarray]] = []
+            for j in range(F):
+                vals: Optional[set] = set()
+                for p in parts:
+                    if p[j] is None or vals is None:
+                        vals = None
+                        break
+                    vals.update(p[j])
+                    if len(vals) > maxc:
+                        vals = None
+                        break
+                merged.append(
+                    np.array(sorted(vals), dtype=np.float64) if vals is not None else None
+                )
+            distinct = merged
+        category_maps: Dict[int, Dict[float, int]] = {}
+        for j, d in enumerate(distinct):
+            if d is None or len(d) > maxc or len(d) < 1:
+                continue
+            vals = sorted(float(v) for v in d)
+            # 0.0 keeps index 0 (sparsity-preserving, per Spark)
+            if 0.0 in vals:
+                vals = [0.0] + [v for v in vals if v != 0.0]
+            category_maps[j] = {v: i for i, v in enumerate(vals)}
+        model = VectorIndexerModel(numFeatures=F, categoryMaps=category_maps)
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class VectorIndexerModel(_VectorIndexerParams, Model):
+    def __init__(self, numFeatures: int = 0,
+                 categoryMaps: Optional[Dict[int, Dict[float, int]]] = None):
+        super().__init__()
+        self.numFeatures = int(numFeatures)
+        self.categoryMaps: Dict[int, Dict[float, int]] = {
+            int(k): {float(a): int(b) for a, b in v.items()}
+            for k, v in (categoryMaps or {}).items()
+        }
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getInputCol())
+        if not isinstance(x, torch.Tensor) or x.dim() != 2:
+            raise TypeError("VectorIndexer input must be a vector column")
+        handle = self.getHandleInvalid()
+        out = x.to(torch.float32).clone()
+        keep = None
+        cat_meta: Dict[int, int] = {}
+        for j, cmap in self.categoryMaps.items():
+            vals = torch.tensor(sorted(cmap.keys()), dtype=torch.float64,
+                                device=x.device)
+            idxs = torch.tensor([cmap[float(v)] for v in vals],
+                                dtype=torch.float32, device=x.device)
+            col = x[:, j].to(torch.float64)
+            pos = torch.searchsorted(vals, col.contiguous())
+            pos_c = pos.clamp_max(len(vals) - 1)
+            hit = vals[pos_c] == col
+            mapped = idxs[pos_c]
+            arity = len(cmap)
+            if bool((~hit).any()):
+                if handle == "error":
+                    bad = col[~hit][:5].cpu().tolist()
+                    raise ValueError(
+                        f"VectorIndexer found unseen values {bad} in feature {j} "
+                        f"(handleInvalid='error')"
+                    )
+                if handle == "keep":
+                    mapped = torch.where(hit, mapped, torch.full_like(mapped, float(arity)))
+                    arity += 1
+                elif handle == "skip":
+                    keep = hit if keep is None else (keep & hit)
+                else:
+                    raise ValueError(f"Unsupported handleInvalid: {handle}")
+            out[:, j] = mapped
+            cat_meta[j] = arity
+        res = dataset.withColumn(self.getOutputCol(), out,
+                                 meta={"categorical": cat_meta} if cat_meta else None)
+        if keep is not None:
+            res = res.filter(keep)
+        return res
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str):
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
 
 
 # ---------------------------------------------------------------------------

@@ -26,9 +26,16 @@ Column = Union[torch.Tensor, np.ndarray]
 class FlowFrame:
     """An eager columnar batch: ordered ``{name: column}`` with equal lengths."""
 
-    def __init__(self, columns: "Dict[str, Any]", device: Optional[Union[str, torch.device]] = None):
+    def __init__(self, columns: "Dict[str, Any]", device: Optional[Union[str, torch.device]] = None,
+                 meta: Optional[Dict[str, dict]] = None):
         self._cols: Dict[str, Column] = {}
         self._nrows: Optional[int] = None
+        # per-column metadata (the stand-in for Spark's ML column attributes:
+        # e.g. {"categorical": {featIdx: arity}} on a features vector, or
+        # {"nominal": numLabels} on a StringIndexer output column)
+        self._meta: Dict[str, dict] = {
+            k: dict(v) for k, v in (meta or {}).items() if k in columns and v
+        }
         self.device = torch.device(device) if device is not None else torch.device("cpu")
         # device is honored, not just recorded: tensor columns are moved so
         # FlowFrame({...}, device="cuda") behaves like .to("cuda") (string
@@ -65,6 +72,23 @@ class FlowFrame:
     @staticmethod
     def from_dict(data: Dict[str, Any], device=None) -> "FlowFrame":
         return FlowFrame(data, device=device)
+
+    # -- column metadata ------------------------------------------------------
+
+    def meta(self, name: str) -> dict:
+        """Metadata dict attached to a column ({} if none)."""
+        return dict(self._meta.get(name, {}))
+
+    def withMeta(self, name: str, meta: dict) -> "FlowFrame":
+        """Same frame with ``meta`` attached to column ``name``."""
+        if name not in self._cols:
+            raise KeyError(name)
+        m = dict(self._meta)
+        if meta:
+            m[name] = dict(meta)
+        else:
+            m.pop(name, None)
+        return FlowFrame(self._cols, device=self.device, meta=m)
 
     # -- basic introspection -------------------------------------------------
 
@@ -109,10 +133,13 @@ class FlowFrame:
 
     # -- transformations -----------------------------------------------------
 
-    def withColumn(self, name: str, col: Any) -> "FlowFrame":
+    def withColumn(self, name: str, col: Any, meta: Optional[dict] = None) -> "FlowFrame":
         cols = dict(self._cols)
         cols[name] = col
-        return FlowFrame(cols, device=self.device)
+        m = {k: v for k, v in self._meta.items() if k != name}
+        if meta:
+            m[name] = dict(meta)
+        return FlowFrame(cols, device=self.device, meta=m)
 
     def select(self, *names: str) -> "FlowFrame":
         flat: List[str] = []
@@ -121,18 +148,20 @@ class FlowFrame:
                 flat.extend(n)
             else:
                 flat.append(n)
-        return FlowFrame({n: self.col(n) for n in flat}, device=self.device)
+        return FlowFrame({n: self.col(n) for n in flat}, device=self.device, meta=self._meta)
 
     def drop(self, *names: str) -> "FlowFrame":
         return FlowFrame(
-            {n: c for n, c in self._cols.items() if n not in names}, device=self.device
+            {n: c for n, c in self._cols.items() if n not in names},
+            device=self.device, meta=self._meta,
         )
 
     def withColumnRenamed(self, old: str, new: str) -> "FlowFrame":
         cols = {}
         for n, c in self._cols.items():
             cols[new if n == old else n] = c
-        return FlowFrame(cols, device=self.device)
+        m = {(new if k == old else k): v for k, v in self._meta.items()}
+        return FlowFrame(cols, device=self.device, meta=m)
 
     def filter(self, mask: Union[torch.Tensor, np.ndarray]) -> "FlowFrame":
         """Row subset by boolean mask (tensor on any device, or numpy)."""
@@ -152,7 +181,7 @@ class FlowFrame:
                 if idx_t is None or idx_t.device != c.device:
                     idx_t = torch.from_numpy(np.nonzero(mask_np)[0]).to(c.device)
                 cols[n] = c.index_select(0, idx_t)
-        return FlowFrame(cols, device=self.device)
+        return FlowFrame(cols, device=self.device, meta=self._meta)
 
     def take_rows(self, indices: np.ndarray) -> "FlowFrame":
         cols: Dict[str, Column] = {}
@@ -166,7 +195,7 @@ class FlowFrame:
                     t = torch.from_numpy(np.ascontiguousarray(indices)).to(c.device)
                     idx_cache[c.device] = t
                 cols[n] = c.index_select(0, t)
-        return FlowFrame(cols, device=self.device)
+        return FlowFrame(cols, device=self.device, meta=self._meta)
 
     def limit(self, n: int) -> "FlowFrame":
         return self.take_rows(np.arange(min(n, self.count()), dtype=np.int64))
@@ -181,7 +210,7 @@ class FlowFrame:
                 cols[n] = np.concatenate([a, np.asarray(b, dtype=object)])
             else:
                 cols[n] = torch.cat([a, b], dim=0)
-        return FlowFrame(cols, device=self.device)
+        return FlowFrame(cols, device=self.device, meta=self._meta)
 
     # -- randomSplit ---------------------------------------------------------
 
@@ -214,7 +243,7 @@ class FlowFrame:
         cols: Dict[str, Column] = {}
         for n, c in self._cols.items():
             cols[n] = c if isinstance(c, np.ndarray) else c.to(device, non_blocking=True)
-        return FlowFrame(cols, device=device)
+        return FlowFrame(cols, device=device, meta=self._meta)
 
     def cpu(self) -> "FlowFrame":
         return self.to("cpu")

@@ -76,6 +76,8 @@ _register("QuantileDiscretizer", "org.apache.spark.ml.feature.QuantileDiscretize
 _register("PCA", "org.apache.spark.ml.feature.PCA")
 _register("PCAModel", "org.apache.spark.ml.feature.PCAModel")
 _register("VectorSlicer", "org.apache.spark.ml.feature.VectorSlicer")
+_register("VectorIndexer", "org.apache.spark.ml.feature.VectorIndexer")
+_register("VectorIndexerModel", "org.apache.spark.ml.feature.VectorIndexerModel")
 _register("Interaction", "org.apache.spark.ml.feature.Interaction")
 _register("KMeans", "org.apache.spark.ml.clustering.KMeans")
 _register("KMeansModel", "org.apache.spark.ml.clustering.KMeansModel")
@@ -365,6 +367,17 @@ def _tree_node_rows(tree, tree_id: int) -> List[Dict[str, Any]]:
         visit(0)
     for i in order:
         internal = tree.feature[i] >= 0
+        mask = int(tree.cat_mask[i]) if internal else 0
+        if mask:
+            # categorical split: the left category set + the feature arity
+            # (Spark CategoricalSplit serialization)
+            lcot = [float(c) for c in range(64) if (mask >> c) & 1]
+            ncat = int(tree.num_cats[i])
+        elif internal:
+            lcot = [float(tree.thr_raw[i])]  # continuous: single threshold
+            ncat = 0
+        else:
+            lcot, ncat = [], -1
         rows.append(
             {
                 "treeID": tree_id,
@@ -378,11 +391,8 @@ def _tree_node_rows(tree, tree_id: int) -> List[Dict[str, Any]]:
                     "rightChild": new_id[tree.right[i]] if internal else -1,
                     "split": {
                         "featureIndex": int(tree.feature[i]) if internal else -1,
-                        # continuous split: single-element threshold list
-                        # (numCategories = 0 marks continuous; categorical
-                        # split-sets are a documented divergence)
-                        "leftCategoriesOrThreshold": [float(tree.thr_raw[i])] if internal else [],
-                        "numCategories": 0 if internal else -1,
+                        "leftCategoriesOrThreshold": lcot,
+                        "numCategories": ncat,
                     },
                 },
             }
@@ -414,8 +424,19 @@ def _trees_from_node_rows(rows: List[Dict[str, Any]], regression: bool = False):
         for nd in nodes:
             i = nd["id"]
             if nd["leftChild"] >= 0:
-                tr.feature[i] = int(nd["split"]["featureIndex"])
-                tr.thr_raw[i] = float(nd["split"]["leftCategoriesOrThreshold"][0])
+                sp = nd["split"]
+                tr.feature[i] = int(sp["featureIndex"])
+                ncat = int(sp.get("numCategories", 0))
+                if ncat > 0:
+                    mask = 0
+                    for c in sp["leftCategoriesOrThreshold"]:
+                        mask |= 1 << int(c)
+                    tr.cat_mask[i] = mask
+                    tr.num_cats[i] = ncat
+                    tr.thr_bin[i] = -1
+                    tr.thr_raw[i] = float("nan")
+                else:
+                    tr.thr_raw[i] = float(sp["leftCategoriesOrThreshold"][0])
                 tr.left[i] = int(nd["leftChild"])
                 tr.right[i] = int(nd["rightChild"])
                 tr.gain[i] = float(nd["gain"])
@@ -445,6 +466,21 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
         ]
     if cls == "MaxAbsScalerModel":
         return [{"maxAbs": vector_struct(instance.maxAbs.numpy())}]
+    if cls == "VectorIndexerModel":
+        # Spark stores Data(numFeatures, categoryMaps: Map[Int, Map[Double,
+        # Int]]); parquet map-of-map types are fragile across readers, so
+        # categoryMaps is flattened to one row per categorical feature with
+        # the category values ordered by their index (index = position) —
+        # marked as an unconfirmed-field divergence (SURVEY.md §7 part 2).
+        rows = [{"numFeatures": int(instance.numFeatures),
+                 "featureIndex": -1, "categories": []}]
+        for j in sorted(instance.categoryMaps):
+            cmap = instance.categoryMaps[j]
+            ordered = [v for v, _ in sorted(cmap.items(), key=lambda t: t[1])]
+            rows.append({"numFeatures": int(instance.numFeatures),
+                         "featureIndex": int(j),
+                         "categories": [float(v) for v in ordered]})
+        return rows
     if cls == "ImputerModel":
         return [{"surrogates": [float(v) for v in instance.surrogates]}]
     if cls == "LogisticRegressionModel":
@@ -658,6 +694,15 @@ def _construct(cls, simple: str, rows, path: str, meta):
         coef = torch.from_numpy(vector_from_struct(r0["coefficients"])).to(torch.float32)
         return cls(coefficients=coef, intercept=float(r0["intercept"]),
                    numFeatures=int(coef.numel()))
+    if simple == "VectorIndexerModel":
+        nf = int(rows[0]["numFeatures"]) if rows else 0
+        cmaps = {}
+        for r0 in rows:
+            if int(r0["featureIndex"]) >= 0:
+                cmaps[int(r0["featureIndex"])] = {
+                    float(v): i for i, v in enumerate(r0["categories"])
+                }
+        return cls(numFeatures=nf, categoryMaps=cmaps)
     if simple == "MultilayerPerceptronClassificationModel":
         layers = meta.get("paramMap", {}).get("layers") or meta.get(
             "defaultParamMap", {}

@@ -16,21 +16,21 @@ void launch_bin_codes(const float*, const float*, uint8_t*, long long, int,
 void launch_tree_hist_cls(const uint8_t*, int*, const int*, const int*,
                           const int*, const uint8_t*, const float*, const int*,
                           float*, const int*, const int*, const int*,
-                          const int*, int, const float*,
+                          const int*, const long long*, int, const float*,
                           unsigned long long, long long, long long, int, int,
                           int, int, int, int, void*);
 void launch_tree_hist_reg(const uint8_t*, int*, const int*, const int*,
                           const int*, const float*, const float*, const int*,
                           float*, const int*, const int*, const int*,
-                          const int*, int, const float*,
+                          const int*, const long long*, int, const float*,
                           unsigned long long, long long, long long, int, int,
                           int, int, int, int, void*);
 void launch_node_advance(const uint8_t*, const int*, int*, const int*,
-                         const int*, const int*, const int*,
+                         const int*, const int*, const int*, const long long*,
                          long long, int, int, void*);
 void launch_tree_predict(const float*, const int*, const float*, const int*,
-                         const int*, const float*, float*, long long, int, int,
-                         int, int, void*);
+                         const int*, const float*, const long long*, float*,
+                         long long, int, int, int, int, void*);
 void launch_logreg_delta(const float*, const int*, const float*, const float*,
                          const float*, float*, float*, long long, int, int,
                          void*);
@@ -85,7 +85,7 @@ torch::Tensor tree_hist_cls(torch::Tensor codes, torch::Tensor nodeid,
                             int64_t B, int64_t C, torch::Tensor fs,
                             torch::Tensor adv_feat, torch::Tensor adv_thr,
                             torch::Tensor adv_left, torch::Tensor adv_right,
-                            int64_t bag_mode,
+                            torch::Tensor adv_cat, int64_t bag_mode,
                             torch::Tensor bag_cdf, int64_t bag_seed,
                             int64_t row_offset) {
   CHK(codes);
@@ -109,7 +109,11 @@ torch::Tensor tree_hist_cls(torch::Tensor codes, torch::Tensor nodeid,
       adv ? adv_feat.data_ptr<int>() : nullptr,
       adv ? adv_thr.data_ptr<int>() : nullptr,
       adv ? adv_left.data_ptr<int>() : nullptr,
-      adv ? adv_right.data_ptr<int>() : nullptr, (int)bag_mode,
+      adv ? adv_right.data_ptr<int>() : nullptr,
+      (adv && adv_cat.numel())
+          ? reinterpret_cast<const long long*>(adv_cat.data_ptr<int64_t>())
+          : nullptr,
+      (int)bag_mode,
       bag_cdf.numel() ? bag_cdf.data_ptr<float>() : nullptr,
       (unsigned long long)bag_seed, (long long)row_offset, N, F,
       T, Fp, (int)B, (int)C, (int)l_max, cur_stream());
@@ -123,7 +127,7 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
                             int64_t B, torch::Tensor fs,
                             torch::Tensor adv_feat, torch::Tensor adv_thr,
                             torch::Tensor adv_left, torch::Tensor adv_right,
-                            int64_t bag_mode,
+                            torch::Tensor adv_cat, int64_t bag_mode,
                             torch::Tensor bag_cdf, int64_t bag_seed,
                             int64_t row_offset) {
   CHK(codes);
@@ -147,7 +151,11 @@ torch::Tensor tree_hist_reg(torch::Tensor codes, torch::Tensor nodeid,
       adv ? adv_feat.data_ptr<int>() : nullptr,
       adv ? adv_thr.data_ptr<int>() : nullptr,
       adv ? adv_left.data_ptr<int>() : nullptr,
-      adv ? adv_right.data_ptr<int>() : nullptr, (int)bag_mode,
+      adv ? adv_right.data_ptr<int>() : nullptr,
+      (adv && adv_cat.numel())
+          ? reinterpret_cast<const long long*>(adv_cat.data_ptr<int64_t>())
+          : nullptr,
+      (int)bag_mode,
       bag_cdf.numel() ? bag_cdf.data_ptr<float>() : nullptr,
       (unsigned long long)bag_seed, (long long)row_offset, N, F,
       T, Fp, (int)B, (int)G, (int)l_max, cur_stream());
@@ -183,9 +191,11 @@ torch::Tensor split_scan(torch::Tensor hist, int64_t imp_code,
 
 torch::Tensor node_advance(torch::Tensor codes, torch::Tensor nodeid,
                            torch::Tensor feat, torch::Tensor thr,
-                           torch::Tensor left, torch::Tensor right) {
+                           torch::Tensor left, torch::Tensor right,
+                           torch::Tensor catmask) {
   CHK(codes);
   CHK(nodeid);
+  TORCH_CHECK(catmask.scalar_type() == torch::kInt64, "catmask must be i64");
   const long long N = codes.size(0);
   const int F = codes.size(1);
   const int T = nodeid.size(0);
@@ -193,16 +203,20 @@ torch::Tensor node_advance(torch::Tensor codes, torch::Tensor nodeid,
   launch_node_advance(codes.data_ptr<uint8_t>(), nodeid.data_ptr<int>(),
                       out.data_ptr<int>(), feat.data_ptr<int>(),
                       thr.data_ptr<int>(), left.data_ptr<int>(),
-                      right.data_ptr<int>(), N, F, T, cur_stream());
+                      right.data_ptr<int>(),
+                      reinterpret_cast<const long long*>(catmask.data_ptr<int64_t>()),
+                      N, F, T, cur_stream());
   return out;
 }
 
 torch::Tensor tree_predict(torch::Tensor X, torch::Tensor feature,
                            torch::Tensor threshold, torch::Tensor left,
-                           torch::Tensor right, torch::Tensor value) {
+                           torch::Tensor right, torch::Tensor value,
+                           torch::Tensor catmask) {
   CHK(X);
   CHK(feature);
   CHK(value);
+  TORCH_CHECK(catmask.scalar_type() == torch::kInt64, "catmask must be i64");
   const long long N = X.size(0);
   const int F = X.size(1);
   const int T = feature.size(0);
@@ -213,6 +227,7 @@ torch::Tensor tree_predict(torch::Tensor X, torch::Tensor feature,
   launch_tree_predict(X.data_ptr<float>(), feature.data_ptr<int>(),
                       threshold.data_ptr<float>(), left.data_ptr<int>(),
                       right.data_ptr<int>(), value.data_ptr<float>(),
+                      reinterpret_cast<const long long*>(catmask.data_ptr<int64_t>()),
                       out.data_ptr<float>(), N, F, T, M, K, cur_stream());
   return out;
 }

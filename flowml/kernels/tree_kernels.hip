@@ -89,6 +89,23 @@ __device__ __forceinline__ float flowml_hash_bag(
   return (float)k;
 }
 
+// Split routing. Continuous: go left iff code <= thr (bin space) or
+// x <= threshold (raw space). Categorical (mask != 0): go left iff the
+// category bit is set (CategoricalSplit.shouldGoLeft; unseen/out-of-range
+// categories go right). Masks are int64 -> arity <= 63, enforced at fit.
+__device__ inline bool flowml_left_code(int code, int thr, long long mask) {
+  if (mask != 0) return code <= 63 && ((mask >> code) & 1LL) != 0;
+  return code <= thr;
+}
+
+__device__ inline bool flowml_left_raw(float x, float thr, long long mask) {
+  if (mask != 0) {
+    if (x < 0.f || x > 63.f) return false;
+    return ((mask >> (int)x) & 1LL) != 0;
+  }
+  return x <= thr;
+}
+
 // ---------------------------------------------------------------------------
 // K2 — classification histogram (LDS-privatized per tree, fused advance/bag)
 // ---------------------------------------------------------------------------
@@ -106,6 +123,7 @@ extern "C" __global__ void flowml_tree_hist_cls(
     const int* __restrict__ adv_feat,    // arena split arrays or null
     const int* __restrict__ adv_thr, const int* __restrict__ adv_left,
     const int* __restrict__ adv_right,
+    const long long* __restrict__ adv_cat,  // arena categorical masks or null
     int bag_mode, const float* __restrict__ bag_cdf,
     unsigned long long bag_seed, long long row_offset,
     long long N, int F, int Fp, int B, int C, int l_max,
@@ -173,7 +191,9 @@ extern "C" __global__ void flowml_tree_hist_cls(
         // them in the histograms) so the final per-leaf stats pass can
         // still address every row
         if (f >= 0) {
-          nid = (crow[f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
+          const long long cm = adv_cat ? adv_cat[nid] : 0LL;
+          nid = flowml_left_code(crow[f], adv_thr[nid], cm) ? adv_left[nid]
+                                                            : adv_right[nid];
           nid_t[r] = nid;
         }
       }
@@ -239,6 +259,7 @@ extern "C" __global__ void flowml_tree_hist_reg(
     float* __restrict__ out,
     const int* __restrict__ adv_feat, const int* __restrict__ adv_thr,
     const int* __restrict__ adv_left, const int* __restrict__ adv_right,
+    const long long* __restrict__ adv_cat,
     int bag_mode, const float* __restrict__ bag_cdf,
     unsigned long long bag_seed, long long row_offset,
     long long N, int F, int Fp, int B, int Gtot,
@@ -297,7 +318,9 @@ extern "C" __global__ void flowml_tree_hist_reg(
       if (nid >= 0) {
         const int f = adv_feat[nid];
         if (f >= 0) {  // keep-at-leaf (see cls kernel)
-          nid = (crow[f] <= adv_thr[nid]) ? adv_left[nid] : adv_right[nid];
+          const long long cm = adv_cat ? adv_cat[nid] : 0LL;
+          nid = flowml_left_code(crow[f], adv_thr[nid], cm) ? adv_left[nid]
+                                                            : adv_right[nid];
           nid_t[r] = nid;
         }
       }
@@ -365,7 +388,8 @@ extern "C" __global__ void flowml_node_advance(
     const uint8_t* __restrict__ codes, const int* __restrict__ nodeid_in,
     int* __restrict__ nodeid_out, const int* __restrict__ feat,
     const int* __restrict__ thr, const int* __restrict__ left,
-    const int* __restrict__ right, long long N, int F) {
+    const int* __restrict__ right, const long long* __restrict__ catmask,
+    long long N, int F) {
   const int t = blockIdx.y;
   const int* in_t = nodeid_in + (long long)t * N;
   int* out_t = nodeid_out + (long long)t * N;
@@ -382,7 +406,8 @@ extern "C" __global__ void flowml_node_advance(
       continue;
     }
     const int code = codes[r * (long long)F + f];
-    out_t[r] = (code <= thr[nid]) ? left[nid] : right[nid];
+    out_t[r] = flowml_left_code(code, thr[nid], catmask[nid]) ? left[nid]
+                                                              : right[nid];
   }
 }
 
@@ -394,6 +419,7 @@ extern "C" __global__ void flowml_tree_predict(
     const float* __restrict__ X, const int* __restrict__ feature,
     const float* __restrict__ threshold, const int* __restrict__ left,
     const int* __restrict__ right, const float* __restrict__ value,
+    const long long* __restrict__ catmask,
     float* __restrict__ out, long long N, int F, int T, int M, int K) {
   for (long long r = blockIdx.x * (long long)blockDim.x + threadIdx.x; r < N;
        r += (long long)gridDim.x * blockDim.x) {
@@ -406,8 +432,10 @@ extern "C" __global__ void flowml_tree_predict(
       int node = 0;
       int f = feature[base + node];
       while (f >= 0) {
-        node = (xr[f] <= threshold[base + node]) ? left[base + node]
-                                                 : right[base + node];
+        node = flowml_left_raw(xr[f], threshold[base + node],
+                               catmask[base + node])
+                   ? left[base + node]
+                   : right[base + node];
         f = feature[base + node];
       }
       const float* v = value + (base + node) * (long long)K;
@@ -555,8 +583,8 @@ extern "C" void launch_tree_hist_cls(
     const uint8_t* codes, int* nodeid, const int* lut, const int* tree_lo,
     const int* tree_cnt, const uint8_t* labels, const float* bag, const int* fs,
     float* out, const int* adv_feat, const int* adv_thr, const int* adv_left,
-    const int* adv_right, int bag_mode, const float* bag_cdf,
-    unsigned long long bag_seed, long long row_offset,
+    const int* adv_right, const long long* adv_cat, int bag_mode,
+    const float* bag_cdf, unsigned long long bag_seed, long long row_offset,
     long long N, int F, int T, int Fp, int B, int C,
     int l_max, void* stream) {
   hist_opt_in_big_lds();
@@ -582,16 +610,17 @@ extern "C" void launch_tree_hist_cls(
   hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(T, tiles), dim3(block), lds,
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
                      labels, bag, fs, out, adv_feat, adv_thr, adv_left,
-                     adv_right, bag_mode, bag_cdf, bag_seed, row_offset, N, F,
-                     Fp, B, C, l_max, rows_per_block, stage_bytes);
+                     adv_right, adv_cat, bag_mode, bag_cdf, bag_seed,
+                     row_offset, N, F, Fp, B, C, l_max, rows_per_block,
+                     stage_bytes);
 }
 
 extern "C" void launch_tree_hist_reg(
     const uint8_t* codes, int* nodeid, const int* lut, const int* tree_lo,
     const int* tree_cnt, const float* targets, const float* bag, const int* fs,
     float* out, const int* adv_feat, const int* adv_thr, const int* adv_left,
-    const int* adv_right, int bag_mode, const float* bag_cdf,
-    unsigned long long bag_seed, long long row_offset,
+    const int* adv_right, const long long* adv_cat, int bag_mode,
+    const float* bag_cdf, unsigned long long bag_seed, long long row_offset,
     long long N, int F, int T, int Fp, int B, int Gtot, int l_max,
     void* stream) {
   hist_opt_in_big_lds();
@@ -615,30 +644,32 @@ extern "C" void launch_tree_hist_reg(
   hipLaunchKernelGGL(flowml_tree_hist_reg, dim3(T, tiles), dim3(block), lds,
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
                      targets, bag, fs, out, adv_feat, adv_thr, adv_left,
-                     adv_right, bag_mode, bag_cdf, bag_seed, row_offset, N, F,
-                     Fp, B, Gtot, l_max, rows_per_block, stage_bytes);
+                     adv_right, adv_cat, bag_mode, bag_cdf, bag_seed,
+                     row_offset, N, F, Fp, B, Gtot, l_max, rows_per_block,
+                     stage_bytes);
 }
 
 extern "C" void launch_node_advance(const uint8_t* codes, const int* nodeid_in,
                                     int* nodeid_out, const int* feat,
                                     const int* thr, const int* left,
-                                    const int* right, long long N, int F,
-                                    int T, void* stream) {
+                                    const int* right, const long long* catmask,
+                                    long long N, int F, int T, void* stream) {
   const int block = 256;
   hipLaunchKernelGGL(flowml_node_advance, dim3(grid_1d(N, block, 2048), T),
                      dim3(block), 0, (hipStream_t)stream, codes, nodeid_in,
-                     nodeid_out, feat, thr, left, right, N, F);
+                     nodeid_out, feat, thr, left, right, catmask, N, F);
 }
 
 extern "C" void launch_tree_predict(const float* X, const int* feature,
                                     const float* threshold, const int* left,
                                     const int* right, const float* value,
-                                    float* out, long long N, int F, int T,
-                                    int M, int K, void* stream) {
+                                    const long long* catmask, float* out,
+                                    long long N, int F, int T, int M, int K,
+                                    void* stream) {
   const int block = 256;
   hipLaunchKernelGGL(flowml_tree_predict, dim3(grid_1d(N, block, 8192)),
                      dim3(block), 0, (hipStream_t)stream, X, feature, threshold,
-                     left, right, value, out, N, F, T, M, K);
+                     left, right, value, catmask, out, N, F, T, M, K);
 }
 
 extern "C" void launch_logreg_delta(const float* X, const int* y,

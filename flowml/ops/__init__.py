@@ -180,10 +180,11 @@ def tree_hist_nodes(
         fs = feat_subset if feat_subset is not None else _empty(dev, torch.int32)
         bg = bag if bag is not None else _empty(dev)
         if advance is not None:
-            af, at, al, ar = advance
+            af, at, al, ar, am = advance
         else:
             e = _empty(dev, torch.int32)
             af = at = al = ar = e
+            am = _empty(dev, torch.int64)
         if hash_bag is not None:
             bm, cdf, bseed, rofs = hash_bag
             cdf = cdf.to(dev)
@@ -193,16 +194,16 @@ def tree_hist_nodes(
             return ext.tree_hist_cls(
                 codes, nodeid, lut, tree_lo, tree_cnt, labels, bg,
                 int(n_groups), int(l_max), int(n_bins), int(n_classes), fs,
-                af, at, al, ar, int(bm), cdf, int(bseed), int(rofs),
+                af, at, al, ar, am, int(bm), cdf, int(bseed), int(rofs),
             )
         return ext.tree_hist_reg(
             codes, nodeid, lut, tree_lo, tree_cnt, targets, bg,
             int(n_groups), int(l_max), int(n_bins), fs,
-            af, at, al, ar, int(bm), cdf, int(bseed), int(rofs),
+            af, at, al, ar, am, int(bm), cdf, int(bseed), int(rofs),
         )
     if advance is not None:
-        af, at, al, ar = advance
-        nodeid.copy_(node_advance(codes, nodeid, af, at, al, ar))
+        af, at, al, ar, am = advance
+        nodeid.copy_(node_advance(codes, nodeid, af, at, al, ar, am))
     return _tree_hist_nodes_torch(
         codes, nodeid, lut, labels, bag, n_groups, n_bins, n_classes,
         feat_subset, targets, hash_bag,
@@ -350,25 +351,37 @@ def node_advance(
     thr: torch.Tensor,     # n_nodes int32: split bin (go left iff code <= thr)
     left: torch.Tensor,    # n_nodes int32 (global child ids)
     right: torch.Tensor,   # n_nodes int32
+    catmask: Optional[torch.Tensor] = None,  # n_nodes int64: categorical
+    # split-set (bit c set = category c left); 0 = continuous node
 ) -> torch.Tensor:
     """Advance every (tree,row) one level; rows at leaves keep their id (K4b)."""
+    if catmask is None:
+        catmask = torch.zeros_like(left, dtype=torch.int64)
     if hip_available(codes):
-        return get_ext().node_advance(codes, nodeid, feat, thr, left, right)
+        return get_ext().node_advance(codes, nodeid, feat, thr, left, right,
+                                      catmask)
     T, N = nodeid.shape
     out = torch.empty_like(nodeid)
     for t in range(T):
-        out[t] = _node_advance_torch(codes, nodeid[t], feat, thr, left, right)
+        out[t] = _node_advance_torch(codes, nodeid[t], feat, thr, left, right,
+                                     catmask)
     return out
 
 
-def _node_advance_torch(codes, nodeid, feat_of, thrbin_of, left_of, right_of):
+def _node_advance_torch(codes, nodeid, feat_of, thrbin_of, left_of, right_of,
+                        catmask_of):
     nid = nodeid.to(torch.int64)
     live = nid >= 0
     nid_c = nid.clamp_min(0)
     f = feat_of.to(torch.int64)[nid_c]
     internal = (f >= 0) & live
     fx = codes.gather(1, f.clamp_min(0).unsqueeze(1)).squeeze(1).to(torch.int32)
-    go_left = fx <= thrbin_of[nid_c]
+    m = catmask_of[nid_c]
+    go_left = torch.where(
+        m != 0,
+        (m >> fx.to(torch.int64).clamp_(0, 63)) & 1 == 1,
+        fx <= thrbin_of[nid_c],
+    )
     nxt = torch.where(go_left, left_of.to(torch.int64)[nid_c], right_of.to(torch.int64)[nid_c])
     # keep-at-leaf: rows at leaves retain their node id (histograms skip
     # them via lut = -1; the final per-leaf stats pass needs them live)
@@ -388,17 +401,25 @@ def tree_predict(
     left: torch.Tensor,      # T×M int32
     right: torch.Tensor,     # T×M int32
     value: torch.Tensor,     # T×M×K float32 leaf vectors
+    catmask: Optional[torch.Tensor] = None,  # T×M int64 split-sets (0 = cont)
 ) -> torch.Tensor:
-    """Sum of per-tree leaf value vectors per row: N×K float32 (K8)."""
+    """Sum of per-tree leaf value vectors per row: N×K float32 (K8).
+
+    Categorical nodes route by set membership: category = int(x), go left
+    iff bit ``category`` of ``catmask`` is set (CategoricalSplit.shouldGoLeft;
+    unseen categories go right, as in Spark)."""
+    if catmask is None:
+        catmask = torch.zeros_like(left, dtype=torch.int64)
     if hip_available(X):
         return get_ext().tree_predict(
             X.contiguous(), feature.contiguous(), threshold.contiguous(),
-            left.contiguous(), right.contiguous(), value.contiguous()
+            left.contiguous(), right.contiguous(), value.contiguous(),
+            catmask.contiguous()
         )
-    return _tree_predict_torch(X, feature, threshold, left, right, value)
+    return _tree_predict_torch(X, feature, threshold, left, right, value, catmask)
 
 
-def _tree_predict_torch(X, feature, threshold, left, right, value):
+def _tree_predict_torch(X, feature, threshold, left, right, value, catmask):
     N, F = X.shape
     T, M = feature.shape
     K = value.shape[2]
@@ -409,13 +430,18 @@ def _tree_predict_torch(X, feature, threshold, left, right, value):
         thr_t = threshold[t]
         left_t = left[t].to(torch.int64)
         right_t = right[t].to(torch.int64)
+        mask_t = catmask[t]
         for _ in range(64):
             f = feat_t[node]
             internal = f >= 0
             if not bool(internal.any()):
                 break
             fx = X.gather(1, torch.clamp(f, min=0).unsqueeze(1)).squeeze(1)
-            go_left = fx <= thr_t[node]
+            m = mask_t[node]
+            cat = fx.to(torch.int64).clamp(0, 63)
+            go_left = torch.where(
+                m != 0, ((m >> cat) & 1 == 1) & (fx >= 0), fx <= thr_t[node]
+            )
             nxt = torch.where(go_left, left_t[node], right_t[node])
             node = torch.where(internal, nxt, node)
         out += value[t][node]

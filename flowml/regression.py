@@ -26,6 +26,8 @@ from flowml.classification.tree import (
     FlatTree,
     GrowthConfig,
     TreeEnsemble,
+    apply_categorical_codes,
+    cat_arity_from_meta,
     find_split_thresholds,
     grow_trees,
     poisson_bag,
@@ -88,6 +90,9 @@ def _fit_reg_forest(est, dataset: FlowFrame, num_trees: int, use_hash_bag: bool)
     B = est.getMaxBins()
     thresholds = find_split_thresholds(X, B, seed)
     codes = ops.bin_codes(X, thresholds)
+    cat_arity = cat_arity_from_meta(dataset, est.getFeaturesCol(), F)
+    if cat_arity is not None:
+        apply_categorical_codes(codes, X, cat_arity)
     hash_bag = None
     if use_hash_bag:
         rate = est.getOrDefault(est.subsamplingRate)
@@ -117,6 +122,7 @@ def _fit_reg_forest(est, dataset: FlowFrame, num_trees: int, use_hash_bag: bool)
         seed=seed,
         is_classification=False,
         n_classes=2,
+        cat_arity=cat_arity,
     )
     trees = grow_trees(codes, y, None, thresholds, cfg, row_weights=row_w,
                        hash_bag=hash_bag)
@@ -233,6 +239,9 @@ class GBTRegressor(_TreeEnsembleParams, _PredictorParams, HasMaxIter, Estimator,
         B = self.getMaxBins()
         thresholds = find_split_thresholds(X, B, seed)
         codes = ops.bin_codes(X, thresholds)
+        cat_arity = cat_arity_from_meta(dataset, self.getFeaturesCol(), F)
+        if cat_arity is not None:
+            apply_categorical_codes(codes, X, cat_arity)
         step = self.getOrDefault(self.stepSize)
         max_iter = self.getOrDefault(self.maxIter)
         subsample = self.getOrDefault(self.subsamplingRate)
@@ -256,6 +265,7 @@ class GBTRegressor(_TreeEnsembleParams, _PredictorParams, HasMaxIter, Estimator,
                 seed=(seed + 1_000_003 * rnd) & 0x7FFFFFFF,
                 is_classification=False,
                 n_classes=2,
+                cat_arity=cat_arity,
             )
             bag = poisson_bag(N, 1, subsample, False, cfg.seed, dev) if subsample < 1.0 else None
             if inst_w is not None:

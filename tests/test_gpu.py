@@ -364,7 +364,8 @@ class TestFusedHistFeatures:
         thr = torch.tensor([3, 0, 0] * T, dtype=torch.int32, device=DEV)
         left = torch.cat([torch.tensor([t * M + 1, -1, -1], dtype=torch.int32) for t in range(T)]).to(DEV)
         right = torch.cat([torch.tensor([t * M + 2, -1, -1], dtype=torch.int32) for t in range(T)]).to(DEV)
-        adv = (feat, thr, left, right)
+        adv = (feat, thr, left, right,
+               torch.zeros(T * M, dtype=torch.int64, device=DEV))
         labels = torch.from_numpy(r.integers(0, C, N).astype(np.uint8)).to(DEV)
         lut = torch.full((T * M,), -1, dtype=torch.int32)
         g = 0
