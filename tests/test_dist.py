@@ -404,3 +404,47 @@ def test_rf_dist_world8_predictions_close_to_single():
     m = RandomForestClassifier(numTrees=4, maxDepth=3, seed=5, bootstrap=False).fit(full)
     pred_single = m.transform(full)["prediction"].numpy()
     assert (pred_dist == pred_single).mean() > 0.99
+
+
+def _rf_categorical_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.classification import RandomForestClassifier
+
+    r = np.random.default_rng(30)
+    n, k = 2000, 5
+    cat = r.integers(0, k, n)
+    x1 = r.normal(size=n)
+    y = np.isin(cat, [0, 2, 4]).astype(np.float64)
+    X = np.column_stack([cat, x1]).astype(np.float32)
+    lo, hi = dist.shard_range(n, rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi]),
+                    "label": y[lo:hi]}).withMeta(
+        "features", {"categorical": {0: k}})
+    m = RandomForestClassifier(numTrees=4, maxDepth=3, seed=5,
+                               bootstrap=False).fit(df)
+    masks = np.concatenate([t.cat_mask for t in m._trees])
+    gathered = dist.all_gather_object(masks)
+    assert all(np.array_equal(g, gathered[0]) for g in gathered), "masks diverged"
+    full = FlowFrame({"features": torch.from_numpy(X), "label": y}).withMeta(
+        "features", {"categorical": {0: k}})
+    return m.transform(full)["prediction"].numpy(), masks
+
+
+def test_rf_categorical_dist_equals_single():
+    pred_dist, masks_dist = _run_dist("_rf_categorical_dist_body")
+    from flowml import FlowFrame
+    from flowml.classification import RandomForestClassifier
+
+    r = np.random.default_rng(30)
+    n, k = 2000, 5
+    cat = r.integers(0, k, n)
+    x1 = r.normal(size=n)
+    y = np.isin(cat, [0, 2, 4]).astype(np.float64)
+    X = np.column_stack([cat, x1]).astype(np.float32)
+    full = FlowFrame({"features": torch.from_numpy(X), "label": y}).withMeta(
+        "features", {"categorical": {0: k}})
+    m = RandomForestClassifier(numTrees=4, maxDepth=3, seed=5, bootstrap=False).fit(full)
+    masks_single = np.concatenate([t.cat_mask for t in m._trees])
+    assert np.array_equal(masks_dist, masks_single)
+    assert (pred_dist == m.transform(full)["prediction"].numpy()).mean() > 0.99
+    assert (masks_single != 0).any()  # the categorical path actually ran
