@@ -318,8 +318,15 @@ def bench_lr(args, device):
 
 
 def bench_infer(args, device):
-    """Config 5: streaming micro-batch ensemble inference (K8)."""
+    """Config 5: streaming micro-batch ensemble inference (K8).
+
+    Multi-rank: the StreamingDriver round-robins a global stream across
+    ranks (each rank replays its own hipGraph) and all-reduces the processed
+    row/class counts, so the reported number is a measured aggregate, not a
+    per-rank extrapolation.
+    """
     from flowml.classification import RandomForestClassifier
+    from flowml.streaming import StreamingClassifier, StreamingDriver
 
     rank = dist.get_rank()
     world = dist.get_world_size()
@@ -328,30 +335,34 @@ def bench_infer(args, device):
     model = RandomForestClassifier(numTrees=args.trees, maxDepth=args.depth,
                                    seed=args.seed).fit(make_df(X, y))
     batch = args.batch
-    Xb, _ = synth_flows(batch, 41, 5, args.seed + 31 + rank, device)
     use_graph = device.type == "cuda"
-    if use_graph:
-        from flowml.streaming import StreamingClassifier
+    sc = StreamingClassifier(model, batch_size=batch, num_features=41,
+                             device=device, use_graph=use_graph)
+    driver = StreamingDriver(sc)
+    # small device-resident batch pool cycled as the "stream" (synthetic)
+    pool = [synth_flows(batch, 41, 5, args.seed + 31 + 101 * p + rank, device)[0]
+            for p in range(2)]
 
-        sc = StreamingClassifier(model, batch_size=batch, num_features=41,
-                                 device=device, use_graph=True)
+    def stream(n_global):
+        for i in range(n_global):
+            yield pool[i % len(pool)]
 
-        def step():
-            sc.infer(Xb)
-    else:
-        ens = model._ensemble.to(device)
-
-        def step():
-            raw = ens.predict_raw(Xb)
-            _ = raw.argmax(dim=1)
-
-    elapsed = timed_loop(step, args.steps, args.warmup, device)
+    driver.run(stream(args.warmup * world))  # warmup (incl. graph replays)
+    res = driver.run(stream(args.steps * world))
+    elapsed = torch.tensor([res["elapsed_s"]], dtype=torch.float64,
+                           device=device if device.type == "cuda" else "cpu")
+    if dist.is_initialized():
+        torch.distributed.all_reduce(elapsed, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(elapsed.item())
+    assert res["rows"] == batch * world * args.steps, (
+        "stream accounting mismatch", res["rows"])
     cfg = {
         "model": f"RandomForest ensemble x{args.trees} inference",
         "micro_batch": batch,
         "features": 41,
-        "hipgraph": use_graph,
+        "hipgraph": use_graph and sc._graph is not None,
         "parallelism": f"dp{world}",
+        "rows_all_reduced": res["rows"],
     }
     return elapsed, batch * world * args.steps, cfg, "float32"
 

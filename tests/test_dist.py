@@ -448,3 +448,28 @@ def test_rf_categorical_dist_equals_single():
     assert np.array_equal(masks_dist, masks_single)
     assert (pred_dist == m.transform(full)["prediction"].numpy()).mean() > 0.99
     assert (masks_single != 0).any()  # the categorical path actually ran
+
+
+def _streaming_driver_body(rank, world):
+    from flowml import FlowFrame
+    from flowml.classification import RandomForestClassifier
+    from flowml.streaming import StreamingClassifier, StreamingDriver
+
+    r = np.random.default_rng(40)
+    X, y = _make_data(n=2000)
+    m = RandomForestClassifier(numTrees=4, maxDepth=3, seed=5, bootstrap=False).fit(
+        FlowFrame({"features": torch.from_numpy(X), "label": y}))
+    sc = StreamingClassifier(m, batch_size=256, num_features=5,
+                             device="cpu", use_graph=False)
+    driver = StreamingDriver(sc)
+    pool = [torch.from_numpy(r.normal(size=(256, 5)).astype(np.float32))
+            for _ in range(3)]
+    res = driver.run(pool[i % 3] for i in range(10))
+    return res
+
+
+def test_streaming_driver_round_robin_aggregates():
+    res = _run_dist("_streaming_driver_body")
+    assert res["rows"] == 10 * 256           # global stream fully covered
+    assert res["rows_local"] == 5 * 256      # each of 2 ranks took half
+    assert sum(res["class_counts"]) == 10 * 256
