@@ -157,7 +157,8 @@ class TestTreePredictGPU:
                 right[t, i] = 2 * i + 2
         args = [X, feature.to(DEV), threshold.to(DEV), left.to(DEV), right.to(DEV), value.to(DEV)]
         gpu = ops.tree_predict(*args).cpu()
-        cpu = ops._tree_predict_torch(X.cpu(), feature, threshold, left, right, value)
+        cpu = ops._tree_predict_torch(X.cpu(), feature, threshold, left, right, value,
+                                      torch.zeros_like(left, dtype=torch.int64))
         assert torch.allclose(gpu, cpu, atol=1e-4)
 
 
