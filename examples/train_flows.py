@@ -69,24 +69,39 @@ def synth_csv(path: str, rows: int = 20_000, seed: int = 7) -> None:
             f.write(",".join(row) + "\n")
 
 
-def build_pipeline(model: str, numeric_cols, n_classes_hint: int = 5) -> Pipeline:
-    stages = [
-        StringIndexer(
-            inputCols=CATEGORICAL,
-            outputCols=[c + "_idx" for c in CATEGORICAL],
-            handleInvalid="keep",
-        ),
-        OneHotEncoder(
-            inputCols=[c + "_idx" for c in CATEGORICAL],
-            outputCols=[c + "_vec" for c in CATEGORICAL],
-            handleInvalid="keep",
-        ),
-        VectorAssembler(
-            inputCols=[c + "_vec" for c in CATEGORICAL] + list(numeric_cols),
-            outputCol="features",
-        ),
-        StringIndexer(inputCol="label", outputCol="labelIdx"),
-    ]
+def build_pipeline(model: str, numeric_cols, n_classes_hint: int = 5,
+                   featurize: str = "onehot") -> Pipeline:
+    indexer = StringIndexer(
+        inputCols=CATEGORICAL,
+        outputCols=[c + "_idx" for c in CATEGORICAL],
+        handleInvalid="keep",
+    )
+    if featurize == "categorical" and model in ("rf", "gbt"):
+        # tree path: skip one-hot — the indexed columns carry nominal
+        # metadata, VectorAssembler marks those feature slots categorical,
+        # and the trees use Spark categorical split-sets on them
+        stages = [
+            indexer,
+            VectorAssembler(
+                inputCols=[c + "_idx" for c in CATEGORICAL] + list(numeric_cols),
+                outputCol="features",
+            ),
+            StringIndexer(inputCol="label", outputCol="labelIdx"),
+        ]
+    else:
+        stages = [
+            indexer,
+            OneHotEncoder(
+                inputCols=[c + "_idx" for c in CATEGORICAL],
+                outputCols=[c + "_vec" for c in CATEGORICAL],
+                handleInvalid="keep",
+            ),
+            VectorAssembler(
+                inputCols=[c + "_vec" for c in CATEGORICAL] + list(numeric_cols),
+                outputCol="features",
+            ),
+            StringIndexer(inputCol="label", outputCol="labelIdx"),
+        ]
     clf_args = dict(featuresCol="features", labelCol="labelIdx")
     if model == "lr":
         clf = LogisticRegression(maxIter=100, **clf_args)
@@ -112,6 +127,9 @@ def main():
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--rows", type=int, default=20_000, help="synthetic rows if no --data")
     p.add_argument("--device", default=None, help="cpu | cuda:N (default: auto)")
+    p.add_argument("--featurize", default="onehot", choices=["onehot", "categorical"],
+                   help="categorical: trees split indexed columns with "
+                        "Spark categorical split-sets instead of one-hot")
     args = p.parse_args()
 
     dist.init_from_env()
@@ -138,7 +156,7 @@ def main():
     train, test = df.randomSplit([1 - args.test_fraction, args.test_fraction], seed=args.seed)
 
     numeric = [c for c in df.columns if c not in CATEGORICAL + ["label"]]
-    pipe = build_pipeline(args.model, numeric)
+    pipe = build_pipeline(args.model, numeric, featurize=args.featurize)
     if args.model == "mlp":
         # resolve feature dim: fit featurization stages on a small head
         probe = Pipeline(stages=pipe.getStages()[:3]).fit(train.limit(512)).transform(train.limit(512))
@@ -151,7 +169,12 @@ def main():
     # map predicted indices back to label strings (IndexToString)
     from flowml.feature import IndexToString
 
-    label_stage = model.stages[3]  # the label StringIndexerModel
+    label_stage = next(
+        st for st in model.stages
+        if type(st).__name__ == "StringIndexerModel" and st.getOrDefault(
+            st.inputCol if st.isSet(st.inputCol) else st.inputCols
+        ) == "label"
+    )  # the label StringIndexerModel
     pred = IndexToString(
         inputCol="prediction", outputCol="predictedLabel", labels=label_stage.labels
     ).transform(pred)

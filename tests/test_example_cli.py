@@ -23,3 +23,19 @@ def test_cli_end_to_end(model, tmp_path):
     assert "test accuracy=" in r.stdout
     assert os.path.isdir(os.path.join(save, "metadata"))
     assert os.path.isdir(os.path.join(save, "stages"))
+
+
+def test_cli_categorical_featurization(tmp_path):
+    """--featurize categorical: trees split the indexed columns with
+    categorical split-sets (no one-hot stage in the pipeline)."""
+    save = str(tmp_path / "m_rf_cat")
+    r = subprocess.run(
+        [sys.executable, CLI, "--rows", "3000", "--model", "rf",
+         "--featurize", "categorical", "--save", save, "--device", "cpu"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert r.returncode == 0, r.stdout[-800:] + r.stderr[-800:]
+    assert "test accuracy=" in r.stdout
+    # saved pipeline has 4 stages (indexer, assembler, label indexer, rf)
+    stages = sorted(os.listdir(os.path.join(save, "stages")))
+    assert len(stages) == 4
