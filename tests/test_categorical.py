@@ -318,3 +318,29 @@ class TestVectorIndexer:
         model = Pipeline(stages=[vi, dt]).fit(df)
         out = model.transform(df)
         assert (out["prediction"].numpy() == y).mean() == 1.0
+
+
+class TestGBTCategoricalPersistence:
+    def test_gbt_round_trip_with_categorical_splits(self, tmp_path):
+        """GBT trees are regression trees internally — categorical splits
+        on them must persist (mean-centroid split-sets, width-3 stats AND
+        numCategories>0 together)."""
+        from flowml.classification import GBTClassificationModel
+
+        r = np.random.default_rng(19)
+        n, k = 3000, 4
+        cat = r.integers(0, k, n)
+        y = np.isin(cat, [1, 3]).astype(np.float64)
+        X = np.column_stack([cat, r.normal(size=n)]).astype(np.float32)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y}).withMeta(
+            "features", {"categorical": {0: k}})
+        m = GBTClassifier(maxIter=4, maxDepth=2, seed=0).fit(df)
+        assert any((t.cat_mask != 0).any() for t in m._trees)
+        p = str(tmp_path / "gbtcat")
+        m.save(p)
+        m2 = GBTClassificationModel.load(p)
+        assert all(t.regression for t in m2._trees)
+        for t1, t2 in zip(m._trees, m2._trees):
+            assert np.array_equal(t1.cat_mask, t2.cat_mask)
+        assert torch.allclose(m.transform(df)["rawPrediction"],
+                              m2.transform(df)["rawPrediction"], atol=1e-5)

@@ -473,3 +473,38 @@ def test_streaming_driver_round_robin_aggregates():
     assert res["rows"] == 10 * 256           # global stream fully covered
     assert res["rows_local"] == 5 * 256      # each of 2 ranks took half
     assert sum(res["class_counts"]) == 10 * 256
+
+
+def _vector_indexer_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.feature import VectorIndexer
+
+    r = np.random.default_rng(33)
+    n = 900
+    X = np.column_stack([
+        r.integers(0, 4, n).astype(np.float64),  # categorical
+        r.normal(size=n),                        # continuous
+    ]).astype(np.float32)
+    lo, hi = dist.shard_range(n, rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi])})
+    m = VectorIndexer(inputCol="features", outputCol="ix", maxCategories=5).fit(df)
+    gathered = dist.all_gather_object(m.categoryMaps)
+    assert all(g == gathered[0] for g in gathered), "categoryMaps diverged"
+    return m.categoryMaps
+
+
+def test_vector_indexer_dist_equals_single():
+    maps_dist = _run_dist("_vector_indexer_dist_body")
+    from flowml import FlowFrame
+    from flowml.feature import VectorIndexer
+
+    r = np.random.default_rng(33)
+    n = 900
+    X = np.column_stack([
+        r.integers(0, 4, n).astype(np.float64),
+        r.normal(size=n),
+    ]).astype(np.float32)
+    m = VectorIndexer(inputCol="features", outputCol="ix", maxCategories=5).fit(
+        FlowFrame({"features": torch.from_numpy(X)}))
+    assert maps_dist == m.categoryMaps
+    assert 0 in maps_dist and 1 not in maps_dist

@@ -39,3 +39,18 @@ def test_cli_categorical_featurization(tmp_path):
     # saved pipeline has 4 stages (indexer, assembler, label indexer, rf)
     stages = sorted(os.listdir(os.path.join(save, "stages")))
     assert len(stages) == 4
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("featurize", ["onehot", "categorical"])
+def test_cli_end_to_end_gpu(featurize, tmp_path):
+    """The app driver end-to-end on a real GPU (HIP kernel path)."""
+    save = str(tmp_path / f"m_gpu_{featurize}")
+    r = subprocess.run(
+        [sys.executable, CLI, "--rows", "20000", "--model", "rf",
+         "--featurize", featurize, "--save", save, "--device", "cuda:0"],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert r.returncode == 0, r.stdout[-800:] + r.stderr[-800:]
+    assert "test accuracy=" in r.stdout
+    assert os.path.isdir(os.path.join(save, "stages"))
