@@ -156,6 +156,56 @@ def main():
                      "interceptVector": dense_vec([0.0, 0.5]),
                      "coefficientMatrix": dense_mat([[1.0, -1.0], [0.5, 0.25]]),
                      "isMultinomial": True}])
+    # ---- GBTClassificationModel: internal regression trees ---------------
+    # width-3 impurityStats here are (sum, sum_sq, n) — the OTHER side of
+    # the width-3 ambiguity; numCategories=0 continuous root, leaf preds
+    # are means (s/n)
+    p = os.path.join(OUT, "gbt")
+    write_meta(p, "org.apache.spark.ml.classification.GBTClassificationModel",
+               "gbtc_5566778899aabbcc",
+               {"maxIter": 2, "maxDepth": 1, "stepSize": 0.1, "seed": 1},
+               {"maxIter": 20, "lossType": "logistic"})
+    rows = []
+    rows.append({"treeID": 0, "nodeData": node(
+        0, 0.125, 0.9375, [2.0, 4.5, 16.0], 0.5, 1, 2, 0, [1.5], 0)})
+    rows.append({"treeID": 0, "nodeData": node(
+        1, 0.75, 0.1, [6.0, 5.0, 8.0], -1.0, -1, -1, -1, [], -1)})
+    rows.append({"treeID": 0, "nodeData": node(
+        2, -0.5, 0.2, [-4.0, 2.5, 8.0], -1.0, -1, -1, -1, [], -1)})
+    rows.append({"treeID": 1, "nodeData": node(
+        0, 0.0625, 0.5, [1.0, 2.0, 16.0], 0.25, 1, 2, 1, [0.0, 2.0], 3)})
+    rows.append({"treeID": 1, "nodeData": node(
+        1, 0.25, 0.05, [2.5, 1.0, 10.0], -1.0, -1, -1, -1, [], -1)})
+    rows.append({"treeID": 1, "nodeData": node(
+        2, -0.25, 0.07, [-1.5, 1.0, 6.0], -1.0, -1, -1, -1, [], -1)})
+    write_data(p, rows)
+    tm = os.path.join(p, "treesMetadata")
+    os.makedirs(tm, exist_ok=True)
+    tmeta = [{"treeID": 0, "metadata": json.dumps({"numFeatures": 2}),
+              "weights": 1.0},
+             {"treeID": 1, "metadata": json.dumps({"numFeatures": 2}),
+              "weights": 0.1}]
+    pq.write_table(pa.Table.from_pylist(tmeta),
+                   os.path.join(tm, "part-00000.parquet"))
+    open(os.path.join(tm, "_SUCCESS"), "w").close()
+
+    # ---- OneHotEncoderModel ----------------------------------------------
+    p = os.path.join(OUT, "onehot")
+    write_meta(p, "org.apache.spark.ml.feature.OneHotEncoderModel",
+               "oneHot_99aabbccddeeff00",
+               {"inputCols": ["pIdx", "sIdx"], "outputCols": ["pVec", "sVec"]},
+               {"dropLast": True, "handleInvalid": "error"})
+    write_data(p, [{"categorySizes": [3, 6]}])
+
+    # ---- StandardScalerModel ---------------------------------------------
+    p = os.path.join(OUT, "scaler")
+    write_meta(p, "org.apache.spark.ml.feature.StandardScalerModel",
+               "stdScal_1234abcd5678ef90",
+               {"inputCol": "features", "outputCol": "scaled", "withMean": True},
+               {"withMean": False, "withStd": True})
+    write_data(p, [{"std": dense_vec([1.0, 2.0, 0.5]),
+                    "mean": dense_vec([-1.0, 0.25, 3.0])}])
+
     print(f"golden fixtures written under {OUT}")
 
 

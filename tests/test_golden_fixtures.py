@@ -97,3 +97,31 @@ class TestGoldenLoad:
         assert "flowml" not in src.replace("flowml's writer", "").replace(
             "flowml.io.load", "").replace("no flowml imports", "").replace(
             "flowml imports", "")
+
+    def test_gbt_regression_trees(self):
+        m = load(_p("gbt"))
+        assert type(m).__name__ == "GBTClassificationModel"
+        assert m.treeWeights == [1.0, 0.1]
+        t0, t1 = m._trees
+        # width-3 stats on a GBT tree mean (s, s2, n) — regression kind
+        assert t0.regression and t1.regression
+        assert t0.count[0] == 16.0            # n = stats[2], not sum
+        assert t0.prediction(1) == pytest.approx(6.0 / 8.0)
+        # tree 1 root is a categorical split {0,2} on feature 1
+        assert t1.cat_mask[0] == 0b101 and t1.num_cats[0] == 3
+        # margin check: x routes tree0 right (f0=2>1.5), tree1 left (f1=0)
+        from flowml.frame import FlowFrame
+        X = torch.tensor([[2.0, 0.0]], dtype=torch.float32)
+        raw = m.transform(FlowFrame({"features": X}))["rawPrediction"][0]
+        margin = 1.0 * (-0.5) + 0.1 * 0.25
+        # flowml convention (gbt.py docstring): predictRaw = [-2F, 2F]
+        assert float(raw[1] - raw[0]) == pytest.approx(4 * margin, abs=1e-6)
+
+    def test_onehot_and_scaler(self):
+        m = load(_p("onehot"))
+        assert m.categorySizes == [3, 6]
+        assert m.getOrDefault(m.outputCols) == ["pVec", "sVec"]
+        sc = load(_p("scaler"))
+        assert np.array_equal(sc.std.numpy(), [1.0, 2.0, 0.5])
+        assert np.array_equal(sc.mean.numpy(), [-1.0, 0.25, 3.0])
+        assert sc.getWithMean() is True  # paramMap overrode the default
