@@ -11,8 +11,8 @@ checkpoint of the ensemble-so-far + RNG state, resumable via ``initModel``).
 
 Spark semantics preserved: leaf value = mean pseudo-residual of the leaf
 (no Newton refinement), margin F(x) = sum(stepSize_i * tree_i(x)),
-probability = sigmoid(2 F(x)), prediction by sign. predictRaw = [-2F, 2F]
-(spark GBTClassificationModel.margin convention).
+probability = sigmoid(2 F(x)), prediction by sign. predictRaw = [-F, F]
+(spark GBTClassificationModel.predictRaw convention).
 """
 
 from __future__ import annotations
@@ -343,11 +343,12 @@ class GBTClassificationModel(_GBTParams, _HasTrainingSummary, ProbabilisticClass
         return self._ensemble.predict_raw(X)[:, 0]
 
     def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
-        m = 2.0 * self._margin(features)
+        m = self._margin(features)
         return torch.stack([-m, m], dim=1)
 
     def raw2probability(self, raw: torch.Tensor) -> torch.Tensor:
-        p1 = torch.sigmoid(raw[:, 1])
+        # Spark LogLoss.computeProbability: p1 = 1 / (1 + exp(-2 margin))
+        p1 = torch.sigmoid(2.0 * raw[:, 1])
         return torch.stack([1 - p1, p1], dim=1)
 
     def write(self):

@@ -114,8 +114,8 @@ class TestGoldenLoad:
         X = torch.tensor([[2.0, 0.0]], dtype=torch.float32)
         raw = m.transform(FlowFrame({"features": X}))["rawPrediction"][0]
         margin = 1.0 * (-0.5) + 0.1 * 0.25
-        # flowml convention (gbt.py docstring): predictRaw = [-2F, 2F]
-        assert float(raw[1] - raw[0]) == pytest.approx(4 * margin, abs=1e-6)
+        # Spark convention: predictRaw = [-F, F]
+        assert float(raw[1] - raw[0]) == pytest.approx(2 * margin, abs=1e-6)
 
     def test_onehot_and_scaler(self):
         m = load(_p("onehot"))
