@@ -422,7 +422,11 @@ def main():
     if torch.cuda.is_available():
         import os
 
-        device = torch.device(f"cuda:{int(os.environ.get('LOCAL_RANK', 0))}")
+        # modulo: lets world > n_gpus oversubscribe ranks onto devices —
+        # used to exercise a real multi-rank RCCL communicator on a 1-GPU
+        # box (correctness de-risk; throughput numbers then share the GPU)
+        idx = int(os.environ.get("LOCAL_RANK", 0)) % torch.cuda.device_count()
+        device = torch.device(f"cuda:{idx}")
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")
