@@ -128,33 +128,11 @@ class TestRcclWorld1:
         MaxAbsScaler(inputCol="vec", outputCol="o").fit(df)
 
 
-class TestRcclWorld2OneGpu:
-    def test_bench_lr_world2_shared_gpu(self):
-        """A REAL multi-rank RCCL communicator on one GPU: two ranks share
-        cuda:0 (RCCL supports >1 rank per device) and the LR bench's
-        gradient all_reduce crosses ranks for real. Validates the world>1
-        collective path on hardware without an 8-GPU lease. Throughput is
-        meaningless here (shared GPU) — only completion and rank agreement
-        are asserted."""
-        if not torch.cuda.is_available():
-            pytest.skip("needs GPU")
-        repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
-        env = dict(os.environ)
-        env.pop("RANK", None), env.pop("WORLD_SIZE", None)
-        r = subprocess.run(
-            [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-             "--nproc-per-node=2", "--master-addr=127.0.0.1",
-             f"--master-port={_free_port()}", "bench.py", "--gpus", "2",
-             "--model", "lr", "--steps", "3", "--warmup", "1",
-             "--rows", "50000"],
-            cwd=repo, env=env, capture_output=True, text=True, timeout=600,
-        )
-        assert r.returncode == 0, r.stdout[-1500:] + r.stderr[-1500:]
-        line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
-        out = json.loads(line)
-        assert out["n_gpus"] == 2
-        assert out["config"]["parallelism"] == "dp2"
-        assert out["config"]["global_rows"] == 100000
+# NOTE: a world=2 communicator on ONE GPU was tried and is NOT possible:
+# RCCL (like NCCL) rejects two ranks on the same device ("duplicate GPU",
+# ncclInvalidUsage) — verified on the MI355X box. Multi-rank RCCL semantics
+# are covered by the gloo world=2/4/8 suites; the live-RCCL coverage above
+# (world=1, incl. in-graph) is the maximum a 1-GPU lease allows.
 
 
 class TestTorchrunWorld1:
