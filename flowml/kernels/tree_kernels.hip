@@ -163,12 +163,109 @@ extern "C" __global__ void flowml_tree_hist_cls(
     for (int i = 0; i < 8; ++i) cdf[i] = bag_cdf[i];
   }
 
+  if (!staged) {
+    // Two rows per thread, states resolved side by side: one row's
+    // nid -> lut -> bag -> gathers is a serial dependent-load chain, and
+    // r1 PMC showed the kernel gather-LATENCY bound (zero LDS bank
+    // conflicts, 8 waves/CU) — pairing rows doubles the independent
+    // chains in flight per thread.
+    for (long long sub = r0; sub < r1; sub += 2LL * blockDim.x) {
+      const long long rA = sub + threadIdx.x;
+      const long long rB = sub + blockDim.x + threadIdx.x;
+      const uint8_t* crowA = codes + rA * F;
+      const uint8_t* crowB = codes + rB * F;
+      int gA = -1, gB = -1, labA = 0, labB = 0;
+      float wA = 1.f, wB = 1.f;
+      if (rA < r1) {
+        int nid = nid_t[rA];
+        if (advancing && nid >= 0) {
+          const int f = adv_feat[nid];
+          // rows KEEP their node id at leaves (lut[leaf] = -1 already
+          // skips them) so the per-leaf stats passes can address them
+          if (f >= 0) {
+            const long long cm = adv_cat ? adv_cat[nid] : 0LL;
+            nid = flowml_left_code(crowA[f], adv_thr[nid], cm)
+                      ? adv_left[nid] : adv_right[nid];
+            nid_t[rA] = nid;
+          }
+        }
+        if (cnt && nid >= 0) gA = lut[nid];
+      }
+      if (rB < r1) {
+        int nid = nid_t[rB];
+        if (advancing && nid >= 0) {
+          const int f = adv_feat[nid];
+          if (f >= 0) {
+            const long long cm = adv_cat ? adv_cat[nid] : 0LL;
+            nid = flowml_left_code(crowB[f], adv_thr[nid], cm)
+                      ? adv_left[nid] : adv_right[nid];
+            nid_t[rB] = nid;
+          }
+        }
+        if (cnt && nid >= 0) gB = lut[nid];
+      }
+      if (bag_mode >= 2) {
+        if (gA >= 0) {
+          wA = flowml_hash_bag(bag_mode, cdf, bag_seed, row_offset + rA, t);
+          if (wA == 0.f) gA = -1;
+        }
+        if (gB >= 0) {
+          wB = flowml_hash_bag(bag_mode, cdf, bag_seed, row_offset + rB, t);
+          if (wB == 0.f) gB = -1;
+        }
+      }
+      if (bag_t) {
+        if (gA >= 0) {
+          const float bw = bag_t[rA];
+          if (bw == 0.f) gA = -1; else wA *= bw;
+        }
+        if (gB >= 0) {
+          const float bw = bag_t[rB];
+          if (bw == 0.f) gB = -1; else wB *= bw;
+        }
+      }
+      if (gA >= 0) labA = labels[rA];
+      if (gB >= 0) labB = labels[rB];
+      const int* fsA = (fs && gA >= 0) ? fs + (long long)gA * Fp : nullptr;
+      const int* fsB = (fs && gB >= 0) ? fs + (long long)gB * Fp : nullptr;
+      // branch on the address space (not a ternary pointer): keeps the LDS
+      // path's atomic a ds_add_f32, not a generic flat_atomic_add_f32
+      if (use_lds) {
+        float* hrowA = gA >= 0 ? s_hist + (long long)(gA - g_lo) * stride_fbc
+                               : nullptr;
+        float* hrowB = gB >= 0 ? s_hist + (long long)(gB - g_lo) * stride_fbc
+                               : nullptr;
+        for (int fp = 0; fp < Fp; ++fp) {
+          if (gA >= 0) {
+            const int f = fsA ? fsA[fp] : fp;
+            atomicAdd(&hrowA[(fp * B + crowA[f]) * C + labA], wA);
+          }
+          if (gB >= 0) {
+            const int f = fsB ? fsB[fp] : fp;
+            atomicAdd(&hrowB[(fp * B + crowB[f]) * C + labB], wB);
+          }
+        }
+      } else {
+        float* hrowA = gA >= 0 ? out + (long long)gA * stride_fbc : nullptr;
+        float* hrowB = gB >= 0 ? out + (long long)gB * stride_fbc : nullptr;
+        for (int fp = 0; fp < Fp; ++fp) {
+          if (gA >= 0) {
+            const int f = fsA ? fsA[fp] : fp;
+            atomicAdd(&hrowA[(fp * B + crowA[f]) * C + labA], wA);
+          }
+          if (gB >= 0) {
+            const int f = fsB ? fsB[fp] : fp;
+            atomicAdd(&hrowB[(fp * B + crowB[f]) * C + labB], wB);
+          }
+        }
+      }
+    }
+  } else
   for (long long sub = r0; sub < r1; sub += blockDim.x) {
     const int rows_in = (int)min((long long)blockDim.x, r1 - sub);
     const long long r = sub + threadIdx.x;
-    const uint8_t* crow;
-    if (staged) {
-      __syncthreads();  // previous sub-tile fully consumed
+    __syncthreads();  // previous sub-tile fully consumed
+    {
       const uint8_t* src = codes + sub * F;
       const int total = rows_in * F;
       const int total16 = total & ~15;
@@ -178,18 +275,13 @@ extern "C" __global__ void flowml_tree_hist_cls(
       for (int i = total16 + threadIdx.x; i < total; i += blockDim.x)
         s_codes[i] = src[i];
       __syncthreads();
-      crow = s_codes + threadIdx.x * F;
-    } else {
-      crow = codes + r * F;
     }
+    const uint8_t* crow = s_codes + threadIdx.x * F;
     if (threadIdx.x >= rows_in) continue;
     int nid = nid_t[r];
     if (advancing) {
       if (nid >= 0) {
         const int f = adv_feat[nid];
-        // rows KEEP their node id at leaves (lut[leaf] = -1 already skips
-        // them in the histograms) so the final per-leaf stats pass can
-        // still address every row
         if (f >= 0) {
           const long long cm = adv_cat ? adv_cat[nid] : 0LL;
           nid = flowml_left_code(crow[f], adv_thr[nid], cm) ? adv_left[nid]
@@ -214,8 +306,6 @@ extern "C" __global__ void flowml_tree_hist_cls(
     }
     const int lab = labels[r];
     const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
-    // branch on the address space (not a ternary pointer): keeps the LDS
-    // path's atomic a ds_add_f32 instead of a generic flat_atomic_add_f32
     if (use_lds) {
       float* hrow = s_hist + (long long)(g - g_lo) * stride_fbc;
       for (int fp = 0; fp < Fp; ++fp) {
