@@ -383,12 +383,124 @@ extern "C" __global__ void flowml_tree_hist_reg(
     for (int i = 0; i < 8; ++i) cdf[i] = bag_cdf[i];
   }
 
+  if (!staged) {
+    // paired rows: two independent nid->lut->bag->gather chains in flight
+    // per thread (same rationale and structure as the cls kernel)
+    for (long long sub = r0; sub < r1; sub += 2LL * blockDim.x) {
+      const long long rA = sub + threadIdx.x;
+      const long long rB = sub + blockDim.x + threadIdx.x;
+      const uint8_t* crowA = codes + rA * F;
+      const uint8_t* crowB = codes + rB * F;
+      int gA = -1, gB = -1;
+      float wA = 1.f, wB = 1.f;
+      if (rA < r1) {
+        int nid = nid_t[rA];
+        if (advancing && nid >= 0) {
+          const int f = adv_feat[nid];
+          if (f >= 0) {  // keep-at-leaf (see cls kernel)
+            const long long cm = adv_cat ? adv_cat[nid] : 0LL;
+            nid = flowml_left_code(crowA[f], adv_thr[nid], cm)
+                      ? adv_left[nid] : adv_right[nid];
+            nid_t[rA] = nid;
+          }
+        }
+        if (cnt && nid >= 0) gA = lut[nid];
+      }
+      if (rB < r1) {
+        int nid = nid_t[rB];
+        if (advancing && nid >= 0) {
+          const int f = adv_feat[nid];
+          if (f >= 0) {
+            const long long cm = adv_cat ? adv_cat[nid] : 0LL;
+            nid = flowml_left_code(crowB[f], adv_thr[nid], cm)
+                      ? adv_left[nid] : adv_right[nid];
+            nid_t[rB] = nid;
+          }
+        }
+        if (cnt && nid >= 0) gB = lut[nid];
+      }
+      if (bag_mode >= 2) {
+        if (gA >= 0) {
+          wA = flowml_hash_bag(bag_mode, cdf, bag_seed, row_offset + rA, t);
+          if (wA == 0.f) gA = -1;
+        }
+        if (gB >= 0) {
+          wB = flowml_hash_bag(bag_mode, cdf, bag_seed, row_offset + rB, t);
+          if (wB == 0.f) gB = -1;
+        }
+      }
+      if (bag_t) {
+        if (gA >= 0) {
+          const float bw = bag_t[rA];
+          if (bw == 0.f) gA = -1; else wA *= bw;
+        }
+        if (gB >= 0) {
+          const float bw = bag_t[rB];
+          if (bw == 0.f) gB = -1; else wB *= bw;
+        }
+      }
+      float wtA = 0.f, wt2A = 0.f, wtB = 0.f, wt2B = 0.f;
+      if (gA >= 0) {
+        const float tv = targets[rA];
+        wtA = wA * tv;
+        wt2A = wtA * tv;
+      }
+      if (gB >= 0) {
+        const float tv = targets[rB];
+        wtB = wB * tv;
+        wt2B = wtB * tv;
+      }
+      const int* fsA = (fs && gA >= 0) ? fs + (long long)gA * Fp : nullptr;
+      const int* fsB = (fs && gB >= 0) ? fs + (long long)gB * Fp : nullptr;
+      // branch on address space: ds_add_f32 on the LDS path (see cls)
+      if (use_lds) {
+        float* hrowA = gA >= 0 ? s_hist + (long long)(gA - g_lo) * stride_fb2
+                               : nullptr;
+        float* hrowB = gB >= 0 ? s_hist + (long long)(gB - g_lo) * stride_fb2
+                               : nullptr;
+        for (int fp = 0; fp < Fp; ++fp) {
+          if (gA >= 0) {
+            const int f = fsA ? fsA[fp] : fp;
+            float* cell = &hrowA[(fp * B + crowA[f]) * 2];
+            atomicAdd(cell + 0, wtA);
+            atomicAdd(cell + 1, wA);
+          }
+          if (gB >= 0) {
+            const int f = fsB ? fsB[fp] : fp;
+            float* cell = &hrowB[(fp * B + crowB[f]) * 2];
+            atomicAdd(cell + 0, wtB);
+            atomicAdd(cell + 1, wB);
+          }
+        }
+        if (gA >= 0) atomicAdd(&s_hist[l_max * stride_fb2 + (gA - g_lo)], wt2A);
+        if (gB >= 0) atomicAdd(&s_hist[l_max * stride_fb2 + (gB - g_lo)], wt2B);
+      } else {
+        float* hrowA = gA >= 0 ? out + (long long)gA * stride_fb2 : nullptr;
+        float* hrowB = gB >= 0 ? out + (long long)gB * stride_fb2 : nullptr;
+        for (int fp = 0; fp < Fp; ++fp) {
+          if (gA >= 0) {
+            const int f = fsA ? fsA[fp] : fp;
+            float* cell = &hrowA[(fp * B + crowA[f]) * 2];
+            atomicAdd(cell + 0, wtA);
+            atomicAdd(cell + 1, wA);
+          }
+          if (gB >= 0) {
+            const int f = fsB ? fsB[fp] : fp;
+            float* cell = &hrowB[(fp * B + crowB[f]) * 2];
+            atomicAdd(cell + 0, wtB);
+            atomicAdd(cell + 1, wB);
+          }
+        }
+        if (gA >= 0) atomicAdd(&out[(long long)Gtot * stride_fb2 + gA], wt2A);
+        if (gB >= 0) atomicAdd(&out[(long long)Gtot * stride_fb2 + gB], wt2B);
+      }
+    }
+  } else
   for (long long sub = r0; sub < r1; sub += blockDim.x) {
     const int rows_in = (int)min((long long)blockDim.x, r1 - sub);
     const long long r = sub + threadIdx.x;
-    const uint8_t* crow;
-    if (staged) {
-      __syncthreads();
+    __syncthreads();
+    {
       const uint8_t* src = codes + sub * F;
       const int total = rows_in * F;
       const int total16 = total & ~15;
@@ -398,10 +510,8 @@ extern "C" __global__ void flowml_tree_hist_reg(
       for (int i = total16 + threadIdx.x; i < total; i += blockDim.x)
         s_codes[i] = src[i];
       __syncthreads();
-      crow = s_codes + threadIdx.x * F;
-    } else {
-      crow = codes + r * F;
     }
+    const uint8_t* crow = s_codes + threadIdx.x * F;
     if (threadIdx.x >= rows_in) continue;
     int nid = nid_t[r];
     if (advancing) {
@@ -433,7 +543,6 @@ extern "C" __global__ void flowml_tree_hist_reg(
     const float wt = w * tv;
     const float wt2 = wt * tv;
     const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
-    // branch on address space: ds_add_f32 on the LDS path (see cls kernel)
     if (use_lds) {
       const int gl = g - g_lo;
       float* hrow = s_hist + (long long)gl * stride_fb2;
