@@ -235,6 +235,32 @@ class FlatTree:
         self._n += 1
         return i
 
+    @classmethod
+    def from_arrays(cls, feature, thr_bin, thr_raw, left, right, impurity,
+                    gain, count, stats, cat_mask, num_cats,
+                    regression: bool = False) -> "FlatTree":
+        """Adopt pre-built per-node arrays WITHOUT copying (they may be
+        views into a shared arena gather). Trees built this way are
+        read-only by convention — the grower never appends post-extract."""
+        t = cls.__new__(cls)
+        n = len(feature)
+        t._n = n
+        t._cap = n
+        t._sw = stats.shape[1] if stats.ndim == 2 else 0
+        t.regression = bool(regression)
+        t._feature = feature
+        t._thr_bin = thr_bin
+        t._thr_raw = thr_raw
+        t._left = left
+        t._right = right
+        t._impurity = impurity
+        t._gain = gain
+        t._count = count
+        t._stats = stats
+        t._cat_mask = cat_mask
+        t._num_cats = num_cats
+        return t
+
     def bulk_add(self, stats2d: np.ndarray, imps: np.ndarray, cnts: np.ndarray) -> int:
         """Append k nodes; returns the base id (new nodes are base..base+k-1)."""
         k = len(imps)
@@ -1108,18 +1134,13 @@ def grow_trees(
     trees: List[FlatTree] = []
     for t in range(T):
         sl = slice(bounds[t], bounds[t + 1])
-        n = bounds[t + 1] - bounds[t]
-        tr = FlatTree(stat_width=S, cap=max(int(n), 4), regression=not is_cls)
-        tr.bulk_add(st_o[sl], imp_o[sl], cnt_o[sl])
-        tr.feature[:] = feat_o[sl]
-        tr.thr_bin[:] = thb_o[sl]
-        tr.thr_raw[:] = thr_o[sl]
-        tr.gain[:] = gain_o[sl]
-        tr.cat_mask[:] = catm_o[sl]
-        tr.num_cats[:] = ncat_o[sl]
-        tr.left[:] = left_o[sl]
-        tr.right[:] = right_o[sl]
-        trees.append(tr)
+        # adopt contiguous slices of the one arena gather as views — no
+        # per-tree copies (was ~0.5 ms of the flagship fit across 100 trees)
+        trees.append(FlatTree.from_arrays(
+            feat_o[sl], thb_o[sl], thr_o[sl], left_o[sl], right_o[sl],
+            imp_o[sl], gain_o[sl], cnt_o[sl], st_o[sl], catm_o[sl],
+            ncat_o[sl], regression=not is_cls,
+        ))
     _acc("extract", t0)
 
     if timing:
