@@ -33,8 +33,11 @@ class FlowFrame:
         # per-column metadata (the stand-in for Spark's ML column attributes:
         # e.g. {"categorical": {featIdx: arity}} on a features vector, or
         # {"nominal": numLabels} on a StringIndexer output column)
+        import copy as _copy
+
         self._meta: Dict[str, dict] = {
-            k: dict(v) for k, v in (meta or {}).items() if k in columns and v
+            k: _copy.deepcopy(v) for k, v in (meta or {}).items()
+            if k in columns and v
         }
         self.device = torch.device(device) if device is not None else torch.device("cpu")
         # device is honored, not just recorded: tensor columns are moved so
@@ -76,8 +79,11 @@ class FlowFrame:
     # -- column metadata ------------------------------------------------------
 
     def meta(self, name: str) -> dict:
-        """Metadata dict attached to a column ({} if none)."""
-        return dict(self._meta.get(name, {}))
+        """Metadata dict attached to a column ({} if none). Deep-copied —
+        mutating the result never changes the frame."""
+        import copy as _copy
+
+        return _copy.deepcopy(self._meta.get(name, {}))
 
     def withMeta(self, name: str, meta: dict) -> "FlowFrame":
         """Same frame with ``meta`` attached to column ``name``."""

@@ -151,3 +151,61 @@ class TestCsvMissingValues:
         assert out.col("b2").tolist() == [3.5, 2.5, 4.5]
         m = StringIndexer(inputCol="proto", outputCol="p").fit(out)
         assert m.transform(out).col("p").tolist() == [0.0, 1.0, 0.0]
+
+
+class TestColumnMetadata:
+    """Per-column metadata (the ML-attribute stand-in) must survive every
+    frame transformation that keeps the column."""
+
+    def _f(self):
+        import torch
+        from flowml.frame import FlowFrame
+
+        f = FlowFrame({"a": torch.arange(6.0), "b": torch.ones(6)})
+        return f.withMeta("a", {"categorical": {0: 3}})
+
+    def test_basic_get_set(self):
+        f = self._f()
+        assert f.meta("a") == {"categorical": {0: 3}}
+        assert f.meta("b") == {}
+        # withMeta with empty dict clears
+        assert f.withMeta("a", {}).meta("a") == {}
+        import pytest as _pt
+        with _pt.raises(KeyError):
+            f.withMeta("nope", {"x": 1})
+
+    def test_propagation(self):
+        import numpy as np
+        import torch
+
+        f = self._f()
+        m = {"categorical": {0: 3}}
+        assert f.select("a").meta("a") == m
+        assert f.drop("b").meta("a") == m
+        assert f.withColumnRenamed("a", "z").meta("z") == m
+        assert f.filter(torch.tensor([1, 0, 1, 0, 1, 0], dtype=torch.bool)).meta("a") == m
+        assert f.take_rows(np.array([0, 2])).meta("a") == m
+        assert f.limit(3).meta("a") == m
+        assert f.union(f).meta("a") == m
+        assert f.to("cpu").meta("a") == m
+        sp = f.randomSplit([0.5, 0.5], seed=1)
+        assert sp[0].meta("a") == m and sp[1].meta("a") == m
+
+    def test_replacement_clears_stale_meta(self):
+        import torch
+
+        f = self._f()
+        # replacing the column's data drops its (now stale) metadata
+        assert f.withColumn("a", torch.zeros(6)).meta("a") == {}
+        # other columns keep theirs
+        assert f.withColumn("c", torch.zeros(6)).meta("a") != {}
+        # explicit meta on withColumn sticks
+        f2 = f.withColumn("c", torch.zeros(6), meta={"nominal": 4})
+        assert f2.meta("c") == {"nominal": 4}
+
+    def test_meta_is_copied_not_aliased(self):
+        f = self._f()
+        d = f.meta("a")
+        d["categorical"][0] = 99   # nested mutation
+        d["new"] = 1               # top-level mutation
+        assert f.meta("a") == {"categorical": {0: 3}}  # deep-copied both ways
