@@ -237,7 +237,10 @@ def bench_mlp(args, device):
             torch.cuda.current_stream(device).wait_stream(s)
             torch.cuda.synchronize(device)
             graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(graph):
+            # thread_local capture mode: the RCCL watchdog thread polls
+            # work events concurrently, which global capture mode forbids
+            # (hipErrorStreamCaptureUnsupported race, seen live)
+            with torch.cuda.graph(graph, capture_error_mode="thread_local"):
                 _train_ops()
         except Exception as e:  # noqa: BLE001
             print(f"# hipGraph capture unavailable ({e}); eager steps", flush=True)

@@ -99,7 +99,9 @@ class StreamingClassifier:
         torch.cuda.current_stream(self.device).wait_stream(s)
         torch.cuda.synchronize(self.device)
         self._graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self._graph):
+        # thread_local: lets the RCCL watchdog (if a process group is
+        # live) query its work events while we capture
+        with torch.cuda.graph(self._graph, capture_error_mode="thread_local"):
             self._static_raw, self._static_pred = self._forward()
 
     @torch.no_grad()

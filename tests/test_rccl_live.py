@@ -78,7 +78,7 @@ class TestRcclWorld1:
         torch.cuda.synchronize()
         acc.zero_()
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
+        with torch.cuda.graph(g, capture_error_mode="thread_local"):
             work()
         acc.zero_()
         for _ in range(5):
@@ -105,7 +105,7 @@ class TestRcclWorld1:
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         g = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g):
+        with torch.cuda.graph(g, capture_error_mode="thread_local"):
             work()
         a.fill_(2.0)
         b.fill_(-1.5)
