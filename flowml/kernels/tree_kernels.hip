@@ -35,6 +35,7 @@
 // host-side chunker (_hist_lds_cap).
 
 #include <hip/hip_runtime.h>
+#include <limits.h>
 #include <stdint.h>
 #include <stdlib.h>
 
@@ -328,6 +329,100 @@ extern "C" __global__ void flowml_tree_hist_cls(
   for (int i = threadIdx.x; i < flush_sz; i += blockDim.x) {
     const float v = s_hist[i];
     if (v != 0.f) atomicAdd(&out_t[i], v);
+  }
+}
+
+// Multi-tree-per-block variant (LDS mode, unstaged): TCC counters showed
+// the one-tree-per-block kernel L2-REQUEST bound (96.6% hit rate, ~103G
+// req/s) — every tree re-fetches each row's 41-B code line. One block
+// handling `tpb` trees reuses the row line (L1-hot) across tpb visits,
+// dividing the code-line request stream by ~tpb. Chosen by the launcher
+// when tpb trees' chunk histograms fit the 150 KB LDS carve together.
+extern "C" __global__ void flowml_tree_hist_cls_tpb(
+    const uint8_t* __restrict__ codes, int* __restrict__ nodeid,
+    const int* __restrict__ lut, const int* __restrict__ tree_lo,
+    const int* __restrict__ tree_cnt, const uint8_t* __restrict__ labels,
+    const float* __restrict__ bag, const int* __restrict__ fs,
+    float* __restrict__ out, const int* __restrict__ adv_feat,
+    const int* __restrict__ adv_thr, const int* __restrict__ adv_left,
+    const int* __restrict__ adv_right,
+    const long long* __restrict__ adv_cat,
+    int bag_mode, const float* __restrict__ bag_cdf,
+    unsigned long long bag_seed, long long row_offset,
+    long long N, int F, int T, int tpb, int Fp, int B, int C,
+    long long rows_per_block) {
+  extern __shared__ float s_hist[];
+  const int t0 = blockIdx.x * tpb;
+  const int t_hi = min(t0 + tpb, T);
+  const bool advancing = adv_feat != nullptr;
+  // contiguous group span of this block's trees (tree_lo of a cnt==0 tree
+  // is 0 by host convention — guard on cnt)
+  int g_base = INT_MAX, g_end = 0;
+  for (int tt = t0; tt < t_hi; ++tt) {
+    const int c = tree_cnt[tt];
+    if (c > 0) {
+      g_base = min(g_base, tree_lo[tt]);
+      g_end = max(g_end, tree_lo[tt] + c);
+    }
+  }
+  const int n_groups = (g_base == INT_MAX) ? 0 : g_end - g_base;
+  if (n_groups == 0 && !advancing) return;
+  const int stride_fbc = Fp * B * C;
+  const int hist_sz = n_groups * stride_fbc;
+  for (int i = threadIdx.x; i < hist_sz; i += blockDim.x) s_hist[i] = 0.f;
+  __syncthreads();
+  const long long r0 = blockIdx.y * rows_per_block;
+  const long long r1 = min(r0 + rows_per_block, N);
+  float cdf[8];
+  if (bag_mode >= 2) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i) cdf[i] = bag_cdf[i];
+  }
+  for (long long r = r0 + threadIdx.x; r < r1; r += blockDim.x) {
+    const uint8_t* crow = codes + r * F;
+    for (int tt = t0; tt < t_hi; ++tt) {
+      int* nt = nodeid + (long long)tt * N;
+      int nid = nt[r];
+      if (advancing) {
+        if (nid >= 0) {
+          const int f = adv_feat[nid];
+          if (f >= 0) {  // keep-at-leaf
+            const long long cm = adv_cat ? adv_cat[nid] : 0LL;
+            nid = flowml_left_code(crow[f], adv_thr[nid], cm) ? adv_left[nid]
+                                                              : adv_right[nid];
+            nt[r] = nid;
+          }
+        }
+        if (n_groups == 0) continue;
+      }
+      if (nid < 0) continue;
+      const int g = lut[nid];
+      if (g < 0) continue;
+      float w = 1.f;
+      if (bag_mode >= 2) {
+        w = flowml_hash_bag(bag_mode, cdf, bag_seed, row_offset + r, tt);
+        if (w == 0.f) continue;
+      }
+      if (bag) {
+        const float bw = bag[(long long)tt * N + r];
+        if (bw == 0.f) continue;
+        w *= bw;
+      }
+      const int lab = labels[r];
+      const int* fs_g = fs ? fs + (long long)g * Fp : nullptr;
+      float* hrow = s_hist + (long long)(g - g_base) * stride_fbc;
+      for (int fp = 0; fp < Fp; ++fp) {
+        const int f = fs_g ? fs_g[fp] : fp;
+        atomicAdd(&hrow[(fp * B + crow[f]) * C + lab], w);
+      }
+    }
+  }
+  if (n_groups == 0) return;
+  __syncthreads();
+  float* out_b = out + (long long)g_base * stride_fbc;
+  for (int i = threadIdx.x; i < hist_sz; i += blockDim.x) {
+    const float v = s_hist[i];
+    if (v != 0.f) atomicAdd(&out_b[i], v);
   }
 }
 
@@ -832,6 +927,30 @@ extern "C" void launch_tree_hist_cls(
   static const int cfg_stage = flowml_env_int("FLOWML_HIST_STAGE", 0);
   int stage_bytes = cfg_stage ? block * F : 0;
   if (hist_b + (size_t)stage_bytes > 150 * 1024) stage_bytes = 0;  // fall back
+  // multi-tree blocks when several trees' chunk hists share the LDS carve:
+  // reuses each row's code line across tpb tree visits (L2-request bound
+  // kernel — see profiles/pmc_r02_summary.md)
+  static const int cfg_tpb = flowml_env_int("FLOWML_HIST_TPB", 8);
+  int tpb = 1;
+  if (l_max > 0 && stage_bytes == 0 && hist_b > 0 && cfg_tpb > 1) {
+    // keep 2 blocks/CU resident (<=75 KB/block): halving occupancy for
+    // tpb=2 at the deepest level measured worse than the traffic saved
+    tpb = (int)((75 * 1024) / hist_b);
+    if (tpb > cfg_tpb) tpb = cfg_tpb;
+    if (tpb > T) tpb = T;
+    if (tpb < 1) tpb = 1;
+  }
+  if (tpb > 1) {
+    const int xblocks = (T + tpb - 1) / tpb;
+    const size_t lds2 = hist_b * tpb;
+    hipLaunchKernelGGL(flowml_tree_hist_cls_tpb, dim3(xblocks, tiles),
+                       dim3(block), lds2, (hipStream_t)stream, codes, nodeid,
+                       lut, tree_lo, tree_cnt, labels, bag, fs, out, adv_feat,
+                       adv_thr, adv_left, adv_right, adv_cat, bag_mode,
+                       bag_cdf, bag_seed, row_offset, N, F, T, tpb, Fp, B, C,
+                       rows_per_block);
+    return;
+  }
   const size_t lds = hist_b + stage_bytes;
   hipLaunchKernelGGL(flowml_tree_hist_cls, dim3(T, tiles), dim3(block), lds,
                      (hipStream_t)stream, codes, nodeid, lut, tree_lo, tree_cnt,
