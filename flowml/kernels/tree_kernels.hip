@@ -930,7 +930,11 @@ extern "C" void launch_tree_hist_cls(
   // multi-tree blocks when several trees' chunk hists share the LDS carve:
   // reuses each row's code line across tpb tree visits (L2-request bound
   // kernel — see profiles/pmc_r02_summary.md)
-  static const int cfg_tpb = flowml_env_int("FLOWML_HIST_TPB", 8);
+  // measured WORSE at default (levels 0-2: 2.85 -> 3.1-3.4 ms): the 96.6%
+  // L2 hit rate already amortizes cross-tree code-line reuse, and the
+  // serial per-row tree loop adds more latency than the saved requests.
+  // Kept behind the env flag as a documented negative result.
+  static const int cfg_tpb = flowml_env_int("FLOWML_HIST_TPB", 1);
   int tpb = 1;
   if (l_max > 0 && stage_bytes == 0 && hist_b > 0 && cfg_tpb > 1) {
     // keep 2 blocks/CU resident (<=75 KB/block): halving occupancy for
