@@ -209,7 +209,7 @@ class StringIndexerModel(_StringIndexerParams, Model):
             n_vals = len(labels) + (1 if handle == "keep" else 0)
             out = out.withColumn(
                 o, torch.from_numpy(idx).to(dataset.device),
-                meta={"nominal": n_vals},
+                meta={"nominal": n_vals, "labels": list(labels)},
             )
         if keep_mask is not None:
             out = out.filter(keep_mask)
@@ -816,7 +816,17 @@ class IndexToString(HasInputCol, HasOutputCol, Transformer):
         col = dataset.col(self.getInputCol())
         if isinstance(col, np.ndarray):
             raise TypeError("IndexToString input must be a numeric index column")
-        labels = self.getLabels()
+        if self.isSet(self.labels):
+            labels = self.getLabels()
+        else:
+            # Spark: with no labels param, read them from the input column's
+            # ML attributes (StringIndexerModel attaches them)
+            labels = dataset.meta(self.getInputCol()).get("labels")
+            if not labels:
+                raise ValueError(
+                    f"IndexToString: no labels set and column "
+                    f"{self.getInputCol()!r} carries no label metadata"
+                )
         idx = col.detach().cpu().numpy().astype(np.int64)
         if idx.size and (idx.min() < 0 or idx.max() >= len(labels)):
             raise ValueError(

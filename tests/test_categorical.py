@@ -420,3 +420,19 @@ def test_mixed_cat_cont_predictions_self_consistent(seed):
                 go_left = X[i, f] <= t.thr_raw[node]
             node = int(t.left[node] if go_left else t.right[node])
         assert pred[i] == float(np.argmax(t.stats[node]))
+
+
+def test_index_to_string_reads_label_metadata():
+    """Spark parity: IndexToString with no labels param reads them from the
+    input column's attributes (attached by StringIndexerModel)."""
+    from flowml.feature import IndexToString, StringIndexer
+
+    df = FlowFrame({"s": np.array(["b", "a", "b", "c"], dtype=object)})
+    m = StringIndexer(inputCol="s", outputCol="i").fit(df)
+    out = m.transform(df)
+    assert out.meta("i")["labels"] == m.labels
+    back = IndexToString(inputCol="i", outputCol="s2").transform(out)
+    assert back["s2"].tolist() == ["b", "a", "b", "c"]
+    with pytest.raises(ValueError, match="no labels"):
+        IndexToString(inputCol="i", outputCol="s2").transform(
+            FlowFrame({"i": out["i"]}))
