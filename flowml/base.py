@@ -14,7 +14,29 @@ from typing import Any, Dict, Optional
 from flowml.param import Param, Params
 
 
-class Estimator(Params, metaclass=ABCMeta):
+class _DefaultParamsWritable:
+    """Default Spark-format persistence for parameter-only stages
+    (DefaultParamsWritable/-Readable): unfitted estimators and stateless
+    transformers save metadata-only dirs and reload by class + params.
+    Fitted Models override nothing — their data writers take precedence via
+    the same GenericWriter dispatch."""
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+class Estimator(_DefaultParamsWritable, Params, metaclass=ABCMeta):
     """Abstract estimator that fits a model on a FlowFrame."""
 
     @abstractmethod
@@ -28,7 +50,7 @@ class Estimator(Params, metaclass=ABCMeta):
         return self._fit(dataset)
 
 
-class Transformer(Params, metaclass=ABCMeta):
+class Transformer(_DefaultParamsWritable, Params, metaclass=ABCMeta):
     """Abstract transformer mapping a FlowFrame to a FlowFrame."""
 
     @abstractmethod

@@ -1,0 +1,71 @@
+"""Generic estimator persistence property: for EVERY registered
+estimator/transformer class, params set to non-default values must
+round-trip through the Spark-format save/load — paramMap vs defaultParamMap
+split included. Catches serialization gaps class-by-class without a
+hand-written test per class."""
+
+import numpy as np
+import pytest
+
+from flowml.io import _CLASS_TO_SPARK, load, _our_class
+from flowml.param import Param
+
+ESTIMATORS = sorted(n for n in _CLASS_TO_SPARK if not n.endswith("Model"))
+
+# classes whose construction/round-trip needs special inputs
+SKIP = {
+    "Pipeline",           # covered by dedicated pipeline round-trip tests
+    "IndexToString",      # labels list param covered in test_categorical
+}
+
+# a non-default probe value per converter-ish type
+def _probe_value(cur):
+    if isinstance(cur, bool):
+        return not cur
+    if isinstance(cur, int):
+        return cur + 3
+    if isinstance(cur, float):
+        if cur != cur:  # NaN default (Imputer.missingValue): NaN != NaN
+            return None
+        return cur + 0.125
+    if isinstance(cur, str):
+        return cur  # enums: keep valid value
+    if isinstance(cur, (list, tuple)):
+        return list(cur)
+    return cur
+
+
+@pytest.mark.parametrize("name", [n for n in ESTIMATORS if n not in SKIP])
+def test_estimator_params_round_trip(name, tmp_path):
+    cls = _our_class(_CLASS_TO_SPARK[name])
+    inst = cls()
+    # set every defaulted param to a probe value; leave non-defaulted unset
+    probed = {}
+    for pname, p in list(inst.__dict__.items()):
+        pass
+    for p, default in list(inst._defaultParamMap.items()):
+        v = _probe_value(default)
+        if v is None:
+            continue
+        try:
+            inst.set(p, v)
+            probed[p.name] = v
+        except Exception:
+            continue  # converter rejected the probe — skip that param
+    # common string cols so transform-time getters don't explode on load
+    path = str(tmp_path / name)
+    try:
+        inst.write().save(path)
+    except AttributeError:
+        pytest.skip(f"{name} has no writer")
+    loaded = load(path)
+    assert type(loaded).__name__ == name
+    assert loaded.uid == inst.uid
+    for p, default in inst._defaultParamMap.items():
+        lp = getattr(loaded, p.name)
+        if p.name in probed:
+            assert loaded.isSet(lp), (name, p.name)
+            assert loaded.getOrDefault(lp) == probed[p.name], (name, p.name)
+        # defaults restored as defaults
+        assert loaded.getOrDefault(lp) == inst.getOrDefault(p) \
+            if p.name not in probed else True
