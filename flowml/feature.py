@@ -448,6 +448,19 @@ class _VectorIndexerParams(HasInputCol, HasOutputCol, HasHandleInvalid):
 class VectorIndexer(_VectorIndexerParams, Estimator):
     """Automatically decide which features are categorical and index them.
 
+    >>> import torch
+    >>> from flowml.frame import FlowFrame
+    >>> X = torch.tensor([[10.0, 0.3], [0.0, -1.2], [10.0, 2.4], [5.0, 0.3]])
+    >>> m = VectorIndexer(inputCol="f", outputCol="ix", maxCategories=3).fit(
+    ...     FlowFrame({"f": X}))
+    >>> m.categoryMaps
+    {0: {0.0: 0, 5.0: 1, 10.0: 2}}
+    >>> out = m.transform(FlowFrame({"f": X}))
+    >>> out["ix"][:, 0].tolist()
+    [2.0, 0.0, 2.0, 1.0]
+    >>> out.meta("ix")
+    {'categorical': {0: 3}}
+
     Spark semantics (spark:.../feature/VectorIndexer.scala): a feature with
     at most ``maxCategories`` distinct values is declared categorical; its
     values are re-indexed with **0.0 kept at index 0** (to preserve
