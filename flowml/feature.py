@@ -450,7 +450,7 @@ class VectorIndexer(_VectorIndexerParams, Estimator):
 
     >>> import torch
     >>> from flowml.frame import FlowFrame
-    >>> X = torch.tensor([[10.0, 0.3], [0.0, -1.2], [10.0, 2.4], [5.0, 0.3]])
+    >>> X = torch.tensor([[10.0, 0.3], [0.0, -1.2], [10.0, 2.4], [5.0, 0.7]])
     >>> m = VectorIndexer(inputCol="f", outputCol="ix", maxCategories=3).fit(
     ...     FlowFrame({"f": X}))
     >>> m.categoryMaps
