@@ -858,3 +858,28 @@ class TestCategoricalGPU:
         p_cpu = m_cpu.transform(df_cpu)["prediction"]
         p_gpu = m_gpu.transform(df_gpu)["prediction"].cpu()
         assert (p_cpu == p_gpu).float().mean() > 0.999
+
+
+class TestMemoryStability:
+    def test_repeated_fits_do_not_leak(self):
+        """30 RF fits + transforms: steady-state device memory must not
+        grow (allocator-level leak guard for the kernel/ensemble paths)."""
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(60)
+        X = torch.from_numpy(r.normal(size=(100_000, 41)).astype(np.float32)).to(DEV)
+        y = torch.from_numpy(r.integers(0, 5, 100_000).astype(np.float64)).to(DEV)
+        df = FlowFrame({"features": X, "label": y}, device=DEV)
+        for _ in range(3):  # warm allocator pools
+            m = RandomForestClassifier(numTrees=20, maxDepth=4, seed=0).fit(df)
+            m.transform(df)
+        torch.cuda.synchronize()
+        base = torch.cuda.memory_allocated()
+        for _ in range(30):
+            m = RandomForestClassifier(numTrees=20, maxDepth=4, seed=0).fit(df)
+            m.transform(df)
+        del m
+        torch.cuda.synchronize()
+        grown = torch.cuda.memory_allocated() - base
+        assert grown < 32 * 1024 * 1024, f"device memory grew {grown/1e6:.1f} MB"
