@@ -964,6 +964,191 @@ class MaxAbsScalerModel(_MaxAbsScalerParams, Model):
 
 
 # ---------------------------------------------------------------------------
+# RobustScaler (spark:.../feature/RobustScaler.scala)
+# ---------------------------------------------------------------------------
+
+
+class _RobustScalerParams(HasInputCol, HasOutputCol):
+    lower: Param[float] = Param(
+        "undefined", "lower", "Lower quantile to calculate quantile range",
+        TypeConverters.toFloat)
+    upper: Param[float] = Param(
+        "undefined", "upper", "Upper quantile to calculate quantile range",
+        TypeConverters.toFloat)
+    withCentering: Param[bool] = Param(
+        "undefined", "withCentering", "Whether to center data with median",
+        TypeConverters.toBoolean)
+    withScaling: Param[bool] = Param(
+        "undefined", "withScaling",
+        "Whether to scale the data to quantile range", TypeConverters.toBoolean)
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(lower=0.25, upper=0.75, withCentering=False,
+                         withScaling=True)
+
+
+class RobustScaler(_RobustScalerParams, Estimator):
+    """Scale features by subtracting the median and dividing by the
+    [lower, upper] quantile range — robust to the heavy-tailed feature
+    distributions of flow records (spark:.../feature/RobustScaler.scala:
+    defaults lower=0.25, upper=0.75, withCentering=False, withScaling=True).
+    Distributed fits use exact type-1 global quantiles over the sharded
+    column (``flowml.dist.dist_quantiles``; Spark uses approxQuantile).
+
+    >>> import torch
+    >>> from flowml.frame import FlowFrame
+    >>> X = torch.tensor([[float(v)] for v in range(1, 10)])
+    >>> m = RobustScaler(inputCol="f", outputCol="s").fit(FlowFrame({"f": X}))
+    >>> (float(m.median[0]), float(m.range[0]))
+    (5.0, 4.0)
+    """
+
+    def __init__(self, *, inputCol: Optional[str] = None,
+                 outputCol: Optional[str] = None, lower: Optional[float] = None,
+                 upper: Optional[float] = None,
+                 withCentering: Optional[bool] = None,
+                 withScaling: Optional[bool] = None):
+        super().__init__()
+        self._set(inputCol=inputCol, outputCol=outputCol, lower=lower,
+                  upper=upper, withCentering=withCentering,
+                  withScaling=withScaling)
+
+    def _fit(self, dataset: FlowFrame) -> "RobustScalerModel":
+        from flowml import dist
+
+        x = dataset.col(self.getInputCol())
+        if not isinstance(x, torch.Tensor) or x.dim() != 2:
+            raise TypeError("RobustScaler input must be a vector column")
+        lo = self.getOrDefault(self.lower)
+        up = self.getOrDefault(self.upper)
+        if not (0 <= lo < up <= 1):
+            raise ValueError("require 0 <= lower < upper <= 1")
+        xd = x.to(torch.float64)
+        F = xd.shape[1]
+        if dist.get_world_size() > 1:
+            med = torch.empty(F, dtype=torch.float64)
+            rng = torch.empty(F, dtype=torch.float64)
+            for j in range(F):
+                q = dist.dist_quantiles(xd[:, j], [lo, 0.5, up])
+                med[j] = q[1]
+                rng[j] = q[2] - q[0]
+        else:
+            qs = torch.quantile(
+                xd, torch.tensor([lo, 0.5, up], dtype=torch.float64,
+                                 device=xd.device), dim=0)
+            med = qs[1].cpu()
+            rng = (qs[2] - qs[0]).cpu()
+        model = RobustScalerModel(median=med.to(torch.float64),
+                                  range_=rng.to(torch.float64))
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class RobustScalerModel(_RobustScalerParams, Model):
+    def __init__(self, median: Optional[torch.Tensor] = None,
+                 range_: Optional[torch.Tensor] = None):
+        super().__init__()
+        self.median = median if median is not None else torch.zeros(0)
+        self.range = range_ if range_ is not None else torch.zeros(0)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getInputCol())
+        med = self.median.to(device=x.device, dtype=x.dtype)
+        rng = self.range.to(device=x.device, dtype=x.dtype)
+        out = x
+        if self.getOrDefault(self.withCentering):
+            out = out - med
+        if self.getOrDefault(self.withScaling):
+            d = torch.where(rng == 0, torch.ones_like(rng), rng)
+            out = torch.where(rng == 0, torch.zeros_like(out), out / d)
+        return dataset.withColumn(self.getOutputCol(), out)
+
+
+# ---------------------------------------------------------------------------
+# ElementwiseProduct / PolynomialExpansion (spark:.../feature/)
+# ---------------------------------------------------------------------------
+
+
+class ElementwiseProduct(HasInputCol, HasOutputCol, Transformer):
+    """Multiply each feature vector by a fixed scaling vector, element-wise
+    (spark:.../feature/ElementwiseProduct.scala)."""
+
+    scalingVec: Param[list] = Param(
+        "undefined", "scalingVec", "vector for hadamard product",
+        TypeConverters.toListFloat)
+
+    def __init__(self, *, inputCol: Optional[str] = None,
+                 outputCol: Optional[str] = None,
+                 scalingVec: Optional[List[float]] = None):
+        super().__init__()
+        self._set(inputCol=inputCol, outputCol=outputCol, scalingVec=scalingVec)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getInputCol())
+        w = torch.tensor(self.getOrDefault(self.scalingVec), dtype=x.dtype,
+                         device=x.device)
+        if x.shape[1] != w.numel():
+            raise ValueError(
+                f"scalingVec has {w.numel()} entries, features have {x.shape[1]}")
+        return dataset.withColumn(self.getOutputCol(), x * w)
+
+
+class PolynomialExpansion(HasInputCol, HasOutputCol, Transformer):
+    """Expand features into a polynomial space of the given degree
+    (spark:.../feature/PolynomialExpansion.scala). Output order matches
+    Spark's recursive expansion: for input (x, y) at degree 2 the output is
+    (x, x², y, xy, y²).
+
+    >>> import torch
+    >>> from flowml.frame import FlowFrame
+    >>> df = FlowFrame({"f": torch.tensor([[2.0, 3.0]])})
+    >>> PolynomialExpansion(inputCol="f", outputCol="p", degree=2
+    ...     ).transform(df)["p"].tolist()
+    [[2.0, 4.0, 3.0, 6.0, 9.0]]
+    """
+
+    degree: Param[int] = Param(
+        "undefined", "degree", "the polynomial degree to expand (>= 1)",
+        TypeConverters.toInt)
+
+    def __init__(self, *, inputCol: Optional[str] = None,
+                 outputCol: Optional[str] = None, degree: Optional[int] = None):
+        super().__init__()
+        self._setDefault(degree=2)
+        self._set(inputCol=inputCol, outputCol=outputCol, degree=degree)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getInputCol())
+        deg = self.getOrDefault(self.degree)
+        if deg < 1:
+            raise ValueError("degree must be >= 1")
+        N, F = x.shape
+        # Spark's order: E(k, d) = ⊕_{p=0..d} v_k^p · E(k−1, d−p) with
+        # E(0, d) = [1]; the global constant term is dropped. For (x, y) at
+        # degree 3 this yields x, x², x³, y, xy, x²y, y², xy², y³ — the
+        # documented PolynomialExpansion sequence.
+        def terms(k: int, d: int):
+            if k == 0:
+                return [()]
+            out = []
+            for p in range(d + 1):
+                for t in terms(k - 1, d - p):
+                    out.append(t + (p,))
+            return out
+
+        cols: List[torch.Tensor] = []
+        for expo in terms(F, deg)[1:]:  # drop the constant
+            c = torch.ones(N, dtype=x.dtype, device=x.device)
+            for j, e in enumerate(expo):
+                if e:
+                    c = c * x[:, j] ** e
+            cols.append(c)
+        return dataset.withColumn(self.getOutputCol(), torch.stack(cols, dim=1))
+
+
+# ---------------------------------------------------------------------------
 # Bucketizer (spark:.../feature/Bucketizer.scala)
 # ---------------------------------------------------------------------------
 

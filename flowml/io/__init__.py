@@ -77,6 +77,10 @@ _register("PCA", "org.apache.spark.ml.feature.PCA")
 _register("PCAModel", "org.apache.spark.ml.feature.PCAModel")
 _register("VectorSlicer", "org.apache.spark.ml.feature.VectorSlicer")
 _register("VectorIndexer", "org.apache.spark.ml.feature.VectorIndexer")
+_register("RobustScaler", "org.apache.spark.ml.feature.RobustScaler")
+_register("RobustScalerModel", "org.apache.spark.ml.feature.RobustScalerModel")
+_register("ElementwiseProduct", "org.apache.spark.ml.feature.ElementwiseProduct")
+_register("PolynomialExpansion", "org.apache.spark.ml.feature.PolynomialExpansion")
 _register("VectorIndexerModel", "org.apache.spark.ml.feature.VectorIndexerModel")
 _register("Interaction", "org.apache.spark.ml.feature.Interaction")
 _register("KMeans", "org.apache.spark.ml.clustering.KMeans")
@@ -464,6 +468,13 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "originalMax": vector_struct(instance.originalMax.numpy()),
             }
         ]
+    if cls == "RobustScalerModel":
+        return [
+            {
+                "range": vector_struct(instance.range.numpy()),
+                "median": vector_struct(instance.median.numpy()),
+            }
+        ]
     if cls == "MaxAbsScalerModel":
         return [{"maxAbs": vector_struct(instance.maxAbs.numpy())}]
     if cls == "VectorIndexerModel":
@@ -644,6 +655,11 @@ def _construct(cls, simple: str, rows, path: str, meta):
         return cls(
             originalMin=torch.from_numpy(vector_from_struct(rows[0]["originalMin"])),
             originalMax=torch.from_numpy(vector_from_struct(rows[0]["originalMax"])),
+        )
+    if simple == "RobustScalerModel":
+        return cls(
+            median=torch.from_numpy(vector_from_struct(rows[0]["median"])),
+            range_=torch.from_numpy(vector_from_struct(rows[0]["range"])),
         )
     if simple == "MaxAbsScalerModel":
         return cls(maxAbs=torch.from_numpy(vector_from_struct(rows[0]["maxAbs"])))

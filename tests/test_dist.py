@@ -508,3 +508,32 @@ def test_vector_indexer_dist_equals_single():
         FlowFrame({"features": torch.from_numpy(X)}))
     assert maps_dist == m.categoryMaps
     assert 0 in maps_dist and 1 not in maps_dist
+
+
+def _robust_scaler_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.feature import RobustScaler
+
+    r = np.random.default_rng(44)
+    n = 701
+    X = r.normal(size=(n, 3)).astype(np.float32)
+    lo, hi = dist.shard_range(n, rank, world)
+    df = FlowFrame({"f": torch.from_numpy(X[lo:hi])})
+    m = RobustScaler(inputCol="f", outputCol="s").fit(df)
+    gathered = dist.all_gather_object((m.median.numpy(), m.range.numpy()))
+    for g in gathered:
+        assert np.array_equal(g[0], gathered[0][0]), "ranks diverged"
+    return m.median.numpy(), m.range.numpy()
+
+
+def test_robust_scaler_dist_type1_quantiles():
+    med, rng = _run_dist("_robust_scaler_dist_body")
+    r = np.random.default_rng(44)
+    n = 701
+    X = np.sort(r.normal(size=(n, 3)).astype(np.float32).astype(np.float64), axis=0)
+    for j in range(3):
+        k50 = int(np.ceil(0.5 * n))
+        k25 = int(np.ceil(0.25 * n))
+        k75 = int(np.ceil(0.75 * n))
+        assert med[j] == X[k50 - 1, j]
+        assert rng[j] == X[k75 - 1, j] - X[k25 - 1, j]

@@ -353,3 +353,66 @@ class TestStringOrderTypes:
         assert labels("frequencyAsc") == ["a", "c", "b"]
         assert labels("alphabetAsc") == ["a", "b", "c"]
         assert labels("alphabetDesc") == ["c", "b", "a"]
+
+
+class TestRobustScaler:
+    def test_fit_transform_quantiles(self):
+        from flowml.feature import RobustScaler
+
+        x = torch.tensor([[float(v)] for v in range(1, 10)])  # 1..9
+        df = FlowFrame({"f": x})
+        m = RobustScaler(inputCol="f", outputCol="s").fit(df)
+        assert float(m.median[0]) == 5.0
+        assert float(m.range[0]) == 4.0  # q75(7) - q25(3) torch linear interp
+        out = m.transform(df)["s"]
+        assert torch.allclose(out, x / 4.0)
+        m2 = m.copy({m.withCentering: True})
+        out2 = m2.transform(df)["s"]
+        assert torch.allclose(out2, (x - 5.0) / 4.0)
+        # scaling off
+        m3 = m.copy({m.withCentering: True, m.withScaling: False})
+        assert torch.allclose(m3.transform(df)["s"], x - 5.0)
+
+    def test_zero_range(self):
+        from flowml.feature import RobustScaler
+
+        x = torch.ones(10, 2)
+        m = RobustScaler(inputCol="f", outputCol="s").fit(FlowFrame({"f": x}))
+        out = m.transform(FlowFrame({"f": x}))["s"]
+        assert torch.all(out == 0)  # constant feature scales to 0
+
+    def test_round_trip(self, tmp_path):
+        from flowml.feature import RobustScaler, RobustScalerModel
+
+        r = np.random.default_rng(3)
+        x = torch.from_numpy(r.normal(size=(200, 4)).astype(np.float32))
+        m = RobustScaler(inputCol="f", outputCol="s",
+                         withCentering=True).fit(FlowFrame({"f": x}))
+        p = str(tmp_path / "rs")
+        m.save(p)
+        m2 = RobustScalerModel.load(p)
+        assert torch.allclose(m2.median, m.median)
+        assert torch.allclose(m2.range, m.range)
+        assert m2.getOrDefault(m2.withCentering) is True
+
+
+class TestElementwiseAndPoly:
+    def test_elementwise_product(self):
+        from flowml.feature import ElementwiseProduct
+
+        df = FlowFrame({"f": torch.tensor([[1.0, 2.0, 3.0]])})
+        out = ElementwiseProduct(inputCol="f", outputCol="o",
+                                 scalingVec=[0.0, 1.0, 2.0]).transform(df)
+        assert out["o"].tolist() == [[0.0, 2.0, 6.0]]
+        with pytest.raises(ValueError):
+            ElementwiseProduct(inputCol="f", outputCol="o",
+                               scalingVec=[1.0]).transform(df)
+
+    def test_polynomial_expansion_spark_order(self):
+        from flowml.feature import PolynomialExpansion
+
+        df = FlowFrame({"f": torch.tensor([[2.0, 3.0]])})
+        out = PolynomialExpansion(inputCol="f", outputCol="p", degree=3).transform(df)
+        # the documented Spark sequence for (x, y) at degree 3
+        assert out["p"].tolist() == [[2.0, 4.0, 8.0, 3.0, 6.0, 12.0,
+                                      9.0, 18.0, 27.0]]
