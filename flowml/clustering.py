@@ -13,6 +13,8 @@ random. ClusteringEvaluator implements Spark's squared-euclidean silhouette
 
 from __future__ import annotations
 
+import math
+
 from typing import List, Optional
 
 import torch
@@ -24,6 +26,7 @@ from flowml.param import (
     HasFeaturesCol,
     HasMaxIter,
     HasPredictionCol,
+    HasProbabilityCol,
     HasSeed,
     HasTol,
     HasWeightCol,
@@ -249,4 +252,185 @@ class ClusteringEvaluator(HasFeaturesCol, HasPredictionCol, Params):
         return float(s.mean())
 
 
-__all__ = ["KMeans", "KMeansModel", "ClusteringEvaluator"]
+# ---------------------------------------------------------------------------
+# GaussianMixture (spark:.../ml/clustering/GaussianMixture.scala)
+# ---------------------------------------------------------------------------
+
+
+class _GaussianMixtureParams(HasFeaturesCol, HasPredictionCol, HasProbabilityCol,
+                             HasMaxIter, HasSeed, HasTol, HasWeightCol):
+    k: Param[int] = Param(
+        "undefined", "k", "Number of independent Gaussians in the mixture model.",
+        TypeConverters.toInt)
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(featuresCol="features", predictionCol="prediction",
+                         probabilityCol="probability", k=2, maxIter=100,
+                         tol=0.01, seed=42)
+
+
+def _gmm_log_prob(X: torch.Tensor, mus: torch.Tensor, covs: torch.Tensor,
+                  log_w: torch.Tensor) -> torch.Tensor:
+    """log [w_k N(x | mu_k, Sigma_k)] per (row, component): N×k float64."""
+    N, F = X.shape
+    k = mus.shape[0]
+    out = torch.empty(N, k, dtype=torch.float64, device=X.device)
+    for j in range(k):
+        L = torch.linalg.cholesky(covs[j])
+        d = X - mus[j]
+        z = torch.linalg.solve_triangular(L, d.T, upper=False)  # F×N
+        maha = (z * z).sum(dim=0)
+        logdet = 2.0 * torch.log(torch.diagonal(L)).sum()
+        out[:, j] = log_w[j] - 0.5 * (
+            maha + logdet + F * math.log(2.0 * math.pi))
+    return out
+
+
+class GaussianMixture(_GaussianMixtureParams, Estimator):
+    """Full-covariance Gaussian mixture fit by EM
+    (spark:.../ml/clustering/GaussianMixture.scala: defaults k=2,
+    maxIter=100, tol=0.01 on the total log-likelihood delta).
+
+    Replicated-state EM: every rank computes responsibilities over its
+    shard, the sufficient statistics (N_k, Σ r x, Σ r x xᵀ, log-likelihood)
+    are all-reduced in one flat buffer, and every rank applies the identical
+    M-step (SURVEY.md §2.5 C2 pattern). A 1e-6 ridge keeps covariances SPD
+    (documented divergence: Spark regularizes implicitly via its sample-cov
+    init only).
+    """
+
+    def __init__(self, **kwargs):
+        super().__init__()
+        self._set(**kwargs)
+
+    def setK(self, v: int):
+        return self.set(self.k, v)
+
+    def _fit(self, dataset: FlowFrame) -> "GaussianMixtureModel":
+        X = dataset.col(self.getFeaturesCol())
+        if not isinstance(X, torch.Tensor) or X.dim() != 2:
+            raise TypeError("GaussianMixture features must be a vector column")
+        Xd = X.to(torch.float64)
+        N, F = Xd.shape
+        k = self.getOrDefault(self.k)
+        if k < 2:
+            raise ValueError("k must be > 1")
+        seed = self.getOrDefault(self.seed)
+        g = torch.Generator(device=Xd.device if Xd.is_cuda else "cpu")
+        g.manual_seed(seed & 0x7FFFFFFF)
+        # global moments for the shared diagonal init covariance
+        mom = torch.cat([
+            torch.tensor([float(N)], dtype=torch.float64, device=Xd.device),
+            Xd.sum(dim=0), (Xd * Xd).sum(dim=0)])
+        dist.all_reduce_(mom)
+        n_tot = float(mom[0])
+        gmean = mom[1:1 + F] / n_tot
+        gvar = (mom[1 + F:] / n_tot - gmean * gmean).clamp_min(1e-6)
+        # init means: Spark samples groups of points per component; here a
+        # seeded rank-0 draw of 5 rows per component, broadcast
+        idx = torch.randint(0, N, (k * 5,), generator=g, device=Xd.device)
+        mus = Xd[idx].view(k, 5, F).mean(dim=1).contiguous()
+        dist.broadcast_(mus)
+        covs = torch.diag_embed(gvar).expand(k, F, F).clone()
+        w = torch.full((k,), 1.0 / k, dtype=torch.float64, device=Xd.device)
+        ll_prev = None
+        tol = self.getOrDefault(self.tol)
+        it = 0
+        ll = 0.0
+        for it in range(self.getMaxIter()):
+            lp = _gmm_log_prob(Xd, mus, covs, torch.log(w))
+            lse = torch.logsumexp(lp, dim=1)
+            r = torch.exp(lp - lse.unsqueeze(1))         # N×k
+            # sufficient stats, one flat all_reduce
+            Nk = r.sum(dim=0)                            # k
+            Sx = r.T @ Xd                                # k×F
+            Sxx = torch.stack([ (Xd * r[:, j:j + 1]).T @ Xd for j in range(k) ])
+            buf = torch.cat([Nk, Sx.reshape(-1), Sxx.reshape(-1),
+                             lse.sum().reshape(1)])
+            dist.all_reduce_(buf)
+            Nk = buf[:k]
+            Sx = buf[k:k + k * F].view(k, F)
+            Sxx = buf[k + k * F:k + k * F + k * F * F].view(k, F, F)
+            ll = float(buf[-1])
+            Nc = Nk.clamp_min(1e-12)
+            w = Nk / n_tot
+            mus = Sx / Nc.unsqueeze(1)
+            covs = Sxx / Nc.view(k, 1, 1) - torch.einsum(
+                "kf,kg->kfg", mus, mus)
+            covs = covs + 1e-6 * torch.eye(F, dtype=torch.float64,
+                                           device=Xd.device)
+            if ll_prev is not None and abs(ll - ll_prev) < tol:
+                ll_prev = ll
+                break
+            ll_prev = ll
+        model = GaussianMixtureModel(weights=w.cpu(), mus=mus.cpu(),
+                                     covs=covs.cpu(), logLikelihood=ll,
+                                     numIter=it + 1)
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class GaussianMixtureModel(_GaussianMixtureParams, Model):
+    def __init__(self, weights: Optional[torch.Tensor] = None,
+                 mus: Optional[torch.Tensor] = None,
+                 covs: Optional[torch.Tensor] = None,
+                 logLikelihood: float = 0.0, numIter: int = 0):
+        super().__init__()
+        self._weights = weights if weights is not None else torch.zeros(0, dtype=torch.float64)
+        self._mus = mus if mus is not None else torch.zeros(0, 0, dtype=torch.float64)
+        self._covs = covs if covs is not None else torch.zeros(0, 0, 0, dtype=torch.float64)
+        self.logLikelihood = float(logLikelihood)
+        self.numIter = int(numIter)
+
+    @property
+    def weights(self) -> List[float]:
+        return [float(v) for v in self._weights]
+
+    @property
+    def gaussians(self):
+        """List of (mean DenseVector, cov DenseMatrix) like Spark's
+        MultivariateGaussian pairs."""
+        from flowml.linalg import DenseMatrix, DenseVector
+
+        out = []
+        for j in range(self._mus.shape[0]):
+            mu = DenseVector(self._mus[j].numpy())
+            F = self._covs.shape[1]
+            cov = DenseMatrix(F, F, self._covs[j].T.reshape(-1).numpy())
+            out.append((mu, cov))
+        return out
+
+    def predictProbability(self, X: torch.Tensor) -> torch.Tensor:
+        Xd = X.to(torch.float64)
+        lp = _gmm_log_prob(Xd, self._mus.to(Xd.device), self._covs.to(Xd.device),
+                           torch.log(self._weights.to(Xd.device)))
+        return torch.exp(lp - torch.logsumexp(lp, dim=1, keepdim=True))
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        X = dataset.col(self.getFeaturesCol())
+        prob = self.predictProbability(X)
+        out = dataset.withColumn(
+            self.getOrDefault(self.probabilityCol), prob.to(torch.float32))
+        return out.withColumn(
+            self.getOrDefault(self.predictionCol),
+            prob.argmax(dim=1).to(torch.float64))
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+__all__ = ["KMeans", "KMeansModel", "GaussianMixture", "GaussianMixtureModel",
+           "ClusteringEvaluator"]

@@ -85,6 +85,8 @@ _register("VectorIndexerModel", "org.apache.spark.ml.feature.VectorIndexerModel"
 _register("Interaction", "org.apache.spark.ml.feature.Interaction")
 _register("KMeans", "org.apache.spark.ml.clustering.KMeans")
 _register("KMeansModel", "org.apache.spark.ml.clustering.KMeansModel")
+_register("GaussianMixture", "org.apache.spark.ml.clustering.GaussianMixture")
+_register("GaussianMixtureModel", "org.apache.spark.ml.clustering.GaussianMixtureModel")
 _register("CrossValidatorModel", "org.apache.spark.ml.tuning.CrossValidatorModel")
 _register("TrainValidationSplitModel", "org.apache.spark.ml.tuning.TrainValidationSplitModel")
 _register("OneVsRestModel", "org.apache.spark.ml.classification.OneVsRestModel")
@@ -515,6 +517,18 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "modelType": instance._mtype,
             }
         ]
+    if cls == "GaussianMixtureModel":
+        # Spark schema: single row {weights: array<double>,
+        # mus: array<VectorUDT>, sigmas: array<MatrixUDT>}
+        return [
+            {
+                "weights": [float(v) for v in instance._weights],
+                "mus": [vector_struct(instance._mus[j].numpy())
+                        for j in range(instance._mus.shape[0])],
+                "sigmas": [matrix_struct(instance._covs[j].numpy())
+                           for j in range(instance._covs.shape[0])],
+            }
+        ]
     if cls == "KMeansModel":
         return [
             {
@@ -682,6 +696,14 @@ def _construct(cls, simple: str, rows, path: str, meta):
                    sigma=sg if sg.numel() else torch.zeros(0),
                    numClasses=th.shape[0], numFeatures=th.shape[1],
                    modelType=r0["modelType"])
+    if simple == "GaussianMixtureModel":
+        r0 = rows[0]
+        mus = torch.stack([torch.from_numpy(vector_from_struct(v))
+                           for v in r0["mus"]])
+        covs = torch.stack([torch.from_numpy(matrix_from_struct(m))
+                            for m in r0["sigmas"]])
+        return cls(weights=torch.tensor(r0["weights"], dtype=torch.float64),
+                   mus=mus, covs=covs)
     if simple == "KMeansModel":
         r0 = rows[0]
         return cls(centers=torch.from_numpy(matrix_from_struct(r0["clusterCenters"])),

@@ -79,3 +79,51 @@ class TestClusteringEvaluator:
         worse = ClusteringEvaluator().evaluate(FlowFrame(
             {"features": torch.from_numpy(X), "prediction": torch.from_numpy(bad)}))
         assert good > worse
+
+
+class TestGaussianMixture:
+    def _blobs(self, seed=0):
+        r = np.random.default_rng(seed)
+        A = r.normal(loc=[-4.0, 0.0], scale=0.5, size=(400, 2))
+        B = r.normal(loc=[4.0, 3.0], scale=1.0, size=(600, 2))
+        return np.vstack([A, B]).astype(np.float32)
+
+    def test_separates_and_matches_sklearn(self):
+        from sklearn.mixture import GaussianMixture as SkGMM
+
+        from flowml.clustering import GaussianMixture
+
+        X = self._blobs()
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = GaussianMixture(k=2, seed=3).fit(df)
+        sk = SkGMM(n_components=2, random_state=0).fit(X.astype(np.float64))
+        ours = sorted(
+            (w, tuple(np.round(mu.toArray(), 1)))
+            for w, (mu, _) in zip(m.weights, m.gaussians)
+        )
+        theirs = sorted(
+            (w, tuple(np.round(mu, 1)))
+            for w, mu in zip(sk.weights_, sk.means_)
+        )
+        for (w1, mu1), (w2, mu2) in zip(ours, theirs):
+            assert abs(w1 - w2) < 0.02
+            assert np.allclose(mu1, mu2, atol=0.3)
+        prob = m.transform(df)["probability"]
+        assert prob.shape == (1000, 2)
+        assert torch.allclose(prob.sum(dim=1),
+                              torch.ones(1000), atol=1e-5)
+
+    def test_round_trip(self, tmp_path):
+        from flowml.clustering import GaussianMixture, GaussianMixtureModel
+
+        X = self._blobs(1)
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = GaussianMixture(k=2, seed=5).fit(df)
+        p = str(tmp_path / "gmm")
+        m.save(p)
+        m2 = GaussianMixtureModel.load(p)
+        assert np.allclose(m2.weights, m.weights)
+        assert torch.allclose(m2._mus, m._mus)
+        assert torch.allclose(m2._covs, m._covs)
+        assert torch.equal(m.transform(df)["prediction"],
+                           m2.transform(df)["prediction"])

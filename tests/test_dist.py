@@ -537,3 +537,29 @@ def test_robust_scaler_dist_type1_quantiles():
         k75 = int(np.ceil(0.75 * n))
         assert med[j] == X[k50 - 1, j]
         assert rng[j] == X[k75 - 1, j] - X[k25 - 1, j]
+
+
+def _gmm_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.clustering import GaussianMixture
+
+    r = np.random.default_rng(55)
+    A = r.normal(loc=[-4.0, 0.0], scale=0.5, size=(400, 2))
+    B = r.normal(loc=[4.0, 3.0], scale=1.0, size=(600, 2))
+    X = np.vstack([A, B]).astype(np.float32)
+    perm = r.permutation(len(X))
+    X = X[perm]
+    lo, hi = dist.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi])})
+    m = GaussianMixture(k=2, seed=7).fit(df)
+    payload = np.concatenate([np.asarray(m.weights),
+                              m._mus.numpy().ravel()])
+    gathered = dist.all_gather_object(payload)
+    assert all(np.allclose(g, gathered[0]) for g in gathered), "ranks diverged"
+    return payload
+
+
+def test_gmm_dist_ranks_agree_and_sane():
+    payload = _run_dist("_gmm_dist_body")
+    w = np.sort(payload[:2])
+    assert abs(w[0] - 0.4) < 0.05 and abs(w.sum() - 1.0) < 1e-9
