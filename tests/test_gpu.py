@@ -883,3 +883,26 @@ class TestMemoryStability:
         torch.cuda.synchronize()
         grown = torch.cuda.memory_allocated() - base
         assert grown < 32 * 1024 * 1024, f"device memory grew {grown/1e6:.1f} MB"
+
+
+class TestClusteringGPU:
+    def test_gmm_fit_gpu_matches_cpu(self):
+        """GMM EM on device (f64 cholesky/solve on ROCm) == CPU fit."""
+        from flowml.clustering import GaussianMixture
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(61)
+        A = r.normal(loc=[-4.0, 0.0, 1.0], scale=0.5, size=(2000, 3))
+        B = r.normal(loc=[4.0, 3.0, -1.0], scale=1.0, size=(3000, 3))
+        X = np.vstack([A, B]).astype(np.float32)
+        df_cpu = FlowFrame({"features": torch.from_numpy(X)})
+        df_gpu = df_cpu.to(DEV)
+        m_cpu = GaussianMixture(k=2, seed=3).fit(df_cpu)
+        m_gpu = GaussianMixture(k=2, seed=3).fit(df_gpu)
+        # same seeded init path modulo device RNG -> compare converged params
+        a = np.sort(np.asarray(m_cpu.weights))
+        b = np.sort(np.asarray(m_gpu.weights))
+        assert np.allclose(a, b, atol=1e-3)
+        mu_c = np.sort(m_cpu._mus.numpy(), axis=0)
+        mu_g = np.sort(m_gpu._mus.cpu().numpy(), axis=0)
+        assert np.allclose(mu_c, mu_g, atol=0.05)
