@@ -36,6 +36,7 @@ from flowml.param import (
     HasHandleInvalid,
     HasInputCol,
     HasInputCols,
+    HasLabelCol,
     HasOutputCol,
     HasOutputCols,
     Param,
@@ -1368,6 +1369,274 @@ class QuantileDiscretizer(HasInputCol, HasOutputCol, HasHandleInvalid, Estimator
         return Bucketizer(splits=splits, inputCol=self.getInputCol(),
                           outputCol=self.getOutputCol(),
                           handleInvalid=self.getHandleInvalid())
+
+
+# ---------------------------------------------------------------------------
+# UnivariateFeatureSelector / ChiSqSelector
+# (spark:.../feature/UnivariateFeatureSelector.scala, ChiSqSelector.scala)
+# ---------------------------------------------------------------------------
+
+
+def _bh_fdr_select(pvals: np.ndarray, alpha: float) -> np.ndarray:
+    """Benjamini-Hochberg: indices passing FDR control at level alpha."""
+    F = len(pvals)
+    order = np.argsort(pvals, kind="stable")
+    ranked = pvals[order]
+    thresh = alpha * (np.arange(1, F + 1) / F)
+    ok = ranked <= thresh
+    if not ok.any():
+        return np.zeros(0, dtype=np.int64)
+    kmax = int(np.nonzero(ok)[0].max())
+    return np.sort(order[: kmax + 1])
+
+
+def _selector_stats(X: torch.Tensor, y: torch.Tensor, feature_type: str,
+                    label_type: str):
+    """Per-feature (score, p-value) with all-reduced sufficient statistics.
+
+    chi2 for categorical/categorical, one-way ANOVA F (f_classif) for
+    continuous/categorical, univariate regression F (f_regression) for
+    continuous/continuous — the three combinations Spark supports.
+    """
+    from scipy import stats as sstats
+
+    from flowml import dist
+
+    Xd = X.to(torch.float64)
+    yd = y.reshape(-1).to(torch.float64)
+    N, F = Xd.shape
+    if feature_type == "categorical" and label_type == "categorical":
+        xa = Xd.cpu().numpy().astype(np.int64)
+        ya = yd.cpu().numpy().astype(np.int64)
+        if xa.min(initial=0) < 0 or ya.min(initial=0) < 0:
+            raise ValueError("categorical values must be non-negative indices")
+        vmax = torch.tensor([float(xa.max(initial=0)),
+                             float(ya.max(initial=0))])
+        dist.all_reduce_(vmax, op="max")
+        V, C = int(vmax[0]) + 1, int(vmax[1]) + 1
+        # contingency counts per feature: F×V×C, flat all_reduce
+        cont = np.zeros((F, V, C), np.float64)
+        for j in range(F):
+            np.add.at(cont[j], (xa[:, j], ya), 1.0)
+        cont_t = torch.from_numpy(cont.reshape(-1))
+        dist.all_reduce_(cont_t)
+        cont = cont_t.numpy().reshape(F, V, C)
+        scores = np.zeros(F)
+        pvals = np.ones(F)
+        for j in range(F):
+            tbl = cont[j]
+            rows = tbl.sum(axis=1)
+            cols = tbl.sum(axis=0)
+            n = tbl.sum()
+            keep_r = rows > 0
+            keep_c = cols > 0
+            t = tbl[np.ix_(keep_r, keep_c)]
+            exp = np.outer(rows[keep_r], cols[keep_c]) / max(n, 1.0)
+            dof = (t.shape[0] - 1) * (t.shape[1] - 1)
+            if dof <= 0:
+                continue
+            scores[j] = float(((t - exp) ** 2 / exp).sum())
+            pvals[j] = float(sstats.chi2.sf(scores[j], dof))
+        return scores, pvals
+    if feature_type == "continuous" and label_type == "categorical":
+        ya = yd.cpu().numpy().astype(np.int64)
+        kmax = torch.tensor([float(ya.max(initial=0))])
+        dist.all_reduce_(kmax, op="max")
+        K = int(kmax[0]) + 1
+        oh = torch.zeros(N, K, dtype=torch.float64, device=Xd.device)
+        oh[torch.arange(N, device=Xd.device), yd.to(torch.int64)] = 1.0
+        buf = torch.cat([
+            oh.sum(dim=0),                     # K class counts
+            (oh.T @ Xd).reshape(-1),           # K×F class sums
+            (oh.T @ (Xd * Xd)).reshape(-1),    # K×F class sum-squares
+        ]).cpu()
+        dist.all_reduce_(buf)
+        nk = buf[:K].numpy()
+        sk = buf[K:K + K * F].view(K, F).numpy()
+        qk = buf[K + K * F:].view(K, F).numpy()
+        n = nk.sum()
+        valid = nk > 0
+        k_eff = int(valid.sum())
+        mean_k = sk[valid] / nk[valid, None]
+        gmean = sk.sum(axis=0) / n
+        ss_between = (nk[valid, None] * (mean_k - gmean) ** 2).sum(axis=0)
+        ss_within = (qk[valid] - nk[valid, None] * mean_k ** 2).sum(axis=0)
+        df_b = k_eff - 1
+        df_w = n - k_eff
+        with np.errstate(divide="ignore", invalid="ignore"):
+            scores = (ss_between / df_b) / np.maximum(ss_within / df_w, 1e-300)
+        pvals = sstats.f.sf(scores, df_b, df_w)
+        return scores, pvals
+    if feature_type == "continuous" and label_type == "continuous":
+        buf = torch.cat([
+            torch.tensor([float(N)], dtype=torch.float64, device=Xd.device),
+            Xd.sum(dim=0), (Xd * Xd).sum(dim=0),
+            (Xd * yd.unsqueeze(1)).sum(dim=0),
+            yd.sum().reshape(1), (yd * yd).sum().reshape(1),
+        ]).cpu()
+        dist.all_reduce_(buf)
+        n = float(buf[0])
+        sx = buf[1:1 + F].numpy()
+        qx = buf[1 + F:1 + 2 * F].numpy()
+        sxy = buf[1 + 2 * F:1 + 3 * F].numpy()
+        sy = float(buf[-2])
+        qy = float(buf[-1])
+        cov = sxy / n - (sx / n) * (sy / n)
+        vx = np.maximum(qx / n - (sx / n) ** 2, 1e-300)
+        vy = max(qy / n - (sy / n) ** 2, 1e-300)
+        r2 = np.clip(cov ** 2 / (vx * vy), 0.0, 1.0 - 1e-15)
+        dfn = n - 2
+        scores = r2 / (1 - r2) * dfn
+        pvals = sstats.f.sf(scores, 1, dfn)
+        return scores, pvals
+    raise ValueError(
+        f"unsupported featureType={feature_type!r} / labelType={label_type!r} "
+        "(Spark supports categorical/categorical, continuous/categorical, "
+        "continuous/continuous)")
+
+
+class _UnivariateFeatureSelectorParams(HasInputCol, HasOutputCol, HasLabelCol):
+    featureType: Param[str] = Param(
+        "undefined", "featureType", "The feature type: categorical|continuous",
+        TypeConverters.toString)
+    labelType: Param[str] = Param(
+        "undefined", "labelType", "The label type: categorical|continuous",
+        TypeConverters.toString)
+    selectionMode: Param[str] = Param(
+        "undefined", "selectionMode",
+        "The selection mode: numTopFeatures|percentile|fpr|fdr|fwe",
+        TypeConverters.toString)
+    selectionThreshold: Param[float] = Param(
+        "undefined", "selectionThreshold",
+        "The upper bound of the features that selector will select",
+        TypeConverters.toFloat)
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(labelCol="label", selectionMode="numTopFeatures")
+
+
+class UnivariateFeatureSelector(_UnivariateFeatureSelectorParams, Estimator):
+    """Select features via univariate statistical tests against the label
+    (spark:.../feature/UnivariateFeatureSelector.scala): chi2 for
+    categorical features + categorical label, ANOVA F-test for continuous +
+    categorical, univariate-regression F-test for continuous + continuous.
+    Sufficient statistics (contingency counts / per-class moments /
+    correlation moments) are all-reduced, so sharded fits agree.
+    selectionMode thresholds default per Spark: numTopFeatures 50,
+    percentile 0.1, fpr/fdr/fwe 0.05. Selection orders by ascending
+    p-value (ties: lower feature index first)."""
+
+    def __init__(self, *, featuresCol: Optional[str] = None,
+                 outputCol: Optional[str] = None,
+                 labelCol: Optional[str] = None,
+                 featureType: Optional[str] = None,
+                 labelType: Optional[str] = None,
+                 selectionMode: Optional[str] = None,
+                 selectionThreshold: Optional[float] = None):
+        super().__init__()
+        self._set(inputCol=featuresCol, outputCol=outputCol, labelCol=labelCol,
+                  featureType=featureType, labelType=labelType,
+                  selectionMode=selectionMode,
+                  selectionThreshold=selectionThreshold)
+
+    def _fit(self, dataset: FlowFrame) -> "UnivariateFeatureSelectorModel":
+        X = dataset.col(self.getOrDefault(self.inputCol))
+        y = dataset.col(self.getOrDefault(self.labelCol))
+        if not isinstance(X, torch.Tensor) or X.dim() != 2:
+            raise TypeError("selector features must be a vector column")
+        ftype = self.getOrDefault(self.featureType)
+        ltype = self.getOrDefault(self.labelType)
+        mode = self.getOrDefault(self.selectionMode)
+        if self.isSet(self.selectionThreshold):
+            thr = self.getOrDefault(self.selectionThreshold)
+        else:
+            thr = {"numTopFeatures": 50.0, "percentile": 0.1,
+                   "fpr": 0.05, "fdr": 0.05, "fwe": 0.05}[mode]
+        scores, pvals = _selector_stats(X, torch.as_tensor(
+            y if isinstance(y, torch.Tensor) else np.asarray(y, np.float64)),
+            ftype, ltype)
+        F = len(pvals)
+        order = np.argsort(pvals, kind="stable")
+        if mode == "numTopFeatures":
+            sel = np.sort(order[: int(thr)])
+        elif mode == "percentile":
+            sel = np.sort(order[: int(thr * F)])
+        elif mode == "fpr":
+            sel = np.nonzero(pvals < thr)[0]
+        elif mode == "fwe":
+            sel = np.nonzero(pvals < thr / F)[0]
+        elif mode == "fdr":
+            sel = _bh_fdr_select(pvals, thr)
+        else:
+            raise ValueError(f"Unsupported selectionMode {mode!r}")
+        model = UnivariateFeatureSelectorModel(
+            selectedFeatures=[int(i) for i in sel])
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class UnivariateFeatureSelectorModel(_UnivariateFeatureSelectorParams, Model):
+    def __init__(self, selectedFeatures: Optional[List[int]] = None):
+        super().__init__()
+        self.selectedFeatures: List[int] = list(selectedFeatures or [])
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getOrDefault(self.inputCol))
+        idx = torch.tensor(self.selectedFeatures, dtype=torch.int64,
+                           device=x.device)
+        return dataset.withColumn(self.getOrDefault(self.outputCol),
+                                  x.index_select(1, idx))
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+class ChiSqSelector(HasInputCol, HasOutputCol, HasLabelCol, Estimator):
+    """Legacy chi-squared feature selector (spark:.../feature/
+    ChiSqSelector.scala, deprecated upstream in favor of
+    UnivariateFeatureSelector): categorical features vs categorical label,
+    numTopFeatures default 50."""
+
+    numTopFeatures: Param[int] = Param(
+        "undefined", "numTopFeatures",
+        "Number of features that selector will select", TypeConverters.toInt)
+
+    def __init__(self, *, featuresCol: Optional[str] = None,
+                 outputCol: Optional[str] = None,
+                 labelCol: Optional[str] = None,
+                 numTopFeatures: Optional[int] = None):
+        super().__init__()
+        self._setDefault(labelCol="label", numTopFeatures=50)
+        self._set(inputCol=featuresCol, outputCol=outputCol, labelCol=labelCol,
+                  numTopFeatures=numTopFeatures)
+
+    def _fit(self, dataset: FlowFrame) -> UnivariateFeatureSelectorModel:
+        sel = UnivariateFeatureSelector(
+            featuresCol=self.getOrDefault(self.inputCol),
+            outputCol=self.getOrDefault(self.outputCol),
+            labelCol=self.getOrDefault(self.labelCol),
+            featureType="categorical", labelType="categorical",
+            selectionMode="numTopFeatures",
+            selectionThreshold=float(self.getOrDefault(self.numTopFeatures)),
+        )
+        model = sel._fit(dataset)
+        model._set(inputCol=self.getOrDefault(self.inputCol),
+                   outputCol=self.getOrDefault(self.outputCol))
+        model._set_parent(self)
+        return model
 
 
 # ---------------------------------------------------------------------------

@@ -78,6 +78,9 @@ _register("PCAModel", "org.apache.spark.ml.feature.PCAModel")
 _register("VectorSlicer", "org.apache.spark.ml.feature.VectorSlicer")
 _register("VectorIndexer", "org.apache.spark.ml.feature.VectorIndexer")
 _register("RobustScaler", "org.apache.spark.ml.feature.RobustScaler")
+_register("UnivariateFeatureSelector", "org.apache.spark.ml.feature.UnivariateFeatureSelector")
+_register("UnivariateFeatureSelectorModel", "org.apache.spark.ml.feature.UnivariateFeatureSelectorModel")
+_register("ChiSqSelector", "org.apache.spark.ml.feature.ChiSqSelector")
 _register("RobustScalerModel", "org.apache.spark.ml.feature.RobustScalerModel")
 _register("ElementwiseProduct", "org.apache.spark.ml.feature.ElementwiseProduct")
 _register("PolynomialExpansion", "org.apache.spark.ml.feature.PolynomialExpansion")
@@ -470,6 +473,8 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "originalMax": vector_struct(instance.originalMax.numpy()),
             }
         ]
+    if cls == "UnivariateFeatureSelectorModel":
+        return [{"selectedFeatures": [int(i) for i in instance.selectedFeatures]}]
     if cls == "RobustScalerModel":
         return [
             {
@@ -670,6 +675,8 @@ def _construct(cls, simple: str, rows, path: str, meta):
             originalMin=torch.from_numpy(vector_from_struct(rows[0]["originalMin"])),
             originalMax=torch.from_numpy(vector_from_struct(rows[0]["originalMax"])),
         )
+    if simple == "UnivariateFeatureSelectorModel":
+        return cls(selectedFeatures=[int(i) for i in rows[0]["selectedFeatures"]])
     if simple == "RobustScalerModel":
         return cls(
             median=torch.from_numpy(vector_from_struct(rows[0]["median"])),

@@ -563,3 +563,28 @@ def test_gmm_dist_ranks_agree_and_sane():
     payload = _run_dist("_gmm_dist_body")
     w = np.sort(payload[:2])
     assert abs(w[0] - 0.4) < 0.05 and abs(w.sum() - 1.0) < 1e-9
+
+
+def _selector_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.feature import UnivariateFeatureSelector
+
+    r = np.random.default_rng(66)
+    n = 800
+    X = r.normal(size=(n, 5)).astype(np.float32)
+    y = r.integers(0, 3, n).astype(np.float64)
+    X[:, 3] += 2 * y
+    lo, hi = dist.shard_range(n, rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi]), "label": y[lo:hi]})
+    m = UnivariateFeatureSelector(
+        featuresCol="features", outputCol="s", labelCol="label",
+        featureType="continuous", labelType="categorical",
+        selectionThreshold=1.0).fit(df)
+    gathered = dist.all_gather_object(m.selectedFeatures)
+    assert all(g == gathered[0] for g in gathered)
+    return m.selectedFeatures
+
+
+def test_selector_dist_equals_single():
+    sel = _run_dist("_selector_dist_body")
+    assert sel == [3]

@@ -416,3 +416,85 @@ class TestElementwiseAndPoly:
         # the documented Spark sequence for (x, y) at degree 3
         assert out["p"].tolist() == [[2.0, 4.0, 8.0, 3.0, 6.0, 12.0,
                                       9.0, 18.0, 27.0]]
+
+
+class TestFeatureSelectors:
+    def _data(self):
+        r = np.random.default_rng(9)
+        X = r.normal(size=(500, 6)).astype(np.float32)
+        y = r.integers(0, 3, 500).astype(np.float64)
+        X[:, 2] += y
+        X[:, 4] += 2 * y
+        return X, y
+
+    def test_anova_top_features(self):
+        from flowml.feature import UnivariateFeatureSelector
+
+        X, y = self._data()
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y})
+        m = UnivariateFeatureSelector(
+            featuresCol="features", outputCol="sel", labelCol="label",
+            featureType="continuous", labelType="categorical",
+            selectionThreshold=2.0).fit(df)
+        assert m.selectedFeatures == [2, 4]
+        out = m.transform(df)["sel"]
+        assert out.shape == (500, 2)
+        assert torch.allclose(out[:, 0], torch.from_numpy(X[:, 2]))
+
+    def test_modes(self):
+        from flowml.feature import UnivariateFeatureSelector
+
+        X, y = self._data()
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y})
+        base = dict(featuresCol="features", outputCol="sel", labelCol="label",
+                    featureType="continuous", labelType="categorical")
+        fpr = UnivariateFeatureSelector(selectionMode="fpr",
+                                        selectionThreshold=1e-6, **base).fit(df)
+        assert set(fpr.selectedFeatures) == {2, 4}
+        fwe = UnivariateFeatureSelector(selectionMode="fwe",
+                                        selectionThreshold=1e-5, **base).fit(df)
+        assert set(fwe.selectedFeatures) == {2, 4}
+        fdr = UnivariateFeatureSelector(selectionMode="fdr",
+                                        selectionThreshold=1e-6, **base).fit(df)
+        assert set(fdr.selectedFeatures) == {2, 4}
+        pct = UnivariateFeatureSelector(selectionMode="percentile",
+                                        selectionThreshold=0.34, **base).fit(df)
+        assert len(pct.selectedFeatures) == 2
+
+    def test_chisq_selector(self):
+        from flowml.feature import ChiSqSelector
+
+        r = np.random.default_rng(10)
+        y = r.integers(0, 3, 600).astype(np.float64)
+        Xc = r.integers(0, 4, size=(600, 4)).astype(np.float32)
+        Xc[:, 1] = (y + r.integers(0, 2, 600)) % 4  # informative
+        df = FlowFrame({"features": torch.from_numpy(Xc), "label": y})
+        m = ChiSqSelector(featuresCol="features", outputCol="sel",
+                          numTopFeatures=1).fit(df)
+        assert m.selectedFeatures == [1]
+
+    def test_round_trip(self, tmp_path):
+        from flowml.feature import (UnivariateFeatureSelector,
+                                    UnivariateFeatureSelectorModel)
+
+        X, y = self._data()
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y})
+        m = UnivariateFeatureSelector(
+            featuresCol="features", outputCol="sel", labelCol="label",
+            featureType="continuous", labelType="categorical",
+            selectionThreshold=2.0).fit(df)
+        p = str(tmp_path / "sel")
+        m.save(p)
+        m2 = UnivariateFeatureSelectorModel.load(p)
+        assert m2.selectedFeatures == [2, 4]
+        assert torch.equal(m.transform(df)["sel"], m2.transform(df)["sel"])
+
+    def test_unsupported_combination(self):
+        from flowml.feature import UnivariateFeatureSelector
+
+        X, y = self._data()
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y})
+        with pytest.raises(ValueError, match="unsupported"):
+            UnivariateFeatureSelector(
+                featuresCol="features", outputCol="s", labelCol="label",
+                featureType="categorical", labelType="continuous").fit(df)
