@@ -253,6 +253,177 @@ class ClusteringEvaluator(HasFeaturesCol, HasPredictionCol, Params):
 
 
 # ---------------------------------------------------------------------------
+# BisectingKMeans (spark:.../ml/clustering/BisectingKMeans.scala)
+# ---------------------------------------------------------------------------
+
+
+class _BisectingKMeansParams(HasFeaturesCol, HasPredictionCol, HasMaxIter,
+                             HasSeed, HasWeightCol):
+    k: Param[int] = Param(
+        "undefined", "k", "The desired number of leaf clusters (> 1).",
+        TypeConverters.toInt)
+    minDivisibleClusterSize: Param[float] = Param(
+        "undefined", "minDivisibleClusterSize",
+        "The minimum number of points (if >= 1.0) or the minimum proportion "
+        "of points (if < 1.0) of a divisible cluster.",
+        TypeConverters.toFloat)
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(featuresCol="features", predictionCol="prediction",
+                         k=4, maxIter=20, minDivisibleClusterSize=1.0, seed=42)
+
+
+class BisectingKMeans(_BisectingKMeansParams, Estimator):
+    """Divisive hierarchical clustering: repeatedly bisect the largest
+    divisible cluster with an inner 2-means until ``k`` leaf clusters exist
+    (spark:.../ml/clustering/BisectingKMeans.scala: defaults k=4,
+    maxIter=20, minDivisibleClusterSize=1.0). Replicated across ranks: the
+    inner Lloyd statistics are all-reduced and the seeded split init is
+    rank-0-broadcast, so every rank holds the identical tree of centers."""
+
+    def __init__(self, **kwargs):
+        super().__init__()
+        self._set(**kwargs)
+
+    def setK(self, v: int):
+        return self.set(self.k, v)
+
+    def _bisect(self, X: torch.Tensor, seed: int):
+        """2-means on the rows of X (may be empty on some ranks)."""
+        N = X.shape[0]
+        F = X.shape[1]
+        g = torch.Generator(device=X.device if X.is_cuda else "cpu")
+        g.manual_seed(seed & 0x7FFFFFFF)
+        if N >= 2:
+            idx = torch.randperm(N, generator=g, device=X.device)[:2]
+            centers = X[idx].clone()
+        else:
+            centers = torch.zeros(2, F, dtype=torch.float64, device=X.device)
+        dist.broadcast_(centers)
+        for _ in range(self.getMaxIter()):
+            if N:
+                d2 = _dist2(X, centers)
+                assign = d2.argmin(dim=1)
+                onehot = torch.zeros(N, 2, dtype=torch.float64, device=X.device)
+                onehot[torch.arange(N, device=X.device), assign] = 1.0
+                buf = torch.cat([onehot.sum(dim=0), (onehot.T @ X).reshape(-1)])
+            else:
+                buf = torch.zeros(2 + 2 * F, dtype=torch.float64, device=X.device)
+            dist.all_reduce_(buf)
+            cnt = buf[:2]
+            sums = buf[2:].view(2, F)
+            new_centers = torch.where(
+                (cnt > 0).unsqueeze(1), sums / cnt.clamp_min(1.0).unsqueeze(1),
+                centers)
+            if torch.allclose(new_centers, centers, atol=1e-12):
+                centers = new_centers
+                break
+            centers = new_centers
+        return centers
+
+    def _fit(self, dataset: FlowFrame) -> "BisectingKMeansModel":
+        X = dataset.col(self.getFeaturesCol())
+        if not isinstance(X, torch.Tensor) or X.dim() != 2:
+            raise TypeError("BisectingKMeans features must be a vector column")
+        Xd = X.to(torch.float64)
+        N, F = Xd.shape
+        k = self.getOrDefault(self.k)
+        if k < 2:
+            raise ValueError("k must be > 1")
+        seed = self.getOrDefault(self.seed)
+        mdcs = self.getOrDefault(self.minDivisibleClusterSize)
+        n_tot = torch.tensor([float(N)], dtype=torch.float64, device=Xd.device)
+        dist.all_reduce_(n_tot)
+        min_size = mdcs if mdcs >= 1.0 else mdcs * float(n_tot)
+        assign = torch.zeros(N, dtype=torch.int64, device=Xd.device)
+        centers = [Xd.sum(dim=0)]
+        c0 = torch.cat([centers[0], torch.tensor([float(N)], dtype=torch.float64,
+                                                 device=Xd.device)])
+        dist.all_reduce_(c0)
+        centers = [c0[:F] / c0[F].clamp_min(1.0)]
+        sizes = [float(c0[F])]
+        divisible = [True]
+        while len(centers) < k and any(
+            d and s >= max(min_size, 2.0) for d, s in zip(divisible, sizes)
+        ):
+            # split the largest divisible cluster (global sizes — identical
+            # on every rank, so the choice is replicated)
+            cand = [i for i, (d, sz) in enumerate(zip(divisible, sizes))
+                    if d and sz >= max(min_size, 2.0)]
+            target = max(cand, key=lambda i: sizes[i])
+            mask = assign == target
+            sub = Xd[mask]
+            cpair = self._bisect(sub, seed + 7919 * len(centers))
+            if N:
+                d2 = _dist2(sub, cpair)
+                side = d2.argmin(dim=1)
+            # new sizes, all-reduced
+            sz = torch.zeros(2, dtype=torch.float64, device=Xd.device)
+            if N and len(sub):
+                sz[0] = float((side == 0).sum())
+                sz[1] = float((side == 1).sum())
+            dist.all_reduce_(sz)
+            if float(sz.min()) == 0.0:
+                divisible[target] = False  # degenerate split: mark leaf
+                continue
+            new_id = len(centers)
+            if N and len(sub):
+                sel = torch.nonzero(mask, as_tuple=False).flatten()
+                assign[sel[side == 1]] = new_id
+            centers[target] = cpair[0]
+            centers.append(cpair[1])
+            sizes[target] = float(sz[0])
+            sizes.append(float(sz[1]))
+            divisible[target] = True
+            divisible.append(True)
+        C = torch.stack(centers)
+        # training cost: sum of squared distances to the final centers
+        if N:
+            cost_t = _dist2(Xd, C).min(dim=1).values.sum().reshape(1)
+        else:
+            cost_t = torch.zeros(1, dtype=torch.float64, device=Xd.device)
+        dist.all_reduce_(cost_t)
+        model = BisectingKMeansModel(centers=C.cpu(),
+                                     trainingCost=float(cost_t))
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class BisectingKMeansModel(_BisectingKMeansParams, Model):
+    def __init__(self, centers: Optional[torch.Tensor] = None,
+                 trainingCost: float = 0.0):
+        super().__init__()
+        self._centers = centers if centers is not None else torch.zeros(
+            0, 0, dtype=torch.float64)
+        self.trainingCost = float(trainingCost)
+
+    def clusterCenters(self):
+        return [c.numpy() for c in self._centers]
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        X = dataset.col(self.getFeaturesCol()).to(torch.float64)
+        C = self._centers.to(X.device)
+        pred = _dist2(X, C).argmin(dim=1).to(torch.float64)
+        return dataset.withColumn(self.getOrDefault(self.predictionCol), pred)
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+# ---------------------------------------------------------------------------
 # GaussianMixture (spark:.../ml/clustering/GaussianMixture.scala)
 # ---------------------------------------------------------------------------
 
@@ -432,5 +603,5 @@ class GaussianMixtureModel(_GaussianMixtureParams, Model):
         return _load(path)
 
 
-__all__ = ["KMeans", "KMeansModel", "GaussianMixture", "GaussianMixtureModel",
-           "ClusteringEvaluator"]
+__all__ = ["KMeans", "KMeansModel", "BisectingKMeans", "BisectingKMeansModel",
+           "GaussianMixture", "GaussianMixtureModel", "ClusteringEvaluator"]

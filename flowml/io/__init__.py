@@ -88,6 +88,8 @@ _register("VectorIndexerModel", "org.apache.spark.ml.feature.VectorIndexerModel"
 _register("Interaction", "org.apache.spark.ml.feature.Interaction")
 _register("KMeans", "org.apache.spark.ml.clustering.KMeans")
 _register("KMeansModel", "org.apache.spark.ml.clustering.KMeansModel")
+_register("BisectingKMeans", "org.apache.spark.ml.clustering.BisectingKMeans")
+_register("BisectingKMeansModel", "org.apache.spark.ml.clustering.BisectingKMeansModel")
 _register("GaussianMixture", "org.apache.spark.ml.clustering.GaussianMixture")
 _register("GaussianMixtureModel", "org.apache.spark.ml.clustering.GaussianMixtureModel")
 _register("CrossValidatorModel", "org.apache.spark.ml.tuning.CrossValidatorModel")
@@ -522,6 +524,9 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "modelType": instance._mtype,
             }
         ]
+    if cls == "BisectingKMeansModel":
+        return [{"clusterCenters": matrix_struct(instance._centers.numpy()),
+                 "trainingCost": float(instance.trainingCost)}]
     if cls == "GaussianMixtureModel":
         # Spark schema: single row {weights: array<double>,
         # mus: array<VectorUDT>, sigmas: array<MatrixUDT>}
@@ -703,6 +708,10 @@ def _construct(cls, simple: str, rows, path: str, meta):
                    sigma=sg if sg.numel() else torch.zeros(0),
                    numClasses=th.shape[0], numFeatures=th.shape[1],
                    modelType=r0["modelType"])
+    if simple == "BisectingKMeansModel":
+        r0 = rows[0]
+        return cls(centers=torch.from_numpy(matrix_from_struct(r0["clusterCenters"])),
+                   trainingCost=float(r0["trainingCost"]))
     if simple == "GaussianMixtureModel":
         r0 = rows[0]
         mus = torch.stack([torch.from_numpy(vector_from_struct(v))

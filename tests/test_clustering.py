@@ -127,3 +127,48 @@ class TestGaussianMixture:
         assert torch.allclose(m2._covs, m._covs)
         assert torch.equal(m.transform(df)["prediction"],
                            m2.transform(df)["prediction"])
+
+
+class TestBisectingKMeans:
+    def _blobs(self):
+        r = np.random.default_rng(2)
+        blobs = [r.normal(loc=m, scale=0.3, size=(150, 2))
+                 for m in ([-4, 0], [4, 0], [0, 4], [0, -4])]
+        return np.vstack(blobs).astype(np.float32)
+
+    def test_separates_blobs(self):
+        from flowml.clustering import BisectingKMeans
+
+        X = self._blobs()
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = BisectingKMeans(k=4, seed=1).fit(df)
+        pred = m.transform(df)["prediction"].numpy()
+        for b in range(4):
+            seg = pred[b * 150:(b + 1) * 150]
+            vals, cnts = np.unique(seg, return_counts=True)
+            assert cnts.max() / len(seg) > 0.98
+        assert len(set(pred[::150])) >= 3
+
+    def test_min_divisible_cluster_size(self):
+        from flowml.clustering import BisectingKMeans
+
+        X = self._blobs()
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        # whole dataset = 600 rows; forbidding division of clusters smaller
+        # than 601 rows leaves a single split attempt impossible after root
+        m = BisectingKMeans(k=4, seed=1, minDivisibleClusterSize=601.0).fit(df)
+        assert len(m.clusterCenters()) == 1
+
+    def test_round_trip(self, tmp_path):
+        from flowml.clustering import BisectingKMeans, BisectingKMeansModel
+
+        X = self._blobs()
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = BisectingKMeans(k=3, seed=4).fit(df)
+        p = str(tmp_path / "bkm")
+        m.save(p)
+        m2 = BisectingKMeansModel.load(p)
+        assert torch.allclose(m2._centers, m._centers)
+        assert m2.trainingCost == m.trainingCost
+        assert torch.equal(m.transform(df)["prediction"],
+                           m2.transform(df)["prediction"])

@@ -588,3 +588,31 @@ def _selector_dist_body(rank, world):
 def test_selector_dist_equals_single():
     sel = _run_dist("_selector_dist_body")
     assert sel == [3]
+
+
+def _bkm_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.clustering import BisectingKMeans
+
+    r = np.random.default_rng(77)
+    blobs = [r.normal(loc=m, scale=0.3, size=(150, 2))
+             for m in ([-4, 0], [4, 0], [0, 4], [0, -4])]
+    X = np.vstack(blobs).astype(np.float32)
+    perm = r.permutation(len(X))
+    X = X[perm]
+    lo, hi = dist.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi])})
+    m = BisectingKMeans(k=4, seed=2).fit(df)
+    c = np.stack(m.clusterCenters())
+    gathered = dist.all_gather_object(c)
+    assert all(np.allclose(g, gathered[0]) for g in gathered), "ranks diverged"
+    return c
+
+
+def test_bisecting_kmeans_dist_ranks_agree():
+    c = _run_dist("_bkm_dist_body")
+    assert c.shape == (4, 2)
+    # the four blob centers recovered (order-free)
+    want = np.array([[-4, 0], [4, 0], [0, 4], [0, -4]], dtype=np.float64)
+    for w in want:
+        assert np.min(np.linalg.norm(c - w, axis=1)) < 0.3
