@@ -109,6 +109,8 @@ _register("DecisionTreeRegressionModel", "org.apache.spark.ml.regression.Decisio
 _register("GBTRegressor", "org.apache.spark.ml.regression.GBTRegressor")
 _register("GBTRegressionModel", "org.apache.spark.ml.regression.GBTRegressionModel")
 _register("LinearRegression", "org.apache.spark.ml.regression.LinearRegression")
+_register("GeneralizedLinearRegression", "org.apache.spark.ml.regression.GeneralizedLinearRegression")
+_register("GeneralizedLinearRegressionModel", "org.apache.spark.ml.regression.GeneralizedLinearRegressionModel")
 _register("LinearRegressionModel", "org.apache.spark.ml.regression.LinearRegressionModel")
 _register("NaiveBayes", "org.apache.spark.ml.classification.NaiveBayes")
 _register("NaiveBayesModel", "org.apache.spark.ml.classification.NaiveBayesModel")
@@ -575,6 +577,13 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "coefficients": vector_struct(instance._coef.to(torch.float64).cpu().numpy()),
             }
         ]
+    if cls == "GeneralizedLinearRegressionModel":
+        return [
+            {
+                "intercept": float(instance.intercept),
+                "coefficients": vector_struct(instance._coef.to(torch.float64).cpu().numpy()),
+            }
+        ]
     if cls == "LinearRegressionModel":
         return [
             {
@@ -743,6 +752,17 @@ def _construct(cls, simple: str, rows, path: str, meta):
         coef = torch.from_numpy(vector_from_struct(r0["coefficients"])).to(torch.float32)
         return cls(coefficients=coef, intercept=float(r0["intercept"]),
                    numFeatures=int(coef.numel()))
+    if simple == "GeneralizedLinearRegressionModel":
+        r0 = rows[0]
+        fam = meta.get("paramMap", {}).get("family", "gaussian")
+        lnk = meta.get("paramMap", {}).get("link")
+        if lnk is None:
+            defaults = {"gaussian": "identity", "binomial": "logit",
+                        "poisson": "log", "gamma": "inverse"}
+            lnk = defaults.get(fam, "identity")
+        return cls(
+            coefficients=torch.from_numpy(vector_from_struct(r0["coefficients"])),
+            intercept=float(r0["intercept"]), family=fam, link=lnk)
     if simple == "LinearRegressionModel":
         r0 = rows[0]
         coef = torch.from_numpy(vector_from_struct(r0["coefficients"])).to(torch.float32)

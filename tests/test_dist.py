@@ -616,3 +616,31 @@ def test_bisecting_kmeans_dist_ranks_agree():
     want = np.array([[-4, 0], [4, 0], [0, 4], [0, -4]], dtype=np.float64)
     for w in want:
         assert np.min(np.linalg.norm(c - w, axis=1)) < 0.3
+
+
+def _glr_dist_body(rank, world):
+    from flowml import FlowFrame, dist
+    from flowml.regression import GeneralizedLinearRegression
+
+    r = np.random.default_rng(88)
+    X = r.normal(size=(2000, 3))
+    y = r.poisson(np.exp(X @ np.array([0.3, -0.2, 0.1]) + 0.4)).astype(np.float64)
+    lo, hi = dist.shard_range(len(X), rank, world)
+    df = FlowFrame({"features": torch.from_numpy(X[lo:hi].astype(np.float32)),
+                    "label": y[lo:hi]})
+    m = GeneralizedLinearRegression(family="poisson").fit(df)
+    return np.concatenate([m._coef.numpy(), [m.intercept]])
+
+
+def test_glr_dist_equals_single():
+    out = _run_dist("_glr_dist_body")
+    from flowml.frame import FlowFrame
+    from flowml.regression import GeneralizedLinearRegression
+
+    r = np.random.default_rng(88)
+    X = r.normal(size=(2000, 3))
+    y = r.poisson(np.exp(X @ np.array([0.3, -0.2, 0.1]) + 0.4)).astype(np.float64)
+    m = GeneralizedLinearRegression(family="poisson").fit(
+        FlowFrame({"features": torch.from_numpy(X.astype(np.float32)), "label": y}))
+    expect = np.concatenate([m._coef.numpy(), [m.intercept]])
+    assert np.allclose(out, expect, atol=1e-9)

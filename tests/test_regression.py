@@ -241,3 +241,95 @@ class TestExplainedVariance:
         ybar = float(lab.mean())
         expect = float(((pred - ybar) ** 2).mean())
         assert v == pytest.approx(expect)
+
+
+class TestGeneralizedLinearRegression:
+    def _xy(self, seed=0, n=3000):
+        r = np.random.default_rng(seed)
+        X = r.normal(size=(n, 4))
+        beta = np.array([0.4, -0.3, 0.2, 0.1])
+        return r, X, beta, X @ beta + 0.5
+
+    def test_poisson_matches_sklearn(self):
+        from sklearn.linear_model import PoissonRegressor
+
+        from flowml.regression import GeneralizedLinearRegression
+
+        r, X, beta, eta = self._xy()
+        y = r.poisson(np.exp(eta)).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X.astype(np.float32)),
+                        "label": y})
+        m = GeneralizedLinearRegression(family="poisson").fit(df)
+        sk = PoissonRegressor(alpha=0.0, max_iter=500, tol=1e-10).fit(X, y)
+        assert np.allclose(m._coef.numpy(), sk.coef_, atol=2e-4)
+        assert abs(m.intercept - sk.intercept_) < 2e-4
+        pred = m.transform(df)["prediction"].numpy()
+        assert np.allclose(pred, np.exp(X @ m._coef.numpy() + m.intercept),
+                           rtol=1e-6)
+
+    def test_binomial_matches_sklearn(self):
+        from sklearn.linear_model import LogisticRegression as SkLR
+
+        from flowml.regression import GeneralizedLinearRegression
+
+        r, X, beta, eta = self._xy(1)
+        y = (r.random(len(X)) < 1 / (1 + np.exp(-eta))).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X.astype(np.float32)),
+                        "label": y})
+        m = GeneralizedLinearRegression(family="binomial").fit(df)
+        sk = SkLR(penalty=None, max_iter=500, tol=1e-10).fit(X, y)
+        assert np.allclose(m._coef.numpy(), sk.coef_[0], atol=5e-4)
+
+    def test_gamma_matches_sklearn(self):
+        from sklearn.linear_model import GammaRegressor
+
+        from flowml.regression import GeneralizedLinearRegression
+
+        r, X, beta, eta = self._xy(2)
+        y = r.gamma(shape=5.0, scale=np.exp(eta) / 5.0)
+        df = FlowFrame({"features": torch.from_numpy(X.astype(np.float32)),
+                        "label": y})
+        m = GeneralizedLinearRegression(family="gamma", link="log").fit(df)
+        sk = GammaRegressor(alpha=0.0, max_iter=1000, tol=1e-12).fit(X, y)
+        assert np.allclose(m._coef.numpy(), sk.coef_, atol=1e-3)
+
+    def test_gaussian_identity_equals_ols(self):
+        from flowml.regression import GeneralizedLinearRegression
+
+        r, X, beta, eta = self._xy(3)
+        y = eta + 0.1 * r.normal(size=len(X))
+        df = FlowFrame({"features": torch.from_numpy(X.astype(np.float32)),
+                        "label": y})
+        m = GeneralizedLinearRegression().fit(df)
+        Xa = np.column_stack([X, np.ones(len(X))])
+        ols, *_ = np.linalg.lstsq(Xa, y, rcond=None)
+        assert np.allclose(m._coef.numpy(), ols[:4], atol=1e-8)
+        assert abs(m.intercept - ols[4]) < 1e-8
+
+    def test_link_prediction_col_and_round_trip(self, tmp_path):
+        from flowml.regression import (GeneralizedLinearRegression,
+                                       GeneralizedLinearRegressionModel)
+
+        r, X, beta, eta = self._xy(4)
+        y = r.poisson(np.exp(eta)).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X.astype(np.float32)),
+                        "label": y})
+        m = GeneralizedLinearRegression(
+            family="poisson", linkPredictionCol="eta").fit(df)
+        out = m.transform(df)
+        assert np.allclose(np.exp(out["eta"].numpy()),
+                           out["prediction"].numpy(), rtol=1e-6)
+        p = str(tmp_path / "glr")
+        m.save(p)
+        m2 = GeneralizedLinearRegressionModel.load(p)
+        assert np.allclose(m2._coef.numpy(), m._coef.numpy())
+        assert m2.getOrDefault(m2.family) == "poisson"
+        assert torch.allclose(m.transform(df)["prediction"],
+                              m2.transform(df)["prediction"])
+
+    def test_unsupported_family(self):
+        from flowml.regression import GeneralizedLinearRegression
+
+        df = FlowFrame({"features": torch.zeros(4, 2), "label": np.ones(4)})
+        with pytest.raises(ValueError, match="tweedie"):
+            GeneralizedLinearRegression(family="tweedie").fit(df)
