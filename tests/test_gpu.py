@@ -906,3 +906,30 @@ class TestClusteringGPU:
         mu_c = np.sort(m_cpu._mus.numpy(), axis=0)
         mu_g = np.sort(m_gpu._mus.cpu().numpy(), axis=0)
         assert np.allclose(mu_c, mu_g, atol=0.05)
+
+    def test_new_families_fit_on_gpu(self):
+        """GLR (f64 solve), BisectingKMeans, selectors on device frames."""
+        from flowml.clustering import BisectingKMeans
+        from flowml.feature import RobustScaler, UnivariateFeatureSelector
+        from flowml.frame import FlowFrame
+        from flowml.regression import GeneralizedLinearRegression
+
+        r = np.random.default_rng(62)
+        X = r.normal(size=(20_000, 5)).astype(np.float32)
+        yp = r.poisson(np.exp(X[:, 0] * 0.3 + 0.2)).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": yp}).to(DEV)
+        m = GeneralizedLinearRegression(family="poisson").fit(df)
+        cpu = GeneralizedLinearRegression(family="poisson").fit(
+            FlowFrame({"features": torch.from_numpy(X), "label": yp}))
+        assert np.allclose(m._coef.numpy(), cpu._coef.numpy(), atol=1e-8)
+        yk = r.integers(0, 3, 20_000).astype(np.float64)
+        dfk = FlowFrame({"features": torch.from_numpy(X), "label": yk}).to(DEV)
+        sel = UnivariateFeatureSelector(
+            featuresCol="features", outputCol="s", labelCol="label",
+            featureType="continuous", labelType="categorical",
+            selectionThreshold=2.0).fit(dfk)
+        assert len(sel.selectedFeatures) == 2
+        rs = RobustScaler(inputCol="features", outputCol="r").fit(df)
+        assert rs.median.shape == (5,)
+        bk = BisectingKMeans(k=3, seed=1).fit(df)
+        assert len(bk.clusterCenters()) == 3
