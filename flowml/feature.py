@@ -40,6 +40,7 @@ from flowml.param import (
     HasOutputCol,
     HasOutputCols,
     Param,
+    Params,
     TypeConverters,
 )
 
@@ -1369,6 +1370,170 @@ class QuantileDiscretizer(HasInputCol, HasOutputCol, HasHandleInvalid, Estimator
         return Bucketizer(splits=splits, inputCol=self.getInputCol(),
                           outputCol=self.getOutputCol(),
                           handleInvalid=self.getHandleInvalid())
+
+
+# ---------------------------------------------------------------------------
+# RFormula (spark:.../feature/RFormula.scala)
+# ---------------------------------------------------------------------------
+
+
+class _RFormulaParams(Params):
+    formula: Param[str] = Param(
+        "undefined", "formula", "R model formula", TypeConverters.toString)
+    featuresCol: Param[str] = Param(
+        "undefined", "featuresCol", "features column name", TypeConverters.toString)
+    labelCol: Param[str] = Param(
+        "undefined", "labelCol", "label column name", TypeConverters.toString)
+    handleInvalid: Param[str] = Param(
+        "undefined", "handleInvalid",
+        "how to handle invalid entries: error|skip|keep", TypeConverters.toString)
+
+    def __init__(self):
+        super().__init__()
+        self._setDefault(featuresCol="features", labelCol="label",
+                         handleInvalid="error")
+
+
+def _parse_r_formula(formula: str, columns: List[str]):
+    """Parse ``label ~ term + term - term`` with ``.`` = all other columns
+    and ``a:b`` numeric interactions. Returns (label, terms) where each term
+    is a single column name or a tuple of names (interaction)."""
+    if "~" not in formula:
+        raise ValueError(f"formula must contain '~': {formula!r}")
+    lhs, rhs = formula.split("~", 1)
+    label = lhs.strip()
+    if not label:
+        raise ValueError("formula needs a label on the left of '~'")
+    include: List = []
+    exclude: set = set()
+    for chunk in rhs.replace("-", "+-").split("+"):
+        t = chunk.strip()
+        if not t:
+            continue
+        neg = t.startswith("-")
+        if neg:
+            t = t[1:].strip()
+        if t == ".":
+            for c in columns:
+                if c != label and c not in include:
+                    include.append(c)
+        elif ":" in t:
+            parts = tuple(p.strip() for p in t.split(":"))
+            (exclude.add(parts) if neg else include.append(parts))
+        else:
+            (exclude.add(t) if neg else include.append(t))
+    terms = [t for t in include if t not in exclude]
+    return label, terms
+
+
+class RFormula(_RFormulaParams, Estimator):
+    """R-style model formula (spark:.../feature/RFormula.scala): string
+    columns are StringIndexed + one-hot encoded (dropLast), numeric columns
+    pass through, ``a:b`` numeric interactions multiply, ``.`` expands to
+    all non-label columns and ``- x`` excludes; a string label is
+    StringIndexed into ``labelCol``.
+
+    >>> import numpy as np, torch
+    >>> from flowml.frame import FlowFrame
+    >>> df = FlowFrame({"clicked": np.array(["no", "yes", "no"], dtype=object),
+    ...                 "country": np.array(["us", "ca", "us"], dtype=object),
+    ...                 "hour": torch.tensor([18.0, 12.0, 15.0])})
+    >>> m = RFormula(formula="clicked ~ country + hour").fit(df)
+    >>> out = m.transform(df)
+    >>> out["features"].shape
+    torch.Size([3, 2])
+    >>> out["label"].tolist()
+    [0.0, 1.0, 0.0]
+    """
+
+    def __init__(self, *, formula: Optional[str] = None,
+                 featuresCol: Optional[str] = None,
+                 labelCol: Optional[str] = None,
+                 handleInvalid: Optional[str] = None):
+        super().__init__()
+        self._set(formula=formula, featuresCol=featuresCol, labelCol=labelCol,
+                  handleInvalid=handleInvalid)
+
+    def setFormula(self, v: str):
+        return self.set(self.formula, v)
+
+    def _fit(self, dataset: FlowFrame) -> "RFormulaModel":
+        label, terms = _parse_r_formula(self.getOrDefault(self.formula),
+                                        dataset.columns)
+        handle = self.getOrDefault(self.handleInvalid)
+        stages: List = []
+        assemble: List[str] = []
+        tmp = dataset
+        for t in terms:
+            if isinstance(t, tuple):
+                for c in t:
+                    if tmp.is_string(c):
+                        raise ValueError(
+                            f"interaction term {':'.join(t)} requires numeric "
+                            f"columns; {c!r} is a string column")
+                assemble.append("__rf_inter_" + "_".join(t))
+                continue
+            if tmp.is_string(t):
+                si = StringIndexer(inputCol=t, outputCol=f"__rf_{t}_idx",
+                                   handleInvalid="keep" if handle == "keep" else handle)
+                sim = si.fit(tmp)
+                tmp = sim.transform(tmp)
+                ohe = OneHotEncoder(inputCol=f"__rf_{t}_idx",
+                                    outputCol=f"__rf_{t}_vec",
+                                    handleInvalid="keep" if handle == "keep" else "error")
+                ohm = ohe.fit(tmp)
+                tmp = ohm.transform(tmp)
+                stages += [sim, ohm]
+                assemble.append(f"__rf_{t}_vec")
+            else:
+                assemble.append(t)
+        label_stage = None
+        if label in dataset.columns and dataset.is_string(label):
+            label_stage = StringIndexer(
+                inputCol=label, outputCol=self.getOrDefault(self.labelCol)
+            ).fit(dataset)
+        model = RFormulaModel(
+            terms=terms, stages=stages, assemble_cols=assemble,
+            label_source=label, label_stage=label_stage)
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class RFormulaModel(_RFormulaParams, Model):
+    def __init__(self, terms=None, stages=None, assemble_cols=None,
+                 label_source: str = "", label_stage=None):
+        super().__init__()
+        self._terms = terms or []
+        self._stages = stages or []
+        self._assemble = assemble_cols or []
+        self._label_source = label_source
+        self._label_stage = label_stage
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        out = dataset
+        for st in self._stages:
+            out = st.transform(out)
+        for t in self._terms:
+            if isinstance(t, tuple):
+                col = out.col(t[0]).reshape(-1).clone()
+                for c in t[1:]:
+                    col = col * out.col(c).reshape(-1)
+                out = out.withColumn("__rf_inter_" + "_".join(t), col)
+        out = VectorAssembler(
+            inputCols=self._assemble,
+            outputCol=self.getOrDefault(self.featuresCol),
+            handleInvalid=self.getOrDefault(self.handleInvalid)
+            if self.getOrDefault(self.handleInvalid) != "error" else "error",
+        ).transform(out)
+        lab = self.getOrDefault(self.labelCol)
+        if self._label_stage is not None:
+            out = self._label_stage.transform(out)
+        elif self._label_source in out.columns and self._label_source != lab:
+            out = out.withColumn(lab, out.col(self._label_source))
+        # drop the internal temp columns
+        drop = [c for c in out.columns if c.startswith("__rf_")]
+        return out.drop(*drop) if drop else out
 
 
 # ---------------------------------------------------------------------------

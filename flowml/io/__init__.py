@@ -77,6 +77,8 @@ _register("PCA", "org.apache.spark.ml.feature.PCA")
 _register("PCAModel", "org.apache.spark.ml.feature.PCAModel")
 _register("VectorSlicer", "org.apache.spark.ml.feature.VectorSlicer")
 _register("VectorIndexer", "org.apache.spark.ml.feature.VectorIndexer")
+_register("RFormula", "org.apache.spark.ml.feature.RFormula")
+_register("RFormulaModel", "org.apache.spark.ml.feature.RFormulaModel")
 _register("RobustScaler", "org.apache.spark.ml.feature.RobustScaler")
 _register("UnivariateFeatureSelector", "org.apache.spark.ml.feature.UnivariateFeatureSelector")
 _register("UnivariateFeatureSelectorModel", "org.apache.spark.ml.feature.UnivariateFeatureSelectorModel")
@@ -312,6 +314,21 @@ def _table_from_rows(rows: List[Dict[str, Any]]):
 def _save_instance(instance, path: str) -> None:
     cls = type(instance).__name__
     os.makedirs(path, exist_ok=True)
+    if cls == "RFormulaModel":
+        # Spark nests the fitted featurization inside the model dir; here:
+        # metadata (params + resolved terms) + stages/<i>_<uid>/ + labelStage/
+        save_metadata(instance, path, extra={
+            "resolvedTerms": [list(t) if isinstance(t, tuple) else t
+                              for t in instance._terms],
+            "assembleCols": list(instance._assemble),
+            "labelSource": instance._label_source,
+            "hasLabelStage": instance._label_stage is not None,
+        })
+        _save_pipeline_stages(instance._stages, path)
+        if instance._label_stage is not None:
+            _save_instance(instance._label_stage,
+                           os.path.join(path, "labelStage"))
+        return
     if cls == "Pipeline":
         _save_pipeline_stages(instance.getStages(), path)
         save_metadata(instance, path)
@@ -645,6 +662,27 @@ def load(path: str):
             inst = cls(bestModel=best, avgMetrics=metrics)
         else:
             inst = cls(bestModel=best, validationMetrics=metrics)
+        _apply_params(inst, meta)
+        return inst
+
+    if simple == "RFormulaModel":
+        stages_dir = os.path.join(path, "stages")
+        stages = []
+        if os.path.isdir(stages_dir):
+            entries = sorted(os.listdir(stages_dir),
+                             key=lambda e: int(e.split("_", 1)[0]))
+            stages = [load(os.path.join(stages_dir, e)) for e in entries]
+        label_stage = None
+        if meta.get("hasLabelStage") and os.path.isdir(
+            os.path.join(path, "labelStage")
+        ):
+            label_stage = load(os.path.join(path, "labelStage"))
+        terms = [tuple(t) if isinstance(t, list) else t
+                 for t in meta.get("resolvedTerms", [])]
+        inst = cls(terms=terms, stages=stages,
+                   assemble_cols=list(meta.get("assembleCols", [])),
+                   label_source=meta.get("labelSource", ""),
+                   label_stage=label_stage)
         _apply_params(inst, meta)
         return inst
 

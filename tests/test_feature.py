@@ -498,3 +498,70 @@ class TestFeatureSelectors:
             UnivariateFeatureSelector(
                 featuresCol="features", outputCol="s", labelCol="label",
                 featureType="categorical", labelType="continuous").fit(df)
+
+
+class TestRFormula:
+    def _df(self):
+        r = np.random.default_rng(11)
+        return FlowFrame({
+            "clicked": np.array(r.choice(["no", "yes"], 40), dtype=object),
+            "country": np.array(r.choice(["us", "ca", "mx"], 40), dtype=object),
+            "hour": torch.from_numpy(r.normal(size=40).astype(np.float32)),
+            "bytes": torch.from_numpy(r.normal(size=40).astype(np.float32)),
+        })
+
+    def test_dot_and_exclusion(self):
+        from flowml.feature import RFormula
+
+        df = self._df()
+        m = RFormula(formula="clicked ~ . - bytes").fit(df)
+        out = m.transform(df)
+        # country one-hot (3 cats, dropLast -> 2) + hour
+        assert out["features"].shape == (40, 3)
+        assert set(out["label"].tolist()) <= {0.0, 1.0}
+        # internal temp columns are dropped
+        assert not [c for c in out.columns if c.startswith("__rf_")]
+
+    def test_interaction_terms(self):
+        from flowml.feature import RFormula
+
+        df = self._df()
+        m = RFormula(formula="clicked ~ hour:bytes").fit(df)
+        out = m.transform(df)
+        expect = df["hour"] * df["bytes"]
+        assert torch.allclose(out["features"][:, 0], expect)
+        with pytest.raises(ValueError, match="numeric"):
+            RFormula(formula="clicked ~ hour:country").fit(df)
+
+    def test_numeric_label_passthrough(self):
+        from flowml.feature import RFormula
+
+        df = self._df().withColumn("target", torch.arange(40.0))
+        m = RFormula(formula="target ~ hour + bytes").fit(df)
+        out = m.transform(df)
+        assert torch.equal(out["label"], torch.arange(40.0))
+
+    def test_trains_a_classifier_end_to_end(self):
+        from flowml.classification import LogisticRegression
+        from flowml.feature import RFormula
+        from flowml.pipeline import Pipeline
+
+        df = self._df()
+        model = Pipeline(stages=[
+            RFormula(formula="clicked ~ ."),
+            LogisticRegression(maxIter=10),
+        ]).fit(df)
+        assert "prediction" in model.transform(df).columns
+
+    def test_round_trip(self, tmp_path):
+        from flowml.feature import RFormula, RFormulaModel
+
+        df = self._df()
+        m = RFormula(formula="clicked ~ country + hour:bytes").fit(df)
+        p = str(tmp_path / "rf")
+        m.save(p)
+        m2 = RFormulaModel.load(p)
+        a = m.transform(df)
+        b = m2.transform(df)
+        assert torch.equal(a["features"], b["features"])
+        assert torch.equal(a["label"], b["label"])
