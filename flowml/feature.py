@@ -1523,8 +1523,7 @@ class RFormulaModel(_RFormulaParams, Model):
         out = VectorAssembler(
             inputCols=self._assemble,
             outputCol=self.getOrDefault(self.featuresCol),
-            handleInvalid=self.getOrDefault(self.handleInvalid)
-            if self.getOrDefault(self.handleInvalid) != "error" else "error",
+            handleInvalid=self.getOrDefault(self.handleInvalid),
         ).transform(out)
         lab = self.getOrDefault(self.labelCol)
         if self._label_stage is not None:
@@ -1579,6 +1578,11 @@ def _selector_stats(X: torch.Tensor, y: torch.Tensor, feature_type: str,
                              float(ya.max(initial=0))])
         dist.all_reduce_(vmax, op="max")
         V, C = int(vmax[0]) + 1, int(vmax[1]) + 1
+        if F * V * C > 100_000_000:
+            raise ValueError(
+                f"chi2 contingency would need {F}x{V}x{C} cells — a feature "
+                "declared categorical has huge arity; use "
+                "featureType='continuous' for continuous columns")
         # contingency counts per feature: F×V×C, flat all_reduce
         cont = np.zeros((F, V, C), np.float64)
         for j in range(F):
