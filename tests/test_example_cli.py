@@ -54,3 +54,16 @@ def test_cli_end_to_end_gpu(featurize, tmp_path):
     assert r.returncode == 0, r.stdout[-800:] + r.stderr[-800:]
     assert "test accuracy=" in r.stdout
     assert os.path.isdir(os.path.join(save, "stages"))
+
+
+def test_anomaly_example_cpu():
+    cli = os.path.join(REPO, "examples", "detect_anomalies.py")
+    r = subprocess.run(
+        [sys.executable, cli, "--rows", "6000", "--device", "cpu"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert r.returncode == 0, r.stdout[-500:] + r.stderr[-500:]
+    line = [l for l in r.stdout.splitlines() if "precision" in l][0]
+    prec = float(line.split("precision=")[1].split()[0])
+    rec = float(line.split("recall=")[1].split()[0])
+    assert prec > 0.9 and rec > 0.9, line
