@@ -59,7 +59,7 @@ def test_cli_end_to_end_gpu(featurize, tmp_path):
 def test_anomaly_example_cpu():
     cli = os.path.join(REPO, "examples", "detect_anomalies.py")
     r = subprocess.run(
-        [sys.executable, cli, "--rows", "6000", "--device", "cpu"],
+        [sys.executable, cli, "--rows", "8000", "--device", "cpu"],
         capture_output=True, text=True, timeout=300, cwd=REPO,
     )
     assert r.returncode == 0, r.stdout[-500:] + r.stderr[-500:]
