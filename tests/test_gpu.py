@@ -933,3 +933,26 @@ class TestClusteringGPU:
         assert rs.median.shape == (5,)
         bk = BisectingKMeans(k=3, seed=1).fit(df)
         assert len(bk.clusterCenters()) == 3
+
+
+class TestFullScaleDeterminism:
+    def test_flagship_rf_fit_bitwise_repeatable(self):
+        """Two identical flagship-shaped fits (1.25M rows × 100 trees) give
+        BIT-IDENTICAL forests — integer histogram counts make growth exact,
+        so any nondeterminism (atomics ordering, RNG drift) shows here."""
+        from flowml.classification import RandomForestClassifier
+        from flowml.frame import FlowFrame
+
+        g = torch.Generator(device=DEV).manual_seed(5)
+        X = torch.randn(1_250_000, 41, device=DEV, generator=g)
+        y = (X[:, 0] + X[:, 5] > 0).to(torch.float64) + 2.0 * (
+            X[:, 3] > 1).to(torch.float64)
+        df = FlowFrame({"features": X, "label": y}, device=DEV)
+        m1 = RandomForestClassifier(numTrees=100, maxDepth=5, seed=9).fit(df)
+        m2 = RandomForestClassifier(numTrees=100, maxDepth=5, seed=9).fit(df)
+        for t1, t2 in zip(m1._trees, m2._trees):
+            assert np.array_equal(t1.feature, t2.feature)
+            assert np.array_equal(t1.thr_raw[t1.feature >= 0],
+                                  t2.thr_raw[t2.feature >= 0])
+            assert np.array_equal(t1.stats, t2.stats)
+            assert np.array_equal(t1.left, t2.left)
