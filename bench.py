@@ -418,7 +418,32 @@ def main():
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--report-f1-curve", action="store_true",
                    help="print RF fit-time/macro-F1 at depths 5/8/10 and exit")
+    p.add_argument("--profile", action="store_true",
+                   help="re-exec this bench under rocprofv3 --kernel-trace "
+                        "--stats (output under gpurun_out/prof_bench_<model>)")
     args = p.parse_args()
+
+    if args.profile:
+        # rocprof wired into the bench harness. Re-exec
+        # under rocprofv3 from /tmp (pool rule) with the same arguments.
+        import os
+        import shutil
+        import subprocess
+        import sys as _sys
+
+        if shutil.which("rocprofv3") is None:
+            raise SystemExit("--profile needs rocprofv3 on PATH (GPU box)")
+        here = os.path.dirname(os.path.abspath(__file__))
+        outdir = os.path.join(here, "gpurun_out", f"prof_bench_{args.model}")
+        argv = [a for a in _sys.argv[1:] if a != "--profile"]
+        env = dict(os.environ, TMPDIR="/tmp")
+        r = subprocess.run(
+            ["rocprofv3", "--kernel-trace", "--stats", "--output-format",
+             "csv", "-d", outdir, "--",
+             _sys.executable, os.path.abspath(__file__), *argv],
+            cwd="/tmp", env=env,
+        )
+        raise SystemExit(r.returncode)
 
     dist.init_from_env()
     world = dist.get_world_size()
