@@ -819,9 +819,71 @@ class GeneralizedLinearRegression(_GLRParams, Estimator, _RegressorIOMixin):
         model = GeneralizedLinearRegressionModel(
             coefficients=coef.to(torch.float64).cpu(), intercept=intercept,
             family=fam, link=link, deviance=dev_prev or 0.0, numIter=it + 1)
+        # training summary (GeneralizedLinearRegressionTrainingSummary):
+        # Fisher information at convergence -> covariance of beta.
+        # Pearson X² = Σ w (y-mu)²/V(mu), dispersion fixed at 1 for
+        # binomial/poisson else X²/df; z-test p-values when dispersion is
+        # fixed, t-test otherwise (Spark's convention).
+        gp = gprime(mu)
+        irls_w = 1.0 / (varfn(mu) * gp * gp).clamp_min(1e-300)
+        pearson = ((y - mu) ** 2 / varfn(mu).clamp_min(1e-300))
+        if w is not None:
+            irls_w = irls_w * w
+            pearson = pearson * w
+        Xw = Xa * irls_w.unsqueeze(1)
+        buf2 = torch.cat([(Xw.T @ Xa).reshape(-1), pearson.sum().reshape(1),
+                          torch.tensor([float(N)], dtype=torch.float64,
+                                       device=X.device)])
+        dist.all_reduce_(buf2)
+        A_final = buf2[: P * P].view(P, P)
+        pearson_x2 = float(buf2[P * P])
+        n_tot = float(buf2[P * P + 1])
+        df_resid = max(n_tot - P, 1.0)
+        fixed_disp = fam in ("binomial", "poisson")
+        dispersion = 1.0 if fixed_disp else pearson_x2 / df_resid
+        try:
+            cov = dispersion * torch.linalg.inv(A_final)
+            se = torch.sqrt(torch.diagonal(cov).clamp_min(0))
+            tvals = beta / se.clamp_min(1e-300)
+            import numpy as np
+            import scipy.stats as _st
+
+            tv = tvals.cpu().numpy()
+            if fixed_disp:
+                pv = 2.0 * _st.norm.sf(np.abs(tv))
+            else:
+                pv = 2.0 * _st.t.sf(np.abs(tv), df_resid)
+            model._summary = GeneralizedLinearRegressionTrainingSummary(
+                coefficientStandardErrors=[float(v) for v in se],
+                tValues=[float(v) for v in tv],
+                pValues=[float(v) for v in pv],
+                dispersion=float(dispersion), deviance=float(dev_prev or 0.0),
+                residualDegreeOfFreedom=int(df_resid),
+                numIterations=it + 1)
+        except Exception as _e:  # singular information matrix: no summary
+            import os as _os
+            if _os.environ.get("FLOWML_DEBUG", "0") == "1":
+                raise
+            model._summary = None
         self._copyValues(model)
         model._set_parent(self)
         return model
+
+
+class GeneralizedLinearRegressionTrainingSummary:
+    """Spark GeneralizedLinearRegressionTrainingSummary subset: inference
+    statistics from the Fisher information at convergence. Ordering follows
+    Spark: coefficients first, intercept LAST."""
+
+    def __init__(self, coefficientStandardErrors, tValues, pValues,
+                 dispersion, deviance, residualDegreeOfFreedom, numIterations):
+        self.coefficientStandardErrors = coefficientStandardErrors
+        self.tValues = tValues
+        self.pValues = pValues
+        self.dispersion = dispersion
+        self.deviance = deviance
+        self.residualDegreeOfFreedom = residualDegreeOfFreedom
+        self.numIterations = numIterations
 
 
 class GeneralizedLinearRegressionModel(_GLRParams, Model, _RegressorIOMixin):
@@ -836,7 +898,18 @@ class GeneralizedLinearRegressionModel(_GLRParams, Model, _RegressorIOMixin):
         self._link = link
         self.deviance = float(deviance)
         self.numIter = int(numIter)
+        self._summary = None
         self._set(family=family, link=link)
+
+    @property
+    def summary(self) -> "GeneralizedLinearRegressionTrainingSummary":
+        if self._summary is None:
+            raise RuntimeError(
+                "No training summary available (absent after load, like Spark)")
+        return self._summary
+
+    def hasSummary(self) -> bool:
+        return self._summary is not None
 
     @property
     def coefficients(self):
@@ -859,6 +932,7 @@ class GeneralizedLinearRegressionModel(_GLRParams, Model, _RegressorIOMixin):
 __all__ = [
     "RegressionModel",
     "GeneralizedLinearRegression", "GeneralizedLinearRegressionModel",
+    "GeneralizedLinearRegressionTrainingSummary",
     "LinearRegression", "LinearRegressionModel", "LinearRegressionTrainingSummary",
     "DecisionTreeRegressor", "DecisionTreeRegressionModel",
     "RandomForestRegressor", "RandomForestRegressionModel",

@@ -333,3 +333,67 @@ class TestGeneralizedLinearRegression:
         df = FlowFrame({"features": torch.zeros(4, 2), "label": np.ones(4)})
         with pytest.raises(ValueError, match="tweedie"):
             GeneralizedLinearRegression(family="tweedie").fit(df)
+
+
+class TestGLRSummary:
+    def test_gaussian_summary_matches_ols(self):
+        import scipy.stats as st
+
+        from flowml.regression import GeneralizedLinearRegression
+
+        r = np.random.default_rng(0)
+        X = r.normal(size=(500, 3))
+        y = X @ np.array([1.0, -2.0, 0.0]) + 0.7 + r.normal(scale=1.5, size=500)
+        df = FlowFrame({"features": torch.from_numpy(X.astype(np.float32)),
+                        "label": y})
+        m = GeneralizedLinearRegression().fit(df)
+        s = m.summary
+        Xa = np.column_stack([X, np.ones(500)])
+        beta = np.concatenate([m._coef.numpy(), [m.intercept]])
+        resid = y - Xa @ beta
+        sigma2 = resid @ resid / 496
+        se = np.sqrt(np.diag(sigma2 * np.linalg.inv(Xa.T @ Xa)))
+        assert np.allclose(s.coefficientStandardErrors, se, rtol=1e-4)
+        assert np.allclose(s.tValues, beta / se, rtol=1e-4)
+        assert np.allclose(
+            s.pValues, 2 * st.t.sf(np.abs(beta / se), 496),
+            rtol=1e-3, atol=1e-12)
+        assert s.pValues[2] > 0.05 and s.pValues[0] < 1e-6  # x2 is a null
+        assert abs(s.dispersion - sigma2) < 1e-6
+        assert s.residualDegreeOfFreedom == 496
+
+    def test_poisson_summary_z_inference(self):
+        from flowml.regression import GeneralizedLinearRegression
+
+        r = np.random.default_rng(1)
+        X = r.normal(size=(4000, 3))
+        y = r.poisson(np.exp(X @ np.array([0.5, 0.0, -0.3]) + 0.2)).astype(
+            np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X.astype(np.float32)),
+                        "label": y})
+        m = GeneralizedLinearRegression(family="poisson").fit(df)
+        s = m.summary
+        assert s.dispersion == 1.0  # fixed for poisson
+        assert s.pValues[1] > 0.01      # null coefficient
+        assert s.pValues[0] < 1e-10 and s.pValues[2] < 1e-10
+        # +-2 SE covers the truth for the informative coefficients
+        for i, truth in ((0, 0.5), (2, -0.3)):
+            assert abs(m._coef[i].item() - truth) < 3 * s.coefficientStandardErrors[i]
+
+    def test_summary_absent_after_load(self, tmp_path):
+        from flowml.regression import (GeneralizedLinearRegression,
+                                       GeneralizedLinearRegressionModel)
+
+        r = np.random.default_rng(2)
+        X = r.normal(size=(200, 2))
+        y = X[:, 0] + r.normal(size=200)
+        df = FlowFrame({"features": torch.from_numpy(X.astype(np.float32)),
+                        "label": y})
+        m = GeneralizedLinearRegression().fit(df)
+        assert m.hasSummary()
+        p = str(tmp_path / "g")
+        m.save(p)
+        m2 = GeneralizedLinearRegressionModel.load(p)
+        assert not m2.hasSummary()
+        with pytest.raises(RuntimeError, match="summary"):
+            _ = m2.summary
