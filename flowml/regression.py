@@ -454,6 +454,42 @@ class LinearRegression(_LinearRegressionParams, Estimator, _RegressorIOMixin):
             labelCol=self.getLabelCol(),
             predictionCol=self.getOrDefault(self.predictionCol),
         )
+        if reg == 0:
+            # inference statistics (Spark exposes stdErrors/tValues/pValues
+            # only for unregularized fits): OLS covariance from the
+            # all-reduced normal equations on the RAW features
+            Xd = X.to(torch.float64)
+            P = F + (1 if fit_intercept else 0)
+            Xa = torch.cat([Xd, torch.ones(N, 1, dtype=torch.float64,
+                                           device=dev)], dim=1) if fit_intercept else Xd
+            beta_raw = torch.cat([
+                coef.to(torch.float64).to(dev),
+                torch.tensor([intercept], dtype=torch.float64, device=dev)
+            ]) if fit_intercept else coef.to(torch.float64).to(dev)
+            r_vec = y.to(torch.float64) - Xa @ beta_raw
+            wv = w.to(torch.float64) if w is not None else None
+            Xw = Xa * wv.unsqueeze(1) if wv is not None else Xa
+            rss = (wv * r_vec * r_vec).sum() if wv is not None else (r_vec * r_vec).sum()
+            buf = torch.cat([(Xw.T @ Xa).reshape(-1), rss.reshape(1),
+                             torch.tensor([float(N)], dtype=torch.float64,
+                                          device=dev)])
+            dist.all_reduce_(buf)
+            try:
+                import numpy as _np
+                import scipy.stats as _st
+
+                A = buf[: P * P].view(P, P)
+                df_resid = max(float(buf[-1]) - P, 1.0)
+                sigma2 = float(buf[P * P]) / df_resid
+                cov = sigma2 * torch.linalg.inv(A)
+                se = torch.sqrt(torch.diagonal(cov).clamp_min(0)).cpu().numpy()
+                tv = beta_raw.cpu().numpy() / _np.maximum(se, 1e-300)
+                pv = 2.0 * _st.t.sf(_np.abs(tv), df_resid)
+                model._summary._inference = (
+                    [float(v) for v in se], [float(v) for v in tv],
+                    [float(v) for v in pv], int(df_resid))
+            except Exception:
+                pass
         return model
 
 
@@ -504,6 +540,32 @@ class LinearRegressionTrainingSummary:
         self.labelCol = labelCol
         self.predictionCol = predictionCol
         self._pred = None
+        self._inference = None  # (se, t, p, df) when regParam == 0
+
+    def _infer_stats(self, i):
+        if self._inference is None:
+            raise RuntimeError(
+                "No Std. Error of coefficients available for this "
+                "LinearRegressionModel (regularized fit — Spark parity)")
+        return self._inference[i]
+
+    @property
+    def coefficientStandardErrors(self):
+        """OLS standard errors, coefficients first, intercept last
+        (unregularized fits only — like Spark)."""
+        return self._infer_stats(0)
+
+    @property
+    def tValues(self):
+        return self._infer_stats(1)
+
+    @property
+    def pValues(self):
+        return self._infer_stats(2)
+
+    @property
+    def degreesOfFreedom(self):
+        return self._infer_stats(3)
 
     @property
     def predictions(self):

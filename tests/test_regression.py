@@ -397,3 +397,32 @@ class TestGLRSummary:
         assert not m2.hasSummary()
         with pytest.raises(RuntimeError, match="summary"):
             _ = m2.summary
+
+
+class TestLinearRegressionInferenceStats:
+    def test_ols_stats_match(self):
+        import scipy.stats as st
+
+        r = np.random.default_rng(0)
+        X = r.normal(size=(500, 3))
+        y = X @ np.array([1.0, -2.0, 0.0]) + 0.7 + r.normal(scale=1.5, size=500)
+        df = FlowFrame({"features": torch.from_numpy(X.astype(np.float32)),
+                        "label": y})
+        m = LinearRegression(maxIter=200).fit(df)
+        s = m.summary
+        Xa = np.column_stack([X, np.ones(500)])
+        ols, *_ = np.linalg.lstsq(Xa, y, rcond=None)
+        resid = y - Xa @ ols
+        se = np.sqrt(np.diag((resid @ resid / 496) * np.linalg.inv(Xa.T @ Xa)))
+        assert np.allclose(s.coefficientStandardErrors, se, rtol=1e-2)
+        assert s.degreesOfFreedom == 496
+        assert s.pValues[2] > 0.05 and s.pValues[0] < 1e-6
+
+    def test_regularized_raises_like_spark(self):
+        r = np.random.default_rng(1)
+        X = r.normal(size=(100, 2)).astype(np.float32)
+        y = X[:, 0].astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y})
+        m = LinearRegression(regParam=0.1).fit(df)
+        with pytest.raises(RuntimeError, match="Std. Error"):
+            _ = m.summary.pValues
