@@ -403,6 +403,8 @@ class TestLinearRegressionInferenceStats:
     def test_ols_stats_match(self):
         import scipy.stats as st
 
+        from flowml.regression import LinearRegression
+
         r = np.random.default_rng(0)
         X = r.normal(size=(500, 3))
         y = X @ np.array([1.0, -2.0, 0.0]) + 0.7 + r.normal(scale=1.5, size=500)
@@ -419,6 +421,8 @@ class TestLinearRegressionInferenceStats:
         assert s.pValues[2] > 0.05 and s.pValues[0] < 1e-6
 
     def test_regularized_raises_like_spark(self):
+        from flowml.regression import LinearRegression
+
         r = np.random.default_rng(1)
         X = r.normal(size=(100, 2)).astype(np.float32)
         y = X[:, 0].astype(np.float64)
