@@ -30,7 +30,8 @@ from flowml.classification.base import (
     _label_tensor,
 )
 from flowml.frame import FlowFrame
-from flowml.classification.summary import ClassificationTrainingSummary
+from flowml.classification.summary import (BinaryClassificationTrainingSummary,
+                                           ClassificationTrainingSummary)
 from flowml.linalg import DenseMatrix, DenseVector
 from flowml.optim import lbfgs, owlqn
 from flowml.param import (
@@ -412,14 +413,26 @@ class LogisticRegressionModel(ProbabilisticClassificationModel, _LogisticRegress
         return _load(path)
 
 
-class LogisticRegressionTrainingSummary(ClassificationTrainingSummary):
+class LogisticRegressionTrainingSummary(BinaryClassificationTrainingSummary):
     """Mirror of Spark's training summary: ``objectiveHistory`` /
-    ``totalIterations`` plus the shared training-set metrics
-    (accuracy, weighted*/byLabel, predictions)."""
+    ``totalIterations`` plus the shared training-set metrics — and, for
+    BINARY models, the threshold curves (roc/pr/…ByThreshold) of
+    BinaryLogisticRegressionTrainingSummary. Curve accessors raise on
+    multiclass models, like Spark's summary split."""
 
     def __init__(self, objectiveHistory: List[float], totalIterations: int,
                  model=None, dataset=None, labelCol: str = "label",
-                 predictionCol: str = "prediction"):
-        super().__init__(model, dataset, labelCol, predictionCol)
+                 predictionCol: str = "prediction",
+                 probabilityCol: str = "probability"):
+        super().__init__(model, dataset, labelCol, predictionCol,
+                         probabilityCol)
         self.objectiveHistory = objectiveHistory
         self.totalIterations = totalIterations
+
+    def _curves(self):
+        if getattr(self._model, "numClasses", 2) != 2:
+            raise RuntimeError(
+                "binary-only metric: this LogisticRegressionModel is "
+                "multiclass (Spark exposes these on "
+                "BinaryLogisticRegressionTrainingSummary only)")
+        return super()._curves()

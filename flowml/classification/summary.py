@@ -92,6 +92,103 @@ class ClassificationTrainingSummary:
         return [self._eval("falsePositiveRateByLabel", metricLabel=l) for l in self.labels]
 
 
+class BinaryClassificationTrainingSummary(ClassificationTrainingSummary):
+    """Binary summary with threshold curves
+    (spark BinaryLogisticRegressionTrainingSummary): ROC / PR curves and
+    precision/recall/F-measure by threshold, computed from the probability
+    column over the training predictions."""
+
+    def __init__(self, model, dataset, labelCol: str, predictionCol: str,
+                 probabilityCol: str = "probability"):
+        super().__init__(model, dataset, labelCol, predictionCol)
+        self.probabilityCol = probabilityCol
+        self._curve_cache = None
+
+    def _curves(self):
+        if self._curve_cache is not None:
+            return self._curve_cache
+        import torch
+
+        pred = self.predictions
+        prob = pred.col(self.probabilityCol)
+        score = prob[:, 1] if prob.dim() == 2 else prob.reshape(-1)
+        y = pred.col(self.labelCol).reshape(-1).to(torch.float64)
+        score = score.to(torch.float64)
+        order = torch.argsort(score, descending=True)
+        y = y[order]
+        s_sorted = score[order]
+        tp = torch.cumsum(y, dim=0)
+        fp = torch.cumsum(1 - y, dim=0)
+        # one curve point per distinct score (last index of each run)
+        last = torch.ones_like(y, dtype=torch.bool)
+        last[:-1] = s_sorted[:-1] != s_sorted[1:]
+        thr = s_sorted[last]
+        tp = tp[last]
+        fp = fp[last]
+        P = float(y.sum())
+        Nn = float((1 - y).sum())
+        self._curve_cache = (thr, tp, fp, P, Nn)
+        return self._curve_cache
+
+    @property
+    def roc(self):
+        """FlowFrame{FPR, TPR} anchored at (0,0) and (1,1)."""
+        import torch
+
+        from flowml.frame import FlowFrame
+
+        thr, tp, fp, P, Nn = self._curves()
+        tpr = torch.cat([tp.new_zeros(1), tp / max(P, 1e-300), tp.new_ones(1)])
+        fpr = torch.cat([fp.new_zeros(1), fp / max(Nn, 1e-300), fp.new_ones(1)])
+        return FlowFrame({"FPR": fpr, "TPR": tpr})
+
+    @property
+    def areaUnderROC(self) -> float:
+        import torch
+
+        r = self.roc
+        return float(torch.trapz(r.col("TPR"), r.col("FPR")))
+
+    @property
+    def pr(self):
+        """FlowFrame{recall, precision} with the (0, p1) start point."""
+        import torch
+
+        from flowml.frame import FlowFrame
+
+        thr, tp, fp, P, Nn = self._curves()
+        prec = tp / (tp + fp)
+        rec = tp / max(P, 1e-300)
+        return FlowFrame({
+            "recall": torch.cat([rec.new_zeros(1), rec]),
+            "precision": torch.cat([prec[:1], prec]),
+        })
+
+    def _by_threshold(self, values):
+        from flowml.frame import FlowFrame
+
+        thr, *_ = self._curves()
+        return FlowFrame({"threshold": thr, "metric": values})
+
+    @property
+    def precisionByThreshold(self):
+        thr, tp, fp, P, Nn = self._curves()
+        return self._by_threshold(tp / (tp + fp))
+
+    @property
+    def recallByThreshold(self):
+        thr, tp, fp, P, Nn = self._curves()
+        return self._by_threshold(tp / max(P, 1e-300))
+
+    def fMeasureByThreshold(self, beta: float = 1.0):
+        thr, tp, fp, P, Nn = self._curves()
+        prec = tp / (tp + fp)
+        rec = tp / max(P, 1e-300)
+        b2 = beta * beta
+        f = (1 + b2) * prec * rec / (b2 * prec + rec).clamp_min(1e-300)
+        return self._by_threshold(f)
+
+
 class _HasTrainingSummary:
     """Mixin giving models Spark's summary/hasSummary contract."""
 

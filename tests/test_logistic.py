@@ -141,3 +141,40 @@ def test_fit_intercept_false():
     df, _, _ = _binary_df()
     m = LogisticRegression(maxIter=50, fitIntercept=False).fit(df)
     assert m.intercept == 0.0
+
+
+class TestBinarySummaryCurves:
+    def test_roc_pr_threshold_curves(self):
+        from flowml.evaluation import BinaryClassificationEvaluator
+
+        r = np.random.default_rng(0)
+        X = r.normal(size=(800, 4)).astype(np.float32)
+        y = (X[:, 0] - X[:, 2] + 0.4 * r.normal(size=800) > 0).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y})
+        m = LogisticRegression(maxIter=60).fit(df)
+        s = m.summary
+        # AUC agrees with the evaluator on the same score
+        ev = BinaryClassificationEvaluator(
+            rawPredictionCol="probability").evaluate(m.transform(df))
+        assert abs(s.areaUnderROC - ev) < 1e-9
+        roc = s.roc
+        assert float(roc["FPR"][0]) == 0.0 and float(roc["TPR"][-1]) == 1.0
+        assert (roc["TPR"][1:] >= roc["TPR"][:-1]).all()  # monotone
+        pr = s.pr
+        assert pr.columns == ["recall", "precision"]
+        f1 = s.fMeasureByThreshold()
+        assert 0.7 < float(f1["metric"].max()) <= 1.0
+        p_at = s.precisionByThreshold
+        r_at = s.recallByThreshold
+        assert len(p_at) == len(r_at) == len(f1)
+
+    def test_multiclass_curves_raise(self):
+        r = np.random.default_rng(1)
+        X = r.normal(size=(300, 4)).astype(np.float32)
+        y = r.integers(0, 3, 300).astype(np.float64)
+        m = LogisticRegression(maxIter=5).fit(
+            FlowFrame({"features": torch.from_numpy(X), "label": y}))
+        with pytest.raises(RuntimeError, match="binary-only"):
+            _ = m.summary.areaUnderROC
+        # shared multiclass metrics still work
+        assert 0.0 <= m.summary.accuracy <= 1.0
