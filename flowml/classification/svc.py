@@ -11,6 +11,8 @@ from __future__ import annotations
 import torch
 
 from flowml import dist
+from flowml.classification.summary import (BinaryClassificationTrainingSummary,
+                                           _HasTrainingSummary)
 from flowml.classification.base import (
     Classifier,
     ClassificationModel,
@@ -113,10 +115,16 @@ class LinearSVC(_LinearSVCParams, Classifier):
                                numFeatures=F)
         self._copyValues(model)
         model._set_parent(self)
+        # Spark LinearSVCTrainingSummary: binary curves scored on the raw
+        # margin column (LinearSVC has no probability column)
+        model._summary = BinaryClassificationTrainingSummary(
+            model, dataset, self.getLabelCol(),
+            self.getOrDefault(self.predictionCol),
+            probabilityCol=self.getOrDefault(self.rawPredictionCol))
         return model
 
 
-class LinearSVCModel(_LinearSVCParams, ClassificationModel):
+class LinearSVCModel(_LinearSVCParams, _HasTrainingSummary, ClassificationModel):
     def __init__(self, coefficients=None, intercept: float = 0.0, numFeatures: int = 0):
         super().__init__()
         if coefficients is None:

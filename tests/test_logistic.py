@@ -178,3 +178,17 @@ class TestBinarySummaryCurves:
             _ = m.summary.areaUnderROC
         # shared multiclass metrics still work
         assert 0.0 <= m.summary.accuracy <= 1.0
+
+
+def test_linear_svc_summary_curves():
+    from flowml.classification import LinearSVC
+
+    r = np.random.default_rng(2)
+    X = r.normal(size=(500, 4)).astype(np.float32)
+    y = (X[:, 0] - X[:, 1] + 0.3 * r.normal(size=500) > 0).astype(np.float64)
+    df = FlowFrame({"features": torch.from_numpy(X), "label": y})
+    m = LinearSVC(maxIter=40).fit(df)
+    s = m.summary
+    assert s.areaUnderROC > 0.9
+    assert 0.8 < s.accuracy <= 1.0
+    assert len(s.precisionByThreshold) > 10
