@@ -274,6 +274,15 @@ class RandomForestClassificationModel(_TreeEnsembleParams, _HasTrainingSummary, 
         return [1.0] * len(self._trees)
 
     @property
+    def toDebugString(self) -> str:
+        from flowml.classification.tree import ensemble_debug_string
+
+        return ensemble_debug_string(
+            f"RandomForestClassificationModel: uid={self.uid}, "
+            f"numTrees={len(self._trees)}, numClasses={self.numClasses}, "
+            f"numFeatures={self.numFeatures}", self._trees)
+
+    @property
     def featureImportances(self) -> DenseVector:
         return DenseVector(self._ensemble.feature_importances(self.numFeatures))
 

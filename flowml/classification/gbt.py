@@ -342,6 +342,36 @@ class GBTClassificationModel(_GBTParams, _HasTrainingSummary, ProbabilisticClass
     def _margin(self, X: torch.Tensor) -> torch.Tensor:
         return self._ensemble.predict_raw(X)[:, 0]
 
+    def evaluateEachIteration(self, dataset, loss: str = "logistic"):
+        """Per-boosting-stage training loss (spark
+        GBTClassificationModel.evaluateEachIteration): cumulative margin
+        after each tree, mean logistic loss log(1 + exp(-2 y' F_m))."""
+        if loss != "logistic":
+            raise ValueError("only loss='logistic' is supported")
+        from flowml.classification.base import _features_tensor, _label_tensor
+        from flowml.classification.tree import TreeEnsemble
+
+        X = _features_tensor(dataset, self.getFeaturesCol()).to(torch.float32)
+        y = _label_tensor(dataset, self.getLabelCol()).to(torch.float32)
+        ysign = (2.0 * y - 1.0).to(X.device)
+        margin = torch.zeros(len(y), dtype=torch.float32, device=X.device)
+        out = []
+        for tree, w in zip(self._trees, self._tree_weights):
+            single = TreeEnsemble([tree], 2, mode="scalar").to(X.device)
+            margin = margin + w * single.predict_raw(X)[:, 0]
+            out.append(float(torch.log1p(
+                torch.exp(-2.0 * ysign * margin)).mean()))
+        return out
+
+    @property
+    def toDebugString(self) -> str:
+        from flowml.classification.tree import ensemble_debug_string
+
+        return ensemble_debug_string(
+            f"GBTClassificationModel: uid={self.uid}, "
+            f"numTrees={len(self._trees)}, numFeatures={self.numFeatures}",
+            self._trees, self._tree_weights)
+
     def predictRaw(self, features: torch.Tensor) -> torch.Tensor:
         m = self._margin(features)
         return torch.stack([-m, m], dim=1)

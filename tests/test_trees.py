@@ -372,3 +372,40 @@ class TestWeightCol:
         pred = m.transform(df)["prediction"]
         rec1 = float(((pred == 1) & (y == 1)).sum() / (y == 1).sum())
         assert rec1 > 0.9
+
+
+class TestDebugStringsAndIterEval:
+    def test_to_debug_string(self):
+        r = np.random.default_rng(0)
+        X = r.normal(size=(800, 4)).astype(np.float32)
+        y = (X[:, 0] > 0).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y})
+        m = RandomForestClassifier(numTrees=2, maxDepth=2, seed=0).fit(df)
+        dbg = m.toDebugString
+        assert dbg.startswith("RandomForestClassificationModel")
+        assert "If (feature" in dbg and "Predict:" in dbg
+        assert dbg.count("Tree ") == 2
+
+    def test_categorical_debug_string(self):
+        r = np.random.default_rng(1)
+        cat = r.integers(0, 4, 800)
+        y = np.isin(cat, [0, 2]).astype(np.float64)
+        X = np.column_stack([cat, r.normal(size=800)]).astype(np.float32)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y}).withMeta(
+            "features", {"categorical": {0: 4}})
+        m = RandomForestClassifier(numTrees=1, maxDepth=1, seed=0,
+                                   bootstrap=False).fit(df)
+        dbg = m.toDebugString
+        assert "in {" in dbg  # categorical split renders as set membership
+
+    def test_gbt_evaluate_each_iteration(self):
+        r = np.random.default_rng(2)
+        X = r.normal(size=(2000, 4)).astype(np.float32)
+        y = (X[:, 0] > 0).astype(np.float64)
+        df = FlowFrame({"features": torch.from_numpy(X), "label": y})
+        m = GBTClassifier(maxIter=6, maxDepth=2, seed=0).fit(df)
+        losses = m.evaluateEachIteration(df)
+        assert len(losses) == 6
+        assert losses[-1] < losses[0]  # boosting reduces training loss
+        with pytest.raises(ValueError):
+            m.evaluateEachIteration(df, loss="squared")
