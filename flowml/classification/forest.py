@@ -319,6 +319,16 @@ class DecisionTreeClassificationModel(_TreeEnsembleParams, _HasTrainingSummary, 
         self._ensemble = TreeEnsemble([self._tree], numClasses, mode="prob_sum")
 
     @property
+    def toDebugString(self) -> str:
+        from flowml.classification.tree import tree_debug_string
+
+        return (f"DecisionTreeClassificationModel: uid={self.uid}, "
+                f"depth={self.depth}, numNodes={self.numNodes}, "
+                f"numClasses={self.numClasses}, "
+                f"numFeatures={self.numFeatures}\n"
+                + tree_debug_string(self._tree))
+
+    @property
     def depth(self) -> int:
         return self._tree.depth()
 
