@@ -158,6 +158,15 @@ class RandomForestRegressionModel(_TreeEnsembleParams, RegressionModel, _Regress
         return len(self._trees)
 
     @property
+    def toDebugString(self) -> str:
+        from flowml.classification.tree import ensemble_debug_string
+
+        return ensemble_debug_string(
+            f"RandomForestRegressionModel: uid={self.uid}, "
+            f"numTrees={len(self._trees)}, numFeatures={self.numFeatures}",
+            self._trees)
+
+    @property
     def featureImportances(self):
         from flowml.linalg import DenseVector
 
@@ -197,6 +206,15 @@ class DecisionTreeRegressionModel(_TreeEnsembleParams, RegressionModel, _Regress
     @property
     def numNodes(self) -> int:
         return self._trees[0].n_nodes if self._trees else 0
+
+    @property
+    def toDebugString(self) -> str:
+        from flowml.classification.tree import tree_debug_string
+
+        return (f"DecisionTreeRegressionModel: uid={self.uid}, "
+                f"depth={self.depth}, numNodes={self.numNodes}, "
+                f"numFeatures={self.numFeatures}\n"
+                + (tree_debug_string(self._trees[0]) if self._trees else ""))
 
     @property
     def featureImportances(self):
@@ -303,6 +321,15 @@ class GBTRegressionModel(_TreeEnsembleParams, RegressionModel, _RegressorIOMixin
 
     def getNumTrees(self) -> int:
         return len(self._trees)
+
+    @property
+    def toDebugString(self) -> str:
+        from flowml.classification.tree import ensemble_debug_string
+
+        return ensemble_debug_string(
+            f"GBTRegressionModel: uid={self.uid}, "
+            f"numTrees={len(self._trees)}, numFeatures={self.numFeatures}",
+            self._trees, self.treeWeights)
 
     @property
     def featureImportances(self):

@@ -430,3 +430,23 @@ class TestLinearRegressionInferenceStats:
         m = LinearRegression(regParam=0.1).fit(df)
         with pytest.raises(RuntimeError, match="Std. Error"):
             _ = m.summary.pValues
+
+
+def test_regression_models_debug_strings():
+    from flowml.regression import (DecisionTreeRegressor, GBTRegressor,
+                                   RandomForestRegressor)
+
+    r = np.random.default_rng(0)
+    X = r.normal(size=(400, 3)).astype(np.float32)
+    y = (X[:, 0] * 2).astype(np.float64)
+    df = FlowFrame({"features": torch.from_numpy(X),
+                    "label": torch.from_numpy(y)})
+    for est, head in ((RandomForestRegressor(numTrees=2, maxDepth=2, seed=0),
+                       "RandomForestRegressionModel"),
+                      (DecisionTreeRegressor(maxDepth=2, seed=0),
+                       "DecisionTreeRegressionModel"),
+                      (GBTRegressor(maxIter=2, maxDepth=2, seed=0),
+                       "GBTRegressionModel")):
+        dbg = est.fit(df).toDebugString
+        assert dbg.startswith(head)
+        assert "Predict:" in dbg
