@@ -153,6 +153,9 @@ class KMeans(_KMeansParams, Estimator):
                             numIter=it + 1)
         self._copyValues(model)
         model._set_parent(self)
+        model._summary = ClusteringSummary(
+            model, dataset, self.getOrDefault(self.predictionCol), k, it + 1,
+            trainingCost=cost)
         return model
 
 
@@ -160,9 +163,19 @@ class KMeansModel(_KMeansParams, Model):
     def __init__(self, centers: Optional[torch.Tensor] = None,
                  trainingCost: float = 0.0, numIter: int = 0):
         super().__init__()
+        self._summary = None
         self._centers = centers if centers is not None else torch.zeros(0, 0, dtype=torch.float64)
         self.trainingCost = trainingCost
         self.numIter = numIter
+
+    @property
+    def summary(self) -> "ClusteringSummary":
+        if self._summary is None:
+            raise RuntimeError("No training summary available (absent after load)")
+        return self._summary
+
+    def hasSummary(self) -> bool:
+        return self._summary is not None
 
     def clusterCenters(self) -> List:
         return [c.numpy() for c in self._centers]
@@ -250,6 +263,38 @@ class ClusteringEvaluator(HasFeaturesCol, HasPredictionCol, Params):
             torch.zeros_like(own),
         )
         return float(s.mean())
+
+
+class ClusteringSummary:
+    """spark ClusteringSummary subset: lazy predictions + clusterSizes +
+    numIter (KMeans adds trainingCost, GaussianMixture adds logLikelihood)."""
+
+    def __init__(self, model, dataset, predictionCol: str, k: int,
+                 numIter: int, **extra):
+        self._model = model
+        self._dataset = dataset
+        self.predictionCol = predictionCol
+        self.k = int(k)
+        self.numIter = int(numIter)
+        for kk, v in extra.items():
+            setattr(self, kk, v)
+        self._pred = None
+
+    @property
+    def predictions(self):
+        if self._pred is None:
+            self._pred = self._model.transform(self._dataset)
+        return self._pred
+
+    @property
+    def clusterSizes(self):
+        pred = self.predictions.col(self.predictionCol).to(torch.int64)
+        sizes = torch.bincount(pred, minlength=self.k)
+        from flowml import dist as _d
+
+        sizes = sizes.to(torch.float64)
+        _d.all_reduce_(sizes)
+        return [int(v) for v in sizes]
 
 
 # ---------------------------------------------------------------------------
@@ -388,6 +433,9 @@ class BisectingKMeans(_BisectingKMeansParams, Estimator):
                                      trainingCost=float(cost_t))
         self._copyValues(model)
         model._set_parent(self)
+        model._summary = ClusteringSummary(
+            model, dataset, self.getOrDefault(self.predictionCol),
+            len(centers), 0, trainingCost=float(cost_t))
         return model
 
 
@@ -395,9 +443,19 @@ class BisectingKMeansModel(_BisectingKMeansParams, Model):
     def __init__(self, centers: Optional[torch.Tensor] = None,
                  trainingCost: float = 0.0):
         super().__init__()
+        self._summary = None
         self._centers = centers if centers is not None else torch.zeros(
             0, 0, dtype=torch.float64)
         self.trainingCost = float(trainingCost)
+
+    @property
+    def summary(self) -> "ClusteringSummary":
+        if self._summary is None:
+            raise RuntimeError("No training summary available (absent after load)")
+        return self._summary
+
+    def hasSummary(self) -> bool:
+        return self._summary is not None
 
     def clusterCenters(self):
         return [c.numpy() for c in self._centers]
@@ -540,6 +598,10 @@ class GaussianMixture(_GaussianMixtureParams, Estimator):
                                      numIter=it + 1)
         self._copyValues(model)
         model._set_parent(self)
+        model._summary = ClusteringSummary(
+            model, dataset, self.getOrDefault(self.predictionCol), k, it + 1,
+            logLikelihood=ll,
+            probabilityCol=self.getOrDefault(self.probabilityCol))
         return model
 
 
@@ -549,11 +611,21 @@ class GaussianMixtureModel(_GaussianMixtureParams, Model):
                  covs: Optional[torch.Tensor] = None,
                  logLikelihood: float = 0.0, numIter: int = 0):
         super().__init__()
+        self._summary = None
         self._weights = weights if weights is not None else torch.zeros(0, dtype=torch.float64)
         self._mus = mus if mus is not None else torch.zeros(0, 0, dtype=torch.float64)
         self._covs = covs if covs is not None else torch.zeros(0, 0, 0, dtype=torch.float64)
         self.logLikelihood = float(logLikelihood)
         self.numIter = int(numIter)
+
+    @property
+    def summary(self) -> "ClusteringSummary":
+        if self._summary is None:
+            raise RuntimeError("No training summary available (absent after load)")
+        return self._summary
+
+    def hasSummary(self) -> bool:
+        return self._summary is not None
 
     @property
     def weights(self) -> List[float]:
@@ -604,4 +676,5 @@ class GaussianMixtureModel(_GaussianMixtureParams, Model):
 
 
 __all__ = ["KMeans", "KMeansModel", "BisectingKMeans", "BisectingKMeansModel",
-           "GaussianMixture", "GaussianMixtureModel", "ClusteringEvaluator"]
+           "GaussianMixture", "GaussianMixtureModel", "ClusteringSummary",
+           "ClusteringEvaluator"]

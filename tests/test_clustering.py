@@ -172,3 +172,28 @@ class TestBisectingKMeans:
         assert m2.trainingCost == m.trainingCost
         assert torch.equal(m.transform(df)["prediction"],
                            m2.transform(df)["prediction"])
+
+
+def test_clustering_summaries():
+    from flowml.clustering import BisectingKMeans, GaussianMixture, KMeans
+
+    r = np.random.default_rng(3)
+    X = np.vstack([r.normal(loc=m, scale=0.4, size=(100, 2))
+                   for m in ([-3, 0], [3, 0], [0, 3])]).astype(np.float32)
+    df = FlowFrame({"features": torch.from_numpy(X)})
+    km = KMeans(k=3, seed=1).fit(df)
+    s = km.summary
+    assert sum(s.clusterSizes) == 300 and len(s.clusterSizes) == 3
+    assert s.trainingCost > 0 and s.numIter >= 1
+    gm = GaussianMixture(k=3, seed=1).fit(df)
+    assert np.isfinite(gm.summary.logLikelihood)
+    assert gm.summary.probabilityCol == "probability"
+    bk = BisectingKMeans(k=3, seed=1).fit(df)
+    assert sum(bk.summary.clusterSizes) == 300
+    # absent after load
+    import tempfile, os
+    d = tempfile.mkdtemp()
+    km.save(os.path.join(d, "km"))
+    from flowml.clustering import KMeansModel
+    km2 = KMeansModel.load(os.path.join(d, "km"))
+    assert not km2.hasSummary()
