@@ -1224,6 +1224,12 @@ class _ImputerParams(HasInputCol, HasInputCols, HasOutputCol, HasOutputCols):
         "placeholder for missing values (default NaN)",
         TypeConverters.toFloat,
     )
+    relativeError: Param[float] = Param(
+        "undefined", "relativeError",
+        "relative error for the approximate quantile algorithm (API parity; "
+        "the median here is computed exactly, so this is accepted and "
+        "unused)", TypeConverters.toFloat,
+    )
 
     def _in_out_cols(self):
         if self.isSet(self.inputCols):
@@ -1236,11 +1242,12 @@ class Imputer(_ImputerParams, Estimator):
 
     def __init__(self, *, inputCol: Optional[str] = None, outputCol: Optional[str] = None,
                  inputCols: Optional[list] = None, outputCols: Optional[list] = None,
-                 strategy: Optional[str] = None, missingValue: Optional[float] = None):
+                 strategy: Optional[str] = None, missingValue: Optional[float] = None,
+                 relativeError: Optional[float] = None):
         super().__init__()
-        self._setDefault(strategy="mean", missingValue=float("nan"))
+        self._setDefault(strategy="mean", missingValue=float("nan", relativeError=0.001))
         self._set(inputCol=inputCol, outputCol=outputCol, inputCols=inputCols,
-                  outputCols=outputCols, strategy=strategy, missingValue=missingValue)
+                  outputCols=outputCols, strategy=strategy, missingValue=missingValue, relativeError=relativeError)
 
     def _fit(self, dataset: FlowFrame) -> "ImputerModel":
         from flowml import dist
@@ -1338,13 +1345,21 @@ class QuantileDiscretizer(HasInputCol, HasOutputCol, HasHandleInvalid, Estimator
         "Number of buckets (quantiles) into which data points are grouped (>= 2).",
         TypeConverters.toInt,
     )
+    relativeError: Param[float] = Param(
+        "undefined", "relativeError",
+        "relative error for the approximate quantile algorithm (API parity; "
+        "quantiles here are exact — torch.quantile locally, type-1 bisection "
+        "distributed — so this is accepted and unused)",
+        TypeConverters.toFloat,
+    )
 
     def __init__(self, *, inputCol: Optional[str] = None, outputCol: Optional[str] = None,
-                 numBuckets: Optional[int] = None, handleInvalid: Optional[str] = None):
+                 numBuckets: Optional[int] = None, handleInvalid: Optional[str] = None,
+                 relativeError: Optional[float] = None):
         super().__init__()
-        self._setDefault(numBuckets=2, handleInvalid="error")
+        self._setDefault(numBuckets=2, handleInvalid="error", relativeError=0.001)
         self._set(inputCol=inputCol, outputCol=outputCol, numBuckets=numBuckets,
-                  handleInvalid=handleInvalid)
+                  handleInvalid=handleInvalid, relativeError=relativeError)
 
     def _fit(self, dataset: FlowFrame) -> "Bucketizer":
         x = dataset.col(self.getInputCol())
