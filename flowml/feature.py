@@ -1245,7 +1245,8 @@ class Imputer(_ImputerParams, Estimator):
                  strategy: Optional[str] = None, missingValue: Optional[float] = None,
                  relativeError: Optional[float] = None):
         super().__init__()
-        self._setDefault(strategy="mean", missingValue=float("nan", relativeError=0.001))
+        self._setDefault(strategy="mean", missingValue=float("nan"),
+                         relativeError=0.001)
         self._set(inputCol=inputCol, outputCol=outputCol, inputCols=inputCols,
                   outputCols=outputCols, strategy=strategy, missingValue=missingValue, relativeError=relativeError)
 
