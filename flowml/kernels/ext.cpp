@@ -46,6 +46,8 @@ void launch_gemm_bias_act_bf16(const void*, const void*, const float*, void*,
 void launch_gemm_bias_act_f32(const float*, const float*, const float*, float*,
                               long long, int, int, int, void*);
 void launch_act_backward_bf16(const void*, void*, long long, int, void*);
+void launch_gemm_dgrad_act_bf16(const void*, const void*, const void*, void*,
+                                long long, int, int, int, void*);
 void launch_wgrad_bias_bf16(const void*, const void*, float*, long long, int,
                             int, int, long long, int, void*);
 void wgrad_bias_plan(long long, int*, long long*, int*);
@@ -232,6 +234,26 @@ torch::Tensor tree_predict(torch::Tensor X, torch::Tensor feature,
   return out;
 }
 
+torch::Tensor gemm_dgrad_act(torch::Tensor D, torch::Tensor Wt,
+                             torch::Tensor H, int64_t act) {
+  CHK(D);
+  CHK(Wt);
+  CHK(H);
+  TORCH_CHECK(D.scalar_type() == torch::kBFloat16, "D must be bf16");
+  const long long N = D.size(0);
+  const int Kp = D.size(1);
+  const int M = Wt.size(0);
+  TORCH_CHECK(Wt.size(1) == Kp, "Wt must be M x Kp");
+  TORCH_CHECK(Kp % 32 == 0, "Kp must be a multiple of 32");
+  TORCH_CHECK(M <= 128, "M <= 128");
+  TORCH_CHECK(H.size(0) == N && H.size(1) == M, "H must be N x M");
+  auto out = torch::empty({N, M}, D.options());
+  launch_gemm_dgrad_act_bf16(D.data_ptr(), Wt.data_ptr(), H.data_ptr(),
+                             out.data_ptr(), N, M, Kp, (int)act,
+                             cur_stream());
+  return out;
+}
+
 std::tuple<torch::Tensor, torch::Tensor> logreg_delta(torch::Tensor X,
                                                       torch::Tensor y,
                                                       torch::Tensor w,
@@ -394,6 +416,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("node_advance", &node_advance, "K4b node-id advance");
   m.def("split_scan", &split_scan, "K3 split-gain scan (packed)");
   m.def("split_scan_reg", &split_scan_reg, "K3b variance scan (2-wide + node s2)");
+  m.def("gemm_dgrad_act", &gemm_dgrad_act);
   m.def("tree_predict", &tree_predict, "K8 ensemble inference");
   m.def("logreg_delta", &logreg_delta, "K1 fused logistic delta");
   m.def("wgrad_f32", &wgrad_f32, "split-K f32 wgrad + bias grad (LR)");

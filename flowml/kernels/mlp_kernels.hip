@@ -92,6 +92,75 @@ extern "C" __global__ __launch_bounds__(256) void flowml_gemm_bias_act_bf16(
   }
 }
 
+// dgrad with fused activation-backward: out[N,M] = (D[N,Kp] @ Wt[M,Kp]^T)
+// * act'(H[N,M]).  Wt is the layer weight TRANSPOSED once per step (tiny);
+// H is the layer's forward activation. Replaces the rocBLAS dgrad GEMM +
+// the separate act_backward pass (one write of delta instead of
+// write+read+write — the two act_backward launches were 0.28 ms of the
+// 1.47 ms step).
+extern "C" __global__ __launch_bounds__(256) void flowml_gemm_dgrad_act_bf16(
+    const __bf16* __restrict__ D, const __bf16* __restrict__ Wt,
+    const __bf16* __restrict__ H, __bf16* __restrict__ out, long long N,
+    int M, int Kp, int act) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long long row_base = (long long)blockIdx.x * 64 + wave * 16;
+  if (row_base >= N) return;
+  const int i_in_tile = lane & 15;
+  const int kgrp = lane >> 4;
+  const int nfrag = (M + 15) / 16;
+
+  f32x4 acc[MAX_NFRAG];
+#pragma unroll
+  for (int nf = 0; nf < MAX_NFRAG; ++nf) acc[nf] = {0.f, 0.f, 0.f, 0.f};
+
+  const long long a_row = row_base + i_in_tile;
+  const bool a_valid = a_row < N;
+  const __bf16* a_base = D + a_row * Kp + kgrp * 8;
+
+  for (int kk = 0; kk < Kp; kk += 32) {
+    bf16x8 a_frag;
+    if (a_valid) {
+      a_frag = *reinterpret_cast<const bf16x8*>(a_base + kk);
+    } else {
+      a_frag = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+    }
+    for (int nf = 0; nf < nfrag; ++nf) {
+      const int n = nf * 16 + i_in_tile;
+      bf16x8 b_frag;
+      if (n < M) {
+        b_frag = *reinterpret_cast<const bf16x8*>(Wt + (long long)n * Kp + kk + kgrp * 8);
+      } else {
+        b_frag = bf16x8{0, 0, 0, 0, 0, 0, 0, 0};
+      }
+      acc[nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc[nf], 0, 0, 0);
+    }
+  }
+
+  const int col_in = lane & 15;
+  const int row_off = (lane >> 4) * 4;
+  for (int nf = 0; nf < nfrag; ++nf) {
+    const int n = nf * 16 + col_in;
+    if (n >= M) continue;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const long long r = row_base + row_off + reg;
+      if (r < N) {
+        const float h = (float)H[r * M + n];
+        float g;
+        if (act == 1) {
+          g = h > 0.f ? acc[nf][reg] : 0.f;            // relu'
+        } else if (act == 2) {
+          g = acc[nf][reg] * h * (1.f - h);            // sigmoid'
+        } else {
+          g = acc[nf][reg];
+        }
+        out[r * M + n] = (__bf16)g;
+      }
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // f32 path: exact f32 MFMA (v_mfma_f32_16x16x4_f32). Kp%4==0.
 // A lane map: A[i=l%16][k=l/16]; B[k=l/16][n=l%16]; D as above.
@@ -186,6 +255,17 @@ extern "C" void launch_gemm_bias_act_f32(const float* X, const float* W,
   const int blocks = (int)((N + 63) / 64);
   hipLaunchKernelGGL(flowml_gemm_bias_act_f32, dim3(blocks), dim3(256), 0,
                      (hipStream_t)stream, X, W, b, out, N, M, Kp, act);
+}
+
+extern "C" void launch_gemm_dgrad_act_bf16(const void* D, const void* Wt,
+                                           const void* H, void* out,
+                                           long long N, int M, int Kp,
+                                           int act, void* stream) {
+  const long long blocks = (N + 63) / 64;
+  hipLaunchKernelGGL(flowml_gemm_dgrad_act_bf16, dim3((unsigned)blocks),
+                     dim3(256), 0, (hipStream_t)stream, (const __bf16*)D,
+                     (const __bf16*)Wt, (const __bf16*)H, (__bf16*)out, N, M,
+                     Kp, act);
 }
 
 extern "C" void launch_act_backward_bf16(const void* h, void* delta,

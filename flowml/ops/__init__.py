@@ -598,13 +598,27 @@ def mlp_grad(
         if i == 0 and in_features is not None and gWs[0].shape[1] != in_features:
             gWs[0] = gWs[0][:, :in_features].contiguous()  # drop X pad columns
         if i > 0:
-            dx = delta @ Ws[i]                # K7 dgrad (plain GEMM: rocBLAS)
             h_i = acts[i]
-            if dx.is_cuda and dx.dtype == torch.bfloat16 and hip_available(dx):
-                delta = get_ext().act_backward_(
-                    h_i.contiguous(), dx.contiguous(), _act_code(activation)
-                )
+            out_i, in_i = Ws[i].shape
+            if (
+                delta.is_cuda and delta.dtype == torch.bfloat16
+                and h_i.dtype == torch.bfloat16 and in_i <= 128
+                and hip_available(delta)
+            ):
+                # fused dgrad+act': one MFMA kernel writes the next delta
+                # directly (was rocBLAS dgrad + a separate act_backward
+                # read-modify-write — ~0.28 ms of the bf16 step)
+                Kp = ((out_i + 31) // 32) * 32
+                d_in = delta
+                Wt = Ws[i].T
+                if Kp != out_i:
+                    d_in = torch.nn.functional.pad(delta, (0, Kp - out_i))
+                    Wt = torch.nn.functional.pad(Wt, (0, Kp - out_i))
+                delta = get_ext().gemm_dgrad_act(
+                    d_in.contiguous(), Wt.contiguous(), h_i.contiguous(),
+                    _act_code(activation))
             else:
+                dx = delta @ Ws[i]            # K7 dgrad (plain GEMM)
                 if activation == "relu":
                     dact = (h_i > 0).to(dx.dtype)
                 else:

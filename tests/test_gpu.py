@@ -956,3 +956,68 @@ class TestFullScaleDeterminism:
                                   t2.thr_raw[t2.feature >= 0])
             assert np.array_equal(t1.stats, t2.stats)
             assert np.array_equal(t1.left, t2.left)
+
+
+class TestFusedDgradAct:
+    @pytest.mark.parametrize("act", ["relu", "sigmoid"])
+    def test_mlp_grad_matches_torch_reference(self, act):
+        """Fused dgrad+act' path == the plain torch composition (fp32 ref)."""
+        from flowml import ops
+
+        r = np.random.default_rng(70)
+        N = 50_000
+        layers = [41, 128, 64, 5]
+        Kp = 64
+        X = torch.zeros(N, Kp, dtype=torch.bfloat16, device=DEV)
+        X[:, :41] = torch.from_numpy(
+            r.normal(size=(N, 41)).astype(np.float32)).to(DEV).to(torch.bfloat16)
+        y = torch.from_numpy(r.integers(0, 5, N)).to(DEV)
+        Ws, bs = [], []
+        dims = [Kp, 128, 64, 5]
+        for i in range(3):
+            Ws.append(torch.from_numpy(
+                (r.normal(size=(dims[i + 1], dims[i])) * 0.2).astype(np.float32)
+            ).to(DEV).to(torch.bfloat16))
+            bs.append(torch.zeros(dims[i + 1], device=DEV))
+        loss_g, gW_g, gB_g = ops.mlp_grad(X, y, Ws, bs, activation=act,
+                                          in_features=41)
+        # fp32 torch reference of the same math
+        Xf = X.to(torch.float32)
+        h = Xf
+        acts = [Xf]
+        for i in range(3):
+            h = h @ Ws[i].to(torch.float32).T + bs[i]
+            if i < 2:
+                h = torch.relu(h) if act == "relu" else torch.sigmoid(h)
+            acts.append(h)
+        p = torch.softmax(h, dim=1)
+        delta = p.clone()
+        delta[torch.arange(N, device=DEV), y] -= 1.0
+        gW_ref = []
+        for i in range(2, -1, -1):
+            gW_ref.append(delta.T @ acts[i])
+            if i > 0:
+                dx = delta @ Ws[i].to(torch.float32)
+                hi = acts[i]
+                dact = (hi > 0).float() if act == "relu" else hi * (1 - hi)
+                delta = dx * dact
+        gW_ref = gW_ref[::-1]
+        for i in range(3):
+            g = gW_g[i].to(torch.float32)
+            ref = gW_ref[i][:, :g.shape[1]]
+            denom = ref.abs().mean().clamp_min(1e-3)
+            assert (g - ref).abs().mean() / denom < 0.05, (i, act)
+
+    def test_deterministic(self):
+        from flowml import ops
+
+        r = np.random.default_rng(71)
+        D = torch.from_numpy(r.normal(size=(30_000, 64)).astype(np.float32)
+                             ).to(DEV).to(torch.bfloat16)
+        Wt = torch.from_numpy(r.normal(size=(128, 64)).astype(np.float32)
+                              ).to(DEV).to(torch.bfloat16)
+        H = torch.from_numpy(r.normal(size=(30_000, 128)).astype(np.float32)
+                             ).to(DEV).to(torch.bfloat16)
+        a = ops.get_ext().gemm_dgrad_act(D, Wt, H, 1)
+        b = ops.get_ext().gemm_dgrad_act(D, Wt, H, 1)
+        assert torch.equal(a, b)
