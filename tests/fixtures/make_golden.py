@@ -206,6 +206,28 @@ def main():
     write_data(p, [{"std": dense_vec([1.0, 2.0, 0.5]),
                     "mean": dense_vec([-1.0, 0.25, 3.0])}])
 
+    # ---- GeneralizedLinearRegressionModel --------------------------------
+    p = os.path.join(OUT, "glr")
+    write_meta(p, "org.apache.spark.ml.regression.GeneralizedLinearRegressionModel",
+               "glm_4455667788990011",
+               {"family": "poisson", "link": "log", "maxIter": 10},
+               {"family": "gaussian", "maxIter": 25, "tol": 1e-06})
+    write_data(p, [{"intercept": 0.25,
+                    "coefficients": dense_vec([0.5, -0.125, 2.0])}])
+
+    # ---- GaussianMixtureModel --------------------------------------------
+    p = os.path.join(OUT, "gmm")
+    write_meta(p, "org.apache.spark.ml.clustering.GaussianMixtureModel",
+               "GaussianMixture_aa11bb22cc33dd44",
+               {"k": 2, "seed": 11},
+               {"k": 2, "maxIter": 100, "tol": 0.01})
+    write_data(p, [{
+        "weights": [0.25, 0.75],
+        "mus": [dense_vec([0.0, 1.0]), dense_vec([4.0, -2.0])],
+        "sigmas": [dense_mat([[1.0, 0.1], [0.1, 2.0]]),
+                   dense_mat([[0.5, 0.0], [0.0, 0.25]])],
+    }])
+
     print(f"golden fixtures written under {OUT}")
 
 

@@ -125,3 +125,27 @@ class TestGoldenLoad:
         assert np.array_equal(sc.std.numpy(), [1.0, 2.0, 0.5])
         assert np.array_equal(sc.mean.numpy(), [-1.0, 0.25, 3.0])
         assert sc.getWithMean() is True  # paramMap overrode the default
+
+    def test_glr(self):
+        m = load(_p("glr"))
+        assert type(m).__name__ == "GeneralizedLinearRegressionModel"
+        assert m.getOrDefault(m.family) == "poisson"
+        assert np.array_equal(m._coef.numpy(), [0.5, -0.125, 2.0])
+        assert m.intercept == 0.25
+        from flowml.frame import FlowFrame
+        X = torch.tensor([[1.0, 0.0, 0.0]], dtype=torch.float32)
+        out = m.transform(FlowFrame({"features": X}))["prediction"]
+        assert float(out[0]) == pytest.approx(np.exp(0.75), rel=1e-6)
+
+    def test_gmm(self):
+        m = load(_p("gmm"))
+        assert type(m).__name__ == "GaussianMixtureModel"
+        assert m.weights == [0.25, 0.75]
+        mus = m._mus.numpy()
+        assert np.array_equal(mus, [[0.0, 1.0], [4.0, -2.0]])
+        covs = m._covs.numpy()
+        assert np.array_equal(covs[0], [[1.0, 0.1], [0.1, 2.0]])
+        from flowml.frame import FlowFrame
+        X = torch.tensor([[4.0, -2.0]], dtype=torch.float32)
+        prob = m.transform(FlowFrame({"features": X}))["probability"]
+        assert float(prob[0, 1]) > 0.99  # at the heavy component's mean
