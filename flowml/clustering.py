@@ -177,6 +177,13 @@ class KMeansModel(_KMeansParams, Model):
     def hasSummary(self) -> bool:
         return self._summary is not None
 
+    def predict(self, features) -> int:
+        """Cluster index for a single vector (spark KMeansModel.predict)."""
+        v = torch.as_tensor(
+            features.toArray() if hasattr(features, "toArray") else features,
+            dtype=torch.float64).reshape(1, -1)
+        return int(_dist2(v, self._centers.to(v.device)).argmin())
+
     def clusterCenters(self) -> List:
         return [c.numpy() for c in self._centers]
 
@@ -457,6 +464,13 @@ class BisectingKMeansModel(_BisectingKMeansParams, Model):
     def hasSummary(self) -> bool:
         return self._summary is not None
 
+    def predict(self, features) -> int:
+        """Cluster index for a single vector (spark KMeansModel.predict)."""
+        v = torch.as_tensor(
+            features.toArray() if hasattr(features, "toArray") else features,
+            dtype=torch.float64).reshape(1, -1)
+        return int(_dist2(v, self._centers.to(v.device)).argmin())
+
     def clusterCenters(self):
         return [c.numpy() for c in self._centers]
 
@@ -644,6 +658,13 @@ class GaussianMixtureModel(_GaussianMixtureParams, Model):
             cov = DenseMatrix(F, F, self._covs[j].T.reshape(-1).numpy())
             out.append((mu, cov))
         return out
+
+    def predict(self, features) -> int:
+        """Most-likely component for a single vector."""
+        v = torch.as_tensor(
+            features.toArray() if hasattr(features, "toArray") else features,
+            dtype=torch.float32).reshape(1, -1)
+        return int(self.predictProbability(v).argmax())
 
     def predictProbability(self, X: torch.Tensor) -> torch.Tensor:
         Xd = X.to(torch.float64)

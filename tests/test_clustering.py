@@ -197,3 +197,17 @@ def test_clustering_summaries():
     from flowml.clustering import KMeansModel
     km2 = KMeansModel.load(os.path.join(d, "km"))
     assert not km2.hasSummary()
+
+
+def test_single_vector_predict():
+    from flowml.clustering import BisectingKMeans, GaussianMixture, KMeans
+    from flowml.linalg import DenseVector
+
+    r = np.random.default_rng(4)
+    X = np.vstack([r.normal(loc=m, scale=0.3, size=(100, 2))
+                   for m in ([-3, 0], [3, 0])]).astype(np.float32)
+    df = FlowFrame({"features": torch.from_numpy(X)})
+    for est in (KMeans(k=2, seed=1), GaussianMixture(k=2, seed=1),
+                BisectingKMeans(k=2, seed=1)):
+        m = est.fit(df)
+        assert m.predict(DenseVector([-3.0, 0.0])) != m.predict([3.0, 0.0])
