@@ -570,10 +570,14 @@ class GaussianMixture(_GaussianMixtureParams, Estimator):
         n_tot = float(mom[0])
         gmean = mom[1:1 + F] / n_tot
         gvar = (mom[1 + F:] / n_tot - gmean * gmean).clamp_min(1e-6)
-        # init means: Spark samples groups of points per component; here a
-        # seeded rank-0 draw of 5 rows per component, broadcast
-        idx = torch.randint(0, N, (k * 5,), generator=g, device=Xd.device)
-        mus = Xd[idx].view(k, 5, F).mean(dim=1).contiguous()
+        # init means from a short KMeans warm start (3 Lloyd iterations,
+        # replicated/seeded — the same all-reduced path as KMeans.fit).
+        # Documented divergence: Spark averages random sample slices, which
+        # collapses to a saddle on symmetric mixtures; k-means init is
+        # sklearn's default and strictly more robust.
+        warm = KMeans(k=k, maxIter=3, seed=seed,
+                      featuresCol=self.getFeaturesCol()).fit(dataset)
+        mus = warm._centers.to(device=Xd.device, dtype=torch.float64).contiguous()
         dist.broadcast_(mus)
         covs = torch.diag_embed(gvar).expand(k, F, F).clone()
         w = torch.full((k,), 1.0 / k, dtype=torch.float64, device=Xd.device)
