@@ -644,3 +644,32 @@ def test_glr_dist_equals_single():
         FlowFrame({"features": torch.from_numpy(X.astype(np.float32)), "label": y}))
     expect = np.concatenate([m._coef.numpy(), [m.intercept]])
     assert np.allclose(out, expect, atol=1e-9)
+
+
+def test_bench_under_torchrun_world2_cpu():
+    """The driver's exact launch shape (torchrun → bench.py) at world=2 on
+    gloo: rendezvous, sharded synth data, all-reduced fit, one JSON line."""
+    import json
+    import socket
+    import subprocess
+    import sys
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.pop("RANK", None), env.pop("WORLD_SIZE", None)
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node=2", "--master-addr=127.0.0.1",
+         f"--master-port={port}", "bench.py", "--gpus", "2", "--model", "rf",
+         "--steps", "2", "--warmup", "1", "--rows", "3000", "--trees", "8"],
+        cwd=repo, env=env, capture_output=True, text=True, timeout=420,
+    )
+    assert r.returncode == 0, r.stdout[-1000:] + r.stderr[-1000:]
+    out = json.loads([l for l in r.stdout.splitlines() if l.startswith("{")][-1])
+    assert out["n_gpus"] == 2
+    assert out["config"]["parallelism"] == "dp2"
+    assert out["config"]["global_rows"] == 6000
