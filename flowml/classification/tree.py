@@ -1237,6 +1237,9 @@ class TreeEnsemble:
         self._right = torch.from_numpy(right)
         self._catmask = torch.from_numpy(catmask)
         self._value = torch.from_numpy(value)
+        self._packed = ops.pack_tree_nodes(
+            self._feature, self._threshold, self._left, self._right,
+            self._catmask)
         self._tree_weights: Optional[torch.Tensor] = None  # GBT stepSize weights
 
     def set_tree_weights(self, w: Sequence[float]) -> None:
@@ -1249,6 +1252,7 @@ class TreeEnsemble:
         self._right = self._right.to(device)
         self._catmask = self._catmask.to(device)
         self._value = self._value.to(device)
+        self._packed = self._packed.to(device)
         if self._tree_weights is not None:
             self._tree_weights = self._tree_weights.to(device)
         return self
@@ -1261,7 +1265,7 @@ class TreeEnsemble:
             value = (value * self._tree_weights.view(-1, 1, 1)).contiguous()
         return ops.tree_predict(
             X.to(torch.float32), self._feature, self._threshold, self._left,
-            self._right, value, self._catmask
+            self._right, value, self._catmask, packed=self._packed
         )
 
     def feature_importances(self, num_features: int) -> np.ndarray:

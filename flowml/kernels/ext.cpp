@@ -28,8 +28,8 @@ void launch_tree_hist_reg(const uint8_t*, int*, const int*, const int*,
 void launch_node_advance(const uint8_t*, const int*, int*, const int*,
                          const int*, const int*, const int*, const long long*,
                          long long, int, int, void*);
-void launch_tree_predict(const float*, const int*, const float*, const int*,
-                         const int*, const float*, const long long*, float*,
+void launch_tree_predict(const float*, const void*, const float*,
+                         const long long*, float*,
                          long long, int, int, int, int, void*);
 void launch_logreg_delta(const float*, const int*, const float*, const float*,
                          const float*, float*, float*, long long, int, int,
@@ -211,24 +211,24 @@ torch::Tensor node_advance(torch::Tensor codes, torch::Tensor nodeid,
   return out;
 }
 
-torch::Tensor tree_predict(torch::Tensor X, torch::Tensor feature,
-                           torch::Tensor threshold, torch::Tensor left,
-                           torch::Tensor right, torch::Tensor value,
-                           torch::Tensor catmask) {
+torch::Tensor tree_predict(torch::Tensor X, torch::Tensor packed,
+                           torch::Tensor value, torch::Tensor catmask) {
   CHK(X);
-  CHK(feature);
+  CHK(packed);
   CHK(value);
+  TORCH_CHECK(packed.scalar_type() == torch::kInt32, "packed must be i32");
+  TORCH_CHECK(packed.dim() == 3 && packed.size(2) == 4,
+              "packed must be T x M x 4");
   TORCH_CHECK(catmask.scalar_type() == torch::kInt64, "catmask must be i64");
   const long long N = X.size(0);
   const int F = X.size(1);
-  const int T = feature.size(0);
-  const int M = feature.size(1);
+  const int T = packed.size(0);
+  const int M = packed.size(1);
   const int K = value.size(2);
   TORCH_CHECK(K <= 16, "tree_predict supports up to 16 classes");
   auto out = torch::empty({N, K}, X.options().dtype(torch::kFloat32));
-  launch_tree_predict(X.data_ptr<float>(), feature.data_ptr<int>(),
-                      threshold.data_ptr<float>(), left.data_ptr<int>(),
-                      right.data_ptr<int>(), value.data_ptr<float>(),
+  launch_tree_predict(X.data_ptr<float>(), packed.data_ptr<int>(),
+                      value.data_ptr<float>(),
                       reinterpret_cast<const long long*>(catmask.data_ptr<int64_t>()),
                       out.data_ptr<float>(), N, F, T, M, K, cur_stream());
   return out;

@@ -709,6 +709,12 @@ extern "C" __global__ void flowml_node_advance(
 // K8 — batched ensemble inference
 // ---------------------------------------------------------------------------
 
+// Packed-node inference layout: one 16-B uint4 per node = {feature (bit 30
+// flags a categorical split), threshold bits, left, right} — ONE L1/L2
+// access per visit instead of four scattered SoA loads (feature/threshold/
+// left/right). catmask stays a side array read only at flagged nodes.
+#define FLOWML_CAT_FLAG (1 << 30)
+
 // FLOWML_TPAR trees traverse concurrently per thread: a single tree's walk
 // is a serial chain of dependent loads (feature -> compare -> child ->
 // feature, ~2 loads x depth levels), so one-tree-at-a-time is latency-bound
@@ -719,10 +725,8 @@ extern "C" __global__ void flowml_node_advance(
 #define FLOWML_TPAR 4
 
 extern "C" __global__ void flowml_tree_predict(
-    const float* __restrict__ X, const int* __restrict__ feature,
-    const float* __restrict__ threshold, const int* __restrict__ left,
-    const int* __restrict__ right, const float* __restrict__ value,
-    const long long* __restrict__ catmask,
+    const float* __restrict__ X, const uint4* __restrict__ packed,
+    const float* __restrict__ value, const long long* __restrict__ catmask,
     float* __restrict__ out, long long N, int F, int T, int M, int K) {
   for (long long r = blockIdx.x * (long long)blockDim.x + threadIdx.x; r < N;
        r += (long long)gridDim.x * blockDim.x) {
@@ -733,25 +737,34 @@ extern "C" __global__ void flowml_tree_predict(
     for (int t0 = 0; t0 < T; t0 += FLOWML_TPAR) {
       const int nt = min(FLOWML_TPAR, T - t0);
       int node[FLOWML_TPAR];
-      int f[FLOWML_TPAR];
+      uint4 nd[FLOWML_TPAR];
 #pragma unroll
       for (int j = 0; j < FLOWML_TPAR; ++j) {
         node[j] = 0;
-        f[j] = (j < nt) ? feature[(long long)(t0 + j) * M] : -1;
+        nd[j] = (j < nt) ? packed[(long long)(t0 + j) * M]
+                         : uint4{(unsigned)-1, 0, 0, 0};
       }
       bool live = true;
       while (live) {
         live = false;
 #pragma unroll
         for (int j = 0; j < FLOWML_TPAR; ++j) {
-          if (f[j] < 0) continue;
+          int f = (int)nd[j].x;
+          if (f < 0) continue;
           const long long base = (long long)(t0 + j) * M;
-          const int n = flowml_left_raw(xr[f[j]], threshold[base + node[j]],
-                                        catmask[base + node[j]])
-                            ? left[base + node[j]]
-                            : right[base + node[j]];
+          bool go_left;
+          const float thr = __uint_as_float(nd[j].y);
+          if (f & FLOWML_CAT_FLAG) {
+            f &= ~FLOWML_CAT_FLAG;
+            const float x = xr[f];
+            const long long m = catmask[base + node[j]];
+            go_left = (x >= 0.f && x <= 63.f) && (((m >> (int)x) & 1LL) != 0);
+          } else {
+            go_left = xr[f] <= thr;
+          }
+          const int n = go_left ? (int)nd[j].z : (int)nd[j].w;
           node[j] = n;
-          f[j] = feature[base + n];
+          nd[j] = packed[base + n];
           live = true;
         }
       }
@@ -1009,16 +1022,15 @@ extern "C" void launch_node_advance(const uint8_t* codes, const int* nodeid_in,
                      nodeid_out, feat, thr, left, right, catmask, N, F);
 }
 
-extern "C" void launch_tree_predict(const float* X, const int* feature,
-                                    const float* threshold, const int* left,
-                                    const int* right, const float* value,
+extern "C" void launch_tree_predict(const float* X, const void* packed,
+                                    const float* value,
                                     const long long* catmask, float* out,
                                     long long N, int F, int T, int M, int K,
                                     void* stream) {
   const int block = 256;
   hipLaunchKernelGGL(flowml_tree_predict, dim3(grid_1d(N, block, 8192)),
-                     dim3(block), 0, (hipStream_t)stream, X, feature, threshold,
-                     left, right, value, catmask, out, N, F, T, M, K);
+                     dim3(block), 0, (hipStream_t)stream, X,
+                     (const uint4*)packed, value, catmask, out, N, F, T, M, K);
 }
 
 extern "C" void launch_logreg_delta(const float* X, const int* y,
