@@ -53,11 +53,19 @@ def _features_tensor(dataset: FlowFrame, col: str) -> torch.Tensor:
     return x
 
 
-def _label_tensor(dataset: FlowFrame, col: str) -> torch.Tensor:
+def _label_tensor(dataset: FlowFrame, col: str,
+                  classification: bool = False) -> torch.Tensor:
     y = dataset.col(col)
     if not isinstance(y, torch.Tensor):
         raise TypeError(f"label column {col!r} must be numeric")
-    return y.reshape(-1)
+    y = y.reshape(-1)
+    if classification and len(y) and bool((y < 0).any()):
+        # Spark: "Labels MUST be in [0, maxNumClasses)" — fail clearly
+        # instead of crashing in the label indexing later
+        raise ValueError(
+            f"classification labels in column {col!r} must be non-negative "
+            f"integers (found min {float(y.min())})")
+    return y
 
 
 class Classifier(_ClassifierParams, Estimator):

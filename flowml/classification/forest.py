@@ -138,7 +138,7 @@ class RandomForestClassifier(_TreeEnsembleParams, ProbabilisticClassifier):
 
     def _fit(self, dataset: FlowFrame) -> "RandomForestClassificationModel":
         X = _features_tensor(dataset, self.getFeaturesCol()).to(torch.float32)
-        y = _label_tensor(dataset, self.getLabelCol()).to(torch.int64)
+        y = _label_tensor(dataset, self.getLabelCol(), classification=True).to(torch.int64)
         dev = X.device
         N, F = X.shape
         kmax = torch.tensor([float(y.max().item()) if N else 0.0], device=dev)

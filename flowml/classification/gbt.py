@@ -153,7 +153,7 @@ class GBTClassifier(_GBTParams, ProbabilisticClassifier):
 
     def _fit(self, dataset: FlowFrame) -> "GBTClassificationModel":
         X = _features_tensor(dataset, self.getFeaturesCol()).to(torch.float32)
-        y01 = _label_tensor(dataset, self.getLabelCol()).to(torch.float32)
+        y01 = _label_tensor(dataset, self.getLabelCol(), classification=True).to(torch.float32)
         dev = X.device
         N, F = X.shape
         if bool((y01.max() > 1).item()) if N else False:
@@ -352,7 +352,7 @@ class GBTClassificationModel(_GBTParams, _HasTrainingSummary, ProbabilisticClass
         from flowml.classification.tree import TreeEnsemble
 
         X = _features_tensor(dataset, self.getFeaturesCol()).to(torch.float32)
-        y = _label_tensor(dataset, self.getLabelCol()).to(torch.float32)
+        y = _label_tensor(dataset, self.getLabelCol(), classification=True).to(torch.float32)
         ysign = (2.0 * y - 1.0).to(X.device)
         margin = torch.zeros(len(y), dtype=torch.float32, device=X.device)
         out = []

@@ -177,7 +177,7 @@ class LogisticRegression(_LogisticRegressionParams, ProbabilisticClassifier):
 
     def _fit(self, dataset: FlowFrame) -> "LogisticRegressionModel":
         X = _features_tensor(dataset, self.getFeaturesCol()).to(torch.float32)
-        y = _label_tensor(dataset, self.getLabelCol()).to(torch.int64)
+        y = _label_tensor(dataset, self.getLabelCol(), classification=True).to(torch.int64)
         dev = X.device
         N, F = X.shape
         w = None

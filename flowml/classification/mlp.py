@@ -164,7 +164,7 @@ class MultilayerPerceptronClassifier(_MLPParams, ProbabilisticClassifier):
         if not layers or len(layers) < 2:
             raise ValueError("layers must have at least input and output sizes")
         X = _features_tensor(dataset, self.getFeaturesCol()).to(torch.float32)
-        y = _label_tensor(dataset, self.getLabelCol()).to(torch.int64)
+        y = _label_tensor(dataset, self.getLabelCol(), classification=True).to(torch.int64)
         dev = X.device
         if X.shape[1] != layers[0]:
             raise ValueError(

@@ -45,7 +45,7 @@ class NaiveBayes(_NaiveBayesParams, ProbabilisticClassifier):
 
     def _fit(self, dataset: FlowFrame) -> "NaiveBayesModel":
         X = _features_tensor(dataset, self.getFeaturesCol()).to(torch.float64)
-        y = _label_tensor(dataset, self.getLabelCol()).to(torch.int64)
+        y = _label_tensor(dataset, self.getLabelCol(), classification=True).to(torch.int64)
         dev = X.device
         N, F = X.shape
         mtype = self.getOrDefault(self.modelType)

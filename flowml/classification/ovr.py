@@ -38,7 +38,7 @@ class OneVsRest(_ClassifierParams, Estimator):
     def _fit(self, dataset: FlowFrame) -> "OneVsRestModel":
         if self.classifier is None:
             raise ValueError("OneVsRest requires a classifier to be set")
-        y = _label_tensor(dataset, self.getLabelCol())
+        y = _label_tensor(dataset, self.getLabelCol(), classification=True)
         num_classes = int(y.max().item()) + 1 if len(y) else 2
         label_col = self.getLabelCol()
         models: List[ClassificationModel] = []

@@ -55,7 +55,7 @@ class LinearSVC(_LinearSVCParams, Classifier):
         from flowml.optim import lbfgs
 
         X = _features_tensor(dataset, self.getFeaturesCol()).to(torch.float32)
-        y = _label_tensor(dataset, self.getLabelCol()).to(torch.float32)
+        y = _label_tensor(dataset, self.getLabelCol(), classification=True).to(torch.float32)
         dev = X.device
         N, F = X.shape
         if bool((y.max() > 1).item()) if N else False:

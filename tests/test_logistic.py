@@ -192,3 +192,15 @@ def test_linear_svc_summary_curves():
     assert s.areaUnderROC > 0.9
     assert 0.8 < s.accuracy <= 1.0
     assert len(s.precisionByThreshold) > 10
+
+
+def test_negative_labels_clear_error():
+    for est in ("LogisticRegression", "RandomForestClassifier",
+                "MultilayerPerceptronClassifier"):
+        import flowml.classification as C
+
+        cls = getattr(C, est)
+        kw = {"layers": [2, 4, 2]} if est.startswith("Multilayer") else {}
+        with pytest.raises(ValueError, match="non-negative"):
+            cls(**kw).fit(FlowFrame({"features": torch.zeros(4, 2),
+                                     "label": np.array([-1.0, 0, 1, 0])}))
