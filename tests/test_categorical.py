@@ -436,3 +436,43 @@ def test_index_to_string_reads_label_metadata():
     with pytest.raises(ValueError, match="no labels"):
         IndexToString(inputCol="i", outputCol="s2").transform(
             FlowFrame({"i": out["i"]}))
+
+
+def test_max_arity_63_mask_boundary():
+    """Arity 63 (the int64-mask ceiling): splits using the top category
+    (bit 62) must round-trip and route correctly."""
+    r = np.random.default_rng(20)
+    n, k = 6000, 63
+    cat = r.integers(0, k, n)
+    left_set = set(range(0, 63, 2))          # evens left (incl. 62)
+    y = np.isin(cat, list(left_set)).astype(np.float64)
+    X = np.column_stack([cat, r.normal(size=n)]).astype(np.float32)
+    df = FlowFrame({"features": torch.from_numpy(X), "label": y}).withMeta(
+        "features", {"categorical": {0: k}})
+    m = DecisionTreeClassifier(maxDepth=1, maxBins=64, seed=0).fit(df)
+    t = m._tree
+    assert t.feature[0] == 0 and t.cat_mask[0] != 0
+    mask = int(t.cat_mask[0])
+    got_left = {c for c in range(63) if (mask >> c) & 1}
+    # the split is the planted partition or its complement
+    assert got_left in (left_set, set(range(63)) - left_set)
+    pred = m.transform(df)["prediction"].numpy()
+    assert (pred == y).mean() == 1.0
+    # persistence round-trip keeps the full-width mask
+    import tempfile, os as _os
+    d = tempfile.mkdtemp()
+    m.save(_os.path.join(d, "t"))
+    from flowml.classification import DecisionTreeClassificationModel
+    m2 = DecisionTreeClassificationModel.load(_os.path.join(d, "t"))
+    assert int(m2._tree.cat_mask[0]) == mask
+
+
+def test_arity_64_rejected():
+    r = np.random.default_rng(21)
+    X = np.column_stack([r.integers(0, 64, 500),
+                         r.normal(size=500)]).astype(np.float32)
+    y = r.integers(0, 2, 500).astype(np.float64)
+    df = FlowFrame({"features": torch.from_numpy(X), "label": y}).withMeta(
+        "features", {"categorical": {0: 64}})
+    with pytest.raises(ValueError, match="arity"):
+        DecisionTreeClassifier(maxBins=128, seed=0).fit(df)
