@@ -1021,3 +1021,24 @@ class TestFusedDgradAct:
         a = ops.get_ext().gemm_dgrad_act(D, Wt, H, 1)
         b = ops.get_ext().gemm_dgrad_act(D, Wt, H, 1)
         assert torch.equal(a, b)
+
+    def test_arity63_mask_boundary_gpu(self):
+        """bit-62 categorical masks route identically on the HIP kernels."""
+        from flowml.classification import DecisionTreeClassifier
+        from flowml.frame import FlowFrame
+
+        r = np.random.default_rng(53)
+        n, k = 60_000, 63
+        cat = r.integers(0, k, n)
+        y = np.isin(cat, list(range(0, 63, 2))).astype(np.float64)
+        X = np.column_stack([cat, r.normal(size=n)]).astype(np.float32)
+        meta = {"categorical": {0: k}}
+        df_cpu = FlowFrame({"features": torch.from_numpy(X), "label": y}).withMeta("features", meta)
+        df_gpu = df_cpu.to(DEV).withMeta("features", meta)
+        m_cpu = DecisionTreeClassifier(maxDepth=1, maxBins=64, seed=0).fit(df_cpu)
+        m_gpu = DecisionTreeClassifier(maxDepth=1, maxBins=64, seed=0).fit(df_gpu)
+        assert int(m_cpu._tree.cat_mask[0]) == int(m_gpu._tree.cat_mask[0])
+        p_cpu = m_cpu.transform(df_cpu)["prediction"]
+        p_gpu = m_gpu.transform(df_gpu)["prediction"].cpu()
+        assert torch.equal(p_cpu, p_gpu)
+        assert (p_cpu.numpy() == y).mean() == 1.0
