@@ -339,8 +339,16 @@ def bench_infer(args, device):
                                    seed=args.seed).fit(make_df(X, y))
     batch = args.batch
     use_graph = device.type == "cuda"
-    sc = StreamingClassifier(model, batch_size=batch, num_features=41,
-                             device=device, use_graph=use_graph)
+    featurizer = None
+    if getattr(args, "with_featurizer", False):
+        from flowml.feature import StandardScaler
+
+        featurizer = StandardScaler(inputCol="features",
+                                    outputCol="scaled").fit(make_df(X, y))
+    sc = StreamingClassifier(
+        model, batch_size=batch, num_features=41, device=device,
+        use_graph=use_graph, featurizer=featurizer,
+        feature_col="scaled" if featurizer is not None else "features")
     driver = StreamingDriver(sc)
     # small device-resident batch pool cycled as the "stream" (synthetic)
     pool = [synth_flows(batch, 41, 5, args.seed + 31 + 101 * p + rank, device)[0]
@@ -364,6 +372,7 @@ def bench_infer(args, device):
         "micro_batch": batch,
         "features": 41,
         "hipgraph": use_graph and sc._graph is not None,
+        "featurizer_in_graph": featurizer is not None,
         "parallelism": f"dp{world}",
         "rows_all_reduced": res["rows"],
     }
@@ -416,6 +425,14 @@ def main():
     p.add_argument("--depth", type=int, default=5)
     p.add_argument("--batch", type=int, default=262144, help="infer micro-batch")
     p.add_argument("--seed", type=int, default=42)
+    p.add_argument("--scaling", default="weak", choices=["weak", "strong"],
+                   help="weak: per-GPU rows fixed (default, the driver "
+                        "contract); strong: --rows is the GLOBAL row count, "
+                        "divided across ranks")
+    p.add_argument("--with-featurizer", action="store_true",
+                   help="infer mode: capture a StandardScaler featurization "
+                        "stage INSIDE the hipGraph before the ensemble "
+                        "forward (SURVEY K8 featurize+forward shape)")
     p.add_argument("--report-f1-curve", action="store_true",
                    help="print RF fit-time/macro-F1 at depths 5/8/10 and exit")
     p.add_argument("--profile", action="store_true",
@@ -469,8 +486,17 @@ def main():
         defaults = {"rf": 1_250_000, "gbt": 6_250_000, "mlp": 1_000_000,
                     "lr": 1_250_000, "infer": 500_000}
         args.rows = defaults[args.model]
+        if args.scaling == "strong":
+            args.rows *= 8  # global default = the 8-GPU weak total
         if device.type == "cpu":
             args.rows = min(args.rows, 20_000)  # keep the no-GPU sanity path fast
+    if args.scaling == "strong":
+        # fixed TOTAL work: this rank's shard of the global row count
+        from flowml.dist import shard_range
+
+        lo, hi = shard_range(args.rows)
+        args.global_rows_strong = args.rows
+        args.rows = hi - lo
 
     if args.report_f1_curve:
         if device.type == "cpu":
@@ -507,7 +533,7 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
             "higher_is_better": True,
-            "scaling": "weak",
+            "scaling": args.scaling,
             "vs_baseline": None,
             "dtype": dtype,
             "data": "synthetic",
