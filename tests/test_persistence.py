@@ -275,3 +275,65 @@ class TestMegaPipelineRoundtrip:
         lm = PipelineModel.load(str(tmp_path / "mega"))
         assert torch.equal(lm.transform(df)["prediction"], pred)
         assert len(lm.stages) == 8
+
+
+def test_round2_stage_families_in_one_pipeline(tmp_path):
+    """Round-2 additions in one PipelineModel: RobustScaler,
+    UnivariateFeatureSelector, VectorIndexer-categorical trees — bitwise
+    round-trip of predictions."""
+    from flowml import FlowFrame, Pipeline, PipelineModel
+    from flowml.classification import RandomForestClassifier
+    from flowml.feature import (RobustScaler, StringIndexer,
+                                UnivariateFeatureSelector, VectorAssembler)
+
+    r = np.random.default_rng(5)
+    n = 600
+    proto = np.array(["tcp", "udp", "icmp"])[r.integers(0, 3, n)]
+    num = torch.from_numpy(r.normal(size=(n, 5)).astype(np.float32))
+    label = torch.from_numpy(r.integers(0, 2, n).astype(np.float64))
+    df = FlowFrame({"proto": proto, "num": num, "label": label})
+    pipe = Pipeline(stages=[
+        StringIndexer(inputCol="proto", outputCol="p_idx"),
+        VectorAssembler(inputCols=["p_idx", "num"], outputCol="raw"),
+        RobustScaler(inputCol="raw", outputCol="scaled", withCentering=True),
+        UnivariateFeatureSelector(
+            featuresCol="scaled", outputCol="features", labelCol="label",
+            featureType="continuous", labelType="categorical",
+            selectionThreshold=4.0),
+        RandomForestClassifier(numTrees=4, maxDepth=3, seed=0),
+    ])
+    pm = pipe.fit(df)
+    pred = pm.transform(df)["prediction"]
+    pm.save(str(tmp_path / "mega2"))
+    lm = PipelineModel.load(str(tmp_path / "mega2"))
+    assert torch.equal(lm.transform(df)["prediction"], pred)
+    assert [type(s).__name__ for s in lm.stages] == [
+        "StringIndexerModel", "VectorAssembler", "RobustScalerModel",
+        "UnivariateFeatureSelectorModel", "RandomForestClassificationModel"]
+
+
+def test_rformula_gmm_pipeline_round_trip(tmp_path):
+    """RFormula + GaussianMixture as a pipeline (unsupervised tail)."""
+    from flowml import FlowFrame, Pipeline, PipelineModel
+    from flowml.clustering import GaussianMixture
+    from flowml.feature import RFormula
+
+    r = np.random.default_rng(6)
+    n = 400
+    df = FlowFrame({
+        "kind": np.array(["a", "b"])[r.integers(0, 2, n)],
+        "x": torch.from_numpy(np.concatenate(
+            [r.normal(-3, 0.5, n // 2), r.normal(3, 0.5, n - n // 2)]
+        ).astype(np.float32)),
+        "y0": torch.from_numpy(r.normal(size=n).astype(np.float32)),
+        "lbl": torch.from_numpy(r.integers(0, 2, n).astype(np.float64)),
+    })
+    pipe = Pipeline(stages=[
+        RFormula(formula="lbl ~ kind + x + y0"),
+        GaussianMixture(k=2, seed=4),
+    ])
+    pm = pipe.fit(df)
+    pred = pm.transform(df)["prediction"]
+    pm.save(str(tmp_path / "rfgmm"))
+    lm = PipelineModel.load(str(tmp_path / "rfgmm"))
+    assert torch.equal(lm.transform(df)["prediction"], pred)
