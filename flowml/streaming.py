@@ -144,10 +144,16 @@ class StreamingDriver:
         self.rank = dist.get_rank()
         self.world = dist.get_world_size()
 
-    def run(self, batches, n_classes: Optional[int] = None) -> dict:
+    def run(self, batches, n_classes: Optional[int] = None,
+            count_classes: bool = True) -> dict:
         """Consume a global stream (iterable of tensors); returns aggregate
         {"rows": global rows scored, "class_counts": global per-class counts,
-        "rows_local": this rank's share, "elapsed_s", "flows_per_s"}."""
+        "rows_local": this rank's share, "elapsed_s", "flows_per_s"}.
+
+        ``count_classes=False`` skips the per-batch bincount (measured
+        ~0.25 ms/262k batch — 30% of the graph-replay time) and returns
+        ``class_counts=None``; row counts are still all-reduced.
+        """
         import time as _time
 
         from flowml import dist
@@ -166,9 +172,10 @@ class StreamingDriver:
                 continue
             pred, _raw = self.sc.infer(X.to(dev, non_blocking=True))
             n_local += pred.numel()
-            counts += torch.bincount(
-                pred.reshape(-1).to(torch.int64), minlength=C
-            ).to(counts.dtype)[:C]
+            if count_classes:
+                counts += torch.bincount(
+                    pred.reshape(-1).to(torch.int64), minlength=C
+                ).to(counts.dtype)[:C]
         if dev.type == "cuda":
             torch.cuda.synchronize(dev)
         dist.barrier()
@@ -179,7 +186,7 @@ class StreamingDriver:
         return {
             "rows": rows,
             "rows_local": n_local,
-            "class_counts": agg[:-1].cpu().tolist(),
+            "class_counts": agg[:-1].cpu().tolist() if count_classes else None,
             "elapsed_s": elapsed,
             "flows_per_s": rows / elapsed if elapsed > 0 else 0.0,
         }
