@@ -45,7 +45,6 @@ void launch_gemm_bias_act_bf16(const void*, const void*, const float*, void*,
                                long long, int, int, int, void*);
 void launch_gemm_bias_act_f32(const float*, const float*, const float*, float*,
                               long long, int, int, int, void*);
-void launch_act_backward_bf16(const void*, void*, long long, int, void*);
 void launch_gemm_dgrad_act_bf16(const void*, const void*, const void*, void*,
                                 long long, int, int, int, void*);
 void launch_wgrad_bias_bf16(const void*, const void*, float*, long long, int,
@@ -371,15 +370,6 @@ std::tuple<torch::Tensor, torch::Tensor> wgrad_bias(torch::Tensor delta,
   return {gW, gB};
 }
 
-torch::Tensor act_backward_(torch::Tensor h, torch::Tensor delta,
-                            int64_t act) {
-  CHK(h);
-  CHK(delta);
-  TORCH_CHECK(h.scalar_type() == torch::kBFloat16, "bf16 only");
-  launch_act_backward_bf16(h.data_ptr(), delta.data_ptr(), h.numel(),
-                           (int)act, cur_stream());
-  return delta;
-}
 
 
 // deterministic split-K f32 wgrad: gradW = Dᵀ·X (+bias grad), for the LR
@@ -422,6 +412,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("wgrad_f32", &wgrad_f32, "split-K f32 wgrad + bias grad (LR)");
   m.def("softmax_xent_grad", &softmax_xent_grad, "K6 softmax/xent grad");
   m.def("gemm_bias_act", &gemm_bias_act, "K5 fused MFMA GEMM+bias+act");
-  m.def("act_backward_", &act_backward_, "K7 activation backward (in-place)");
   m.def("wgrad_bias", &wgrad_bias, "K7 split-N wgrad + fused bias grad");
 }

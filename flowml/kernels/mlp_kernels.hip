@@ -216,17 +216,6 @@ extern "C" __global__ __launch_bounds__(256) void flowml_gemm_bias_act_f32(
 // relu: h>0 ? 1 : 0 ; sigmoid: h*(1-h) — h is the post-activation value.
 // ---------------------------------------------------------------------------
 
-extern "C" __global__ void flowml_act_backward_bf16(
-    const __bf16* __restrict__ h, __bf16* __restrict__ delta, long long total,
-    int act) {
-  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x;
-       i < total; i += (long long)gridDim.x * blockDim.x) {
-    const float hv = (float)h[i];
-    const float dv = (float)delta[i];
-    const float g = (act == 1) ? (hv > 0.f ? 1.f : 0.f) : hv * (1.f - hv);
-    delta[i] = (__bf16)(dv * g);
-  }
-}
 
 
 static int flowml_env_int_mlp(const char* name, int dflt) {
@@ -268,17 +257,6 @@ extern "C" void launch_gemm_dgrad_act_bf16(const void* D, const void* Wt,
                      Kp, act);
 }
 
-extern "C" void launch_act_backward_bf16(const void* h, void* delta,
-                                         long long total, int act,
-                                         void* stream) {
-  const int block = 256;
-  long long g = (total + block - 1) / block;
-  if (g > 4096) g = 4096;
-  if (g < 1) g = 1;
-  hipLaunchKernelGGL(flowml_act_backward_bf16, dim3((int)g), dim3(block), 0,
-                     (hipStream_t)stream, (const __bf16*)h, (__bf16*)delta,
-                     total, act);
-}
 
 // ---------------------------------------------------------------------------
 // K7 — wgrad with fused bias-grad: gW[K,F] = deltaᵀ[K,N] @ act[N,F],
