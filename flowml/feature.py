@@ -1591,8 +1591,9 @@ def _selector_stats(X: torch.Tensor, y: torch.Tensor, feature_type: str,
         if xa.min(initial=0) < 0 or ya.min(initial=0) < 0:
             raise ValueError("categorical values must be non-negative indices")
         vmax = torch.tensor([float(xa.max(initial=0)),
-                             float(ya.max(initial=0))])
+                             float(ya.max(initial=0))], device=X.device)
         dist.all_reduce_(vmax, op="max")
+        vmax = vmax.cpu()
         V, C = int(vmax[0]) + 1, int(vmax[1]) + 1
         if F * V * C > 100_000_000:
             raise ValueError(
@@ -1603,9 +1604,11 @@ def _selector_stats(X: torch.Tensor, y: torch.Tensor, feature_type: str,
         cont = np.zeros((F, V, C), np.float64)
         for j in range(F):
             np.add.at(cont[j], (xa[:, j], ya), 1.0)
-        cont_t = torch.from_numpy(cont.reshape(-1))
+        # all_reduce on the features' device (NCCL needs device tensors;
+        # gloo handles CPU) — then host math
+        cont_t = torch.from_numpy(cont.reshape(-1)).to(X.device)
         dist.all_reduce_(cont_t)
-        cont = cont_t.numpy().reshape(F, V, C)
+        cont = cont_t.cpu().numpy().reshape(F, V, C)
         scores = np.zeros(F)
         pvals = np.ones(F)
         for j in range(F):
@@ -1625,8 +1628,9 @@ def _selector_stats(X: torch.Tensor, y: torch.Tensor, feature_type: str,
         return scores, pvals
     if feature_type == "continuous" and label_type == "categorical":
         ya = yd.cpu().numpy().astype(np.int64)
-        kmax = torch.tensor([float(ya.max(initial=0))])
+        kmax = torch.tensor([float(ya.max(initial=0))], device=Xd.device)
         dist.all_reduce_(kmax, op="max")
+        kmax = kmax.cpu()
         K = int(kmax[0]) + 1
         oh = torch.zeros(N, K, dtype=torch.float64, device=Xd.device)
         oh[torch.arange(N, device=Xd.device), yd.to(torch.int64)] = 1.0
@@ -1634,8 +1638,9 @@ def _selector_stats(X: torch.Tensor, y: torch.Tensor, feature_type: str,
             oh.sum(dim=0),                     # K class counts
             (oh.T @ Xd).reshape(-1),           # K×F class sums
             (oh.T @ (Xd * Xd)).reshape(-1),    # K×F class sum-squares
-        ]).cpu()
-        dist.all_reduce_(buf)
+        ])
+        dist.all_reduce_(buf)                  # on-device (NCCL-safe)
+        buf = buf.cpu()
         nk = buf[:K].numpy()
         sk = buf[K:K + K * F].view(K, F).numpy()
         qk = buf[K + K * F:].view(K, F).numpy()
@@ -1658,8 +1663,9 @@ def _selector_stats(X: torch.Tensor, y: torch.Tensor, feature_type: str,
             Xd.sum(dim=0), (Xd * Xd).sum(dim=0),
             (Xd * yd.unsqueeze(1)).sum(dim=0),
             yd.sum().reshape(1), (yd * yd).sum().reshape(1),
-        ]).cpu()
-        dist.all_reduce_(buf)
+        ])
+        dist.all_reduce_(buf)                  # on-device (NCCL-safe)
+        buf = buf.cpu()
         n = float(buf[0])
         sx = buf[1:1 + F].numpy()
         qx = buf[1 + F:1 + 2 * F].numpy()
