@@ -127,6 +127,38 @@ class TestRcclWorld1:
         assert np.allclose(mm.originalMax.numpy(), x.max(0).values.double().cpu().numpy())
         MaxAbsScaler(inputCol="vec", outputCol="o").fit(df)
 
+    def test_round2_estimators_under_live_comm(self, nccl_world1):
+        """Round-2 estimators whose stat reduces must be NCCL-device-safe:
+        RobustScaler (dist quantile machinery), UnivariateFeatureSelector
+        (on-device contingency/moment reduces), GMM/GLR flat buffers."""
+        from flowml.feature import RobustScaler, UnivariateFeatureSelector
+        from flowml.frame import FlowFrame
+        from flowml.regression import GeneralizedLinearRegression
+
+        r = np.random.default_rng(3)
+        X = torch.from_numpy(r.normal(size=(20_000, 5)).astype(np.float32)).to("cuda:0")
+        yk = torch.from_numpy(r.integers(0, 3, 20_000).astype(np.float64)).to("cuda:0")
+        df = FlowFrame({"features": X, "label": yk}, device="cuda:0")
+        rs = RobustScaler(inputCol="features", outputCol="s").fit(df)
+        assert rs.range.shape == (5,)
+        sel = UnivariateFeatureSelector(
+            featuresCol="features", outputCol="s", labelCol="label",
+            featureType="continuous", labelType="categorical",
+            selectionThreshold=2.0).fit(df)
+        assert len(sel.selectedFeatures) == 2
+        # categorical/categorical chi2 path (device contingency reduce)
+        Xc = torch.from_numpy(r.integers(0, 4, (20_000, 3)).astype(np.float32)).to("cuda:0")
+        dfc = FlowFrame({"features": Xc, "label": yk}, device="cuda:0")
+        sel2 = UnivariateFeatureSelector(
+            featuresCol="features", outputCol="s", labelCol="label",
+            featureType="categorical", labelType="categorical",
+            selectionThreshold=1.0).fit(dfc)
+        assert len(sel2.selectedFeatures) == 1
+        yp = torch.from_numpy(
+            r.poisson(2.0, 20_000).astype(np.float64)).to("cuda:0")
+        GeneralizedLinearRegression(family="poisson").fit(
+            FlowFrame({"features": X, "label": yp}, device="cuda:0"))
+
 
 # NOTE: a world=2 communicator on ONE GPU was tried and is NOT possible:
 # RCCL (like NCCL) rejects two ranks on the same device ("duplicate GPU",
