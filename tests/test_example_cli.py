@@ -67,3 +67,42 @@ def test_anomaly_example_cpu():
     prec = float(line.split("precision=")[1].split()[0])
     rec = float(line.split("recall=")[1].split()[0])
     assert prec > 0.9 and rec > 0.9, line
+
+
+def test_serve_example_cpu(tmp_path):
+    """train -> save -> load -> serve loop (examples/serve_flows.py)."""
+    save = str(tmp_path / "m")
+    r1 = subprocess.run(
+        [sys.executable, CLI, "--rows", "3000", "--model", "rf",
+         "--save", save, "--device", "cpu"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert r1.returncode == 0, r1.stderr[-500:]
+    serve = os.path.join(REPO, "examples", "serve_flows.py")
+    r2 = subprocess.run(
+        [sys.executable, serve, "--model-dir", save, "--batch", "1024",
+         "--batches", "5", "--device", "cpu"],
+        capture_output=True, text=True, timeout=300, cwd=REPO,
+    )
+    assert r2.returncode == 0, r2.stdout[-500:] + r2.stderr[-500:]
+    assert "served 5120 flows" in r2.stdout
+    assert "class counts:" in r2.stdout
+
+
+@pytest.mark.gpu
+def test_serve_example_gpu(tmp_path):
+    save = str(tmp_path / "m")
+    r1 = subprocess.run(
+        [sys.executable, CLI, "--rows", "20000", "--model", "rf",
+         "--save", save, "--device", "cuda:0"],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert r1.returncode == 0, r1.stderr[-500:]
+    serve = os.path.join(REPO, "examples", "serve_flows.py")
+    r2 = subprocess.run(
+        [sys.executable, serve, "--model-dir", save, "--batch", "65536",
+         "--batches", "40", "--device", "cuda:0"],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert r2.returncode == 0, r2.stdout[-500:] + r2.stderr[-500:]
+    assert "hipGraph=True" in r2.stdout
