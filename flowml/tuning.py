@@ -91,12 +91,20 @@ class CrossValidator(_ValidatorParams, Estimator):
     numFolds: Param[int] = Param(
         "undefined", "numFolds", "number of folds for cross validation (>= 2)", TypeConverters.toInt
     )
+    foldCol: Param[str] = Param(
+        "undefined", "foldCol",
+        "Param for the column name of user specified fold number. Once this "
+        "is specified, CrossValidator won't do random k-fold split. Note "
+        "that this column should be integer type with range [0, numFolds).",
+        TypeConverters.toString,
+    )
 
     def __init__(self, *, estimator=None, estimatorParamMaps=None, evaluator=None,
-                 numFolds: Optional[int] = None, seed: Optional[int] = None):
+                 numFolds: Optional[int] = None, seed: Optional[int] = None,
+                 foldCol: Optional[str] = None):
         super().__init__()
-        self._setDefault(numFolds=3)
-        self._set(numFolds=numFolds, seed=seed)
+        self._setDefault(numFolds=3, foldCol="")
+        self._set(numFolds=numFolds, seed=seed, foldCol=foldCol)
         if estimator is not None:
             self.estimator = estimator
         if estimatorParamMaps is not None:
@@ -109,8 +117,17 @@ class CrossValidator(_ValidatorParams, Estimator):
         k = self.getOrDefault(self.numFolds)
         seed = self.getOrDefault(self.seed)
         n = dataset.count()
-        rng = np.random.default_rng(seed)
-        fold_of = rng.integers(0, k, n)  # Spark: random fold assignment
+        fold_col = self.getOrDefault(self.foldCol)
+        if fold_col:
+            # Spark 3.1 foldCol: user-specified integer folds in [0, k)
+            fc = dataset.col(fold_col)
+            fold_of = fc.detach().cpu().numpy().astype(np.int64).reshape(-1)
+            if fold_of.size and (fold_of.min() < 0 or fold_of.max() >= k):
+                raise ValueError(
+                    f"foldCol {fold_col!r} values must be in [0, {k})")
+        else:
+            rng = np.random.default_rng(seed)
+            fold_of = rng.integers(0, k, n)  # Spark: random fold assignment
         n_maps = len(self.estimatorParamMaps)
         metrics = np.zeros(n_maps, dtype=np.float64)
         for fold in range(k):

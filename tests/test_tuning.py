@@ -145,3 +145,29 @@ class TestValidatorModelPersistence:
         lm = CrossValidatorModel.load(str(tmp_path / "cvm"))
         assert lm.avgMetrics == pytest.approx(m.avgMetrics)
         assert torch.equal(lm.transform(df)["prediction"], m.transform(df)["prediction"])
+
+
+def test_cross_validator_fold_col():
+    from flowml.classification import LogisticRegression
+    from flowml.evaluation import MulticlassClassificationEvaluator
+    from flowml.tuning import CrossValidator, ParamGridBuilder
+
+    r = np.random.default_rng(0)
+    X = r.normal(size=(300, 4)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float64)
+    folds = np.arange(300) % 3
+    df = FlowFrame({"features": torch.from_numpy(X), "label": y,
+                    "fold": folds.astype(np.float64)})
+    lr = LogisticRegression(maxIter=10)
+    grid = ParamGridBuilder().addGrid(lr.regParam, [0.0, 0.1]).build()
+    cv = CrossValidator(estimator=lr, estimatorParamMaps=grid,
+                        evaluator=MulticlassClassificationEvaluator(
+                            metricName="accuracy"),
+                        numFolds=3, foldCol="fold")
+    m = cv.fit(df)
+    assert len(m.avgMetrics) == 2 and max(m.avgMetrics) > 0.8
+    bad = df.withColumn("fold", torch.full((300,), 5.0))
+    with pytest.raises(ValueError, match="foldCol"):
+        CrossValidator(estimator=lr, estimatorParamMaps=grid,
+                       evaluator=MulticlassClassificationEvaluator(),
+                       numFolds=3, foldCol="fold").fit(bad)
