@@ -1831,6 +1831,119 @@ class ChiSqSelector(HasInputCol, HasOutputCol, HasLabelCol, Estimator):
 
 
 # ---------------------------------------------------------------------------
+# VarianceThresholdSelector / VectorSizeHint (spark:.../feature/)
+# ---------------------------------------------------------------------------
+
+
+class VarianceThresholdSelector(HasOutputCol, Estimator):
+    """Drop features whose (sample) variance is <= varianceThreshold
+    (spark:.../feature/VarianceThresholdSelector.scala, default 0 — removes
+    constant features). Moments are all-reduced, so sharded fits agree."""
+
+    featuresCol: Param[str] = Param(
+        "undefined", "featuresCol", "features column name",
+        TypeConverters.toString)
+    varianceThreshold: Param[float] = Param(
+        "undefined", "varianceThreshold",
+        "Param for variance threshold. Features with a (sample) variance "
+        "not greater than this threshold will be removed.",
+        TypeConverters.toFloat)
+
+    def __init__(self, *, featuresCol: Optional[str] = None,
+                 outputCol: Optional[str] = None,
+                 varianceThreshold: Optional[float] = None):
+        super().__init__()
+        self._setDefault(featuresCol="features", varianceThreshold=0.0)
+        self._set(featuresCol=featuresCol, outputCol=outputCol,
+                  varianceThreshold=varianceThreshold)
+
+    def _fit(self, dataset: FlowFrame) -> "VarianceThresholdSelectorModel":
+        from flowml import dist
+
+        x = dataset.col(self.getOrDefault(self.featuresCol))
+        if not isinstance(x, torch.Tensor) or x.dim() != 2:
+            raise TypeError("VarianceThresholdSelector input must be a vector column")
+        xd = x.to(torch.float64)
+        F = xd.shape[1]
+        buf = torch.cat([
+            torch.tensor([float(xd.shape[0])], dtype=torch.float64,
+                         device=xd.device),
+            xd.sum(dim=0), (xd * xd).sum(dim=0)])
+        dist.all_reduce_(buf)
+        n = float(buf[0])
+        mean = buf[1:1 + F] / max(n, 1.0)
+        var = (buf[1 + F:] - n * mean * mean) / max(n - 1.0, 1.0)  # sample var
+        thr = self.getOrDefault(self.varianceThreshold)
+        sel = [int(i) for i in torch.nonzero(var.cpu() > thr).flatten()]
+        model = VarianceThresholdSelectorModel(selectedFeatures=sel)
+        self._copyValues(model)
+        model._set_parent(self)
+        return model
+
+
+class VarianceThresholdSelectorModel(HasOutputCol, Model):
+    featuresCol: Param[str] = Param(
+        "undefined", "featuresCol", "features column name",
+        TypeConverters.toString)
+
+    def __init__(self, selectedFeatures: Optional[List[int]] = None):
+        super().__init__()
+        self._setDefault(featuresCol="features")
+        self.selectedFeatures: List[int] = list(selectedFeatures or [])
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        x = dataset.col(self.getOrDefault(self.featuresCol))
+        idx = torch.tensor(self.selectedFeatures, dtype=torch.int64,
+                           device=x.device)
+        return dataset.withColumn(self.getOrDefault(self.outputCol),
+                                  x.index_select(1, idx))
+
+    def write(self):
+        from flowml.io import GenericWriter
+
+        return GenericWriter(self)
+
+    def save(self, path: str) -> None:
+        self.write().save(path)
+
+    @classmethod
+    def load(cls, path: str):
+        from flowml.io import load as _load
+
+        return _load(path)
+
+
+class VectorSizeHint(HasInputCol, HasHandleInvalid, Transformer):
+    """Validate (and annotate) a vector column's size
+    (spark:.../feature/VectorSizeHint.scala). Dense tensors always carry
+    their size here, so this validates and stamps {"size": n} metadata."""
+
+    size: Param[int] = Param(
+        "undefined", "size", "the size of the vector that this hint is for",
+        TypeConverters.toInt)
+
+    def __init__(self, *, inputCol: Optional[str] = None,
+                 size: Optional[int] = None,
+                 handleInvalid: Optional[str] = None):
+        super().__init__()
+        self._setDefault(handleInvalid="error")
+        self._set(inputCol=inputCol, size=size, handleInvalid=handleInvalid)
+
+    def _transform(self, dataset: FlowFrame) -> FlowFrame:
+        col = self.getInputCol()
+        x = dataset.col(col)
+        want = self.getOrDefault(self.size)
+        if not isinstance(x, torch.Tensor) or x.dim() != 2:
+            raise TypeError("VectorSizeHint input must be a vector column")
+        if x.shape[1] != want and self.getHandleInvalid() == "error":
+            raise ValueError(
+                f"column {col!r} has vector size {x.shape[1]}, expected {want}")
+        meta = dataset.meta(col)
+        meta["size"] = int(want)
+        return dataset.withMeta(col, meta)
+
+
+# ---------------------------------------------------------------------------
 # PCA (spark:.../feature/PCA.scala)
 # ---------------------------------------------------------------------------
 

@@ -78,6 +78,9 @@ _register("PCAModel", "org.apache.spark.ml.feature.PCAModel")
 _register("VectorSlicer", "org.apache.spark.ml.feature.VectorSlicer")
 _register("VectorIndexer", "org.apache.spark.ml.feature.VectorIndexer")
 _register("RFormula", "org.apache.spark.ml.feature.RFormula")
+_register("VarianceThresholdSelector", "org.apache.spark.ml.feature.VarianceThresholdSelector")
+_register("VarianceThresholdSelectorModel", "org.apache.spark.ml.feature.VarianceThresholdSelectorModel")
+_register("VectorSizeHint", "org.apache.spark.ml.feature.VectorSizeHint")
 _register("RFormulaModel", "org.apache.spark.ml.feature.RFormulaModel")
 _register("RobustScaler", "org.apache.spark.ml.feature.RobustScaler")
 _register("UnivariateFeatureSelector", "org.apache.spark.ml.feature.UnivariateFeatureSelector")
@@ -494,6 +497,8 @@ def _data_rows(instance, path: str) -> Optional[List[Dict[str, Any]]]:
                 "originalMax": vector_struct(instance.originalMax.numpy()),
             }
         ]
+    if cls == "VarianceThresholdSelectorModel":
+        return [{"selectedFeatures": [int(i) for i in instance.selectedFeatures]}]
     if cls == "UnivariateFeatureSelectorModel":
         return [{"selectedFeatures": [int(i) for i in instance.selectedFeatures]}]
     if cls == "RobustScalerModel":
@@ -727,6 +732,8 @@ def _construct(cls, simple: str, rows, path: str, meta):
             originalMin=torch.from_numpy(vector_from_struct(rows[0]["originalMin"])),
             originalMax=torch.from_numpy(vector_from_struct(rows[0]["originalMax"])),
         )
+    if simple == "VarianceThresholdSelectorModel":
+        return cls(selectedFeatures=[int(i) for i in rows[0]["selectedFeatures"]])
     if simple == "UnivariateFeatureSelectorModel":
         return cls(selectedFeatures=[int(i) for i in rows[0]["selectedFeatures"]])
     if simple == "RobustScalerModel":

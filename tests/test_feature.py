@@ -565,3 +565,36 @@ class TestRFormula:
         b = m2.transform(df)
         assert torch.equal(a["features"], b["features"])
         assert torch.equal(a["label"], b["label"])
+
+
+class TestVarianceThresholdAndSizeHint:
+    def test_variance_selector(self, tmp_path):
+        from flowml.feature import (VarianceThresholdSelector,
+                                    VarianceThresholdSelectorModel)
+
+        r = np.random.default_rng(0)
+        X = np.column_stack([
+            r.normal(size=200),            # var ~1
+            np.full(200, 3.0),             # constant -> dropped at default 0
+            r.normal(scale=0.01, size=200) # tiny var
+        ]).astype(np.float32)
+        df = FlowFrame({"features": torch.from_numpy(X)})
+        m = VarianceThresholdSelector(outputCol="s").fit(df)
+        assert m.selectedFeatures == [0, 2]
+        m2 = VarianceThresholdSelector(outputCol="s",
+                                       varianceThreshold=0.01).fit(df)
+        assert m2.selectedFeatures == [0]
+        p = str(tmp_path / "vts")
+        m.save(p)
+        m3 = VarianceThresholdSelectorModel.load(p)
+        assert m3.selectedFeatures == [0, 2]
+        assert m3.transform(df)["s"].shape == (200, 2)
+
+    def test_vector_size_hint(self):
+        from flowml.feature import VectorSizeHint
+
+        df = FlowFrame({"f": torch.zeros(5, 3)})
+        out = VectorSizeHint(inputCol="f", size=3).transform(df)
+        assert out.meta("f")["size"] == 3
+        with pytest.raises(ValueError, match="vector size"):
+            VectorSizeHint(inputCol="f", size=4).transform(df)
