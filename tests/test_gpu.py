@@ -1042,3 +1042,25 @@ class TestFusedDgradAct:
         p_gpu = m_gpu.transform(df_gpu)["prediction"].cpu()
         assert torch.equal(p_cpu, p_gpu)
         assert (p_cpu.numpy() == y).mean() == 1.0
+
+
+class TestProfilingGPU:
+    def test_phase_timer_cuda_events(self, tmp_path):
+        from flowml.utils.profiling import PhaseTimer
+
+        t = PhaseTimer()
+        assert t.use_cuda
+        x = torch.randn(1_000_000, device=DEV)
+        with t.phase("square"):
+            for _ in range(10):
+                x = x * x
+        torch.cuda.synchronize()
+        t.flush()
+        path = str(tmp_path / "trace.json")
+        t.dump_chrome_trace(path)
+        import json as _json
+
+        tr = _json.load(open(path))
+        evs = tr["traceEvents"] if isinstance(tr, dict) else tr
+        assert any(e.get("name") == "square" and e.get("dur", 0) > 0
+                   for e in evs)
