@@ -243,3 +243,23 @@ class TestMlpGrad:
         ref.backward()
         assert float(loss) == pytest.approx(float(ref.detach()), rel=1e-4)
         assert torch.allclose(gWs[0], W1.grad, atol=1e-4)
+
+
+def test_pack_tree_nodes_encoding():
+    """Packed K8 node layout: CAT_FLAG on categorical internals, float bits
+    preserved, leaves stay -1."""
+    from flowml import ops
+
+    feature = torch.tensor([[2, -1, 5]], dtype=torch.int32)
+    threshold = torch.tensor([[1.5, 0.0, -3.25]], dtype=torch.float32)
+    left = torch.tensor([[1, -1, 0]], dtype=torch.int32)
+    right = torch.tensor([[2, -1, 1]], dtype=torch.int32)
+    catmask = torch.tensor([[0, 0, 0b101]], dtype=torch.int64)
+    p = ops.pack_tree_nodes(feature, threshold, left, right, catmask)
+    assert p.shape == (1, 3, 4) and p.dtype == torch.int32
+    assert int(p[0, 0, 0]) == 2                       # continuous: bare id
+    assert int(p[0, 1, 0]) == -1                      # leaf untouched
+    assert int(p[0, 2, 0]) == 5 | ops.TREE_CAT_FLAG   # categorical flagged
+    assert p[0, :, 1].view(torch.float32).tolist() == [1.5, 0.0, -3.25]
+    assert p[0, :, 2].tolist() == [1, -1, 0]
+    assert p[0, :, 3].tolist() == [2, -1, 1]
