@@ -108,7 +108,7 @@ def main():
     if rank == 0:
         top = int(np.argmax(res["class_counts"]))
         print(f"served {res['rows']} flows in {res['elapsed_s']:.2f}s = "
-              f"{res['flows_per_s']/1e6:.1f}M flows/s "
+              f"{res['flows_per_s']/1e6:.3f}M flows/s "
               f"(world={world}, hipGraph={sc._graph is not None})")
         print(f"class counts: {[int(c) for c in res['class_counts']]} "
               f"(modal class index {top})")
