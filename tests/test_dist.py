@@ -673,3 +673,23 @@ def test_bench_under_torchrun_world2_cpu():
     assert out["n_gpus"] == 2
     assert out["config"]["parallelism"] == "dp2"
     assert out["config"]["global_rows"] == 6000
+
+
+def test_bench_f1_curve_contract():
+    """`bench.py --report-f1-curve` prints one JSON line per depth."""
+    import json
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--report-f1-curve", "--trees", "5",
+         "--rows", "3000"],
+        cwd=repo, capture_output=True, text=True, timeout=420,
+    )
+    assert r.returncode == 0, r.stderr[-500:]
+    lines = [json.loads(l) for l in r.stdout.splitlines() if l.startswith("{")]
+    depths = [l["maxDepth"] for l in lines if l.get("curve") == "rf_f1_vs_depth"]
+    assert depths == [5, 8, 10]
+    for l in lines:
+        assert l["fit_ms"] > 0 and 0 <= l["macro_f1"] <= 1
